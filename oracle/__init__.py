@@ -1,0 +1,245 @@
+"""oracle — TEST INFRASTRUCTURE ONLY.
+
+CPU parity oracle for the examl_amd HIP hot path.  May be imported only by
+tests/, __graft_entry__.smoke() and bench.py's cpu_baseline leg — never by
+product code.  See oracle/oracle.c header.
+
+Two libraries:
+  * liboracle.so  — our own C restatement of the reference hot path.
+  * _ref/libref.so — the reference's kernels compiled in place from
+    /root/reference (golden-vector source; only used where present).
+"""
+
+import ctypes
+import os
+
+import numpy as np
+
+_HERE = os.path.dirname(os.path.abspath(__file__))
+
+TIP_TIP, TIP_INNER, INNER_INNER = 0, 1, 2
+TWOTOTHE256 = 2.0 ** 256
+MINLIKELIHOOD = 2.0 ** -256
+ZMIN, ZMAX = 1.0e-15, 1.0 - 1.0e-6
+
+_c_d = ctypes.POINTER(ctypes.c_double)
+_c_i = ctypes.POINTER(ctypes.c_int)
+_c_u8 = ctypes.POINTER(ctypes.c_ubyte)
+_c_u32 = ctypes.POINTER(ctypes.c_uint)
+
+
+def aligned(shape, dtype=np.float64, alignment=64):
+    """numpy array aligned for the reference's AVX loads (BYTE_ALIGNMENT)."""
+    dtype = np.dtype(dtype)
+    n = int(np.prod(shape))
+    buf = np.zeros(n * dtype.itemsize + alignment, dtype=np.uint8)
+    off = (-buf.ctypes.data) % alignment
+    return buf[off:off + n * dtype.itemsize].view(dtype).reshape(shape)
+
+
+def _dp(a):
+    assert a.dtype == np.float64 and a.flags.c_contiguous
+    return a.ctypes.data_as(_c_d)
+
+
+def _ip(a):
+    assert a.dtype == np.int32 and a.flags.c_contiguous
+    return a.ctypes.data_as(_c_i)
+
+
+def _u8p(a):
+    assert a.dtype == np.uint8 and a.flags.c_contiguous
+    return a.ctypes.data_as(_c_u8)
+
+
+def _load(path):
+    return ctypes.CDLL(path)
+
+
+_orc = _load(os.path.join(_HERE, "liboracle.so"))
+
+_REF_PATH = os.path.join(_HERE, "_ref", "libref.so")
+_ref = _load(_REF_PATH) if os.path.exists(_REF_PATH) else None
+
+
+def have_ref():
+    return _ref is not None
+
+
+# ---------------------------------------------------------------------------
+# Oracle (our restatement)
+# ---------------------------------------------------------------------------
+
+def make_p(z1, z2, rates, EI, EIGN, num_cats, states):
+    """left/right P-matrix pair; z1/z2 are log-branch-lengths (pre-clamped)."""
+    sq = states * states
+    left = aligned(num_cats * sq)
+    right = aligned(num_cats * sq)
+    _orc.oracle_make_p(
+        ctypes.c_double(z1), ctypes.c_double(z2), _dp(rates), _dp(EI),
+        _dp(EIGN), ctypes.c_int(num_cats), _dp(left), _dp(right),
+        ctypes.c_int(states))
+    return left, right
+
+
+def calc_diagptable(z, states, num_cats, rates, EIGN):
+    diag = aligned(num_cats * states)
+    _orc.oracle_calc_diagptable(
+        ctypes.c_double(z), ctypes.c_int(states), ctypes.c_int(num_cats),
+        _dp(rates), _dp(EIGN), _dp(diag))
+    return diag
+
+
+def newview_dna_gamma(tip_case, x1, x2, extEV, tipVector, tipX1, tipX2, n,
+                      left, right, wgt, lib=None):
+    lib = lib or _orc
+    fn = (lib.oracle_newview_dna_gamma if lib is _orc
+          else lib.newviewGTRGAMMA_AVX)
+    x3 = aligned(n * 16)
+    inc = ctypes.c_int(0)
+    nullp = ctypes.cast(None, _c_d)
+    nullb = ctypes.cast(None, _c_u8)
+    fn(ctypes.c_int(tip_case),
+       _dp(x1) if x1 is not None else nullp,
+       _dp(x2) if x2 is not None else nullp,
+       _dp(x3), _dp(extEV), _dp(tipVector),
+       _u8p(tipX1) if tipX1 is not None else nullb,
+       _u8p(tipX2) if tipX2 is not None else nullb,
+       ctypes.c_int(n), _dp(left), _dp(right), _ip(wgt),
+       ctypes.byref(inc))
+    return x3, inc.value
+
+
+def evaluate_dna_gamma(wgt, x1, x2, tipVector, tipX1, n, diag, lib=None):
+    lib = lib or _orc
+    fn = (lib.oracle_evaluate_dna_gamma if lib is _orc
+          else lib.evaluateGTRGAMMA)
+    fn.restype = ctypes.c_double
+    nullp = ctypes.cast(None, _c_d)
+    nullb = ctypes.cast(None, _c_u8)
+    return fn(_ip(wgt),
+              _dp(x1) if x1 is not None else nullp,
+              _dp(x2), _dp(tipVector),
+              _u8p(tipX1) if tipX1 is not None else nullb,
+              ctypes.c_int(n), _dp(diag))
+
+
+def sum_dna_gamma(tip_case, x1, x2, tipVector, tipX1, tipX2, n, lib=None):
+    lib = lib or _orc
+    fn = lib.oracle_sum_dna_gamma if lib is _orc else lib.sumGAMMA
+    sumtable = aligned(n * 16)
+    nullp = ctypes.cast(None, _c_d)
+    nullb = ctypes.cast(None, _c_u8)
+    if lib is _orc:
+        fn(ctypes.c_int(tip_case), _dp(sumtable),
+           _dp(x1) if x1 is not None else nullp,
+           _dp(x2) if x2 is not None else nullp,
+           _dp(tipVector),
+           _u8p(tipX1) if tipX1 is not None else nullb,
+           _u8p(tipX2) if tipX2 is not None else nullb, ctypes.c_int(n))
+    else:
+        # reference sumGAMMA(tipCase, sumtable, x1, x2, tipVector, tipX1, tipX2, n)
+        fn(ctypes.c_int(tip_case), _dp(sumtable),
+           _dp(x1) if x1 is not None else nullp,
+           _dp(x2) if x2 is not None else nullp,
+           _dp(tipVector),
+           _u8p(tipX1) if tipX1 is not None else nullb,
+           _u8p(tipX2) if tipX2 is not None else nullb, ctypes.c_int(n))
+    return sumtable
+
+
+def core_dna_gamma(n, sumtable, EIGN, gammaRates, lz, wgt, lib=None):
+    lib = lib or _orc
+    d1 = ctypes.c_double(0.0)
+    d2 = ctypes.c_double(0.0)
+    if lib is _orc:
+        _orc.oracle_core_dna_gamma(
+            ctypes.c_int(n), _dp(sumtable), ctypes.byref(d1),
+            ctypes.byref(d2), _dp(EIGN), _dp(gammaRates),
+            ctypes.c_double(lz), _ip(wgt))
+    else:
+        # reference coreGTRGAMMA(upper, sumtable, ext_dlnLdlz, ext_d2lnLdlz2,
+        #                        EIGN, gammaRates, lz, wgt)
+        lib.coreGTRGAMMA(
+            ctypes.c_int(n), _dp(sumtable), ctypes.byref(d1),
+            ctypes.byref(d2), _dp(EIGN), _dp(gammaRates),
+            ctypes.c_double(lz), _ip(wgt))
+    return d1.value, d2.value
+
+
+def make_gamma_cats(alpha, k=4):
+    rates = aligned(k)
+    _orc.oracle_make_gamma_cats(ctypes.c_double(alpha), _dp(rates),
+                                ctypes.c_int(k))
+    return rates
+
+
+def init_gtr_dna(frequencies, rates):
+    """EIGN/EV/EI/tipVector for DNA GTR (states=4, 16 ambiguity codes)."""
+    n = 4
+    value_vector = np.arange(16, dtype=np.uint32)  # bitVectorIdentity[0..15]
+    EIGN = aligned(n)
+    EV = aligned(n * n)
+    EI = aligned(n * n)
+    tipVector = aligned(16 * n)
+    _orc.oracle_init_gtr(
+        ctypes.c_int(n), value_vector.ctypes.data_as(_c_u32),
+        ctypes.c_int(16), _dp(EIGN), _dp(EV), _dp(EI),
+        _dp(np.ascontiguousarray(frequencies, dtype=np.float64)),
+        _dp(np.ascontiguousarray(rates, dtype=np.float64)), _dp(tipVector))
+    return EIGN, EV, EI, tipVector
+
+
+# ---------------------------------------------------------------------------
+# Reference (_ref) direct-call wrappers for golden generation/validation
+# ---------------------------------------------------------------------------
+
+def ref_make_p(z1, z2, rates, EI, EIGN, num_cats, states):
+    assert _ref is not None
+    sq = states * states
+    left = aligned(num_cats * sq)
+    right = aligned(num_cats * sq)
+    # makeP(z1, z2, rptr, EI, EIGN, numberOfCategories, left, right,
+    #       saveMem, maxCat, states) — newviewGenericSpecial.c:78
+    _ref.makeP(ctypes.c_double(z1), ctypes.c_double(z2), _dp(rates), _dp(EI),
+               _dp(EIGN), ctypes.c_int(num_cats), _dp(left), _dp(right),
+               ctypes.c_int(0), ctypes.c_int(num_cats), ctypes.c_int(states))
+    return left, right
+
+
+def ref_calc_diagptable(z, states, num_cats, rates, EIGN):
+    assert _ref is not None
+    diag = aligned(num_cats * states)
+    _ref.calcDiagptable(ctypes.c_double(z), ctypes.c_int(states),
+                        ctypes.c_int(num_cats), _dp(rates), _dp(EIGN),
+                        _dp(diag))
+    return diag
+
+
+def ref_init_gtr_dna(frequencies, rates):
+    assert _ref is not None
+    n = 4
+    value_vector = np.arange(16, dtype=np.uint32)
+    EIGN = aligned(n)
+    EV = aligned(n * n)
+    EI = aligned(n * n)
+    tipVector = aligned(16 * n)
+    # initGeneric(n, valueVector, valueVectorLength, ext_EIGN, EV, EI,
+    #             frequencies, ext_initialRates, tipVector, model) —
+    # models.c:3234, exposed via -Dstatic=
+    _ref.initGeneric(
+        ctypes.c_int(n), value_vector.ctypes.data_as(_c_u32),
+        ctypes.c_int(16), _dp(EIGN), _dp(EV), _dp(EI),
+        _dp(np.ascontiguousarray(frequencies, dtype=np.float64)),
+        _dp(np.ascontiguousarray(rates, dtype=np.float64)), _dp(tipVector),
+        ctypes.c_int(0))
+    return EIGN, EV, EI, tipVector
+
+
+def ref_make_gamma_cats(alpha, k=4):
+    assert _ref is not None
+    rates = aligned(k)
+    # makeGammaCats(alpha, gammaRates, K, useMedian) — models.c:3795
+    _ref.makeGammaCats(ctypes.c_double(alpha), _dp(rates), ctypes.c_int(k),
+                       ctypes.c_int(0))
+    return rates

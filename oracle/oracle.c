@@ -1,0 +1,718 @@
+/* ============================================================================
+ * oracle/oracle.c — TEST INFRASTRUCTURE ONLY.
+ *
+ * CPU restatement of ExaML's per-site conditional-likelihood hot path
+ * (the Felsenstein pruning core), used exclusively as the parity oracle for
+ * the MI355X HIP kernels in examl_amd/csrc/.  Nothing in the product path
+ * may import, link or call this library; only tests/, __graft_entry__.smoke()
+ * and bench.py's cpu_baseline leg use it (and there only as the checker).
+ *
+ * Parity is PINNED: every function cites the reference implementation it
+ * restates (file:line into /root/reference/), and tests/golden/ holds vectors
+ * generated from the reference's own compiled kernels (oracle/_ref, see
+ * Makefile) that this restatement must reproduce bit-for-bit.
+ *
+ * All arithmetic is fp64 and the summation ORDER deliberately mirrors the
+ * reference's AVX/SSE3 pairwise (hadd) order so results are bit-identical,
+ * not merely close.
+ * ==========================================================================*/
+
+#include <math.h>
+#include <stdlib.h>
+#include <string.h>
+#include <assert.h>
+
+/* Constants — reference: examl/axml.h:88,94,110-117 */
+#define ORC_TWOTOTHE256 \
+  115792089237316195423570985008687907853269984665640564039457584007913129639936.0
+#define ORC_MINLIKELIHOOD (1.0 / ORC_TWOTOTHE256)
+#define ORC_ZMIN 1.0E-15
+#define ORC_ZMAX (1.0 - 1.0E-6)
+#define ORC_MAX_TIP_EV 0.999999999
+
+/* tipCase values — reference: examl/axml.h:302-304 */
+#define ORC_TIP_TIP 0
+#define ORC_TIP_INNER 1
+#define ORC_INNER_INNER 2
+
+#define EXPORT __attribute__((visibility("default")))
+
+/* --------------------------------------------------------------------------
+ * makeP — P(z) = exp(rate_c * EIGN_k * log z) * EI, column 0 == 1.
+ * Restates examl/newviewGenericSpecial.c:78 (makeP), saveMem path omitted
+ * (out of scope).  z1/z2 here are ALREADY log-transformed branch
+ * lengths (the caller applies the zmin clamp + log, as
+ * newviewGenericSpecial.c:982-983 does).
+ * ------------------------------------------------------------------------*/
+EXPORT void oracle_make_p(double z1, double z2, const double *rptr,
+                          const double *EI, const double *EIGN,
+                          int numberOfCategories, double *left, double *right,
+                          int states) {
+  int i, j, k;
+  int statesSquare = states * states;
+  double d1[64], d2[64];
+  assert(states <= 64);
+  for (i = 0; i < numberOfCategories; i++) {
+    for (j = 1; j < states; j++) {
+      d1[j] = exp(rptr[i] * (EIGN[j] * z1));
+      d2[j] = exp(rptr[i] * (EIGN[j] * z2));
+    }
+    for (j = 0; j < states; j++) {
+      left[statesSquare * i + states * j] = 1.0;
+      right[statesSquare * i + states * j] = 1.0;
+      for (k = 1; k < states; k++) {
+        left[statesSquare * i + states * j + k] = d1[k] * EI[states * j + k];
+        right[statesSquare * i + states * j + k] = d2[k] * EI[states * j + k];
+      }
+    }
+  }
+}
+
+/* --------------------------------------------------------------------------
+ * calcDiagptable — diag[c*states+l] = exp(rate_c * EIGN_l * log z), col 0 = 1.
+ * Restates examl/evaluateGenericSpecial.c:80.  Takes the RAW branch length z
+ * (clamp + log happen here, as in the reference).
+ * ------------------------------------------------------------------------*/
+EXPORT void oracle_calc_diagptable(double z, int states,
+                                   int numberOfCategories, const double *rptr,
+                                   const double *EIGN, double *diagptable) {
+  int i, l;
+  double lz = (z < ORC_ZMIN) ? log(ORC_ZMIN) : log(z);
+  for (i = 0; i < numberOfCategories; i++) {
+    diagptable[i * states] = 1.0;
+    for (l = 1; l < states; l++)
+      diagptable[i * states + l] = exp(rptr[i] * (EIGN[l] * lz));
+  }
+}
+
+/* pairwise 4-sum matching _mm256_hadd_pd+permute (avxLikelihood.c:33-61):
+ * (a0+a1) + (a2+a3) */
+static inline double hadd4d(const double *a) {
+  return (a[0] + a[1]) + (a[2] + a[3]);
+}
+
+/* --------------------------------------------------------------------------
+ * newview, DNA GTRGAMMA (states=4, 4 gamma cats, span=16).
+ * Restates examl/avxLikelihood.c:64 (newviewGTRGAMMA_AVX) including its
+ * exact scaling rule: a site is rescaled by 2^256 iff ALL 16 span entries
+ * have |x| < 2^-256 (checked on the unscaled values, avxLikelihood.c:223-242),
+ * and the TIP_TIP case performs NO scaling check (avxLikelihood.c:85-157).
+ * Layout: x[site*16 + cat*4 + state]; left/right[cat*16 + row*4 + col];
+ * extEV[row*4 + state] (row-major eigenvector matrix, models.c:3372-3376);
+ * tipVector[code*4 + state], codes 1..15.
+ * ------------------------------------------------------------------------*/
+EXPORT void oracle_newview_dna_gamma(int tipCase, const double *x1,
+                                     const double *x2, double *x3,
+                                     const double *extEV,
+                                     const double *tipVector,
+                                     const unsigned char *tipX1,
+                                     const unsigned char *tipX2, int n,
+                                     const double *left, const double *right,
+                                     const int *wgt, int *scalerIncrement) {
+  int i, k, l, s;
+  int addScale = 0;
+  /* ump[code][cat*16 + row*4 + lane] in the AVX code stores the SAME scalar
+   * in all 4 lanes (hadd3 broadcast, avxLikelihood.c:89-124); we store one. */
+  double umpX1[16 * 16], umpX2[16 * 16];
+
+  switch (tipCase) {
+  case ORC_TIP_TIP: {
+    for (i = 1; i < 16; i++) {
+      const double *tv = &tipVector[i * 4];
+      for (k = 0; k < 4; k++) /* cat (j in ref) */
+        for (l = 0; l < 4; l++) { /* row (k in ref) */
+          double p[4];
+          for (s = 0; s < 4; s++) p[s] = left[k * 16 + l * 4 + s] * tv[s];
+          umpX1[i * 16 + k * 4 + l] = hadd4d(p);
+          for (s = 0; s < 4; s++) p[s] = right[k * 16 + l * 4 + s] * tv[s];
+          umpX2[i * 16 + k * 4 + l] = hadd4d(p);
+        }
+    }
+    for (i = 0; i < n; i++) {
+      const double *uX1 = &umpX1[16 * tipX1[i]];
+      const double *uX2 = &umpX2[16 * tipX2[i]];
+      for (k = 0; k < 4; k++) {
+        double xv[4] = {0, 0, 0, 0};
+        for (l = 0; l < 4; l++) {
+          double t = uX1[k * 4 + l] * uX2[k * 4 + l];
+          for (s = 0; s < 4; s++) xv[s] += t * extEV[l * 4 + s];
+        }
+        for (s = 0; s < 4; s++) x3[16 * i + 4 * k + s] = xv[s];
+      }
+      /* NO scaling in TIP_TIP (matches avxLikelihood.c:85-157) */
+    }
+  } break;
+  case ORC_TIP_INNER: {
+    for (i = 1; i < 16; i++) {
+      const double *tv = &tipVector[i * 4];
+      for (k = 0; k < 4; k++)
+        for (l = 0; l < 4; l++) {
+          double p[4];
+          for (s = 0; s < 4; s++) p[s] = left[k * 16 + l * 4 + s] * tv[s];
+          umpX1[i * 16 + k * 4 + l] = hadd4d(p);
+        }
+    }
+    for (i = 0; i < n; i++) {
+      const double *uX1 = &umpX1[16 * tipX1[i]];
+      double xv[16];
+      int scale = 1;
+      for (k = 0; k < 4; k++) {
+        const double *xvr = &x2[i * 16 + k * 4];
+        double acc[4] = {0, 0, 0, 0};
+        for (l = 0; l < 4; l++) {
+          double p[4];
+          for (s = 0; s < 4; s++) p[s] = xvr[s] * right[k * 16 + l * 4 + s];
+          double t = uX1[k * 4 + l] * hadd4d(p);
+          for (s = 0; s < 4; s++) acc[s] += t * extEV[l * 4 + s];
+        }
+        for (s = 0; s < 4; s++) xv[k * 4 + s] = acc[s];
+        if (scale) {
+          for (s = 0; s < 4; s++)
+            if (!(fabs(acc[s]) < ORC_MINLIKELIHOOD)) { scale = 0; break; }
+        }
+      }
+      if (scale) {
+        for (s = 0; s < 16; s++) xv[s] *= ORC_TWOTOTHE256;
+        addScale += wgt[i];
+      }
+      for (s = 0; s < 16; s++) x3[16 * i + s] = xv[s];
+    }
+  } break;
+  case ORC_INNER_INNER: {
+    for (i = 0; i < n; i++) {
+      double xv[16];
+      int scale = 1;
+      for (k = 0; k < 4; k++) {
+        const double *xvl = &x1[i * 16 + k * 4];
+        const double *xvr = &x2[i * 16 + k * 4];
+        double acc[4] = {0, 0, 0, 0};
+        for (l = 0; l < 4; l++) {
+          double pl[4], pr[4];
+          for (s = 0; s < 4; s++) {
+            pl[s] = xvl[s] * left[k * 16 + l * 4 + s];
+            pr[s] = xvr[s] * right[k * 16 + l * 4 + s];
+          }
+          /* hadd4 (avxLikelihood.c:33) multiplies the two pairwise sums */
+          double t = hadd4d(pl) * hadd4d(pr);
+          for (s = 0; s < 4; s++) acc[s] += t * extEV[l * 4 + s];
+        }
+        for (s = 0; s < 4; s++) xv[k * 4 + s] = acc[s];
+        if (scale) {
+          for (s = 0; s < 4; s++)
+            if (!(fabs(acc[s]) < ORC_MINLIKELIHOOD)) { scale = 0; break; }
+        }
+      }
+      if (scale) {
+        for (s = 0; s < 16; s++) xv[s] *= ORC_TWOTOTHE256;
+        addScale += wgt[i];
+      }
+      for (s = 0; s < 16; s++) x3[16 * i + s] = xv[s];
+    }
+  } break;
+  default:
+    assert(0);
+  }
+  *scalerIncrement = addScale;
+}
+
+/* --------------------------------------------------------------------------
+ * evaluate, DNA GTRGAMMA.  Restates examl/evaluateGenericSpecial.c:1879
+ * (evaluateGTRGAMMA, SSE3): term_i = sum_{c,k} x1*x2*diag, accumulated in
+ * the SSE even/odd lane split; lnL += wgt_i * log(0.25*|term_i|).
+ * tipX1 == NULL means INNER_INNER at the root branch.
+ * Returns the partition log likelihood WITHOUT the scaler-undo term (the
+ * caller adds (gs_p+gs_q)*log(minlikelihood), evaluateGenericSpecial.c:830).
+ * ------------------------------------------------------------------------*/
+EXPORT double oracle_evaluate_dna_gamma(const int *wptr, const double *x1_start,
+                                        const double *x2_start,
+                                        const double *tipVector,
+                                        const unsigned char *tipX1, int n,
+                                        const double *diagptable) {
+  double sum = 0.0;
+  int i, j;
+  if (tipX1) {
+    for (i = 0; i < n; i++) {
+      const double *x1 = &tipVector[4 * tipX1[i]];
+      const double *x2 = &x2_start[16 * i];
+      double t0 = 0.0, t1 = 0.0; /* SSE lane 0 / lane 1 accumulators */
+      for (j = 0; j < 4; j++) {
+        t0 += x1[0] * x2[j * 4 + 0] * diagptable[j * 4 + 0];
+        t1 += x1[1] * x2[j * 4 + 1] * diagptable[j * 4 + 1];
+        t0 += x1[2] * x2[j * 4 + 2] * diagptable[j * 4 + 2];
+        t1 += x1[3] * x2[j * 4 + 3] * diagptable[j * 4 + 3];
+      }
+      sum += wptr[i] * log(0.25 * fabs(t0 + t1));
+    }
+  } else {
+    for (i = 0; i < n; i++) {
+      const double *x1 = &x1_start[16 * i];
+      const double *x2 = &x2_start[16 * i];
+      double t0 = 0.0, t1 = 0.0;
+      for (j = 0; j < 4; j++) {
+        t0 += x1[j * 4 + 0] * x2[j * 4 + 0] * diagptable[j * 4 + 0];
+        t1 += x1[j * 4 + 1] * x2[j * 4 + 1] * diagptable[j * 4 + 1];
+        t0 += x1[j * 4 + 2] * x2[j * 4 + 2] * diagptable[j * 4 + 2];
+        t1 += x1[j * 4 + 3] * x2[j * 4 + 3] * diagptable[j * 4 + 3];
+      }
+      sum += wptr[i] * log(0.25 * fabs(t0 + t1));
+    }
+  }
+  return sum;
+}
+
+/* --------------------------------------------------------------------------
+ * sumGAMMA — sum[i,c,k] = x1'[i,c,k] * x2'[i,c,k] with tip expansion.
+ * Restates examl/makenewzGenericSpecial.c:1798 (sumGAMMA, SSE3).
+ * ------------------------------------------------------------------------*/
+EXPORT void oracle_sum_dna_gamma(int tipCase, double *sumtable,
+                                 const double *x1_start, const double *x2_start,
+                                 const double *tipVector,
+                                 const unsigned char *tipX1,
+                                 const unsigned char *tipX2, int n) {
+  int i, j, k;
+  switch (tipCase) {
+  case ORC_TIP_TIP:
+    for (i = 0; i < n; i++) {
+      const double *x1 = &tipVector[4 * tipX1[i]];
+      const double *x2 = &tipVector[4 * tipX2[i]];
+      for (j = 0; j < 4; j++)
+        for (k = 0; k < 4; k++)
+          sumtable[i * 16 + j * 4 + k] = x1[k] * x2[k];
+    }
+    break;
+  case ORC_TIP_INNER:
+    for (i = 0; i < n; i++) {
+      const double *x1 = &tipVector[4 * tipX1[i]];
+      const double *x2 = &x2_start[16 * i];
+      for (j = 0; j < 4; j++)
+        for (k = 0; k < 4; k++)
+          sumtable[i * 16 + j * 4 + k] = x1[k] * x2[j * 4 + k];
+    }
+    break;
+  case ORC_INNER_INNER:
+    for (i = 0; i < n; i++) {
+      const double *x1 = &x1_start[16 * i];
+      const double *x2 = &x2_start[16 * i];
+      for (j = 0; j < 4; j++)
+        for (k = 0; k < 4; k++)
+          sumtable[i * 16 + j * 4 + k] = x1[j * 4 + k] * x2[j * 4 + k];
+    }
+    break;
+  default:
+    assert(0);
+  }
+}
+
+/* --------------------------------------------------------------------------
+ * coreGTRGAMMA — per-site 1st/2nd log-likelihood derivatives wrt the
+ * log branch length lz.  Restates examl/makenewzGenericSpecial.c:2309
+ * (SSE3 even/odd lane accumulation preserved).
+ * ------------------------------------------------------------------------*/
+EXPORT void oracle_core_dna_gamma(int upper, const double *sumtable,
+                                  double *ext_dlnLdlz, double *ext_d2lnLdlz2,
+                                  const double *EIGN, const double *gammaRates,
+                                  double lz, const int *wgt) {
+  double dlnLdlz = 0.0, d2lnLdlz2 = 0.0;
+  double d0[16], d1[16], d2[16];
+  int i, j, l;
+  for (i = 0; i < 4; i++) {
+    double ki = gammaRates[i], kisqr = ki * ki;
+    d0[i * 4] = 1.0;
+    d1[i * 4] = 0.0;
+    d2[i * 4] = 0.0;
+    for (l = 1; l < 4; l++) {
+      d0[i * 4 + l] = exp(EIGN[l] * ki * lz);
+      d1[i * 4 + l] = EIGN[l] * ki;
+      d2[i * 4 + l] = EIGN[l] * EIGN[l] * kisqr;
+    }
+  }
+  for (i = 0; i < upper; i++) {
+    const double *sum = &sumtable[i * 16];
+    double a0e = 0, a0o = 0, a1e = 0, a1o = 0, a2e = 0, a2o = 0;
+    for (j = 0; j < 4; j++) {
+      for (l = 0; l < 4; l += 2) {
+        double te = d0[j * 4 + l] * sum[j * 4 + l];
+        double to = d0[j * 4 + l + 1] * sum[j * 4 + l + 1];
+        a0e += te;            a0o += to;
+        a1e += te * d1[j * 4 + l];     a1o += to * d1[j * 4 + l + 1];
+        a2e += te * d2[j * 4 + l];     a2o += to * d2[j * 4 + l + 1];
+      }
+    }
+    double inv_Li = 1.0 / fabs(a0e + a0o);
+    double dlnLidlz = (a1e + a1o) * inv_Li;
+    double d2lnLidlz2 = (a2e + a2o) * inv_Li;
+    dlnLdlz += wgt[i] * dlnLidlz;
+    d2lnLdlz2 += wgt[i] * (d2lnLidlz2 - dlnLidlz * dlnLidlz);
+  }
+  *ext_dlnLdlz = dlnLdlz;
+  *ext_d2lnLdlz2 = d2lnLdlz2;
+}
+
+/* ==========================================================================
+ * Model preparation (host math, runs once per model-parameter change).
+ * ==========================================================================*/
+
+/* LnGamma — Pike & Hill (1966) Algorithm 291, as used at models.c:3589 */
+static double orc_LnGamma(double alpha) {
+  double x = alpha, f = 0.0, z, result;
+  if (x < 7.0) {
+    f = 1.0;
+    z = alpha - 1.0;
+    while ((z = z + 1.0) < 7.0) f *= z;
+    x = z;
+    f = -log(f);
+  }
+  z = 1 / (x * x);
+  result = f + (x - 0.5) * log(x) - x + .918938533204673 +
+           (((-.000595238095238 * z + .000793650793651) * z -
+             .002777777777778) * z + .083333333333333) / x;
+  return result;
+}
+
+/* IncompleteGamma — Bhattacharjee (1970) AS32, as used at models.c:3627 */
+static double orc_IncompleteGamma(double x, double alpha,
+                                  double ln_gamma_alpha) {
+  int i;
+  double p = alpha, g = ln_gamma_alpha;
+  double accurate = 1e-8, overflow = 1e30;
+  double factor, gin = 0, rn = 0, a = 0, b = 0, an = 0, dif = 0, term = 0,
+         pn[6];
+  if (x == 0) return 0;
+  if (x < 0 || p <= 0) return -1;
+  factor = exp(p * log(x) - x - g);
+  if (!(x > 1 && x >= p)) {
+    /* series expansion */
+    gin = 1; term = 1; rn = p;
+    do { rn++; term *= x / rn; gin += term; } while (term > accurate);
+    gin *= factor / p;
+    return gin;
+  }
+  /* continued fraction */
+  a = 1 - p; b = a + x + 1; term = 0;
+  pn[0] = 1; pn[1] = x; pn[2] = x + 1; pn[3] = x * b;
+  gin = pn[2] / pn[3];
+  for (;;) {
+    a++; b += 2; term++;
+    an = a * term;
+    for (i = 0; i < 2; i++) pn[i + 4] = b * pn[i + 2] - an * pn[i];
+    if (pn[5] != 0) {
+      rn = pn[4] / pn[5];
+      dif = fabs(gin - rn);
+      if (dif <= accurate && dif <= accurate * rn) { gin = rn; break; }
+      gin = rn;
+    }
+    for (i = 0; i < 4; i++) pn[i] = pn[i + 2];
+    if (fabs(pn[4]) >= overflow)
+      for (i = 0; i < 4; i++) pn[i] /= overflow;
+  }
+  return 1 - factor * gin;
+}
+
+/* PointNormal — Odeh & Evans (1974) AS70, as used at models.c:3700 */
+static double orc_PointNormal(double prob) {
+  double a0 = -.322232431088, a1 = -1, a2 = -.342242088547,
+         a3 = -.0204231210245, a4 = -.453642210148e-4, b0 = .0993484626060,
+         b1 = .588581570495, b2 = .531103462366, b3 = .103537752850,
+         b4 = .0038560700634;
+  double y, z = 0, p = prob, p1;
+  p1 = (p < 0.5 ? p : 1 - p);
+  if (p1 < 1e-20) return -9999;
+  y = sqrt(log(1 / (p1 * p1)));
+  z = y + ((((y * a4 + a3) * y + a2) * y + a1) * y + a0) /
+          ((((y * b4 + b3) * y + b2) * y + b1) * y + b0);
+  return (p < 0.5 ? -z : z);
+}
+
+/* PointChi2 — Best & Roberts (1975) AS91, as used at models.c:3725 */
+static double orc_PointChi2(double prob, double v) {
+  double e = .5e-6, aa = .6931471805, p = prob, g;
+  double xx, c, ch, a = 0, q = 0, p1 = 0, p2 = 0, t = 0, x = 0, b = 0, s1, s2,
+             s3, s4, s5, s6;
+  if (p < .000002 || p > .999998 || v <= 0) return -1;
+  g = orc_LnGamma(v / 2);
+  xx = v / 2;
+  c = xx - 1;
+  if (v < -1.24 * log(p)) {
+    ch = pow((p * xx * exp(g + xx * aa)), 1 / xx);
+    if (ch - e < 0) return ch;
+  } else if (v <= .32) {
+    ch = 0.4;
+    a = log(1 - p);
+    for (;;) {
+      q = ch;
+      p1 = 1 + ch * (4.67 + ch);
+      p2 = ch * (6.73 + ch * (6.66 + ch));
+      t = -0.5 + (4.67 + 2 * ch) / p1 - (6.73 + ch * (13.32 + 3 * ch)) / p2;
+      ch -= (1 - exp(a + g + .5 * ch + c * aa) * p2 / p1) / t;
+      if (fabs(q / ch - 1) - .01 <= 0) break;
+    }
+  } else {
+    x = orc_PointNormal(p);
+    p1 = 0.222222 / v;
+    ch = v * pow((x * sqrt(p1) + 1 - p1), 3.0);
+    if (ch > 2.2 * v + 6) ch = -2 * (log(1 - p) - c * log(.5 * ch) + g);
+  }
+  do {
+    q = ch;
+    p1 = .5 * ch;
+    if ((t = orc_IncompleteGamma(p1, xx, g)) < 0.0) return -1;
+    p2 = p - t;
+    t = p2 * exp(xx * aa + g + p1 - c * log(ch));
+    b = t / ch;
+    a = 0.5 * t - b * c;
+    s1 = (210 + a * (140 + a * (105 + a * (84 + a * (70 + 60 * a))))) / 420;
+    s2 = (420 + a * (735 + a * (966 + a * (1141 + 1278 * a)))) / 2520;
+    s3 = (210 + a * (462 + a * (707 + 932 * a))) / 2520;
+    s4 = (252 + a * (672 + 1182 * a) + c * (294 + a * (889 + 1740 * a))) / 5040;
+    s5 = (84 + 264 * a + c * (175 + 606 * a)) / 2520;
+    s6 = (120 + c * (346 + 127 * c)) / 5040;
+    ch += t * (1 + 0.5 * t * s1 -
+               b * c * (s1 - b * (s2 - b * (s3 - b * (s4 - b * (s5 - b * s6))))));
+  } while (fabs(q / ch - 1) > e);
+  return ch;
+}
+
+/* makeGammaCats (mean rates, useMedian=FALSE default) — models.c:3795 */
+EXPORT void oracle_make_gamma_cats(double alpha, double *gammaRates, int K) {
+  double factor = alpha / alpha * K, lnga1, alfa = alpha, beta = alpha;
+  double gammaProbs[32];
+  int i;
+  lnga1 = orc_LnGamma(alfa + 1);
+  for (i = 0; i < K - 1; i++)
+    gammaProbs[i] = orc_PointChi2((i + 1.0) / K, 2.0 * alfa) / (2.0 * beta);
+  for (i = 0; i < K - 1; i++)
+    gammaProbs[i] = orc_IncompleteGamma(gammaProbs[i] * beta, alfa + 1, lnga1);
+  gammaRates[0] = gammaProbs[0] * factor;
+  gammaRates[K - 1] = (1 - gammaProbs[K - 2]) * factor;
+  for (i = 1; i < K - 1; i++)
+    gammaRates[i] = (gammaProbs[i] - gammaProbs[i - 1]) * factor;
+}
+
+/* --------------------------------------------------------------------------
+ * Symmetric eigensolver: Householder tridiagonalization + QL.
+ * Index-for-index restatement of the reference's mytred2 / mytqli
+ * (models.c:3068 / models.c:3151) so that eigenvalue ORDER, eigenvector
+ * SIGNS and every intermediate rounding match bit-for-bit (the reference
+ * uses a column-major working convention and the classic pythag-free
+ * rotation form; a textbook tred2/tqli differs in both).  After the pair,
+ * the ROWS of `a` are the eigenvectors of the input (validated against
+ * oracle/_ref in tests/test_oracle_cpu.py).
+ * a: n*n row-major flat array (a[i*n+j] == the reference's a[i][j]).
+ * ------------------------------------------------------------------------*/
+static void orc_tred2(double *a, const int n, double *d, double *e) {
+  int l, k, j, i;
+  double scale, hh, h, g, f;
+  for (i = n; i > 1; i--) {
+    l = i - 1;
+    h = 0.0;
+    scale = 0.0;
+    if (l > 1) {
+      for (k = 1; k <= l; k++) scale += fabs(a[(k - 1) * n + (i - 1)]);
+      if (scale == 0.0)
+        e[i - 1] = a[(l - 1) * n + (i - 1)];
+      else {
+        for (k = 1; k <= l; k++) {
+          a[(k - 1) * n + (i - 1)] /= scale;
+          h += a[(k - 1) * n + (i - 1)] * a[(k - 1) * n + (i - 1)];
+        }
+        f = a[(l - 1) * n + (i - 1)];
+        g = ((f > 0) ? -sqrt(h) : sqrt(h));
+        e[i - 1] = scale * g;
+        h -= f * g;
+        a[(l - 1) * n + (i - 1)] = f - g;
+        f = 0.0;
+        for (j = 1; j <= l; j++) {
+          a[(i - 1) * n + (j - 1)] = a[(j - 1) * n + (i - 1)] / h;
+          g = 0.0;
+          for (k = 1; k <= j; k++)
+            g += a[(k - 1) * n + (j - 1)] * a[(k - 1) * n + (i - 1)];
+          for (k = j + 1; k <= l; k++)
+            g += a[(j - 1) * n + (k - 1)] * a[(k - 1) * n + (i - 1)];
+          e[j - 1] = g / h;
+          f += e[j - 1] * a[(j - 1) * n + (i - 1)];
+        }
+        hh = f / (h + h);
+        for (j = 1; j <= l; j++) {
+          f = a[(j - 1) * n + (i - 1)];
+          g = e[j - 1] - hh * f;
+          e[j - 1] = g;
+          for (k = 1; k <= j; k++)
+            a[(k - 1) * n + (j - 1)] -=
+                (f * e[k - 1] + g * a[(k - 1) * n + (i - 1)]);
+        }
+      }
+    } else
+      e[i - 1] = a[(l - 1) * n + (i - 1)];
+    d[i - 1] = h;
+  }
+  d[0] = 0.0;
+  e[0] = 0.0;
+  for (i = 1; i <= n; i++) {
+    l = i - 1;
+    if (d[i - 1] != 0.0) {
+      for (j = 1; j <= l; j++) {
+        g = 0.0;
+        for (k = 1; k <= l; k++)
+          g += a[(k - 1) * n + (i - 1)] * a[(j - 1) * n + (k - 1)];
+        for (k = 1; k <= l; k++)
+          a[(j - 1) * n + (k - 1)] -= g * a[(i - 1) * n + (k - 1)];
+      }
+    }
+    d[i - 1] = a[(i - 1) * n + (i - 1)];
+    a[(i - 1) * n + (i - 1)] = 1.0;
+    for (j = 1; j <= l; j++)
+      a[(i - 1) * n + (j - 1)] = a[(j - 1) * n + (i - 1)] = 0.0;
+  }
+}
+
+static void orc_tqli(double *d, double *e, const int n, double *z) {
+  int m, l, iter, i, k;
+  double s, r, p, g, f, dd, c, b;
+  for (i = 2; i <= n; i++) e[i - 2] = e[i - 1];
+  e[n - 1] = 0.0;
+  for (l = 1; l <= n; l++) {
+    iter = 0;
+    do {
+      for (m = l; m <= n - 1; m++) {
+        dd = fabs(d[m - 1]) + fabs(d[m]);
+        if (fabs(e[m - 1]) + dd == dd) break;
+      }
+      if (m != l) {
+        assert(iter < 30);
+        iter++;
+        g = (d[l] - d[l - 1]) / (2.0 * e[l - 1]);
+        r = sqrt((g * g) + 1.0);
+        g = d[m - 1] - d[l - 1] + e[l - 1] / (g + ((g < 0) ? -fabs(r) : fabs(r)));
+        s = c = 1.0;
+        p = 0.0;
+        for (i = m - 1; i >= l; i--) {
+          f = s * e[i - 1];
+          b = c * e[i - 1];
+          if (fabs(f) >= fabs(g)) {
+            c = g / f;
+            r = sqrt((c * c) + 1.0);
+            e[i] = f * r;
+            c *= (s = 1.0 / r);
+          } else {
+            s = f / g;
+            r = sqrt((s * s) + 1.0);
+            e[i] = g * r;
+            s *= (c = 1.0 / r);
+          }
+          g = d[i] - p;
+          r = (d[i - 1] - g) * s + 2.0 * c * b;
+          p = s * r;
+          d[i] = g + p;
+          g = c * r - b;
+          for (k = 1; k <= n; k++) {
+            f = z[i * n + (k - 1)];
+            z[i * n + (k - 1)] = s * z[(i - 1) * n + (k - 1)] + c * f;
+            z[(i - 1) * n + (k - 1)] = c * z[(i - 1) * n + (k - 1)] - s * f;
+          }
+        }
+        d[l - 1] = d[l - 1] - p;
+        e[l - 1] = g;
+        e[m - 1] = 0.0;
+      }
+    } while (m != l);
+  }
+}
+
+/* --------------------------------------------------------------------------
+ * GTR model initialization: symmetrized rate matrix -> eigendecomposition ->
+ * EIGN / EV / EI / tipVector.  Restates examl/models.c:3234 (initGeneric):
+ * a[i][j] = rate_ij*sqrt(f_i f_j), a[i][i] = -sum_j rate_ij f_j;
+ * fracchange normalization; the near-zero eigenvalue moved to index 0 with
+ * its eigenvector normalized to sum 1 (models.c:3336-3352); EIGN negated and
+ * scaled by 1/fracchange; EV[i*n+j] = EIGV[i][j]; EI row scaling by
+ * invfreq (models.c:3381-3388); tipVector = per-ambiguity-code sums of
+ * eigenvector rows clamped to MAX_TIP_EV (models.c:3410-3436).
+ *
+ * valueVector: length vlen ambiguity-code bitmasks (identity 0..15 for DNA,
+ * per getBitVector).  rates: upper-triangle initial rates (6 for DNA).
+ * ------------------------------------------------------------------------*/
+EXPORT void oracle_init_gtr(int n, const unsigned int *valueVector, int vlen,
+                            double *ext_EIGN, double *EV, double *EI,
+                            const double *frequencies, const double *rates,
+                            double *tipVector) {
+  double a[64 * 64], d[64], e[64], EIGV[64 * 64], invfreq[64], EIGN[64];
+  double fracchange = 0.0;
+  double r[64 * 64];
+  int i, j, k, m, l;
+  assert(n <= 64);
+
+  memset(r, 0, sizeof(double) * n * n);
+  i = 0;
+  for (j = 0; j < n - 1; j++)
+    for (k = j + 1; k < n; k++) r[j * n + k] = rates[i++];
+  for (j = 0; j < n; j++) {
+    r[j * n + j] = 0.0;
+    for (k = 0; k < j; k++) r[j * n + k] = r[k * n + j];
+  }
+  for (j = 0; j < n; j++)
+    for (k = 0; k < n; k++)
+      fracchange += frequencies[j] * r[j * n + k] * frequencies[k];
+
+  memset(a, 0, sizeof(double) * n * n);
+  m = 0;
+  for (i = 0; i < n; i++)
+    for (j = i + 1; j < n; j++) {
+      double factor = rates[m++];
+      a[i * n + j] = a[j * n + i] =
+          factor * sqrt(frequencies[i] * frequencies[j]);
+      a[i * n + i] -= factor * frequencies[j];
+      a[j * n + j] -= factor * frequencies[i];
+    }
+
+  orc_tred2(a, n, d, e);
+  orc_tqli(d, e, n, a);
+
+  /* columns of a are eigenvectors; postprocess exactly as models.c:3327+ */
+  for (i = 0; i < n; i++)
+    for (j = 0; j < n; j++) a[i * n + j] *= sqrt(frequencies[j]);
+
+  for (i = 0; i < n; i++) {
+    if (d[i] > -1e-8) {
+      if (i != 0) {
+        double tmp = d[i], sum = 0;
+        d[i] = d[0];
+        d[0] = tmp;
+        for (j = 0; j < n; j++) {
+          tmp = a[i * n + j];
+          a[i * n + j] = a[0 * n + j];
+          sum += (a[0 * n + j] = tmp);
+        }
+        for (j = 0; j < n; j++) a[0 * n + j] /= sum;
+      }
+      break;
+    }
+  }
+  for (i = 0; i < n; i++) {
+    EIGN[i] = -d[i];
+    for (j = 0; j < n; j++) EIGV[i * n + j] = a[j * n + i];
+    invfreq[i] = 1 / EIGV[i * n + 0];
+  }
+  ext_EIGN[0] = 0.0;
+  for (l = 1; l < n; l++) {
+    ext_EIGN[l] = EIGN[l] * (1.0 / fracchange);
+    assert(ext_EIGN[l] > 0.0);
+  }
+  for (i = 0; i < n; i++)
+    for (j = 0; j < n; j++) EV[i * n + j] = EIGV[i * n + j];
+  for (i = 0; i < n; i++)
+    for (j = 0; j < n; j++)
+      EI[i * n + j] = (j == 0) ? 1.0 : EV[i * n + j] * invfreq[i];
+
+  for (i = 0; i < vlen; i++) {
+    unsigned int value = valueVector[i];
+    for (j = 0; j < n; j++) tipVector[i * n + j] = 0;
+    if (value > 0)
+      for (j = 0; j < n; j++)
+        if ((value >> j) & 1)
+          for (l = 0; l < n; l++) tipVector[i * n + l] += EIGV[j * n + l];
+  }
+  for (i = 0; i < vlen; i++)
+    for (j = 0; j < n; j++)
+      if (tipVector[i * n + j] > ORC_MAX_TIP_EV)
+        tipVector[i * n + j] = ORC_MAX_TIP_EV;
+}
