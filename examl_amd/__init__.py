@@ -1,0 +1,102 @@
+"""examl_amd — MI355X-native (gfx950/CDNA4) implementation of ExaML's
+per-site conditional-likelihood hot path.
+
+The compute path is libexaml_hip.so (hand-written HIP kernels behind the
+C-ABI in include/examl_hip.h).  This package is the host-side mirror of the
+reference's likelihood entry points (newviewGeneric / evaluateGeneric /
+makenewzGeneric shapes); PyTorch is used only for device memory, streams and
+the RCCL all-reduce.
+
+The HIP extension is REQUIRED on a GPU machine: there is no CPU fallback in
+the product path (the CPU restatement under oracle/ is test infrastructure
+and must never be imported from here).
+"""
+
+import ctypes
+import os
+
+__version__ = "0.1"
+
+_HERE = os.path.dirname(os.path.abspath(__file__))
+_LIB_PATH = os.path.join(_HERE, "libexaml_hip.so")
+
+TIP_TIP, TIP_INNER, INNER_INNER = 0, 1, 2
+ZMIN, ZMAX = 1.0e-15, 1.0 - 1.0e-6
+
+
+class TravEntry(ctypes.Structure):
+    """examl_hip_trav_entry (include/examl_hip.h) — one post-order CLV
+    update, mirroring the reference's traversalInfo (examl/axml.h:434)."""
+    _fields_ = [
+        ("tipCase", ctypes.c_int),
+        ("pNumber", ctypes.c_int),
+        ("qNumber", ctypes.c_int),
+        ("rNumber", ctypes.c_int),
+        ("x1Slot", ctypes.c_int),
+        ("x2Slot", ctypes.c_int),
+        ("x3Slot", ctypes.c_int),
+        ("qz", ctypes.c_double),
+        ("rz", ctypes.c_double),
+    ]
+
+
+def _bind(lib):
+    d = ctypes.c_double
+    i = ctypes.c_int
+    l = ctypes.c_long
+    p = ctypes.c_void_p
+    lib.examl_hip_version.restype = ctypes.c_char_p
+    lib.examl_hip_last_error_string.restype = ctypes.c_char_p
+    lib.examl_host_make_p.argtypes = [d, d, p, p, p, i, p, p, i]
+    lib.examl_host_calc_diagptable.argtypes = [d, i, i, p, p, p]
+    lib.examl_host_core_dtables_dna.argtypes = [p, p, d, p]
+    lib.examl_host_init_gtr_dna.argtypes = [p, p, p, p, p, p]
+    lib.examl_host_make_gamma_cats.argtypes = [d, p, i]
+    lib.examl_hip_newview_dna_gamma.argtypes = \
+        [i, p, p, p, p, p, p, p, l, p, p, p, p, p]
+    lib.examl_hip_evaluate_dna_gamma.argtypes = \
+        [p, p, p, p, p, l, p, p, p, d, p, p]
+    lib.examl_hip_sum_dna_gamma.argtypes = [i, p, p, p, p, p, p, l, p]
+    lib.examl_hip_core_dna_gamma.argtypes = [l, p, p, p, p, p]
+    lib.examl_hip_newview_traversal_dna_gamma.argtypes = \
+        [p, i, p, p, p, p, p, p, l, p, l, p, l, p, p, p, p]
+    lib.examl_hip_evaluate_root_dna_gamma.argtypes = \
+        [i, i, i, i, i, i, d, p, p, p, p, l, p, l, p, l, p, p, p, p]
+    lib.examl_hip_sum_root_dna_gamma.argtypes = \
+        [i, i, i, i, i, p, p, l, p, l, p, l, p]
+    lib.examl_hip_core_root_dna_gamma.argtypes = [l, p, p, p, d, p, p, p, p]
+    return lib
+
+
+_lib = None
+_load_error = None
+try:
+    _lib = _bind(ctypes.CDLL(_LIB_PATH))
+except OSError as e:  # pragma: no cover
+    _load_error = e
+
+
+def lib():
+    """The HIP extension.  Fails loudly if it is missing — no fallback."""
+    if _lib is None:
+        raise RuntimeError(
+            f"examl_amd: HIP extension not available at {_LIB_PATH} "
+            f"({_load_error}); build it with __graft_entry__.build()")
+    return _lib
+
+
+def check(rc, what):
+    if rc != 0:
+        raise RuntimeError(
+            f"examl_amd: {what} failed (rc={rc}): "
+            f"{lib().examl_hip_last_error_string().decode()}")
+
+
+from .model import DnaGtrModel          # noqa: E402
+from .tree import PhyloTree             # noqa: E402
+from .engine import DnaGammaEngine      # noqa: E402
+
+__all__ = [
+    "lib", "check", "TravEntry", "DnaGtrModel", "PhyloTree",
+    "DnaGammaEngine", "TIP_TIP", "TIP_INNER", "INNER_INNER", "ZMIN", "ZMAX",
+]

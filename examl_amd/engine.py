@@ -1,0 +1,219 @@
+"""Device engine for one DNA GTRGAMMA partition slice: owns the HBM-resident
+state (CLVs, tips, weights, model vectors, scaler counts) and drives the HIP
+kernels through the C-ABI (include/examl_hip.h).
+
+This is the GPU replacement for the per-partition bodies of
+newviewIterative / evaluateIterative / makenewzIterative+execCore
+(SURVEY.md §8a rows a3-a5).  All buffers are torch CUDA tensors (device
+memory + stream plumbing only); kernels run on torch's current stream so
+torch.distributed (RCCL) collectives compose in stream order.
+"""
+
+import ctypes
+import math
+
+import numpy as np
+import torch
+
+from . import INNER_INNER, TIP_INNER, TIP_TIP, TravEntry, ZMIN, ZMAX, check, lib
+
+
+def _vp(t):
+    return ctypes.c_void_p(t.data_ptr())
+
+
+def _np_vp(a):
+    return a.ctypes.data_as(ctypes.c_void_p)
+
+
+class DnaGammaEngine:
+    SPAN = 16  # 4 states x 4 gamma cats
+
+    def __init__(self, tips, wgt, model, device="cuda", max_ops=None):
+        """tips: uint8 [ntips+1, width] (row 0 unused; ambiguity codes 1..15,
+        the yVector of examl/axml.h:599); wgt: int32 [width]; model:
+        DnaGtrModel."""
+        assert tips.dtype == np.uint8 and tips.ndim == 2
+        self.ntips = tips.shape[0] - 1
+        self.width = tips.shape[1]
+        self.ninner = self.ntips - 2
+        self.model = model
+        self.device = torch.device(device)
+        if self.device.type == "cuda" and not torch.cuda.is_available():
+            raise RuntimeError("examl_amd: CUDA/HIP device not available")
+        n_ops = max_ops or (self.ninner + 8)
+
+        dev = self.device
+        self.d_tips = torch.from_numpy(np.ascontiguousarray(tips)).to(dev)
+        self.d_wgt = torch.from_numpy(
+            np.ascontiguousarray(wgt, dtype=np.int32)).to(dev)
+        self.d_EV = torch.from_numpy(model.EV).to(dev)
+        self.d_tipVector = torch.from_numpy(model.tipVector).to(dev)
+        # CLV pool: one slot per inner node (xVector, device-resident for the
+        # life of the run — the GPU answer to the lazy allocation of
+        # newviewGenericSpecial.c:1200-1215)
+        self.d_clv = torch.empty((self.ninner, self.width * self.SPAN),
+                                 dtype=torch.float64, device=dev)
+        # per-node scaler counts (globalScaler, axml.h:601); tips stay 0
+        self.d_scalers = torch.zeros(2 * self.ntips, dtype=torch.int32,
+                                     device=dev)
+        self.d_pbuf = torch.empty(n_ops * 128, dtype=torch.float64, device=dev)
+        self.d_inc = torch.empty(n_ops, dtype=torch.int32, device=dev)
+        self.d_diag = torch.empty(16, dtype=torch.float64, device=dev)
+        self.d_dtab = torch.empty(48, dtype=torch.float64, device=dev)
+        self.d_lnl = torch.zeros(1, dtype=torch.float64, device=dev)
+        self.d_out2 = torch.zeros(2, dtype=torch.float64, device=dev)
+        self.d_sum = None  # sumBuffer (axml.h:558), allocated on first use
+        self._max_ops = n_ops
+
+    # -- plumbing -----------------------------------------------------------
+
+    def _stream(self):
+        if self.device.type == "cuda":
+            return ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
+        return ctypes.c_void_p(0)
+
+    def sync(self):
+        if self.device.type == "cuda":
+            torch.cuda.synchronize(self.device)
+
+    # -- newviewIterative ---------------------------------------------------
+
+    def newview_traversal(self, entries):
+        """Run the post-order CLV updates (newviewIterative,
+        newviewGenericSpecial.c:917)."""
+        if not entries:
+            return
+        assert len(entries) <= self._max_ops, "grow max_ops"
+        arr = (TravEntry * len(entries))(*entries)
+        m = self.model
+        check(lib().examl_hip_newview_traversal_dna_gamma(
+            ctypes.cast(arr, ctypes.c_void_p), len(entries),
+            _np_vp(m.EIGN), _np_vp(m.EI), _np_vp(m.gammaRates),
+            _vp(self.d_EV), _vp(self.d_tipVector), _vp(self.d_clv),
+            ctypes.c_long(self.width * self.SPAN), _vp(self.d_tips),
+            ctypes.c_long(self.width), _vp(self.d_wgt),
+            ctypes.c_long(self.width), _vp(self.d_scalers), _vp(self.d_inc),
+            _vp(self.d_pbuf), self._stream()), "newview_traversal")
+
+    # -- evaluateIterative --------------------------------------------------
+
+    def _root_case(self, tree, p, q):
+        """Resolve the root branch operands like evaluateIterative
+        (evaluateGenericSpecial.c:612-668): tip goes to the tip slot, the
+        other side's CLV to x2."""
+        p_tip, q_tip = tree.is_tip(p), tree.is_tip(q)
+        assert not (p_tip and q_tip)
+        if q_tip:
+            return TIP_INNER, -1, tree.clv_slot(p), q, -1, p, q
+        if p_tip:
+            return TIP_INNER, -1, tree.clv_slot(q), p, -1, p, q
+        return (INNER_INNER, tree.clv_slot(p), tree.clv_slot(q), -1, -1, p, q)
+
+    def evaluate_root(self, tree, p, q, z, all_reduce=False):
+        """lnL at the branch p--q (evaluateIterative's per-partition body +
+        the C1 all-reduce, evaluateGenericSpecial.c:403/969).  Returns the
+        device scalar tensor (call .item() to sync)."""
+        tc, x1s, x2s, tslot, _, pn, qn = self._root_case(tree, p, q)
+        self.d_lnl.zero_()
+        m = self.model
+        check(lib().examl_hip_evaluate_root_dna_gamma(
+            tc, pn, qn, x1s, x2s, tslot, ctypes.c_double(z),
+            _np_vp(m.EIGN), _np_vp(m.gammaRates), _vp(self.d_tipVector),
+            _vp(self.d_clv), ctypes.c_long(self.width * self.SPAN),
+            _vp(self.d_tips), ctypes.c_long(self.width), _vp(self.d_wgt),
+            ctypes.c_long(self.width), _vp(self.d_scalers), _vp(self.d_diag),
+            _vp(self.d_lnl), self._stream()), "evaluate_root")
+        if all_reduce and torch.distributed.is_initialized():
+            torch.distributed.all_reduce(self.d_lnl)
+        return self.d_lnl
+
+    def full_lnl(self, tree, root_edge=None, all_reduce=False):
+        """Full-traversal evaluateGeneric (evaluateGenericSpecial.c:897)."""
+        entries, (p, q, z) = tree.full_traversal(root_edge)
+        self.newview_traversal(entries)
+        return self.evaluate_root(tree, p, q, z, all_reduce=all_reduce)
+
+    # -- makenewz (NR branch-length optimization) ---------------------------
+
+    def _ensure_sum(self):
+        if self.d_sum is None:
+            self.d_sum = torch.empty(self.width * self.SPAN,
+                                     dtype=torch.float64, device=self.device)
+
+    def sum_root(self, tree, p, q):
+        """sumBuffer precompute at branch p--q (makenewzIterative,
+        makenewzGenericSpecial.c:628; CLVs must already face the branch)."""
+        self._ensure_sum()
+        p_tip, q_tip = tree.is_tip(p), tree.is_tip(q)
+        if p_tip and q_tip:
+            tc, x1s, x2s, t1, t2 = TIP_TIP, -1, -1, p, q
+        elif q_tip:
+            tc, x1s, x2s, t1, t2 = TIP_INNER, -1, tree.clv_slot(p), q, -1
+        elif p_tip:
+            tc, x1s, x2s, t1, t2 = TIP_INNER, -1, tree.clv_slot(q), p, -1
+        else:
+            tc, x1s, x2s, t1, t2 = (INNER_INNER, tree.clv_slot(p),
+                                    tree.clv_slot(q), -1, -1)
+        check(lib().examl_hip_sum_root_dna_gamma(
+            tc, x1s, x2s, t1, t2, _vp(self.d_tipVector), _vp(self.d_clv),
+            ctypes.c_long(self.width * self.SPAN), _vp(self.d_tips),
+            ctypes.c_long(self.width), _vp(self.d_sum),
+            ctypes.c_long(self.width), self._stream()), "sum_root")
+
+    def core_derivs(self, lz, all_reduce=False):
+        """execCore (makenewzGenericSpecial.c:849) + the C2 all-reduce
+        (:1244).  Returns (dlnLdlz, d2lnLdlz2) as host floats."""
+        m = self.model
+        self.d_out2.zero_()
+        check(lib().examl_hip_core_root_dna_gamma(
+            ctypes.c_long(self.width), _vp(self.d_sum), _np_vp(m.EIGN),
+            _np_vp(m.gammaRates), ctypes.c_double(lz), _vp(self.d_wgt),
+            _vp(self.d_dtab), _vp(self.d_out2), self._stream()),
+            "core_root")
+        if all_reduce and torch.distributed.is_initialized():
+            torch.distributed.all_reduce(self.d_out2)
+        v = self.d_out2.cpu()
+        return float(v[0]), float(v[1])
+
+    def makenewz(self, tree, p, q, z0, maxiter=64, all_reduce=False):
+        """Newton-Raphson branch length at p--q, restating topLevelMakenewz
+        (makenewzGenericSpecial.c:1133) for the joint-branch-length case
+        (numBranches=1).  CLVs must already face the branch (the caller runs
+        a traversal to p--q first, as makenewzGeneric:1385 does)."""
+        self.sum_root(tree, p, q)
+        z = float(z0)
+        zprev = z
+        zstep = (1.0 - ZMAX) * z + ZMIN
+        curvat_ok = True
+        outer_converged = False
+        it = maxiter
+        while not outer_converged:
+            if curvat_ok:
+                curvat_ok = False
+                zprev = z
+                zstep = (1.0 - ZMAX) * z + ZMIN
+            z = min(max(z, ZMIN), ZMAX)
+            lz = math.log(z)
+            dlnL, d2lnL = self.core_derivs(lz, all_reduce=all_reduce)
+            if (d2lnL >= 0.0) and (z < ZMAX):
+                zprev = z = 0.37 * z + 0.63  # bad curvature, shorten branch
+                continue
+            curvat_ok = True
+            if d2lnL < 0.0:
+                tantmp = -dlnL / d2lnL
+                if tantmp < 100:
+                    z *= math.exp(tantmp)
+                    z = max(z, ZMIN)
+                    z = min(z, 0.25 * zprev + 0.75)
+                else:
+                    z = 0.25 * zprev + 0.75
+            z = min(z, ZMAX)
+            it -= 1
+            if abs(z - zprev) > zstep:
+                if it < -20:
+                    z = float(z0)
+                    outer_converged = True
+            else:
+                outer_converged = True
+        return z

@@ -1,0 +1,204 @@
+/* ============================================================================
+ * examl_hip.h — C-ABI boundary of the MI355X-native ExaML likelihood core.
+ *
+ * This is the drop-in surface replacing ExaML's L0/L1 likelihood kernels
+ * (SURVEY.md §8b).  Each entry point cites the reference function it
+ * replaces (file:line into the upstream ExaML tree).  The host side of
+ * ExaML (searchAlgo.c / optimizeModel.c, plain C) drives these through the
+ * same call shapes its own dispatch layer uses; INTEGRATION.md shows the
+ * binding a maintainer would add.
+ *
+ * Conventions:
+ *   - "dev_" prefixed pointers are DEVICE (HIP) pointers; all others are
+ *     host pointers.  No torch types anywhere.
+ *   - `stream` is a hipStream_t passed as void* (0 = default stream).
+ *   - All launches are stream-ordered and asynchronous; scalar results land
+ *     in device buffers the caller reads back (or feeds to RCCL).
+ *   - Return 0 on success, nonzero HIP error code otherwise;
+ *     examl_hip_last_error_string() describes the last failure.
+ *   - Memory layout is the reference's: CLV x[site*span + cat*states + state]
+ *     fp64 with span = 4*states (GAMMA); P-matrices left/right
+ *     [cat*states^2 + row*states + col]; tipVector[code*states + state];
+ *     per-node scaler counts as unsigned int (axml.h:601 globalScaler).
+ * ==========================================================================*/
+
+#ifndef EXAML_HIP_H
+#define EXAML_HIP_H
+
+#include <stddef.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* tipCase values — reference examl/axml.h:302-304 */
+#define EXAML_TIP_TIP 0
+#define EXAML_TIP_INNER 1
+#define EXAML_INNER_INNER 2
+
+const char *examl_hip_version(void);
+const char *examl_hip_last_error_string(void);
+
+/* ---------------------------------------------------------------------------
+ * Host-side model math (runs once per model-parameter change; feeds the
+ * kernels).  These replace the corresponding host functions in the
+ * reference and keep bit-identical arithmetic.
+ * ------------------------------------------------------------------------ */
+
+/* replaces makeP (examl/newviewGenericSpecial.c:78); z1/z2 are
+ * log-transformed branch lengths (caller clamps to zmin and takes log as
+ * newviewGenericSpecial.c:982-983 does). */
+void examl_host_make_p(double z1, double z2, const double *rates,
+                       const double *EI, const double *EIGN, int numCats,
+                       double *left, double *right, int states);
+
+/* replaces calcDiagptable (examl/evaluateGenericSpecial.c:80); z is the raw
+ * branch length (clamp+log inside). */
+void examl_host_calc_diagptable(double z, int states, int numCats,
+                                const double *rates, const double *EIGN,
+                                double *diag);
+
+/* d0/d1/d2 tables of coreGTRGAMMA (examl/makenewzGenericSpecial.c:2330-2346)
+ * for states=4: out48 = {d0[16], d1[16], d2[16]}. */
+void examl_host_core_dtables_dna(const double *EIGN, const double *gammaRates,
+                                 double lz, double *out48);
+
+/* replaces initReversibleGTR -> initGeneric (examl/models.c:3462/3234) for
+ * DNA (states=4, 16 ambiguity codes): emits EIGN[4], EV[16], EI[16],
+ * tipVector[64]. */
+void examl_host_init_gtr_dna(const double *frequencies, const double *rates6,
+                             double *EIGN, double *EV, double *EI,
+                             double *tipVector);
+
+/* replaces makeGammaCats (examl/models.c:3795), mean-rate (useMedian=FALSE)
+ * discrete gamma. */
+void examl_host_make_gamma_cats(double alpha, double *gammaRates, int K);
+
+/* ---------------------------------------------------------------------------
+ * L0 device kernels (one launch each).  DNA GTRGAMMA: states=4, span=16.
+ * ------------------------------------------------------------------------ */
+
+/* replaces newviewGTRGAMMA_AVX (examl/avxLikelihood.c:64, decl
+ * examl/axml.h:1360): x3 = EV . ((P_L x1) o (P_R x2)) per (site, gamma cat),
+ * with the reference's exact 2^-256 underflow rescale rule (all 16 span
+ * entries below threshold; no scaling in TIP_TIP) and scaler counts
+ * accumulated as sum of wgt[site] into *dev_scalerInc (device, caller
+ * zeroes). x1/x2/x3/tipX1/tipX2/left/right/wgt/EV/tipVector are device
+ * pointers. */
+int examl_hip_newview_dna_gamma(int tipCase, const double *dev_x1,
+                                const double *dev_x2, double *dev_x3,
+                                const double *dev_EV,
+                                const double *dev_tipVector,
+                                const unsigned char *dev_tipX1,
+                                const unsigned char *dev_tipX2, long n,
+                                const double *dev_left,
+                                const double *dev_right, const int *dev_wgt,
+                                unsigned int *dev_scalerInc, void *stream);
+
+/* replaces evaluateGTRGAMMA (examl/evaluateGenericSpecial.c:1879) plus the
+ * scaler undo at :830: atomically accumulates
+ *   sum_i wgt[i]*log(0.25*|term_i|)  +  (gs_p + gs_q) * log_minlik
+ * into *dev_lnl (caller zeroes).  dev_tipX1 == NULL selects the
+ * inner-inner body.  dev_gsP/dev_gsQ point at the two nodes' scaler counts
+ * (device; pass NULL,NULL to skip the undo term). */
+int examl_hip_evaluate_dna_gamma(const int *dev_wgt, const double *dev_x1,
+                                 const double *dev_x2,
+                                 const double *dev_tipVector,
+                                 const unsigned char *dev_tipX1, long n,
+                                 const double *dev_diag,
+                                 const unsigned int *dev_gsP,
+                                 const unsigned int *dev_gsQ,
+                                 double log_minlik, double *dev_lnl,
+                                 void *stream);
+
+/* replaces sumGAMMA (examl/makenewzGenericSpecial.c:1798):
+ * dev_sum[i,c,k] = x1'[i,c,k] * x2'[i,c,k] with tip expansion. */
+int examl_hip_sum_dna_gamma(int tipCase, double *dev_sum,
+                            const double *dev_x1, const double *dev_x2,
+                            const double *dev_tipVector,
+                            const unsigned char *dev_tipX1,
+                            const unsigned char *dev_tipX2, long n,
+                            void *stream);
+
+/* replaces coreGTRGAMMA (examl/makenewzGenericSpecial.c:2309): accumulates
+ * {dlnLdlz, d2lnLdlz2} into dev_out2[2] (caller zeroes).  dev_dtables holds
+ * the 48 doubles from examl_host_core_dtables_dna, uploaded by the caller. */
+int examl_hip_core_dna_gamma(long n, const double *dev_sum,
+                             const double *dev_dtables, const int *dev_wgt,
+                             double *dev_out2, void *stream);
+
+/* ---------------------------------------------------------------------------
+ * L1 batched executors (the newviewIterative-shaped path: one host call per
+ * traversal, kernels chained on `stream`).
+ * ------------------------------------------------------------------------ */
+
+/* One post-order CLV update op — the traversalInfo entry of
+ * examl/axml.h:434-442 with CLV slot / tip-row bindings resolved by the
+ * caller (the xVector/yVector indexing of newviewIterative,
+ * examl/newviewGenericSpecial.c:1221-1261). */
+typedef struct {
+  int tipCase;           /* EXAML_TIP_* */
+  int pNumber, qNumber, rNumber; /* node ids, index dev_scalers */
+  int x1Slot, x2Slot, x3Slot;    /* CLV slots; for tip operands the slot is
+                                    the tip row in dev_tips (x1Slot for q's
+                                    tip, x2Slot for r when TIP_TIP) */
+  double qz, rz;                 /* raw branch lengths */
+} examl_hip_trav_entry;
+
+/* replaces the per-partition body of newviewIterative
+ * (examl/newviewGenericSpecial.c:917): computes all P-matrix pairs on the
+ * host (makeP), uploads them in one copy, launches one newview kernel per
+ * entry in post-order, then one finalize kernel that applies the recursive
+ * scaler accumulation globalScaler[p] = gs[q] + gs[r] + inc
+ * (newviewGenericSpecial.c:1503-1510; gs of tip nodes stays 0).
+ * dev_pbuf: >= numOps*128 doubles; dev_inc: >= numOps uints (zeroed here). */
+int examl_hip_newview_traversal_dna_gamma(
+    const examl_hip_trav_entry *ops, int numOps, const double *EIGN,
+    const double *EI, const double *gammaRates, const double *dev_EV,
+    const double *dev_tipVector, double *dev_clv, long clvStrideDoubles,
+    const unsigned char *dev_tips, long tipStrideBytes, const int *dev_wgt,
+    long n, unsigned int *dev_scalers, unsigned int *dev_inc,
+    double *dev_pbuf, void *stream);
+
+/* replaces the per-partition body of evaluateIterative
+ * (examl/evaluateGenericSpecial.c:403): calcDiagptable on the host for the
+ * root branch z, upload, then the evaluate kernel including the scaler
+ * undo.  rootTipCase: EXAML_TIP_INNER (tipSlot = tip row of the tip node)
+ * or EXAML_INNER_INNER.  dev_diag_scratch: >= 16 doubles.  *dev_lnl is
+ * accumulated (caller zeroes; feed to RCCL all-reduce afterwards —
+ * replaces the MPI_Allreduce at evaluateGenericSpecial.c:969). */
+int examl_hip_evaluate_root_dna_gamma(
+    int rootTipCase, int pNumber, int qNumber, int x1Slot, int x2Slot,
+    int tipSlot, double z, const double *EIGN, const double *gammaRates,
+    const double *dev_tipVector, double *dev_clv, long clvStrideDoubles,
+    const unsigned char *dev_tips, long tipStrideBytes, const int *dev_wgt,
+    long n, const unsigned int *dev_scalers, double *dev_diag_scratch,
+    double *dev_lnl, void *stream);
+
+/* replaces the per-partition body of makenewzIterative's sum precompute
+ * (examl/makenewzGenericSpecial.c:628,673-839) for the branch p--q. */
+int examl_hip_sum_root_dna_gamma(int rootTipCase, int x1Slot, int x2Slot,
+                                 int tipSlot, int tipSlot2,
+                                 const double *dev_tipVector, double *dev_clv,
+                                 long clvStrideDoubles,
+                                 const unsigned char *dev_tips,
+                                 long tipStrideBytes, double *dev_sum, long n,
+                                 void *stream);
+
+/* replaces the per-partition body of execCore
+ * (examl/makenewzGenericSpecial.c:849): host dtables + upload + core
+ * kernel; dev_dtab_scratch >= 48 doubles; dev_out2 accumulated (caller
+ * zeroes; reduce across ranks afterwards — replaces the MPI_Allreduce at
+ * makenewzGenericSpecial.c:1244). */
+int examl_hip_core_root_dna_gamma(long n, const double *dev_sum,
+                                  const double *EIGN,
+                                  const double *gammaRates, double lz,
+                                  const int *dev_wgt,
+                                  double *dev_dtab_scratch, double *dev_out2,
+                                  void *stream);
+
+#ifdef __cplusplus
+}
+#endif
+
+#endif /* EXAML_HIP_H */

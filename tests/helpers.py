@@ -1,0 +1,154 @@
+"""Shared test helpers: synthetic data and the ORACLE-side executor that
+replays a product traversal on the CPU restatement (the checker for GPU
+parity and host-logic tests)."""
+
+import math
+
+import numpy as np
+
+import oracle as O
+from examl_amd import INNER_INNER, TIP_INNER, TIP_TIP
+
+
+def make_synthetic(ntips, width, seed=42):
+    """Random tip matrix (codes 1..15 heavily biased to pure bases A/C/G/T =
+    1/2/4/8) + pattern weights; row 0 unused (ExaML yVector is 1-based)."""
+    rng = np.random.default_rng(seed)
+    tips = np.zeros((ntips + 1, width), dtype=np.uint8)
+    pure = np.array([1, 2, 4, 8], dtype=np.uint8)
+    base = pure[rng.integers(0, 4, width)]
+    for t in range(1, ntips + 1):
+        row = base.copy()
+        mut = rng.random(width) < 0.10
+        row[mut] = pure[rng.integers(0, 4, mut.sum())]
+        amb = rng.random(width) < 0.01
+        row[amb] = rng.integers(1, 16, amb.sum()).astype(np.uint8)
+        tips[t] = row
+    wgt = np.ones(width, dtype=np.int32)
+    return tips, wgt
+
+
+def _model_arrays(model):
+    """Aligned copies of a DnaGtrModel's vectors for the oracle's AVX-path
+    reference calls."""
+    def al(a):
+        out = O.aligned(a.shape)
+        out[:] = a
+        return out
+    return (al(model.EIGN), al(model.EV), al(model.EI), al(model.tipVector),
+            al(model.gammaRates))
+
+
+def oracle_full_lnl(entries, root, tree, model, tips, wgt,
+                    return_state=False):
+    """Replay the product's traversal entries through the CPU oracle
+    (newview per entry + recursive scalers + evaluate at the root),
+    restating evaluateGeneric end to end."""
+    EIGN, EV, EI, tipVector, g = _model_arrays(model)
+    width = tips.shape[1]
+    ntips = tips.shape[0] - 1
+    clv = {}
+    scalers = np.zeros(2 * ntips, dtype=np.int64)
+    wgt = np.ascontiguousarray(wgt, dtype=np.int32)
+
+    for e in entries:
+        qz = math.log(e.qz) if e.qz > O.ZMIN else math.log(O.ZMIN)
+        rz = math.log(e.rz) if e.rz > O.ZMIN else math.log(O.ZMIN)
+        left, right = O.make_p(qz, rz, g, EI, EIGN, 4, 4)
+        if e.tipCase == TIP_TIP:
+            x3, inc = O.newview_dna_gamma(
+                TIP_TIP, None, None, EV, tipVector,
+                np.ascontiguousarray(tips[e.x1Slot]),
+                np.ascontiguousarray(tips[e.x2Slot]), width, left, right, wgt)
+        elif e.tipCase == TIP_INNER:
+            x3, inc = O.newview_dna_gamma(
+                TIP_INNER, None, clv[e.x2Slot], EV, tipVector,
+                np.ascontiguousarray(tips[e.x1Slot]), None, width, left,
+                right, wgt)
+        else:
+            x3, inc = O.newview_dna_gamma(
+                INNER_INNER, clv[e.x1Slot], clv[e.x2Slot], EV, tipVector,
+                None, None, width, left, right, wgt)
+        clv[e.x3Slot] = x3
+        scalers[e.pNumber] = scalers[e.qNumber] + scalers[e.rNumber] + inc
+
+    p, q, z = root
+    diag = O.calc_diagptable(z, 4, 4, g, EIGN)
+    p_tip, q_tip = tree.is_tip(p), tree.is_tip(q)
+    if q_tip:
+        lnl = O.evaluate_dna_gamma(wgt, None, clv[tree.clv_slot(p)],
+                                   tipVector, np.ascontiguousarray(tips[q]),
+                                   width, diag)
+    elif p_tip:
+        lnl = O.evaluate_dna_gamma(wgt, None, clv[tree.clv_slot(q)],
+                                   tipVector, np.ascontiguousarray(tips[p]),
+                                   width, diag)
+    else:
+        lnl = O.evaluate_dna_gamma(wgt, clv[tree.clv_slot(p)],
+                                   clv[tree.clv_slot(q)], tipVector, None,
+                                   width, diag)
+    lnl += float(scalers[p] + scalers[q]) * math.log(O.MINLIKELIHOOD)
+    if return_state:
+        return lnl, clv, scalers
+    return lnl
+
+
+def oracle_makenewz(entries, root, tree, model, tips, wgt, z0, maxiter=64):
+    """CPU restatement of topLevelMakenewz (numBranches=1) over the oracle
+    sum/core kernels — the checker for DnaGammaEngine.makenewz."""
+    EIGN, EV, EI, tipVector, g = _model_arrays(model)
+    width = tips.shape[1]
+    _, clv, _ = oracle_full_lnl(entries, root, tree, model, tips, wgt,
+                                return_state=True)
+    p, q, _ = root
+    p_tip, q_tip = tree.is_tip(p), tree.is_tip(q)
+    wgt = np.ascontiguousarray(wgt, dtype=np.int32)
+    if p_tip and q_tip:
+        st = O.sum_dna_gamma(TIP_TIP, None, None, tipVector,
+                             np.ascontiguousarray(tips[p]),
+                             np.ascontiguousarray(tips[q]), width)
+    elif q_tip:
+        st = O.sum_dna_gamma(TIP_INNER, None, clv[tree.clv_slot(p)],
+                             tipVector, np.ascontiguousarray(tips[q]), None,
+                             width)
+    elif p_tip:
+        st = O.sum_dna_gamma(TIP_INNER, None, clv[tree.clv_slot(q)],
+                             tipVector, np.ascontiguousarray(tips[p]), None,
+                             width)
+    else:
+        st = O.sum_dna_gamma(INNER_INNER, clv[tree.clv_slot(p)],
+                             clv[tree.clv_slot(q)], tipVector, None, None,
+                             width)
+
+    z = float(z0)
+    zprev, zstep = z, (1.0 - O.ZMAX) * z + O.ZMIN
+    curvat_ok, outer_converged, it = True, False, maxiter
+    while not outer_converged:
+        if curvat_ok:
+            curvat_ok = False
+            zprev = z
+            zstep = (1.0 - O.ZMAX) * z + O.ZMIN
+        z = min(max(z, O.ZMIN), O.ZMAX)
+        lz = math.log(z)
+        dlnL, d2lnL = O.core_dna_gamma(width, st, EIGN, g, lz, wgt)
+        if (d2lnL >= 0.0) and (z < O.ZMAX):
+            zprev = z = 0.37 * z + 0.63
+            continue
+        curvat_ok = True
+        if d2lnL < 0.0:
+            tantmp = -dlnL / d2lnL
+            if tantmp < 100:
+                z *= math.exp(tantmp)
+                z = max(z, O.ZMIN)
+                z = min(z, 0.25 * zprev + 0.75)
+            else:
+                z = 0.25 * zprev + 0.75
+        z = min(z, O.ZMAX)
+        it -= 1
+        if abs(z - zprev) > zstep:
+            if it < -20:
+                z = float(z0)
+                outer_converged = True
+        else:
+            outer_converged = True
+    return z
