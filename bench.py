@@ -1,0 +1,226 @@
+#!/usr/bin/env python3
+"""Headline benchmark: ExaML's north-star workload on MI355X.
+
+Workload (BASELINE.json configs[1]): synthetic DNA, 50 taxa x 1,000,000
+sites per GPU, single partition, GTRGAMMA, fp64.  One "step" = one full-tree
+evaluateGeneric: a full post-order traversal (48 newview CLV updates over
+all sites) + the root lnL evaluation + the per-partition lnL all-reduce
+(RCCL when world>1) — the per-partition body of SURVEY.md §3.1.
+
+  python bench.py --gpus N --steps K --warmup W
+
+N>1 is launched by the driver as one rank per GPU via torch.distributed.run;
+sites shard one-million-per-rank (weak scaling — per-GPU work fixed, the
+reference's site sharding with the C1 all-reduce in the loop).
+
+Output: ONE JSON line from rank 0 (see the repo contract), including
+  roofline     — the dominant kernel (newview INNER_INNER), algorithmic
+                 bytes/launch over its HIP-event launch time vs 8 TB/s HBM3E
+  cpu_baseline — the reference's own AVX kernel (oracle/_ref/libref.so)
+                 timed on this box's host cores (kind "reference"), or the
+                 oracle C restatement (kind "port") if _ref is absent.
+"""
+
+import argparse
+import ctypes
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import torch  # noqa: E402
+
+import examl_amd as ea  # noqa: E402
+from examl_amd.synthetic import make_alignment  # noqa: E402
+
+NTAXA = 50
+SITES_PER_GPU = 1_000_000
+# algorithmic HBM traffic per site for newview:
+#   INNER_INNER: 2x16 fp64 read + 16 fp64 write + 4 B wgt = 388 B
+BYTES_PER_SITE_II = 388.0
+HBM_PEAK = 8.0e12  # B/s, MI355X HBM3E spec
+
+
+def log(msg):
+    if int(os.environ.get("RANK", "0")) == 0:
+        print(msg, file=sys.stderr, flush=True)
+
+
+def cpu_baseline_leg(width=200_000, budget_s=10.0):
+    """Time the reference AVX newview (INNER_INNER) single-threaded on this
+    host, bounded to ~budget_s; returns the cpu_baseline dict."""
+    try:
+        import oracle as O
+        lib = O._ref if O.have_ref() else None
+        kind = "reference" if lib is not None else "port"
+        rng = np.random.default_rng(1)
+        x1 = O.aligned(width * 16)
+        x1[:] = rng.uniform(0.01, 1.0, width * 16)
+        x2 = O.aligned(width * 16)
+        x2[:] = rng.uniform(0.01, 1.0, width * 16)
+        wgt = np.ones(width, dtype=np.int32)
+        EIGN, EV, EI, tipVector = O.init_gtr_dna(
+            [0.25] * 4, [1.0, 2.0, 0.8, 1.1, 3.0, 1.0])
+        g = O.make_gamma_cats(0.5)
+        left, right = O.make_p(np.log(0.9), np.log(0.7), g, EI, EIGN, 4, 4)
+        # one calibration call, then fill the budget
+        t0 = time.perf_counter()
+        O.newview_dna_gamma(ea.INNER_INNER, x1, x2, EV, tipVector, None,
+                            None, width, left, right, wgt, lib=lib)
+        per = time.perf_counter() - t0
+        reps = max(3, int(budget_s / max(per, 1e-3)))
+        t0 = time.perf_counter()
+        for _ in range(reps):
+            O.newview_dna_gamma(ea.INNER_INNER, x1, x2, EV, tipVector, None,
+                                None, width, left, right, wgt, lib=lib)
+        el = time.perf_counter() - t0
+        return {
+            "value": reps * width / el,
+            "unit": "site-updates/s",
+            "cores": 1,
+            "kind": kind,
+            "sample": (f"newviewGTRGAMMA_AVX INNER_INNER, {width} sites x "
+                       f"{reps} reps, 1 thread"
+                       if kind == "reference" else
+                       f"oracle newview II, {width} sites x {reps} reps, "
+                       f"1 thread"),
+        }
+    except Exception as e:  # pragma: no cover
+        log(f"cpu_baseline failed: {e}")
+        return None
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=60)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--sites", type=int, default=SITES_PER_GPU)
+    ap.add_argument("--taxa", type=int, default=NTAXA)
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+        dist = dist_mod
+        torch.cuda.set_device(local_rank)
+        dist.init_process_group("nccl")
+    device = torch.device(f"cuda:{local_rank}")
+
+    ntips, width = args.taxa, args.sites
+    log(f"generating synthetic alignment: {ntips} taxa x {width} sites "
+        f"(rank {rank}/{world})")
+    tips, wgt = make_alignment(ntips, width, seed=42 + rank)
+    model = ea.DnaGtrModel([0.28, 0.22, 0.24, 0.26],
+                           [1.2, 2.9, 0.7, 1.0, 3.2, 1.0], alpha=0.6)
+    tree = ea.PhyloTree.random(ntips, seed=7)
+    eng = ea.DnaGammaEngine(tips, wgt, model, device=device)
+    entries, (p, q, z) = tree.full_traversal()
+    n_ops = len(entries)
+    tc_counts = [sum(1 for e in entries if e.tipCase == t) for t in range(3)]
+
+    def step():
+        eng.newview_traversal(entries)
+        lnl = eng.evaluate_root(tree, p, q, z, all_reduce=world > 1)
+        return lnl
+
+    # warmup
+    for _ in range(args.warmup):
+        step()
+    eng.sync()
+    lnl0 = float(eng.d_lnl.cpu())
+
+    L = ea.lib()
+    L.examl_hip_profile_reset()
+    L.examl_hip_profile_enable(1)
+
+    if dist:
+        dist.barrier()
+    eng.sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    eng.sync()
+    if dist:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    L.examl_hip_profile_enable(0)
+    ms = np.zeros(3)
+    cnt = np.zeros(3, dtype=np.int64)
+    L.examl_hip_profile_get(ms.ctypes.data_as(ctypes.c_void_p),
+                            cnt.ctypes.data_as(ctypes.c_void_p))
+
+    # max over ranks
+    if dist:
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.cpu())
+
+    if rank != 0:
+        return
+
+    # roofline of the dominant kernel (newview INNER_INNER)
+    ii_ms, ii_n = float(ms[2]), int(cnt[2])
+    achieved = (BYTES_PER_SITE_II * width * ii_n) / (ii_ms * 1e-3) \
+        if ii_ms > 0 else None
+    roofline = {
+        "bound": "hbm",
+        "achieved": achieved / 1e9 if achieved else None,
+        "peak": HBM_PEAK / 1e9,
+        "unit": "GB/s",
+        "frac": achieved / HBM_PEAK if achieved else None,
+        "traffic": None,  # PMC traffic: see profiles/ (rocprofv3 --pmc runs)
+    }
+
+    cpu = None
+    if not args.no_cpu_baseline and world == 1 \
+            and not os.environ.get("EXAML_BENCH_NO_CPU"):
+        log("timing CPU baseline (reference AVX kernel, 1 core)...")
+        cpu = cpu_baseline_leg()
+
+    site_updates = float(args.steps) * n_ops * width * world
+    value = site_updates / elapsed
+    out = {
+        "metric": "newview site-lnL updates/sec",
+        "value": value,
+        "unit": "site-updates/s",
+        "n_gpus": world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": elapsed / args.steps * 1e3,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "f64",
+        "data": "synthetic",
+        "config": {
+            "workload": "synthetic_dna_50taxa_1Msites_gtrgamma "
+                        "(BASELINE.json configs[1]; full-tree evaluateGeneric"
+                        " per step)",
+            "taxa": ntips,
+            "sites_per_gpu": width,
+            "partitions": 1,
+            "newview_ops_per_step": n_ops,
+            "tipcase_counts": {"TT": tc_counts[0], "TI": tc_counts[1],
+                               "II": tc_counts[2]},
+            "full_tree_eval_ms": elapsed / args.steps * 1e3,
+            "lnl": lnl0,
+            "parallelism": f"dp{world} site-sharded, 1 RCCL all-reduce/step",
+        },
+        "roofline": roofline,
+        "cpu_baseline": cpu,
+    }
+    print(json.dumps(out), flush=True)
+
+
+if __name__ == "__main__":
+    main()

@@ -65,6 +65,9 @@ def _bind(lib):
     lib.examl_hip_sum_root_dna_gamma.argtypes = \
         [i, i, i, i, i, p, p, l, p, l, p, l, p]
     lib.examl_hip_core_root_dna_gamma.argtypes = [l, p, p, p, d, p, p, p, p]
+    lib.examl_hip_profile_enable.argtypes = [i]
+    lib.examl_hip_profile_reset.argtypes = []
+    lib.examl_hip_profile_get.argtypes = [p, p]
     return lib
 
 

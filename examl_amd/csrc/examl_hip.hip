@@ -430,6 +430,7 @@ extern "C" int examl_hip_newview_dna_gamma(
     void *stream) {
   if (n <= 0) return 0;
   hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError(); /* clear stale per-thread error (e.g. torch probes) */
   const int grid = grid_for(n * 4);
   /* left/right must be contiguous (P = left | right); the launcher copies
    * are avoided by requiring the caller to pass left==P, right==P+64 when
@@ -471,6 +472,7 @@ extern "C" int examl_hip_evaluate_dna_gamma(
     double *lnl, void *stream) {
   if (n <= 0) return 0;
   hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError(); /* clear stale per-thread error (e.g. torch probes) */
   const int grid = grid_for(n * 4);
   if (tipX1)
     hipLaunchKernelGGL((k_evaluate_dna_gamma<true>), dim3(grid),
@@ -492,6 +494,7 @@ extern "C" int examl_hip_sum_dna_gamma(int tipCase, double *sum,
                                        void *stream) {
   if (n <= 0) return 0;
   hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError(); /* clear stale per-thread error (e.g. torch probes) */
   const int grid = grid_for(n * 4);
   switch (tipCase) {
   case EXAML_TIP_TIP:
@@ -522,10 +525,69 @@ extern "C" int examl_hip_core_dna_gamma(long n, const double *sum,
                                         double *out2, void *stream) {
   if (n <= 0) return 0;
   hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError(); /* clear stale per-thread error (e.g. torch probes) */
   hipLaunchKernelGGL(k_core_dna_gamma, dim3(grid_for(n * 4)), dim3(NV_BLOCK),
                      0, s, sum, dtab, wgt, n, out2);
   CHK(hipGetLastError());
   return 0;
+}
+
+/* --- kernel-time profiling (HIP events on the launch stream) --------------
+ * bench.py's roofline leg: when enabled, every newview launch in the
+ * traversal executor is bracketed by hipEvent pairs, accumulated per
+ * tipCase on examl_hip_profile_get (which synchronizes the device).
+ */
+#include <vector>
+
+struct ProfEv {
+  hipEvent_t a, b;
+  int tc;
+};
+static bool g_prof_on = false;
+static std::vector<ProfEv> g_prof_pend;
+static std::vector<std::pair<hipEvent_t, hipEvent_t>> g_prof_pool;
+static double g_prof_ms[3] = {0, 0, 0};
+static long g_prof_cnt[3] = {0, 0, 0};
+
+static void prof_flush() {
+  for (auto &e : g_prof_pend) {
+    float ms = 0;
+    hipEventSynchronize(e.b);
+    hipEventElapsedTime(&ms, e.a, e.b);
+    g_prof_ms[e.tc] += ms;
+    g_prof_cnt[e.tc] += 1;
+    g_prof_pool.push_back({e.a, e.b});
+  }
+  g_prof_pend.clear();
+}
+
+static void prof_begin(hipEvent_t *a, hipEvent_t *b) {
+  if (!g_prof_pool.empty()) {
+    *a = g_prof_pool.back().first;
+    *b = g_prof_pool.back().second;
+    g_prof_pool.pop_back();
+  } else {
+    hipEventCreate(a);
+    hipEventCreate(b);
+  }
+}
+
+extern "C" void examl_hip_profile_enable(int on) { g_prof_on = on != 0; }
+
+extern "C" void examl_hip_profile_reset(void) {
+  prof_flush();
+  for (int i = 0; i < 3; i++) {
+    g_prof_ms[i] = 0;
+    g_prof_cnt[i] = 0;
+  }
+}
+
+extern "C" void examl_hip_profile_get(double *ms_by_tc, long *cnt_by_tc) {
+  prof_flush();
+  for (int i = 0; i < 3; i++) {
+    ms_by_tc[i] = g_prof_ms[i];
+    cnt_by_tc[i] = g_prof_cnt[i];
+  }
 }
 
 /* --- batched traversal (newviewIterative body) --------------------------- */
@@ -539,6 +601,7 @@ extern "C" int examl_hip_newview_traversal_dna_gamma(
     void *stream) {
   if (numOps <= 0 || n <= 0) return 0;
   hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError(); /* clear stale per-thread error (e.g. torch probes) */
 
   /* 1. all P-matrix pairs on the host, one block of 128 doubles per op
    *    (newviewGenericSpecial.c:982-1044: clamp to zmin, log, makeP) */
@@ -566,6 +629,11 @@ extern "C" int examl_hip_newview_traversal_dna_gamma(
   const int grid = grid_for(n * 4);
   for (int e = 0; e < numOps; e++) {
     const examl_hip_trav_entry *op = &ops[e];
+    hipEvent_t ev_a = nullptr, ev_b = nullptr;
+    if (g_prof_on) {
+      prof_begin(&ev_a, &ev_b);
+      hipEventRecord(ev_a, s);
+    }
     const double *P = dev_pbuf + (long)e * 128;
     double *x3 = dev_clv + (long)op->x3Slot * clvStride;
     const double *x1 = nullptr, *x2 = nullptr;
@@ -598,6 +666,11 @@ extern "C" int examl_hip_newview_traversal_dna_gamma(
       return -1;
     }
     CHK(hipGetLastError());
+    if (g_prof_on) {
+      hipEventRecord(ev_b, s);
+      g_prof_pend.push_back({ev_a, ev_b, op->tipCase});
+      if (g_prof_pend.size() > 2048) prof_flush();
+    }
   }
 
   /* 3. recursive scaler accumulation, chunked through by-value kernargs */
@@ -628,6 +701,7 @@ extern "C" int examl_hip_evaluate_root_dna_gamma(
     void *stream) {
   if (n <= 0) return 0;
   hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError(); /* clear stale per-thread error (e.g. torch probes) */
   double hostDiag[16];
   examl_host_calc_diagptable(z, 4, 4, gammaRates, EIGN, hostDiag);
   CHK(hipMemcpyAsync(dev_diag, hostDiag, sizeof(hostDiag),
@@ -695,6 +769,7 @@ extern "C" int examl_hip_core_root_dna_gamma(long n, const double *dev_sum,
                                              double *dev_out2, void *stream) {
   if (n <= 0) return 0;
   hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError(); /* clear stale per-thread error (e.g. torch probes) */
   double host48[48];
   examl_host_core_dtables_dna(EIGN, gammaRates, lz, host48);
   CHK(hipMemcpyAsync(dev_dtab, host48, sizeof(host48), hipMemcpyHostToDevice,

@@ -39,6 +39,14 @@ extern "C" {
 const char *examl_hip_version(void);
 const char *examl_hip_last_error_string(void);
 
+/* Kernel-time profiling for the bench's roofline leg: when enabled, every
+ * newview launch of the traversal executor is bracketed with hipEvent pairs
+ * on its launch stream; _get synchronizes outstanding pairs and returns the
+ * accumulated {ms, launch count} per tipCase (index = EXAML_TIP_*). */
+void examl_hip_profile_enable(int on);
+void examl_hip_profile_reset(void);
+void examl_hip_profile_get(double *ms_by_tc, long *cnt_by_tc);
+
 /* ---------------------------------------------------------------------------
  * Host-side model math (runs once per model-parameter change; feeds the
  * kernels).  These replace the corresponding host functions in the

@@ -11,21 +11,9 @@ from examl_amd import INNER_INNER, TIP_INNER, TIP_TIP
 
 
 def make_synthetic(ntips, width, seed=42):
-    """Random tip matrix (codes 1..15 heavily biased to pure bases A/C/G/T =
-    1/2/4/8) + pattern weights; row 0 unused (ExaML yVector is 1-based)."""
-    rng = np.random.default_rng(seed)
-    tips = np.zeros((ntips + 1, width), dtype=np.uint8)
-    pure = np.array([1, 2, 4, 8], dtype=np.uint8)
-    base = pure[rng.integers(0, 4, width)]
-    for t in range(1, ntips + 1):
-        row = base.copy()
-        mut = rng.random(width) < 0.10
-        row[mut] = pure[rng.integers(0, 4, mut.sum())]
-        amb = rng.random(width) < 0.01
-        row[amb] = rng.integers(1, 16, amb.sum()).astype(np.uint8)
-        tips[t] = row
-    wgt = np.ones(width, dtype=np.int32)
-    return tips, wgt
+    """Seeded synthetic alignment (delegates to the product generator)."""
+    from examl_amd.synthetic import make_alignment
+    return make_alignment(ntips, width, seed=seed)
 
 
 def _model_arrays(model):
