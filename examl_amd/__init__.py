@@ -15,6 +15,14 @@ and must never be imported from here).
 import ctypes
 import os
 
+# Load torch (and thus its bundled HIP runtime, soname libamdhip64.so.7)
+# BEFORE dlopen'ing our C-ABI library: our DT_NEEDED "libamdhip64.so.7" then
+# resolves to the already-loaded instance and the process has ONE HIP
+# runtime.  In the other order torch's DT_NEEDED "libamdhip64.so" loads a
+# second runtime whose HSA instance sees no device (observed: rc=100
+# hipErrorNoDevice from every call of ours while torch works).
+import torch  # noqa: F401
+
 __version__ = "0.1"
 
 _HERE = os.path.dirname(os.path.abspath(__file__))
