@@ -172,13 +172,22 @@ def main():
     ii_ms, ii_n = float(ms[2]), int(cnt[2])
     achieved = (BYTES_PER_SITE_II * width * ii_n) / (ii_ms * 1e-3) \
         if ii_ms > 0 else None
+    # PMC-measured HBM traffic for this exact workload (collected in a
+    # separate rocprofv3 --pmc pass, corrected per MI355X_MICROARCH.md §HBM;
+    # see profiles/r01_pmc_traffic.json)
+    traffic = None
+    cal = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                       "profiles", "r01_pmc_traffic.json")
+    if width == SITES_PER_GPU and os.path.exists(cal):
+        with open(cal) as f:
+            traffic = json.load(f)["traffic_bytes_per_launch"]
     roofline = {
         "bound": "hbm",
         "achieved": achieved / 1e9 if achieved else None,
         "peak": HBM_PEAK / 1e9,
         "unit": "GB/s",
         "frac": achieved / HBM_PEAK if achieved else None,
-        "traffic": None,  # PMC traffic: see profiles/ (rocprofv3 --pmc runs)
+        "traffic": traffic,
     }
 
     cpu = None

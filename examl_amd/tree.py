@@ -77,6 +77,23 @@ class PhyloTree:
         assert next_inner == 2 * ntips - 1
         return t
 
+    @staticmethod
+    def caterpillar(ntips, z=DEFAULTZ):
+        """Maximally deep (chain) topology — drives the 2^-256 rescale path
+        (CLV magnitudes shrink multiplicatively with depth)."""
+        t = PhyloTree(ntips)
+        first = ntips + 1
+        t.add_edge(1, first, z)
+        t.add_edge(2, first, z)
+        prev = first
+        for k in range(3, ntips):
+            m = ntips + k - 1
+            t.add_edge(prev, m, z)
+            t.add_edge(k, m, z)
+            prev = m
+        t.add_edge(ntips, prev, z)
+        return t
+
     # -- traversal ----------------------------------------------------------
 
     def _collect(self, node, parent, out):

@@ -156,10 +156,11 @@ def test_full_pipeline_vs_oracle(dev):
 def test_full_pipeline_scaling_depth(dev):
     """Deep caterpillar-ish tree with long branches -> guaranteed 2^-256
     rescales; lnL must still match the oracle."""
-    ntips, width = 40, 2048
+    ntips, width = 500, 128
     tips, wgt = make_synthetic(ntips, width, seed=99)
     model = ea.DnaGtrModel.jukes_cantor(alpha=0.3)
-    tree = ea.PhyloTree.random(ntips, seed=3, z=0.35)
+    # 498-op traversal: also exercises the chunked scaler finalize
+    tree = ea.PhyloTree.caterpillar(ntips, z=0.5)
     eng = ea.DnaGammaEngine(tips, wgt, model, device=dev)
     entries, root = tree.full_traversal()
     lnl = eng.full_lnl(tree).item()

@@ -76,6 +76,21 @@ def test_oracle_makenewz_improves_lnl(setup):
     assert z_opt != z0
 
 
+def test_caterpillar_rescale_path():
+    """A 500-deep chain tree drives CLVs through the 2^-256 rescale; the
+    recursive scaler accounting must keep lnL finite and negative."""
+    ntips, width = 500, 64
+    tips, wgt = make_synthetic(ntips, width, seed=99)
+    model = ea.DnaGtrModel.jukes_cantor(alpha=0.3)
+    tree = ea.PhyloTree.caterpillar(ntips, z=0.5)
+    entries, root = tree.full_traversal()
+    assert len(entries) == ntips - 2
+    lnl, _, sc = oracle_full_lnl(entries, root, tree, model, tips, wgt,
+                                 return_state=True)
+    assert sc.max() > 0
+    assert np.isfinite(lnl) and lnl < 0
+
+
 def test_weights_scale_lnl(setup):
     """Pattern-compression weights: doubling every weight doubles lnL."""
     _, _, tips, wgt, model, tree = setup
