@@ -35,13 +35,15 @@ sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 import torch  # noqa: E402
 
 import examl_amd as ea  # noqa: E402
-from examl_amd.synthetic import make_alignment  # noqa: E402
+from examl_amd.synthetic import make_alignment, make_alignment_aa  # noqa: E402
 
 NTAXA = 50
 SITES_PER_GPU = 1_000_000
-# algorithmic HBM traffic per site for newview:
-#   INNER_INNER: 2x16 fp64 read + 16 fp64 write + 4 B wgt = 388 B
-BYTES_PER_SITE_II = 388.0
+SITES_PER_GPU_PROT = 200_000
+# algorithmic HBM traffic per site for newview INNER_INNER:
+#   DNA: 2x16 fp64 read + 16 fp64 write + 4 B wgt = 388 B
+#   protein: 2x80 fp64 read + 80 fp64 write + 4 B wgt = 1924 B
+BYTES_PER_SITE_II = {4: 388.0, 20: 1924.0}
 HBM_PEAK = 8.0e12  # B/s, MI355X HBM3E spec
 
 
@@ -50,44 +52,57 @@ def log(msg):
         print(msg, file=sys.stderr, flush=True)
 
 
-def cpu_baseline_leg(width=200_000, budget_s=10.0):
+def cpu_baseline_leg(width=200_000, budget_s=10.0, protein=False):
     """Time the reference AVX newview (INNER_INNER) single-threaded on this
     host, bounded to ~budget_s; returns the cpu_baseline dict."""
     try:
         import oracle as O
         lib = O._ref if O.have_ref() else None
         kind = "reference" if lib is not None else "port"
+        st = 20 if protein else 4
+        span = 4 * st
+        if protein:
+            width = min(width, 20_000)
         rng = np.random.default_rng(1)
-        x1 = O.aligned(width * 16)
-        x1[:] = rng.uniform(0.01, 1.0, width * 16)
-        x2 = O.aligned(width * 16)
-        x2[:] = rng.uniform(0.01, 1.0, width * 16)
+        x1 = O.aligned(width * span)
+        x1[:] = rng.uniform(0.01, 1.0, width * span)
+        x2 = O.aligned(width * span)
+        x2[:] = rng.uniform(0.01, 1.0, width * span)
         wgt = np.ones(width, dtype=np.int32)
-        EIGN, EV, EI, tipVector = O.init_gtr_dna(
-            [0.25] * 4, [1.0, 2.0, 0.8, 1.1, 3.0, 1.0])
+        if protein:
+            lg = np.load(os.path.join(os.path.dirname(
+                os.path.abspath(__file__)), "examl_amd", "data",
+                "lg_model.npz"))
+            EIGN, EV, EI, tipVector = O.init_gtr_aa(lg["frequencies"],
+                                                    lg["rates190"])
+            nv = O.newview_prot_gamma
+            name = "newviewGTRGAMMAPROT_AVX"
+        else:
+            EIGN, EV, EI, tipVector = O.init_gtr_dna(
+                [0.25] * 4, [1.0, 2.0, 0.8, 1.1, 3.0, 1.0])
+            nv = O.newview_dna_gamma
+            name = "newviewGTRGAMMA_AVX"
         g = O.make_gamma_cats(0.5)
-        left, right = O.make_p(np.log(0.9), np.log(0.7), g, EI, EIGN, 4, 4)
+        left, right = O.make_p(np.log(0.9), np.log(0.7), g, EI, EIGN, 4, st)
         # one calibration call, then fill the budget
         t0 = time.perf_counter()
-        O.newview_dna_gamma(ea.INNER_INNER, x1, x2, EV, tipVector, None,
-                            None, width, left, right, wgt, lib=lib)
+        nv(ea.INNER_INNER, x1, x2, EV, tipVector, None, None, width, left,
+           right, wgt, lib=lib)
         per = time.perf_counter() - t0
         reps = max(3, int(budget_s / max(per, 1e-3)))
         t0 = time.perf_counter()
         for _ in range(reps):
-            O.newview_dna_gamma(ea.INNER_INNER, x1, x2, EV, tipVector, None,
-                                None, width, left, right, wgt, lib=lib)
+            nv(ea.INNER_INNER, x1, x2, EV, tipVector, None, None, width,
+               left, right, wgt, lib=lib)
         el = time.perf_counter() - t0
         return {
             "value": reps * width / el,
             "unit": "site-updates/s",
             "cores": 1,
             "kind": kind,
-            "sample": (f"newviewGTRGAMMA_AVX INNER_INNER, {width} sites x "
-                       f"{reps} reps, 1 thread"
-                       if kind == "reference" else
-                       f"oracle newview II, {width} sites x {reps} reps, "
-                       f"1 thread"),
+            "sample": f"{name} INNER_INNER, {width} sites x {reps} reps, "
+                      f"1 thread" + ("" if kind == "reference"
+                                     else " (oracle restatement)"),
         }
     except Exception as e:  # pragma: no cover
         log(f"cpu_baseline failed: {e}")
@@ -99,8 +114,10 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=60)
     ap.add_argument("--warmup", type=int, default=10)
-    ap.add_argument("--sites", type=int, default=SITES_PER_GPU)
+    ap.add_argument("--sites", type=int, default=0)
     ap.add_argument("--taxa", type=int, default=NTAXA)
+    ap.add_argument("--protein", action="store_true",
+                    help="config 4: 50 taxa x 200k sites, LG+GAMMA")
     ap.add_argument("--no-cpu-baseline", action="store_true")
     args = ap.parse_args()
 
@@ -115,12 +132,18 @@ def main():
         dist.init_process_group("nccl")
     device = torch.device(f"cuda:{local_rank}")
 
-    ntips, width = args.taxa, args.sites
+    ntips = args.taxa
+    width = args.sites or (SITES_PER_GPU_PROT if args.protein
+                           else SITES_PER_GPU)
     log(f"generating synthetic alignment: {ntips} taxa x {width} sites "
-        f"(rank {rank}/{world})")
-    tips, wgt = make_alignment(ntips, width, seed=42 + rank)
-    model = ea.DnaGtrModel([0.28, 0.22, 0.24, 0.26],
-                           [1.2, 2.9, 0.7, 1.0, 3.2, 1.0], alpha=0.6)
+        f"(rank {rank}/{world}, {'LG+GAMMA' if args.protein else 'GTRGAMMA'})")
+    if args.protein:
+        tips, wgt = make_alignment_aa(ntips, width, seed=42 + rank)
+        model = ea.ProtGtrModel.lg(alpha=0.75)
+    else:
+        tips, wgt = make_alignment(ntips, width, seed=42 + rank)
+        model = ea.DnaGtrModel([0.28, 0.22, 0.24, 0.26],
+                               [1.2, 2.9, 0.7, 1.0, 3.2, 1.0], alpha=0.6)
     tree = ea.PhyloTree.random(ntips, seed=7)
     eng = ea.DnaGammaEngine(tips, wgt, model, device=device)
     entries, (p, q, z) = tree.full_traversal()
@@ -170,7 +193,8 @@ def main():
 
     # roofline of the dominant kernel (newview INNER_INNER)
     ii_ms, ii_n = float(ms[2]), int(cnt[2])
-    achieved = (BYTES_PER_SITE_II * width * ii_n) / (ii_ms * 1e-3) \
+    bps = BYTES_PER_SITE_II[model.states]
+    achieved = (bps * width * ii_n) / (ii_ms * 1e-3) \
         if ii_ms > 0 else None
     # PMC-measured HBM traffic for this exact workload (collected in a
     # separate rocprofv3 --pmc pass, corrected per MI355X_MICROARCH.md §HBM;
@@ -178,7 +202,7 @@ def main():
     traffic = None
     cal = os.path.join(os.path.dirname(os.path.abspath(__file__)),
                        "profiles", "r01_pmc_traffic.json")
-    if width == SITES_PER_GPU and os.path.exists(cal):
+    if width == SITES_PER_GPU and model.states == 4 and os.path.exists(cal):
         with open(cal) as f:
             traffic = json.load(f)["traffic_bytes_per_launch"]
     roofline = {
@@ -194,7 +218,7 @@ def main():
     if not args.no_cpu_baseline and world == 1 \
             and not os.environ.get("EXAML_BENCH_NO_CPU"):
         log("timing CPU baseline (reference AVX kernel, 1 core)...")
-        cpu = cpu_baseline_leg()
+        cpu = cpu_baseline_leg(protein=args.protein)
 
     site_updates = float(args.steps) * n_ops * width * world
     value = site_updates / elapsed
@@ -212,9 +236,11 @@ def main():
         "dtype": "f64",
         "data": "synthetic",
         "config": {
-            "workload": "synthetic_dna_50taxa_1Msites_gtrgamma "
-                        "(BASELINE.json configs[1]; full-tree evaluateGeneric"
-                        " per step)",
+            "workload": ("synthetic_prot_50taxa_200ksites_lg_gamma "
+                         "(BASELINE.json configs[3])" if args.protein else
+                         "synthetic_dna_50taxa_1Msites_gtrgamma "
+                         "(BASELINE.json configs[1])")
+                        + "; full-tree evaluateGeneric per step",
             "taxa": ntips,
             "sites_per_gpu": width,
             "partitions": 1,

@@ -73,6 +73,22 @@ def _bind(lib):
     lib.examl_hip_sum_root_dna_gamma.argtypes = \
         [i, i, i, i, i, p, p, l, p, l, p, l, p]
     lib.examl_hip_core_root_dna_gamma.argtypes = [l, p, p, p, d, p, p, p, p]
+    # protein surface (same shapes, states=20)
+    lib.examl_host_init_gtr_aa.argtypes = [p, p, p, p, p, p]
+    lib.examl_host_core_dtables_prot.argtypes = [p, p, d, p]
+    lib.examl_hip_newview_prot_gamma.argtypes = \
+        [i, p, p, p, p, p, p, p, l, p, p, p, p, p]
+    lib.examl_hip_evaluate_prot_gamma.argtypes = \
+        [p, p, p, p, p, l, p, p, p, d, p, p]
+    lib.examl_hip_sum_prot_gamma.argtypes = [i, p, p, p, p, p, p, l, p]
+    lib.examl_hip_core_prot_gamma.argtypes = [l, p, p, p, p, p]
+    lib.examl_hip_newview_traversal_prot_gamma.argtypes = \
+        [p, i, p, p, p, p, p, p, l, p, l, p, l, p, p, p, p]
+    lib.examl_hip_evaluate_root_prot_gamma.argtypes = \
+        [i, i, i, i, i, i, d, p, p, p, p, l, p, l, p, l, p, p, p, p]
+    lib.examl_hip_sum_root_prot_gamma.argtypes = \
+        [i, i, i, i, i, p, p, l, p, l, p, l, p]
+    lib.examl_hip_core_root_prot_gamma.argtypes = [l, p, p, p, d, p, p, p, p]
     lib.examl_hip_profile_enable.argtypes = [i]
     lib.examl_hip_profile_reset.argtypes = []
     lib.examl_hip_profile_get.argtypes = [p, p]
@@ -103,11 +119,11 @@ def check(rc, what):
             f"{lib().examl_hip_last_error_string().decode()}")
 
 
-from .model import DnaGtrModel          # noqa: E402
+from .model import DnaGtrModel, ProtGtrModel  # noqa: E402
 from .tree import PhyloTree             # noqa: E402
 from .engine import DnaGammaEngine      # noqa: E402
 
 __all__ = [
-    "lib", "check", "TravEntry", "DnaGtrModel", "PhyloTree",
+    "lib", "check", "TravEntry", "DnaGtrModel", "ProtGtrModel", "PhyloTree",
     "DnaGammaEngine", "TIP_TIP", "TIP_INNER", "INNER_INNER", "ZMIN", "ZMAX",
 ]

@@ -334,6 +334,287 @@ __global__ __launch_bounds__(NV_BLOCK) void k_core_dna_gamma(
   }
 }
 
+/* ===========================================================================
+ * Protein (20-state) GTRGAMMA kernels — span 80, tip codes 1..22.
+ * Same thread <-> (site, cat) mapping; x rows held in registers (160 B per
+ * operand per thread; the wave's loads cover a contiguous 10 KiB region so
+ * L1/L2 line reuse keeps HBM traffic algorithmic), P/EV/ump tables in LDS.
+ * Summation order matches newviewGTRGAMMAPROT_AVX (avxLikelihood.c:1312):
+ * 20-dots as four lane accumulators over five 4-chunks, (t0+t1)+(t2+t3).
+ * ==========================================================================*/
+
+__device__ __forceinline__ double dot20o(const double *a, const double *b) {
+  double t0 = 0, t1 = 0, t2 = 0, t3 = 0;
+#pragma unroll
+  for (int c = 0; c < 20; c += 4) {
+    t0 += a[c] * b[c];
+    t1 += a[c + 1] * b[c + 1];
+    t2 += a[c + 2] * b[c + 2];
+    t3 += a[c + 3] * b[c + 3];
+  }
+  return (t0 + t1) + (t2 + t3);
+}
+
+template <int TC>
+__global__ __launch_bounds__(NV_BLOCK) void k_newview_prot_gamma(
+    const double *__restrict__ x1, const double *__restrict__ x2,
+    double *__restrict__ x3, const double *__restrict__ P,
+    const double *__restrict__ EV, const double *__restrict__ tipVec,
+    const unsigned char *__restrict__ tipX1,
+    const unsigned char *__restrict__ tipX2, const int *__restrict__ wgt,
+    long n, unsigned int *__restrict__ scalerInc) {
+  __shared__ double sL[1600], sR[1600], sEV[400];
+  __shared__ double sTV[TC != EXAML_INNER_INNER ? 460 : 1];
+  __shared__ double sU1[TC != EXAML_INNER_INNER ? 1840 : 1];
+  __shared__ double sU2[TC == EXAML_TIP_TIP ? 1840 : 1];
+
+  const int tid = threadIdx.x;
+  for (int j = tid; j < 1600; j += NV_BLOCK) {
+    sL[j] = P[j];
+    sR[j] = P[1600 + j];
+  }
+  for (int j = tid; j < 400; j += NV_BLOCK) sEV[j] = EV[j];
+  if (TC != EXAML_INNER_INNER)
+    for (int j = tid; j < 460; j += NV_BLOCK) sTV[j] = tipVec[j];
+  __syncthreads();
+
+  if (TC != EXAML_INNER_INNER) {
+    /* ump tables (avxLikelihood.c:1355-1389): entry (code, cat*20+row) */
+    for (int j = tid; j < 23 * 80; j += NV_BLOCK) {
+      const int code = j / 80, k = j % 80;
+      sU1[j] = dot20o(&sTV[20 * code], &sL[k * 20]);
+      if (TC == EXAML_TIP_TIP) sU2[j] = dot20o(&sTV[20 * code], &sR[k * 20]);
+    }
+    __syncthreads();
+  }
+
+  const long units = n * 4;
+  const int lane = tid & 63;
+  for (long idx = (long)blockIdx.x * NV_BLOCK + tid; idx < units;
+       idx += (long)gridDim.x * NV_BLOCK) {
+    const long site = idx >> 2;
+    const int cat = (int)(idx & 3);
+    double xl[20], xr[20], acc[20];
+    int code1 = 0, code2 = 0;
+    if (TC == EXAML_INNER_INNER) {
+#pragma unroll
+      for (int s = 0; s < 20; s += 4) {
+        const double4 a = *reinterpret_cast<const double4 *>(&x1[idx * 20 + s]);
+        const double4 b = *reinterpret_cast<const double4 *>(&x2[idx * 20 + s]);
+        xl[s] = a.x; xl[s + 1] = a.y; xl[s + 2] = a.z; xl[s + 3] = a.w;
+        xr[s] = b.x; xr[s + 1] = b.y; xr[s + 2] = b.z; xr[s + 3] = b.w;
+      }
+    } else if (TC == EXAML_TIP_INNER) {
+      code1 = tipX1[site];
+#pragma unroll
+      for (int s = 0; s < 20; s += 4) {
+        const double4 b = *reinterpret_cast<const double4 *>(&x2[idx * 20 + s]);
+        xr[s] = b.x; xr[s + 1] = b.y; xr[s + 2] = b.z; xr[s + 3] = b.w;
+      }
+    } else {
+      code1 = tipX1[site];
+      code2 = tipX2[site];
+    }
+#pragma unroll
+    for (int s = 0; s < 20; s++) acc[s] = 0.0;
+    for (int l = 0; l < 20; l++) {
+      double u1, u2;
+      if (TC == EXAML_INNER_INNER) {
+        u1 = dot20o(xl, &sL[cat * 400 + l * 20]);
+        u2 = dot20o(xr, &sR[cat * 400 + l * 20]);
+      } else if (TC == EXAML_TIP_INNER) {
+        u1 = sU1[80 * code1 + cat * 20 + l];
+        u2 = dot20o(xr, &sR[cat * 400 + l * 20]);
+      } else {
+        u1 = sU1[80 * code1 + cat * 20 + l];
+        u2 = sU2[80 * code2 + cat * 20 + l];
+      }
+      const double t = u1 * u2;
+#pragma unroll
+      for (int s = 0; s < 20; s++) acc[s] += t * sEV[l * 20 + s];
+    }
+
+    if (TC != EXAML_TIP_TIP) {
+      bool small = true;
+#pragma unroll
+      for (int s = 0; s < 20; s++)
+        small &= (fabs(acc[s]) < MINLIKELIHOOD);
+      const unsigned long long m = __ballot(small);
+      if (((m >> (lane & ~3)) & 0xFULL) == 0xFULL) {
+#pragma unroll
+        for (int s = 0; s < 20; s++) acc[s] *= TWOTOTHE256;
+        if ((lane & 3) == 0)
+          atomicAdd(scalerInc, (unsigned int)wgt[site]);
+      }
+    }
+#pragma unroll
+    for (int s = 0; s < 20; s += 4)
+      *reinterpret_cast<double4 *>(&x3[idx * 20 + s]) =
+          make_double4(acc[s], acc[s + 1], acc[s + 2], acc[s + 3]);
+  }
+}
+
+/* evaluateGTRGAMMAPROT (evaluateGenericSpecial.c:1393) + scaler undo */
+template <bool TIP>
+__global__ __launch_bounds__(NV_BLOCK) void k_evaluate_prot_gamma(
+    const double *__restrict__ x1, const double *__restrict__ x2,
+    const double *__restrict__ tipVec, const unsigned char *__restrict__ tipX1,
+    const int *__restrict__ wgt, const double *__restrict__ diag, long n,
+    const unsigned int *__restrict__ gsP, const unsigned int *__restrict__ gsQ,
+    double log_minlik, double *__restrict__ lnlOut) {
+  __shared__ double sD[80], sTV[TIP ? 460 : 1], sRed[NV_BLOCK / 64];
+  const int tid = threadIdx.x;
+  for (int j = tid; j < 80; j += NV_BLOCK) sD[j] = diag[j];
+  if (TIP)
+    for (int j = tid; j < 460; j += NV_BLOCK) sTV[j] = tipVec[j];
+  __syncthreads();
+
+  const long units = n * 4;
+  const int lane = tid & 63;
+  double acc = 0.0;
+  for (long idx = (long)blockIdx.x * NV_BLOCK + tid; idx < units;
+       idx += (long)gridDim.x * NV_BLOCK) {
+    const long site = idx >> 2;
+    const int cat = (int)(idx & 3);
+    const double *le =
+        TIP ? &sTV[20 * tipX1[site]] : &x1[idx * 20];
+    double t0 = 0, t1 = 0;
+#pragma unroll
+    for (int l = 0; l < 20; l += 2) {
+      const double2 b = *reinterpret_cast<const double2 *>(&x2[idx * 20 + l]);
+      t0 += le[l] * b.x * sD[cat * 20 + l];
+      t1 += le[l + 1] * b.y * sD[cat * 20 + l + 1];
+    }
+    double p = t0 + t1;
+    p += __shfl_xor(p, 1);
+    p += __shfl_xor(p, 2);
+    if ((lane & 3) == 0) acc += (double)wgt[site] * log(0.25 * fabs(p));
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off);
+  if (lane == 0) sRed[tid >> 6] = acc;
+  __syncthreads();
+  if (tid == 0) {
+    double s = 0;
+#pragma unroll
+    for (int w = 0; w < NV_BLOCK / 64; w++) s += sRed[w];
+    if (blockIdx.x == 0 && gsP != nullptr)
+      s += ((double)(*gsP) + (double)(*gsQ)) * log_minlik;
+    atomicAdd(lnlOut, s);
+  }
+}
+
+/* sumGAMMAPROT (makenewzGenericSpecial.c:2083) */
+template <int TC>
+__global__ __launch_bounds__(NV_BLOCK) void k_sum_prot_gamma(
+    double *__restrict__ sum, const double *__restrict__ x1,
+    const double *__restrict__ x2, const double *__restrict__ tipVec,
+    const unsigned char *__restrict__ tipX1,
+    const unsigned char *__restrict__ tipX2, long n) {
+  __shared__ double sTV[TC != EXAML_INNER_INNER ? 460 : 1];
+  const int tid = threadIdx.x;
+  if (TC != EXAML_INNER_INNER) {
+    for (int j = tid; j < 460; j += NV_BLOCK) sTV[j] = tipVec[j];
+    __syncthreads();
+  }
+  const long units = n * 4;
+  for (long idx = (long)blockIdx.x * NV_BLOCK + tid; idx < units;
+       idx += (long)gridDim.x * NV_BLOCK) {
+    const long site = idx >> 2;
+#pragma unroll
+    for (int l = 0; l < 20; l += 2) {
+      double a0, a1, b0, b1;
+      if (TC == EXAML_TIP_TIP) {
+        a0 = sTV[20 * tipX1[site] + l];
+        a1 = sTV[20 * tipX1[site] + l + 1];
+        b0 = sTV[20 * tipX2[site] + l];
+        b1 = sTV[20 * tipX2[site] + l + 1];
+      } else if (TC == EXAML_TIP_INNER) {
+        a0 = sTV[20 * tipX1[site] + l];
+        a1 = sTV[20 * tipX1[site] + l + 1];
+        const double2 b = *reinterpret_cast<const double2 *>(&x2[idx * 20 + l]);
+        b0 = b.x;
+        b1 = b.y;
+      } else {
+        const double2 a = *reinterpret_cast<const double2 *>(&x1[idx * 20 + l]);
+        const double2 b = *reinterpret_cast<const double2 *>(&x2[idx * 20 + l]);
+        a0 = a.x;
+        a1 = a.y;
+        b0 = b.x;
+        b1 = b.y;
+      }
+      *reinterpret_cast<double2 *>(&sum[idx * 20 + l]) =
+          make_double2(a0 * b0, a1 * b1);
+    }
+  }
+}
+
+/* coreGTRGAMMAPROT (makenewzGenericSpecial.c:2581); dtab = {d0,d1,d2}[80] */
+__global__ __launch_bounds__(NV_BLOCK) void k_core_prot_gamma(
+    const double *__restrict__ sum, const double *__restrict__ dtab,
+    const int *__restrict__ wgt, long n, double *__restrict__ out2) {
+  __shared__ double sD0[80], sD1[80], sD2[80], sRed[2][NV_BLOCK / 64];
+  const int tid = threadIdx.x;
+  for (int j = tid; j < 80; j += NV_BLOCK) {
+    sD0[j] = dtab[j];
+    sD1[j] = dtab[80 + j];
+    sD2[j] = dtab[160 + j];
+  }
+  __syncthreads();
+
+  const long units = n * 4;
+  const int lane = tid & 63;
+  double accD1 = 0.0, accD2 = 0.0;
+  for (long idx = (long)blockIdx.x * NV_BLOCK + tid; idx < units;
+       idx += (long)gridDim.x * NV_BLOCK) {
+    const long site = idx >> 2;
+    const int cat = (int)(idx & 3);
+    double a0 = 0, a1 = 0, a2 = 0;
+#pragma unroll
+    for (int l = 0; l < 20; l += 2) {
+      const double2 s2 = *reinterpret_cast<const double2 *>(&sum[idx * 20 + l]);
+      const double te = sD0[cat * 20 + l] * s2.x;
+      const double to = sD0[cat * 20 + l + 1] * s2.y;
+      a0 += te + to;
+      a1 += te * sD1[cat * 20 + l] + to * sD1[cat * 20 + l + 1];
+      a2 += te * sD2[cat * 20 + l] + to * sD2[cat * 20 + l + 1];
+    }
+    a0 += __shfl_xor(a0, 1);
+    a0 += __shfl_xor(a0, 2);
+    a1 += __shfl_xor(a1, 1);
+    a1 += __shfl_xor(a1, 2);
+    a2 += __shfl_xor(a2, 1);
+    a2 += __shfl_xor(a2, 2);
+    if ((lane & 3) == 0) {
+      const double inv = 1.0 / fabs(a0);
+      const double d1 = a1 * inv, d2 = a2 * inv;
+      const double w = (double)wgt[site];
+      accD1 += w * d1;
+      accD2 += w * (d2 - d1 * d1);
+    }
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    accD1 += __shfl_down(accD1, off);
+    accD2 += __shfl_down(accD2, off);
+  }
+  if (lane == 0) {
+    sRed[0][tid >> 6] = accD1;
+    sRed[1][tid >> 6] = accD2;
+  }
+  __syncthreads();
+  if (tid == 0) {
+    double s1 = 0, s2 = 0;
+#pragma unroll
+    for (int w = 0; w < NV_BLOCK / 64; w++) {
+      s1 += sRed[0][w];
+      s2 += sRed[1][w];
+    }
+    atomicAdd(&out2[0], s1);
+    atomicAdd(&out2[1], s2);
+  }
+}
+
 /* --- scaler finalize -------------------------------------------------------
  * Applies globalScaler[p] = gs[q] + gs[r] + inc in post order
  * (newviewGenericSpecial.c:1503-1510); gs of tip nodes stays 0 so the
@@ -408,6 +689,25 @@ extern "C" void examl_host_core_dtables_dna(const double *EIGN,
       d0[i * 4 + l] = exp(EIGN[l] * ki * lz);
       d1[i * 4 + l] = EIGN[l] * ki;
       d2[i * 4 + l] = EIGN[l] * EIGN[l] * kisqr;
+    }
+  }
+}
+
+extern "C" void examl_host_core_dtables_prot(const double *EIGN,
+                                             const double *gammaRates,
+                                             double lz, double *out240) {
+  /* restates the diagptable0/1/2 setup of coreGTRGAMMAPROT,
+   * examl/makenewzGenericSpecial.c:2594-2609 */
+  double *d0 = out240, *d1 = out240 + 80, *d2 = out240 + 160;
+  for (int i = 0; i < 4; i++) {
+    const double ki = gammaRates[i], kisqr = ki * ki;
+    d0[i * 20] = 1.0;
+    d1[i * 20] = 0.0;
+    d2[i * 20] = 0.0;
+    for (int l = 1; l < 20; l++) {
+      d0[i * 20 + l] = exp(EIGN[l] * ki * lz);
+      d1[i * 20 + l] = EIGN[l] * ki;
+      d2[i * 20 + l] = EIGN[l] * EIGN[l] * kisqr;
     }
   }
 }
@@ -776,6 +1076,291 @@ extern "C" int examl_hip_core_root_dna_gamma(long n, const double *dev_sum,
                      s));
   return examl_hip_core_dna_gamma(n, dev_sum, dev_dtab, dev_wgt, dev_out2,
                                   stream);
+}
+
+/* ===========================================================================
+ * Protein launchers + executors (states=20; same shapes as the DNA ones)
+ * ==========================================================================*/
+
+extern "C" int examl_hip_newview_prot_gamma(
+    int tipCase, const double *x1, const double *x2, double *x3,
+    const double *EV, const double *tipVec, const unsigned char *tipX1,
+    const unsigned char *tipX2, long n, const double *left,
+    const double *right, const int *wgt, unsigned int *scalerInc,
+    void *stream) {
+  if (n <= 0) return 0;
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  const int grid = grid_for(n * 4);
+  if (right != left + 1600) {
+    snprintf(g_err, sizeof(g_err),
+             "newview_prot: right must be left+1600 (one P block)");
+    return -1;
+  }
+  switch (tipCase) {
+  case EXAML_TIP_TIP:
+    hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_TIP>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec,
+                       tipX1, tipX2, wgt, n, scalerInc);
+    break;
+  case EXAML_TIP_INNER:
+    hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_INNER>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec,
+                       tipX1, tipX2, wgt, n, scalerInc);
+    break;
+  case EXAML_INNER_INNER:
+    hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_INNER_INNER>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec,
+                       tipX1, tipX2, wgt, n, scalerInc);
+    break;
+  default:
+    snprintf(g_err, sizeof(g_err), "newview_prot: bad tipCase %d", tipCase);
+    return -1;
+  }
+  CHK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int examl_hip_evaluate_prot_gamma(
+    const int *wgt, const double *x1, const double *x2, const double *tipVec,
+    const unsigned char *tipX1, long n, const double *diag,
+    const unsigned int *gsP, const unsigned int *gsQ, double log_minlik,
+    double *lnl, void *stream) {
+  if (n <= 0) return 0;
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  const int grid = grid_for(n * 4);
+  if (tipX1)
+    hipLaunchKernelGGL((k_evaluate_prot_gamma<true>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, x1, x2, tipVec, tipX1, wgt, diag,
+                       n, gsP, gsQ, log_minlik, lnl);
+  else
+    hipLaunchKernelGGL((k_evaluate_prot_gamma<false>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, x1, x2, tipVec, tipX1, wgt, diag,
+                       n, gsP, gsQ, log_minlik, lnl);
+  CHK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int examl_hip_sum_prot_gamma(int tipCase, double *sum,
+                                        const double *x1, const double *x2,
+                                        const double *tipVec,
+                                        const unsigned char *tipX1,
+                                        const unsigned char *tipX2, long n,
+                                        void *stream) {
+  if (n <= 0) return 0;
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  const int grid = grid_for(n * 4);
+  switch (tipCase) {
+  case EXAML_TIP_TIP:
+    hipLaunchKernelGGL((k_sum_prot_gamma<EXAML_TIP_TIP>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, sum, x1, x2, tipVec, tipX1,
+                       tipX2, n);
+    break;
+  case EXAML_TIP_INNER:
+    hipLaunchKernelGGL((k_sum_prot_gamma<EXAML_TIP_INNER>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, sum, x1, x2, tipVec, tipX1,
+                       tipX2, n);
+    break;
+  case EXAML_INNER_INNER:
+    hipLaunchKernelGGL((k_sum_prot_gamma<EXAML_INNER_INNER>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, sum, x1, x2, tipVec, tipX1,
+                       tipX2, n);
+    break;
+  default:
+    snprintf(g_err, sizeof(g_err), "sum_prot: bad tipCase %d", tipCase);
+    return -1;
+  }
+  CHK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int examl_hip_core_prot_gamma(long n, const double *sum,
+                                         const double *dtab, const int *wgt,
+                                         double *out2, void *stream) {
+  if (n <= 0) return 0;
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  hipLaunchKernelGGL(k_core_prot_gamma, dim3(grid_for(n * 4)), dim3(NV_BLOCK),
+                     0, s, sum, dtab, wgt, n, out2);
+  CHK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int examl_hip_newview_traversal_prot_gamma(
+    const examl_hip_trav_entry *ops, int numOps, const double *EIGN,
+    const double *EI, const double *gammaRates, const double *dev_EV,
+    const double *dev_tipVec, double *dev_clv, long clvStride,
+    const unsigned char *dev_tips, long tipStride, const int *dev_wgt, long n,
+    unsigned int *dev_scalers, unsigned int *dev_inc, double *dev_pbuf,
+    void *stream) {
+  if (numOps <= 0 || n <= 0) return 0;
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+
+  static thread_local double *hostP = nullptr;
+  static thread_local int hostPCap = 0;
+  if (numOps > hostPCap) {
+    free(hostP);
+    hostP = (double *)malloc((size_t)numOps * 3200 * sizeof(double));
+    hostPCap = numOps;
+  }
+  for (int e = 0; e < numOps; e++) {
+    double qz = ops[e].qz, rz = ops[e].rz;
+    qz = (qz > ZMIN) ? log(qz) : log(ZMIN);
+    rz = (rz > ZMIN) ? log(rz) : log(ZMIN);
+    examl_host_make_p(qz, rz, gammaRates, EI, EIGN, 4, &hostP[e * 3200],
+                      &hostP[e * 3200 + 1600], 20);
+  }
+  CHK(hipMemcpyAsync(dev_pbuf, hostP, (size_t)numOps * 3200 * sizeof(double),
+                     hipMemcpyHostToDevice, s));
+  CHK(hipMemsetAsync(dev_inc, 0, (size_t)numOps * sizeof(unsigned int), s));
+
+  const int grid = grid_for(n * 4);
+  for (int e = 0; e < numOps; e++) {
+    const examl_hip_trav_entry *op = &ops[e];
+    hipEvent_t ev_a = nullptr, ev_b = nullptr;
+    if (g_prof_on) {
+      prof_begin(&ev_a, &ev_b);
+      hipEventRecord(ev_a, s);
+    }
+    const double *P = dev_pbuf + (long)e * 3200;
+    double *x3 = dev_clv + (long)op->x3Slot * clvStride;
+    const double *x1 = nullptr, *x2 = nullptr;
+    const unsigned char *t1 = nullptr, *t2 = nullptr;
+    switch (op->tipCase) {
+    case EXAML_TIP_TIP:
+      t1 = dev_tips + (long)op->x1Slot * tipStride;
+      t2 = dev_tips + (long)op->x2Slot * tipStride;
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_TIP>), dim3(grid),
+                         dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV,
+                         dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
+      break;
+    case EXAML_TIP_INNER:
+      t1 = dev_tips + (long)op->x1Slot * tipStride;
+      x2 = dev_clv + (long)op->x2Slot * clvStride;
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_INNER>), dim3(grid),
+                         dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV,
+                         dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
+      break;
+    case EXAML_INNER_INNER:
+      x1 = dev_clv + (long)op->x1Slot * clvStride;
+      x2 = dev_clv + (long)op->x2Slot * clvStride;
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_INNER_INNER>),
+                         dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P,
+                         dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
+      break;
+    default:
+      snprintf(g_err, sizeof(g_err), "traversal_prot: bad tipCase %d",
+               op->tipCase);
+      return -1;
+    }
+    CHK(hipGetLastError());
+    if (g_prof_on) {
+      hipEventRecord(ev_b, s);
+      g_prof_pend.push_back({ev_a, ev_b, op->tipCase});
+      if (g_prof_pend.size() > 2048) prof_flush();
+    }
+  }
+
+  for (int base = 0; base < numOps; base += FIN_CHUNK) {
+    FinMeta m;
+    m.count = (numOps - base < FIN_CHUNK) ? (numOps - base) : FIN_CHUNK;
+    m.base = base;
+    for (int e = 0; e < m.count; e++) {
+      m.p[e] = ops[base + e].pNumber;
+      m.q[e] = ops[base + e].qNumber;
+      m.r[e] = ops[base + e].rNumber;
+    }
+    hipLaunchKernelGGL(k_scaler_finalize, dim3(1), dim3(64), 0, s, m, dev_inc,
+                       dev_scalers);
+    CHK(hipGetLastError());
+  }
+  return 0;
+}
+
+extern "C" int examl_hip_evaluate_root_prot_gamma(
+    int rootTipCase, int pNumber, int qNumber, int x1Slot, int x2Slot,
+    int tipSlot, double z, const double *EIGN, const double *gammaRates,
+    const double *dev_tipVec, double *dev_clv, long clvStride,
+    const unsigned char *dev_tips, long tipStride, const int *dev_wgt, long n,
+    const unsigned int *dev_scalers, double *dev_diag, double *dev_lnl,
+    void *stream) {
+  if (n <= 0) return 0;
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  double hostDiag[80];
+  examl_host_calc_diagptable(z, 20, 4, gammaRates, EIGN, hostDiag);
+  CHK(hipMemcpyAsync(dev_diag, hostDiag, sizeof(hostDiag),
+                     hipMemcpyHostToDevice, s));
+  const double log_minlik = log(MINLIKELIHOOD);
+  const unsigned int *gsP = dev_scalers ? dev_scalers + pNumber : nullptr;
+  const unsigned int *gsQ = dev_scalers ? dev_scalers + qNumber : nullptr;
+  const int grid = grid_for(n * 4);
+  if (rootTipCase == EXAML_TIP_INNER) {
+    const unsigned char *t1 = dev_tips + (long)tipSlot * tipStride;
+    const double *x2 = dev_clv + (long)x2Slot * clvStride;
+    hipLaunchKernelGGL((k_evaluate_prot_gamma<true>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, nullptr, x2, dev_tipVec, t1,
+                       dev_wgt, dev_diag, n, gsP, gsQ, log_minlik, dev_lnl);
+  } else if (rootTipCase == EXAML_INNER_INNER) {
+    const double *x1 = dev_clv + (long)x1Slot * clvStride;
+    const double *x2 = dev_clv + (long)x2Slot * clvStride;
+    hipLaunchKernelGGL((k_evaluate_prot_gamma<false>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, x1, x2, dev_tipVec, nullptr,
+                       dev_wgt, dev_diag, n, gsP, gsQ, log_minlik, dev_lnl);
+  } else {
+    snprintf(g_err, sizeof(g_err), "evaluate_root_prot: bad tipCase %d",
+             rootTipCase);
+    return -1;
+  }
+  CHK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int examl_hip_sum_root_prot_gamma(
+    int rootTipCase, int x1Slot, int x2Slot, int tipSlot, int tipSlot2,
+    const double *dev_tipVec, double *dev_clv, long clvStride,
+    const unsigned char *dev_tips, long tipStride, double *dev_sum, long n,
+    void *stream) {
+  const double *x1 = nullptr, *x2 = nullptr;
+  const unsigned char *t1 = nullptr, *t2 = nullptr;
+  switch (rootTipCase) {
+  case EXAML_TIP_TIP:
+    t1 = dev_tips + (long)tipSlot * tipStride;
+    t2 = dev_tips + (long)tipSlot2 * tipStride;
+    break;
+  case EXAML_TIP_INNER:
+    t1 = dev_tips + (long)tipSlot * tipStride;
+    x2 = dev_clv + (long)x2Slot * clvStride;
+    break;
+  case EXAML_INNER_INNER:
+    x1 = dev_clv + (long)x1Slot * clvStride;
+    x2 = dev_clv + (long)x2Slot * clvStride;
+    break;
+  default:
+    snprintf(g_err, sizeof(g_err), "sum_root_prot: bad tipCase %d",
+             rootTipCase);
+    return -1;
+  }
+  return examl_hip_sum_prot_gamma(rootTipCase, dev_sum, x1, x2, dev_tipVec,
+                                  t1, t2, n, stream);
+}
+
+extern "C" int examl_hip_core_root_prot_gamma(
+    long n, const double *dev_sum, const double *EIGN,
+    const double *gammaRates, double lz, const int *dev_wgt,
+    double *dev_dtab, double *dev_out2, void *stream) {
+  if (n <= 0) return 0;
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  double host240[240];
+  examl_host_core_dtables_prot(EIGN, gammaRates, lz, host240);
+  CHK(hipMemcpyAsync(dev_dtab, host240, sizeof(host240),
+                     hipMemcpyHostToDevice, s));
+  return examl_hip_core_prot_gamma(n, dev_sum, dev_dtab, dev_wgt, dev_out2,
+                                   stream);
 }
 
 #undef CHK

@@ -76,10 +76,9 @@ static double IncompleteGamma_(double x, double alpha, double ln_gamma_alpha) {
     if (pn[5] != 0) {
       rn = pn[4] / pn[5];
       dif = fabs(gin - rn);
-      if (dif <= accurate && dif <= accurate * rn) {
-        gin = rn;
-        break;
-      }
+      /* NOTE: on convergence the reference keeps the PREVIOUS gin
+       * (models.c:3648-3651: the l42 jump precedes the l34 update) */
+      if (dif <= accurate && dif <= accurate * rn) break;
       gin = rn;
     }
     for (i = 0; i < 4; i++) pn[i] = pn[i + 2];
@@ -289,19 +288,18 @@ static void tqli_(double *d, double *e, const int n, double *z) {
   }
 }
 
-extern "C" void examl_host_init_gtr_dna(const double *frequencies,
-                                        const double *rates6, double *EIGN_out,
-                                        double *EV, double *EI,
-                                        double *tipVector) {
-  const int n = 4, vlen = 16;
-  double a[16], d[4], e[4], EIGV[16], invfreq[4], EIGN[4], r[16];
+static void init_gtr_generic(int n, const unsigned int *valueVector, int vlen,
+                             const double *frequencies, const double *rates,
+                             double *EIGN_out, double *EV, double *EI,
+                             double *tipVector) {
+  double a[400], d[20], e[20], EIGV[400], invfreq[20], EIGN[20], r[400];
   double fracchange = 0.0;
   int i, j, k, m, l;
 
-  memset(r, 0, sizeof(r));
+  memset(r, 0, sizeof(double) * n * n);
   i = 0;
   for (j = 0; j < n - 1; j++)
-    for (k = j + 1; k < n; k++) r[j * n + k] = rates6[i++];
+    for (k = j + 1; k < n; k++) r[j * n + k] = rates[i++];
   for (j = 0; j < n; j++) {
     r[j * n + j] = 0.0;
     for (k = 0; k < j; k++) r[j * n + k] = r[k * n + j];
@@ -310,11 +308,11 @@ extern "C" void examl_host_init_gtr_dna(const double *frequencies,
     for (k = 0; k < n; k++)
       fracchange += frequencies[j] * r[j * n + k] * frequencies[k];
 
-  memset(a, 0, sizeof(a));
+  memset(a, 0, sizeof(double) * n * n);
   m = 0;
   for (i = 0; i < n; i++)
     for (j = i + 1; j < n; j++) {
-      const double factor = rates6[m++];
+      const double factor = rates[m++];
       a[i * n + j] = a[j * n + i] =
           factor * sqrt(frequencies[i] * frequencies[j]);
       a[i * n + i] -= factor * frequencies[j];
@@ -360,7 +358,7 @@ extern "C" void examl_host_init_gtr_dna(const double *frequencies,
       EI[i * n + j] = (j == 0) ? 1.0 : EV[i * n + j] * invfreq[i];
 
   for (i = 0; i < vlen; i++) {
-    const unsigned int value = (unsigned int)i; /* bitVectorIdentity */
+    const unsigned int value = valueVector[i];
     for (j = 0; j < n; j++) tipVector[i * n + j] = 0;
     if (value > 0)
       for (j = 0; j < n; j++)
@@ -370,4 +368,29 @@ extern "C" void examl_host_init_gtr_dna(const double *frequencies,
   for (i = 0; i < vlen; i++)
     for (j = 0; j < n; j++)
       if (tipVector[i * n + j] > MAX_TIP_EV) tipVector[i * n + j] = MAX_TIP_EV;
+}
+
+/* DNA: bitVectorIdentity[0..15] (globalVariables.h:80), getUndetermined=15 */
+extern "C" void examl_host_init_gtr_dna(const double *frequencies,
+                                        const double *rates6, double *EIGN,
+                                        double *EV, double *EI,
+                                        double *tipVector) {
+  unsigned int vv[16];
+  for (int i = 0; i < 16; i++) vv[i] = (unsigned int)i;
+  init_gtr_generic(4, vv, 16, frequencies, rates6, EIGN, EV, EI, tipVector);
+}
+
+/* AA: bitVectorAA[23] (globalVariables.h:95) — 20 single-residue bits plus
+ * B = D|N, Z = E|Q, X = all */
+extern "C" void examl_host_init_gtr_aa(const double *frequencies,
+                                       const double *rates190, double *EIGN,
+                                       double *EV, double *EI,
+                                       double *tipVector) {
+  unsigned int vv[23];
+  for (int i = 0; i < 20; i++) vv[i] = 1u << i;
+  vv[20] = (1u << 2) | (1u << 3);  /* B: D or N */
+  vv[21] = (1u << 5) | (1u << 6);  /* Z: E or Q */
+  vv[22] = 0xFFFFFu;               /* X */
+  init_gtr_generic(20, vv, 23, frequencies, rates190, EIGN, EV, EI,
+                   tipVector);
 }

@@ -27,13 +27,17 @@ def _np_vp(a):
 
 
 class DnaGammaEngine:
-    SPAN = 16  # 4 states x 4 gamma cats
+    """Engine for one GTRGAMMA partition slice; `model.states` selects the
+    DNA (4-state) or protein (20-state) kernel family."""
 
     def __init__(self, tips, wgt, model, device="cuda", max_ops=None):
-        """tips: uint8 [ntips+1, width] (row 0 unused; ambiguity codes 1..15,
-        the yVector of examl/axml.h:599); wgt: int32 [width]; model:
-        DnaGtrModel."""
+        """tips: uint8 [ntips+1, width] (row 0 unused; ambiguity codes —
+        1..15 DNA / 1..22 AA — the yVector of examl/axml.h:599);
+        wgt: int32 [width]; model: DnaGtrModel or ProtGtrModel."""
         assert tips.dtype == np.uint8 and tips.ndim == 2
+        self.states = model.states
+        self.SPAN = 4 * self.states
+        self._sfx = "dna" if self.states == 4 else "prot"
         self.ntips = tips.shape[0] - 1
         self.width = tips.shape[1]
         self.ninner = self.ntips - 2
@@ -57,16 +61,22 @@ class DnaGammaEngine:
         # per-node scaler counts (globalScaler, axml.h:601); tips stay 0
         self.d_scalers = torch.zeros(2 * self.ntips, dtype=torch.int32,
                                      device=dev)
-        self.d_pbuf = torch.empty(n_ops * 128, dtype=torch.float64, device=dev)
+        self.d_pbuf = torch.empty(n_ops * 8 * self.states * self.states,
+                                  dtype=torch.float64, device=dev)
         self.d_inc = torch.empty(n_ops, dtype=torch.int32, device=dev)
-        self.d_diag = torch.empty(16, dtype=torch.float64, device=dev)
-        self.d_dtab = torch.empty(48, dtype=torch.float64, device=dev)
+        self.d_diag = torch.empty(4 * self.states, dtype=torch.float64,
+                                  device=dev)
+        self.d_dtab = torch.empty(12 * self.states, dtype=torch.float64,
+                                  device=dev)
         self.d_lnl = torch.zeros(1, dtype=torch.float64, device=dev)
         self.d_out2 = torch.zeros(2, dtype=torch.float64, device=dev)
         self.d_sum = None  # sumBuffer (axml.h:558), allocated on first use
         self._max_ops = n_ops
 
     # -- plumbing -----------------------------------------------------------
+
+    def _fn(self, name):
+        return getattr(lib(), f"examl_hip_{name}_{self._sfx}_gamma")
 
     def _stream(self):
         if self.device.type == "cuda":
@@ -87,7 +97,7 @@ class DnaGammaEngine:
         assert len(entries) <= self._max_ops, "grow max_ops"
         arr = (TravEntry * len(entries))(*entries)
         m = self.model
-        check(lib().examl_hip_newview_traversal_dna_gamma(
+        check(self._fn("newview_traversal")(
             ctypes.cast(arr, ctypes.c_void_p), len(entries),
             _np_vp(m.EIGN), _np_vp(m.EI), _np_vp(m.gammaRates),
             _vp(self.d_EV), _vp(self.d_tipVector), _vp(self.d_clv),
@@ -117,7 +127,7 @@ class DnaGammaEngine:
         tc, x1s, x2s, tslot, _, pn, qn = self._root_case(tree, p, q)
         self.d_lnl.zero_()
         m = self.model
-        check(lib().examl_hip_evaluate_root_dna_gamma(
+        check(self._fn("evaluate_root")(
             tc, pn, qn, x1s, x2s, tslot, ctypes.c_double(z),
             _np_vp(m.EIGN), _np_vp(m.gammaRates), _vp(self.d_tipVector),
             _vp(self.d_clv), ctypes.c_long(self.width * self.SPAN),
@@ -155,7 +165,7 @@ class DnaGammaEngine:
         else:
             tc, x1s, x2s, t1, t2 = (INNER_INNER, tree.clv_slot(p),
                                     tree.clv_slot(q), -1, -1)
-        check(lib().examl_hip_sum_root_dna_gamma(
+        check(self._fn("sum_root")(
             tc, x1s, x2s, t1, t2, _vp(self.d_tipVector), _vp(self.d_clv),
             ctypes.c_long(self.width * self.SPAN), _vp(self.d_tips),
             ctypes.c_long(self.width), _vp(self.d_sum),
@@ -166,7 +176,7 @@ class DnaGammaEngine:
         (:1244).  Returns (dlnLdlz, d2lnLdlz2) as host floats."""
         m = self.model
         self.d_out2.zero_()
-        check(lib().examl_hip_core_root_dna_gamma(
+        check(self._fn("core_root")(
             ctypes.c_long(self.width), _vp(self.d_sum), _np_vp(m.EIGN),
             _np_vp(m.gammaRates), ctypes.c_double(lz), _vp(self.d_wgt),
             _vp(self.d_dtab), _vp(self.d_out2), self._stream()),

@@ -17,6 +17,9 @@ class DnaGtrModel:
     """EIGN / EV / EI / tipVector / gammaRates for one DNA GTRGAMMA
     partition (the pInfo model block, examl/axml.h:533-629)."""
 
+    states = 4
+    n_codes = 16
+
     def __init__(self, frequencies, rates6, alpha):
         self.frequencies = np.ascontiguousarray(frequencies, dtype=np.float64)
         self.rates6 = np.ascontiguousarray(rates6, dtype=np.float64)
@@ -41,3 +44,40 @@ class DnaGtrModel:
     @staticmethod
     def jukes_cantor(alpha=1.0):
         return DnaGtrModel([0.25] * 4, [1.0] * 6, alpha)
+
+
+class ProtGtrModel:
+    """20-state (protein) GTRGAMMA model block — the AA_DATA branch of
+    initReversibleGTR (examl/models.c:3495) with explicit exchangeability
+    rates; ProtGtrModel.lg() loads the LG matrix (examl_amd/data)."""
+
+    states = 20
+    n_codes = 23
+
+    def __init__(self, frequencies, rates190, alpha):
+        self.frequencies = np.ascontiguousarray(frequencies, dtype=np.float64)
+        self.rates190 = np.ascontiguousarray(rates190, dtype=np.float64)
+        self.alpha = float(alpha)
+        assert self.frequencies.shape == (20,)
+        assert self.rates190.shape == (190,)
+        self.EIGN = np.zeros(20)
+        self.EV = np.zeros(400)
+        self.EI = np.zeros(400)
+        self.tipVector = np.zeros(23 * 20)
+        self.gammaRates = np.zeros(4)
+        L = lib()
+        L.examl_host_init_gtr_aa(_dp(self.frequencies), _dp(self.rates190),
+                                 _dp(self.EIGN), _dp(self.EV), _dp(self.EI),
+                                 _dp(self.tipVector))
+        L.examl_host_make_gamma_cats(self.alpha, _dp(self.gammaRates), 4)
+
+    def set_alpha(self, alpha):
+        self.alpha = float(alpha)
+        lib().examl_host_make_gamma_cats(self.alpha, _dp(self.gammaRates), 4)
+
+    @staticmethod
+    def lg(alpha=0.8):
+        import os
+        d = np.load(os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                                 "data", "lg_model.npz"))
+        return ProtGtrModel(d["frequencies"], d["rates190"], alpha)

@@ -205,6 +205,89 @@ int examl_hip_core_root_dna_gamma(long n, const double *dev_sum,
                                   double *dev_dtab_scratch, double *dev_out2,
                                   void *stream);
 
+/* ---------------------------------------------------------------------------
+ * Protein (20-state) GTRGAMMA surface — span 80, tip codes 1..22.  Each
+ * function replaces the 20-state counterpart of the DNA one above:
+ *   newview  — newviewGTRGAMMAPROT_AVX (examl/avxLikelihood.c:1312)
+ *   evaluate — evaluateGTRGAMMAPROT (examl/evaluateGenericSpecial.c:1393)
+ *   sum      — sumGAMMAPROT (examl/makenewzGenericSpecial.c:2083)
+ *   core     — coreGTRGAMMAPROT (examl/makenewzGenericSpecial.c:2581)
+ * P blocks are 3200 doubles (left|right), diag 80, dtables 240.
+ * ------------------------------------------------------------------------ */
+
+/* replaces initReversibleGTR for AA_DATA/GTR (examl/models.c:3495 ->
+ * initGeneric with bitVectorAA); rates190 = upper-triangle exchangeabilities
+ * (e.g. the LG model, examl_amd/data/lg_model.npz). */
+void examl_host_init_gtr_aa(const double *frequencies, const double *rates190,
+                            double *EIGN, double *EV, double *EI,
+                            double *tipVector);
+
+void examl_host_core_dtables_prot(const double *EIGN,
+                                  const double *gammaRates, double lz,
+                                  double *out240);
+
+int examl_hip_newview_prot_gamma(int tipCase, const double *dev_x1,
+                                 const double *dev_x2, double *dev_x3,
+                                 const double *dev_EV,
+                                 const double *dev_tipVector,
+                                 const unsigned char *dev_tipX1,
+                                 const unsigned char *dev_tipX2, long n,
+                                 const double *dev_left,
+                                 const double *dev_right, const int *dev_wgt,
+                                 unsigned int *dev_scalerInc, void *stream);
+
+int examl_hip_evaluate_prot_gamma(const int *dev_wgt, const double *dev_x1,
+                                  const double *dev_x2,
+                                  const double *dev_tipVector,
+                                  const unsigned char *dev_tipX1, long n,
+                                  const double *dev_diag,
+                                  const unsigned int *dev_gsP,
+                                  const unsigned int *dev_gsQ,
+                                  double log_minlik, double *dev_lnl,
+                                  void *stream);
+
+int examl_hip_sum_prot_gamma(int tipCase, double *dev_sum,
+                             const double *dev_x1, const double *dev_x2,
+                             const double *dev_tipVector,
+                             const unsigned char *dev_tipX1,
+                             const unsigned char *dev_tipX2, long n,
+                             void *stream);
+
+int examl_hip_core_prot_gamma(long n, const double *dev_sum,
+                              const double *dev_dtables, const int *dev_wgt,
+                              double *dev_out2, void *stream);
+
+int examl_hip_newview_traversal_prot_gamma(
+    const examl_hip_trav_entry *ops, int numOps, const double *EIGN,
+    const double *EI, const double *gammaRates, const double *dev_EV,
+    const double *dev_tipVector, double *dev_clv, long clvStrideDoubles,
+    const unsigned char *dev_tips, long tipStrideBytes, const int *dev_wgt,
+    long n, unsigned int *dev_scalers, unsigned int *dev_inc,
+    double *dev_pbuf, void *stream);
+
+int examl_hip_evaluate_root_prot_gamma(
+    int rootTipCase, int pNumber, int qNumber, int x1Slot, int x2Slot,
+    int tipSlot, double z, const double *EIGN, const double *gammaRates,
+    const double *dev_tipVector, double *dev_clv, long clvStrideDoubles,
+    const unsigned char *dev_tips, long tipStrideBytes, const int *dev_wgt,
+    long n, const unsigned int *dev_scalers, double *dev_diag_scratch,
+    double *dev_lnl, void *stream);
+
+int examl_hip_sum_root_prot_gamma(int rootTipCase, int x1Slot, int x2Slot,
+                                  int tipSlot, int tipSlot2,
+                                  const double *dev_tipVector,
+                                  double *dev_clv, long clvStrideDoubles,
+                                  const unsigned char *dev_tips,
+                                  long tipStrideBytes, double *dev_sum,
+                                  long n, void *stream);
+
+int examl_hip_core_root_prot_gamma(long n, const double *dev_sum,
+                                   const double *EIGN,
+                                   const double *gammaRates, double lz,
+                                   const int *dev_wgt,
+                                   double *dev_dtab_scratch,
+                                   double *dev_out2, void *stream);
+
 #ifdef __cplusplus
 }
 #endif

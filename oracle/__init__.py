@@ -190,6 +190,107 @@ def init_gtr_dna(frequencies, rates):
     return EIGN, EV, EI, tipVector
 
 
+BIT_VECTOR_AA = np.array([1 << i for i in range(20)] + [12, 96, 0xFFFFF],
+                         dtype=np.uint32)  # globalVariables.h:95
+
+
+def init_gtr_aa(frequencies, rates190):
+    n = 20
+    EIGN = aligned(n)
+    EV = aligned(n * n)
+    EI = aligned(n * n)
+    tipVector = aligned(23 * n)
+    _orc.oracle_init_gtr(
+        ctypes.c_int(n), BIT_VECTOR_AA.ctypes.data_as(_c_u32),
+        ctypes.c_int(23), _dp(EIGN), _dp(EV), _dp(EI),
+        _dp(np.ascontiguousarray(frequencies, dtype=np.float64)),
+        _dp(np.ascontiguousarray(rates190, dtype=np.float64)), _dp(tipVector))
+    return EIGN, EV, EI, tipVector
+
+
+def newview_prot_gamma(tip_case, x1, x2, extEV, tipVector, tipX1, tipX2, n,
+                       left, right, wgt, lib=None):
+    lib = lib or _orc
+    fn = (lib.oracle_newview_prot_gamma if lib is _orc
+          else lib.newviewGTRGAMMAPROT_AVX)
+    x3 = aligned(n * 80)
+    inc = ctypes.c_int(0)
+    nullp = ctypes.cast(None, _c_d)
+    nullb = ctypes.cast(None, _c_u8)
+    fn(ctypes.c_int(tip_case),
+       _dp(x1) if x1 is not None else nullp,
+       _dp(x2) if x2 is not None else nullp,
+       _dp(x3), _dp(extEV), _dp(tipVector),
+       _u8p(tipX1) if tipX1 is not None else nullb,
+       _u8p(tipX2) if tipX2 is not None else nullb,
+       ctypes.c_int(n), _dp(left), _dp(right), _ip(wgt), ctypes.byref(inc))
+    return x3, inc.value
+
+
+def evaluate_prot_gamma(wgt, x1, x2, tipVector, tipX1, n, diag, lib=None):
+    lib = lib or _orc
+    fn = (lib.oracle_evaluate_prot_gamma if lib is _orc
+          else lib.evaluateGTRGAMMAPROT)
+    fn.restype = ctypes.c_double
+    nullp = ctypes.cast(None, _c_d)
+    nullb = ctypes.cast(None, _c_u8)
+    return fn(_ip(wgt),
+              _dp(x1) if x1 is not None else nullp,
+              _dp(x2), _dp(tipVector),
+              _u8p(tipX1) if tipX1 is not None else nullb,
+              ctypes.c_int(n), _dp(diag))
+
+
+def sum_prot_gamma(tip_case, x1, x2, tipVector, tipX1, tipX2, n, lib=None):
+    lib = lib or _orc
+    fn = lib.oracle_sum_prot_gamma if lib is _orc else lib.sumGAMMAPROT
+    sumtable = aligned(n * 80)
+    nullp = ctypes.cast(None, _c_d)
+    nullb = ctypes.cast(None, _c_u8)
+    fn(ctypes.c_int(tip_case), _dp(sumtable),
+       _dp(x1) if x1 is not None else nullp,
+       _dp(x2) if x2 is not None else nullp,
+       _dp(tipVector),
+       _u8p(tipX1) if tipX1 is not None else nullb,
+       _u8p(tipX2) if tipX2 is not None else nullb, ctypes.c_int(n))
+    return sumtable
+
+
+def core_prot_gamma(n, sumtable, EIGN, gammaRates, lz, wgt, lib=None):
+    lib = lib or _orc
+    d1 = ctypes.c_double(0.0)
+    d2 = ctypes.c_double(0.0)
+    if lib is _orc:
+        _orc.oracle_core_prot_gamma(
+            ctypes.c_int(n), _dp(sumtable), ctypes.byref(d1),
+            ctypes.byref(d2), _dp(EIGN), _dp(gammaRates),
+            ctypes.c_double(lz), _ip(wgt))
+    else:
+        # reference coreGTRGAMMAPROT(gammaRates, EIGN, sumtable, upper, wgt,
+        #                            ext_dlnLdlz, ext_d2lnLdlz2, lz) —
+        # makenewzGenericSpecial.c:2581 (note the different arg order)
+        lib.coreGTRGAMMAPROT(_dp(gammaRates), _dp(EIGN), _dp(sumtable),
+                             ctypes.c_int(n), _ip(wgt), ctypes.byref(d1),
+                             ctypes.byref(d2), ctypes.c_double(lz))
+    return d1.value, d2.value
+
+
+def ref_init_gtr_aa(frequencies, rates190):
+    assert _ref is not None
+    n = 20
+    EIGN = aligned(n)
+    EV = aligned(n * n)
+    EI = aligned(n * n)
+    tipVector = aligned(23 * n)
+    _ref.initGeneric(
+        ctypes.c_int(n), BIT_VECTOR_AA.ctypes.data_as(_c_u32),
+        ctypes.c_int(23), _dp(EIGN), _dp(EV), _dp(EI),
+        _dp(np.ascontiguousarray(frequencies, dtype=np.float64)),
+        _dp(np.ascontiguousarray(rates190, dtype=np.float64)), _dp(tipVector),
+        ctypes.c_int(0))
+    return EIGN, EV, EI, tipVector
+
+
 # ---------------------------------------------------------------------------
 # Reference (_ref) direct-call wrappers for golden generation/validation
 # ---------------------------------------------------------------------------

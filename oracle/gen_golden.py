@@ -119,6 +119,64 @@ def main():
     cases["model"] = np.bytes_(name.encode())
     np.savez_compressed(os.path.join(GOLDEN, "kernels_dna_gamma.npz"),
                         **cases)
+
+    # ---- protein (LG+GAMMA) ----------------------------------------------
+    lg = np.load(os.path.join(os.path.dirname(__file__), "..", "examl_amd",
+                              "data", "lg_model.npz"))
+    alpha_aa = 0.8
+    EIGN, EV, EI, tipVector = O.ref_init_gtr_aa(lg["frequencies"],
+                                                lg["rates190"])
+    g = O.ref_make_gamma_cats(alpha_aa)
+    pc = {
+        "alpha": np.float64(alpha_aa),
+        "EIGN": EIGN, "EV": EV, "EI": EI, "tipVector": tipVector,
+        "gammaRates": np.asarray(g),
+    }
+    z_q, z_r = 0.77, 0.21
+    lzq, lzr = np.log(z_q), np.log(z_r)
+    left, right = O.ref_make_p(lzq, lzr, g, EI, EIGN, 4, 20)
+    pc["z_q"], pc["z_r"] = np.float64(z_q), np.float64(z_r)
+    pc["left"], pc["right"] = left, right
+    n = 320
+    for tag, mag in [("norm", 1.0), ("tiny", 1e-80)]:
+        x1 = O.aligned(n * 80); x1[:] = rng.uniform(0.01, 1.0, n * 80) * mag
+        x2 = O.aligned(n * 80); x2[:] = rng.uniform(0.01, 1.0, n * 80) * mag
+        wgt = np.ascontiguousarray(rng.integers(1, 5, n), dtype=np.int32)
+        t1 = np.ascontiguousarray(rng.integers(1, 23, n), dtype=np.uint8)
+        t2 = np.ascontiguousarray(rng.integers(1, 23, n), dtype=np.uint8)
+        pc[f"{tag}_x1"], pc[f"{tag}_x2"] = x1, x2
+        pc[f"{tag}_wgt"], pc[f"{tag}_tipX1"], pc[f"{tag}_tipX2"] = wgt, t1, t2
+        for tc, a1, a2, u1, u2 in [
+            (O.TIP_TIP, None, None, t1, t2),
+            (O.TIP_INNER, None, x2, t1, None),
+            (O.INNER_INNER, x1, x2, None, None),
+        ]:
+            x3, inc = O.newview_prot_gamma(tc, a1, a2, EV, tipVector, u1, u2,
+                                           n, left, right, wgt, lib=ref)
+            pc[f"{tag}_newview_tc{tc}_x3"] = x3
+            pc[f"{tag}_newview_tc{tc}_inc"] = np.int64(inc)
+    x1, x2 = pc["norm_x1"], pc["norm_x2"]
+    wgt, t1, t2 = pc["norm_wgt"], pc["norm_tipX1"], pc["norm_tipX2"]
+    z_root = 0.69
+    diag = O.ref_calc_diagptable(z_root, 20, 4, g, EIGN)
+    pc["z_root"], pc["diag"] = np.float64(z_root), diag
+    pc["eval_II"] = np.float64(O.evaluate_prot_gamma(
+        wgt, x1, x2, tipVector, None, n, diag, lib=ref))
+    pc["eval_TIP"] = np.float64(O.evaluate_prot_gamma(
+        wgt, None, x2, tipVector, t1, n, diag, lib=ref))
+    lz = np.log(0.55)
+    pc["lz_core"] = np.float64(lz)
+    for tc, a1, a2, u1, u2 in [
+        (O.TIP_TIP, None, None, t1, t2),
+        (O.TIP_INNER, None, x2, t1, None),
+        (O.INNER_INNER, x1, x2, None, None),
+    ]:
+        st = O.sum_prot_gamma(tc, a1, a2, tipVector, u1, u2, n, lib=ref)
+        pc[f"sum_tc{tc}"] = st
+        d1, d2 = O.core_prot_gamma(n, st, EIGN, g, lz, wgt, lib=ref)
+        pc[f"core_tc{tc}_d1"] = np.float64(d1)
+        pc[f"core_tc{tc}_d2"] = np.float64(d2)
+    np.savez_compressed(os.path.join(GOLDEN, "kernels_prot_gamma.npz"), **pc)
     print("golden fixtures written to", GOLDEN)
 
 

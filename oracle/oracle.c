@@ -349,6 +349,196 @@ EXPORT void oracle_core_dna_gamma(int upper, const double *sumtable,
 }
 
 /* ==========================================================================
+ * Protein (20-state) GTRGAMMA kernels — span 80, tip codes 1..22.
+ * ==========================================================================*/
+
+/* 20-wide dot in the AVX lane order (avxLikelihood.c:1366-1383 etc.):
+ * four lane accumulators over five 4-chunks, then (t0+t1)+(t2+t3). */
+static inline double dot20_avx(const double *a, const double *b) {
+  double t0 = 0, t1 = 0, t2 = 0, t3 = 0;
+  int c;
+  for (c = 0; c < 20; c += 4) {
+    t0 += a[c] * b[c];
+    t1 += a[c + 1] * b[c + 1];
+    t2 += a[c + 2] * b[c + 2];
+    t3 += a[c + 3] * b[c + 3];
+  }
+  return (t0 + t1) + (t2 + t3);
+}
+
+/* --------------------------------------------------------------------------
+ * newview, protein GTRGAMMA.  Restates avxLikelihood.c:1312
+ * (newviewGTRGAMMAPROT_AVX): per (site, cat):
+ *   u1[l] = dot20(P_L[cat,l,:], x1'), u2[l] = dot20(P_R[cat,l,:], x2')
+ *   x3[s] = sum_l (u1[l]*u2[l]) * EV[l,s]   (sequential l)
+ * Tip operands via 23x80 ump tables; scaling over all 80 span entries,
+ * no scaling in TIP_TIP (avxLikelihood.c:1349-1456/1603-1630/1784-1797).
+ * left/right layout: [cat*400 + row*20 + col]; tipVector[code*20+s].
+ * ------------------------------------------------------------------------*/
+EXPORT void oracle_newview_prot_gamma(int tipCase, const double *x1,
+                                      const double *x2, double *x3,
+                                      const double *extEV,
+                                      const double *tipVector,
+                                      const unsigned char *tipX1,
+                                      const unsigned char *tipX2, int n,
+                                      const double *left, const double *right,
+                                      const int *wgt, int *scalerIncrement) {
+  int i, k, l, s;
+  int addScale = 0;
+  static double umpX1[23 * 80], umpX2[23 * 80];
+
+  if (tipCase != ORC_INNER_INNER) {
+    for (i = 0; i < 23; i++) {
+      const double *v = &tipVector[20 * i];
+      for (k = 0; k < 80; k++) {
+        umpX1[80 * i + k] = dot20_avx(v, &left[k * 20]);
+        if (tipCase == ORC_TIP_TIP)
+          umpX2[80 * i + k] = dot20_avx(v, &right[k * 20]);
+      }
+    }
+  }
+
+  for (i = 0; i < n; i++) {
+    double xv[80];
+    for (k = 0; k < 4; k++) { /* cat */
+      double acc[20];
+      for (s = 0; s < 20; s++) acc[s] = 0.0;
+      for (l = 0; l < 20; l++) {
+        double u1, u2;
+        if (tipCase == ORC_TIP_TIP) {
+          u1 = umpX1[80 * tipX1[i] + k * 20 + l];
+          u2 = umpX2[80 * tipX2[i] + k * 20 + l];
+        } else if (tipCase == ORC_TIP_INNER) {
+          u1 = umpX1[80 * tipX1[i] + k * 20 + l];
+          u2 = dot20_avx(&x2[80 * i + 20 * k], &right[k * 400 + l * 20]);
+        } else {
+          u1 = dot20_avx(&x1[80 * i + 20 * k], &left[k * 400 + l * 20]);
+          u2 = dot20_avx(&x2[80 * i + 20 * k], &right[k * 400 + l * 20]);
+        }
+        const double t = u1 * u2;
+        for (s = 0; s < 20; s++) acc[s] += t * extEV[20 * l + s];
+      }
+      for (s = 0; s < 20; s++) xv[k * 20 + s] = acc[s];
+    }
+    if (tipCase != ORC_TIP_TIP) {
+      int scale = 1;
+      for (l = 0; scale && l < 80; l++)
+        if (!(fabs(xv[l]) < ORC_MINLIKELIHOOD)) scale = 0;
+      if (scale) {
+        for (l = 0; l < 80; l++) xv[l] *= ORC_TWOTOTHE256;
+        addScale += wgt[i];
+      }
+    }
+    for (l = 0; l < 80; l++) x3[80 * i + l] = xv[l];
+  }
+  *scalerIncrement = addScale;
+}
+
+/* --------------------------------------------------------------------------
+ * evaluate, protein GTRGAMMA.  Restates evaluateGenericSpecial.c:1393
+ * (evaluateGTRGAMMAPROT, SSE even/odd lane split over all (cat,state)).
+ * ------------------------------------------------------------------------*/
+EXPORT double oracle_evaluate_prot_gamma(const int *wptr,
+                                         const double *x1_start,
+                                         const double *x2_start,
+                                         const double *tipVector,
+                                         const unsigned char *tipX1, int n,
+                                         const double *diagptable) {
+  double sum = 0.0;
+  int i, j, l;
+  for (i = 0; i < n; i++) {
+    double t0 = 0.0, t1 = 0.0;
+    for (j = 0; j < 4; j++) {
+      const double *d = &diagptable[j * 20];
+      const double *le = tipX1 ? &tipVector[20 * tipX1[i]]
+                               : &x1_start[80 * i + 20 * j];
+      const double *ri = &x2_start[80 * i + 20 * j];
+      for (l = 0; l < 20; l += 2) {
+        t0 += le[l] * ri[l] * d[l];
+        t1 += le[l + 1] * ri[l + 1] * d[l + 1];
+      }
+    }
+    sum += wptr[i] * log(0.25 * fabs(t0 + t1));
+  }
+  return sum;
+}
+
+/* sumGAMMAPROT — makenewzGenericSpecial.c:2083 (pure elementwise) */
+EXPORT void oracle_sum_prot_gamma(int tipCase, double *sumtable,
+                                  const double *x1_start,
+                                  const double *x2_start,
+                                  const double *tipVector,
+                                  const unsigned char *tipX1,
+                                  const unsigned char *tipX2, int n) {
+  int i, l, k;
+  for (i = 0; i < n; i++) {
+    for (l = 0; l < 4; l++) {
+      const double *le, *ri;
+      switch (tipCase) {
+      case ORC_TIP_TIP:
+        le = &tipVector[20 * tipX1[i]];
+        ri = &tipVector[20 * tipX2[i]];
+        break;
+      case ORC_TIP_INNER:
+        le = &tipVector[20 * tipX1[i]];
+        ri = &x2_start[80 * i + l * 20];
+        break;
+      default:
+        le = &x1_start[80 * i + l * 20];
+        ri = &x2_start[80 * i + l * 20];
+      }
+      for (k = 0; k < 20; k++)
+        sumtable[i * 80 + l * 20 + k] = le[k] * ri[k];
+    }
+  }
+}
+
+/* coreGTRGAMMAPROT — makenewzGenericSpecial.c:2581 (SSE even/odd order) */
+EXPORT void oracle_core_prot_gamma(int upper, const double *sumtable,
+                                   double *ext_dlnLdlz, double *ext_d2lnLdlz2,
+                                   const double *EIGN,
+                                   const double *gammaRates, double lz,
+                                   const int *wgt) {
+  double dlnLdlz = 0.0, d2lnLdlz2 = 0.0;
+  double d0[80], d1[80], d2[80];
+  int i, j, l;
+  for (i = 0; i < 4; i++) {
+    double ki = gammaRates[i], kisqr = ki * ki;
+    d0[i * 20] = 1.0;
+    d1[i * 20] = 0.0;
+    d2[i * 20] = 0.0;
+    for (l = 1; l < 20; l++) {
+      d0[i * 20 + l] = exp(EIGN[l] * ki * lz);
+      d1[i * 20 + l] = EIGN[l] * ki;
+      d2[i * 20 + l] = EIGN[l] * EIGN[l] * kisqr;
+    }
+  }
+  for (i = 0; i < upper; i++) {
+    const double *sum = &sumtable[i * 80];
+    double a0e = 0, a0o = 0, a1e = 0, a1o = 0, a2e = 0, a2o = 0;
+    for (j = 0; j < 4; j++) {
+      for (l = 0; l < 20; l += 2) {
+        const double te = d0[j * 20 + l] * sum[j * 20 + l];
+        const double to = d0[j * 20 + l + 1] * sum[j * 20 + l + 1];
+        a0e += te;
+        a0o += to;
+        a1e += te * d1[j * 20 + l];
+        a1o += to * d1[j * 20 + l + 1];
+        a2e += te * d2[j * 20 + l];
+        a2o += to * d2[j * 20 + l + 1];
+      }
+    }
+    const double inv_Li = 1.0 / fabs(a0e + a0o);
+    const double dlnLidlz = (a1e + a1o) * inv_Li;
+    const double d2lnLidlz2 = (a2e + a2o) * inv_Li;
+    dlnLdlz += wgt[i] * dlnLidlz;
+    d2lnLdlz2 += wgt[i] * (d2lnLidlz2 - dlnLidlz * dlnLidlz);
+  }
+  *ext_dlnLdlz = dlnLdlz;
+  *ext_d2lnLdlz2 = d2lnLdlz2;
+}
+
+/* ==========================================================================
  * Model preparation (host math, runs once per model-parameter change).
  * ==========================================================================*/
 
@@ -398,7 +588,9 @@ static double orc_IncompleteGamma(double x, double alpha,
     if (pn[5] != 0) {
       rn = pn[4] / pn[5];
       dif = fabs(gin - rn);
-      if (dif <= accurate && dif <= accurate * rn) { gin = rn; break; }
+      /* NOTE: on convergence the reference keeps the PREVIOUS gin
+       * (models.c:3648-3651: the l42 jump precedes the l34 update) */
+      if (dif <= accurate && dif <= accurate * rn) break;
       gin = rn;
     }
     for (i = 0; i < 4; i++) pn[i] = pn[i + 2];

@@ -31,8 +31,11 @@ def oracle_full_lnl(entries, root, tree, model, tips, wgt,
                     return_state=False):
     """Replay the product's traversal entries through the CPU oracle
     (newview per entry + recursive scalers + evaluate at the root),
-    restating evaluateGeneric end to end."""
+    restating evaluateGeneric end to end.  model.states picks DNA/protein."""
     EIGN, EV, EI, tipVector, g = _model_arrays(model)
+    st = model.states
+    nv = O.newview_dna_gamma if st == 4 else O.newview_prot_gamma
+    ev = O.evaluate_dna_gamma if st == 4 else O.evaluate_prot_gamma
     width = tips.shape[1]
     ntips = tips.shape[0] - 1
     clv = {}
@@ -42,49 +45,65 @@ def oracle_full_lnl(entries, root, tree, model, tips, wgt,
     for e in entries:
         qz = math.log(e.qz) if e.qz > O.ZMIN else math.log(O.ZMIN)
         rz = math.log(e.rz) if e.rz > O.ZMIN else math.log(O.ZMIN)
-        left, right = O.make_p(qz, rz, g, EI, EIGN, 4, 4)
+        left, right = O.make_p(qz, rz, g, EI, EIGN, 4, st)
         if e.tipCase == TIP_TIP:
-            x3, inc = O.newview_dna_gamma(
+            x3, inc = nv(
                 TIP_TIP, None, None, EV, tipVector,
                 np.ascontiguousarray(tips[e.x1Slot]),
                 np.ascontiguousarray(tips[e.x2Slot]), width, left, right, wgt)
         elif e.tipCase == TIP_INNER:
-            x3, inc = O.newview_dna_gamma(
+            x3, inc = nv(
                 TIP_INNER, None, clv[e.x2Slot], EV, tipVector,
                 np.ascontiguousarray(tips[e.x1Slot]), None, width, left,
                 right, wgt)
         else:
-            x3, inc = O.newview_dna_gamma(
+            x3, inc = nv(
                 INNER_INNER, clv[e.x1Slot], clv[e.x2Slot], EV, tipVector,
                 None, None, width, left, right, wgt)
         clv[e.x3Slot] = x3
         scalers[e.pNumber] = scalers[e.qNumber] + scalers[e.rNumber] + inc
 
     p, q, z = root
-    diag = O.calc_diagptable(z, 4, 4, g, EIGN)
+    diag = O.calc_diagptable(z, st, 4, g, EIGN)
     p_tip, q_tip = tree.is_tip(p), tree.is_tip(q)
     if q_tip:
-        lnl = O.evaluate_dna_gamma(wgt, None, clv[tree.clv_slot(p)],
-                                   tipVector, np.ascontiguousarray(tips[q]),
-                                   width, diag)
+        lnl = ev(wgt, None, clv[tree.clv_slot(p)],
+                 tipVector, np.ascontiguousarray(tips[q]), width, diag)
     elif p_tip:
-        lnl = O.evaluate_dna_gamma(wgt, None, clv[tree.clv_slot(q)],
-                                   tipVector, np.ascontiguousarray(tips[p]),
-                                   width, diag)
+        lnl = ev(wgt, None, clv[tree.clv_slot(q)],
+                 tipVector, np.ascontiguousarray(tips[p]), width, diag)
     else:
-        lnl = O.evaluate_dna_gamma(wgt, clv[tree.clv_slot(p)],
-                                   clv[tree.clv_slot(q)], tipVector, None,
-                                   width, diag)
+        lnl = ev(wgt, clv[tree.clv_slot(p)],
+                 clv[tree.clv_slot(q)], tipVector, None, width, diag)
     lnl += float(scalers[p] + scalers[q]) * math.log(O.MINLIKELIHOOD)
     if return_state:
         return lnl, clv, scalers
     return lnl
 
 
+def make_synthetic_aa(ntips, width, seed=42):
+    """Seeded synthetic protein alignment (codes 1..22)."""
+    rng = np.random.default_rng(seed)
+    tips = np.zeros((ntips + 1, width), dtype=np.uint8)
+    base = rng.integers(1, 21, width).astype(np.uint8)
+    for t in range(1, ntips + 1):
+        row = base.copy()
+        mut = rng.random(width) < 0.15
+        row[mut] = rng.integers(1, 21, int(mut.sum())).astype(np.uint8)
+        amb = rng.random(width) < 0.01
+        row[amb] = rng.integers(1, 23, int(amb.sum())).astype(np.uint8)
+        tips[t] = row
+    wgt = np.ones(width, dtype=np.int32)
+    return tips, wgt
+
+
 def oracle_makenewz(entries, root, tree, model, tips, wgt, z0, maxiter=64):
     """CPU restatement of topLevelMakenewz (numBranches=1) over the oracle
     sum/core kernels — the checker for DnaGammaEngine.makenewz."""
     EIGN, EV, EI, tipVector, g = _model_arrays(model)
+    states = model.states
+    sum_fn = O.sum_dna_gamma if states == 4 else O.sum_prot_gamma
+    core_fn = O.core_dna_gamma if states == 4 else O.core_prot_gamma
     width = tips.shape[1]
     _, clv, _ = oracle_full_lnl(entries, root, tree, model, tips, wgt,
                                 return_state=True)
@@ -92,21 +111,18 @@ def oracle_makenewz(entries, root, tree, model, tips, wgt, z0, maxiter=64):
     p_tip, q_tip = tree.is_tip(p), tree.is_tip(q)
     wgt = np.ascontiguousarray(wgt, dtype=np.int32)
     if p_tip and q_tip:
-        st = O.sum_dna_gamma(TIP_TIP, None, None, tipVector,
-                             np.ascontiguousarray(tips[p]),
-                             np.ascontiguousarray(tips[q]), width)
+        st = sum_fn(TIP_TIP, None, None, tipVector,
+                    np.ascontiguousarray(tips[p]),
+                    np.ascontiguousarray(tips[q]), width)
     elif q_tip:
-        st = O.sum_dna_gamma(TIP_INNER, None, clv[tree.clv_slot(p)],
-                             tipVector, np.ascontiguousarray(tips[q]), None,
-                             width)
+        st = sum_fn(TIP_INNER, None, clv[tree.clv_slot(p)],
+                    tipVector, np.ascontiguousarray(tips[q]), None, width)
     elif p_tip:
-        st = O.sum_dna_gamma(TIP_INNER, None, clv[tree.clv_slot(q)],
-                             tipVector, np.ascontiguousarray(tips[p]), None,
-                             width)
+        st = sum_fn(TIP_INNER, None, clv[tree.clv_slot(q)],
+                    tipVector, np.ascontiguousarray(tips[p]), None, width)
     else:
-        st = O.sum_dna_gamma(INNER_INNER, clv[tree.clv_slot(p)],
-                             clv[tree.clv_slot(q)], tipVector, None, None,
-                             width)
+        st = sum_fn(INNER_INNER, clv[tree.clv_slot(p)],
+                    clv[tree.clv_slot(q)], tipVector, None, None, width)
 
     z = float(z0)
     zprev, zstep = z, (1.0 - O.ZMAX) * z + O.ZMIN
@@ -118,7 +134,7 @@ def oracle_makenewz(entries, root, tree, model, tips, wgt, z0, maxiter=64):
             zstep = (1.0 - O.ZMAX) * z + O.ZMIN
         z = min(max(z, O.ZMIN), O.ZMAX)
         lz = math.log(z)
-        dlnL, d2lnL = O.core_dna_gamma(width, st, EIGN, g, lz, wgt)
+        dlnL, d2lnL = core_fn(width, st, EIGN, g, lz, wgt)
         if (d2lnL >= 0.0) and (z < O.ZMAX):
             zprev = z = 0.37 * z + 0.63
             continue
