@@ -87,6 +87,14 @@ class DnaGammaEngine:
         if self.device.type == "cuda":
             torch.cuda.synchronize(self.device)
 
+    def upload_model(self):
+        """Re-upload EV/tipVector after a model-parameter change
+        (changeModelParameters -> initReversibleGTR,
+        optimizeModel.c:419-449).  EIGN/EI/gammaRates are host-side inputs
+        of each call and need no upload."""
+        self.d_EV.copy_(torch.from_numpy(self.model.EV))
+        self.d_tipVector.copy_(torch.from_numpy(self.model.tipVector))
+
     # -- newviewIterative ---------------------------------------------------
 
     def newview_traversal(self, entries):

@@ -41,6 +41,14 @@ class DnaGtrModel:
         self.alpha = float(alpha)
         lib().examl_host_make_gamma_cats(self.alpha, _dp(self.gammaRates), 4)
 
+    def reinit(self):
+        """Recompute the eigendecomposition after mutating rates6 or
+        frequencies (the initReversibleGTR call of changeModelParameters,
+        optimizeModel.c:424/448)."""
+        lib().examl_host_init_gtr_dna(_dp(self.frequencies), _dp(self.rates6),
+                                      _dp(self.EIGN), _dp(self.EV),
+                                      _dp(self.EI), _dp(self.tipVector))
+
     @staticmethod
     def jukes_cantor(alpha=1.0):
         return DnaGtrModel([0.25] * 4, [1.0] * 6, alpha)

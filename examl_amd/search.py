@@ -1,0 +1,634 @@
+"""The search/optimizer layer (ExaML's L3) restated over the C-ABI engines:
+branch-length smoothing (searchAlgo.c: update/smooth/smoothTree/
+treeEvaluate) and model-parameter optimization (optimizeModel.c:
+brakGeneric/brentGeneric/optParamGeneric/optRates/optAlphas/optBaseFreqs/
+modOpt), faithful to the reference's control flow so that the -f E
+(TREE_EVALUATION) pipeline reproduces the reference's final lnL on the
+same inputs.
+
+Backend-agnostic: drives any per-partition engine object exposing
+newview_traversal / evaluate_root / sum_root / core_derivs / upload_model
+and a mutable .model — the HIP DnaGammaEngine or the test-only oracle
+engine."""
+
+import math
+
+import numpy as np
+
+from . import INNER_INNER, TIP_INNER, TIP_TIP, TravEntry, ZMIN, ZMAX
+
+# constants — examl/axml.h:89-95,164-197, optimizeModel.c:45-49
+SMOOTHINGS = 32
+NEWZPERCYCLE = 1
+DELTAZ = 0.00001
+ALPHA_MIN, ALPHA_MAX = 0.02, 1000.0
+RATE_MIN, RATE_MAX = 0.0000001, 1000000.0
+FREQ_MIN = 0.001
+ITMAX = 100
+BRENT_ZEPS = 1.0e-5
+BRENT_CGOLD = 0.3819660
+MNBRAK_GOLD = 1.618034
+MNBRAK_TINY = 1.0e-20
+MNBRAK_GLIMIT = 100.0
+UNLIKELY = -1.0e300
+
+RATE_F, ALPHA_F, FREQ_F = 0, 1, 2
+
+
+def _sign(a, b):
+    return abs(a) if b > 0.0 else -abs(a)
+
+
+class TreeSearch:
+    """One tree + one engine per partition (joint branch lengths,
+    numBranches=1)."""
+
+    def __init__(self, tree, engines, opt_freq_flags=None):
+        self.tree = tree
+        self.engines = engines
+        self.M = len(engines)
+        self.execute_model = [True] * self.M
+        self.per_partition_lnl = [0.0] * self.M
+        self.likelihood = None
+        self.oriented = {}  # inner node -> the parent neighbor its CLV faces
+        self.start = 1  # tr->start = tr->nodep[1]
+        self.partition_smoothed = True
+        self.partition_converged = False
+        self.opt_freq_flags = opt_freq_flags or [False] * self.M
+        # freqExponents state (models.c:4227: init 0.0)
+        self.freq_exponents = [np.zeros(e.model.states) for e in engines]
+
+    # ------------------------------------------------------------------
+    # traversal construction (computeTraversalInfo,
+    # newviewGenericSpecial.c:691)
+    # ------------------------------------------------------------------
+
+    def _children(self, p, parent):
+        t = self.tree
+        nbrs = [w for w in t.adj[p] if w != parent]
+        assert len(nbrs) == 2
+        q, r = nbrs
+        return q, r
+
+    def _collect(self, p, parent, partial, out):
+        t = self.tree
+        if t.is_tip(p):
+            return
+        q, r = self._children(p, parent)
+        q_tip, r_tip = t.is_tip(q), t.is_tip(r)
+        if q_tip and r_tip:
+            tc = TIP_TIP
+        elif q_tip or r_tip:
+            if r_tip:  # tip data stored for q (newviewGenericSpecial.c:744)
+                q, r = r, q
+            if self.oriented.get(r) != p or not partial:
+                self._collect(r, p, partial, out)
+            tc = TIP_INNER
+        else:
+            if self.oriented.get(q) != p or not partial:
+                self._collect(q, p, partial, out)
+            if self.oriented.get(r) != p or not partial:
+                self._collect(r, p, partial, out)
+            tc = INNER_INNER
+        e = TravEntry()
+        e.tipCase = tc
+        e.pNumber, e.qNumber, e.rNumber = p, q, r
+        e.qz, e.rz = t.get_z(p, q), t.get_z(p, r)
+        e.x3Slot = t.clv_slot(p)
+        e.x1Slot = q if t.is_tip(q) else t.clv_slot(q)
+        e.x2Slot = r if t.is_tip(r) else t.clv_slot(r)
+        out.append(e)
+        self.oriented[p] = parent
+
+    def _run(self, entries):
+        if not entries:
+            return
+        for m, eng in enumerate(self.engines):
+            if self.execute_model[m]:
+                eng.newview_traversal(entries)
+
+    # ------------------------------------------------------------------
+    # L1 entry points over the engines
+    # ------------------------------------------------------------------
+
+    def newview_generic(self, p, parent):
+        """newviewGeneric(tr, p, FALSE) with p->back == parent
+        (newviewGenericSpecial.c:1523)."""
+        out = []
+        self._collect(p, parent, True, out)
+        self._run(out)
+
+    def evaluate_generic(self, full=True, p=None):
+        """evaluateGeneric(tr, p (default tr->start), fullTraversal)
+        (evaluateGenericSpecial.c:897)."""
+        t = self.tree
+        if p is None:
+            p = self.start
+        q = next(iter(t.adj[p]))  # p->back
+        out = []
+        self._collect(p, q, not full, out)
+        self._collect(q, p, not full, out)
+        self._run(out)
+        z = t.get_z(p, q)
+        for m, eng in enumerate(self.engines):
+            if self.execute_model[m]:
+                self.per_partition_lnl[m] = float(
+                    eng.evaluate_root(t, p, q, z))
+        self.likelihood = sum(self.per_partition_lnl)
+        return self.likelihood
+
+    def makenewz_generic(self, p, q, z0, maxiter):
+        """makenewzGeneric (makenewzGenericSpecial.c:1355, mask=FALSE) +
+        topLevelMakenewz (:1133) for numBranches=1, derivatives summed over
+        executing partitions (execCore, :1075)."""
+        out = []
+        self._collect(p, q, True, out)
+        self._collect(q, p, True, out)
+        self._run(out)
+        for m, eng in enumerate(self.engines):
+            if self.execute_model[m]:
+                eng.sum_root(self.tree, p, q)
+
+        z = float(z0)
+        zprev = z
+        zstep = (1.0 - ZMAX) * z + ZMIN
+        curvat_ok = True
+        outer_converged = False
+        it = maxiter
+        while not outer_converged:
+            if curvat_ok:
+                curvat_ok = False
+                zprev = z
+                zstep = (1.0 - ZMAX) * z + ZMIN
+            z = min(max(z, ZMIN), ZMAX)
+            lz = math.log(z)
+            dlnL = d2lnL = 0.0
+            for m, eng in enumerate(self.engines):
+                if self.execute_model[m]:
+                    a, b = eng.core_derivs(lz)
+                    dlnL += a
+                    d2lnL += b
+            if (d2lnL >= 0.0) and (z < ZMAX):
+                zprev = z = 0.37 * z + 0.63
+                continue
+            curvat_ok = True
+            if d2lnL < 0.0:
+                tantmp = -dlnL / d2lnL
+                if tantmp < 100:
+                    z *= math.exp(tantmp)
+                    z = max(z, ZMIN)
+                    z = min(z, 0.25 * zprev + 0.75)
+                else:
+                    z = 0.25 * zprev + 0.75
+            z = min(z, ZMAX)
+            it -= 1
+            if abs(z - zprev) > zstep:
+                if it < -20:
+                    z = float(z0)
+                    outer_converged = True
+            else:
+                outer_converged = True
+        return z
+
+    # ------------------------------------------------------------------
+    # branch-length smoothing (searchAlgo.c:127-270,2635)
+    # ------------------------------------------------------------------
+
+    def update(self, p, parent):
+        """update(tr, p) with q = p->back = parent (searchAlgo.c:127)."""
+        t = self.tree
+        z0 = t.get_z(p, parent)
+        z = self.makenewz_generic(p, parent, z0, NEWZPERCYCLE)
+        if not self.partition_converged:
+            if abs(z - z0) > DELTAZ:
+                self.partition_smoothed = False
+            t.set_z(p, parent, z)
+
+    def smooth(self, p, parent):
+        """smooth(tr, p) with p->back == parent (searchAlgo.c:196)."""
+        t = self.tree
+        self.update(p, parent)
+        if not t.is_tip(p):
+            for w in self._children(p, parent):
+                self.smooth(w, p)
+            self.newview_generic(p, parent)
+
+    def smooth_tree(self, maxtimes):
+        """smoothTree (searchAlgo.c:237)."""
+        t = self.tree
+        p = self.start
+        self.partition_converged = False
+        while maxtimes > 0:
+            maxtimes -= 1
+            self.partition_smoothed = True
+            self.smooth(next(iter(t.adj[p])), p)  # smooth(tr, p->back)
+            # p == tr->start is a tip: the second descent is skipped
+            if self.partition_smoothed:  # allSmoothed
+                self.partition_converged = True
+                break
+        self.partition_converged = False
+
+    def tree_evaluate(self, smooth_factor):
+        """treeEvaluate (searchAlgo.c:2635)."""
+        self.smooth_tree(int(SMOOTHINGS * smooth_factor))
+        return self.evaluate_generic(full=True)
+
+    # ------------------------------------------------------------------
+    # model-parameter optimization (optimizeModel.c)
+    # ------------------------------------------------------------------
+
+    def _change_param(self, m, rate_number, value, which):
+        """changeModelParameters (optimizeModel.c:419)."""
+        eng = self.engines[m]
+        model = eng.model
+        if which == RATE_F:
+            # setRateModel (optimizeModel.c:78); DNA: position < 5
+            rates = model.rates6 if model.states == 4 else model.rates190
+            assert 0 <= rate_number < len(rates) - (1 if model.states == 4
+                                                    else 0)
+            assert RATE_MIN <= value <= RATE_MAX
+            rates[rate_number] = value
+            model.reinit()
+            eng.upload_model()
+        elif which == ALPHA_F:
+            model.set_alpha(value)
+        elif which == FREQ_F:
+            w = self.freq_exponents[m]
+            w[rate_number] = value
+            ew = np.exp(w)
+            model.frequencies[:] = ew / ew.sum()
+            model.reinit()
+            eng.upload_model()
+        else:
+            raise AssertionError(which)
+
+    def _evaluate_change(self, rate_number, values, converged, which,
+                         groups, valid):
+        """evaluateChange (optimizeModel.c:464): apply parameter probes,
+        mask converged groups and invalid partitions, full-traversal
+        evaluate, return -sum(perPartitionLH) per group."""
+        in_valid_group = set()
+        pos = 0
+        for gi, g in enumerate(groups):
+            if not valid[gi]:
+                continue
+            for m in g:
+                in_valid_group.add(m)
+                if converged[pos]:
+                    self.execute_model[m] = False
+                else:
+                    self._change_param(m, rate_number, values[pos], which)
+            pos += 1
+        for m in range(self.M):
+            if m not in in_valid_group:
+                self.execute_model[m] = False
+        self.evaluate_generic(full=True)
+        results = np.zeros(pos)
+        pos = 0
+        for gi, g in enumerate(groups):
+            if not valid[gi]:
+                continue
+            results[pos] = -sum(self.per_partition_lnl[m] for m in g)
+            pos += 1
+        self.execute_model = [True] * self.M
+        return results
+
+    def _brak(self, param, ax, bx, cx, fa, fb, fc, lim_inf, lim_sup,
+              rate_number, which, groups, valid):
+        """brakGeneric (optimizeModel.c:800), numpy over the valid groups."""
+        n = len(ax)
+        converged = np.zeros(n, dtype=bool)
+        state = np.zeros(n, dtype=int)
+        end_state = np.zeros(n, dtype=int)
+        u = np.zeros(n)
+        ulim = np.zeros(n)
+
+        np.clip(ax, lim_inf, lim_sup, out=ax)
+        param[:] = ax
+        fa[:] = self._evaluate_change(rate_number, param, converged, which,
+                                      groups, valid)
+        np.clip(bx, lim_inf, lim_sup, out=bx)
+        param[:] = bx
+        fb[:] = self._evaluate_change(rate_number, param, converged, which,
+                                      groups, valid)
+        for i in range(n):
+            if fb[i] > fa[i]:
+                ax[i], bx[i] = bx[i], ax[i]
+                fa[i], fb[i] = fb[i], fa[i]
+            cx[i] = bx[i] + MNBRAK_GOLD * (bx[i] - ax[i])
+            cx[i] = min(max(cx[i], lim_inf[i]), lim_sup[i])
+            param[i] = cx[i]
+        fc[:] = self._evaluate_change(rate_number, param, converged, which,
+                                      groups, valid)
+
+        while True:
+            if converged.all():
+                np.clip(ax, lim_inf, lim_sup, out=ax)
+                np.clip(bx, lim_inf, lim_sup, out=bx)
+                np.clip(cx, lim_inf, lim_sup, out=cx)
+                return
+            for i in range(n):
+                if converged[i]:
+                    continue
+                if state[i] == 0:
+                    end_state[i] = 0
+                    if not (fb[i] > fc[i]):
+                        converged[i] = True
+                    else:
+                        ax[i] = min(max(ax[i], lim_inf[i]), lim_sup[i])
+                        bx[i] = min(max(bx[i], lim_inf[i]), lim_sup[i])
+                        cx[i] = min(max(cx[i], lim_inf[i]), lim_sup[i])
+                        r_ = (bx[i] - ax[i]) * (fb[i] - fc[i])
+                        q_ = (bx[i] - cx[i]) * (fb[i] - fa[i])
+                        u[i] = bx[i] - ((bx[i] - cx[i]) * q_ -
+                                        (bx[i] - ax[i]) * r_) / \
+                            (2.0 * _sign(max(abs(q_ - r_), MNBRAK_TINY),
+                                         q_ - r_))
+                        ulim[i] = bx[i] + MNBRAK_GLIMIT * (cx[i] - bx[i])
+                        u[i] = min(max(u[i], lim_inf[i]), lim_sup[i])
+                        ulim[i] = min(max(ulim[i], lim_inf[i]), lim_sup[i])
+                        if (bx[i] - u[i]) * (u[i] - cx[i]) > 0.0:
+                            u[i] = min(max(u[i], lim_inf[i]), lim_sup[i])
+                            param[i] = u[i]
+                            end_state[i] = 1
+                        elif (cx[i] - u[i]) * (u[i] - ulim[i]) > 0.0:
+                            u[i] = min(max(u[i], lim_inf[i]), lim_sup[i])
+                            param[i] = u[i]
+                            end_state[i] = 2
+                        elif (u[i] - ulim[i]) * (ulim[i] - cx[i]) >= 0.0:
+                            u[i] = ulim[i]
+                            u[i] = min(max(u[i], lim_inf[i]), lim_sup[i])
+                            param[i] = u[i]
+                            end_state[i] = 0
+                        else:
+                            u[i] = cx[i] + MNBRAK_GOLD * (cx[i] - bx[i])
+                            u[i] = min(max(u[i], lim_inf[i]), lim_sup[i])
+                            param[i] = u[i]
+                            end_state[i] = 0
+                elif state[i] == 1:
+                    end_state[i] = 0
+                elif state[i] == 2:
+                    end_state[i] = 3
+                else:
+                    raise AssertionError
+            temp = self._evaluate_change(rate_number, param, converged,
+                                         which, groups, valid)
+            for i in range(n):
+                if converged[i]:
+                    continue
+                if end_state[i] == 0:
+                    fu = temp[i]
+                    ax[i], bx[i], cx[i] = bx[i], cx[i], u[i]
+                    fa[i], fb[i], fc[i] = fb[i], fc[i], fu
+                    state[i] = 0
+                elif end_state[i] == 1:
+                    fu = temp[i]
+                    if fu < fc[i]:
+                        ax[i] = bx[i]
+                        bx[i] = u[i]
+                        fa[i] = fb[i]
+                        fb[i] = fu
+                        converged[i] = True
+                    elif fu > fb[i]:
+                        cx[i] = u[i]
+                        fc[i] = fu
+                        converged[i] = True
+                    else:
+                        u[i] = cx[i] + MNBRAK_GOLD * (cx[i] - bx[i])
+                        u[i] = min(max(u[i], lim_inf[i]), lim_sup[i])
+                        param[i] = u[i]
+                        state[i] = 1
+                elif end_state[i] == 2:
+                    fu = temp[i]
+                    if fu < fc[i]:
+                        bx[i], cx[i], u[i] = cx[i], u[i], \
+                            cx[i] + MNBRAK_GOLD * (cx[i] - bx[i])
+                        fb[i], fc[i] = fc[i], fu
+                        state[i] = 2
+                    else:
+                        state[i] = 0
+                        ax[i], bx[i], cx[i] = bx[i], cx[i], u[i]
+                        fa[i], fb[i], fc[i] = fb[i], fc[i], fu
+                elif end_state[i] == 3:
+                    fb[i], fc[i], fu = fc[i], temp[i], temp[i]
+                    ax[i], bx[i], cx[i] = bx[i], cx[i], u[i]
+                    fa[i], fb[i], fc[i] = fb[i], fc[i], fu
+                    state[i] = 0
+                else:
+                    raise AssertionError
+
+    def _brent(self, ax, bx, cx, fb, tol, rate_number, which, groups, valid,
+               lim_inf, lim_sup):
+        """brentGeneric (optimizeModel.c:582); returns (xmin, result)."""
+        n = len(ax)
+        a = np.minimum(ax, cx)
+        b = np.maximum(ax, cx)
+        x = bx.copy()
+        w = bx.copy()
+        v = bx.copy()
+        fw = fb.copy()
+        fv = fb.copy()
+        fx = fb.copy()
+        d = np.zeros(n)
+        e = np.zeros(n)
+        u = np.zeros(n)
+        xmin = np.zeros(n)
+        result = np.zeros(n)
+        converged = np.zeros(n, dtype=bool)
+
+        for _ in range(1, ITMAX + 1):
+            if converged.all():
+                return xmin, result
+            for i in range(n):
+                if converged[i]:
+                    continue
+                xm = 0.5 * (a[i] + b[i])
+                tol1 = tol * abs(x[i]) + BRENT_ZEPS
+                tol2 = 2.0 * tol1
+                if abs(x[i] - xm) <= (tol2 - 0.5 * (b[i] - a[i])):
+                    result[i] = -fx[i]
+                    xmin[i] = x[i]
+                    converged[i] = True
+                    continue
+                if abs(e[i]) > tol1:
+                    r_ = (x[i] - w[i]) * (fx[i] - fv[i])
+                    q_ = (x[i] - v[i]) * (fx[i] - fw[i])
+                    p_ = (x[i] - v[i]) * q_ - (x[i] - w[i]) * r_
+                    q_ = 2.0 * (q_ - r_)
+                    if q_ > 0.0:
+                        p_ = -p_
+                    q_ = abs(q_)
+                    etemp = e[i]
+                    e[i] = d[i]
+                    if (abs(p_) >= abs(0.5 * q_ * etemp)
+                            or p_ <= q_ * (a[i] - x[i])
+                            or p_ >= q_ * (b[i] - x[i])):
+                        e[i] = a[i] - x[i] if x[i] >= xm else b[i] - x[i]
+                        d[i] = BRENT_CGOLD * e[i]
+                    else:
+                        d[i] = p_ / q_
+                        u[i] = x[i] + d[i]
+                        if u[i] - a[i] < tol2 or b[i] - u[i] < tol2:
+                            d[i] = _sign(tol1, xm - x[i])
+                else:
+                    e[i] = a[i] - x[i] if x[i] >= xm else b[i] - x[i]
+                    d[i] = BRENT_CGOLD * e[i]
+                u[i] = (x[i] + d[i]) if abs(d[i]) >= tol1 \
+                    else (x[i] + _sign(tol1, d[i]))
+            fu = self._evaluate_change(rate_number, u, converged, which,
+                                       groups, valid)
+            for i in range(n):
+                if converged[i]:
+                    continue
+                if fu[i] <= fx[i]:
+                    if u[i] >= x[i]:
+                        a[i] = x[i]
+                    else:
+                        b[i] = x[i]
+                    v[i], w[i], x[i] = w[i], x[i], u[i]
+                    fv[i], fw[i], fx[i] = fw[i], fx[i], fu[i]
+                else:
+                    if u[i] < x[i]:
+                        a[i] = u[i]
+                    else:
+                        b[i] = u[i]
+                    if fu[i] <= fw[i] or w[i] == x[i]:
+                        v[i] = w[i]
+                        w[i] = u[i]
+                        fv[i] = fw[i]
+                        fw[i] = fu[i]
+                    elif fu[i] <= fv[i] or v[i] == x[i] or v[i] == w[i]:
+                        v[i] = u[i]
+                        fv[i] = fu[i]
+        raise AssertionError("Too many iterations in BRENT")
+
+    def _opt_param_generic(self, groups, valid, rate_number, lim_inf_s,
+                           lim_sup_s, which, model_epsilon):
+        """optParamGeneric (optimizeModel.c:1283)."""
+        self.evaluate_generic(full=True)
+        vg = [g for gi, g in enumerate(groups) if valid[gi]]
+        n = len(vg)
+        if n == 0:
+            return
+        start_values = np.zeros(n)
+        start_lh = np.zeros(n)
+        lim_inf = np.zeros(n)
+        lim_sup = np.zeros(n)
+        for pos, g in enumerate(vg):
+            for m in g:
+                start_lh[pos] += self.per_partition_lnl[m]
+                model = self.engines[m].model
+                if which == ALPHA_F:
+                    lim_inf[pos], lim_sup[pos] = lim_inf_s, lim_sup_s
+                    start_values[pos] = model.alpha
+                elif which == RATE_F:
+                    lim_inf[pos], lim_sup[pos] = lim_inf_s, lim_sup_s
+                    rates = (model.rates6 if model.states == 4
+                             else model.rates190)
+                    start_values[pos] = rates[rate_number]
+                elif which == FREQ_F:
+                    lim_inf[pos] = self._min_freq(m, rate_number, lim_inf_s)
+                    lim_sup[pos] = self._max_freq(m, rate_number, lim_sup_s)
+                    start_values[pos] = self.freq_exponents[m][rate_number]
+        a = np.clip(start_values + 0.1, lim_inf, lim_sup)
+        b = np.clip(start_values - 0.1, lim_inf, lim_sup)
+        c = np.zeros(n)
+        fa = np.zeros(n)
+        fb = np.zeros(n)
+        fc = np.zeros(n)
+        param = np.zeros(n)
+        self._brak(param, a, b, c, fa, fb, fc, lim_inf, lim_sup, rate_number,
+                   which, groups, valid)
+        xmin, end_lh = self._brent(a, b, c, fb, model_epsilon, rate_number,
+                                   which, groups, valid, lim_inf, lim_sup)
+        for pos, g in enumerate(vg):
+            val = start_values[pos] if start_lh[pos] > end_lh[pos] \
+                else xmin[pos]
+            for m in g:
+                self._change_param(m, rate_number, val, which)
+
+    def _min_freq(self, m, which_freq, absolute_min):
+        """minFreq (optimizeModel.c:1222)."""
+        w = self.freq_exponents[m]
+        c = sum(math.exp(w[i]) for i in range(len(w)) if i != which_freq)
+        return max(math.log(FREQ_MIN) + math.log(c) - math.log(1.0 -
+                                                               FREQ_MIN),
+                   absolute_min)
+
+    def _max_freq(self, m, which_freq, absolute_max):
+        """maxFreq (optimizeModel.c:1248): symmetric cap so one frequency
+        cannot exceed 1 - (states-1)*FREQ_MIN."""
+        w = self.freq_exponents[m]
+        states = len(w)
+        c = sum(math.exp(w[i]) for i in range(states) if i != which_freq)
+        return min(math.log(1.0 - (states - 1) * FREQ_MIN) + math.log(c)
+                   - math.log((states - 1) * FREQ_MIN), absolute_max)
+
+    def opt_rates_generic(self, model_epsilon):
+        """optRatesGeneric + optRates (optimizeModel.c:1634/1603):
+        per-partition (unlinked) groups; DNA rate numbers 0..4."""
+        groups = [[m] for m in range(self.M)]
+        dna = [self.engines[m].model.states == 4 for m in range(self.M)]
+        if any(dna):
+            n_rates = (4 * 4 - 4) // 2 - 1
+            for rn in range(n_rates):
+                self._opt_param_generic(groups, dna, rn, RATE_MIN, RATE_MAX,
+                                        RATE_F, model_epsilon)
+        # AA GTR rate optimization only applies to protModels==GTR
+        # partitions (AAisGTR, optimizeModel.c:1614) — LG et al. are fixed.
+
+    def opt_alphas_generic(self, model_epsilon):
+        """optAlphasGeneric (optimizeModel.c:1136)."""
+        groups = [[m] for m in range(self.M)]
+        valid = [True] * self.M
+        self._opt_param_generic(groups, valid, -1, ALPHA_MIN, ALPHA_MAX,
+                                ALPHA_F, model_epsilon)
+
+    def opt_base_freqs(self, model_epsilon):
+        """optBaseFreqs + optFreqs (optimizeModel.c:1501/1594)."""
+        groups = [[m] for m in range(self.M)]
+        dna_valid = [self.engines[m].model.states == 4
+                     and self.opt_freq_flags[m] for m in range(self.M)]
+        if any(dna_valid):
+            for rn in range(4):
+                self._opt_param_generic(groups, dna_valid, rn, -1000000.0,
+                                        200.0, FREQ_F, model_epsilon)
+        aa_valid = [self.engines[m].model.states == 20
+                    and self.opt_freq_flags[m] for m in range(self.M)]
+        if any(aa_valid):
+            for rn in range(20):
+                self._opt_param_generic(groups, aa_valid, rn, -1000000.0,
+                                        200.0, FREQ_F, model_epsilon)
+
+    def mod_opt(self, likelihood_epsilon=0.1, model_epsilon=0.0001,
+                log=None):
+        """modOpt (optimizeModel.c:2963) for the GAMMA model."""
+        self.evaluate_generic(full=True)
+        while True:
+            current = self.likelihood
+            self.opt_rates_generic(model_epsilon)
+            self.evaluate_generic(full=True)
+            self.tree_evaluate(0.0625)
+            self.evaluate_generic(full=True)
+            self.opt_base_freqs(model_epsilon)
+            self.evaluate_generic(full=True)
+            self.tree_evaluate(0.0625)
+            self.opt_alphas_generic(model_epsilon)
+            self.evaluate_generic(full=True)
+            self.tree_evaluate(0.1)
+            if log:
+                log(f"modOpt pass: {current:.6f} -> {self.likelihood:.6f}")
+            if abs(current - self.likelihood) <= likelihood_epsilon:
+                break
+        return self.likelihood
+
+    def tree_evaluation_mode(self, log=None):
+        """The -f E (slow TREE_EVALUATION) flow for one tree
+        (axml.c:2316-2331): evaluate, treeEvaluate(1), modOpt(0.1)."""
+        self.evaluate_generic(full=True)
+        if log:
+            log(f"initial lnL = {self.likelihood:.6f}")
+        self.tree_evaluate(1.0)
+        if log:
+            log(f"after treeEvaluate = {self.likelihood:.6f}")
+        return self.mod_opt(0.1, log=log)

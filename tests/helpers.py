@@ -97,6 +97,109 @@ def make_synthetic_aa(ntips, width, seed=42):
     return tips, wgt
 
 
+class OracleEngine:
+    """CPU engine with the SAME interface as DnaGammaEngine, executing the
+    oracle kernels — lets examl_amd.search.TreeSearch run end to end on the
+    CPU restatement (the checker for the GPU search path, and the vehicle
+    for the testData/49 -f E parity anchor)."""
+
+    def __init__(self, tips, wgt, model):
+        self.model = model
+        self.ntips = tips.shape[0] - 1
+        self.width = tips.shape[1]
+        self.tips = tips
+        self.wgt = np.ascontiguousarray(wgt, dtype=np.int32)
+        self.clv = {}
+        self.scalers = np.zeros(2 * self.ntips, dtype=np.int64)
+        self.sumtable = None
+        self._nv = O.newview_dna_gamma if model.states == 4 \
+            else O.newview_prot_gamma
+        self._ev = O.evaluate_dna_gamma if model.states == 4 \
+            else O.evaluate_prot_gamma
+        self._sum = O.sum_dna_gamma if model.states == 4 else O.sum_prot_gamma
+        self._core = O.core_dna_gamma if model.states == 4 \
+            else O.core_prot_gamma
+
+    def upload_model(self):
+        pass  # oracle reads model arrays directly per call
+
+    def _arrays(self):
+        return _model_arrays(self.model)
+
+    def newview_traversal(self, entries):
+        EIGN, EV, EI, tipVector, g = self._arrays()
+        st = self.model.states
+        for e in entries:
+            qz = math.log(e.qz) if e.qz > O.ZMIN else math.log(O.ZMIN)
+            rz = math.log(e.rz) if e.rz > O.ZMIN else math.log(O.ZMIN)
+            left, right = O.make_p(qz, rz, g, EI, EIGN, 4, st)
+            if e.tipCase == TIP_TIP:
+                x3, inc = self._nv(
+                    TIP_TIP, None, None, EV, tipVector,
+                    np.ascontiguousarray(self.tips[e.x1Slot]),
+                    np.ascontiguousarray(self.tips[e.x2Slot]), self.width,
+                    left, right, self.wgt)
+            elif e.tipCase == TIP_INNER:
+                x3, inc = self._nv(
+                    TIP_INNER, None, self.clv[e.x2Slot], EV, tipVector,
+                    np.ascontiguousarray(self.tips[e.x1Slot]), None,
+                    self.width, left, right, self.wgt)
+            else:
+                x3, inc = self._nv(
+                    INNER_INNER, self.clv[e.x1Slot], self.clv[e.x2Slot], EV,
+                    tipVector, None, None, self.width, left, right, self.wgt)
+            self.clv[e.x3Slot] = x3
+            self.scalers[e.pNumber] = (self.scalers[e.qNumber] +
+                                       self.scalers[e.rNumber] + inc)
+
+    def evaluate_root(self, tree, p, q, z):
+        EIGN, EV, EI, tipVector, g = self._arrays()
+        st = self.model.states
+        diag = O.calc_diagptable(z, st, 4, g, EIGN)
+        p_tip, q_tip = tree.is_tip(p), tree.is_tip(q)
+        if q_tip:
+            lnl = self._ev(self.wgt, None, self.clv[tree.clv_slot(p)],
+                           tipVector, np.ascontiguousarray(self.tips[q]),
+                           self.width, diag)
+        elif p_tip:
+            lnl = self._ev(self.wgt, None, self.clv[tree.clv_slot(q)],
+                           tipVector, np.ascontiguousarray(self.tips[p]),
+                           self.width, diag)
+        else:
+            lnl = self._ev(self.wgt, self.clv[tree.clv_slot(p)],
+                           self.clv[tree.clv_slot(q)], tipVector, None,
+                           self.width, diag)
+        lnl += float(self.scalers[p] + self.scalers[q]) * \
+            math.log(O.MINLIKELIHOOD)
+        return lnl
+
+    def sum_root(self, tree, p, q):
+        EIGN, EV, EI, tipVector, g = self._arrays()
+        p_tip, q_tip = tree.is_tip(p), tree.is_tip(q)
+        if p_tip and q_tip:
+            self.sumtable = self._sum(
+                TIP_TIP, None, None, tipVector,
+                np.ascontiguousarray(self.tips[p]),
+                np.ascontiguousarray(self.tips[q]), self.width)
+        elif q_tip:
+            self.sumtable = self._sum(
+                TIP_INNER, None, self.clv[tree.clv_slot(p)], tipVector,
+                np.ascontiguousarray(self.tips[q]), None, self.width)
+        elif p_tip:
+            self.sumtable = self._sum(
+                TIP_INNER, None, self.clv[tree.clv_slot(q)], tipVector,
+                np.ascontiguousarray(self.tips[p]), None, self.width)
+        else:
+            self.sumtable = self._sum(
+                INNER_INNER, self.clv[tree.clv_slot(p)],
+                self.clv[tree.clv_slot(q)], tipVector, None, None,
+                self.width)
+
+    def core_derivs(self, lz):
+        EIGN, EV, EI, tipVector, g = self._arrays()
+        return self._core(self.width, self.sumtable, EIGN, g, lz, self.wgt)
+
+
 def oracle_makenewz(entries, root, tree, model, tips, wgt, z0, maxiter=64):
     """CPU restatement of topLevelMakenewz (numBranches=1) over the oracle
     sum/core kernels — the checker for DnaGammaEngine.makenewz."""
