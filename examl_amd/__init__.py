@@ -92,6 +92,8 @@ def _bind(lib):
     lib.examl_hip_profile_enable.argtypes = [i]
     lib.examl_hip_profile_reset.argtypes = []
     lib.examl_hip_profile_get.argtypes = [p, p]
+    lib.examl_hip_use_graphs.argtypes = [i]
+    lib.examl_hip_graphs_clear.argtypes = []
     return lib
 
 

@@ -890,7 +890,258 @@ extern "C" void examl_hip_profile_get(double *ms_by_tc, long *cnt_by_tc) {
   }
 }
 
+/* --- hipGraph cache for traversal replays ----------------------------------
+ * A full traversal is a fixed sequence of ~n-2 kernel launches whose
+ * ARGUMENTS only change when the traversal shape (slots/tipCases) changes;
+ * branch lengths flow through the P-matrix CONTENT, which is re-uploaded by
+ * the captured memcpy node from the same pinned host buffer each replay.
+ * Search-mode Brent probes re-evaluate the identical full traversal dozens
+ * of times, and the bench repeats one shape every step — both hit the cache
+ * and replay with one hipGraphLaunch instead of ~50 API calls.
+ * Bypassed while HIP-event profiling is enabled (events are not captured).
+ */
+struct TravGraph {
+  unsigned long long key;
+  hipGraphExec_t exec;
+};
+static thread_local std::vector<TravGraph> g_graphs;
+static bool g_use_graphs = true;
+
+extern "C" void examl_hip_use_graphs(int on) { g_use_graphs = on != 0; }
+
+extern "C" void examl_hip_graphs_clear(void) {
+  for (auto &g : g_graphs) hipGraphExecDestroy(g.exec);
+  g_graphs.clear();
+}
+
+static unsigned long long trav_key(const examl_hip_trav_entry *ops,
+                                   int numOps, long n, const void *clv,
+                                   const void *tips, const void *pbuf,
+                                   const void *ev, const void *tipvec,
+                                   const void *wgt, const void *scalers,
+                                   const void *inc, const void *stream,
+                                   int states) {
+  unsigned long long h = 1469598103934665603ULL;
+  auto mix = [&h](unsigned long long v) {
+    h ^= v;
+    h *= 1099511628211ULL;
+  };
+  mix((unsigned long long)numOps);
+  mix((unsigned long long)n);
+  mix((unsigned long long)(uintptr_t)clv);
+  mix((unsigned long long)(uintptr_t)tips);
+  mix((unsigned long long)(uintptr_t)pbuf);
+  mix((unsigned long long)(uintptr_t)ev);
+  mix((unsigned long long)(uintptr_t)tipvec);
+  mix((unsigned long long)(uintptr_t)wgt);
+  mix((unsigned long long)(uintptr_t)scalers);
+  mix((unsigned long long)(uintptr_t)inc);
+  mix((unsigned long long)(uintptr_t)stream);
+  mix((unsigned long long)states);
+  for (int e = 0; e < numOps; e++) {
+    mix(((unsigned long long)ops[e].tipCase << 48) ^
+        ((unsigned long long)(unsigned)ops[e].pNumber << 32) ^
+        ((unsigned long long)(unsigned)ops[e].x1Slot << 16) ^
+        (unsigned long long)(unsigned)ops[e].x2Slot);
+    mix(((unsigned long long)(unsigned)ops[e].qNumber << 32) ^
+        ((unsigned long long)(unsigned)ops[e].rNumber << 16) ^
+        (unsigned long long)(unsigned)ops[e].x3Slot);
+  }
+  return h;
+}
+
+static hipGraphExec_t trav_graph_find(unsigned long long key) {
+  for (auto &g : g_graphs)
+    if (g.key == key) return g.exec;
+  return nullptr;
+}
+
+static void trav_graph_store(unsigned long long key, hipGraphExec_t exec) {
+  if (g_graphs.size() >= 32) {
+    for (auto &g : g_graphs) hipGraphExecDestroy(g.exec);
+    g_graphs.clear();
+  }
+  g_graphs.push_back({key, exec});
+}
+
+/* pinned host staging for the P blocks (stable address: the captured
+ * memcpy node re-reads it on every replay) */
+static thread_local double *g_hostP = nullptr;
+static thread_local size_t g_hostPCap = 0;
+
+static double *hostP_get(size_t doubles) {
+  if (doubles > g_hostPCap) {
+    if (g_hostP) hipHostFree(g_hostP);
+    size_t cap = doubles * 2;
+    if (hipHostMalloc((void **)&g_hostP, cap * sizeof(double)) !=
+        hipSuccess) {
+      g_hostP = (double *)malloc(cap * sizeof(double));
+    }
+    g_hostPCap = cap;
+  }
+  return g_hostP;
+}
+
 /* --- batched traversal (newviewIterative body) --------------------------- */
+
+template <int STATES>
+static int traversal_impl(const examl_hip_trav_entry *ops, int numOps,
+                          const double *EIGN, const double *EI,
+                          const double *gammaRates, const double *dev_EV,
+                          const double *dev_tipVec, double *dev_clv,
+                          long clvStride, const unsigned char *dev_tips,
+                          long tipStride, const int *dev_wgt, long n,
+                          unsigned int *dev_scalers, unsigned int *dev_inc,
+                          double *dev_pbuf, hipStream_t s) {
+  constexpr int PBLK = 8 * STATES * STATES; /* left|right, 4 cats */
+  double *hostP = hostP_get((size_t)numOps * PBLK);
+
+  /* 1. all P-matrix pairs on the host (newviewGenericSpecial.c:982-1044) */
+  for (int e = 0; e < numOps; e++) {
+    double qz = ops[e].qz, rz = ops[e].rz;
+    qz = (qz > ZMIN) ? log(qz) : log(ZMIN);
+    rz = (rz > ZMIN) ? log(rz) : log(ZMIN);
+    examl_host_make_p(qz, rz, gammaRates, EI, EIGN, 4, &hostP[e * PBLK],
+                      &hostP[e * PBLK + PBLK / 2], STATES);
+  }
+
+  /* graph fast path: identical traversal shape -> replay (P content flows
+   * through the captured memcpy from the pinned hostP buffer) */
+  const bool want_graph = g_use_graphs && !g_prof_on && s != nullptr;
+  unsigned long long key = 0;
+  bool capturing = false;
+  if (want_graph) {
+    key = trav_key(ops, numOps, n, dev_clv, dev_tips, dev_pbuf, dev_EV,
+                   dev_tipVec, dev_wgt, dev_scalers, dev_inc, (void *)s,
+                   STATES);
+    hipGraphExec_t exec = trav_graph_find(key);
+    if (exec) {
+      CHK(hipGraphLaunch(exec, s));
+      return 0;
+    }
+    capturing =
+        hipStreamBeginCapture(s, hipStreamCaptureModeThreadLocal) ==
+        hipSuccess;
+    (void)hipGetLastError();
+  }
+
+  int rc = 0;
+  do {
+    hipError_t err = hipMemcpyAsync(dev_pbuf, hostP,
+                                    (size_t)numOps * PBLK * sizeof(double),
+                                    hipMemcpyHostToDevice, s);
+    if (err != hipSuccess) { rc = set_err(err, "pbuf upload"); break; }
+    err = hipMemsetAsync(dev_inc, 0, (size_t)numOps * sizeof(unsigned int),
+                         s);
+    if (err != hipSuccess) { rc = set_err(err, "inc memset"); break; }
+
+    /* 2. one newview kernel per entry, post order on one stream */
+    const int grid = grid_for(n * 4);
+    for (int e = 0; e < numOps && rc == 0; e++) {
+      const examl_hip_trav_entry *op = &ops[e];
+      hipEvent_t ev_a = nullptr, ev_b = nullptr;
+      if (g_prof_on) {
+        prof_begin(&ev_a, &ev_b);
+        hipEventRecord(ev_a, s);
+      }
+      const double *P = dev_pbuf + (long)e * PBLK;
+      double *x3 = dev_clv + (long)op->x3Slot * clvStride;
+      const double *x1 = nullptr, *x2 = nullptr;
+      const unsigned char *t1 = nullptr, *t2 = nullptr;
+      switch (op->tipCase) {
+      case EXAML_TIP_TIP:
+        t1 = dev_tips + (long)op->x1Slot * tipStride;
+        t2 = dev_tips + (long)op->x2Slot * tipStride;
+        if (STATES == 4)
+          hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_TIP_TIP>), dim3(grid),
+                             dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV,
+                             dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
+        else
+          hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_TIP>),
+                             dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P,
+                             dev_EV, dev_tipVec, t1, t2, dev_wgt, n,
+                             dev_inc + e);
+        break;
+      case EXAML_TIP_INNER:
+        t1 = dev_tips + (long)op->x1Slot * tipStride;
+        x2 = dev_clv + (long)op->x2Slot * clvStride;
+        if (STATES == 4)
+          hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_TIP_INNER>),
+                             dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P,
+                             dev_EV, dev_tipVec, t1, t2, dev_wgt, n,
+                             dev_inc + e);
+        else
+          hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_INNER>),
+                             dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P,
+                             dev_EV, dev_tipVec, t1, t2, dev_wgt, n,
+                             dev_inc + e);
+        break;
+      case EXAML_INNER_INNER:
+        x1 = dev_clv + (long)op->x1Slot * clvStride;
+        x2 = dev_clv + (long)op->x2Slot * clvStride;
+        if (STATES == 4)
+          hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_INNER_INNER>),
+                             dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P,
+                             dev_EV, dev_tipVec, t1, t2, dev_wgt, n,
+                             dev_inc + e);
+        else
+          hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_INNER_INNER>),
+                             dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P,
+                             dev_EV, dev_tipVec, t1, t2, dev_wgt, n,
+                             dev_inc + e);
+        break;
+      default:
+        snprintf(g_err, sizeof(g_err), "traversal: bad tipCase %d",
+                 op->tipCase);
+        rc = -1;
+        break;
+      }
+      if (rc == 0) {
+        err = hipGetLastError();
+        if (err != hipSuccess) { rc = set_err(err, "newview launch"); break; }
+      }
+      if (g_prof_on) {
+        hipEventRecord(ev_b, s);
+        g_prof_pend.push_back({ev_a, ev_b, op->tipCase});
+        if (g_prof_pend.size() > 2048) prof_flush();
+      }
+    }
+    if (rc != 0) break;
+
+    /* 3. recursive scaler accumulation (newviewGenericSpecial.c:1503) */
+    for (int base = 0; base < numOps && rc == 0; base += FIN_CHUNK) {
+      FinMeta m;
+      m.count = (numOps - base < FIN_CHUNK) ? (numOps - base) : FIN_CHUNK;
+      m.base = base;
+      for (int e = 0; e < m.count; e++) {
+        m.p[e] = ops[base + e].pNumber;
+        m.q[e] = ops[base + e].qNumber;
+        m.r[e] = ops[base + e].rNumber;
+      }
+      hipLaunchKernelGGL(k_scaler_finalize, dim3(1), dim3(64), 0, s, m,
+                         dev_inc, dev_scalers);
+      err = hipGetLastError();
+      if (err != hipSuccess) rc = set_err(err, "finalize launch");
+    }
+  } while (0);
+
+  if (capturing) {
+    hipGraph_t graph = nullptr;
+    hipError_t err = hipStreamEndCapture(s, &graph);
+    if (rc != 0) {
+      if (graph) hipGraphDestroy(graph);
+      return rc;
+    }
+    if (err != hipSuccess) return set_err(err, "end capture");
+    hipGraphExec_t exec = nullptr;
+    err = hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0);
+    hipGraphDestroy(graph);
+    if (err != hipSuccess) return set_err(err, "graph instantiate");
+    trav_graph_store(key, exec);
+    CHK(hipGraphLaunch(exec, s));
+  }
+  return rc;
+}
 
 extern "C" int examl_hip_newview_traversal_dna_gamma(
     const examl_hip_trav_entry *ops, int numOps, const double *EIGN,
@@ -900,94 +1151,11 @@ extern "C" int examl_hip_newview_traversal_dna_gamma(
     unsigned int *dev_scalers, unsigned int *dev_inc, double *dev_pbuf,
     void *stream) {
   if (numOps <= 0 || n <= 0) return 0;
-  hipStream_t s = (hipStream_t)stream;
-  (void)hipGetLastError(); /* clear stale per-thread error (e.g. torch probes) */
-
-  /* 1. all P-matrix pairs on the host, one block of 128 doubles per op
-   *    (newviewGenericSpecial.c:982-1044: clamp to zmin, log, makeP) */
-  static thread_local double *hostP = nullptr;
-  static thread_local int hostPCap = 0;
-  if (numOps > hostPCap) {
-    free(hostP);
-    hostP = (double *)malloc((size_t)numOps * 128 * sizeof(double));
-    hostPCap = numOps;
-  }
-  for (int e = 0; e < numOps; e++) {
-    double qz = ops[e].qz, rz = ops[e].rz;
-    qz = (qz > ZMIN) ? log(qz) : log(ZMIN);
-    rz = (rz > ZMIN) ? log(rz) : log(ZMIN);
-    examl_host_make_p(qz, rz, gammaRates, EI, EIGN, 4, &hostP[e * 128],
-                      &hostP[e * 128 + 64], 4);
-  }
-  /* one upload for the whole traversal (hipMemcpyAsync from pageable memory
-   * is host-synchronous, so hostP is reusable on return) */
-  CHK(hipMemcpyAsync(dev_pbuf, hostP, (size_t)numOps * 128 * sizeof(double),
-                     hipMemcpyHostToDevice, s));
-  CHK(hipMemsetAsync(dev_inc, 0, (size_t)numOps * sizeof(unsigned int), s));
-
-  /* 2. one newview kernel per entry, post order on one stream */
-  const int grid = grid_for(n * 4);
-  for (int e = 0; e < numOps; e++) {
-    const examl_hip_trav_entry *op = &ops[e];
-    hipEvent_t ev_a = nullptr, ev_b = nullptr;
-    if (g_prof_on) {
-      prof_begin(&ev_a, &ev_b);
-      hipEventRecord(ev_a, s);
-    }
-    const double *P = dev_pbuf + (long)e * 128;
-    double *x3 = dev_clv + (long)op->x3Slot * clvStride;
-    const double *x1 = nullptr, *x2 = nullptr;
-    const unsigned char *t1 = nullptr, *t2 = nullptr;
-    switch (op->tipCase) {
-    case EXAML_TIP_TIP:
-      t1 = dev_tips + (long)op->x1Slot * tipStride;
-      t2 = dev_tips + (long)op->x2Slot * tipStride;
-      hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_TIP_TIP>), dim3(grid),
-                         dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV,
-                         dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
-      break;
-    case EXAML_TIP_INNER:
-      t1 = dev_tips + (long)op->x1Slot * tipStride;
-      x2 = dev_clv + (long)op->x2Slot * clvStride;
-      hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_TIP_INNER>), dim3(grid),
-                         dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV,
-                         dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
-      break;
-    case EXAML_INNER_INNER:
-      x1 = dev_clv + (long)op->x1Slot * clvStride;
-      x2 = dev_clv + (long)op->x2Slot * clvStride;
-      hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_INNER_INNER>), dim3(grid),
-                         dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV,
-                         dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
-      break;
-    default:
-      snprintf(g_err, sizeof(g_err), "traversal: bad tipCase %d",
-               op->tipCase);
-      return -1;
-    }
-    CHK(hipGetLastError());
-    if (g_prof_on) {
-      hipEventRecord(ev_b, s);
-      g_prof_pend.push_back({ev_a, ev_b, op->tipCase});
-      if (g_prof_pend.size() > 2048) prof_flush();
-    }
-  }
-
-  /* 3. recursive scaler accumulation, chunked through by-value kernargs */
-  for (int base = 0; base < numOps; base += FIN_CHUNK) {
-    FinMeta m;
-    m.count = (numOps - base < FIN_CHUNK) ? (numOps - base) : FIN_CHUNK;
-    m.base = base;
-    for (int e = 0; e < m.count; e++) {
-      m.p[e] = ops[base + e].pNumber;
-      m.q[e] = ops[base + e].qNumber;
-      m.r[e] = ops[base + e].rNumber;
-    }
-    hipLaunchKernelGGL(k_scaler_finalize, dim3(1), dim3(64), 0, s, m, dev_inc,
-                       dev_scalers);
-    CHK(hipGetLastError());
-  }
-  return 0;
+  (void)hipGetLastError();
+  return traversal_impl<4>(ops, numOps, EIGN, EI, gammaRates, dev_EV,
+                           dev_tipVec, dev_clv, clvStride, dev_tips,
+                           tipStride, dev_wgt, n, dev_scalers, dev_inc,
+                           dev_pbuf, (hipStream_t)stream);
 }
 
 /* --- evaluate at the root (evaluateIterative body) ------------------------ */
@@ -1196,88 +1364,11 @@ extern "C" int examl_hip_newview_traversal_prot_gamma(
     unsigned int *dev_scalers, unsigned int *dev_inc, double *dev_pbuf,
     void *stream) {
   if (numOps <= 0 || n <= 0) return 0;
-  hipStream_t s = (hipStream_t)stream;
   (void)hipGetLastError();
-
-  static thread_local double *hostP = nullptr;
-  static thread_local int hostPCap = 0;
-  if (numOps > hostPCap) {
-    free(hostP);
-    hostP = (double *)malloc((size_t)numOps * 3200 * sizeof(double));
-    hostPCap = numOps;
-  }
-  for (int e = 0; e < numOps; e++) {
-    double qz = ops[e].qz, rz = ops[e].rz;
-    qz = (qz > ZMIN) ? log(qz) : log(ZMIN);
-    rz = (rz > ZMIN) ? log(rz) : log(ZMIN);
-    examl_host_make_p(qz, rz, gammaRates, EI, EIGN, 4, &hostP[e * 3200],
-                      &hostP[e * 3200 + 1600], 20);
-  }
-  CHK(hipMemcpyAsync(dev_pbuf, hostP, (size_t)numOps * 3200 * sizeof(double),
-                     hipMemcpyHostToDevice, s));
-  CHK(hipMemsetAsync(dev_inc, 0, (size_t)numOps * sizeof(unsigned int), s));
-
-  const int grid = grid_for(n * 4);
-  for (int e = 0; e < numOps; e++) {
-    const examl_hip_trav_entry *op = &ops[e];
-    hipEvent_t ev_a = nullptr, ev_b = nullptr;
-    if (g_prof_on) {
-      prof_begin(&ev_a, &ev_b);
-      hipEventRecord(ev_a, s);
-    }
-    const double *P = dev_pbuf + (long)e * 3200;
-    double *x3 = dev_clv + (long)op->x3Slot * clvStride;
-    const double *x1 = nullptr, *x2 = nullptr;
-    const unsigned char *t1 = nullptr, *t2 = nullptr;
-    switch (op->tipCase) {
-    case EXAML_TIP_TIP:
-      t1 = dev_tips + (long)op->x1Slot * tipStride;
-      t2 = dev_tips + (long)op->x2Slot * tipStride;
-      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_TIP>), dim3(grid),
-                         dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV,
-                         dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
-      break;
-    case EXAML_TIP_INNER:
-      t1 = dev_tips + (long)op->x1Slot * tipStride;
-      x2 = dev_clv + (long)op->x2Slot * clvStride;
-      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_INNER>), dim3(grid),
-                         dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV,
-                         dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
-      break;
-    case EXAML_INNER_INNER:
-      x1 = dev_clv + (long)op->x1Slot * clvStride;
-      x2 = dev_clv + (long)op->x2Slot * clvStride;
-      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_INNER_INNER>),
-                         dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P,
-                         dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
-      break;
-    default:
-      snprintf(g_err, sizeof(g_err), "traversal_prot: bad tipCase %d",
-               op->tipCase);
-      return -1;
-    }
-    CHK(hipGetLastError());
-    if (g_prof_on) {
-      hipEventRecord(ev_b, s);
-      g_prof_pend.push_back({ev_a, ev_b, op->tipCase});
-      if (g_prof_pend.size() > 2048) prof_flush();
-    }
-  }
-
-  for (int base = 0; base < numOps; base += FIN_CHUNK) {
-    FinMeta m;
-    m.count = (numOps - base < FIN_CHUNK) ? (numOps - base) : FIN_CHUNK;
-    m.base = base;
-    for (int e = 0; e < m.count; e++) {
-      m.p[e] = ops[base + e].pNumber;
-      m.q[e] = ops[base + e].qNumber;
-      m.r[e] = ops[base + e].rNumber;
-    }
-    hipLaunchKernelGGL(k_scaler_finalize, dim3(1), dim3(64), 0, s, m, dev_inc,
-                       dev_scalers);
-    CHK(hipGetLastError());
-  }
-  return 0;
+  return traversal_impl<20>(ops, numOps, EIGN, EI, gammaRates, dev_EV,
+                            dev_tipVec, dev_clv, clvStride, dev_tips,
+                            tipStride, dev_wgt, n, dev_scalers, dev_inc,
+                            dev_pbuf, (hipStream_t)stream);
 }
 
 extern "C" int examl_hip_evaluate_root_prot_gamma(

@@ -47,6 +47,12 @@ void examl_hip_profile_enable(int on);
 void examl_hip_profile_reset(void);
 void examl_hip_profile_get(double *ms_by_tc, long *cnt_by_tc);
 
+/* hipGraph replay of repeated traversal shapes (on by default; bypassed
+ * while profiling is enabled).  examl_hip_graphs_clear drops all cached
+ * executable graphs (e.g. before freeing the device buffers they bind). */
+void examl_hip_use_graphs(int on);
+void examl_hip_graphs_clear(void);
+
 /* ---------------------------------------------------------------------------
  * Host-side model math (runs once per model-parameter change; feeds the
  * kernels).  These replace the corresponding host functions in the
