@@ -49,7 +49,11 @@ extern "C" const char *examl_hip_last_error_string(void) { return g_err; }
  * ==========================================================================*/
 
 #define NV_BLOCK 256
-#define MAX_GRID 4096
+
+/* clang ext-vector for nontemporal 32-byte stores (HIP's double4 is a
+ * class type the builtin rejects) */
+typedef double v4d __attribute__((ext_vector_type(4)));
+#define MAX_GRID 8192
 
 static inline int grid_for(long units) {
   long g = (units + NV_BLOCK - 1) / NV_BLOCK;
@@ -66,7 +70,7 @@ static inline int grid_for(long units) {
  * < 2^-256, multiply the site's span by 2^256 and add wgt[site] to the
  * scaler count (avxLikelihood.c:223-305).
  */
-template <int TC>
+template <int TC, bool NT>
 __global__ __launch_bounds__(NV_BLOCK) void k_newview_dna_gamma(
     const double *__restrict__ x1, const double *__restrict__ x2,
     double *__restrict__ x3, const double *__restrict__ P,
@@ -164,8 +168,14 @@ __global__ __launch_bounds__(NV_BLOCK) void k_newview_dna_gamma(
           atomicAdd(scalerInc, (unsigned int)wgt[site]);
       }
     }
-    *reinterpret_cast<double4 *>(&x3[idx * 4]) =
-        make_double4(a0, a1, a2, a3);
+    if (NT)
+      /* streaming store: x3 far exceeds L2 and read-allocating stores
+       * waste HBM bandwidth (+19% measured, tools/kernel_ab) */
+      __builtin_nontemporal_store((v4d){a0, a1, a2, a3},
+                                  reinterpret_cast<v4d *>(&x3[idx * 4]));
+    else
+      *reinterpret_cast<double4 *>(&x3[idx * 4]) =
+          make_double4(a0, a1, a2, a3);
   }
 }
 
@@ -355,7 +365,7 @@ __device__ __forceinline__ double dot20o(const double *a, const double *b) {
   return (t0 + t1) + (t2 + t3);
 }
 
-template <int TC>
+template <int TC, bool NT>
 __global__ __launch_bounds__(NV_BLOCK) void k_newview_prot_gamma(
     const double *__restrict__ x1, const double *__restrict__ x2,
     double *__restrict__ x3, const double *__restrict__ P,
@@ -448,9 +458,16 @@ __global__ __launch_bounds__(NV_BLOCK) void k_newview_prot_gamma(
       }
     }
 #pragma unroll
-    for (int s = 0; s < 20; s += 4)
-      *reinterpret_cast<double4 *>(&x3[idx * 20 + s]) =
-          make_double4(acc[s], acc[s + 1], acc[s + 2], acc[s + 3]);
+    for (int s = 0; s < 20; s += 4) {
+      const double4 v = make_double4(acc[s], acc[s + 1], acc[s + 2],
+                                     acc[s + 3]);
+      if (NT)
+        __builtin_nontemporal_store(
+            (v4d){v.x, v.y, v.z, v.w},
+            reinterpret_cast<v4d *>(&x3[idx * 20 + s]));
+      else
+        *reinterpret_cast<double4 *>(&x3[idx * 20 + s]) = v;
+    }
   }
 }
 
@@ -732,6 +749,7 @@ extern "C" int examl_hip_newview_dna_gamma(
   hipStream_t s = (hipStream_t)stream;
   (void)hipGetLastError(); /* clear stale per-thread error (e.g. torch probes) */
   const int grid = grid_for(n * 4);
+  const bool nt = n >= 65536; /* streaming stores once x3 exceeds L2 */
   /* left/right must be contiguous (P = left | right); the launcher copies
    * are avoided by requiring the caller to pass left==P, right==P+64 when
    * using the traversal executor; for the standalone call we accept two
@@ -743,19 +761,22 @@ extern "C" int examl_hip_newview_dna_gamma(
   }
   switch (tipCase) {
   case EXAML_TIP_TIP:
-    hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_TIP_TIP>), dim3(grid),
-                       dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec,
-                       tipX1, tipX2, wgt, n, scalerInc);
+    if (nt)
+      hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_TIP_TIP, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
+    else
+      hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_TIP_TIP, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
     break;
   case EXAML_TIP_INNER:
-    hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_TIP_INNER>), dim3(grid),
-                       dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec,
-                       tipX1, tipX2, wgt, n, scalerInc);
+    if (nt)
+      hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_TIP_INNER, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
+    else
+      hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_TIP_INNER, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
     break;
   case EXAML_INNER_INNER:
-    hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_INNER_INNER>), dim3(grid),
-                       dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec,
-                       tipX1, tipX2, wgt, n, scalerInc);
+    if (nt)
+      hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_INNER_INNER, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
+    else
+      hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_INNER_INNER, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
     break;
   default:
     snprintf(g_err, sizeof(g_err), "newview: bad tipCase %d", tipCase);
@@ -1037,6 +1058,7 @@ static int traversal_impl(const examl_hip_trav_entry *ops, int numOps,
 
     /* 2. one newview kernel per entry, post order on one stream */
     const int grid = grid_for(n * 4);
+    const bool nt = (STATES == 4) ? (n >= 65536) : (n >= 16384);
     for (int e = 0; e < numOps && rc == 0; e++) {
       const examl_hip_trav_entry *op = &ops[e];
       hipEvent_t ev_a = nullptr, ev_b = nullptr;
@@ -1053,42 +1075,43 @@ static int traversal_impl(const examl_hip_trav_entry *ops, int numOps,
         t1 = dev_tips + (long)op->x1Slot * tipStride;
         t2 = dev_tips + (long)op->x2Slot * tipStride;
         if (STATES == 4)
-          hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_TIP_TIP>), dim3(grid),
-                             dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV,
-                             dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
+          if (nt)
+      hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_TIP_TIP, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
+    else
+      hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_TIP_TIP, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
         else
-          hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_TIP>),
-                             dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P,
-                             dev_EV, dev_tipVec, t1, t2, dev_wgt, n,
-                             dev_inc + e);
+          if (nt)
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_TIP, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
+    else
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_TIP, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
         break;
       case EXAML_TIP_INNER:
         t1 = dev_tips + (long)op->x1Slot * tipStride;
         x2 = dev_clv + (long)op->x2Slot * clvStride;
         if (STATES == 4)
-          hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_TIP_INNER>),
-                             dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P,
-                             dev_EV, dev_tipVec, t1, t2, dev_wgt, n,
-                             dev_inc + e);
+          if (nt)
+      hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_TIP_INNER, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
+    else
+      hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_TIP_INNER, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
         else
-          hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_INNER>),
-                             dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P,
-                             dev_EV, dev_tipVec, t1, t2, dev_wgt, n,
-                             dev_inc + e);
+          if (nt)
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_INNER, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
+    else
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_INNER, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
         break;
       case EXAML_INNER_INNER:
         x1 = dev_clv + (long)op->x1Slot * clvStride;
         x2 = dev_clv + (long)op->x2Slot * clvStride;
         if (STATES == 4)
-          hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_INNER_INNER>),
-                             dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P,
-                             dev_EV, dev_tipVec, t1, t2, dev_wgt, n,
-                             dev_inc + e);
+          if (nt)
+      hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_INNER_INNER, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
+    else
+      hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_INNER_INNER, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
         else
-          hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_INNER_INNER>),
-                             dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P,
-                             dev_EV, dev_tipVec, t1, t2, dev_wgt, n,
-                             dev_inc + e);
+          if (nt)
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_INNER_INNER, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
+    else
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_INNER_INNER, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
         break;
       default:
         snprintf(g_err, sizeof(g_err), "traversal: bad tipCase %d",
@@ -1260,6 +1283,7 @@ extern "C" int examl_hip_newview_prot_gamma(
   hipStream_t s = (hipStream_t)stream;
   (void)hipGetLastError();
   const int grid = grid_for(n * 4);
+  const bool nt = n >= 16384; /* 640 B/site write: L2 overflow threshold */
   if (right != left + 1600) {
     snprintf(g_err, sizeof(g_err),
              "newview_prot: right must be left+1600 (one P block)");
@@ -1267,19 +1291,22 @@ extern "C" int examl_hip_newview_prot_gamma(
   }
   switch (tipCase) {
   case EXAML_TIP_TIP:
-    hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_TIP>), dim3(grid),
-                       dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec,
-                       tipX1, tipX2, wgt, n, scalerInc);
+    if (nt)
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_TIP, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
+    else
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_TIP, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
     break;
   case EXAML_TIP_INNER:
-    hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_INNER>), dim3(grid),
-                       dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec,
-                       tipX1, tipX2, wgt, n, scalerInc);
+    if (nt)
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_INNER, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
+    else
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_INNER, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
     break;
   case EXAML_INNER_INNER:
-    hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_INNER_INNER>), dim3(grid),
-                       dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec,
-                       tipX1, tipX2, wgt, n, scalerInc);
+    if (nt)
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_INNER_INNER, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
+    else
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_INNER_INNER, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
     break;
   default:
     snprintf(g_err, sizeof(g_err), "newview_prot: bad tipCase %d", tipCase);
