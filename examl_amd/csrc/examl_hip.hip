@@ -1058,7 +1058,7 @@ static int traversal_impl(const examl_hip_trav_entry *ops, int numOps,
 
     /* 2. one newview kernel per entry, post order on one stream */
     const int grid = grid_for(n * 4);
-    const bool nt = (STATES == 4) ? (n >= 65536) : (n >= 16384);
+    const bool nt = (STATES == 4) && (n >= 65536);
     for (int e = 0; e < numOps && rc == 0; e++) {
       const examl_hip_trav_entry *op = &ops[e];
       hipEvent_t ev_a = nullptr, ev_b = nullptr;
@@ -1283,7 +1283,8 @@ extern "C" int examl_hip_newview_prot_gamma(
   hipStream_t s = (hipStream_t)stream;
   (void)hipGetLastError();
   const int grid = grid_for(n * 4);
-  const bool nt = n >= 16384; /* 640 B/site write: L2 overflow threshold */
+  const bool nt = false; /* NT hurts the protein kernel: its five strided
+    32-B stores per thread defeat write-combining (measured 2x slower) */
   if (right != left + 1600) {
     snprintf(g_err, sizeof(g_err),
              "newview_prot: right must be left+1600 (one P block)");
