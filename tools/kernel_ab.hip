@@ -160,6 +160,8 @@ double timeit(F f, int reps) {
   return ms / reps;
 }
 
+extern "C" void run_appended_probes();
+
 int main() {
   const long n = N_SITES;
   double *x1, *x2, *x3, *P, *EV;
@@ -226,5 +228,194 @@ int main() {
     printf("%-22s %8.2f us  %6.2f TB/s (stream ceiling, 384 B/site)\n",
            "copy-add ceiling", ms * 1e3, 384.0 * n / (ms * 1e-3) / 1e12);
   }
+  run_appended_probes();
   return 0;
+}
+
+/* ===== appended probes: protein II variants + f64 MFMA rate ============= */
+
+/* P1: production protein shape (interleaved cats, LDS P/EV) */
+__global__ __launch_bounds__(BLOCK) void kp_base(
+    const double *__restrict__ x1, const double *__restrict__ x2,
+    double *__restrict__ x3, const double *__restrict__ P,
+    const double *__restrict__ EV, long n) {
+  __shared__ double sL[1600], sR[1600], sEV[400];
+  const int tid = threadIdx.x;
+  for (int j = tid; j < 1600; j += BLOCK) { sL[j] = P[j]; sR[j] = P[1600+j]; }
+  for (int j = tid; j < 400; j += BLOCK) sEV[j] = EV[j];
+  __syncthreads();
+  const long units = n * 4;
+  for (long idx = (long)blockIdx.x * BLOCK + tid; idx < units;
+       idx += (long)gridDim.x * BLOCK) {
+    const int cat = (int)(idx & 3);
+    double xl[20], xr[20], acc[20];
+#pragma unroll
+    for (int s2 = 0; s2 < 20; s2 += 4) {
+      const double4 a = *reinterpret_cast<const double4 *>(&x1[idx*20+s2]);
+      const double4 b = *reinterpret_cast<const double4 *>(&x2[idx*20+s2]);
+      xl[s2]=a.x; xl[s2+1]=a.y; xl[s2+2]=a.z; xl[s2+3]=a.w;
+      xr[s2]=b.x; xr[s2+1]=b.y; xr[s2+2]=b.z; xr[s2+3]=b.w;
+    }
+#pragma unroll
+    for (int s2 = 0; s2 < 20; s2++) acc[s2] = 0;
+    for (int l = 0; l < 20; l++) {
+      double t0=0,t1=0,t2=0,t3=0, r0=0,r1=0,r2=0,r3=0;
+#pragma unroll
+      for (int c = 0; c < 20; c += 4) {
+        t0 += xl[c]*sL[cat*400+l*20+c];   t1 += xl[c+1]*sL[cat*400+l*20+c+1];
+        t2 += xl[c+2]*sL[cat*400+l*20+c+2]; t3 += xl[c+3]*sL[cat*400+l*20+c+3];
+        r0 += xr[c]*sR[cat*400+l*20+c];   r1 += xr[c+1]*sR[cat*400+l*20+c+1];
+        r2 += xr[c+2]*sR[cat*400+l*20+c+2]; r3 += xr[c+3]*sR[cat*400+l*20+c+3];
+      }
+      const double t = ((t0+t1)+(t2+t3)) * ((r0+r1)+(r2+r3));
+#pragma unroll
+      for (int s2 = 0; s2 < 20; s2++) acc[s2] += t * sEV[l*20+s2];
+    }
+#pragma unroll
+    for (int s2 = 0; s2 < 20; s2 += 4)
+      *reinterpret_cast<double4 *>(&x3[idx*20+s2]) =
+          make_double4(acc[s2], acc[s2+1], acc[s2+2], acc[s2+3]);
+  }
+}
+
+/* P2: cat-uniform wave -> P/EV via uniform (scalar) global loads */
+__global__ __launch_bounds__(BLOCK) void kp_scalar(
+    const double *__restrict__ x1, const double *__restrict__ x2,
+    double *__restrict__ x3, const double *__restrict__ P,
+    const double *__restrict__ EV, long n) {
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int cat = wave & 3; /* uniform per wave */
+  const long nw = (long)gridDim.x * (BLOCK / 64 / 4); /* site-groups of 64 */
+  const long wg = (long)blockIdx.x * (BLOCK / 64 / 4) + (wave >> 2);
+  const double *Lc = P + cat * 400;
+  const double *Rc = P + 1600 + cat * 400;
+  for (long base = wg * 64; base < n; base += nw * 64) {
+    const long site = base + lane;
+    if (site >= n) break;
+    const long idx = site * 4 + cat;
+    double xl[20], xr[20], acc[20];
+#pragma unroll
+    for (int s2 = 0; s2 < 20; s2 += 4) {
+      const double4 a = *reinterpret_cast<const double4 *>(&x1[idx*20+s2]);
+      const double4 b = *reinterpret_cast<const double4 *>(&x2[idx*20+s2]);
+      xl[s2]=a.x; xl[s2+1]=a.y; xl[s2+2]=a.z; xl[s2+3]=a.w;
+      xr[s2]=b.x; xr[s2+1]=b.y; xr[s2+2]=b.z; xr[s2+3]=b.w;
+    }
+#pragma unroll
+    for (int s2 = 0; s2 < 20; s2++) acc[s2] = 0;
+    for (int l = 0; l < 20; l++) {
+      double t0=0,t1=0,t2=0,t3=0, r0=0,r1=0,r2=0,r3=0;
+#pragma unroll
+      for (int c = 0; c < 20; c += 4) {
+        t0 += xl[c]*Lc[l*20+c];     t1 += xl[c+1]*Lc[l*20+c+1];
+        t2 += xl[c+2]*Lc[l*20+c+2]; t3 += xl[c+3]*Lc[l*20+c+3];
+        r0 += xr[c]*Rc[l*20+c];     r1 += xr[c+1]*Rc[l*20+c+1];
+        r2 += xr[c+2]*Rc[l*20+c+2]; r3 += xr[c+3]*Rc[l*20+c+3];
+      }
+      const double t = ((t0+t1)+(t2+t3)) * ((r0+r1)+(r2+r3));
+#pragma unroll
+      for (int s2 = 0; s2 < 20; s2++) acc[s2] += t * EV[l*20+s2];
+    }
+#pragma unroll
+    for (int s2 = 0; s2 < 20; s2 += 4)
+      *reinterpret_cast<double4 *>(&x3[idx*20+s2]) =
+          make_double4(acc[s2], acc[s2+1], acc[s2+2], acc[s2+3]);
+  }
+}
+
+/* f64 MFMA 16x16x4 throughput probe */
+typedef double d4v __attribute__((ext_vector_type(4)));
+__global__ __launch_bounds__(256) void k_mfma64(double *out, double a0) {
+  d4v c = {0, 0, 0, 0};
+  double a = a0 + threadIdx.x, b = a0 * 1.0001 + threadIdx.x;
+  d4v c1 = {1, 1, 1, 1}, c2 = {2, 2, 2, 2}, c3 = {3, 3, 3, 3};
+  for (int i = 0; i < 2048; i++) {
+    c = __builtin_amdgcn_mfma_f64_16x16x4f64(a, b, c, 0, 0, 0);
+    c1 = __builtin_amdgcn_mfma_f64_16x16x4f64(a, b, c1, 0, 0, 0);
+    c2 = __builtin_amdgcn_mfma_f64_16x16x4f64(a, b, c2, 0, 0, 0);
+    c3 = __builtin_amdgcn_mfma_f64_16x16x4f64(a, b, c3, 0, 0, 0);
+  }
+  out[threadIdx.x + blockIdx.x * 256] = c[0] + c1[1] + c2[2] + c3[3];
+}
+
+/* f64 VALU FMA probe */
+__global__ __launch_bounds__(256) void k_valu64(double *out, double a0) {
+  double a = a0 + threadIdx.x, b = a0 * 1.0001;
+  double c0=0,c1=1,c2=2,c3=3,c4=4,c5=5,c6=6,c7=7;
+  for (int i = 0; i < 4096; i++) {
+    c0 = fma(a, b, c0); c1 = fma(a, b, c1); c2 = fma(a, b, c2);
+    c3 = fma(a, b, c3); c4 = fma(a, b, c4); c5 = fma(a, b, c5);
+    c6 = fma(a, b, c6); c7 = fma(a, b, c7);
+  }
+  out[threadIdx.x + blockIdx.x * 256] = c0+c1+c2+c3+c4+c5+c6+c7;
+}
+
+extern "C" void run_appended_probes() {
+  const long n = 200000;
+  double *x1, *x2, *x3, *P, *EV;
+  CHK(hipMalloc(&x1, n * 80 * 8));
+  CHK(hipMalloc(&x2, n * 80 * 8));
+  CHK(hipMalloc(&x3, n * 80 * 8));
+  CHK(hipMalloc(&P, 3200 * 8));
+  CHK(hipMalloc(&EV, 400 * 8));
+  double *h = (double *)malloc(n * 80 * 8);
+  for (long i = 0; i < n * 80; i++) h[i] = 0.1 + (i % 97) * 0.009;
+  CHK(hipMemcpy(x1, h, n * 80 * 8, hipMemcpyHostToDevice));
+  CHK(hipMemcpy(x2, h, n * 80 * 8, hipMemcpyHostToDevice));
+  double hp[3200], hev[400];
+  for (int i = 0; i < 3200; i++) hp[i] = 0.2 + (i % 53) * 0.003;
+  for (int i = 0; i < 400; i++) hev[i] = 0.3 + (i % 31) * 0.01;
+  CHK(hipMemcpy(P, hp, sizeof(hp), hipMemcpyHostToDevice));
+  CHK(hipMemcpy(EV, hev, sizeof(hev), hipMemcpyHostToDevice));
+
+  printf("\nprotein II (no scaling branch), %ld sites, 1924 B/site\n", n);
+  for (int grid : {2048, 4096, 8192}) {
+    double ms = timeit([&] {
+      hipLaunchKernelGGL(kp_base, dim3(grid), dim3(BLOCK), 0, 0, x1, x2, x3,
+                         P, EV, n);
+    }, 10);
+    printf("P1 lds base  g%-6d %8.1f us  %6.2f TB/s  %5.1f Gsites/s\n", grid,
+           ms * 1e3, 1920.0 * n / (ms * 1e-3) / 1e12, n / (ms * 1e-3) / 1e9);
+  }
+  for (int grid : {2048, 4096, 8192, 12500}) {
+    double ms = timeit([&] {
+      hipLaunchKernelGGL(kp_scalar, dim3(grid), dim3(BLOCK), 0, 0, x1, x2,
+                         x3, P, EV, n);
+    }, 10);
+    printf("P2 scalarP   g%-6d %8.1f us  %6.2f TB/s  %5.1f Gsites/s\n", grid,
+           ms * 1e3, 1920.0 * n / (ms * 1e-3) / 1e12, n / (ms * 1e-3) / 1e9);
+  }
+  /* verify P2 == P1 bitwise */
+  {
+    double *x4;
+    CHK(hipMalloc(&x4, n * 80 * 8));
+    hipLaunchKernelGGL(kp_base, dim3(4096), dim3(BLOCK), 0, 0, x1, x2, x3, P, EV, n);
+    hipLaunchKernelGGL(kp_scalar, dim3(4096), dim3(BLOCK), 0, 0, x1, x2, x4, P, EV, n);
+    CHK(hipDeviceSynchronize());
+    double *h3 = (double *)malloc(n * 80 * 8), *h4 = (double *)malloc(n * 80 * 8);
+    CHK(hipMemcpy(h3, x3, n * 80 * 8, hipMemcpyDeviceToHost));
+    CHK(hipMemcpy(h4, x4, n * 80 * 8, hipMemcpyDeviceToHost));
+    long bad = 0;
+    for (long i = 0; i < n * 80; i++) if (h3[i] != h4[i]) bad++;
+    printf("P1 vs P2 bit-diff count: %ld\n", bad);
+  }
+  double *out;
+  CHK(hipMalloc(&out, 8192 * 256 * 8));
+  {
+    double ms = timeit([&] {
+      hipLaunchKernelGGL(k_mfma64, dim3(2048), dim3(256), 0, 0, out, 1.0001);
+    }, 5);
+    /* flops: grid*256 threads /64 lanes = waves; per wave iter: 4 mfma * 2048 flop */
+    double waves = 2048.0 * 256 / 64;
+    double fl = waves * 2048.0 * 4 * 2048;
+    printf("f64 MFMA 16x16x4: %8.2f ms -> %6.1f TFLOP/s\n", ms, fl / (ms * 1e-3) / 1e12);
+  }
+  {
+    double ms = timeit([&] {
+      hipLaunchKernelGGL(k_valu64, dim3(2048), dim3(256), 0, 0, out, 1.0001);
+    }, 5);
+    double fl = 2048.0 * 256 * 4096.0 * 8 * 2;
+    printf("f64 VALU fma:     %8.2f ms -> %6.1f TFLOP/s\n", ms, fl / (ms * 1e-3) / 1e12);
+  }
 }
