@@ -63,32 +63,34 @@ def _bind(lib):
     lib.examl_hip_newview_dna_gamma.argtypes = \
         [i, p, p, p, p, p, p, p, l, p, p, p, p, p]
     lib.examl_hip_evaluate_dna_gamma.argtypes = \
-        [p, p, p, p, p, l, p, p, p, d, p, p]
+        [p, p, p, p, p, l, p, p, p, d, p, p, p]
     lib.examl_hip_sum_dna_gamma.argtypes = [i, p, p, p, p, p, p, l, p]
-    lib.examl_hip_core_dna_gamma.argtypes = [l, p, p, p, p, p]
+    lib.examl_hip_core_dna_gamma.argtypes = [l, p, p, p, p, p, p]
     lib.examl_hip_newview_traversal_dna_gamma.argtypes = \
         [p, i, p, p, p, p, p, p, l, p, l, p, l, p, p, p, p]
     lib.examl_hip_evaluate_root_dna_gamma.argtypes = \
-        [i, i, i, i, i, i, d, p, p, p, p, l, p, l, p, l, p, p, p, p]
+        [i, i, i, i, i, i, d, p, p, p, p, l, p, l, p, l, p, p, p, p, p]
     lib.examl_hip_sum_root_dna_gamma.argtypes = \
         [i, i, i, i, i, p, p, l, p, l, p, l, p]
-    lib.examl_hip_core_root_dna_gamma.argtypes = [l, p, p, p, d, p, p, p, p]
+    lib.examl_hip_core_root_dna_gamma.argtypes = \
+        [l, p, p, p, d, p, p, p, p, p]
     # protein surface (same shapes, states=20)
     lib.examl_host_init_gtr_aa.argtypes = [p, p, p, p, p, p]
     lib.examl_host_core_dtables_prot.argtypes = [p, p, d, p]
     lib.examl_hip_newview_prot_gamma.argtypes = \
         [i, p, p, p, p, p, p, p, l, p, p, p, p, p]
     lib.examl_hip_evaluate_prot_gamma.argtypes = \
-        [p, p, p, p, p, l, p, p, p, d, p, p]
+        [p, p, p, p, p, l, p, p, p, d, p, p, p]
     lib.examl_hip_sum_prot_gamma.argtypes = [i, p, p, p, p, p, p, l, p]
-    lib.examl_hip_core_prot_gamma.argtypes = [l, p, p, p, p, p]
+    lib.examl_hip_core_prot_gamma.argtypes = [l, p, p, p, p, p, p]
     lib.examl_hip_newview_traversal_prot_gamma.argtypes = \
         [p, i, p, p, p, p, p, p, l, p, l, p, l, p, p, p, p]
     lib.examl_hip_evaluate_root_prot_gamma.argtypes = \
-        [i, i, i, i, i, i, d, p, p, p, p, l, p, l, p, l, p, p, p, p]
+        [i, i, i, i, i, i, d, p, p, p, p, l, p, l, p, l, p, p, p, p, p]
     lib.examl_hip_sum_root_prot_gamma.argtypes = \
         [i, i, i, i, i, p, p, l, p, l, p, l, p]
-    lib.examl_hip_core_root_prot_gamma.argtypes = [l, p, p, p, d, p, p, p, p]
+    lib.examl_hip_core_root_prot_gamma.argtypes = \
+        [l, p, p, p, d, p, p, p, p, p]
     lib.examl_hip_profile_enable.argtypes = [i]
     lib.examl_hip_profile_reset.argtypes = []
     lib.examl_hip_profile_get.argtypes = [p, p]

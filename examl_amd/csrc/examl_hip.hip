@@ -191,8 +191,7 @@ __global__ __launch_bounds__(NV_BLOCK) void k_evaluate_dna_gamma(
     const double *__restrict__ x1, const double *__restrict__ x2,
     const double *__restrict__ tipVec, const unsigned char *__restrict__ tipX1,
     const int *__restrict__ wgt, const double *__restrict__ diag, long n,
-    const unsigned int *__restrict__ gsP, const unsigned int *__restrict__ gsQ,
-    double log_minlik, double *__restrict__ lnlOut) {
+    double *__restrict__ partials) {
   __shared__ double sD[16], sTV[64], sRed[NV_BLOCK / 64];
   const int tid = threadIdx.x;
   if (tid < 16) sD[tid] = diag[tid];
@@ -231,9 +230,7 @@ __global__ __launch_bounds__(NV_BLOCK) void k_evaluate_dna_gamma(
     double s = 0;
 #pragma unroll
     for (int w = 0; w < NV_BLOCK / 64; w++) s += sRed[w];
-    if (blockIdx.x == 0 && gsP != nullptr)
-      s += ((double)(*gsP) + (double)(*gsQ)) * log_minlik;
-    atomicAdd(lnlOut, s);
+    partials[blockIdx.x] = s; /* deterministic 2-pass reduction */
   }
 }
 
@@ -282,7 +279,7 @@ __global__ __launch_bounds__(NV_BLOCK) void k_sum_dna_gamma(
  */
 __global__ __launch_bounds__(NV_BLOCK) void k_core_dna_gamma(
     const double *__restrict__ sum, const double *__restrict__ dtab,
-    const int *__restrict__ wgt, long n, double *__restrict__ out2) {
+    const int *__restrict__ wgt, long n, double *__restrict__ partials) {
   __shared__ double sD0[16], sD1[16], sD2[16], sRed[2][NV_BLOCK / 64];
   const int tid = threadIdx.x;
   if (tid < 16) {
@@ -339,8 +336,8 @@ __global__ __launch_bounds__(NV_BLOCK) void k_core_dna_gamma(
       s1 += sRed[0][w];
       s2 += sRed[1][w];
     }
-    atomicAdd(&out2[0], s1);
-    atomicAdd(&out2[1], s2);
+    partials[2 * blockIdx.x] = s1; /* deterministic 2-pass reduction */
+    partials[2 * blockIdx.x + 1] = s2;
   }
 }
 
@@ -477,8 +474,7 @@ __global__ __launch_bounds__(NV_BLOCK) void k_evaluate_prot_gamma(
     const double *__restrict__ x1, const double *__restrict__ x2,
     const double *__restrict__ tipVec, const unsigned char *__restrict__ tipX1,
     const int *__restrict__ wgt, const double *__restrict__ diag, long n,
-    const unsigned int *__restrict__ gsP, const unsigned int *__restrict__ gsQ,
-    double log_minlik, double *__restrict__ lnlOut) {
+    double *__restrict__ partials) {
   __shared__ double sD[80], sTV[TIP ? 460 : 1], sRed[NV_BLOCK / 64];
   const int tid = threadIdx.x;
   for (int j = tid; j < 80; j += NV_BLOCK) sD[j] = diag[j];
@@ -515,9 +511,7 @@ __global__ __launch_bounds__(NV_BLOCK) void k_evaluate_prot_gamma(
     double s = 0;
 #pragma unroll
     for (int w = 0; w < NV_BLOCK / 64; w++) s += sRed[w];
-    if (blockIdx.x == 0 && gsP != nullptr)
-      s += ((double)(*gsP) + (double)(*gsQ)) * log_minlik;
-    atomicAdd(lnlOut, s);
+    partials[blockIdx.x] = s; /* deterministic 2-pass reduction */
   }
 }
 
@@ -569,7 +563,7 @@ __global__ __launch_bounds__(NV_BLOCK) void k_sum_prot_gamma(
 /* coreGTRGAMMAPROT (makenewzGenericSpecial.c:2581); dtab = {d0,d1,d2}[80] */
 __global__ __launch_bounds__(NV_BLOCK) void k_core_prot_gamma(
     const double *__restrict__ sum, const double *__restrict__ dtab,
-    const int *__restrict__ wgt, long n, double *__restrict__ out2) {
+    const int *__restrict__ wgt, long n, double *__restrict__ partials) {
   __shared__ double sD0[80], sD1[80], sD2[80], sRed[2][NV_BLOCK / 64];
   const int tid = threadIdx.x;
   for (int j = tid; j < 80; j += NV_BLOCK) {
@@ -627,8 +621,63 @@ __global__ __launch_bounds__(NV_BLOCK) void k_core_prot_gamma(
       s1 += sRed[0][w];
       s2 += sRed[1][w];
     }
-    atomicAdd(&out2[0], s1);
-    atomicAdd(&out2[1], s2);
+    partials[2 * blockIdx.x] = s1; /* deterministic 2-pass reduction */
+    partials[2 * blockIdx.x + 1] = s2;
+  }
+}
+
+/* --- deterministic final reductions ----------------------------------------
+ * One 256-thread block folds the per-block partials in a FIXED order, so
+ * lnL and the NR derivatives are bit-reproducible across runs (the
+ * reference's rationale for its Reduce+Bcast alternative,
+ * makenewzGenericSpecial.c:1242).  The scaler undo (gs_p+gs_q)*log_minlik
+ * (evaluateGenericSpecial.c:830) is applied here.
+ */
+__global__ __launch_bounds__(NV_BLOCK) void k_reduce_lnl(
+    const double *__restrict__ partials, int nblocks,
+    const unsigned int *__restrict__ gsP, const unsigned int *__restrict__ gsQ,
+    double log_minlik, double *__restrict__ lnlOut) {
+  __shared__ double sred[NV_BLOCK];
+  const int tid = threadIdx.x;
+  double v = 0;
+  for (int i = tid; i < nblocks; i += NV_BLOCK) v += partials[i];
+  sred[tid] = v;
+  __syncthreads();
+  for (int off = NV_BLOCK / 2; off > 0; off >>= 1) {
+    if (tid < off) sred[tid] += sred[tid + off];
+    __syncthreads();
+  }
+  if (tid == 0) {
+    double s = sred[0];
+    if (gsP != nullptr)
+      s += ((double)(*gsP) + (double)(*gsQ)) * log_minlik;
+    *lnlOut += s;
+  }
+}
+
+__global__ __launch_bounds__(NV_BLOCK) void k_reduce_2(
+    const double *__restrict__ partials, int nblocks,
+    double *__restrict__ out2) {
+  __shared__ double sred[2][NV_BLOCK];
+  const int tid = threadIdx.x;
+  double v1 = 0, v2 = 0;
+  for (int i = tid; i < nblocks; i += NV_BLOCK) {
+    v1 += partials[2 * i];
+    v2 += partials[2 * i + 1];
+  }
+  sred[0][tid] = v1;
+  sred[1][tid] = v2;
+  __syncthreads();
+  for (int off = NV_BLOCK / 2; off > 0; off >>= 1) {
+    if (tid < off) {
+      sred[0][tid] += sred[0][tid + off];
+      sred[1][tid] += sred[1][tid + off];
+    }
+    __syncthreads();
+  }
+  if (tid == 0) {
+    out2[0] += sred[0][0];
+    out2[1] += sred[1][0];
   }
 }
 
@@ -790,7 +839,7 @@ extern "C" int examl_hip_evaluate_dna_gamma(
     const int *wgt, const double *x1, const double *x2, const double *tipVec,
     const unsigned char *tipX1, long n, const double *diag,
     const unsigned int *gsP, const unsigned int *gsQ, double log_minlik,
-    double *lnl, void *stream) {
+    double *dev_partials, double *lnl, void *stream) {
   if (n <= 0) return 0;
   hipStream_t s = (hipStream_t)stream;
   (void)hipGetLastError(); /* clear stale per-thread error (e.g. torch probes) */
@@ -798,11 +847,14 @@ extern "C" int examl_hip_evaluate_dna_gamma(
   if (tipX1)
     hipLaunchKernelGGL((k_evaluate_dna_gamma<true>), dim3(grid),
                        dim3(NV_BLOCK), 0, s, x1, x2, tipVec, tipX1, wgt, diag,
-                       n, gsP, gsQ, log_minlik, lnl);
+                       n, dev_partials);
   else
     hipLaunchKernelGGL((k_evaluate_dna_gamma<false>), dim3(grid),
                        dim3(NV_BLOCK), 0, s, x1, x2, tipVec, tipX1, wgt, diag,
-                       n, gsP, gsQ, log_minlik, lnl);
+                       n, dev_partials);
+  CHK(hipGetLastError());
+  hipLaunchKernelGGL(k_reduce_lnl, dim3(1), dim3(NV_BLOCK), 0, s,
+                     dev_partials, grid, gsP, gsQ, log_minlik, lnl);
   CHK(hipGetLastError());
   return 0;
 }
@@ -843,12 +895,17 @@ extern "C" int examl_hip_sum_dna_gamma(int tipCase, double *sum,
 
 extern "C" int examl_hip_core_dna_gamma(long n, const double *sum,
                                         const double *dtab, const int *wgt,
-                                        double *out2, void *stream) {
+                                        double *dev_partials, double *out2,
+                                        void *stream) {
   if (n <= 0) return 0;
   hipStream_t s = (hipStream_t)stream;
   (void)hipGetLastError(); /* clear stale per-thread error (e.g. torch probes) */
-  hipLaunchKernelGGL(k_core_dna_gamma, dim3(grid_for(n * 4)), dim3(NV_BLOCK),
-                     0, s, sum, dtab, wgt, n, out2);
+  const int grid = grid_for(n * 4);
+  hipLaunchKernelGGL(k_core_dna_gamma, dim3(grid), dim3(NV_BLOCK), 0, s, sum,
+                     dtab, wgt, n, dev_partials);
+  CHK(hipGetLastError());
+  hipLaunchKernelGGL(k_reduce_2, dim3(1), dim3(NV_BLOCK), 0, s, dev_partials,
+                     grid, out2);
   CHK(hipGetLastError());
   return 0;
 }
@@ -1188,8 +1245,8 @@ extern "C" int examl_hip_evaluate_root_dna_gamma(
     int tipSlot, double z, const double *EIGN, const double *gammaRates,
     const double *dev_tipVec, double *dev_clv, long clvStride,
     const unsigned char *dev_tips, long tipStride, const int *dev_wgt, long n,
-    const unsigned int *dev_scalers, double *dev_diag, double *dev_lnl,
-    void *stream) {
+    const unsigned int *dev_scalers, double *dev_diag, double *dev_partials,
+    double *dev_lnl, void *stream) {
   if (n <= 0) return 0;
   hipStream_t s = (hipStream_t)stream;
   (void)hipGetLastError(); /* clear stale per-thread error (e.g. torch probes) */
@@ -1206,18 +1263,21 @@ extern "C" int examl_hip_evaluate_root_dna_gamma(
     const double *x2 = dev_clv + (long)x2Slot * clvStride;
     hipLaunchKernelGGL((k_evaluate_dna_gamma<true>), dim3(grid),
                        dim3(NV_BLOCK), 0, s, nullptr, x2, dev_tipVec, t1,
-                       dev_wgt, dev_diag, n, gsP, gsQ, log_minlik, dev_lnl);
+                       dev_wgt, dev_diag, n, dev_partials);
   } else if (rootTipCase == EXAML_INNER_INNER) {
     const double *x1 = dev_clv + (long)x1Slot * clvStride;
     const double *x2 = dev_clv + (long)x2Slot * clvStride;
     hipLaunchKernelGGL((k_evaluate_dna_gamma<false>), dim3(grid),
                        dim3(NV_BLOCK), 0, s, x1, x2, dev_tipVec, nullptr,
-                       dev_wgt, dev_diag, n, gsP, gsQ, log_minlik, dev_lnl);
+                       dev_wgt, dev_diag, n, dev_partials);
   } else {
     snprintf(g_err, sizeof(g_err), "evaluate_root: bad tipCase %d",
              rootTipCase);
     return -1;
   }
+  CHK(hipGetLastError());
+  hipLaunchKernelGGL(k_reduce_lnl, dim3(1), dim3(NV_BLOCK), 0, s,
+                     dev_partials, grid, gsP, gsQ, log_minlik, dev_lnl);
   CHK(hipGetLastError());
   return 0;
 }
@@ -1257,6 +1317,7 @@ extern "C" int examl_hip_core_root_dna_gamma(long n, const double *dev_sum,
                                              const double *gammaRates,
                                              double lz, const int *dev_wgt,
                                              double *dev_dtab,
+                                             double *dev_partials,
                                              double *dev_out2, void *stream) {
   if (n <= 0) return 0;
   hipStream_t s = (hipStream_t)stream;
@@ -1265,8 +1326,8 @@ extern "C" int examl_hip_core_root_dna_gamma(long n, const double *dev_sum,
   examl_host_core_dtables_dna(EIGN, gammaRates, lz, host48);
   CHK(hipMemcpyAsync(dev_dtab, host48, sizeof(host48), hipMemcpyHostToDevice,
                      s));
-  return examl_hip_core_dna_gamma(n, dev_sum, dev_dtab, dev_wgt, dev_out2,
-                                  stream);
+  return examl_hip_core_dna_gamma(n, dev_sum, dev_dtab, dev_wgt,
+                                  dev_partials, dev_out2, stream);
 }
 
 /* ===========================================================================
@@ -1321,7 +1382,7 @@ extern "C" int examl_hip_evaluate_prot_gamma(
     const int *wgt, const double *x1, const double *x2, const double *tipVec,
     const unsigned char *tipX1, long n, const double *diag,
     const unsigned int *gsP, const unsigned int *gsQ, double log_minlik,
-    double *lnl, void *stream) {
+    double *dev_partials, double *lnl, void *stream) {
   if (n <= 0) return 0;
   hipStream_t s = (hipStream_t)stream;
   (void)hipGetLastError();
@@ -1329,11 +1390,14 @@ extern "C" int examl_hip_evaluate_prot_gamma(
   if (tipX1)
     hipLaunchKernelGGL((k_evaluate_prot_gamma<true>), dim3(grid),
                        dim3(NV_BLOCK), 0, s, x1, x2, tipVec, tipX1, wgt, diag,
-                       n, gsP, gsQ, log_minlik, lnl);
+                       n, dev_partials);
   else
     hipLaunchKernelGGL((k_evaluate_prot_gamma<false>), dim3(grid),
                        dim3(NV_BLOCK), 0, s, x1, x2, tipVec, tipX1, wgt, diag,
-                       n, gsP, gsQ, log_minlik, lnl);
+                       n, dev_partials);
+  CHK(hipGetLastError());
+  hipLaunchKernelGGL(k_reduce_lnl, dim3(1), dim3(NV_BLOCK), 0, s,
+                     dev_partials, grid, gsP, gsQ, log_minlik, lnl);
   CHK(hipGetLastError());
   return 0;
 }
@@ -1374,12 +1438,17 @@ extern "C" int examl_hip_sum_prot_gamma(int tipCase, double *sum,
 
 extern "C" int examl_hip_core_prot_gamma(long n, const double *sum,
                                          const double *dtab, const int *wgt,
-                                         double *out2, void *stream) {
+                                         double *dev_partials, double *out2,
+                                         void *stream) {
   if (n <= 0) return 0;
   hipStream_t s = (hipStream_t)stream;
   (void)hipGetLastError();
-  hipLaunchKernelGGL(k_core_prot_gamma, dim3(grid_for(n * 4)), dim3(NV_BLOCK),
-                     0, s, sum, dtab, wgt, n, out2);
+  const int grid = grid_for(n * 4);
+  hipLaunchKernelGGL(k_core_prot_gamma, dim3(grid), dim3(NV_BLOCK), 0, s,
+                     sum, dtab, wgt, n, dev_partials);
+  CHK(hipGetLastError());
+  hipLaunchKernelGGL(k_reduce_2, dim3(1), dim3(NV_BLOCK), 0, s, dev_partials,
+                     grid, out2);
   CHK(hipGetLastError());
   return 0;
 }
@@ -1404,8 +1473,8 @@ extern "C" int examl_hip_evaluate_root_prot_gamma(
     int tipSlot, double z, const double *EIGN, const double *gammaRates,
     const double *dev_tipVec, double *dev_clv, long clvStride,
     const unsigned char *dev_tips, long tipStride, const int *dev_wgt, long n,
-    const unsigned int *dev_scalers, double *dev_diag, double *dev_lnl,
-    void *stream) {
+    const unsigned int *dev_scalers, double *dev_diag, double *dev_partials,
+    double *dev_lnl, void *stream) {
   if (n <= 0) return 0;
   hipStream_t s = (hipStream_t)stream;
   (void)hipGetLastError();
@@ -1422,18 +1491,21 @@ extern "C" int examl_hip_evaluate_root_prot_gamma(
     const double *x2 = dev_clv + (long)x2Slot * clvStride;
     hipLaunchKernelGGL((k_evaluate_prot_gamma<true>), dim3(grid),
                        dim3(NV_BLOCK), 0, s, nullptr, x2, dev_tipVec, t1,
-                       dev_wgt, dev_diag, n, gsP, gsQ, log_minlik, dev_lnl);
+                       dev_wgt, dev_diag, n, dev_partials);
   } else if (rootTipCase == EXAML_INNER_INNER) {
     const double *x1 = dev_clv + (long)x1Slot * clvStride;
     const double *x2 = dev_clv + (long)x2Slot * clvStride;
     hipLaunchKernelGGL((k_evaluate_prot_gamma<false>), dim3(grid),
                        dim3(NV_BLOCK), 0, s, x1, x2, dev_tipVec, nullptr,
-                       dev_wgt, dev_diag, n, gsP, gsQ, log_minlik, dev_lnl);
+                       dev_wgt, dev_diag, n, dev_partials);
   } else {
     snprintf(g_err, sizeof(g_err), "evaluate_root_prot: bad tipCase %d",
              rootTipCase);
     return -1;
   }
+  CHK(hipGetLastError());
+  hipLaunchKernelGGL(k_reduce_lnl, dim3(1), dim3(NV_BLOCK), 0, s,
+                     dev_partials, grid, gsP, gsQ, log_minlik, dev_lnl);
   CHK(hipGetLastError());
   return 0;
 }
@@ -1470,7 +1542,7 @@ extern "C" int examl_hip_sum_root_prot_gamma(
 extern "C" int examl_hip_core_root_prot_gamma(
     long n, const double *dev_sum, const double *EIGN,
     const double *gammaRates, double lz, const int *dev_wgt,
-    double *dev_dtab, double *dev_out2, void *stream) {
+    double *dev_dtab, double *dev_partials, double *dev_out2, void *stream) {
   if (n <= 0) return 0;
   hipStream_t s = (hipStream_t)stream;
   (void)hipGetLastError();
@@ -1478,8 +1550,8 @@ extern "C" int examl_hip_core_root_prot_gamma(
   examl_host_core_dtables_prot(EIGN, gammaRates, lz, host240);
   CHK(hipMemcpyAsync(dev_dtab, host240, sizeof(host240),
                      hipMemcpyHostToDevice, s));
-  return examl_hip_core_prot_gamma(n, dev_sum, dev_dtab, dev_wgt, dev_out2,
-                                   stream);
+  return examl_hip_core_prot_gamma(n, dev_sum, dev_dtab, dev_wgt,
+                                   dev_partials, dev_out2, stream);
 }
 
 #undef CHK

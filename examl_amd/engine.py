@@ -68,6 +68,8 @@ class DnaGammaEngine:
                                   device=dev)
         self.d_dtab = torch.empty(12 * self.states, dtype=torch.float64,
                                   device=dev)
+        self.d_partials = torch.empty(2 * 8192, dtype=torch.float64,
+                                      device=dev)
         self.d_lnl = torch.zeros(1, dtype=torch.float64, device=dev)
         self.d_out2 = torch.zeros(2, dtype=torch.float64, device=dev)
         self.d_sum = None  # sumBuffer (axml.h:558), allocated on first use
@@ -141,7 +143,8 @@ class DnaGammaEngine:
             _vp(self.d_clv), ctypes.c_long(self.width * self.SPAN),
             _vp(self.d_tips), ctypes.c_long(self.width), _vp(self.d_wgt),
             ctypes.c_long(self.width), _vp(self.d_scalers), _vp(self.d_diag),
-            _vp(self.d_lnl), self._stream()), "evaluate_root")
+            _vp(self.d_partials), _vp(self.d_lnl), self._stream()),
+            "evaluate_root")
         if all_reduce and torch.distributed.is_initialized():
             torch.distributed.all_reduce(self.d_lnl)
         return self.d_lnl
@@ -187,8 +190,8 @@ class DnaGammaEngine:
         check(self._fn("core_root")(
             ctypes.c_long(self.width), _vp(self.d_sum), _np_vp(m.EIGN),
             _np_vp(m.gammaRates), ctypes.c_double(lz), _vp(self.d_wgt),
-            _vp(self.d_dtab), _vp(self.d_out2), self._stream()),
-            "core_root")
+            _vp(self.d_dtab), _vp(self.d_partials), _vp(self.d_out2),
+            self._stream()), "core_root")
         if all_reduce and torch.distributed.is_initialized():
             torch.distributed.all_reduce(self.d_out2)
         v = self.d_out2.cpu()

@@ -122,8 +122,8 @@ int examl_hip_evaluate_dna_gamma(const int *dev_wgt, const double *dev_x1,
                                  const double *dev_diag,
                                  const unsigned int *dev_gsP,
                                  const unsigned int *dev_gsQ,
-                                 double log_minlik, double *dev_lnl,
-                                 void *stream);
+                                 double log_minlik, double *dev_partials,
+                                 double *dev_lnl, void *stream);
 
 /* replaces sumGAMMA (examl/makenewzGenericSpecial.c:1798):
  * dev_sum[i,c,k] = x1'[i,c,k] * x2'[i,c,k] with tip expansion. */
@@ -139,7 +139,8 @@ int examl_hip_sum_dna_gamma(int tipCase, double *dev_sum,
  * the 48 doubles from examl_host_core_dtables_dna, uploaded by the caller. */
 int examl_hip_core_dna_gamma(long n, const double *dev_sum,
                              const double *dev_dtables, const int *dev_wgt,
-                             double *dev_out2, void *stream);
+                             double *dev_partials, double *dev_out2,
+                             void *stream);
 
 /* ---------------------------------------------------------------------------
  * L1 batched executors (the newviewIterative-shaped path: one host call per
@@ -187,7 +188,7 @@ int examl_hip_evaluate_root_dna_gamma(
     const double *dev_tipVector, double *dev_clv, long clvStrideDoubles,
     const unsigned char *dev_tips, long tipStrideBytes, const int *dev_wgt,
     long n, const unsigned int *dev_scalers, double *dev_diag_scratch,
-    double *dev_lnl, void *stream);
+    double *dev_partials, double *dev_lnl, void *stream);
 
 /* replaces the per-partition body of makenewzIterative's sum precompute
  * (examl/makenewzGenericSpecial.c:628,673-839) for the branch p--q. */
@@ -208,7 +209,8 @@ int examl_hip_core_root_dna_gamma(long n, const double *dev_sum,
                                   const double *EIGN,
                                   const double *gammaRates, double lz,
                                   const int *dev_wgt,
-                                  double *dev_dtab_scratch, double *dev_out2,
+                                  double *dev_dtab_scratch,
+                                  double *dev_partials, double *dev_out2,
                                   void *stream);
 
 /* ---------------------------------------------------------------------------
@@ -249,8 +251,8 @@ int examl_hip_evaluate_prot_gamma(const int *dev_wgt, const double *dev_x1,
                                   const double *dev_diag,
                                   const unsigned int *dev_gsP,
                                   const unsigned int *dev_gsQ,
-                                  double log_minlik, double *dev_lnl,
-                                  void *stream);
+                                  double log_minlik, double *dev_partials,
+                                  double *dev_lnl, void *stream);
 
 int examl_hip_sum_prot_gamma(int tipCase, double *dev_sum,
                              const double *dev_x1, const double *dev_x2,
@@ -261,7 +263,8 @@ int examl_hip_sum_prot_gamma(int tipCase, double *dev_sum,
 
 int examl_hip_core_prot_gamma(long n, const double *dev_sum,
                               const double *dev_dtables, const int *dev_wgt,
-                              double *dev_out2, void *stream);
+                              double *dev_partials, double *dev_out2,
+                              void *stream);
 
 int examl_hip_newview_traversal_prot_gamma(
     const examl_hip_trav_entry *ops, int numOps, const double *EIGN,
@@ -277,7 +280,7 @@ int examl_hip_evaluate_root_prot_gamma(
     const double *dev_tipVector, double *dev_clv, long clvStrideDoubles,
     const unsigned char *dev_tips, long tipStrideBytes, const int *dev_wgt,
     long n, const unsigned int *dev_scalers, double *dev_diag_scratch,
-    double *dev_lnl, void *stream);
+    double *dev_partials, double *dev_lnl, void *stream);
 
 int examl_hip_sum_root_prot_gamma(int rootTipCase, int x1Slot, int x2Slot,
                                   int tipSlot, int tipSlot2,
@@ -292,7 +295,8 @@ int examl_hip_core_root_prot_gamma(long n, const double *dev_sum,
                                    const double *gammaRates, double lz,
                                    const int *dev_wgt,
                                    double *dev_dtab_scratch,
-                                   double *dev_out2, void *stream);
+                                   double *dev_partials, double *dev_out2,
+                                   void *stream);
 
 #ifdef __cplusplus
 }

@@ -81,17 +81,20 @@ def test_evaluate_kernel_vs_golden(kern, model, dev):
     d_wgt = _to_dev(kern["norm_wgt"], dev)
     d_diag = _to_dev(kern["diag"], dev)
     d_lnl = torch.zeros(1, dtype=torch.float64, device=dev)
+    d_part = torch.zeros(8192, dtype=torch.float64, device=dev)
     ea.check(ea.lib().examl_hip_evaluate_dna_gamma(
         vp(d_wgt), vp(d_x1), vp(d_x2), vp(d_tv), ctypes.c_void_p(0),
         ctypes.c_long(n), vp(d_diag), ctypes.c_void_p(0), ctypes.c_void_p(0),
-        ctypes.c_double(0.0), vp(d_lnl), ctypes.c_void_p(0)), "evaluate")
+        ctypes.c_double(0.0), vp(d_part), vp(d_lnl), ctypes.c_void_p(0)),
+        "evaluate")
     torch.cuda.synchronize()
     assert np.isclose(d_lnl.item(), float(kern["eval_II"]), rtol=1e-12)
     d_lnl.zero_()
     ea.check(ea.lib().examl_hip_evaluate_dna_gamma(
         vp(d_wgt), ctypes.c_void_p(0), vp(d_x2), vp(d_tv), vp(d_t1),
         ctypes.c_long(n), vp(d_diag), ctypes.c_void_p(0), ctypes.c_void_p(0),
-        ctypes.c_double(0.0), vp(d_lnl), ctypes.c_void_p(0)), "evaluate")
+        ctypes.c_double(0.0), vp(d_part), vp(d_lnl), ctypes.c_void_p(0)),
+        "evaluate")
     torch.cuda.synchronize()
     assert np.isclose(d_lnl.item(), float(kern["eval_TIP"]), rtol=1e-12)
 
@@ -119,12 +122,13 @@ def test_sum_and_core_kernels_vs_golden(kern, model, dev, tc):
     d_wgt = _to_dev(kern["norm_wgt"], dev)
     d_dtab = torch.zeros(48, dtype=torch.float64, device=dev)
     d_out2 = torch.zeros(2, dtype=torch.float64, device=dev)
+    d_part = torch.zeros(2 * 8192, dtype=torch.float64, device=dev)
     ea.check(ea.lib().examl_hip_core_root_dna_gamma(
         ctypes.c_long(n), vp(d_sum),
         model.EIGN.ctypes.data_as(ctypes.c_void_p),
         model.gammaRates.ctypes.data_as(ctypes.c_void_p),
         ctypes.c_double(float(kern["lz_core"])), vp(d_wgt), vp(d_dtab),
-        vp(d_out2), ctypes.c_void_p(0)), "core")
+        vp(d_part), vp(d_out2), ctypes.c_void_p(0)), "core")
     torch.cuda.synchronize()
     out = d_out2.cpu().numpy()
     assert np.isclose(out[0], float(kern[f"core_tc{tc}_d1"]), rtol=1e-11)
@@ -201,3 +205,26 @@ def test_makenewz_vs_oracle(dev):
     before = eng.evaluate_root(tree, p, q, z0).item()
     after = eng.evaluate_root(tree, p, q, z_gpu).item()
     assert after >= before - 0.01
+
+
+def test_lnl_bitwise_deterministic(dev):
+    """The two-pass fixed-order reductions make lnL and the NR derivatives
+    bit-reproducible across repeated evaluations and engine instances (the
+    reference's determinism rationale, makenewzGenericSpecial.c:1242)."""
+    ntips, width = 20, 100000
+    tips, wgt = make_synthetic(ntips, width, seed=8)
+    model = ea.DnaGtrModel([0.3, 0.2, 0.25, 0.25],
+                           [1.1, 2.4, 0.8, 0.9, 3.0, 1.0], alpha=0.7)
+    tree = ea.PhyloTree.random(ntips, seed=2, rng_z=True)
+    vals = []
+    derivs = []
+    for _ in range(2):
+        eng = ea.DnaGammaEngine(tips, wgt, model, device=dev)
+        entries, (p, q, z) = tree.full_traversal()
+        eng.newview_traversal(entries)
+        vals.append(eng.evaluate_root(tree, p, q, z).item())
+        vals.append(eng.evaluate_root(tree, p, q, z).item())
+        eng.sum_root(tree, p, q)
+        derivs.append(eng.core_derivs(float(np.log(z))))
+    assert vals[0] == vals[1] == vals[2] == vals[3]
+    assert derivs[0] == derivs[1]

@@ -75,17 +75,18 @@ def test_evaluate_prot_kernel_vs_golden(kern, dev):
     d_wgt = _to_dev(kern["norm_wgt"], dev)
     d_diag = _to_dev(kern["diag"], dev)
     d_lnl = torch.zeros(1, dtype=torch.float64, device=dev)
+    d_part = torch.zeros(8192, dtype=torch.float64, device=dev)
     null = ctypes.c_void_p(0)
     ea.check(ea.lib().examl_hip_evaluate_prot_gamma(
         vp(d_wgt), vp(d_x1), vp(d_x2), vp(d_tv), null, ctypes.c_long(n),
-        vp(d_diag), null, null, ctypes.c_double(0.0), vp(d_lnl),
+        vp(d_diag), null, null, ctypes.c_double(0.0), vp(d_part), vp(d_lnl),
         ctypes.c_void_p(0)), "evaluate_prot")
     torch.cuda.synchronize()
     assert np.isclose(d_lnl.item(), float(kern["eval_II"]), rtol=1e-12)
     d_lnl.zero_()
     ea.check(ea.lib().examl_hip_evaluate_prot_gamma(
         vp(d_wgt), null, vp(d_x2), vp(d_tv), vp(d_t1), ctypes.c_long(n),
-        vp(d_diag), null, null, ctypes.c_double(0.0), vp(d_lnl),
+        vp(d_diag), null, null, ctypes.c_double(0.0), vp(d_part), vp(d_lnl),
         ctypes.c_void_p(0)), "evaluate_prot")
     torch.cuda.synchronize()
     assert np.isclose(d_lnl.item(), float(kern["eval_TIP"]), rtol=1e-12)
@@ -115,6 +116,7 @@ def test_sum_core_prot_kernels(kern, dev, tc):
     d_wgt = _to_dev(kern["norm_wgt"], dev)
     d_dtab = torch.zeros(240, dtype=torch.float64, device=dev)
     d_out2 = torch.zeros(2, dtype=torch.float64, device=dev)
+    d_part = torch.zeros(2 * 8192, dtype=torch.float64, device=dev)
     EIGN = np.ascontiguousarray(kern["EIGN"])
     g = np.ascontiguousarray(kern["gammaRates"])
     ea.check(ea.lib().examl_hip_core_root_prot_gamma(
@@ -122,7 +124,7 @@ def test_sum_core_prot_kernels(kern, dev, tc):
         EIGN.ctypes.data_as(ctypes.c_void_p),
         g.ctypes.data_as(ctypes.c_void_p),
         ctypes.c_double(float(kern["lz_core"])), vp(d_wgt), vp(d_dtab),
-        vp(d_out2), ctypes.c_void_p(0)), "core_prot")
+        vp(d_part), vp(d_out2), ctypes.c_void_p(0)), "core_prot")
     torch.cuda.synchronize()
     out = d_out2.cpu().numpy()
     assert np.isclose(out[0], float(kern[f"core_tc{tc}_d1"]), rtol=1e-11)
