@@ -118,6 +118,10 @@ def main():
     ap.add_argument("--taxa", type=int, default=NTAXA)
     ap.add_argument("--protein", action="store_true",
                     help="config 4: 50 taxa x 200k sites, LG+GAMMA")
+    ap.add_argument("--partitions", type=int, default=1,
+                    help="partitions per GPU, each on its own HIP stream "
+                         "(config 3 shard shape: --partitions 16 "
+                         "--sites 125000)")
     ap.add_argument("--no-cpu-baseline", action="store_true")
     args = ap.parse_args()
 
@@ -145,15 +149,39 @@ def main():
         model = ea.DnaGtrModel([0.28, 0.22, 0.24, 0.26],
                                [1.2, 2.9, 0.7, 1.0, 3.2, 1.0], alpha=0.6)
     tree = ea.PhyloTree.random(ntips, seed=7)
-    eng = ea.DnaGammaEngine(tips, wgt, model, device=device)
+    P = args.partitions
+    assert width % P == 0
+    pw = width // P
+    engines = [ea.DnaGammaEngine(np.ascontiguousarray(tips[:, i*pw:(i+1)*pw]),
+                                 wgt[i*pw:(i+1)*pw], model, device=device)
+               for i in range(P)]
+    eng = engines[0]
+    streams = [torch.cuda.Stream(device=device) for _ in range(P)] \
+        if P > 1 else []
     entries, (p, q, z) = tree.full_traversal()
     n_ops = len(entries)
     tc_counts = [sum(1 for e in entries if e.tipCase == t) for t in range(3)]
 
     def step():
-        eng.newview_traversal(entries)
-        lnl = eng.evaluate_root(tree, p, q, z, all_reduce=world > 1)
-        return lnl
+        if P == 1:
+            eng.newview_traversal(entries)
+            return eng.evaluate_root(tree, p, q, z, all_reduce=world > 1)
+        # partitioned: one HIP stream per partition (the per-partition loop
+        # of newviewIterative/evaluateIterative runs concurrently), then ONE
+        # all-reduce of the per-partition lnL vector (the C1 collective)
+        cur = torch.cuda.current_stream(device)
+        for e_, st in zip(engines, streams):
+            st.wait_stream(cur)
+            with torch.cuda.stream(st):
+                e_.newview_traversal(entries)
+                e_.evaluate_root(tree, p, q, z)
+        for st in streams:
+            cur.wait_stream(st)
+        lnl_vec = torch.cat([e_.d_lnl for e_ in engines])
+        if world > 1:
+            import torch.distributed as dd
+            dd.all_reduce(lnl_vec)
+        return lnl_vec.sum()
 
     # warmup (also captures the traversal's hipGraph)
     for _ in range(args.warmup):
@@ -249,7 +277,7 @@ def main():
                         + "; full-tree evaluateGeneric per step",
             "taxa": ntips,
             "sites_per_gpu": width,
-            "partitions": 1,
+            "partitions": P,
             "newview_ops_per_step": n_ops,
             "tipcase_counts": {"TT": tc_counts[0], "TI": tc_counts[1],
                                "II": tc_counts[2]},
