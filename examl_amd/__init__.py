@@ -94,6 +94,14 @@ def _bind(lib):
     lib.examl_hip_profile_enable.argtypes = [i]
     lib.examl_hip_profile_reset.argtypes = []
     lib.examl_hip_profile_get.argtypes = [p, p]
+    lib.examl_host_core_dtables_dna_cat.argtypes = [p, p, i, d, p]
+    lib.examl_hip_newview_dna_cat.argtypes = \
+        [i, p, p, p, p, p, p, p, p, l, p, i, p, p, p]
+    lib.examl_hip_evaluate_dna_cat.argtypes = \
+        [p, p, p, p, p, p, l, p, i, p, p, d, p, p, p]
+    lib.examl_hip_sum_dna_cat.argtypes = [i, p, p, p, p, p, p, l, p]
+    lib.examl_hip_core_root_dna_cat.argtypes = \
+        [l, p, p, p, i, d, p, p, p, p, p, p]
     lib.examl_hip_use_graphs.argtypes = [i]
     lib.examl_hip_graphs_clear.argtypes = []
     return lib

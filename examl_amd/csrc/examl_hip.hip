@@ -342,6 +342,229 @@ __global__ __launch_bounds__(NV_BLOCK) void k_core_dna_gamma(
 }
 
 /* ===========================================================================
+ * DNA CAT (PSR) kernels — span 4, per-site rate category cptr[i]
+ * (the -m PSR mode).  One thread per site (32 B/lane
+ * coalesced); P rows (numCats<=25 pairs) and EV/tipVector staged in LDS;
+ * the rescale decision is thread-local (a site's whole span lives in one
+ * lane), counts via atomicAdd (rare).
+ * ==========================================================================*/
+
+#define MAX_CAT 25 /* maxCategories default, axml.h */
+
+template <int TC, bool NT>
+__global__ __launch_bounds__(NV_BLOCK) void k_newview_dna_cat(
+    const double *__restrict__ EV, const int *__restrict__ cptr,
+    const double *__restrict__ x1, const double *__restrict__ x2,
+    double *__restrict__ x3, const double *__restrict__ tipVec,
+    const unsigned char *__restrict__ tipX1,
+    const unsigned char *__restrict__ tipX2, const int *__restrict__ wgt,
+    long n, const double *__restrict__ P, int numCats,
+    unsigned int *__restrict__ scalerInc) {
+  __shared__ double sL[MAX_CAT * 16], sR[MAX_CAT * 16], sEV[16], sTV[64];
+  const int tid = threadIdx.x;
+  for (int j = tid; j < numCats * 16; j += NV_BLOCK) {
+    sL[j] = P[j];
+    sR[j] = P[numCats * 16 + j];
+  }
+  if (tid < 16) sEV[tid] = EV[tid];
+  if (tid < 64) sTV[tid] = tipVec[tid];
+  __syncthreads();
+
+  for (long i = (long)blockIdx.x * NV_BLOCK + tid; i < n;
+       i += (long)gridDim.x * NV_BLOCK) {
+    const int cat = cptr[i];
+    const double *le = &sL[cat * 16];
+    const double *ri = &sR[cat * 16];
+    double a[4], b[4];
+    if (TC == EXAML_TIP_TIP) {
+      const double *t1 = &sTV[4 * tipX1[i]];
+      const double *t2 = &sTV[4 * tipX2[i]];
+#pragma unroll
+      for (int s = 0; s < 4; s++) {
+        a[s] = t1[s];
+        b[s] = t2[s];
+      }
+    } else if (TC == EXAML_TIP_INNER) {
+      const double *t1 = &sTV[4 * tipX1[i]];
+      const double4 xr = *reinterpret_cast<const double4 *>(&x2[i * 4]);
+      a[0] = t1[0]; a[1] = t1[1]; a[2] = t1[2]; a[3] = t1[3];
+      b[0] = xr.x; b[1] = xr.y; b[2] = xr.z; b[3] = xr.w;
+    } else {
+      const double4 xl = *reinterpret_cast<const double4 *>(&x1[i * 4]);
+      const double4 xr = *reinterpret_cast<const double4 *>(&x2[i * 4]);
+      a[0] = xl.x; a[1] = xl.y; a[2] = xl.z; a[3] = xl.w;
+      b[0] = xr.x; b[1] = xr.y; b[2] = xr.z; b[3] = xr.w;
+    }
+    double v0 = 0, v1 = 0, v2 = 0, v3 = 0;
+#pragma unroll
+    for (int l = 0; l < 4; l++) {
+      const double u1 = (a[0] * le[l * 4] + a[1] * le[l * 4 + 1]) +
+                        (a[2] * le[l * 4 + 2] + a[3] * le[l * 4 + 3]);
+      const double u2 = (b[0] * ri[l * 4] + b[1] * ri[l * 4 + 1]) +
+                        (b[2] * ri[l * 4 + 2] + b[3] * ri[l * 4 + 3]);
+      const double t = u1 * u2;
+      v0 += t * sEV[l * 4 + 0];
+      v1 += t * sEV[l * 4 + 1];
+      v2 += t * sEV[l * 4 + 2];
+      v3 += t * sEV[l * 4 + 3];
+    }
+    if (TC != EXAML_TIP_TIP) {
+      if ((fabs(v0) < MINLIKELIHOOD) & (fabs(v1) < MINLIKELIHOOD) &
+          (fabs(v2) < MINLIKELIHOOD) & (fabs(v3) < MINLIKELIHOOD)) {
+        v0 *= TWOTOTHE256;
+        v1 *= TWOTOTHE256;
+        v2 *= TWOTOTHE256;
+        v3 *= TWOTOTHE256;
+        atomicAdd(scalerInc, (unsigned int)wgt[i]);
+      }
+    }
+    if (NT)
+      __builtin_nontemporal_store((v4d){v0, v1, v2, v3},
+                                  reinterpret_cast<v4d *>(&x3[i * 4]));
+    else
+      *reinterpret_cast<double4 *>(&x3[i * 4]) = make_double4(v0, v1, v2, v3);
+  }
+}
+
+/* evaluateGTRCAT (evaluateGenericSpecial.c:1988): per-site diag row,
+ * no 0.25 factor */
+template <bool TIP>
+__global__ __launch_bounds__(NV_BLOCK) void k_evaluate_dna_cat(
+    const int *__restrict__ cptr, const double *__restrict__ x1,
+    const double *__restrict__ x2, const double *__restrict__ tipVec,
+    const unsigned char *__restrict__ tipX1, const int *__restrict__ wgt,
+    const double *__restrict__ diag, int numCats, long n,
+    double *__restrict__ partials) {
+  __shared__ double sD[MAX_CAT * 4], sTV[TIP ? 64 : 1], sRed[NV_BLOCK / 64];
+  const int tid = threadIdx.x;
+  for (int j = tid; j < numCats * 4; j += NV_BLOCK) sD[j] = diag[j];
+  if (TIP && tid < 64) sTV[tid] = tipVec[tid];
+  __syncthreads();
+  const int lane = tid & 63;
+  double acc = 0.0;
+  for (long i = (long)blockIdx.x * NV_BLOCK + tid; i < n;
+       i += (long)gridDim.x * NV_BLOCK) {
+    const double *d = &sD[4 * cptr[i]];
+    const double4 b = *reinterpret_cast<const double4 *>(&x2[i * 4]);
+    double t0, t1;
+    if (TIP) {
+      const double *t = &sTV[4 * tipX1[i]];
+      t0 = t[0] * b.x * d[0] + t[2] * b.z * d[2];
+      t1 = t[1] * b.y * d[1] + t[3] * b.w * d[3];
+    } else {
+      const double4 a = *reinterpret_cast<const double4 *>(&x1[i * 4]);
+      t0 = a.x * b.x * d[0] + a.z * b.z * d[2];
+      t1 = a.y * b.y * d[1] + a.w * b.w * d[3];
+    }
+    acc += (double)wgt[i] * log(fabs(t0 + t1));
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off);
+  if (lane == 0) sRed[tid >> 6] = acc;
+  __syncthreads();
+  if (tid == 0) {
+    double s = 0;
+#pragma unroll
+    for (int w = 0; w < NV_BLOCK / 64; w++) s += sRed[w];
+    partials[blockIdx.x] = s;
+  }
+}
+
+/* sumCAT (makenewzGenericSpecial.c:1850) */
+template <int TC>
+__global__ __launch_bounds__(NV_BLOCK) void k_sum_dna_cat(
+    double *__restrict__ sum, const double *__restrict__ x1,
+    const double *__restrict__ x2, const double *__restrict__ tipVec,
+    const unsigned char *__restrict__ tipX1,
+    const unsigned char *__restrict__ tipX2, long n) {
+  __shared__ double sTV[64];
+  const int tid = threadIdx.x;
+  if (TC != EXAML_INNER_INNER && tid < 64) sTV[tid] = tipVec[tid];
+  if (TC != EXAML_INNER_INNER) __syncthreads();
+  for (long i = (long)blockIdx.x * NV_BLOCK + tid; i < n;
+       i += (long)gridDim.x * NV_BLOCK) {
+    double4 a, b;
+    if (TC == EXAML_TIP_TIP) {
+      const double *t1 = &sTV[4 * tipX1[i]];
+      const double *t2 = &sTV[4 * tipX2[i]];
+      a = make_double4(t1[0], t1[1], t1[2], t1[3]);
+      b = make_double4(t2[0], t2[1], t2[2], t2[3]);
+    } else if (TC == EXAML_TIP_INNER) {
+      const double *t1 = &sTV[4 * tipX1[i]];
+      a = make_double4(t1[0], t1[1], t1[2], t1[3]);
+      b = *reinterpret_cast<const double4 *>(&x2[i * 4]);
+    } else {
+      a = *reinterpret_cast<const double4 *>(&x1[i * 4]);
+      b = *reinterpret_cast<const double4 *>(&x2[i * 4]);
+    }
+    *reinterpret_cast<double4 *>(&sum[i * 4]) =
+        make_double4(a.x * b.x, a.y * b.y, a.z * b.z, a.w * b.w);
+  }
+}
+
+/* coreGTRCAT (makenewzGenericSpecial.c:2402).  dtab layout (host-built by
+ * examl_host_core_dtables_dna_cat): d[numCats*4] | e1[4] | e2[4] |
+ * rptr[numCats]. */
+__global__ __launch_bounds__(NV_BLOCK) void k_core_dna_cat(
+    const double *__restrict__ sum, const double *__restrict__ dtab,
+    const int *__restrict__ wgt, const int *__restrict__ cptr, int numCats,
+    long n, double *__restrict__ partials) {
+  __shared__ double sD[MAX_CAT * 4], sE1[4], sE2[4], sRp[MAX_CAT],
+      sRed[2][NV_BLOCK / 64];
+  const int tid = threadIdx.x;
+  for (int j = tid; j < numCats * 4; j += NV_BLOCK) sD[j] = dtab[j];
+  if (tid < 4) {
+    sE1[tid] = dtab[numCats * 4 + tid];
+    sE2[tid] = dtab[numCats * 4 + 4 + tid];
+  }
+  for (int j = tid; j < numCats; j += NV_BLOCK)
+    sRp[j] = dtab[numCats * 4 + 8 + j];
+  __syncthreads();
+  const int lane = tid & 63;
+  double accD1 = 0.0, accD2 = 0.0;
+  for (long i = (long)blockIdx.x * NV_BLOCK + tid; i < n;
+       i += (long)gridDim.x * NV_BLOCK) {
+    const int cat = cptr[i];
+    const double *d1 = &sD[4 * cat];
+    const double r = sRp[cat];
+    const double w = (double)wgt[i];
+    const double wr1 = r * w, wr2 = r * r * w;
+    const double4 s4 = *reinterpret_cast<const double4 *>(&sum[i * 4]);
+    const double te0 = d1[0] * s4.x, to0 = d1[1] * s4.y;
+    const double te1 = d1[2] * s4.z, to1 = d1[3] * s4.w;
+    const double a0 = (te0 + te1) + (to0 + to1);
+    const double a1 = (te0 * sE1[0] + te1 * sE1[2]) +
+                      (to0 * sE1[1] + to1 * sE1[3]);
+    const double a2 = (te0 * sE2[0] + te1 * sE2[2]) +
+                      (to0 * sE2[1] + to1 * sE2[3]);
+    const double inv = 1.0 / fabs(a0);
+    const double dl = a1 * inv, d2l = a2 * inv;
+    accD1 += wr1 * dl;
+    accD2 += wr2 * (d2l - dl * dl);
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    accD1 += __shfl_down(accD1, off);
+    accD2 += __shfl_down(accD2, off);
+  }
+  if (lane == 0) {
+    sRed[0][tid >> 6] = accD1;
+    sRed[1][tid >> 6] = accD2;
+  }
+  __syncthreads();
+  if (tid == 0) {
+    double s1 = 0, s2 = 0;
+#pragma unroll
+    for (int w = 0; w < NV_BLOCK / 64; w++) {
+      s1 += sRed[0][w];
+      s2 += sRed[1][w];
+    }
+    partials[2 * blockIdx.x] = s1;
+    partials[2 * blockIdx.x + 1] = s2;
+  }
+}
+
+/* ===========================================================================
  * Protein (20-state) GTRGAMMA kernels — span 80, tip codes 1..22.
  * Same thread <-> (site, cat) mapping; x rows held in registers (160 B per
  * operand per thread; the wave's loads cover a contiguous 10 KiB region so
@@ -775,6 +998,30 @@ extern "C" void examl_host_core_dtables_prot(const double *EIGN,
       d1[i * 20 + l] = EIGN[l] * ki;
       d2[i * 20 + l] = EIGN[l] * EIGN[l] * kisqr;
     }
+  }
+}
+
+extern "C" void examl_host_core_dtables_dna_cat(const double *EIGN,
+                                                const double *rptr,
+                                                int numCats, double lz,
+                                                double *out) {
+  /* restates the d/e1/e2 setup of coreGTRCAT,
+   * examl/makenewzGenericSpecial.c:2425-2450; layout
+   * d[numCats*4] | e1[4] | e2[4] | rptr[numCats] */
+  double *d = out, *e1 = out + numCats * 4, *e2 = e1 + 4, *rp = e2 + 4;
+  const double dd1 = EIGN[1] * lz, dd2 = EIGN[2] * lz, dd3 = EIGN[3] * lz;
+  for (int i = 0; i < numCats; i++) {
+    d[i * 4 + 0] = 1.0;
+    d[i * 4 + 1] = exp(dd1 * rptr[i]);
+    d[i * 4 + 2] = exp(dd2 * rptr[i]);
+    d[i * 4 + 3] = exp(dd3 * rptr[i]);
+    rp[i] = rptr[i];
+  }
+  e1[0] = 0.0;
+  e2[0] = 0.0;
+  for (int l = 1; l < 4; l++) {
+    e1[l] = EIGN[l];
+    e2[l] = EIGN[l] * EIGN[l];
   }
 }
 
@@ -1328,6 +1575,165 @@ extern "C" int examl_hip_core_root_dna_gamma(long n, const double *dev_sum,
                      s));
   return examl_hip_core_dna_gamma(n, dev_sum, dev_dtab, dev_wgt,
                                   dev_partials, dev_out2, stream);
+}
+
+/* ===========================================================================
+ * DNA CAT (PSR) launchers — L0 surface (the per-partition CAT dispatch of
+ * newviewIterative/evaluateIterative/makenewzIterative; the CAT traversal
+ * executor and optimizeRateCategories host loop are round-2 work)
+ * ==========================================================================*/
+
+extern "C" int examl_hip_newview_dna_cat(
+    int tipCase, const double *dev_EV, const int *dev_cptr, const double *x1,
+    const double *x2, double *x3, const double *dev_tipVec,
+    const unsigned char *tipX1, const unsigned char *tipX2, long n,
+    const double *dev_P, int numCats, const int *dev_wgt,
+    unsigned int *dev_scalerInc, void *stream) {
+  if (n <= 0) return 0;
+  if (numCats > MAX_CAT) {
+    snprintf(g_err, sizeof(g_err), "newview_cat: numCats %d > %d", numCats,
+             MAX_CAT);
+    return -1;
+  }
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  const int grid = grid_for(n);
+  const bool nt = n >= 262144; /* 32 B/site write */
+  switch (tipCase) {
+  case EXAML_TIP_TIP:
+    if (nt)
+      hipLaunchKernelGGL((k_newview_dna_cat<EXAML_TIP_TIP, true>), dim3(grid),
+                         dim3(NV_BLOCK), 0, s, dev_EV, dev_cptr, x1, x2, x3,
+                         dev_tipVec, tipX1, tipX2, dev_wgt, n, dev_P, numCats,
+                         dev_scalerInc);
+    else
+      hipLaunchKernelGGL((k_newview_dna_cat<EXAML_TIP_TIP, false>),
+                         dim3(grid), dim3(NV_BLOCK), 0, s, dev_EV, dev_cptr,
+                         x1, x2, x3, dev_tipVec, tipX1, tipX2, dev_wgt, n,
+                         dev_P, numCats, dev_scalerInc);
+    break;
+  case EXAML_TIP_INNER:
+    if (nt)
+      hipLaunchKernelGGL((k_newview_dna_cat<EXAML_TIP_INNER, true>),
+                         dim3(grid), dim3(NV_BLOCK), 0, s, dev_EV, dev_cptr,
+                         x1, x2, x3, dev_tipVec, tipX1, tipX2, dev_wgt, n,
+                         dev_P, numCats, dev_scalerInc);
+    else
+      hipLaunchKernelGGL((k_newview_dna_cat<EXAML_TIP_INNER, false>),
+                         dim3(grid), dim3(NV_BLOCK), 0, s, dev_EV, dev_cptr,
+                         x1, x2, x3, dev_tipVec, tipX1, tipX2, dev_wgt, n,
+                         dev_P, numCats, dev_scalerInc);
+    break;
+  case EXAML_INNER_INNER:
+    if (nt)
+      hipLaunchKernelGGL((k_newview_dna_cat<EXAML_INNER_INNER, true>),
+                         dim3(grid), dim3(NV_BLOCK), 0, s, dev_EV, dev_cptr,
+                         x1, x2, x3, dev_tipVec, tipX1, tipX2, dev_wgt, n,
+                         dev_P, numCats, dev_scalerInc);
+    else
+      hipLaunchKernelGGL((k_newview_dna_cat<EXAML_INNER_INNER, false>),
+                         dim3(grid), dim3(NV_BLOCK), 0, s, dev_EV, dev_cptr,
+                         x1, x2, x3, dev_tipVec, tipX1, tipX2, dev_wgt, n,
+                         dev_P, numCats, dev_scalerInc);
+    break;
+  default:
+    snprintf(g_err, sizeof(g_err), "newview_cat: bad tipCase %d", tipCase);
+    return -1;
+  }
+  CHK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int examl_hip_evaluate_dna_cat(
+    const int *dev_cptr, const int *dev_wgt, const double *x1,
+    const double *x2, const double *dev_tipVec, const unsigned char *tipX1,
+    long n, const double *dev_diag, int numCats, const unsigned int *gsP,
+    const unsigned int *gsQ, double log_minlik, double *dev_partials,
+    double *dev_lnl, void *stream) {
+  if (n <= 0) return 0;
+  if (numCats > MAX_CAT) {
+    snprintf(g_err, sizeof(g_err), "evaluate_cat: numCats %d > %d", numCats,
+             MAX_CAT);
+    return -1;
+  }
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  const int grid = grid_for(n);
+  if (tipX1)
+    hipLaunchKernelGGL((k_evaluate_dna_cat<true>), dim3(grid), dim3(NV_BLOCK),
+                       0, s, dev_cptr, x1, x2, dev_tipVec, tipX1, dev_wgt,
+                       dev_diag, numCats, n, dev_partials);
+  else
+    hipLaunchKernelGGL((k_evaluate_dna_cat<false>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, dev_cptr, x1, x2, dev_tipVec,
+                       tipX1, dev_wgt, dev_diag, numCats, n, dev_partials);
+  CHK(hipGetLastError());
+  hipLaunchKernelGGL(k_reduce_lnl, dim3(1), dim3(NV_BLOCK), 0, s,
+                     dev_partials, grid, gsP, gsQ, log_minlik, dev_lnl);
+  CHK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int examl_hip_sum_dna_cat(int tipCase, double *dev_sum,
+                                     const double *x1, const double *x2,
+                                     const double *dev_tipVec,
+                                     const unsigned char *tipX1,
+                                     const unsigned char *tipX2, long n,
+                                     void *stream) {
+  if (n <= 0) return 0;
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  const int grid = grid_for(n);
+  switch (tipCase) {
+  case EXAML_TIP_TIP:
+    hipLaunchKernelGGL((k_sum_dna_cat<EXAML_TIP_TIP>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, dev_sum, x1, x2, dev_tipVec,
+                       tipX1, tipX2, n);
+    break;
+  case EXAML_TIP_INNER:
+    hipLaunchKernelGGL((k_sum_dna_cat<EXAML_TIP_INNER>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, dev_sum, x1, x2, dev_tipVec,
+                       tipX1, tipX2, n);
+    break;
+  case EXAML_INNER_INNER:
+    hipLaunchKernelGGL((k_sum_dna_cat<EXAML_INNER_INNER>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, dev_sum, x1, x2, dev_tipVec,
+                       tipX1, tipX2, n);
+    break;
+  default:
+    snprintf(g_err, sizeof(g_err), "sum_cat: bad tipCase %d", tipCase);
+    return -1;
+  }
+  CHK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int examl_hip_core_root_dna_cat(
+    long n, const double *dev_sum, const double *EIGN, const double *rptr,
+    int numCats, double lz, const int *dev_wgt, const int *dev_cptr,
+    double *dev_dtab, double *dev_partials, double *dev_out2, void *stream) {
+  if (n <= 0) return 0;
+  if (numCats > MAX_CAT) {
+    snprintf(g_err, sizeof(g_err), "core_cat: numCats %d > %d", numCats,
+             MAX_CAT);
+    return -1;
+  }
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  double host[MAX_CAT * 4 + 8 + MAX_CAT];
+  examl_host_core_dtables_dna_cat(EIGN, rptr, numCats, lz, host);
+  CHK(hipMemcpyAsync(dev_dtab, host,
+                     (size_t)(numCats * 4 + 8 + numCats) * sizeof(double),
+                     hipMemcpyHostToDevice, s));
+  const int grid = grid_for(n);
+  hipLaunchKernelGGL(k_core_dna_cat, dim3(grid), dim3(NV_BLOCK), 0, s,
+                     dev_sum, dev_dtab, dev_wgt, dev_cptr, numCats, n,
+                     dev_partials);
+  CHK(hipGetLastError());
+  hipLaunchKernelGGL(k_reduce_2, dim3(1), dim3(NV_BLOCK), 0, s, dev_partials,
+                     grid, dev_out2);
+  CHK(hipGetLastError());
+  return 0;
 }
 
 /* ===========================================================================

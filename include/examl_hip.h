@@ -214,6 +214,56 @@ int examl_hip_core_root_dna_gamma(long n, const double *dev_sum,
                                   void *stream);
 
 /* ---------------------------------------------------------------------------
+ * DNA CAT (PSR, -m PSR) surface — span 4, per-site rate category cptr[i],
+ * numCats <= 25 rate categories (maxCategories).  Each function replaces:
+ *   newview  — newviewGTRCAT_AVX (examl/avxLikelihood.c:326)
+ *   evaluate — evaluateGTRCAT (examl/evaluateGenericSpecial.c:1988; no 0.25)
+ *   sum      — sumCAT (examl/makenewzGenericSpecial.c:1850)
+ *   core     — coreGTRCAT (examl/makenewzGenericSpecial.c:2402)
+ * P blocks are numCats*32 doubles (left|right); diag numCats*4; the core
+ * dtable scratch is numCats*4 + 8 + numCats doubles.  The CAT traversal
+ * executor and the optimizeRateCategories/evaluatePartialGeneric host loop
+ * are scheduled for round 2.
+ * ------------------------------------------------------------------------ */
+
+void examl_host_core_dtables_dna_cat(const double *EIGN, const double *rptr,
+                                     int numCats, double lz, double *out);
+
+int examl_hip_newview_dna_cat(int tipCase, const double *dev_EV,
+                              const int *dev_cptr, const double *dev_x1,
+                              const double *dev_x2, double *dev_x3,
+                              const double *dev_tipVector,
+                              const unsigned char *dev_tipX1,
+                              const unsigned char *dev_tipX2, long n,
+                              const double *dev_P, int numCats,
+                              const int *dev_wgt,
+                              unsigned int *dev_scalerInc, void *stream);
+
+int examl_hip_evaluate_dna_cat(const int *dev_cptr, const int *dev_wgt,
+                               const double *dev_x1, const double *dev_x2,
+                               const double *dev_tipVector,
+                               const unsigned char *dev_tipX1, long n,
+                               const double *dev_diag, int numCats,
+                               const unsigned int *dev_gsP,
+                               const unsigned int *dev_gsQ, double log_minlik,
+                               double *dev_partials, double *dev_lnl,
+                               void *stream);
+
+int examl_hip_sum_dna_cat(int tipCase, double *dev_sum, const double *dev_x1,
+                          const double *dev_x2,
+                          const double *dev_tipVector,
+                          const unsigned char *dev_tipX1,
+                          const unsigned char *dev_tipX2, long n,
+                          void *stream);
+
+int examl_hip_core_root_dna_cat(long n, const double *dev_sum,
+                                const double *EIGN, const double *rptr,
+                                int numCats, double lz, const int *dev_wgt,
+                                const int *dev_cptr, double *dev_dtab_scratch,
+                                double *dev_partials, double *dev_out2,
+                                void *stream);
+
+/* ---------------------------------------------------------------------------
  * Protein (20-state) GTRGAMMA surface — span 80, tip codes 1..22.  Each
  * function replaces the 20-state counterpart of the DNA one above:
  *   newview  — newviewGTRGAMMAPROT_AVX (examl/avxLikelihood.c:1312)

@@ -190,6 +190,64 @@ def init_gtr_dna(frequencies, rates):
     return EIGN, EV, EI, tipVector
 
 
+def newview_dna_cat(tip_case, EV, cptr, x1, x2, tipVector, tipX1, tipX2, n,
+                    left, right, wgt, lib=None):
+    lib = lib or _orc
+    fn = (lib.oracle_newview_dna_cat if lib is _orc
+          else lib.newviewGTRCAT_AVX)
+    x3 = aligned(n * 4)
+    inc = ctypes.c_int(0)
+    nullp = ctypes.cast(None, _c_d)
+    nullb = ctypes.cast(None, _c_u8)
+    fn(ctypes.c_int(tip_case), _dp(EV), _ip(cptr),
+       _dp(x1) if x1 is not None else nullp,
+       _dp(x2) if x2 is not None else nullp,
+       _dp(x3), _dp(tipVector),
+       _u8p(tipX1) if tipX1 is not None else nullb,
+       _u8p(tipX2) if tipX2 is not None else nullb,
+       ctypes.c_int(n), _dp(left), _dp(right), _ip(wgt), ctypes.byref(inc))
+    return x3, inc.value
+
+
+def evaluate_dna_cat(cptr, wgt, x1, x2, tipVector, tipX1, n, diag, lib=None):
+    lib = lib or _orc
+    fn = (lib.oracle_evaluate_dna_cat if lib is _orc else lib.evaluateGTRCAT)
+    fn.restype = ctypes.c_double
+    nullp = ctypes.cast(None, _c_d)
+    nullb = ctypes.cast(None, _c_u8)
+    return fn(_ip(cptr), _ip(wgt),
+              _dp(x1) if x1 is not None else nullp,
+              _dp(x2), _dp(tipVector),
+              _u8p(tipX1) if tipX1 is not None else nullb,
+              ctypes.c_int(n), _dp(diag))
+
+
+def sum_dna_cat(tip_case, x1, x2, tipVector, tipX1, tipX2, n, lib=None):
+    lib = lib or _orc
+    fn = lib.oracle_sum_dna_cat if lib is _orc else lib.sumCAT
+    sumtable = aligned(n * 4)
+    nullp = ctypes.cast(None, _c_d)
+    nullb = ctypes.cast(None, _c_u8)
+    fn(ctypes.c_int(tip_case), _dp(sumtable),
+       _dp(x1) if x1 is not None else nullp,
+       _dp(x2) if x2 is not None else nullp,
+       _dp(tipVector),
+       _u8p(tipX1) if tipX1 is not None else nullb,
+       _u8p(tipX2) if tipX2 is not None else nullb, ctypes.c_int(n))
+    return sumtable
+
+
+def core_dna_cat(n, num_cats, sumtable, wgt, rptr, EIGN, cptr, lz, lib=None):
+    lib = lib or _orc
+    fn = lib.oracle_core_dna_cat if lib is _orc else lib.coreGTRCAT
+    d1 = ctypes.c_double(0.0)
+    d2 = ctypes.c_double(0.0)
+    fn(ctypes.c_int(n), ctypes.c_int(num_cats), _dp(sumtable),
+       ctypes.byref(d1), ctypes.byref(d2), _ip(wgt), _dp(rptr), _dp(EIGN),
+       _ip(cptr), ctypes.c_double(lz))
+    return d1.value, d2.value
+
+
 BIT_VECTOR_AA = np.array([1 << i for i in range(20)] + [12, 96, 0xFFFFF],
                          dtype=np.uint32)  # globalVariables.h:95
 

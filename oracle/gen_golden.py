@@ -180,5 +180,71 @@ def main():
     print("golden fixtures written to", GOLDEN)
 
 
+
+def gen_cat():
+    """CAT (PSR) golden fixtures — reference kernels via _ref."""
+    assert O.have_ref()
+    rng = np.random.default_rng(424242)
+    ref = O._ref
+    d = np.load(os.path.join(GOLDEN, "model_dna.npz"))
+    EIGN = O.aligned(4); EIGN[:] = d["m1_EIGN"]
+    EV = O.aligned(16); EV[:] = d["m1_EV"]
+    EI = O.aligned(16); EI[:] = d["m1_EI"]
+    tipVector = O.aligned(64); tipVector[:] = d["m1_tipVector"]
+    num_cats = 9
+    rptr = O.aligned(num_cats)
+    rptr[:] = rng.uniform(0.03, 5.0, num_cats)
+    z_q, z_r = 0.84, 0.18
+    left, right = O.ref_make_p(np.log(z_q), np.log(z_r), rptr, EI, EIGN,
+                               num_cats, 4)
+    n = 768
+    cc = {"num_cats": np.int64(num_cats), "rptr": np.asarray(rptr),
+          "left": np.asarray(left), "right": np.asarray(right),
+          "z_q": np.float64(z_q), "z_r": np.float64(z_r)}
+    cptr = np.ascontiguousarray(rng.integers(0, num_cats, n), np.int32)
+    cc["cptr"] = cptr
+    for tag, mag in [("norm", 1.0), ("tiny", 1e-80)]:
+        x1 = O.aligned(n * 4); x1[:] = rng.uniform(0.01, 1.0, n * 4) * mag
+        x2 = O.aligned(n * 4); x2[:] = rng.uniform(0.01, 1.0, n * 4) * mag
+        wgt = np.ascontiguousarray(rng.integers(1, 5, n), np.int32)
+        t1 = np.ascontiguousarray(rng.integers(1, 16, n), np.uint8)
+        t2 = np.ascontiguousarray(rng.integers(1, 16, n), np.uint8)
+        cc[f"{tag}_x1"], cc[f"{tag}_x2"] = x1, x2
+        cc[f"{tag}_wgt"], cc[f"{tag}_tipX1"], cc[f"{tag}_tipX2"] = wgt, t1, t2
+        for tc, a1, a2, u1, u2 in [
+            (O.TIP_TIP, None, None, t1, t2),
+            (O.TIP_INNER, None, x2, t1, None),
+            (O.INNER_INNER, x1, x2, None, None),
+        ]:
+            x3, inc = O.newview_dna_cat(tc, EV, cptr, a1, a2, tipVector, u1,
+                                        u2, n, left, right, wgt, lib=ref)
+            cc[f"{tag}_newview_tc{tc}_x3"] = x3
+            cc[f"{tag}_newview_tc{tc}_inc"] = np.int64(inc)
+    x1, x2 = cc["norm_x1"], cc["norm_x2"]
+    wgt, t1, t2 = cc["norm_wgt"], cc["norm_tipX1"], cc["norm_tipX2"]
+    z_root = 0.71
+    diag = O.ref_calc_diagptable(z_root, 4, num_cats, rptr, EIGN)
+    cc["z_root"], cc["diag"] = np.float64(z_root), np.asarray(diag)
+    cc["eval_II"] = np.float64(O.evaluate_dna_cat(
+        cptr, wgt, x1, x2, tipVector, None, n, diag, lib=ref))
+    cc["eval_TIP"] = np.float64(O.evaluate_dna_cat(
+        cptr, wgt, None, x2, tipVector, t1, n, diag, lib=ref))
+    lz = np.log(0.52)
+    cc["lz_core"] = np.float64(lz)
+    for tc, a1, a2, u1, u2 in [
+        (O.TIP_TIP, None, None, t1, t2),
+        (O.TIP_INNER, None, x2, t1, None),
+        (O.INNER_INNER, x1, x2, None, None),
+    ]:
+        st = O.sum_dna_cat(tc, a1, a2, tipVector, u1, u2, n, lib=ref)
+        cc[f"sum_tc{tc}"] = np.asarray(st)
+        d1, d2 = O.core_dna_cat(n, num_cats, st, wgt, rptr, EIGN, cptr, lz,
+                                lib=ref)
+        cc[f"core_tc{tc}_d1"] = np.float64(d1)
+        cc[f"core_tc{tc}_d2"] = np.float64(d2)
+    np.savez_compressed(os.path.join(GOLDEN, "kernels_dna_cat.npz"), **cc)
+    print("CAT golden fixtures written")
+
+
 if __name__ == "__main__":
-    main()
+    gen_cat()

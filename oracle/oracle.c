@@ -349,6 +349,164 @@ EXPORT void oracle_core_dna_gamma(int upper, const double *sumtable,
 }
 
 /* ==========================================================================
+ * DNA CAT (PSR) kernels — span 4, per-site rate category cptr[i].
+ * ==========================================================================*/
+
+/* --------------------------------------------------------------------------
+ * newview, DNA CAT.  Restates avxLikelihood.c:326 (newviewGTRCAT_AVX):
+ * per site, P rows at left/right[cptr[i]*16]; vv = EV . ((L x1) o (R x2));
+ * rescale when all 4 |vv| < 2^-256 (TIP_INNER / INNER_INNER only).
+ * ------------------------------------------------------------------------*/
+EXPORT void oracle_newview_dna_cat(int tipCase, const double *EV,
+                                   const int *cptr, const double *x1_start,
+                                   const double *x2_start, double *x3_start,
+                                   const double *tipVector,
+                                   const unsigned char *tipX1,
+                                   const unsigned char *tipX2, int n,
+                                   const double *left, const double *right,
+                                   const int *wgt, int *scalerIncrement) {
+  int i, l, s;
+  int addScale = 0;
+  for (i = 0; i < n; i++) {
+    const double *le = &left[cptr[i] * 16];
+    const double *ri = &right[cptr[i] * 16];
+    const double *x1, *x2;
+    switch (tipCase) {
+    case ORC_TIP_TIP:
+      x1 = &tipVector[4 * tipX1[i]];
+      x2 = &tipVector[4 * tipX2[i]];
+      break;
+    case ORC_TIP_INNER:
+      x1 = &tipVector[4 * tipX1[i]];
+      x2 = &x2_start[4 * i];
+      break;
+    default:
+      x1 = &x1_start[4 * i];
+      x2 = &x2_start[4 * i];
+    }
+    double vv[4] = {0, 0, 0, 0};
+    for (l = 0; l < 4; l++) {
+      double pl[4], pr[4];
+      for (s = 0; s < 4; s++) {
+        pl[s] = x1[s] * le[l * 4 + s];
+        pr[s] = x2[s] * ri[l * 4 + s];
+      }
+      const double t = hadd4d(pl) * hadd4d(pr);
+      for (s = 0; s < 4; s++) vv[s] += t * EV[l * 4 + s];
+    }
+    if (tipCase != ORC_TIP_TIP) {
+      int scale = 1;
+      for (s = 0; s < 4; s++)
+        if (!(fabs(vv[s]) < ORC_MINLIKELIHOOD)) { scale = 0; break; }
+      if (scale) {
+        for (s = 0; s < 4; s++) vv[s] *= ORC_TWOTOTHE256;
+        addScale += wgt[i];
+      }
+    }
+    for (s = 0; s < 4; s++) x3_start[4 * i + s] = vv[s];
+  }
+  *scalerIncrement = addScale;
+}
+
+/* evaluateGTRCAT (evaluateGenericSpecial.c:1988): per-site diag row at
+ * diagptable[4*cptr[i]]; NO 0.25 factor (single category per site). */
+EXPORT double oracle_evaluate_dna_cat(const int *cptr, const int *wptr,
+                                      const double *x1_start,
+                                      const double *x2_start,
+                                      const double *tipVector,
+                                      const unsigned char *tipX1, int n,
+                                      const double *diagptable) {
+  double sum = 0.0;
+  int i;
+  for (i = 0; i < n; i++) {
+    const double *x1 = tipX1 ? &tipVector[4 * tipX1[i]] : &x1_start[4 * i];
+    const double *x2 = &x2_start[4 * i];
+    const double *d = &diagptable[4 * cptr[i]];
+    const double t0 = x1[0] * x2[0] * d[0] + x1[2] * x2[2] * d[2];
+    const double t1 = x1[1] * x2[1] * d[1] + x1[3] * x2[3] * d[3];
+    sum += wptr[i] * log(fabs(t0 + t1));
+  }
+  return sum;
+}
+
+/* sumCAT (makenewzGenericSpecial.c:1850) */
+EXPORT void oracle_sum_dna_cat(int tipCase, double *sumtable,
+                               const double *x1_start, const double *x2_start,
+                               const double *tipVector,
+                               const unsigned char *tipX1,
+                               const unsigned char *tipX2, int n) {
+  int i, k;
+  for (i = 0; i < n; i++) {
+    const double *x1, *x2;
+    switch (tipCase) {
+    case ORC_TIP_TIP:
+      x1 = &tipVector[4 * tipX1[i]];
+      x2 = &tipVector[4 * tipX2[i]];
+      break;
+    case ORC_TIP_INNER:
+      x1 = &tipVector[4 * tipX1[i]];
+      x2 = &x2_start[4 * i];
+      break;
+    default:
+      x1 = &x1_start[4 * i];
+      x2 = &x2_start[4 * i];
+    }
+    for (k = 0; k < 4; k++) sumtable[i * 4 + k] = x1[k] * x2[k];
+  }
+}
+
+/* coreGTRCAT (makenewzGenericSpecial.c:2402): per-site rate r=rptr[cptr],
+ * weights wr1=r*wgt, wr2=r^2*wgt; e1=EIGN, e2=EIGN^2 (NOT rate-scaled). */
+EXPORT void oracle_core_dna_cat(int upper, int numberOfCategories,
+                                const double *sumtable, double *ext_dlnLdlz,
+                                double *ext_d2lnLdlz2, const int *wgt,
+                                const double *rptr, const double *EIGN,
+                                const int *cptr, double lz) {
+  double e1[4], e2[4], d[25 * 4];
+  double dlnLdlz = 0.0, d2lnLdlz2 = 0.0;
+  int i, l;
+  e1[0] = 0.0;
+  e2[0] = 0.0;
+  for (l = 1; l < 4; l++) {
+    e1[l] = EIGN[l];
+    e2[l] = EIGN[l] * EIGN[l];
+  }
+  {
+    const double dd1 = EIGN[1] * lz, dd2 = EIGN[2] * lz, dd3 = EIGN[3] * lz;
+    for (i = 0; i < numberOfCategories; i++) {
+      d[i * 4 + 0] = 1.0;
+      d[i * 4 + 1] = exp(dd1 * rptr[i]);
+      d[i * 4 + 2] = exp(dd2 * rptr[i]);
+      d[i * 4 + 3] = exp(dd3 * rptr[i]);
+    }
+  }
+  for (i = 0; i < upper; i++) {
+    const double *s = &sumtable[4 * i];
+    const double *d1 = &d[4 * cptr[i]];
+    const double r = rptr[cptr[i]];
+    const double wr1 = r * wgt[i], wr2 = r * r * wgt[i];
+    double a0e = 0, a0o = 0, a1e = 0, a1o = 0, a2e = 0, a2o = 0;
+    for (l = 0; l < 4; l += 2) {
+      const double te = d1[l] * s[l];
+      const double to = d1[l + 1] * s[l + 1];
+      a0e += te;
+      a0o += to;
+      a1e += te * e1[l];
+      a1o += to * e1[l + 1];
+      a2e += te * e2[l];
+      a2o += to * e2[l + 1];
+    }
+    const double inv_Li = 1.0 / fabs(a0e + a0o);
+    const double dlnLidlz = (a1e + a1o) * inv_Li;
+    const double d2lnLidlz2 = (a2e + a2o) * inv_Li;
+    dlnLdlz += wr1 * dlnLidlz;
+    d2lnLdlz2 += wr2 * (d2lnLidlz2 - dlnLidlz * dlnLidlz);
+  }
+  *ext_dlnLdlz = dlnLdlz;
+  *ext_d2lnLdlz2 = d2lnLdlz2;
+}
+
+/* ==========================================================================
  * Protein (20-state) GTRGAMMA kernels — span 80, tip codes 1..22.
  * ==========================================================================*/
 

@@ -1,0 +1,204 @@
+"""CAT (PSR, -m PSR) kernel parity — SURVEY §8f row 1.  CPU: the oracle
+restatement vs the reference golden vectors (bit-exact).  GPU: the HIP CAT
+kernels vs the same goldens."""
+
+import ctypes
+import os
+
+import numpy as np
+import pytest
+
+import examl_amd as ea
+import oracle as O
+
+
+@pytest.fixture(scope="module")
+def kern(golden_dir):
+    return np.load(os.path.join(golden_dir, "kernels_dna_cat.npz"))
+
+
+def _al(a):
+    out = O.aligned(a.shape, a.dtype)
+    out[:] = a
+    return out
+
+
+@pytest.fixture(scope="module")
+def model(golden_dir):
+    d = np.load(os.path.join(golden_dir, "model_dna.npz"))
+    return ea.DnaGtrModel(d["m1_freqs"], d["m1_rates6"], float(d["m1_alpha"]))
+
+
+# ---------------- CPU: oracle vs golden -----------------------------------
+
+@pytest.mark.parametrize("tag", ["norm", "tiny"])
+@pytest.mark.parametrize("tc", [O.TIP_TIP, O.TIP_INNER, O.INNER_INNER])
+def test_oracle_newview_cat(kern, model, tag, tc):
+    EV = _al(model.EV)
+    tipVector = _al(model.tipVector)
+    left = _al(kern["left"])
+    right = _al(kern["right"])
+    cptr = np.ascontiguousarray(kern["cptr"])
+    x1 = _al(kern[f"{tag}_x1"])
+    x2 = _al(kern[f"{tag}_x2"])
+    wgt = np.ascontiguousarray(kern[f"{tag}_wgt"])
+    t1 = np.ascontiguousarray(kern[f"{tag}_tipX1"])
+    t2 = np.ascontiguousarray(kern[f"{tag}_tipX2"])
+    n = len(wgt)
+    args = {
+        O.TIP_TIP: (None, None, t1, t2),
+        O.TIP_INNER: (None, x2, t1, None),
+        O.INNER_INNER: (x1, x2, None, None),
+    }[tc]
+    x3, inc = O.newview_dna_cat(tc, EV, cptr, args[0], args[1], tipVector,
+                                args[2], args[3], n, left, right, wgt)
+    assert inc == int(kern[f"{tag}_newview_tc{tc}_inc"])
+    assert np.array_equal(x3, kern[f"{tag}_newview_tc{tc}_x3"])
+
+
+def test_oracle_evaluate_sum_core_cat(kern, model):
+    tipVector = _al(model.tipVector)
+    EIGN = _al(model.EIGN)
+    rptr = _al(kern["rptr"])
+    num_cats = int(kern["num_cats"])
+    cptr = np.ascontiguousarray(kern["cptr"])
+    x1 = _al(kern["norm_x1"])
+    x2 = _al(kern["norm_x2"])
+    wgt = np.ascontiguousarray(kern["norm_wgt"])
+    t1 = np.ascontiguousarray(kern["norm_tipX1"])
+    t2 = np.ascontiguousarray(kern["norm_tipX2"])
+    diag = _al(kern["diag"])
+    n = len(wgt)
+    assert O.evaluate_dna_cat(cptr, wgt, x1, x2, tipVector, None, n,
+                              diag) == float(kern["eval_II"])
+    assert O.evaluate_dna_cat(cptr, wgt, None, x2, tipVector, t1, n,
+                              diag) == float(kern["eval_TIP"])
+    # makeP with numCats != 4 must also match
+    left, right = O.make_p(np.log(float(kern["z_q"])),
+                           np.log(float(kern["z_r"])), rptr, _al(model.EI),
+                           EIGN, num_cats, 4)
+    assert np.array_equal(left, kern["left"])
+    assert np.array_equal(right, kern["right"])
+    for tc, a1, a2, u1, u2 in [
+        (O.TIP_TIP, None, None, t1, t2),
+        (O.TIP_INNER, None, x2, t1, None),
+        (O.INNER_INNER, x1, x2, None, None),
+    ]:
+        st = O.sum_dna_cat(tc, a1, a2, tipVector, u1, u2, n)
+        assert np.array_equal(st, kern[f"sum_tc{tc}"])
+        d1, d2 = O.core_dna_cat(n, num_cats, st, wgt, rptr, EIGN, cptr,
+                                float(kern["lz_core"]))
+        assert d1 == float(kern[f"core_tc{tc}_d1"])
+        assert d2 == float(kern[f"core_tc{tc}_d2"])
+
+
+# ---------------- GPU: HIP kernels vs golden -------------------------------
+
+def vp(t):
+    return ctypes.c_void_p(t.data_ptr())
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("tag", ["norm", "tiny"])
+@pytest.mark.parametrize("tc", [ea.TIP_TIP, ea.TIP_INNER, ea.INNER_INNER])
+def test_gpu_newview_cat_bit_exact(kern, model, tag, tc):
+    import torch
+    dev = torch.device("cuda:0")
+
+    def td(a):
+        return torch.from_numpy(np.ascontiguousarray(a)).to(dev)
+
+    n = len(kern[f"{tag}_wgt"])
+    num_cats = int(kern["num_cats"])
+    d_x1 = td(kern[f"{tag}_x1"])
+    d_x2 = td(kern[f"{tag}_x2"])
+    d_x3 = torch.zeros(n * 4, dtype=torch.float64, device=dev)
+    d_P = td(np.concatenate([kern["left"], kern["right"]]))
+    d_EV = td(model.EV)
+    d_tv = td(model.tipVector)
+    d_cptr = td(kern["cptr"])
+    d_t1 = td(kern[f"{tag}_tipX1"])
+    d_t2 = td(kern[f"{tag}_tipX2"])
+    d_wgt = td(kern[f"{tag}_wgt"])
+    d_inc = torch.zeros(1, dtype=torch.int32, device=dev)
+    null = ctypes.c_void_p(0)
+    ea.check(ea.lib().examl_hip_newview_dna_cat(
+        tc, vp(d_EV), vp(d_cptr),
+        vp(d_x1) if tc == ea.INNER_INNER else null,
+        vp(d_x2) if tc != ea.TIP_TIP else null,
+        vp(d_x3), vp(d_tv),
+        vp(d_t1) if tc != ea.INNER_INNER else null,
+        vp(d_t2) if tc == ea.TIP_TIP else null,
+        ctypes.c_long(n), vp(d_P), num_cats, vp(d_wgt), vp(d_inc),
+        ctypes.c_void_p(0)), "newview_cat")
+    torch.cuda.synchronize()
+    assert int(d_inc.item()) == int(kern[f"{tag}_newview_tc{tc}_inc"])
+    assert np.array_equal(d_x3.cpu().numpy(), kern[f"{tag}_newview_tc{tc}_x3"])
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("tc", [ea.TIP_TIP, ea.TIP_INNER, ea.INNER_INNER])
+def test_gpu_evaluate_sum_core_cat(kern, model, tc):
+    import torch
+    dev = torch.device("cuda:0")
+
+    def td(a):
+        return torch.from_numpy(np.ascontiguousarray(a)).to(dev)
+
+    n = len(kern["norm_wgt"])
+    num_cats = int(kern["num_cats"])
+    d_x1 = td(kern["norm_x1"])
+    d_x2 = td(kern["norm_x2"])
+    d_tv = td(model.tipVector)
+    d_cptr = td(kern["cptr"])
+    d_t1 = td(kern["norm_tipX1"])
+    d_t2 = td(kern["norm_tipX2"])
+    d_wgt = td(kern["norm_wgt"])
+    d_diag = td(kern["diag"])
+    d_lnl = torch.zeros(1, dtype=torch.float64, device=dev)
+    d_part = torch.zeros(2 * 8192, dtype=torch.float64, device=dev)
+    null = ctypes.c_void_p(0)
+    if tc == ea.INNER_INNER:  # evaluate only has tip/inner bodies
+        ea.check(ea.lib().examl_hip_evaluate_dna_cat(
+            vp(d_cptr), vp(d_wgt), vp(d_x1), vp(d_x2), vp(d_tv), null,
+            ctypes.c_long(n), vp(d_diag), num_cats, null, null,
+            ctypes.c_double(0.0), vp(d_part), vp(d_lnl), ctypes.c_void_p(0)),
+            "evaluate_cat")
+        torch.cuda.synchronize()
+        assert np.isclose(d_lnl.item(), float(kern["eval_II"]), rtol=1e-12)
+    elif tc == ea.TIP_INNER:
+        ea.check(ea.lib().examl_hip_evaluate_dna_cat(
+            vp(d_cptr), vp(d_wgt), null, vp(d_x2), vp(d_tv), vp(d_t1),
+            ctypes.c_long(n), vp(d_diag), num_cats, null, null,
+            ctypes.c_double(0.0), vp(d_part), vp(d_lnl), ctypes.c_void_p(0)),
+            "evaluate_cat")
+        torch.cuda.synchronize()
+        assert np.isclose(d_lnl.item(), float(kern["eval_TIP"]), rtol=1e-12)
+
+    d_sum = torch.zeros(n * 4, dtype=torch.float64, device=dev)
+    ea.check(ea.lib().examl_hip_sum_dna_cat(
+        tc, vp(d_sum),
+        vp(d_x1) if tc == ea.INNER_INNER else null,
+        vp(d_x2) if tc != ea.TIP_TIP else null,
+        vp(d_tv),
+        vp(d_t1) if tc != ea.INNER_INNER else null,
+        vp(d_t2) if tc == ea.TIP_TIP else null,
+        ctypes.c_long(n), ctypes.c_void_p(0)), "sum_cat")
+    torch.cuda.synchronize()
+    assert np.array_equal(d_sum.cpu().numpy(), kern[f"sum_tc{tc}"])
+
+    EIGN = np.ascontiguousarray(model.EIGN)
+    rptr = np.ascontiguousarray(kern["rptr"])
+    d_dtab = torch.zeros(num_cats * 4 + 8 + num_cats, dtype=torch.float64,
+                         device=dev)
+    d_out2 = torch.zeros(2, dtype=torch.float64, device=dev)
+    ea.check(ea.lib().examl_hip_core_root_dna_cat(
+        ctypes.c_long(n), vp(d_sum),
+        EIGN.ctypes.data_as(ctypes.c_void_p),
+        rptr.ctypes.data_as(ctypes.c_void_p), num_cats,
+        ctypes.c_double(float(kern["lz_core"])), vp(d_wgt), vp(d_cptr),
+        vp(d_dtab), vp(d_part), vp(d_out2), ctypes.c_void_p(0)), "core_cat")
+    torch.cuda.synchronize()
+    out = d_out2.cpu().numpy()
+    assert np.isclose(out[0], float(kern[f"core_tc{tc}_d1"]), rtol=1e-11)
+    assert np.isclose(out[1], float(kern[f"core_tc{tc}_d2"]), rtol=1e-11)
