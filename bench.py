@@ -122,6 +122,9 @@ def main():
                     help="partitions per GPU, each on its own HIP stream "
                          "(config 3 shard shape: --partitions 16 "
                          "--sites 125000)")
+    ap.add_argument("--fast-math", action="store_true",
+                    help="FMA protein newview (the reference's _FMA build "
+                         "class; ~1 ulp/op vs the bit-exact default)")
     ap.add_argument("--no-cpu-baseline", action="store_true")
     args = ap.parse_args()
 
@@ -148,6 +151,8 @@ def main():
         tips, wgt = make_alignment(ntips, width, seed=42 + rank)
         model = ea.DnaGtrModel([0.28, 0.22, 0.24, 0.26],
                                [1.2, 2.9, 0.7, 1.0, 3.2, 1.0], alpha=0.6)
+    if args.fast_math:
+        ea.lib().examl_hip_fast_math(1)
     tree = ea.PhyloTree.random(ntips, seed=7)
     P = args.partitions
     assert width % P == 0
@@ -284,6 +289,7 @@ def main():
             "full_tree_eval_ms": elapsed / args.steps * 1e3,
             "lnl": lnl0,
             "parallelism": f"dp{world} site-sharded, 1 RCCL all-reduce/step",
+            "fast_math": bool(args.fast_math),
         },
         "roofline": roofline,
         "cpu_baseline": cpu,

@@ -104,6 +104,7 @@ def _bind(lib):
         [l, p, p, p, i, d, p, p, p, p, p, p]
     lib.examl_hip_use_graphs.argtypes = [i]
     lib.examl_hip_graphs_clear.argtypes = []
+    lib.examl_hip_fast_math.argtypes = [i]
     return lib
 
 

@@ -573,19 +573,27 @@ __global__ __launch_bounds__(NV_BLOCK) void k_core_dna_cat(
  * 20-dots as four lane accumulators over five 4-chunks, (t0+t1)+(t2+t3).
  * ==========================================================================*/
 
+template <bool FAST>
 __device__ __forceinline__ double dot20o(const double *a, const double *b) {
   double t0 = 0, t1 = 0, t2 = 0, t3 = 0;
 #pragma unroll
   for (int c = 0; c < 20; c += 4) {
-    t0 += a[c] * b[c];
-    t1 += a[c + 1] * b[c + 1];
-    t2 += a[c + 2] * b[c + 2];
-    t3 += a[c + 3] * b[c + 3];
+    if (FAST) { /* fused accumulate: the reference's _FMA build class */
+      t0 = fma(a[c], b[c], t0);
+      t1 = fma(a[c + 1], b[c + 1], t1);
+      t2 = fma(a[c + 2], b[c + 2], t2);
+      t3 = fma(a[c + 3], b[c + 3], t3);
+    } else {
+      t0 += a[c] * b[c];
+      t1 += a[c + 1] * b[c + 1];
+      t2 += a[c + 2] * b[c + 2];
+      t3 += a[c + 3] * b[c + 3];
+    }
   }
   return (t0 + t1) + (t2 + t3);
 }
 
-template <int TC, bool NT>
+template <int TC, bool NT, bool FAST>
 __global__ __launch_bounds__(NV_BLOCK) void k_newview_prot_gamma(
     const double *__restrict__ x1, const double *__restrict__ x2,
     double *__restrict__ x3, const double *__restrict__ P,
@@ -612,8 +620,9 @@ __global__ __launch_bounds__(NV_BLOCK) void k_newview_prot_gamma(
     /* ump tables (avxLikelihood.c:1355-1389): entry (code, cat*20+row) */
     for (int j = tid; j < 23 * 80; j += NV_BLOCK) {
       const int code = j / 80, k = j % 80;
-      sU1[j] = dot20o(&sTV[20 * code], &sL[k * 20]);
-      if (TC == EXAML_TIP_TIP) sU2[j] = dot20o(&sTV[20 * code], &sR[k * 20]);
+      sU1[j] = dot20o<FAST>(&sTV[20 * code], &sL[k * 20]);
+      if (TC == EXAML_TIP_TIP)
+        sU2[j] = dot20o<FAST>(&sTV[20 * code], &sR[k * 20]);
     }
     __syncthreads();
   }
@@ -650,18 +659,23 @@ __global__ __launch_bounds__(NV_BLOCK) void k_newview_prot_gamma(
     for (int l = 0; l < 20; l++) {
       double u1, u2;
       if (TC == EXAML_INNER_INNER) {
-        u1 = dot20o(xl, &sL[cat * 400 + l * 20]);
-        u2 = dot20o(xr, &sR[cat * 400 + l * 20]);
+        u1 = dot20o<FAST>(xl, &sL[cat * 400 + l * 20]);
+        u2 = dot20o<FAST>(xr, &sR[cat * 400 + l * 20]);
       } else if (TC == EXAML_TIP_INNER) {
         u1 = sU1[80 * code1 + cat * 20 + l];
-        u2 = dot20o(xr, &sR[cat * 400 + l * 20]);
+        u2 = dot20o<FAST>(xr, &sR[cat * 400 + l * 20]);
       } else {
         u1 = sU1[80 * code1 + cat * 20 + l];
         u2 = sU2[80 * code2 + cat * 20 + l];
       }
       const double t = u1 * u2;
 #pragma unroll
-      for (int s = 0; s < 20; s++) acc[s] += t * sEV[l * 20 + s];
+      for (int s = 0; s < 20; s++) {
+        if (FAST)
+          acc[s] = fma(t, sEV[l * 20 + s], acc[s]);
+        else
+          acc[s] += t * sEV[l * 20 + s];
+      }
     }
 
     if (TC != EXAML_TIP_TIP) {
@@ -1231,6 +1245,9 @@ struct TravGraph {
 };
 static thread_local std::vector<TravGraph> g_graphs;
 static bool g_use_graphs = true;
+static bool g_fast_math = false;
+
+extern "C" void examl_hip_fast_math(int on) { g_fast_math = on != 0; }
 
 extern "C" void examl_hip_use_graphs(int on) { g_use_graphs = on != 0; }
 
@@ -1263,6 +1280,7 @@ static unsigned long long trav_key(const examl_hip_trav_entry *ops,
   mix((unsigned long long)(uintptr_t)inc);
   mix((unsigned long long)(uintptr_t)stream);
   mix((unsigned long long)states);
+  mix((unsigned long long)(g_fast_math ? 1 : 0));
   for (int e = 0; e < numOps; e++) {
     mix(((unsigned long long)ops[e].tipCase << 48) ^
         ((unsigned long long)(unsigned)ops[e].pNumber << 32) ^
@@ -1385,9 +1403,15 @@ static int traversal_impl(const examl_hip_trav_entry *ops, int numOps,
       hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_TIP_TIP, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
         else
           if (nt)
-      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_TIP, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
+      if (g_fast_math)
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_TIP, true, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
     else
-      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_TIP, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_TIP, true, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
+    else
+      if (g_fast_math)
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_TIP, false, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
+    else
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_TIP, false, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
         break;
       case EXAML_TIP_INNER:
         t1 = dev_tips + (long)op->x1Slot * tipStride;
@@ -1399,9 +1423,15 @@ static int traversal_impl(const examl_hip_trav_entry *ops, int numOps,
       hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_TIP_INNER, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
         else
           if (nt)
-      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_INNER, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
+      if (g_fast_math)
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_INNER, true, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
     else
-      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_INNER, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_INNER, true, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
+    else
+      if (g_fast_math)
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_INNER, false, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
+    else
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_INNER, false, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
         break;
       case EXAML_INNER_INNER:
         x1 = dev_clv + (long)op->x1Slot * clvStride;
@@ -1413,9 +1443,15 @@ static int traversal_impl(const examl_hip_trav_entry *ops, int numOps,
       hipLaunchKernelGGL((k_newview_dna_gamma<EXAML_INNER_INNER, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
         else
           if (nt)
-      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_INNER_INNER, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
+      if (g_fast_math)
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_INNER_INNER, true, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
     else
-      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_INNER_INNER, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_INNER_INNER, true, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
+    else
+      if (g_fast_math)
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_INNER_INNER, false, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
+    else
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_INNER_INNER, false, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P, dev_EV, dev_tipVec, t1, t2, dev_wgt, n, dev_inc + e);
         break;
       default:
         snprintf(g_err, sizeof(g_err), "traversal: bad tipCase %d",
@@ -1760,21 +1796,39 @@ extern "C" int examl_hip_newview_prot_gamma(
   switch (tipCase) {
   case EXAML_TIP_TIP:
     if (nt)
-      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_TIP, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
+      if (g_fast_math)
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_TIP, true, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
     else
-      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_TIP, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_TIP, true, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
+    else
+      if (g_fast_math)
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_TIP, false, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
+    else
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_TIP, false, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
     break;
   case EXAML_TIP_INNER:
     if (nt)
-      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_INNER, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
+      if (g_fast_math)
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_INNER, true, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
     else
-      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_INNER, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_INNER, true, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
+    else
+      if (g_fast_math)
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_INNER, false, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
+    else
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_TIP_INNER, false, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
     break;
   case EXAML_INNER_INNER:
     if (nt)
-      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_INNER_INNER, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
+      if (g_fast_math)
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_INNER_INNER, true, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
     else
-      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_INNER_INNER, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_INNER_INNER, true, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
+    else
+      if (g_fast_math)
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_INNER_INNER, false, true>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
+    else
+      hipLaunchKernelGGL((k_newview_prot_gamma<EXAML_INNER_INNER, false, false>), dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, left, EV, tipVec, tipX1, tipX2, wgt, n, scalerInc);
     break;
   default:
     snprintf(g_err, sizeof(g_err), "newview_prot: bad tipCase %d", tipCase);

@@ -53,6 +53,12 @@ void examl_hip_profile_get(double *ms_by_tc, long *cnt_by_tc);
 void examl_hip_use_graphs(int on);
 void examl_hip_graphs_clear(void);
 
+/* Opt-in fused-multiply-add variant of the protein newview kernel (the
+ * reference's own _FMA build class, avxLikelihood.c:17-19): ~1 ulp/op
+ * difference vs the default bit-exact no-FMA path, roughly half the VALU
+ * instructions.  Off by default; parity tests run with it off. */
+void examl_hip_fast_math(int on);
+
 /* ---------------------------------------------------------------------------
  * Host-side model math (runs once per model-parameter change; feeds the
  * kernels).  These replace the corresponding host functions in the

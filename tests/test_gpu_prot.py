@@ -163,3 +163,20 @@ def test_prot_makenewz_vs_oracle(dev):
     z_gpu = eng.makenewz(tree, p, q, z0)
     z_ref = oracle_makenewz(entries, root, tree, model, tips, wgt, z0)
     assert abs(z_gpu - z_ref) < 1e-9
+
+
+def test_prot_fast_math_variant(dev):
+    """The opt-in FMA protein kernel (the reference's _FMA build class)
+    agrees with the bit-exact path to fp64-roundoff accuracy."""
+    ntips, width = 14, 4096
+    tips, wgt = make_synthetic_aa(ntips, width, seed=9)
+    model = ea.ProtGtrModel.lg(alpha=0.7)
+    tree = ea.PhyloTree.random(ntips, seed=3, rng_z=True)
+    eng = ea.DnaGammaEngine(tips, wgt, model, device=dev)
+    exact = eng.full_lnl(tree).item()
+    ea.lib().examl_hip_fast_math(1)
+    try:
+        fast = eng.full_lnl(tree).item()
+    finally:
+        ea.lib().examl_hip_fast_math(0)
+    assert abs(fast - exact) / abs(exact) < 1e-12
