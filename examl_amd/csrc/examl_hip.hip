@@ -1300,7 +1300,7 @@ static hipGraphExec_t trav_graph_find(unsigned long long key) {
 }
 
 static void trav_graph_store(unsigned long long key, hipGraphExec_t exec) {
-  if (g_graphs.size() >= 32) {
+  if (g_graphs.size() >= 64) {
     for (auto &g : g_graphs) hipGraphExecDestroy(g.exec);
     g_graphs.clear();
   }
@@ -1350,7 +1350,11 @@ static int traversal_impl(const examl_hip_trav_entry *ops, int numOps,
 
   /* graph fast path: identical traversal shape -> replay (P content flows
    * through the captured memcpy from the pinned hostP buffer) */
-  const bool want_graph = g_use_graphs && !g_prof_on && s != nullptr;
+  /* tiny partial traversals (search-mode smoothing probes) change shape
+   * every call: capturing them would thrash the cache at ~1 ms per
+   * capture, so only repeated LARGE shapes go through graphs */
+  const bool want_graph =
+      g_use_graphs && !g_prof_on && s != nullptr && numOps >= 8;
   unsigned long long key = 0;
   bool capturing = false;
   if (want_graph) {

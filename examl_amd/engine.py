@@ -197,6 +197,18 @@ class DnaGammaEngine:
         v = self.d_out2.cpu()
         return float(v[0]), float(v[1])
 
+    def core_derivs_async(self, lz):
+        """Launch execCore without reading back (the search layer batches
+        one sync over all partitions)."""
+        m = self.model
+        self.d_out2.zero_()
+        check(self._fn("core_root")(
+            ctypes.c_long(self.width), _vp(self.d_sum), _np_vp(m.EIGN),
+            _np_vp(m.gammaRates), ctypes.c_double(lz), _vp(self.d_wgt),
+            _vp(self.d_dtab), _vp(self.d_partials), _vp(self.d_out2),
+            self._stream()), "core_root")
+        return self.d_out2
+
     def makenewz(self, tree, p, q, z0, maxiter=64, all_reduce=False):
         """Newton-Raphson branch length at p--q, restating topLevelMakenewz
         (makenewzGenericSpecial.c:1133) for the joint-branch-length case

@@ -83,6 +83,19 @@ class ProtGtrModel:
         self.alpha = float(alpha)
         lib().examl_host_make_gamma_cats(self.alpha, _dp(self.gammaRates), 4)
 
+    def reinit(self):
+        lib().examl_host_init_gtr_aa(_dp(self.frequencies),
+                                     _dp(self.rates190), _dp(self.EIGN),
+                                     _dp(self.EV), _dp(self.EI),
+                                     _dp(self.tipVector))
+
+    def set_matrix(self, rates190, frequencies):
+        """Swap the substitution matrix + base frequencies (the AUTO model
+        search, optimizeModel.c:2631-2632) and re-run initReversibleGTR."""
+        self.rates190[:] = rates190
+        self.frequencies[:] = frequencies
+        self.reinit()
+
     @staticmethod
     def lg(alpha=0.8):
         import os

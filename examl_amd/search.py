@@ -43,7 +43,8 @@ class TreeSearch:
     """One tree + one engine per partition (joint branch lengths,
     numBranches=1)."""
 
-    def __init__(self, tree, engines, opt_freq_flags=None):
+    def __init__(self, tree, engines, opt_freq_flags=None,
+                 auto_flags=None, empirical_freqs=None):
         self.tree = tree
         self.engines = engines
         self.M = len(engines)
@@ -55,6 +56,11 @@ class TreeSearch:
         self.partition_smoothed = True
         self.partition_converged = False
         self.opt_freq_flags = opt_freq_flags or [False] * self.M
+        # AUTO protein model selection state (optimizeModel.c:2669)
+        self.auto_flags = auto_flags or [False] * self.M
+        self.empirical_freqs = empirical_freqs or [None] * self.M
+        self.auto_prot_models = [4] * self.M  # WAG default, models.c:4222
+        self.prot_freqs = [1] * self.M  # AUTO: 1 = fixed/model freqs
         # freqExponents state (models.c:4227: init 0.0)
         self.freq_exponents = [np.zeros(e.model.states) for e in engines]
 
@@ -130,10 +136,19 @@ class TreeSearch:
         self._collect(q, p, not full, out)
         self._run(out)
         z = t.get_z(p, q)
+        # launch all partitions, then ONE host sync for the readbacks
+        outs = []
         for m, eng in enumerate(self.engines):
             if self.execute_model[m]:
-                self.per_partition_lnl[m] = float(
-                    eng.evaluate_root(t, p, q, z))
+                outs.append((m, eng.evaluate_root(t, p, q, z)))
+        if outs and not isinstance(outs[0][1], float):
+            import torch
+            vals = torch.cat([o[1] for o in outs]).cpu()
+            for k, (m, _) in enumerate(outs):
+                self.per_partition_lnl[m] = float(vals[k])
+        else:
+            for m, v in outs:
+                self.per_partition_lnl[m] = v
         self.likelihood = sum(self.per_partition_lnl)
         return self.likelihood
 
@@ -163,9 +178,16 @@ class TreeSearch:
             z = min(max(z, ZMIN), ZMAX)
             lz = math.log(z)
             dlnL = d2lnL = 0.0
-            for m, eng in enumerate(self.engines):
-                if self.execute_model[m]:
-                    a, b = eng.core_derivs(lz)
+            outs = [eng.core_derivs_async(lz)
+                    for m, eng in enumerate(self.engines)
+                    if self.execute_model[m]]
+            if outs and not isinstance(outs[0], tuple):
+                import torch
+                vals = torch.stack(outs).cpu()
+                dlnL = float(vals[:, 0].sum())
+                d2lnL = float(vals[:, 1].sum())
+            else:
+                for a, b in outs:
                     dlnL += a
                     d2lnL += b
             if (d2lnL >= 0.0) and (z < ZMAX):
@@ -600,6 +622,97 @@ class TreeSearch:
                 self._opt_param_generic(groups, aa_valid, rn, -1000000.0,
                                         200.0, FREQ_F, model_epsilon)
 
+    # -- AUTO protein model selection (optimizeModel.c:2606-2900) ----------
+
+    def _aa_table(self):
+        import os
+        if not hasattr(self, "_aa_data"):
+            self._aa_data = np.load(os.path.join(
+                os.path.dirname(os.path.abspath(__file__)), "data",
+                "aa_models.npz"))
+        return self._aa_data
+
+    def _set_auto_matrix(self, m, index, prot_freqs):
+        """autoProtModels = index; protFreqs semantics for AUTO partitions:
+        1 = the matrix's own frequencies, 0 = empirical
+        (models.c:3528-3534)."""
+        tab = self._aa_table()
+        freqs = tab["frequencies"][index] if prot_freqs \
+            else self.empirical_freqs[m]
+        self.engines[m].model.set_matrix(tab["rates190"][index], freqs)
+        self.engines[m].upload_model()
+        self.auto_prot_models[m] = index
+        self.prot_freqs[m] = prot_freqs
+
+    def reset_branches(self):
+        """resetBranches (optimizeModel.c:2509): every z to defaultz."""
+        from .tree import DEFAULTZ
+        for a, b in self.tree.edges():
+            self.tree.set_z(a, b, DEFAULTZ)
+
+    def _save_tree_z(self):
+        return {e: self.tree.get_z(*e) for e in self.tree.edges()}
+
+    def _restore_tree_z(self, saved):
+        for (a, b), z in saved.items():
+            self.tree.set_z(a, b, z)
+
+    def _opt_model_pass(self, fixed_freqs):
+        """optModel (optimizeModel.c:2606): try every candidate matrix on
+        all AUTO partitions simultaneously; returns (bestIndex, bestScores)
+        per partition."""
+        best_index = [-1] * self.M
+        best_scores = [UNLIKELY] * self.M
+        for i in range(19):  # AUTO = 19 candidate matrices (axml.h:261)
+            for m in range(self.M):
+                if self.auto_flags[m]:
+                    self._set_auto_matrix(m, i, 1 if fixed_freqs else 0)
+            self.reset_branches()
+            self.evaluate_generic(full=True)
+            self.tree_evaluate(0.5)
+            for m in range(self.M):
+                if self.auto_flags[m] and \
+                        self.per_partition_lnl[m] > best_scores[m]:
+                    best_scores[m] = self.per_partition_lnl[m]
+                    best_index[m] = i
+        return best_index, best_scores
+
+    def auto_protein(self, log=None):
+        """autoProtein (optimizeModel.c:2669), ML criterion (the default,
+        axml.c:982)."""
+        if not any(self.auto_flags):
+            return
+        saved_z = self._save_tree_z()
+        old_index = list(self.auto_prot_models)
+        old_freqs = list(self.prot_freqs)
+        self.evaluate_generic(full=True)
+        start_lh = self.likelihood
+        bi_fixed, bs_fixed = self._opt_model_pass(fixed_freqs=True)
+        bi_emp, bs_emp = self._opt_model_pass(fixed_freqs=False)
+        for m in range(self.M):
+            if not self.auto_flags[m]:
+                continue
+            if bs_fixed[m] > bs_emp[m]:  # AUTO_ML
+                self._set_auto_matrix(m, bi_fixed[m], 1)
+            else:
+                self._set_auto_matrix(m, bi_emp[m], 0)
+            if log:
+                tab = self._aa_table()
+                log(f"AUTO partition {m}: "
+                    f"{tab['names'][self.auto_prot_models[m]]} "
+                    f"({'fixed' if self.prot_freqs[m] else 'empirical'} "
+                    f"freqs)")
+        self.reset_branches()
+        self.evaluate_generic(full=True)
+        self.tree_evaluate(2.0)
+        if self.likelihood < start_lh:
+            for m in range(self.M):
+                if self.auto_flags[m]:
+                    self._set_auto_matrix(m, old_index[m], old_freqs[m])
+            self._restore_tree_z(saved_z)
+            self.evaluate_generic(full=True)
+        assert self.likelihood >= start_lh - 1e-6
+
     def mod_opt(self, likelihood_epsilon=0.1, model_epsilon=0.0001,
                 log=None):
         """modOpt (optimizeModel.c:2963) for the GAMMA model."""
@@ -608,6 +721,7 @@ class TreeSearch:
             current = self.likelihood
             self.opt_rates_generic(model_epsilon)
             self.evaluate_generic(full=True)
+            self.auto_protein(log=log)
             self.tree_evaluate(0.0625)
             self.evaluate_generic(full=True)
             self.opt_base_freqs(model_epsilon)

@@ -199,6 +199,9 @@ class OracleEngine:
         EIGN, EV, EI, tipVector, g = self._arrays()
         return self._core(self.width, self.sumtable, EIGN, g, lz, self.wgt)
 
+    def core_derivs_async(self, lz):
+        return self.core_derivs(lz)
+
 
 def oracle_makenewz(entries, root, tree, model, tips, wgt, z0, maxiter=64):
     """CPU restatement of topLevelMakenewz (numBranches=1) over the oracle
