@@ -1307,6 +1307,18 @@ static void trav_graph_store(unsigned long long key, hipGraphExec_t exec) {
   g_graphs.push_back({key, exec});
 }
 
+/* small pinned staging for per-call diag/dtable uploads (a pageable
+ * hipMemcpyAsync blocks the host ~5-10 us; these sit on every Brent probe
+ * and NR iteration) */
+static thread_local double *g_smallPin = nullptr;
+static double *small_pin(void) {
+  if (!g_smallPin) {
+    if (hipHostMalloc((void **)&g_smallPin, 4096) != hipSuccess)
+      g_smallPin = (double *)malloc(4096);
+  }
+  return g_smallPin;
+}
+
 /* pinned host staging for the P blocks (stable address: the captured
  * memcpy node re-reads it on every replay) */
 static thread_local double *g_hostP = nullptr;
@@ -1537,9 +1549,9 @@ extern "C" int examl_hip_evaluate_root_dna_gamma(
   if (n <= 0) return 0;
   hipStream_t s = (hipStream_t)stream;
   (void)hipGetLastError(); /* clear stale per-thread error (e.g. torch probes) */
-  double hostDiag[16];
+  double *hostDiag = small_pin();
   examl_host_calc_diagptable(z, 4, 4, gammaRates, EIGN, hostDiag);
-  CHK(hipMemcpyAsync(dev_diag, hostDiag, sizeof(hostDiag),
+  CHK(hipMemcpyAsync(dev_diag, hostDiag, 16 * sizeof(double),
                      hipMemcpyHostToDevice, s));
   const double log_minlik = log(MINLIKELIHOOD);
   const unsigned int *gsP = dev_scalers ? dev_scalers + pNumber : nullptr;
@@ -1609,10 +1621,10 @@ extern "C" int examl_hip_core_root_dna_gamma(long n, const double *dev_sum,
   if (n <= 0) return 0;
   hipStream_t s = (hipStream_t)stream;
   (void)hipGetLastError(); /* clear stale per-thread error (e.g. torch probes) */
-  double host48[48];
+  double *host48 = small_pin();
   examl_host_core_dtables_dna(EIGN, gammaRates, lz, host48);
-  CHK(hipMemcpyAsync(dev_dtab, host48, sizeof(host48), hipMemcpyHostToDevice,
-                     s));
+  CHK(hipMemcpyAsync(dev_dtab, host48, 48 * sizeof(double),
+                     hipMemcpyHostToDevice, s));
   return examl_hip_core_dna_gamma(n, dev_sum, dev_dtab, dev_wgt,
                                   dev_partials, dev_out2, stream);
 }
@@ -1942,9 +1954,9 @@ extern "C" int examl_hip_evaluate_root_prot_gamma(
   if (n <= 0) return 0;
   hipStream_t s = (hipStream_t)stream;
   (void)hipGetLastError();
-  double hostDiag[80];
+  double *hostDiag = small_pin();
   examl_host_calc_diagptable(z, 20, 4, gammaRates, EIGN, hostDiag);
-  CHK(hipMemcpyAsync(dev_diag, hostDiag, sizeof(hostDiag),
+  CHK(hipMemcpyAsync(dev_diag, hostDiag, 80 * sizeof(double),
                      hipMemcpyHostToDevice, s));
   const double log_minlik = log(MINLIKELIHOOD);
   const unsigned int *gsP = dev_scalers ? dev_scalers + pNumber : nullptr;
@@ -2010,9 +2022,9 @@ extern "C" int examl_hip_core_root_prot_gamma(
   if (n <= 0) return 0;
   hipStream_t s = (hipStream_t)stream;
   (void)hipGetLastError();
-  double host240[240];
+  double *host240 = small_pin();
   examl_host_core_dtables_prot(EIGN, gammaRates, lz, host240);
-  CHK(hipMemcpyAsync(dev_dtab, host240, sizeof(host240),
+  CHK(hipMemcpyAsync(dev_dtab, host240, 240 * sizeof(double),
                      hipMemcpyHostToDevice, s));
   return examl_hip_core_prot_gamma(n, dev_sum, dev_dtab, dev_wgt,
                                    dev_partials, dev_out2, stream);
