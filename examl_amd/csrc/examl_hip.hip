@@ -1307,18 +1307,6 @@ static void trav_graph_store(unsigned long long key, hipGraphExec_t exec) {
   g_graphs.push_back({key, exec});
 }
 
-/* small pinned staging for per-call diag/dtable uploads (a pageable
- * hipMemcpyAsync blocks the host ~5-10 us; these sit on every Brent probe
- * and NR iteration) */
-static thread_local double *g_smallPin = nullptr;
-static double *small_pin(void) {
-  if (!g_smallPin) {
-    if (hipHostMalloc((void **)&g_smallPin, 4096) != hipSuccess)
-      g_smallPin = (double *)malloc(4096);
-  }
-  return g_smallPin;
-}
-
 /* pinned host staging for the P blocks (stable address: the captured
  * memcpy node re-reads it on every replay) */
 static thread_local double *g_hostP = nullptr;
@@ -1549,9 +1537,12 @@ extern "C" int examl_hip_evaluate_root_dna_gamma(
   if (n <= 0) return 0;
   hipStream_t s = (hipStream_t)stream;
   (void)hipGetLastError(); /* clear stale per-thread error (e.g. torch probes) */
-  double *hostDiag = small_pin();
+  /* stack buffer: a pageable hipMemcpyAsync is host-synchronous, so the
+   * buffer is safely reusable on return (a shared pinned buffer would race
+   * with its own in-flight copies) */
+  double hostDiag[16];
   examl_host_calc_diagptable(z, 4, 4, gammaRates, EIGN, hostDiag);
-  CHK(hipMemcpyAsync(dev_diag, hostDiag, 16 * sizeof(double),
+  CHK(hipMemcpyAsync(dev_diag, hostDiag, sizeof(hostDiag),
                      hipMemcpyHostToDevice, s));
   const double log_minlik = log(MINLIKELIHOOD);
   const unsigned int *gsP = dev_scalers ? dev_scalers + pNumber : nullptr;
@@ -1621,10 +1612,10 @@ extern "C" int examl_hip_core_root_dna_gamma(long n, const double *dev_sum,
   if (n <= 0) return 0;
   hipStream_t s = (hipStream_t)stream;
   (void)hipGetLastError(); /* clear stale per-thread error (e.g. torch probes) */
-  double *host48 = small_pin();
+  double host48[48];
   examl_host_core_dtables_dna(EIGN, gammaRates, lz, host48);
-  CHK(hipMemcpyAsync(dev_dtab, host48, 48 * sizeof(double),
-                     hipMemcpyHostToDevice, s));
+  CHK(hipMemcpyAsync(dev_dtab, host48, sizeof(host48), hipMemcpyHostToDevice,
+                     s));
   return examl_hip_core_dna_gamma(n, dev_sum, dev_dtab, dev_wgt,
                                   dev_partials, dev_out2, stream);
 }
@@ -1954,9 +1945,9 @@ extern "C" int examl_hip_evaluate_root_prot_gamma(
   if (n <= 0) return 0;
   hipStream_t s = (hipStream_t)stream;
   (void)hipGetLastError();
-  double *hostDiag = small_pin();
+  double hostDiag[80];
   examl_host_calc_diagptable(z, 20, 4, gammaRates, EIGN, hostDiag);
-  CHK(hipMemcpyAsync(dev_diag, hostDiag, 80 * sizeof(double),
+  CHK(hipMemcpyAsync(dev_diag, hostDiag, sizeof(hostDiag),
                      hipMemcpyHostToDevice, s));
   const double log_minlik = log(MINLIKELIHOOD);
   const unsigned int *gsP = dev_scalers ? dev_scalers + pNumber : nullptr;
@@ -2022,9 +2013,9 @@ extern "C" int examl_hip_core_root_prot_gamma(
   if (n <= 0) return 0;
   hipStream_t s = (hipStream_t)stream;
   (void)hipGetLastError();
-  double *host240 = small_pin();
+  double host240[240];
   examl_host_core_dtables_prot(EIGN, gammaRates, lz, host240);
-  CHK(hipMemcpyAsync(dev_dtab, host240, 240 * sizeof(double),
+  CHK(hipMemcpyAsync(dev_dtab, host240, sizeof(host240),
                      hipMemcpyHostToDevice, s));
   return examl_hip_core_prot_gamma(n, dev_sum, dev_dtab, dev_wgt,
                                    dev_partials, dev_out2, stream);
