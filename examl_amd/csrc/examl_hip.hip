@@ -1307,22 +1307,48 @@ static void trav_graph_store(unsigned long long key, hipGraphExec_t exec) {
   g_graphs.push_back({key, exec});
 }
 
-/* pinned host staging for the P blocks (stable address: the captured
- * memcpy node re-reads it on every replay) */
-static thread_local double *g_hostP = nullptr;
-static thread_local size_t g_hostPCap = 0;
+/* Pinned host staging for the P blocks, ONE SLOT PER ENGINE (keyed by its
+ * device P buffer) with a completion event:
+ *  - the address is stable, so a cached graph's memcpy node can re-read it
+ *    on every replay;
+ *  - hipMemcpyAsync from PINNED memory is truly asynchronous, so before
+ *    REFILLING a slot we must wait for its previous copy — the event is
+ *    recorded right after the copy (inside the capture, so replays
+ *    re-record it too).  A single shared buffer raced across partitions:
+ *    engine B's host-side refill corrupted engine A's still-queued copy.
+ */
+struct HostPSlot {
+  const void *key;
+  double *buf;
+  size_t cap;
+  hipEvent_t ev;
+  bool ev_valid;
+};
+static thread_local std::vector<HostPSlot> g_hostP_slots;
 
-static double *hostP_get(size_t doubles) {
-  if (doubles > g_hostPCap) {
-    if (g_hostP) hipHostFree(g_hostP);
-    size_t cap = doubles * 2;
-    if (hipHostMalloc((void **)&g_hostP, cap * sizeof(double)) !=
-        hipSuccess) {
-      g_hostP = (double *)malloc(cap * sizeof(double));
+static HostPSlot *hostP_get(const void *dev_pbuf, size_t doubles) {
+  HostPSlot *slot = nullptr;
+  for (auto &sl : g_hostP_slots)
+    if (sl.key == dev_pbuf) {
+      slot = &sl;
+      break;
     }
-    g_hostPCap = cap;
+  if (!slot) {
+    g_hostP_slots.push_back({dev_pbuf, nullptr, 0, nullptr, false});
+    slot = &g_hostP_slots.back();
+    hipEventCreateWithFlags(&slot->ev, hipEventDisableTiming);
   }
-  return g_hostP;
+  if (slot->ev_valid) hipEventSynchronize(slot->ev);
+  if (doubles > slot->cap) {
+    if (slot->buf) hipHostFree(slot->buf);
+    slot->cap = doubles * 2;
+    if (hipHostMalloc((void **)&slot->buf, slot->cap * sizeof(double)) !=
+        hipSuccess)
+      slot->buf = (double *)malloc(slot->cap * sizeof(double));
+    /* cached graphs may hold the old address */
+    examl_hip_graphs_clear();
+  }
+  return slot;
 }
 
 /* --- batched traversal (newviewIterative body) --------------------------- */
@@ -1337,7 +1363,8 @@ static int traversal_impl(const examl_hip_trav_entry *ops, int numOps,
                           unsigned int *dev_scalers, unsigned int *dev_inc,
                           double *dev_pbuf, hipStream_t s) {
   constexpr int PBLK = 8 * STATES * STATES; /* left|right, 4 cats */
-  double *hostP = hostP_get((size_t)numOps * PBLK);
+  HostPSlot *pslot = hostP_get(dev_pbuf, (size_t)numOps * PBLK);
+  double *hostP = pslot->buf;
 
   /* 1. all P-matrix pairs on the host (newviewGenericSpecial.c:982-1044) */
   for (int e = 0; e < numOps; e++) {
@@ -1378,6 +1405,10 @@ static int traversal_impl(const examl_hip_trav_entry *ops, int numOps,
                                     (size_t)numOps * PBLK * sizeof(double),
                                     hipMemcpyHostToDevice, s);
     if (err != hipSuccess) { rc = set_err(err, "pbuf upload"); break; }
+    /* completion marker for the NEXT refill of this slot (captured as a
+     * graph node when capturing, so replays re-record it) */
+    hipEventRecord(pslot->ev, s);
+    pslot->ev_valid = true;
     err = hipMemsetAsync(dev_inc, 0, (size_t)numOps * sizeof(unsigned int),
                          s);
     if (err != hipSuccess) { rc = set_err(err, "inc memset"); break; }
@@ -1499,11 +1530,21 @@ static int traversal_impl(const examl_hip_trav_entry *ops, int numOps,
       if (graph) hipGraphDestroy(graph);
       return rc;
     }
-    if (err != hipSuccess) return set_err(err, "end capture");
     hipGraphExec_t exec = nullptr;
-    err = hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0);
-    hipGraphDestroy(graph);
-    if (err != hipSuccess) return set_err(err, "graph instantiate");
+    if (err == hipSuccess) {
+      err = hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0);
+      hipGraphDestroy(graph);
+    }
+    if (err != hipSuccess) {
+      /* the captured sequence never executed: disable graphs and run it
+       * for real */
+      (void)hipGetLastError();
+      g_use_graphs = false;
+      return traversal_impl<STATES>(ops, numOps, EIGN, EI, gammaRates,
+                                    dev_EV, dev_tipVec, dev_clv, clvStride,
+                                    dev_tips, tipStride, dev_wgt, n,
+                                    dev_scalers, dev_inc, dev_pbuf, s);
+    }
     trav_graph_store(key, exec);
     CHK(hipGraphLaunch(exec, s));
   }
