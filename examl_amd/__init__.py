@@ -102,6 +102,12 @@ def _bind(lib):
     lib.examl_hip_sum_dna_cat.argtypes = [i, p, p, p, p, p, p, l, p]
     lib.examl_hip_core_root_dna_cat.argtypes = \
         [l, p, p, p, i, d, p, p, p, p, p, p]
+    lib.examl_hip_newview_traversal_dna_cat.argtypes = \
+        [p, i, p, p, p, i, p, p, p, p, l, p, l, p, l, p, p, p, p]
+    lib.examl_hip_evaluate_root_dna_cat_x.argtypes = \
+        [i, i, i, i, i, i, d, p, p, i, p, p, p, l, p, l, p, l, p, p, p, p, p]
+    lib.examl_hip_sum_root_dna_cat.argtypes = \
+        [i, i, i, i, i, p, p, l, p, l, p, l, p]
     lib.examl_hip_use_graphs.argtypes = [i]
     lib.examl_hip_graphs_clear.argtypes = []
     lib.examl_hip_fast_math.argtypes = [i]
@@ -134,9 +140,10 @@ def check(rc, what):
 
 from .model import DnaGtrModel, ProtGtrModel  # noqa: E402
 from .tree import PhyloTree             # noqa: E402
-from .engine import DnaGammaEngine      # noqa: E402
+from .engine import DnaCatEngine, DnaGammaEngine  # noqa: E402
 
 __all__ = [
     "lib", "check", "TravEntry", "DnaGtrModel", "ProtGtrModel", "PhyloTree",
-    "DnaGammaEngine", "TIP_TIP", "TIP_INNER", "INNER_INNER", "ZMIN", "ZMAX",
+    "DnaGammaEngine", "DnaCatEngine", "TIP_TIP", "TIP_INNER",
+    "INNER_INNER", "ZMIN", "ZMAX",
 ]

@@ -1820,6 +1820,238 @@ extern "C" int examl_hip_core_root_dna_cat(
   return 0;
 }
 
+/* --- CAT executors (newviewIterative / evaluateIterative / makenewz CAT
+ * bodies; per-site rate categories dev_cptr, numCats P pairs per op) ------ */
+
+extern "C" int examl_hip_newview_traversal_dna_cat(
+    const examl_hip_trav_entry *ops, int numOps, const double *EIGN,
+    const double *EI, const double *perSiteRates, int numCats,
+    const double *dev_EV, const double *dev_tipVec, const int *dev_cptr,
+    double *dev_clv, long clvStride, const unsigned char *dev_tips,
+    long tipStride, const int *dev_wgt, long n, unsigned int *dev_scalers,
+    unsigned int *dev_inc, double *dev_pbuf, void *stream) {
+  if (numOps <= 0 || n <= 0) return 0;
+  if (numCats > MAX_CAT) {
+    snprintf(g_err, sizeof(g_err), "traversal_cat: numCats %d > %d", numCats,
+             MAX_CAT);
+    return -1;
+  }
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  const int PBLK = numCats * 32; /* left|right, numCats categories */
+  HostPSlot *pslot = hostP_get(dev_pbuf, (size_t)numOps * PBLK);
+  double *hostP = pslot->buf;
+  for (int e = 0; e < numOps; e++) {
+    double qz = ops[e].qz, rz = ops[e].rz;
+    qz = (qz > ZMIN) ? log(qz) : log(ZMIN);
+    rz = (rz > ZMIN) ? log(rz) : log(ZMIN);
+    examl_host_make_p(qz, rz, perSiteRates, EI, EIGN, numCats,
+                      &hostP[e * PBLK], &hostP[e * PBLK + PBLK / 2], 4);
+  }
+
+  const bool want_graph =
+      g_use_graphs && !g_prof_on && s != nullptr && numOps >= 8;
+  unsigned long long key = 0;
+  bool capturing = false;
+  if (want_graph) {
+    key = trav_key(ops, numOps, n, dev_clv, dev_tips, dev_pbuf, dev_EV,
+                   dev_tipVec, dev_wgt, dev_scalers, dev_inc, (void *)s,
+                   1000 + numCats /* distinguish CAT shapes */);
+    hipGraphExec_t exec = trav_graph_find(key);
+    if (exec) {
+      CHK(hipGraphLaunch(exec, s));
+      return 0;
+    }
+    capturing =
+        hipStreamBeginCapture(s, hipStreamCaptureModeThreadLocal) ==
+        hipSuccess;
+    (void)hipGetLastError();
+  }
+
+  int rc = 0;
+  do {
+    hipError_t err = hipMemcpyAsync(dev_pbuf, hostP,
+                                    (size_t)numOps * PBLK * sizeof(double),
+                                    hipMemcpyHostToDevice, s);
+    if (err != hipSuccess) { rc = set_err(err, "cat pbuf upload"); break; }
+    hipEventRecord(pslot->ev, s);
+    pslot->ev_valid = true;
+    err = hipMemsetAsync(dev_inc, 0, (size_t)numOps * sizeof(unsigned int),
+                         s);
+    if (err != hipSuccess) { rc = set_err(err, "cat inc memset"); break; }
+
+    const int grid = grid_for(n);
+    const bool nt = n >= 262144;
+    for (int e = 0; e < numOps && rc == 0; e++) {
+      const examl_hip_trav_entry *op = &ops[e];
+      const double *P = dev_pbuf + (long)e * PBLK;
+      double *x3 = dev_clv + (long)op->x3Slot * clvStride;
+      const double *x1 = nullptr, *x2 = nullptr;
+      const unsigned char *t1 = nullptr, *t2 = nullptr;
+      switch (op->tipCase) {
+      case EXAML_TIP_TIP:
+        t1 = dev_tips + (long)op->x1Slot * tipStride;
+        t2 = dev_tips + (long)op->x2Slot * tipStride;
+        if (nt)
+          hipLaunchKernelGGL((k_newview_dna_cat<EXAML_TIP_TIP, true>),
+                             dim3(grid), dim3(NV_BLOCK), 0, s, dev_EV,
+                             dev_cptr, x1, x2, x3, dev_tipVec, t1, t2,
+                             dev_wgt, n, P, numCats, dev_inc + e);
+        else
+          hipLaunchKernelGGL((k_newview_dna_cat<EXAML_TIP_TIP, false>),
+                             dim3(grid), dim3(NV_BLOCK), 0, s, dev_EV,
+                             dev_cptr, x1, x2, x3, dev_tipVec, t1, t2,
+                             dev_wgt, n, P, numCats, dev_inc + e);
+        break;
+      case EXAML_TIP_INNER:
+        t1 = dev_tips + (long)op->x1Slot * tipStride;
+        x2 = dev_clv + (long)op->x2Slot * clvStride;
+        if (nt)
+          hipLaunchKernelGGL((k_newview_dna_cat<EXAML_TIP_INNER, true>),
+                             dim3(grid), dim3(NV_BLOCK), 0, s, dev_EV,
+                             dev_cptr, x1, x2, x3, dev_tipVec, t1, t2,
+                             dev_wgt, n, P, numCats, dev_inc + e);
+        else
+          hipLaunchKernelGGL((k_newview_dna_cat<EXAML_TIP_INNER, false>),
+                             dim3(grid), dim3(NV_BLOCK), 0, s, dev_EV,
+                             dev_cptr, x1, x2, x3, dev_tipVec, t1, t2,
+                             dev_wgt, n, P, numCats, dev_inc + e);
+        break;
+      case EXAML_INNER_INNER:
+        x1 = dev_clv + (long)op->x1Slot * clvStride;
+        x2 = dev_clv + (long)op->x2Slot * clvStride;
+        if (nt)
+          hipLaunchKernelGGL((k_newview_dna_cat<EXAML_INNER_INNER, true>),
+                             dim3(grid), dim3(NV_BLOCK), 0, s, dev_EV,
+                             dev_cptr, x1, x2, x3, dev_tipVec, t1, t2,
+                             dev_wgt, n, P, numCats, dev_inc + e);
+        else
+          hipLaunchKernelGGL((k_newview_dna_cat<EXAML_INNER_INNER, false>),
+                             dim3(grid), dim3(NV_BLOCK), 0, s, dev_EV,
+                             dev_cptr, x1, x2, x3, dev_tipVec, t1, t2,
+                             dev_wgt, n, P, numCats, dev_inc + e);
+        break;
+      default:
+        snprintf(g_err, sizeof(g_err), "traversal_cat: bad tipCase %d",
+                 op->tipCase);
+        rc = -1;
+        break;
+      }
+      if (rc == 0) {
+        err = hipGetLastError();
+        if (err != hipSuccess) rc = set_err(err, "cat newview launch");
+      }
+    }
+    if (rc != 0) break;
+    for (int base = 0; base < numOps && rc == 0; base += FIN_CHUNK) {
+      FinMeta m;
+      m.count = (numOps - base < FIN_CHUNK) ? (numOps - base) : FIN_CHUNK;
+      m.base = base;
+      for (int e = 0; e < m.count; e++) {
+        m.p[e] = ops[base + e].pNumber;
+        m.q[e] = ops[base + e].qNumber;
+        m.r[e] = ops[base + e].rNumber;
+      }
+      hipLaunchKernelGGL(k_scaler_finalize, dim3(1), dim3(64), 0, s, m,
+                         dev_inc, dev_scalers);
+      err = hipGetLastError();
+      if (err != hipSuccess) rc = set_err(err, "cat finalize launch");
+    }
+  } while (0);
+
+  if (capturing) {
+    hipGraph_t graph = nullptr;
+    hipError_t err = hipStreamEndCapture(s, &graph);
+    if (rc != 0) {
+      if (graph) hipGraphDestroy(graph);
+      return rc;
+    }
+    hipGraphExec_t exec = nullptr;
+    if (err == hipSuccess) {
+      err = hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0);
+      hipGraphDestroy(graph);
+    }
+    if (err != hipSuccess) {
+      (void)hipGetLastError();
+      g_use_graphs = false;
+      return examl_hip_newview_traversal_dna_cat(
+          ops, numOps, EIGN, EI, perSiteRates, numCats, dev_EV, dev_tipVec,
+          dev_cptr, dev_clv, clvStride, dev_tips, tipStride, dev_wgt, n,
+          dev_scalers, dev_inc, dev_pbuf, (void *)s);
+    }
+    trav_graph_store(key, exec);
+    CHK(hipGraphLaunch(exec, s));
+  }
+  return rc;
+}
+
+extern "C" int examl_hip_evaluate_root_dna_cat_x(
+    int rootTipCase, int pNumber, int qNumber, int x1Slot, int x2Slot,
+    int tipSlot, double z, const double *EIGN, const double *perSiteRates,
+    int numCats, const double *dev_tipVec, const int *dev_cptr,
+    double *dev_clv, long clvStride, const unsigned char *dev_tips,
+    long tipStride, const int *dev_wgt, long n,
+    const unsigned int *dev_scalers, double *dev_diag, double *dev_partials,
+    double *dev_lnl, void *stream) {
+  if (n <= 0) return 0;
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  double hostDiag[MAX_CAT * 4];
+  examl_host_calc_diagptable(z, 4, numCats, perSiteRates, EIGN, hostDiag);
+  CHK(hipMemcpyAsync(dev_diag, hostDiag,
+                     (size_t)numCats * 4 * sizeof(double),
+                     hipMemcpyHostToDevice, s));
+  const double log_minlik = log(MINLIKELIHOOD);
+  const unsigned int *gsP = dev_scalers ? dev_scalers + pNumber : nullptr;
+  const unsigned int *gsQ = dev_scalers ? dev_scalers + qNumber : nullptr;
+  const double *x1 = nullptr, *x2 = nullptr;
+  const unsigned char *t1 = nullptr;
+  if (rootTipCase == EXAML_TIP_INNER) {
+    t1 = dev_tips + (long)tipSlot * tipStride;
+    x2 = dev_clv + (long)x2Slot * clvStride;
+  } else if (rootTipCase == EXAML_INNER_INNER) {
+    x1 = dev_clv + (long)x1Slot * clvStride;
+    x2 = dev_clv + (long)x2Slot * clvStride;
+  } else {
+    snprintf(g_err, sizeof(g_err), "evaluate_root_cat: bad tipCase %d",
+             rootTipCase);
+    return -1;
+  }
+  return examl_hip_evaluate_dna_cat(dev_cptr, dev_wgt, x1, x2, dev_tipVec,
+                                    t1, n, dev_diag, numCats, gsP, gsQ,
+                                    log_minlik, dev_partials, dev_lnl,
+                                    (void *)s);
+}
+
+extern "C" int examl_hip_sum_root_dna_cat(
+    int rootTipCase, int x1Slot, int x2Slot, int tipSlot, int tipSlot2,
+    const double *dev_tipVec, double *dev_clv, long clvStride,
+    const unsigned char *dev_tips, long tipStride, double *dev_sum, long n,
+    void *stream) {
+  const double *x1 = nullptr, *x2 = nullptr;
+  const unsigned char *t1 = nullptr, *t2 = nullptr;
+  switch (rootTipCase) {
+  case EXAML_TIP_TIP:
+    t1 = dev_tips + (long)tipSlot * tipStride;
+    t2 = dev_tips + (long)tipSlot2 * tipStride;
+    break;
+  case EXAML_TIP_INNER:
+    t1 = dev_tips + (long)tipSlot * tipStride;
+    x2 = dev_clv + (long)x2Slot * clvStride;
+    break;
+  case EXAML_INNER_INNER:
+    x1 = dev_clv + (long)x1Slot * clvStride;
+    x2 = dev_clv + (long)x2Slot * clvStride;
+    break;
+  default:
+    snprintf(g_err, sizeof(g_err), "sum_root_cat: bad tipCase %d",
+             rootTipCase);
+    return -1;
+  }
+  return examl_hip_sum_dna_cat(rootTipCase, dev_sum, x1, x2, dev_tipVec, t1,
+                               t2, n, stream);
+}
+
 /* ===========================================================================
  * Protein launchers + executors (states=20; same shapes as the DNA ones)
  * ==========================================================================*/

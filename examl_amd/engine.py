@@ -250,3 +250,100 @@ class DnaGammaEngine:
             else:
                 outer_converged = True
         return z
+
+
+class DnaCatEngine(DnaGammaEngine):
+    """CAT (PSR, -m PSR) variant: span 4, per-site rate category cptr and
+    perSiteRates (pInfo.rateCategory / perSiteRates, axml.h:591-592) in
+    place of the 4 discrete gamma rates."""
+
+    def __init__(self, tips, wgt, model, cptr, per_site_rates, device="cuda",
+                 max_ops=None):
+        self.cptr = np.ascontiguousarray(cptr, dtype=np.int32)
+        self.per_site_rates = np.ascontiguousarray(per_site_rates,
+                                                   dtype=np.float64)
+        super().__init__(tips, wgt, model, device=device, max_ops=max_ops)
+        self.SPAN = 4
+        self.num_cats = len(self.per_site_rates)
+        dev = self.device
+        self.d_cptr = torch.from_numpy(self.cptr).to(dev)
+        # re-size CAT-specific buffers (base class sized them for GAMMA)
+        self.d_clv = torch.empty((self.ninner, self.width * 4),
+                                 dtype=torch.float64, device=dev)
+        self.d_pbuf = torch.empty(self._max_ops * self.num_cats * 32,
+                                  dtype=torch.float64, device=dev)
+        self.d_diag = torch.empty(self.num_cats * 4, dtype=torch.float64,
+                                  device=dev)
+        self.d_dtab = torch.empty(self.num_cats * 4 + 8 + self.num_cats,
+                                  dtype=torch.float64, device=dev)
+
+    def newview_traversal(self, entries):
+        if not entries:
+            return
+        assert len(entries) <= self._max_ops, "grow max_ops"
+        arr = (TravEntry * len(entries))(*entries)
+        m = self.model
+        check(lib().examl_hip_newview_traversal_dna_cat(
+            ctypes.cast(arr, ctypes.c_void_p), len(entries),
+            _np_vp(m.EIGN), _np_vp(m.EI), _np_vp(self.per_site_rates),
+            self.num_cats, _vp(self.d_EV), _vp(self.d_tipVector),
+            _vp(self.d_cptr), _vp(self.d_clv),
+            ctypes.c_long(self.width * 4), _vp(self.d_tips),
+            ctypes.c_long(self.width), _vp(self.d_wgt),
+            ctypes.c_long(self.width), _vp(self.d_scalers), _vp(self.d_inc),
+            _vp(self.d_pbuf), self._stream()), "newview_traversal_cat")
+
+    def evaluate_root(self, tree, p, q, z, all_reduce=False):
+        tc, x1s, x2s, tslot, _, pn, qn = self._root_case(tree, p, q)
+        self.d_lnl.zero_()
+        m = self.model
+        check(lib().examl_hip_evaluate_root_dna_cat_x(
+            tc, pn, qn, x1s, x2s, tslot, ctypes.c_double(z),
+            _np_vp(m.EIGN), _np_vp(self.per_site_rates), self.num_cats,
+            _vp(self.d_tipVector), _vp(self.d_cptr), _vp(self.d_clv),
+            ctypes.c_long(self.width * 4), _vp(self.d_tips),
+            ctypes.c_long(self.width), _vp(self.d_wgt),
+            ctypes.c_long(self.width), _vp(self.d_scalers), _vp(self.d_diag),
+            _vp(self.d_partials), _vp(self.d_lnl), self._stream()),
+            "evaluate_root_cat")
+        if all_reduce and torch.distributed.is_initialized():
+            torch.distributed.all_reduce(self.d_lnl)
+        return self.d_lnl
+
+    def sum_root(self, tree, p, q):
+        if self.d_sum is None:
+            self.d_sum = torch.empty(self.width * 4, dtype=torch.float64,
+                                     device=self.device)
+        p_tip, q_tip = tree.is_tip(p), tree.is_tip(q)
+        if p_tip and q_tip:
+            tc, x1s, x2s, t1, t2 = TIP_TIP, -1, -1, p, q
+        elif q_tip:
+            tc, x1s, x2s, t1, t2 = TIP_INNER, -1, tree.clv_slot(p), q, -1
+        elif p_tip:
+            tc, x1s, x2s, t1, t2 = TIP_INNER, -1, tree.clv_slot(q), p, -1
+        else:
+            tc, x1s, x2s, t1, t2 = (INNER_INNER, tree.clv_slot(p),
+                                    tree.clv_slot(q), -1, -1)
+        check(lib().examl_hip_sum_root_dna_cat(
+            tc, x1s, x2s, t1, t2, _vp(self.d_tipVector), _vp(self.d_clv),
+            ctypes.c_long(self.width * 4), _vp(self.d_tips),
+            ctypes.c_long(self.width), _vp(self.d_sum),
+            ctypes.c_long(self.width), self._stream()), "sum_root_cat")
+
+    def core_derivs_async(self, lz):
+        m = self.model
+        self.d_out2.zero_()
+        check(lib().examl_hip_core_root_dna_cat(
+            ctypes.c_long(self.width), _vp(self.d_sum), _np_vp(m.EIGN),
+            _np_vp(self.per_site_rates), self.num_cats, ctypes.c_double(lz),
+            _vp(self.d_wgt), _vp(self.d_cptr), _vp(self.d_dtab),
+            _vp(self.d_partials), _vp(self.d_out2), self._stream()),
+            "core_root_cat")
+        return self.d_out2
+
+    def core_derivs(self, lz, all_reduce=False):
+        self.core_derivs_async(lz)
+        if all_reduce and torch.distributed.is_initialized():
+            torch.distributed.all_reduce(self.d_out2)
+        v = self.d_out2.cpu()
+        return float(v[0]), float(v[1])

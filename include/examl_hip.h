@@ -269,6 +269,35 @@ int examl_hip_core_root_dna_cat(long n, const double *dev_sum,
                                 double *dev_partials, double *dev_out2,
                                 void *stream);
 
+/* CAT executors (the newviewIterative/evaluateIterative/makenewzIterative
+ * CAT bodies; dev_pbuf >= numOps*numCats*32 doubles, dev_diag >= numCats*4).
+ * The optimizeRateCategories/evaluatePartialGeneric host loop is round-2. */
+int examl_hip_newview_traversal_dna_cat(
+    const examl_hip_trav_entry *ops, int numOps, const double *EIGN,
+    const double *EI, const double *perSiteRates, int numCats,
+    const double *dev_EV, const double *dev_tipVector, const int *dev_cptr,
+    double *dev_clv, long clvStrideDoubles, const unsigned char *dev_tips,
+    long tipStrideBytes, const int *dev_wgt, long n,
+    unsigned int *dev_scalers, unsigned int *dev_inc, double *dev_pbuf,
+    void *stream);
+
+int examl_hip_evaluate_root_dna_cat_x(
+    int rootTipCase, int pNumber, int qNumber, int x1Slot, int x2Slot,
+    int tipSlot, double z, const double *EIGN, const double *perSiteRates,
+    int numCats, const double *dev_tipVector, const int *dev_cptr,
+    double *dev_clv, long clvStrideDoubles, const unsigned char *dev_tips,
+    long tipStrideBytes, const int *dev_wgt, long n,
+    const unsigned int *dev_scalers, double *dev_diag_scratch,
+    double *dev_partials, double *dev_lnl, void *stream);
+
+int examl_hip_sum_root_dna_cat(int rootTipCase, int x1Slot, int x2Slot,
+                               int tipSlot, int tipSlot2,
+                               const double *dev_tipVector, double *dev_clv,
+                               long clvStrideDoubles,
+                               const unsigned char *dev_tips,
+                               long tipStrideBytes, double *dev_sum, long n,
+                               void *stream);
+
 /* ---------------------------------------------------------------------------
  * Protein (20-state) GTRGAMMA surface — span 80, tip codes 1..22.  Each
  * function replaces the 20-state counterpart of the DNA one above:

@@ -262,3 +262,59 @@ def oracle_makenewz(entries, root, tree, model, tips, wgt, z0, maxiter=64):
         else:
             outer_converged = True
     return z
+
+
+def oracle_cat_full_lnl(entries, root, tree, model, tips, wgt, cptr,
+                        per_site_rates, return_state=False):
+    """CAT replay of a product traversal on the oracle kernels (the
+    checker for DnaCatEngine)."""
+    EIGN, EV, EI, tipVector, _ = _model_arrays(model)
+    rptr = O.aligned(len(per_site_rates))
+    rptr[:] = per_site_rates
+    num_cats = len(per_site_rates)
+    width = tips.shape[1]
+    ntips = tips.shape[0] - 1
+    cptr = np.ascontiguousarray(cptr, dtype=np.int32)
+    wgt = np.ascontiguousarray(wgt, dtype=np.int32)
+    clv = {}
+    scalers = np.zeros(2 * ntips, dtype=np.int64)
+    for e in entries:
+        qz = math.log(e.qz) if e.qz > O.ZMIN else math.log(O.ZMIN)
+        rz = math.log(e.rz) if e.rz > O.ZMIN else math.log(O.ZMIN)
+        left, right = O.make_p(qz, rz, rptr, EI, EIGN, num_cats, 4)
+        if e.tipCase == TIP_TIP:
+            x3, inc = O.newview_dna_cat(
+                TIP_TIP, EV, cptr, None, None, tipVector,
+                np.ascontiguousarray(tips[e.x1Slot]),
+                np.ascontiguousarray(tips[e.x2Slot]), width, left, right,
+                wgt)
+        elif e.tipCase == TIP_INNER:
+            x3, inc = O.newview_dna_cat(
+                TIP_INNER, EV, cptr, None, clv[e.x2Slot], tipVector,
+                np.ascontiguousarray(tips[e.x1Slot]), None, width, left,
+                right, wgt)
+        else:
+            x3, inc = O.newview_dna_cat(
+                INNER_INNER, EV, cptr, clv[e.x1Slot], clv[e.x2Slot],
+                tipVector, None, None, width, left, right, wgt)
+        clv[e.x3Slot] = x3
+        scalers[e.pNumber] = scalers[e.qNumber] + scalers[e.rNumber] + inc
+    p, q, z = root
+    diag = O.calc_diagptable(z, 4, num_cats, rptr, EIGN)
+    p_tip, q_tip = tree.is_tip(p), tree.is_tip(q)
+    if q_tip:
+        lnl = O.evaluate_dna_cat(cptr, wgt, None, clv[tree.clv_slot(p)],
+                                 tipVector, np.ascontiguousarray(tips[q]),
+                                 width, diag)
+    elif p_tip:
+        lnl = O.evaluate_dna_cat(cptr, wgt, None, clv[tree.clv_slot(q)],
+                                 tipVector, np.ascontiguousarray(tips[p]),
+                                 width, diag)
+    else:
+        lnl = O.evaluate_dna_cat(cptr, wgt, clv[tree.clv_slot(p)],
+                                 clv[tree.clv_slot(q)], tipVector, None,
+                                 width, diag)
+    lnl += float(scalers[p] + scalers[q]) * math.log(O.MINLIKELIHOOD)
+    if return_state:
+        return lnl, clv, scalers
+    return lnl
