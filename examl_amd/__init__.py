@@ -108,6 +108,9 @@ def _bind(lib):
         [i, i, i, i, i, i, d, p, p, i, p, p, p, l, p, l, p, l, p, p, p, p, p]
     lib.examl_hip_sum_root_dna_cat.argtypes = \
         [i, i, i, i, i, p, p, l, p, l, p, l, p]
+    lib.examl_host_evaluate_partial_dna_cat.restype = ctypes.c_double
+    lib.examl_host_evaluate_partial_dna_cat.argtypes = \
+        [p, i, i, i, d, l, d, i, p, p, p, p, p, l, i]
     lib.examl_hip_use_graphs.argtypes = [i]
     lib.examl_hip_graphs_clear.argtypes = []
     lib.examl_hip_fast_math.argtypes = [i]

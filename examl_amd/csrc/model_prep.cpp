@@ -394,3 +394,104 @@ extern "C" void examl_host_init_gtr_aa(const double *frequencies,
   init_gtr_generic(20, vv, 23, frequencies, rates190, EIGN, EV, EI,
                    tipVector);
 }
+
+/* ---------------------------------------------------------------------------
+ * evaluatePartialGeneric for DNA CAT — the single-site re-evaluation at an
+ * arbitrary rate used by optimizeRateCategories
+ * (examl/evaluatePartialGenericSpecial.c:259/924/998, restated
+ * index-for-index, including the reference's own quirks: the overwritten
+ * zmin clamp on the root branch and the per-node (not weight-scaled)
+ * rescale counter).  Pure host math: the reference runs this on the CPU
+ * too (one site, the whole tree).
+ *
+ * ops: the post-order entries of the LAST full traversal (root entry
+ * excluded, as tr->td[0].ti[1..]); root = (rootTip, rootQ, root_z).
+ * tips: host yVector block [node row][site], row index = tip node id.
+ * ------------------------------------------------------------------------ */
+extern "C" double examl_host_evaluate_partial_dna_cat(
+    const void *ops_, int numOps, int rootTipNumber, int rootQNumber,
+    double root_z, long site, double ki, int w, const double *EIGN,
+    const double *EI, const double *EV, const double *tipVector,
+    const unsigned char *tips, long tipStride, int mxtips) {
+  /* examl_hip_trav_entry layout without including the HIP header here */
+  struct TE {
+    int tipCase, pNumber, qNumber, rNumber;
+    int x1Slot, x2Slot, x3Slot;
+    double qz, rz;
+  };
+  const TE *ops = (const TE *)ops_;
+  const double ZMIN_ = 1.0E-15;
+  const double TWO256 =
+      115792089237316195423570985008687907853269984665640564039457584007913129639936.0;
+  const double MINLIK = 1.0 / TWO256;
+
+  double *lVector = (double *)malloc(sizeof(double) * 4 * (size_t)mxtips);
+  int scale = 0;
+
+  for (int k = 0; k < numOps; k++) {
+    const TE *t = &ops[k];
+    double qz = t->qz, rz = t->rz;
+    qz = (qz > ZMIN_) ? log(qz) : log(ZMIN_);
+    rz = (rz > ZMIN_) ? log(rz) : log(ZMIN_);
+    /* computeVectorGTRCAT (evaluatePartialGenericSpecial.c:924) */
+    const double *x1, *x2;
+    double *x3 = &lVector[4 * (t->pNumber - mxtips)];
+    switch (t->tipCase) {
+    case 0: /* TIP_TIP */
+      x1 = &tipVector[4 * tips[(long)t->qNumber * tipStride + site]];
+      x2 = &tipVector[4 * tips[(long)t->rNumber * tipStride + site]];
+      break;
+    case 1: /* TIP_INNER */
+      x1 = &tipVector[4 * tips[(long)t->qNumber * tipStride + site]];
+      x2 = &lVector[4 * (t->rNumber - mxtips)];
+      break;
+    default: /* INNER_INNER */
+      x1 = &lVector[4 * (t->qNumber - mxtips)];
+      x2 = &lVector[4 * (t->rNumber - mxtips)];
+    }
+    const double lz1 = qz * ki, lz2 = rz * ki;
+    double d1[3], d2[3], x1px2[4];
+    for (int j = 0; j < 3; j++) {
+      d1[j] = x1[j + 1] * exp(EIGN[j + 1] * lz1);
+      d2[j] = x2[j + 1] * exp(EIGN[j + 1] * lz2);
+    }
+    for (int j = 0; j < 4; j++) {
+      double u1 = x1[0], u2 = x2[0];
+      for (int kk = 0; kk < 3; kk++) {
+        u1 += d1[kk] * EI[j * 4 + kk + 1];
+        u2 += d2[kk] * EI[j * 4 + kk + 1];
+      }
+      x1px2[j] = u1 * u2;
+    }
+    for (int j = 0; j < 4; j++) x3[j] = 0.0;
+    for (int j = 0; j < 4; j++)
+      for (int kk = 0; kk < 4; kk++) x3[kk] += x1px2[j] * EV[4 * j + kk];
+    if (x3[0] < MINLIK && x3[0] > -MINLIK && x3[1] < MINLIK &&
+        x3[1] > -MINLIK && x3[2] < MINLIK && x3[2] > -MINLIK &&
+        x3[3] < MINLIK && x3[3] > -MINLIK) {
+      x3[0] *= TWO256;
+      x3[1] *= TWO256;
+      x3[2] *= TWO256;
+      x3[3] *= TWO256;
+      scale++;
+    }
+  }
+
+  /* evaluatePartialGTRCAT tail (:998); NOTE the reference's dead zmin
+   * clamp — lz is log(qz) regardless — restated as-is */
+  const double *x1 = &tipVector[4 * tips[(long)rootTipNumber * tipStride +
+                                         site]];
+  const double *x2 = &lVector[4 * (rootQNumber - mxtips)];
+  double lz = log(root_z);
+  lz *= ki;
+  const double d0 = exp(EIGN[1] * lz), dd1 = exp(EIGN[2] * lz),
+               dd2 = exp(EIGN[3] * lz);
+  double term = x1[0] * x2[0];
+  term += x1[1] * x2[1] * d0;
+  term += x1[2] * x2[2] * dd1;
+  term += x1[3] * x2[3] * dd2;
+  term = log(fabs(term)) + (scale * log(MINLIK));
+  term = term * w;
+  free(lVector);
+  return term;
+}

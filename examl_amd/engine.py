@@ -47,6 +47,11 @@ class DnaGammaEngine:
             raise RuntimeError("examl_amd: CUDA/HIP device not available")
         n_ops = max_ops or (self.ninner + 8)
 
+        # host copies kept for evaluatePartialGeneric (CAT per-site rate
+        # search, evaluatePartialGenericSpecial.c:259) and updatePerSiteRates
+        self.host_tips = np.ascontiguousarray(tips)
+        self.host_wgt = np.ascontiguousarray(wgt, dtype=np.int32)
+
         dev = self.device
         self.d_tips = torch.from_numpy(np.ascontiguousarray(tips)).to(dev)
         self.d_wgt = torch.from_numpy(
@@ -267,15 +272,30 @@ class DnaCatEngine(DnaGammaEngine):
         self.num_cats = len(self.per_site_rates)
         dev = self.device
         self.d_cptr = torch.from_numpy(self.cptr).to(dev)
-        # re-size CAT-specific buffers (base class sized them for GAMMA)
+        # re-size CAT-specific buffers (base class sized them for GAMMA);
+        # scratch is sized for maxCategories=25 up front so
+        # optimizeRateCategories can grow numberOfCategories without
+        # reallocating (reference allocates for tr->maxCategories,
+        # axml.c:1936)
+        MAXC = 25
         self.d_clv = torch.empty((self.ninner, self.width * 4),
                                  dtype=torch.float64, device=dev)
-        self.d_pbuf = torch.empty(self._max_ops * self.num_cats * 32,
+        self.d_pbuf = torch.empty(self._max_ops * MAXC * 32,
                                   dtype=torch.float64, device=dev)
-        self.d_diag = torch.empty(self.num_cats * 4, dtype=torch.float64,
+        self.d_diag = torch.empty(MAXC * 4, dtype=torch.float64,
                                   device=dev)
-        self.d_dtab = torch.empty(self.num_cats * 4 + 8 + self.num_cats,
-                                  dtype=torch.float64, device=dev)
+        self.d_dtab = torch.empty(MAXC * 4 + 8 + MAXC, dtype=torch.float64,
+                                  device=dev)
+
+    def set_site_rates(self, cptr, per_site_rates):
+        """Install a new rate categorization (the device-side half of
+        optimizeRateCategories' writeback, optimizeModel.c:2465-2470)."""
+        assert len(per_site_rates) <= 25
+        self.cptr = np.ascontiguousarray(cptr, dtype=np.int32)
+        self.per_site_rates = np.ascontiguousarray(per_site_rates,
+                                                   dtype=np.float64)
+        self.num_cats = len(self.per_site_rates)
+        self.d_cptr.copy_(torch.from_numpy(self.cptr))
 
     def newview_traversal(self, entries):
         if not entries:

@@ -44,7 +44,8 @@ class TreeSearch:
     numBranches=1)."""
 
     def __init__(self, tree, engines, opt_freq_flags=None,
-                 auto_flags=None, empirical_freqs=None):
+                 auto_flags=None, empirical_freqs=None, rate_het="GAMMA",
+                 max_categories=25):
         self.tree = tree
         self.engines = engines
         self.M = len(engines)
@@ -63,6 +64,14 @@ class TreeSearch:
         self.prot_freqs = [1] * self.M  # AUTO: 1 = fixed/model freqs
         # freqExponents state (models.c:4227: init 0.0)
         self.freq_exponents = [np.zeros(e.model.states) for e in engines]
+        # CAT (PSR) state (models.c:4194-4201: one category, all rates 1)
+        self.rate_het = rate_het
+        self.max_categories = max_categories
+        self.rate_cat_invocations = 1  # optimizeRateCategoryInvocations
+        if rate_het == "CAT":
+            self.cat_patrat = [np.ones(e.width) for e in engines]
+            self.cat_lhs = [np.zeros(e.width) for e in engines]
+        self._last_full = None
 
     # ------------------------------------------------------------------
     # traversal construction (computeTraversalInfo,
@@ -136,6 +145,10 @@ class TreeSearch:
         self._collect(q, p, not full, out)
         self._run(out)
         z = t.get_z(p, q)
+        if full:
+            # td[0] of the last full traversal: evaluatePartialGeneric walks
+            # exactly these entries (evaluatePartialGenericSpecial.c:259)
+            self._last_full = (out, p, q, z)
         # launch all partitions, then ONE host sync for the readbacks
         outs = []
         for m, eng in enumerate(self.engines):
@@ -713,9 +726,158 @@ class TreeSearch:
             self.evaluate_generic(full=True)
         assert self.likelihood >= start_lh - 1e-6
 
+    # -- CAT per-site rate optimization (optimizeModel.c:2403) -------------
+
+    def _evaluate_partial(self, m, site, ki):
+        """evaluatePartialGeneric (evaluatePartialGenericSpecial.c:259) via
+        the host C implementation over the last full traversal."""
+        import ctypes
+        from . import lib
+        eng = self.engines[m]
+        entries, p, q, z = self._last_full
+        if not hasattr(self, "_epg_ops") or self._epg_ops[1] is not entries:
+            arr = (TravEntry * len(entries))(*entries)
+            self._epg_ops = (arr, entries)
+        arr = self._epg_ops[0]
+        return lib().examl_host_evaluate_partial_dna_cat(
+            ctypes.cast(arr, ctypes.c_void_p), len(entries),
+            ctypes.c_int(p), ctypes.c_int(q), ctypes.c_double(z),
+            ctypes.c_long(site), ctypes.c_double(ki),
+            ctypes.c_int(int(eng.host_wgt[site])),
+            eng.model.EIGN.ctypes.data_as(ctypes.c_void_p),
+            eng.model.EI.ctypes.data_as(ctypes.c_void_p),
+            eng.model.EV.ctypes.data_as(ctypes.c_void_p),
+            eng.model.tipVector.ctypes.data_as(ctypes.c_void_p),
+            eng.host_tips.ctypes.data_as(ctypes.c_void_p),
+            ctypes.c_long(eng.width), ctypes.c_int(self.tree.ntips))
+
+    def optimize_rate_categories(self, max_categories, log=None):
+        """optimizeRateCategories (optimizeModel.c:2403): per-site rate
+        search (optRateCatPthreads :1798), clustering
+        (categorizeTheRates :2171), mean-1 rescale (updatePerSiteRates
+        :2060) and restore-on-regression."""
+        if max_categories == 1:
+            return
+        initial_lh = self.likelihood
+        self.evaluate_generic(full=True)
+        inv = self.rate_cat_invocations
+        if inv == 1:
+            lower_spacing, upper_spacing = 0.5 / inv, 1.0 / inv
+        else:
+            lower_spacing, upper_spacing = 0.05 / inv, 0.1 / inv
+        lower_spacing = max(lower_spacing, 0.001)
+        upper_spacing = max(upper_spacing, 0.001)
+        self.rate_cat_invocations += 1
+
+        backup = [(self.cat_patrat[m].copy(),
+                   self.engines[m].per_site_rates.copy(),
+                   self.engines[m].cptr.copy(),
+                   self.engines[m].num_cats) for m in range(self.M)]
+
+        eps = 0.00001
+        for m, eng in enumerate(self.engines):
+            patrat = self.cat_patrat[m]
+            lhs = self.cat_lhs[m]
+            for i in range(eng.width):
+                r0 = patrat[i]
+                l0 = self._evaluate_partial(m, i, r0)
+                left_lh = right_lh = l0
+                left_rate = right_rate = r0
+                k = 1
+                while True:
+                    if not (r0 - k * lower_spacing > 0.0001):
+                        break
+                    v = self._evaluate_partial(m, i, r0 - k * lower_spacing)
+                    if not (v > left_lh and abs(left_lh - v) > eps):
+                        break
+                    left_lh = v
+                    left_rate = r0 - k * lower_spacing
+                    k += 1
+                k = 1
+                while True:
+                    v = self._evaluate_partial(m, i, r0 + k * upper_spacing)
+                    if not (v > right_lh and abs(right_lh - v) > eps):
+                        break
+                    right_lh = v
+                    right_rate = r0 + k * upper_spacing
+                    k += 1
+                if right_lh > l0 or left_lh > l0:
+                    if right_lh > left_lh:
+                        patrat[i] = right_rate
+                        lhs[i] = right_lh
+                    else:
+                        patrat[i] = left_rate
+                        lhs[i] = left_lh
+                else:
+                    lhs[i] = l0
+
+        # categorizeTheRates (:2171) per partition
+        for m, eng in enumerate(self.engines):
+            patrat = self.cat_patrat[m]
+            lhs = self.cat_lhs[m]
+            rates = [patrat[0]]
+            acc = [lhs[0]]
+            for i in range(1, eng.width):
+                temp = patrat[i]
+                found = False
+                for k in range(len(rates)):
+                    if temp == rates[k] or abs(temp - rates[k]) < 0.001:
+                        acc[k] += lhs[i]
+                        found = True
+                        break
+                if not found:
+                    rates.append(temp)
+                    acc.append(lhs[i])
+            order = sorted(range(len(rates)), key=lambda k: acc[k])
+            rc_rates = [rates[k] for k in order]
+            num = min(len(rc_rates), max_categories)
+            # categorizePartition (:1733): exact/0.001 match among the kept
+            # categories, else nearest
+            new_cptr = np.zeros(eng.width, dtype=np.int32)
+            for i in range(eng.width):
+                temp = patrat[i]
+                found = False
+                for k in range(num):
+                    if temp == rc_rates[k] or abs(temp - rc_rates[k]) < 0.001:
+                        new_cptr[i] = k
+                        found = True
+                        break
+                if not found:
+                    best, bmin = 0, abs(temp - rc_rates[0])
+                    for k in range(1, num):
+                        d = abs(temp - rc_rates[k])
+                        if d < bmin:
+                            bmin, best = d, k
+                    new_cptr[i] = best
+            eng.set_site_rates(new_cptr, np.array(rc_rates[:num]))
+
+        # updatePerSiteRates (:2060), numBranches=1: global mean-1 rescale
+        wsum = rsum = 0.0
+        for m, eng in enumerate(self.engines):
+            w = eng.host_wgt
+            rsum += float((w * eng.per_site_rates[eng.cptr]).sum())
+            wsum += float(w.sum())
+        scaler = 1.0 / (rsum / wsum)
+        for eng in self.engines:
+            eng.set_site_rates(eng.cptr, eng.per_site_rates * scaler)
+
+        self.evaluate_generic(full=True)
+        if self.likelihood < initial_lh:
+            for m, eng in enumerate(self.engines):
+                pat, rates_b, cptr_b, _ = backup[m]
+                self.cat_patrat[m][:] = pat
+                eng.set_site_rates(cptr_b, rates_b)
+            self.evaluate_generic(full=True)
+            assert abs(self.likelihood - initial_lh) < 1e-6
+        if log:
+            log(f"rate categories: lnL {initial_lh:.6f} -> "
+                f"{self.likelihood:.6f} "
+                f"({[e.num_cats for e in self.engines]})")
+
     def mod_opt(self, likelihood_epsilon=0.1, model_epsilon=0.0001,
                 log=None):
-        """modOpt (optimizeModel.c:2963) for the GAMMA model."""
+        """modOpt (optimizeModel.c:2963) for the GAMMA and CAT models."""
+        cat_opt = 0
         self.evaluate_generic(full=True)
         while True:
             current = self.likelihood
@@ -727,9 +889,18 @@ class TreeSearch:
             self.opt_base_freqs(model_epsilon)
             self.evaluate_generic(full=True)
             self.tree_evaluate(0.0625)
-            self.opt_alphas_generic(model_epsilon)
-            self.evaluate_generic(full=True)
-            self.tree_evaluate(0.1)
+            if self.rate_het == "GAMMA":
+                self.opt_alphas_generic(model_epsilon)
+                self.evaluate_generic(full=True)
+                self.tree_evaluate(0.1)
+            elif self.rate_het == "CAT":
+                if cat_opt < 3:
+                    self.evaluate_generic(full=True)
+                    self.optimize_rate_categories(self.max_categories,
+                                                  log=log)
+                    cat_opt += 1
+            else:
+                raise AssertionError(self.rate_het)
             if log:
                 log(f"modOpt pass: {current:.6f} -> {self.likelihood:.6f}")
             if abs(current - self.likelihood) <= likelihood_epsilon:

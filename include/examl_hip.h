@@ -298,6 +298,16 @@ int examl_hip_sum_root_dna_cat(int rootTipCase, int x1Slot, int x2Slot,
                                long tipStrideBytes, double *dev_sum, long n,
                                void *stream);
 
+/* replaces evaluatePartialGeneric for DNA CAT
+ * (examl/evaluatePartialGenericSpecial.c:259): weighted single-site lnL at
+ * an arbitrary rate ki over the LAST full traversal (ops excludes the root
+ * entry).  Pure host math, as in the reference. */
+double examl_host_evaluate_partial_dna_cat(
+    const void *ops, int numOps, int rootTipNumber, int rootQNumber,
+    double root_z, long site, double ki, int w, const double *EIGN,
+    const double *EI, const double *EV, const double *tipVector,
+    const unsigned char *tips, long tipStrideBytes, int mxtips);
+
 /* ---------------------------------------------------------------------------
  * Protein (20-state) GTRGAMMA surface — span 80, tip codes 1..22.  Each
  * function replaces the 20-state counterpart of the DNA one above:

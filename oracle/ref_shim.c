@@ -52,7 +52,10 @@ int getUndetermined(int dataType) {
  * evaluateGTRGAMMA[PROT], sumGAMMA[PROT], coreGTRGAMMA[PROT], makeP,
  * calcDiagptable, initGeneric, makeGammaCats). */
 void getxnode(nodeptr p) { (void)p; abort(); }
-int isTip(int number, int maxTips) { (void)number; (void)maxTips; abort(); }
+/* axml.c isTip — faithful (needed by evaluatePartialGTRCAT's assert) */
+int isTip(int number, int maxTips) {
+  return number > 0 && number <= maxTips;
+}
 void checkPerSiteRates(const tree *const tr) { (void)tr; abort(); }
 void storeExecuteMaskInTraversalDescriptor(tree *tr) { (void)tr; abort(); }
 void storeValuesInTraversalDescriptor(tree *tr, double *v) {

@@ -109,6 +109,8 @@ class OracleEngine:
         self.width = tips.shape[1]
         self.tips = tips
         self.wgt = np.ascontiguousarray(wgt, dtype=np.int32)
+        self.host_tips = np.ascontiguousarray(tips)
+        self.host_wgt = self.wgt
         self.clv = {}
         self.scalers = np.zeros(2 * self.ntips, dtype=np.int64)
         self.sumtable = None
@@ -201,6 +203,106 @@ class OracleEngine:
 
     def core_derivs_async(self, lz):
         return self.core_derivs(lz)
+
+
+class OracleCatEngine(OracleEngine):
+    """CPU CAT (PSR) engine with the DnaCatEngine interface — lets
+    TreeSearch(rate_het="CAT") run fully on the oracle kernels (the checker
+    for the GPU PSR path and the vehicle for the 49 -f E -m PSR anchor)."""
+
+    def __init__(self, tips, wgt, model, cptr, per_site_rates):
+        super().__init__(tips, wgt, model)
+        self.cptr = np.ascontiguousarray(cptr, dtype=np.int32)
+        self.per_site_rates = np.ascontiguousarray(per_site_rates,
+                                                   dtype=np.float64)
+        self.num_cats = len(self.per_site_rates)
+
+    def set_site_rates(self, cptr, per_site_rates):
+        self.cptr = np.ascontiguousarray(cptr, dtype=np.int32)
+        self.per_site_rates = np.ascontiguousarray(per_site_rates,
+                                                   dtype=np.float64)
+        self.num_cats = len(self.per_site_rates)
+
+    def _rptr(self):
+        r = O.aligned(self.num_cats)
+        r[:] = self.per_site_rates
+        return r
+
+    def newview_traversal(self, entries):
+        EIGN, EV, EI, tipVector, _ = self._arrays()
+        rptr = self._rptr()
+        for e in entries:
+            qz = math.log(e.qz) if e.qz > O.ZMIN else math.log(O.ZMIN)
+            rz = math.log(e.rz) if e.rz > O.ZMIN else math.log(O.ZMIN)
+            left, right = O.make_p(qz, rz, rptr, EI, EIGN, self.num_cats, 4)
+            if e.tipCase == TIP_TIP:
+                x3, inc = O.newview_dna_cat(
+                    TIP_TIP, EV, self.cptr, None, None, tipVector,
+                    np.ascontiguousarray(self.tips[e.x1Slot]),
+                    np.ascontiguousarray(self.tips[e.x2Slot]), self.width,
+                    left, right, self.wgt)
+            elif e.tipCase == TIP_INNER:
+                x3, inc = O.newview_dna_cat(
+                    TIP_INNER, EV, self.cptr, None, self.clv[e.x2Slot],
+                    tipVector, np.ascontiguousarray(self.tips[e.x1Slot]),
+                    None, self.width, left, right, self.wgt)
+            else:
+                x3, inc = O.newview_dna_cat(
+                    INNER_INNER, EV, self.cptr, self.clv[e.x1Slot],
+                    self.clv[e.x2Slot], tipVector, None, None, self.width,
+                    left, right, self.wgt)
+            self.clv[e.x3Slot] = x3
+            self.scalers[e.pNumber] = (self.scalers[e.qNumber] +
+                                       self.scalers[e.rNumber] + inc)
+
+    def evaluate_root(self, tree, p, q, z):
+        EIGN, EV, EI, tipVector, _ = self._arrays()
+        diag = O.calc_diagptable(z, 4, self.num_cats, self._rptr(), EIGN)
+        p_tip, q_tip = tree.is_tip(p), tree.is_tip(q)
+        if q_tip:
+            lnl = O.evaluate_dna_cat(
+                self.cptr, self.wgt, None, self.clv[tree.clv_slot(p)],
+                tipVector, np.ascontiguousarray(self.tips[q]), self.width,
+                diag)
+        elif p_tip:
+            lnl = O.evaluate_dna_cat(
+                self.cptr, self.wgt, None, self.clv[tree.clv_slot(q)],
+                tipVector, np.ascontiguousarray(self.tips[p]), self.width,
+                diag)
+        else:
+            lnl = O.evaluate_dna_cat(
+                self.cptr, self.wgt, self.clv[tree.clv_slot(p)],
+                self.clv[tree.clv_slot(q)], tipVector, None, self.width,
+                diag)
+        lnl += float(self.scalers[p] + self.scalers[q]) *             math.log(O.MINLIKELIHOOD)
+        return lnl
+
+    def sum_root(self, tree, p, q):
+        EIGN, EV, EI, tipVector, _ = self._arrays()
+        p_tip, q_tip = tree.is_tip(p), tree.is_tip(q)
+        if p_tip and q_tip:
+            self.sumtable = O.sum_dna_cat(
+                TIP_TIP, None, None, tipVector,
+                np.ascontiguousarray(self.tips[p]),
+                np.ascontiguousarray(self.tips[q]), self.width)
+        elif q_tip:
+            self.sumtable = O.sum_dna_cat(
+                TIP_INNER, None, self.clv[tree.clv_slot(p)], tipVector,
+                np.ascontiguousarray(self.tips[q]), None, self.width)
+        elif p_tip:
+            self.sumtable = O.sum_dna_cat(
+                TIP_INNER, None, self.clv[tree.clv_slot(q)], tipVector,
+                np.ascontiguousarray(self.tips[p]), None, self.width)
+        else:
+            self.sumtable = O.sum_dna_cat(
+                INNER_INNER, self.clv[tree.clv_slot(p)],
+                self.clv[tree.clv_slot(q)], tipVector, None, None,
+                self.width)
+
+    def core_derivs(self, lz):
+        EIGN, EV, EI, tipVector, _ = self._arrays()
+        return O.core_dna_cat(self.width, self.num_cats, self.sumtable,
+                              self.wgt, self._rptr(), EIGN, self.cptr, lz)
 
 
 def oracle_makenewz(entries, root, tree, model, tips, wgt, z0, maxiter=64):

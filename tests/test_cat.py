@@ -253,3 +253,115 @@ def test_gpu_cat_full_pipeline_vs_oracle(model):
                            float(np.log(z0)))
     assert np.isclose(d_gpu[0], d_ref[0], rtol=1e-10)
     assert np.isclose(d_gpu[1], d_ref[1], rtol=1e-10)
+
+
+# ---------------------------------------------------------------------------
+# evaluatePartialGeneric (the CAT per-site rate probe) — product host fn vs
+# the reference's own evaluatePartialGTRCAT
+# ---------------------------------------------------------------------------
+
+@pytest.mark.skipif(not O.have_ref(), reason="reference libref.so not built")
+def test_evaluate_partial_bit_exact_vs_reference():
+    """examl_host_evaluate_partial_dna_cat (model_prep.cpp, restating
+    evaluatePartialGenericSpecial.c:50-257) is BIT-EXACT against the
+    reference's evaluatePartialGTRCAT over a 12-taxon traversal, 5 sites x
+    3 probe rates."""
+    from examl_amd.search import TreeSearch
+    from tests.helpers import make_synthetic, OracleEngine, _model_arrays
+
+    NUMB = 256  # the reference's NUM_BRANCHES (axml.h:126)
+
+    class RefTI(ctypes.Structure):
+        _fields_ = [("tipCase", ctypes.c_int), ("pNumber", ctypes.c_int),
+                    ("qNumber", ctypes.c_int), ("rNumber", ctypes.c_int),
+                    ("qz", ctypes.c_double * NUMB),
+                    ("rz", ctypes.c_double * NUMB)]
+
+    ntips, width = 12, 200
+    tips, wgt = make_synthetic(ntips, width, seed=77)
+    model = ea.DnaGtrModel([0.3, 0.2, 0.26, 0.24],
+                           [1.2, 2.4, 0.7, 0.9, 3.1, 1.0], 1.0)
+    tree = ea.PhyloTree.random(ntips, seed=4, rng_z=True)
+    ts = TreeSearch(tree, [OracleEngine(tips, wgt, model)])
+    ts.evaluate_generic(full=True)
+    p = 1
+    q = next(iter(tree.adj[1]))
+    entries = []
+    ts.oriented.clear()
+    ts._collect(p, q, False, entries)
+    ts._collect(q, p, False, entries)
+    root_z = tree.get_z(p, q)
+
+    # reference td[0]: ti[0] = root record {pNumber, qNumber, qz[0]}
+    # (evaluatePartialGenericSpecial.c:264-270), ti[1..] = newview ops
+    n_ti = len(entries) + 1
+    arr = (RefTI * n_ti)()
+    arr[0].tipCase = 0
+    arr[0].pNumber = p
+    arr[0].qNumber = q
+    arr[0].qz[0] = root_z
+    for k, e in enumerate(entries):
+        arr[k + 1].tipCase = e.tipCase
+        arr[k + 1].pNumber = e.pNumber
+        arr[k + 1].qNumber = e.qNumber
+        arr[k + 1].rNumber = e.rNumber
+        arr[k + 1].qz[0] = e.qz
+        arr[k + 1].rz[0] = e.rz
+    rows = (ctypes.POINTER(ctypes.c_ubyte) * (ntips + 1))()
+    tipsC = np.ascontiguousarray(tips)
+    for t in range(1, ntips + 1):
+        rows[t] = (ctypes.c_ubyte * width).from_buffer(tipsC[t])
+    EIGN, EV, EI, tipVector, _ = _model_arrays(model)
+    ref = O._ref
+    ref.evaluatePartialGTRCAT.restype = ctypes.c_double
+    L = ea.lib()
+
+    def dp(a):
+        return a.ctypes.data_as(ctypes.POINTER(ctypes.c_double))
+
+    ops_arr = (ea.TravEntry * len(entries))(*entries)
+    for site in [0, 3, 17, 100, 199]:
+        for ki in [0.2, 1.0, 2.7]:
+            r = ref.evaluatePartialGTRCAT(
+                ctypes.c_int(site), ctypes.c_double(ki), ctypes.c_int(n_ti),
+                arr, ctypes.c_double(root_z), ctypes.c_int(int(wgt[site])),
+                dp(EIGN), dp(EI), dp(EV), dp(tipVector), rows,
+                ctypes.c_int(0), ctypes.c_int(ntips))
+            m = L.examl_host_evaluate_partial_dna_cat(
+                ctypes.cast(ops_arr, ctypes.c_void_p), len(entries),
+                ctypes.c_int(p), ctypes.c_int(q), ctypes.c_double(root_z),
+                ctypes.c_long(site), ctypes.c_double(ki),
+                ctypes.c_int(int(wgt[site])),
+                model.EIGN.ctypes.data_as(ctypes.c_void_p),
+                model.EI.ctypes.data_as(ctypes.c_void_p),
+                model.EV.ctypes.data_as(ctypes.c_void_p),
+                model.tipVector.ctypes.data_as(ctypes.c_void_p),
+                tipsC.ctypes.data_as(ctypes.c_void_p),
+                ctypes.c_long(width), ctypes.c_int(ntips))
+            assert r == m, (site, ki, r, m)
+
+
+def test_cat_optimizer_small_cpu():
+    """optimizeRateCategories + the CAT modOpt branch on the oracle engines:
+    lnL monotone through categorization, numberOfCategories grows, and the
+    mean per-site rate is 1 after updatePerSiteRates."""
+    from examl_amd.search import TreeSearch
+    from tests.helpers import make_synthetic, OracleCatEngine
+
+    tips, wgt = make_synthetic(ntips=12, width=120, seed=7)
+    tree = ea.PhyloTree.random(12, seed=3)
+    model = ea.DnaGtrModel(np.full(4, 0.25),
+                           np.array([1.1, 2.2, 0.7, 1.3, 0.9, 1.0]), 1.0)
+    eng = OracleCatEngine(tips, wgt, model, np.zeros(120, dtype=np.int32),
+                          np.array([1.0]))
+    ts = TreeSearch(tree, [eng], rate_het="CAT")
+    l0 = ts.evaluate_generic(full=True)
+    ts.optimize_rate_categories(25)
+    assert ts.likelihood >= l0 - 1e-9
+    assert eng.num_cats > 1
+    # updatePerSiteRates (optimizeModel.c:2060): weighted mean rate == 1
+    mean = float((eng.host_wgt * eng.per_site_rates[eng.cptr]).sum()
+                 / eng.host_wgt.sum())
+    assert abs(mean - 1.0) < 1e-10
+    l1 = ts.mod_opt(0.5)
+    assert l1 > l0
