@@ -337,6 +337,7 @@ class TreeSearch:
         end_state = np.zeros(n, dtype=int)
         u = np.zeros(n)
         ulim = np.zeros(n)
+        fu_state = np.zeros(n)  # the reference's persistent fu[] array
 
         np.clip(ax, lim_inf, lim_sup, out=ax)
         param[:] = ax
@@ -412,12 +413,12 @@ class TreeSearch:
                 if converged[i]:
                     continue
                 if end_state[i] == 0:
-                    fu = temp[i]
+                    fu = fu_state[i] = temp[i]
                     ax[i], bx[i], cx[i] = bx[i], cx[i], u[i]
                     fa[i], fb[i], fc[i] = fb[i], fc[i], fu
                     state[i] = 0
                 elif end_state[i] == 1:
-                    fu = temp[i]
+                    fu = fu_state[i] = temp[i]
                     if fu < fc[i]:
                         ax[i] = bx[i]
                         bx[i] = u[i]
@@ -434,20 +435,31 @@ class TreeSearch:
                         param[i] = u[i]
                         state[i] = 1
                 elif end_state[i] == 2:
-                    fu = temp[i]
+                    fu = fu_state[i] = temp[i]
                     if fu < fc[i]:
-                        bx[i], cx[i], u[i] = cx[i], u[i], \
-                            cx[i] + MNBRAK_GOLD * (cx[i] - bx[i])
-                        fb[i], fc[i] = fc[i], fu
+                        # the reference's SHFT(bx,cx,u, cx+GOLD*(cx-bx)) is
+                        # SEQUENTIAL: the new u is computed from the ALREADY
+                        # SHIFTED bx/cx (= old cx/u), no f-shifts, and param
+                        # is left at the old u — the state-2 -> endState-3
+                        # round re-evaluates that same point
+                        # (optimizeModel.c:1046-1050)
+                        bx[i] = cx[i]
+                        cx[i] = u[i]
+                        u[i] = cx[i] + MNBRAK_GOLD * (cx[i] - bx[i])
                         state[i] = 2
                     else:
                         state[i] = 0
                         ax[i], bx[i], cx[i] = bx[i], cx[i], u[i]
                         fa[i], fb[i], fc[i] = fb[i], fc[i], fu
                 elif end_state[i] == 3:
-                    fb[i], fc[i], fu = fc[i], temp[i], temp[i]
+                    # net effect of the reference's three sequential SHFTs
+                    # (optimizeModel.c:1060-1063): fa=old fc, fb=old fu
+                    # (the carried-over value, == temp here), fc=temp
+                    fa[i] = fc[i]
+                    fb[i] = fu_state[i]
+                    fc[i] = temp[i]
+                    fu_state[i] = temp[i]
                     ax[i], bx[i], cx[i] = bx[i], cx[i], u[i]
-                    fa[i], fb[i], fc[i] = fb[i], fc[i], fu
                     state[i] = 0
                 else:
                     raise AssertionError
