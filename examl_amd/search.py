@@ -45,7 +45,7 @@ class TreeSearch:
 
     def __init__(self, tree, engines, opt_freq_flags=None,
                  auto_flags=None, empirical_freqs=None, rate_het="GAMMA",
-                 max_categories=25):
+                 max_categories=25, per_gene_bl=False):
         self.tree = tree
         self.engines = engines
         self.M = len(engines)
@@ -54,8 +54,16 @@ class TreeSearch:
         self.likelihood = None
         self.oriented = {}  # inner node -> the parent neighbor its CLV faces
         self.start = 1  # tr->start = tr->nodep[1]
-        self.partition_smoothed = True
-        self.partition_converged = False
+        # -M (perGeneBranchLengths): numBranches == NumberOfModels
+        # (topLevelMakenewz's assert, makenewzGenericSpecial.c)
+        self.NB = self.M if per_gene_bl else 1
+        if per_gene_bl:
+            tree.expand_branches(self.M)
+            self.partition_smoothed = np.ones(self.NB, dtype=bool)
+            self.partition_converged = np.zeros(self.NB, dtype=bool)
+        else:
+            self.partition_smoothed = True
+            self.partition_converged = False
         self.opt_freq_flags = opt_freq_flags or [False] * self.M
         # AUTO protein model selection state (optimizeModel.c:2669)
         self.auto_flags = auto_flags or [False] * self.M
@@ -109,18 +117,38 @@ class TreeSearch:
         e.tipCase = tc
         e.pNumber, e.qNumber, e.rNumber = p, q, r
         e.qz, e.rz = t.get_z(p, q), t.get_z(p, r)
+        if self.NB > 1:
+            # per-partition branch lengths (qz[i]/rz[i] of axml.h:434)
+            e.qzv, e.rzv = t.get_zv(p, q).copy(), t.get_zv(p, r).copy()
         e.x3Slot = t.clv_slot(p)
         e.x1Slot = q if t.is_tip(q) else t.clv_slot(q)
         e.x2Slot = r if t.is_tip(r) else t.clv_slot(r)
         out.append(e)
         self.oriented[p] = parent
 
+    def _per_partition_entries(self, entries, m):
+        out = []
+        for e in entries:
+            c = TravEntry()
+            c.tipCase, c.pNumber, c.qNumber, c.rNumber = \
+                e.tipCase, e.pNumber, e.qNumber, e.rNumber
+            c.x1Slot, c.x2Slot, c.x3Slot = e.x1Slot, e.x2Slot, e.x3Slot
+            c.qz, c.rz = e.qzv[m], e.rzv[m]
+            out.append(c)
+        return out
+
     def _run(self, entries):
         if not entries:
             return
-        for m, eng in enumerate(self.engines):
-            if self.execute_model[m]:
-                eng.newview_traversal(entries)
+        if self.NB == 1:
+            for m, eng in enumerate(self.engines):
+                if self.execute_model[m]:
+                    eng.newview_traversal(entries)
+        else:
+            for m, eng in enumerate(self.engines):
+                if self.execute_model[m]:
+                    eng.newview_traversal(
+                        self._per_partition_entries(entries, m))
 
     # ------------------------------------------------------------------
     # L1 entry points over the engines
@@ -144,16 +172,19 @@ class TreeSearch:
         self._collect(p, q, not full, out)
         self._collect(q, p, not full, out)
         self._run(out)
-        z = t.get_z(p, q)
+        zv = t.get_zv(p, q)
+        z = float(zv[0])
         if full:
             # td[0] of the last full traversal: evaluatePartialGeneric walks
             # exactly these entries (evaluatePartialGenericSpecial.c:259)
-            self._last_full = (out, p, q, z)
+            self._last_full = (out, p, q, zv.copy())
         # launch all partitions, then ONE host sync for the readbacks
+        # (z = pz[m] per partition under -M, evaluateGenericSpecial.c:449)
         outs = []
         for m, eng in enumerate(self.engines):
             if self.execute_model[m]:
-                outs.append((m, eng.evaluate_root(t, p, q, z)))
+                outs.append((m, eng.evaluate_root(
+                    t, p, q, float(zv[m]) if self.NB > 1 else z)))
         if outs and not isinstance(outs[0][1], float):
             import torch
             vals = torch.cat([o[1] for o in outs]).cpu()
@@ -229,15 +260,114 @@ class TreeSearch:
     # branch-length smoothing (searchAlgo.c:127-270,2635)
     # ------------------------------------------------------------------
 
+    def makenewz_generic_vec(self, p, q, z0, maxiter, mask):
+        """topLevelMakenewz for numBranches == NumberOfModels (-M):
+        per-partition NR with curvatOK/outerConverged masks
+        (makenewzGenericSpecial.c:849-1063) and the partitionConverged
+        execute mask of makenewzGeneric(mask=TRUE) (:1369-1378)."""
+        NB = self.NB
+        out = []
+        self._collect(p, q, True, out)
+        self._collect(q, p, True, out)
+        if mask:
+            for i in range(NB):
+                self.execute_model[i] = not self.partition_converged[i]
+        z = np.array(z0, dtype=float)
+        zprev = z.copy()
+        zstep = np.zeros(NB)
+        corelz = np.zeros(NB)
+        dl = np.zeros(NB)
+        d2 = np.zeros(NB)
+        miter = np.full(NB, maxiter)
+        outer = np.zeros(NB, dtype=bool)
+        curvat = np.ones(NB, dtype=bool)
+        first = True
+        while True:
+            for i in range(NB):
+                if not outer[i] and curvat[i]:
+                    curvat[i] = False
+                    zprev[i] = z[i]
+                    zstep[i] = (1.0 - ZMAX) * z[i] + ZMIN
+            for i in range(NB):
+                if not outer[i] and not curvat[i]:
+                    z[i] = min(max(z[i], ZMIN), ZMAX)
+                    corelz[i] = math.log(z[i])
+            for m in range(NB):
+                if self.execute_model[m]:
+                    self.execute_model[m] = not curvat[m]
+            if first:
+                self._run(out)
+                for m, eng in enumerate(self.engines):
+                    if self.execute_model[m]:
+                        eng.sum_root(self.tree, p, q)
+                first = False
+            dl[:] = 0.0
+            d2[:] = 0.0
+            outs = [(m, eng.core_derivs_async(float(corelz[m])))
+                    for m, eng in enumerate(self.engines)
+                    if self.execute_model[m]]
+            if outs and not isinstance(outs[0][1], tuple):
+                import torch
+                vals = torch.stack([o[1] for o in outs]).cpu()
+                for k, (m, _) in enumerate(outs):
+                    dl[m], d2[m] = float(vals[k][0]), float(vals[k][1])
+            else:
+                for m, v in outs:
+                    dl[m], d2[m] = v
+            for i in range(NB):
+                if not outer[i] and not curvat[i]:
+                    if d2[i] >= 0.0 and z[i] < ZMAX:
+                        zprev[i] = z[i] = 0.37 * z[i] + 0.63
+                    else:
+                        curvat[i] = True
+            for i in range(NB):
+                if curvat[i] and not outer[i]:
+                    if d2[i] < 0.0:
+                        tantmp = -dl[i] / d2[i]
+                        if tantmp < 100:
+                            z[i] *= math.exp(tantmp)
+                            if z[i] < ZMIN:
+                                z[i] = ZMIN
+                            if z[i] > 0.25 * zprev[i] + 0.75:
+                                z[i] = 0.25 * zprev[i] + 0.75
+                        else:
+                            z[i] = 0.25 * zprev[i] + 0.75
+                    if z[i] > ZMAX:
+                        z[i] = ZMAX
+                    miter[i] -= 1
+                    if abs(z[i] - zprev[i]) > zstep[i]:
+                        if miter[i] < -20:
+                            z[i] = z0[i]
+                            outer[i] = True
+                        else:
+                            outer[i] = False
+                    else:
+                        outer[i] = True
+            if outer.all():
+                break
+        self.execute_model = [True] * self.M
+        return z
+
     def update(self, p, parent):
         """update(tr, p) with q = p->back = parent (searchAlgo.c:127)."""
         t = self.tree
-        z0 = t.get_z(p, parent)
-        z = self.makenewz_generic(p, parent, z0, NEWZPERCYCLE)
-        if not self.partition_converged:
-            if abs(z - z0) > DELTAZ:
-                self.partition_smoothed = False
-            t.set_z(p, parent, z)
+        if self.NB == 1:
+            z0 = t.get_z(p, parent)
+            z = self.makenewz_generic(p, parent, z0, NEWZPERCYCLE)
+            if not self.partition_converged:
+                if abs(z - z0) > DELTAZ:
+                    self.partition_smoothed = False
+                t.set_z(p, parent, z)
+        else:
+            z0 = t.get_zv(p, parent).copy()
+            z = self.makenewz_generic_vec(p, parent, z0, NEWZPERCYCLE,
+                                          mask=True)
+            zarr = t.adj[p][parent]  # aliased both directions
+            for i in range(self.NB):
+                if not self.partition_converged[i]:
+                    if abs(z[i] - z0[i]) > DELTAZ:
+                        self.partition_smoothed[i] = False
+                    zarr[i] = z[i]
 
     def smooth(self, p, parent):
         """smooth(tr, p) with p->back == parent (searchAlgo.c:196)."""
@@ -246,22 +376,46 @@ class TreeSearch:
         if not t.is_tip(p):
             for w in self._children(p, parent):
                 self.smooth(w, p)
-            self.newview_generic(p, parent)
+            if self.NB > 1:
+                # masked newview (newviewGenericSpecial.c:1559-1573)
+                self.execute_model = [not c for c in self.partition_converged]
+                self.newview_generic(p, parent)
+                self.execute_model = [True] * self.M
+            else:
+                self.newview_generic(p, parent)
 
     def smooth_tree(self, maxtimes):
         """smoothTree (searchAlgo.c:237)."""
         t = self.tree
         p = self.start
-        self.partition_converged = False
-        while maxtimes > 0:
-            maxtimes -= 1
-            self.partition_smoothed = True
-            self.smooth(next(iter(t.adj[p])), p)  # smooth(tr, p->back)
-            # p == tr->start is a tip: the second descent is skipped
-            if self.partition_smoothed:  # allSmoothed
-                self.partition_converged = True
-                break
-        self.partition_converged = False
+        if self.NB == 1:
+            self.partition_converged = False
+            while maxtimes > 0:
+                maxtimes -= 1
+                self.partition_smoothed = True
+                self.smooth(next(iter(t.adj[p])), p)  # smooth(tr, p->back)
+                # p == tr->start is a tip: the second descent is skipped
+                if self.partition_smoothed:  # allSmoothed
+                    self.partition_converged = True
+                    break
+            self.partition_converged = False
+        else:
+            self.partition_converged[:] = False
+            while maxtimes > 0:
+                maxtimes -= 1
+                self.partition_smoothed[:] = True
+                self.smooth(next(iter(t.adj[p])), p)
+                # allSmoothed (searchAlgo.c:222): flags converged partitions
+                # even when the sweep as a whole is not yet smoothed
+                result = True
+                for i in range(self.NB):
+                    if not self.partition_smoothed[i]:
+                        result = False
+                    else:
+                        self.partition_converged[i] = True
+                if result:
+                    break
+            self.partition_converged[:] = False
 
     def tree_evaluate(self, smooth_factor):
         """treeEvaluate (searchAlgo.c:2635)."""
@@ -746,10 +900,14 @@ class TreeSearch:
         import ctypes
         from . import lib
         eng = self.engines[m]
-        entries, p, q, z = self._last_full
-        if not hasattr(self, "_epg_ops") or self._epg_ops[1] is not entries:
-            arr = (TravEntry * len(entries))(*entries)
-            self._epg_ops = (arr, entries)
+        entries, p, q, zv = self._last_full
+        z = float(zv[m]) if self.NB > 1 else float(zv[0])
+        key = (id(entries), m if self.NB > 1 else 0)
+        if not hasattr(self, "_epg_ops") or self._epg_ops[1] != key:
+            src = entries if self.NB == 1 \
+                else self._per_partition_entries(entries, m)
+            arr = (TravEntry * len(src))(*src)
+            self._epg_ops = (arr, key)
         arr = self._epg_ops[0]
         return lib().examl_host_evaluate_partial_dna_cat(
             ctypes.cast(arr, ctypes.c_void_p), len(entries),

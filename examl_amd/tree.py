@@ -37,11 +37,34 @@ class PhyloTree:
         del self.adj[b][a]
 
     def set_z(self, a, b, z):
-        self.adj[a][b] = z
-        self.adj[b][a] = z
+        """Scalar set: with per-partition branch lengths (-M) this sets
+        ALL partitions' values (treeReadLen semantics, treeIO.c)."""
+        cur = self.adj[a][b]
+        if isinstance(cur, np.ndarray):
+            cur[:] = z
+        else:
+            self.adj[a][b] = z
+            self.adj[b][a] = z
 
     def get_z(self, a, b):
-        return self.adj[a][b]
+        z = self.adj[a][b]
+        return float(z[0]) if isinstance(z, np.ndarray) else z
+
+    # -- per-partition branch lengths (-M, tr->numBranches > 1) ------------
+
+    def expand_branches(self, nb):
+        """Switch every edge to an nb-vector branch length (aliased in both
+        adjacency directions, like p->z/p->back->z sharing updates)."""
+        for a, b in self.edges():
+            z = self.adj[a][b]
+            if not isinstance(z, np.ndarray):
+                v = np.full(nb, float(z))
+                self.adj[a][b] = v
+                self.adj[b][a] = v
+
+    def get_zv(self, a, b):
+        z = self.adj[a][b]
+        return z if isinstance(z, np.ndarray) else np.array([z])
 
     def edges(self):
         out = []

@@ -51,13 +51,10 @@ def _search(parts, tree, engine_cls):
         opt_freq_flags=[bool(p.optimizeBaseFrequencies) for p in parts])
 
 
-def test_full_f_E_pipeline_cpu_oracle(golden_dir):
+def test_full_f_E_pipeline_cpu_oracle(optimized_49_cpu):
     """The complete -f E flow on the CPU oracle engines: final lnL within
     1e-6 relative of the reference's (measured: 6.1e-8)."""
-    from tests.helpers import OracleEngine
-    taxa, parts, tree = _load(golden_dir)
-    ts = _search(parts, tree, OracleEngine)
-    lnl = ts.tree_evaluation_mode()
+    ts, lnl = optimized_49_cpu
     assert abs(lnl - GOLDEN_FINAL_LNL) < TOL_ABS, lnl
     # optimized parameters must be in the reference's ballpark
     # (ExaML_modelFile golden: alpha ~0.29/0.28/..., rate AG ~7.4 gene1)
