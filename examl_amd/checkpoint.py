@@ -1,0 +1,347 @@
+"""The reference's binary checkpoint format (searchAlgo.c:1153
+writeCheckpointInner / :1502 readCheckpoint): read AND write, so a run can
+move between the reference and this framework mid-flight (-R restart).
+
+The format is a raw x86-64 struct dump:
+
+  checkPointState (1320 B)                      axml.h:682
+  [constraintVector, only if constraintTree]
+  tree0, tree1            (treeStringLength B each)  axml.c:583
+  [CAT only] rateCategory (int32  x crunchedLength)
+             patrat       (double x crunchedLength)
+  per model:
+      numberOfCategories  int32
+      perSiteRates        double x maxCategories (25)
+      EIGN,EV,EI          per-dataType lengths (DNA 4/16/16, AA 20/400/400)
+      freqExponents, frequencies   (4 | 20)
+      tipVector                    (64 | 460)
+      substRates                   (6 | 190)
+      weights, weightExponents     (4 + 4, LG4X)
+      [LG4 blocks if protModels is LG4M/LG4X]
+      alpha double, gammaRates double x 4
+      protModels int32, autoProtModels int32
+  [state == MOD_OPT] likelihoods (double x numberOfTrees),
+                     treeStrings (treeStringLength x numberOfTrees)
+  writeTree: start->number int32, nodeBaseAddress (8 B),
+             node[mxtips + 3*(mxtips-1)] x 2080 B          axml.h:492
+
+node layout (2080 B): z[256] doubles, next ptr, back ptr, hash u32,
+number i32, x/xPars/xBips chars.  next/back are the WRITER's raw
+addresses; readTree (searchAlgo.c:1311) relocates them against the stored
+nodeBaseAddress.  Record order (axml.c setupTree:608): records
+0..mxtips-1 are tips 1..mxtips; each inner node i occupies a block of 3
+(ring b+2 -> b+1 -> b -> b+2, nodep[i] = b+2, x=1 on record b).
+
+All offsets verified against the reference compiled in place
+(tools/ckpt_layout.c)."""
+
+import struct
+
+import numpy as np
+
+from .tree import PhyloTree
+
+CKP_SIZE = 1320
+NODE_SIZE = 2080
+MAX_CATEGORIES = 25
+NMLNGTH = 256
+
+# pLengths rows (globalVariables.h): states -> (eign, ev, ei, freq, tipvec,
+# subst)
+_PLEN = {4: (4, 16, 16, 4, 64, 6), 20: (20, 400, 400, 20, 460, 190)}
+
+MOD_OPT = 4  # ckp.state (axml.h:658)
+
+# rateHetModel values as stored by the reference (axml.h): CAT=0, GAMMA=1
+RATE_HET_CAT = 0
+RATE_HET_GAMMA = 1
+
+
+def tree_string_length(mxtips):
+    return mxtips * (NMLNGTH + 128) + 256 + mxtips * 2  # axml.c:583
+
+
+def _node_count(mxtips):
+    return mxtips + 3 * (mxtips - 1)
+
+
+class Checkpoint:
+    pass
+
+
+def read_checkpoint(path, mxtips, states_per_model, rate_het="GAMMA",
+                    crunched_length=None, num_trees=1):
+    """Parse a reference checkpoint into plain arrays.  states_per_model:
+    4 or 20 per partition; crunched_length: total pattern count (CAT
+    only)."""
+    d = open(path, "rb").read()
+    ck = Checkpoint()
+    ck.state = struct.unpack_from("<i", d, 0)[0]
+    ck.optimize_rate_category_invocations = struct.unpack_from("<i", d, 48)[0]
+    ck.accumulated_time = struct.unpack_from("<d", d, 56)[0]
+    ck.cat_opt = struct.unpack_from("<i", d, 180)[0]
+    ck.tree_iteration = struct.unpack_from("<i", d, 184)[0]
+    constraint = struct.unpack_from("<i", d, 8)[0]
+    c = 1248
+    ck.likelihood_epsilon = struct.unpack_from("<d", d, c + 24)[0]
+    ck.categories = struct.unpack_from("<i", d, c + 32)[0]
+    ck.rate_het_model = struct.unpack_from("<i", d, c + 52)[0]
+    ck.per_gene_bl = bool(struct.unpack_from("<i", d, c + 16)[0])
+
+    off = CKP_SIZE
+    assert not constraint, "constraint trees not supported"
+    tsl = tree_string_length(mxtips)
+    ck.tree0 = d[off:off + tsl]
+    off += tsl
+    ck.tree1 = d[off:off + tsl]
+    off += tsl
+
+    if rate_het == "CAT":
+        assert crunched_length is not None
+        ck.rate_category = np.frombuffer(d, np.int32, crunched_length, off)
+        off += 4 * crunched_length
+        ck.patrat = np.frombuffer(d, np.float64, crunched_length, off)
+        off += 8 * crunched_length
+
+    ck.models = []
+    for states in states_per_model:
+        eign, ev, ei, freq, tipvec, subst = _PLEN[states]
+        m = {}
+
+        def rd(n, kind="d"):
+            nonlocal off
+            if kind == "i":
+                v = struct.unpack_from("<i", d, off)[0]
+                off += 4
+            else:
+                v = np.frombuffer(d, np.float64, n, off).copy()
+                off += 8 * n
+            return v
+
+        m["num_cats"] = rd(1, "i")
+        m["per_site_rates"] = rd(MAX_CATEGORIES)
+        m["EIGN"] = rd(eign)
+        m["EV"] = rd(ev)
+        m["EI"] = rd(ei)
+        m["freqExponents"] = rd(freq)
+        m["frequencies"] = rd(freq)
+        m["tipVector"] = rd(tipvec)
+        m["substRates"] = rd(subst)
+        m["weights"] = rd(4)
+        m["weightExponents"] = rd(4)
+        m["alpha"] = float(rd(1)[0])
+        m["gammaRates"] = rd(4)
+        m["protModels"] = rd(1, "i")
+        m["autoProtModels"] = rd(1, "i")
+        assert m["protModels"] not in (20, 21), "LG4 checkpoints unsupported"
+        ck.models.append(m)
+
+    if ck.state == MOD_OPT:
+        ck.likelihoods = np.frombuffer(d, np.float64, num_trees, off).copy()
+        off += 8 * num_trees
+        ck.tree_strings = d[off:off + tsl * num_trees]
+        off += tsl * num_trees
+
+    # readTree (searchAlgo.c:1311)
+    ck.start_number = struct.unpack_from("<i", d, off)[0]
+    off += 4
+    base = struct.unpack_from("<Q", d, off)[0]
+    off += 8
+    x = _node_count(mxtips)
+    recs = []
+    for k in range(x):
+        o = off + k * NODE_SIZE
+        z = np.frombuffer(d, np.float64, 256, o)
+        nxt, bck = struct.unpack_from("<QQ", d, o + 2048)
+        number = struct.unpack_from("<i", d, o + 2068)[0]
+        recs.append((z, nxt, bck, number))
+    off += x * NODE_SIZE
+    assert off == len(d), (off, len(d))
+
+    def idx(ptr):
+        if ptr == 0:
+            return None
+        r = (ptr - base) // NODE_SIZE
+        assert 0 <= r < x and (ptr - base) % NODE_SIZE == 0
+        return int(r)
+
+    nb = len(states_per_model) if ck.per_gene_bl else 1
+    tree = PhyloTree.__new__(PhyloTree)
+    tree.ntips = mxtips
+    tree.nnodes = 2 * mxtips - 1
+    tree.adj = {i: {} for i in range(1, 2 * mxtips - 1)}
+    for z, nxt, bck, number in recs:
+        b = idx(bck)
+        if b is None or not (1 <= number <= 2 * mxtips - 2):
+            continue  # spare ring block (setupTree's inter = mxtips-1)
+        nbr = recs[b][3]
+        if nbr in tree.adj.get(number, {}):
+            continue
+        zv = np.array(z[:nb]) if nb > 1 else float(z[0])
+        tree.adj[number][nbr] = zv
+        tree.adj[nbr][number] = zv
+    ck.tree = tree
+    return ck
+
+
+def write_checkpoint(path, tree, models, mxtips, *, state=MOD_OPT,
+                     cat_opt=0, tree_iteration=0, invocations=1,
+                     rate_het="GAMMA", per_gene_bl=False,
+                     likelihood_epsilon=0.1, rate_category=None,
+                     patrat=None, likelihoods=None, start_number=1,
+                     accumulated_time=1.0):
+    """Emit a checkpoint the reference's readCheckpoint accepts.  models:
+    one dict per partition with the arrays of read_checkpoint's layout
+    (build_model_entry converts our model objects)."""
+    tsl = tree_string_length(mxtips)
+    out = bytearray(CKP_SIZE)
+    struct.pack_into("<i", out, 0, state)
+    struct.pack_into("<i", out, 48, invocations)
+    struct.pack_into("<d", out, 56, accumulated_time)
+    struct.pack_into("<i", out, 180, cat_opt)
+    struct.pack_into("<i", out, 184, tree_iteration)
+    c = 1248
+    struct.pack_into("<d", out, c + 24, likelihood_epsilon)
+    struct.pack_into("<i", out, c + 32, MAX_CATEGORIES)
+    struct.pack_into("<i", out, c + 36, 0)   # mode TREE_EVALUATION
+    struct.pack_into("<i", out, c + 44, 0)   # initialSet
+    struct.pack_into("<i", out, c + 48, 10)  # adef->initial default
+    struct.pack_into("<i", out, c + 52,
+                     RATE_HET_CAT if rate_het == "CAT" else RATE_HET_GAMMA)
+    struct.pack_into("<i", out, c + 16, 1 if per_gene_bl else 0)
+
+    buf = bytearray(bytes(out))
+    buf += bytes(tsl)  # tree0
+    buf += bytes(tsl)  # tree1
+
+    if rate_het == "CAT":
+        buf += np.ascontiguousarray(rate_category, np.int32).tobytes()
+        buf += np.ascontiguousarray(patrat, np.float64).tobytes()
+
+    for m in models:
+        states = 4 if len(m["substRates"]) == 6 else 20
+        buf += struct.pack("<i", m["num_cats"])
+        psr = np.zeros(MAX_CATEGORIES)
+        psr[:len(m["per_site_rates"])] = m["per_site_rates"]
+        buf += psr.tobytes()
+        for key in ("EIGN", "EV", "EI", "freqExponents", "frequencies",
+                    "tipVector", "substRates"):
+            a = np.ascontiguousarray(m[key], np.float64)
+            buf += a.tobytes()
+        buf += np.asarray(m.get("weights", np.full(4, 0.25))).tobytes()
+        buf += np.asarray(m.get("weightExponents", np.zeros(4))).tobytes()
+        buf += struct.pack("<d", m["alpha"])
+        buf += np.ascontiguousarray(m["gammaRates"], np.float64).tobytes()
+        buf += struct.pack("<ii", m.get("protModels", 0 if states == 4
+                                        else m.get("protModels", 0)),
+                           m.get("autoProtModels", 2))
+
+    if state == MOD_OPT:
+        lk = likelihoods if likelihoods is not None else [0.0]
+        buf += np.asarray(lk, np.float64).tobytes()
+        buf += bytes(tsl * len(lk))
+
+    # writeTree image: tips at records 0..mxtips-1; inner node i at block
+    # b = mxtips + 3*(i - mxtips - 1), ring b+2 -> b+1 -> b -> b+2,
+    # nodep[i] = b+2 (setupTree, axml.c:608)
+    x = _node_count(mxtips)
+    base = NODE_SIZE  # arbitrary nonzero "address"
+
+    def addr(r):
+        return base + r * NODE_SIZE
+
+    nb = len(models) if per_gene_bl else 1
+
+    def zvec(a, b):
+        zv = tree.get_zv(a, b)
+        z = np.full(256, 0.9)
+        z[:nb] = zv[:nb] if len(zv) >= nb else float(zv[0])
+        return z
+
+    # slot assignment: inner node's ring records in order [b+2, b+1, b]
+    # take its (up to 3) neighbors in adjacency order
+    rec_back = [-1] * x          # record -> record index of back (or -1)
+    rec_z = [None] * x
+    rec_number = [0] * x
+    rec_next = [0] * x
+    rec_x = [0] * x
+    inner_slots = {}
+    for i in range(1, mxtips + 1):
+        rec_number[i - 1] = i
+        rec_next[i - 1] = addr(i - 1)  # p->next = p for tips
+    # setupTree allocates inter = mxtips-1 inner blocks (one spare ring
+    # beyond the mxtips-2 used inner nodes) — number & ring the spare too
+    for i in range(mxtips + 1, 2 * mxtips):
+        b = mxtips + 3 * (i - mxtips - 1)
+        for k in (b, b + 1, b + 2):
+            rec_number[k] = i
+            rec_back[k] = -1
+        rec_next[b + 2] = addr(b + 1)
+        rec_next[b + 1] = addr(b)
+        rec_next[b] = addr(b + 2)
+        rec_x[b] = 1
+        inner_slots[i] = [b + 2, b + 1, b]
+    used = {i: 0 for i in inner_slots}
+
+    def take_slot(i):
+        if i <= mxtips:
+            return i - 1
+        s = inner_slots[i][used[i]]
+        used[i] += 1
+        return s
+
+    seen = set()
+    for a in tree.adj:
+        for bn in tree.adj[a]:
+            if (bn, a) in seen:
+                continue
+            seen.add((a, bn))
+            ra, rb = take_slot(a), take_slot(bn)
+            rec_back[ra] = rb
+            rec_back[rb] = ra
+            rec_z[ra] = rec_z[rb] = zvec(a, bn)
+
+    buf += struct.pack("<i", start_number)
+    buf += struct.pack("<Q", base)
+    for r in range(x):
+        z = rec_z[r] if rec_z[r] is not None else np.full(256, 0.9)
+        rec = bytearray(NODE_SIZE)
+        rec[0:2048] = z.tobytes()
+        struct.pack_into("<Q", rec, 2048, rec_next[r])
+        struct.pack_into("<Q", rec, 2056,
+                         0 if rec_back[r] < 0 else addr(rec_back[r]))
+        struct.pack_into("<i", rec, 2068, rec_number[r])
+        rec[2072] = rec_x[r]
+        buf += bytes(rec)
+    open(path, "wb").write(bytes(buf))
+
+
+def build_model_entry(model, num_cats=1, per_site_rates=(1.0,),
+                      freq_exponents=None):
+    """Convert one of our model objects into the per-model dict of the
+    checkpoint layout."""
+    states = model.states
+    if states == 4:
+        subst = model.rates6
+        prot = 0
+        auto = 2
+    else:
+        subst = model.rates190
+        prot = getattr(model, "prot_model_id", 19)
+        auto = getattr(model, "auto_prot_model", 2)
+    return {
+        "num_cats": num_cats,
+        "per_site_rates": np.asarray(per_site_rates, float),
+        "EIGN": model.EIGN[:states],
+        "EV": model.EV,
+        "EI": model.EI,
+        "freqExponents": (freq_exponents if freq_exponents is not None
+                          else np.zeros(states)),
+        "frequencies": model.frequencies,
+        "tipVector": model.tipVector,
+        "substRates": subst,
+        "alpha": model.alpha,
+        "gammaRates": model.gammaRates,
+        "protModels": prot,
+        "autoProtModels": auto,
+    }
