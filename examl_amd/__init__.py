@@ -91,6 +91,18 @@ def _bind(lib):
         [i, i, i, i, i, p, p, l, p, l, p, l, p]
     lib.examl_hip_core_root_prot_gamma.argtypes = \
         [l, p, p, p, d, p, p, p, p, p]
+    # LG4 (per-category matrices)
+    lib.examl_host_make_p_lg4.argtypes = [d, d, p, p, p, p, p]
+    lib.examl_host_calc_diag_lg4.argtypes = [d, p, p, p]
+    lib.examl_host_core_dtables_prot_lg4.argtypes = [p, p, d, p]
+    lib.examl_hip_newview_traversal_prot_lg4.argtypes = \
+        [p, i, p, p, p, p, p, p, l, p, l, p, l, p, p, p, p]
+    lib.examl_hip_evaluate_root_prot_lg4.argtypes = \
+        [i, i, i, i, i, i, d, p, p, p, p, p, l, p, l, p, l, p, p, p, p, p]
+    lib.examl_hip_sum_root_prot_lg4.argtypes = \
+        [i, i, i, i, i, p, p, l, p, l, p, l, p]
+    lib.examl_hip_core_root_prot_lg4.argtypes = \
+        [l, p, p, p, p, d, p, p, p, p, p]
     lib.examl_hip_profile_enable.argtypes = [i]
     lib.examl_hip_profile_reset.argtypes = []
     lib.examl_hip_profile_get.argtypes = [p, p]
@@ -141,12 +153,13 @@ def check(rc, what):
             f"{lib().examl_hip_last_error_string().decode()}")
 
 
-from .model import DnaGtrModel, ProtGtrModel  # noqa: E402
+from .model import DnaGtrModel, Lg4Model, ProtGtrModel  # noqa: E402
 from .tree import PhyloTree             # noqa: E402
-from .engine import DnaCatEngine, DnaGammaEngine  # noqa: E402
+from .engine import DnaCatEngine, DnaGammaEngine, Lg4Engine  # noqa: E402
 
 __all__ = [
-    "lib", "check", "TravEntry", "DnaGtrModel", "ProtGtrModel", "PhyloTree",
-    "DnaGammaEngine", "DnaCatEngine", "TIP_TIP", "TIP_INNER",
+    "lib", "check", "TravEntry", "DnaGtrModel", "ProtGtrModel",
+    "Lg4Model", "PhyloTree",
+    "DnaGammaEngine", "DnaCatEngine", "Lg4Engine", "TIP_TIP", "TIP_INNER",
     "INNER_INNER", "ZMIN", "ZMAX",
 ]

@@ -2295,3 +2295,570 @@ extern "C" int examl_hip_core_root_prot_gamma(
 }
 
 #undef CHK
+
+/* ===========================================================================
+ * LG4 (LG4M/LG4X) protein kernels — one 20-state matrix per gamma category
+ * (Le/Dang/Gascuel 2012).  newview restates newviewGTRGAMMAPROT_AVX_LG4
+ * (avxLikelihood.c:814, 4-lane dot order); evaluate/sum/core restate the
+ * __SIM_SSE3 generics: evaluateGTRGAMMAPROT_LG4
+ * (evaluateGenericSpecial.c:1164, per-category weights, no 0.25),
+ * sumGAMMAPROT_LG4 (makenewzGenericSpecial.c:1999), coreGTRGAMMAPROT_LG4
+ * (:2489, per-category EIGN + weights).
+ *
+ * Per-category buffers concatenated: EV4 stride 400 (12.8 KB), tipVector4
+ * stride 460 (14.7 KB).  LDS budget per tip case (64 KB limit):
+ *   TIP_TIP:      P 25.6 KB + ump 29.4 KB            (EV from L2)
+ *   TIP_INNER:    P 25.6 KB + ump 14.7 KB + EV 12.8 KB
+ *   INNER_INNER:  P 25.6 KB + EV 12.8 KB
+ * ==========================================================================*/
+
+template <int TC, bool NT>
+__global__ __launch_bounds__(NV_BLOCK) void k_newview_prot_lg4(
+    const double *__restrict__ x1, const double *__restrict__ x2,
+    double *__restrict__ x3, const double *__restrict__ P,
+    const double *__restrict__ EV4, const double *__restrict__ tipVec4,
+    const unsigned char *__restrict__ tipX1,
+    const unsigned char *__restrict__ tipX2, const int *__restrict__ wgt,
+    long n, unsigned int *__restrict__ scalerInc) {
+  __shared__ double sL[1600], sR[1600];
+  __shared__ double sEV[TC != EXAML_TIP_TIP ? 1600 : 1];
+  __shared__ double sU1[TC != EXAML_INNER_INNER ? 1840 : 1];
+  __shared__ double sU2[TC == EXAML_TIP_TIP ? 1840 : 1];
+
+  const int tid = threadIdx.x;
+  for (int j = tid; j < 1600; j += NV_BLOCK) {
+    sL[j] = P[j];
+    sR[j] = P[1600 + j];
+  }
+  if (TC != EXAML_TIP_TIP)
+    for (int j = tid; j < 1600; j += NV_BLOCK) sEV[j] = EV4[j];
+  __syncthreads();
+
+  if (TC != EXAML_INNER_INNER) {
+    /* ump tables: row (code, cat*20+l) uses tipVector[cat] (avx:860-868) */
+    for (int j = tid; j < 23 * 80; j += NV_BLOCK) {
+      const int code = j / 80, k = j % 80;
+      const double *v = &tipVec4[(k / 20) * 460 + 20 * code];
+      sU1[j] = dot20o<false>(v, &sL[k * 20]);
+      if (TC == EXAML_TIP_TIP)
+        sU2[j] = dot20o<false>(v, &sR[k * 20]);
+    }
+    __syncthreads();
+  }
+
+  const long units = n * 4;
+  const int lane = tid & 63;
+  for (long idx = (long)blockIdx.x * NV_BLOCK + tid; idx < units;
+       idx += (long)gridDim.x * NV_BLOCK) {
+    const long site = idx >> 2;
+    const int cat = (int)(idx & 3);
+    const double *EVc =
+        (TC == EXAML_TIP_TIP) ? &EV4[cat * 400] : &sEV[cat * 400];
+    double xl[20], xr[20], acc[20];
+    int code1 = 0, code2 = 0;
+    if (TC == EXAML_INNER_INNER) {
+#pragma unroll
+      for (int s = 0; s < 20; s += 4) {
+        const double4 a = *reinterpret_cast<const double4 *>(&x1[idx * 20 + s]);
+        const double4 b = *reinterpret_cast<const double4 *>(&x2[idx * 20 + s]);
+        xl[s] = a.x; xl[s + 1] = a.y; xl[s + 2] = a.z; xl[s + 3] = a.w;
+        xr[s] = b.x; xr[s + 1] = b.y; xr[s + 2] = b.z; xr[s + 3] = b.w;
+      }
+    } else if (TC == EXAML_TIP_INNER) {
+      code1 = tipX1[site];
+#pragma unroll
+      for (int s = 0; s < 20; s += 4) {
+        const double4 b = *reinterpret_cast<const double4 *>(&x2[idx * 20 + s]);
+        xr[s] = b.x; xr[s + 1] = b.y; xr[s + 2] = b.z; xr[s + 3] = b.w;
+      }
+    } else {
+      code1 = tipX1[site];
+      code2 = tipX2[site];
+    }
+#pragma unroll
+    for (int s = 0; s < 20; s++) acc[s] = 0.0;
+    for (int l = 0; l < 20; l++) {
+      double u1, u2;
+      if (TC == EXAML_INNER_INNER) {
+        u1 = dot20o<false>(xl, &sL[cat * 400 + l * 20]);
+        u2 = dot20o<false>(xr, &sR[cat * 400 + l * 20]);
+      } else if (TC == EXAML_TIP_INNER) {
+        u1 = sU1[80 * code1 + cat * 20 + l];
+        u2 = dot20o<false>(xr, &sR[cat * 400 + l * 20]);
+      } else {
+        u1 = sU1[80 * code1 + cat * 20 + l];
+        u2 = sU2[80 * code2 + cat * 20 + l];
+      }
+      const double t = u1 * u2;
+#pragma unroll
+      for (int s = 0; s < 20; s++) acc[s] += t * EVc[l * 20 + s];
+    }
+
+    if (TC != EXAML_TIP_TIP) {
+      bool small = true;
+#pragma unroll
+      for (int s = 0; s < 20; s++)
+        small &= (fabs(acc[s]) < MINLIKELIHOOD);
+      const unsigned long long m = __ballot(small);
+      if (((m >> (lane & ~3)) & 0xFULL) == 0xFULL) {
+#pragma unroll
+        for (int s = 0; s < 20; s++) acc[s] *= TWOTOTHE256;
+        if ((lane & 3) == 0)
+          atomicAdd(scalerInc, (unsigned int)wgt[site]);
+      }
+    }
+#pragma unroll
+    for (int s = 0; s < 20; s += 4) {
+      const double4 v = make_double4(acc[s], acc[s + 1], acc[s + 2],
+                                     acc[s + 3]);
+      if (NT)
+        __builtin_nontemporal_store(
+            (v4d){v.x, v.y, v.z, v.w},
+            reinterpret_cast<v4d *>(&x3[idx * 20 + s]));
+      else
+        *reinterpret_cast<double4 *>(&x3[idx * 20 + s]) = v;
+    }
+  }
+}
+
+/* diagW = diag[80] | weights[4] */
+template <bool TIP>
+__global__ __launch_bounds__(NV_BLOCK) void k_evaluate_prot_lg4(
+    const double *__restrict__ x1, const double *__restrict__ x2,
+    const double *__restrict__ tipVec4, const unsigned char *__restrict__ tipX1,
+    const int *__restrict__ wgt, const double *__restrict__ diagW, long n,
+    double *__restrict__ partials) {
+  __shared__ double sD[84], sTV[TIP ? 1840 : 1], sRed[NV_BLOCK / 64];
+  const int tid = threadIdx.x;
+  for (int j = tid; j < 84; j += NV_BLOCK) sD[j] = diagW[j];
+  if (TIP)
+    for (int j = tid; j < 1840; j += NV_BLOCK) sTV[j] = tipVec4[j];
+  __syncthreads();
+
+  const long units = n * 4;
+  const int lane = tid & 63;
+  double acc = 0.0;
+  for (long idx = (long)blockIdx.x * NV_BLOCK + tid; idx < units;
+       idx += (long)gridDim.x * NV_BLOCK) {
+    const long site = idx >> 2;
+    const int cat = (int)(idx & 3);
+    const double *le =
+        TIP ? &sTV[cat * 460 + 20 * tipX1[site]] : &x1[idx * 20];
+    double t0 = 0, t1 = 0;
+#pragma unroll
+    for (int l = 0; l < 20; l += 2) {
+      const double2 b = *reinterpret_cast<const double2 *>(&x2[idx * 20 + l]);
+      t0 += le[l] * b.x * sD[cat * 20 + l];
+      t1 += le[l + 1] * b.y * sD[cat * 20 + l + 1];
+    }
+    /* weights fold in per category; no 0.25 (evaluateGTRGAMMAPROT_LG4) */
+    double p = sD[80 + cat] * (t0 + t1);
+    p += __shfl_xor(p, 1);
+    p += __shfl_xor(p, 2);
+    if ((lane & 3) == 0) acc += (double)wgt[site] * log(fabs(p));
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off);
+  if (lane == 0) sRed[tid >> 6] = acc;
+  __syncthreads();
+  if (tid == 0) {
+    double s = 0;
+#pragma unroll
+    for (int w = 0; w < NV_BLOCK / 64; w++) s += sRed[w];
+    partials[blockIdx.x] = s;
+  }
+}
+
+template <int TC>
+__global__ __launch_bounds__(NV_BLOCK) void k_sum_prot_lg4(
+    double *__restrict__ sum, const double *__restrict__ x1,
+    const double *__restrict__ x2, const double *__restrict__ tipVec4,
+    const unsigned char *__restrict__ tipX1,
+    const unsigned char *__restrict__ tipX2, long n) {
+  __shared__ double sTV[TC != EXAML_INNER_INNER ? 1840 : 1];
+  const int tid = threadIdx.x;
+  if (TC != EXAML_INNER_INNER) {
+    for (int j = tid; j < 1840; j += NV_BLOCK) sTV[j] = tipVec4[j];
+    __syncthreads();
+  }
+  const long units = n * 4;
+  for (long idx = (long)blockIdx.x * NV_BLOCK + tid; idx < units;
+       idx += (long)gridDim.x * NV_BLOCK) {
+    const long site = idx >> 2;
+    const int cat = (int)(idx & 3);
+#pragma unroll
+    for (int l = 0; l < 20; l += 2) {
+      double a0, a1, b0, b1;
+      if (TC == EXAML_TIP_TIP) {
+        a0 = sTV[cat * 460 + 20 * tipX1[site] + l];
+        a1 = sTV[cat * 460 + 20 * tipX1[site] + l + 1];
+        b0 = sTV[cat * 460 + 20 * tipX2[site] + l];
+        b1 = sTV[cat * 460 + 20 * tipX2[site] + l + 1];
+      } else if (TC == EXAML_TIP_INNER) {
+        a0 = sTV[cat * 460 + 20 * tipX1[site] + l];
+        a1 = sTV[cat * 460 + 20 * tipX1[site] + l + 1];
+        const double2 b = *reinterpret_cast<const double2 *>(&x2[idx * 20 + l]);
+        b0 = b.x;
+        b1 = b.y;
+      } else {
+        const double2 a = *reinterpret_cast<const double2 *>(&x1[idx * 20 + l]);
+        const double2 b = *reinterpret_cast<const double2 *>(&x2[idx * 20 + l]);
+        a0 = a.x;
+        a1 = a.y;
+        b0 = b.x;
+        b1 = b.y;
+      }
+      *reinterpret_cast<double2 *>(&sum[idx * 20 + l]) =
+          make_double2(a0 * b0, a1 * b1);
+    }
+  }
+}
+
+/* dtabW = {d0,d1,d2}[80 each] | weights[4] (per-category EIGN tables) */
+__global__ __launch_bounds__(NV_BLOCK) void k_core_prot_lg4(
+    const double *__restrict__ sum, const double *__restrict__ dtabW,
+    const int *__restrict__ wgt, long n, double *__restrict__ partials) {
+  __shared__ double sD0[80], sD1[80], sD2[80], sW[4], sRed[2][NV_BLOCK / 64];
+  const int tid = threadIdx.x;
+  for (int j = tid; j < 80; j += NV_BLOCK) {
+    sD0[j] = dtabW[j];
+    sD1[j] = dtabW[80 + j];
+    sD2[j] = dtabW[160 + j];
+  }
+  if (tid < 4) sW[tid] = dtabW[240 + tid];
+  __syncthreads();
+
+  const long units = n * 4;
+  const int lane = tid & 63;
+  double accD1 = 0.0, accD2 = 0.0;
+  for (long idx = (long)blockIdx.x * NV_BLOCK + tid; idx < units;
+       idx += (long)gridDim.x * NV_BLOCK) {
+    const long site = idx >> 2;
+    const int cat = (int)(idx & 3);
+    double a0 = 0, a1 = 0, a2 = 0;
+#pragma unroll
+    for (int l = 0; l < 20; l += 2) {
+      const double2 s2 = *reinterpret_cast<const double2 *>(&sum[idx * 20 + l]);
+      const double te = sD0[cat * 20 + l] * s2.x;
+      const double to = sD0[cat * 20 + l + 1] * s2.y;
+      a0 += te + to;
+      a1 += te * sD1[cat * 20 + l] + to * sD1[cat * 20 + l + 1];
+      a2 += te * sD2[cat * 20 + l] + to * sD2[cat * 20 + l + 1];
+    }
+    /* weights fold per category (coreGTRGAMMAPROT_LG4:2560) */
+    a0 *= sW[cat];
+    a1 *= sW[cat];
+    a2 *= sW[cat];
+    a0 += __shfl_xor(a0, 1);
+    a0 += __shfl_xor(a0, 2);
+    a1 += __shfl_xor(a1, 1);
+    a1 += __shfl_xor(a1, 2);
+    a2 += __shfl_xor(a2, 1);
+    a2 += __shfl_xor(a2, 2);
+    if ((lane & 3) == 0) {
+      const double inv = 1.0 / fabs(a0);
+      const double d1 = a1 * inv, d2 = a2 * inv;
+      const double w = (double)wgt[site];
+      accD1 += w * d1;
+      accD2 += w * (d2 - d1 * d1);
+    }
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    accD1 += __shfl_down(accD1, off);
+    accD2 += __shfl_down(accD2, off);
+  }
+  if (lane == 0) {
+    sRed[0][tid >> 6] = accD1;
+    sRed[1][tid >> 6] = accD2;
+  }
+  __syncthreads();
+  if (tid == 0) {
+    double s1 = 0, s2 = 0;
+#pragma unroll
+    for (int w = 0; w < NV_BLOCK / 64; w++) {
+      s1 += sRed[0][w];
+      s2 += sRed[1][w];
+    }
+    partials[blockIdx.x * 2] = s1;
+    partials[blockIdx.x * 2 + 1] = s2;
+  }
+}
+
+/* ---- LG4 executors ------------------------------------------------------ */
+
+#define CHK(call)                                                              \
+  do {                                                                         \
+    hipError_t _e = (call);                                                    \
+    if (_e != hipSuccess) return set_err(_e, #call);                           \
+  } while (0)
+
+extern "C" void examl_host_make_p_lg4(double z1, double z2,
+                                      const double *gammaRates,
+                                      const double *EI4, const double *EIGN4,
+                                      double *left, double *right);
+extern "C" void examl_host_calc_diag_lg4(double z, const double *gammaRates,
+                                         const double *EIGN4, double *diag);
+extern "C" void examl_host_core_dtables_prot_lg4(const double *EIGN4,
+                                                 const double *gammaRates,
+                                                 double lz, double *dtab);
+
+extern "C" int examl_hip_newview_traversal_prot_lg4(
+    const void *ops_, int numOps, const double *EIGN4, const double *EI4,
+    const double *gammaRates, const double *dev_EV4,
+    const double *dev_tipVec4, double *dev_clv, long clvStride,
+    const unsigned char *dev_tips, long tipStride, const int *dev_wgt, long n,
+    unsigned int *dev_scalers, unsigned int *dev_inc, double *dev_pbuf,
+    void *stream) {
+  const examl_hip_trav_entry *ops = (const examl_hip_trav_entry *)ops_;
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  constexpr int PBLK = 3200; /* 4 cats x 400 x {left,right} */
+  HostPSlot *pslot = hostP_get(dev_pbuf, (size_t)numOps * PBLK);
+  double *hostP = pslot->buf;
+  for (int e = 0; e < numOps; e++) {
+    double qz = ops[e].qz, rz = ops[e].rz;
+    qz = (qz > ZMIN) ? log(qz) : log(ZMIN);
+    rz = (rz > ZMIN) ? log(rz) : log(ZMIN);
+    examl_host_make_p_lg4(qz, rz, gammaRates, EI4, EIGN4, &hostP[e * PBLK],
+                          &hostP[e * PBLK + PBLK / 2]);
+  }
+
+  const bool want_graph =
+      g_use_graphs && !g_prof_on && s != nullptr && numOps >= 8;
+  unsigned long long key = 0;
+  bool capturing = false;
+  if (want_graph) {
+    key = trav_key(ops, numOps, n, dev_clv, dev_tips, dev_pbuf, dev_EV4,
+                   dev_tipVec4, dev_wgt, dev_scalers, dev_inc, (void *)s,
+                   2020 /* LG4 tag */);
+    hipGraphExec_t exec = trav_graph_find(key);
+    if (exec) {
+      CHK(hipGraphLaunch(exec, s));
+      return 0;
+    }
+    capturing =
+        hipStreamBeginCapture(s, hipStreamCaptureModeThreadLocal) ==
+        hipSuccess;
+    (void)hipGetLastError();
+  }
+
+  int rc = 0;
+  do {
+    hipError_t err = hipMemcpyAsync(dev_pbuf, hostP,
+                                    (size_t)numOps * PBLK * sizeof(double),
+                                    hipMemcpyHostToDevice, s);
+    if (err != hipSuccess) { rc = set_err(err, "lg4 pbuf upload"); break; }
+    hipEventRecord(pslot->ev, s);
+    pslot->ev_valid = true;
+    err = hipMemsetAsync(dev_inc, 0, (size_t)numOps * sizeof(unsigned int),
+                         s);
+    if (err != hipSuccess) { rc = set_err(err, "lg4 inc memset"); break; }
+
+    const int grid = grid_for(n * 4);
+    for (int e = 0; e < numOps && rc == 0; e++) {
+      const examl_hip_trav_entry *op = &ops[e];
+      hipEvent_t ev_a = nullptr, ev_b = nullptr;
+      if (g_prof_on) {
+        prof_begin(&ev_a, &ev_b);
+        hipEventRecord(ev_a, s);
+      }
+      const double *P = dev_pbuf + (long)e * PBLK;
+      double *x3 = dev_clv + (long)op->x3Slot * clvStride;
+      const double *x1 = nullptr, *x2 = nullptr;
+      const unsigned char *t1 = nullptr, *t2 = nullptr;
+      switch (op->tipCase) {
+      case EXAML_TIP_TIP:
+        t1 = dev_tips + (long)op->x1Slot * tipStride;
+        t2 = dev_tips + (long)op->x2Slot * tipStride;
+        hipLaunchKernelGGL((k_newview_prot_lg4<EXAML_TIP_TIP, false>),
+                           dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P,
+                           dev_EV4, dev_tipVec4, t1, t2, dev_wgt, n,
+                           dev_inc + e);
+        break;
+      case EXAML_TIP_INNER:
+        t1 = dev_tips + (long)op->x1Slot * tipStride;
+        x2 = dev_clv + (long)op->x2Slot * clvStride;
+        hipLaunchKernelGGL((k_newview_prot_lg4<EXAML_TIP_INNER, false>),
+                           dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P,
+                           dev_EV4, dev_tipVec4, t1, t2, dev_wgt, n,
+                           dev_inc + e);
+        break;
+      case EXAML_INNER_INNER:
+        x1 = dev_clv + (long)op->x1Slot * clvStride;
+        x2 = dev_clv + (long)op->x2Slot * clvStride;
+        hipLaunchKernelGGL((k_newview_prot_lg4<EXAML_INNER_INNER, false>),
+                           dim3(grid), dim3(NV_BLOCK), 0, s, x1, x2, x3, P,
+                           dev_EV4, dev_tipVec4, t1, t2, dev_wgt, n,
+                           dev_inc + e);
+        break;
+      default:
+        snprintf(g_err, sizeof(g_err), "lg4 traversal: bad tipCase %d",
+                 op->tipCase);
+        rc = -1;
+        break;
+      }
+      if (rc == 0) {
+        err = hipGetLastError();
+        if (err != hipSuccess) { rc = set_err(err, "lg4 newview launch"); break; }
+      }
+      if (g_prof_on) {
+        hipEventRecord(ev_b, s);
+        g_prof_pend.push_back({ev_a, ev_b, op->tipCase});
+        if (g_prof_pend.size() > 2048) prof_flush();
+      }
+    }
+    if (rc != 0) break;
+
+    for (int base = 0; base < numOps && rc == 0; base += FIN_CHUNK) {
+      FinMeta m;
+      m.count = (numOps - base < FIN_CHUNK) ? (numOps - base) : FIN_CHUNK;
+      m.base = base;
+      for (int e = 0; e < m.count; e++) {
+        m.p[e] = ops[base + e].pNumber;
+        m.q[e] = ops[base + e].qNumber;
+        m.r[e] = ops[base + e].rNumber;
+      }
+      hipLaunchKernelGGL(k_scaler_finalize, dim3(1), dim3(64), 0, s, m,
+                         dev_inc, dev_scalers);
+      err = hipGetLastError();
+      if (err != hipSuccess) rc = set_err(err, "lg4 scaler finalize");
+    }
+  } while (0);
+
+  if (capturing) {
+    hipGraph_t graph = nullptr;
+    hipError_t err = hipStreamEndCapture(s, &graph);
+    if (rc != 0) {
+      if (graph) hipGraphDestroy(graph);
+      return rc;
+    }
+    hipGraphExec_t exec = nullptr;
+    if (err == hipSuccess) {
+      err = hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0);
+      hipGraphDestroy(graph);
+    }
+    if (err != hipSuccess) {
+      /* the captured sequence never executed: disable graphs and run it
+       * for real */
+      (void)hipGetLastError();
+      g_use_graphs = false;
+      return examl_hip_newview_traversal_prot_lg4(
+          ops_, numOps, EIGN4, EI4, gammaRates, dev_EV4, dev_tipVec4,
+          dev_clv, clvStride, dev_tips, tipStride, dev_wgt, n, dev_scalers,
+          dev_inc, dev_pbuf, stream);
+    }
+    trav_graph_store(key, exec);
+    CHK(hipGraphLaunch(exec, s));
+  }
+  return rc;
+}
+
+extern "C" int examl_hip_evaluate_root_prot_lg4(
+    int rootTipCase, int pNumber, int qNumber, int x1Slot, int x2Slot,
+    int tipSlot, double z, const double *EIGN4, const double *gammaRates,
+    const double *weights, const double *dev_tipVec4, double *dev_clv,
+    long clvStride, const unsigned char *dev_tips, long tipStride,
+    const int *dev_wgt, long n, const unsigned int *dev_scalers,
+    double *dev_diag, double *dev_partials, double *dev_lnl, void *stream) {
+  if (n <= 0) return 0;
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  double hostDiag[84];
+  examl_host_calc_diag_lg4(z, gammaRates, EIGN4, hostDiag);
+  for (int i = 0; i < 4; i++) hostDiag[80 + i] = weights[i];
+  CHK(hipMemcpyAsync(dev_diag, hostDiag, sizeof(hostDiag),
+                     hipMemcpyHostToDevice, s));
+  const double log_minlik = log(MINLIKELIHOOD);
+  const unsigned int *gsP = dev_scalers ? dev_scalers + pNumber : nullptr;
+  const unsigned int *gsQ = dev_scalers ? dev_scalers + qNumber : nullptr;
+  const int grid = grid_for(n * 4);
+  if (rootTipCase == EXAML_TIP_INNER) {
+    const unsigned char *t1 = dev_tips + (long)tipSlot * tipStride;
+    const double *x2 = dev_clv + (long)x2Slot * clvStride;
+    hipLaunchKernelGGL((k_evaluate_prot_lg4<true>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, nullptr, x2, dev_tipVec4, t1,
+                       dev_wgt, dev_diag, n, dev_partials);
+  } else if (rootTipCase == EXAML_INNER_INNER) {
+    const double *x1 = dev_clv + (long)x1Slot * clvStride;
+    const double *x2 = dev_clv + (long)x2Slot * clvStride;
+    hipLaunchKernelGGL((k_evaluate_prot_lg4<false>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, x1, x2, dev_tipVec4, nullptr,
+                       dev_wgt, dev_diag, n, dev_partials);
+  } else {
+    snprintf(g_err, sizeof(g_err), "evaluate_root_lg4: bad tipCase %d",
+             rootTipCase);
+    return -1;
+  }
+  CHK(hipGetLastError());
+  hipLaunchKernelGGL(k_reduce_lnl, dim3(1), dim3(NV_BLOCK), 0, s,
+                     dev_partials, grid, gsP, gsQ, log_minlik, dev_lnl);
+  CHK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int examl_hip_sum_root_prot_lg4(
+    int rootTipCase, int x1Slot, int x2Slot, int tipSlot, int tipSlot2,
+    const double *dev_tipVec4, double *dev_clv, long clvStride,
+    const unsigned char *dev_tips, long tipStride, double *dev_sum, long n,
+    void *stream) {
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  const double *x1 = nullptr, *x2 = nullptr;
+  const unsigned char *t1 = nullptr, *t2 = nullptr;
+  const int grid = grid_for(n * 4);
+  switch (rootTipCase) {
+  case EXAML_TIP_TIP:
+    t1 = dev_tips + (long)tipSlot * tipStride;
+    t2 = dev_tips + (long)tipSlot2 * tipStride;
+    hipLaunchKernelGGL((k_sum_prot_lg4<EXAML_TIP_TIP>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, dev_sum, x1, x2, dev_tipVec4,
+                       t1, t2, n);
+    break;
+  case EXAML_TIP_INNER:
+    t1 = dev_tips + (long)tipSlot * tipStride;
+    x2 = dev_clv + (long)x2Slot * clvStride;
+    hipLaunchKernelGGL((k_sum_prot_lg4<EXAML_TIP_INNER>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, dev_sum, x1, x2, dev_tipVec4,
+                       t1, t2, n);
+    break;
+  case EXAML_INNER_INNER:
+    x1 = dev_clv + (long)x1Slot * clvStride;
+    x2 = dev_clv + (long)x2Slot * clvStride;
+    hipLaunchKernelGGL((k_sum_prot_lg4<EXAML_INNER_INNER>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, dev_sum, x1, x2, dev_tipVec4,
+                       t1, t2, n);
+    break;
+  default:
+    snprintf(g_err, sizeof(g_err), "sum_root_lg4: bad tipCase %d",
+             rootTipCase);
+    return -1;
+  }
+  CHK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int examl_hip_core_root_prot_lg4(
+    long n, const double *dev_sum, const double *EIGN4,
+    const double *gammaRates, const double *weights, double lz,
+    const int *dev_wgt, double *dev_dtab, double *dev_partials,
+    double *dev_out2, void *stream) {
+  if (n <= 0) return 0;
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  double host244[244];
+  examl_host_core_dtables_prot_lg4(EIGN4, gammaRates, lz, host244);
+  for (int i = 0; i < 4; i++) host244[240 + i] = weights[i];
+  CHK(hipMemcpyAsync(dev_dtab, host244, sizeof(host244),
+                     hipMemcpyHostToDevice, s));
+  const int grid = grid_for(n * 4);
+  hipLaunchKernelGGL(k_core_prot_lg4, dim3(grid), dim3(NV_BLOCK), 0, s,
+                     dev_sum, dev_dtab, dev_wgt, n, dev_partials);
+  CHK(hipGetLastError());
+  hipLaunchKernelGGL(k_reduce_2, dim3(1), dim3(NV_BLOCK), 0, s,
+                     dev_partials, grid, dev_out2);
+  CHK(hipGetLastError());
+  return 0;
+}
+
+#undef CHK

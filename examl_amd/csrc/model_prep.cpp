@@ -495,3 +495,63 @@ extern "C" double examl_host_evaluate_partial_dna_cat(
   free(lVector);
   return term;
 }
+
+/* =========================================================================
+ * LG4 (LG4M/LG4X) host model math: per-category matrices.
+ * EIGN4 stride 20 (SCALED, scaleLG4X_EIGN), EI4 stride 400.
+ * =========================================================================*/
+
+/* makeP_FlexLG4 (newviewGenericSpecial.c:170), numStates=20, 4 cats */
+extern "C" void examl_host_make_p_lg4(double z1, double z2,
+                                      const double *gammaRates,
+                                      const double *EI4, const double *EIGN4,
+                                      double *left, double *right) {
+  double d1[20], d2[20];
+  for (int i = 0; i < 4; i++) {
+    const double *EI = EI4 + i * 400;
+    const double *EIGN = EIGN4 + i * 20;
+    for (int j = 1; j < 20; j++) {
+      d1[j] = exp(gammaRates[i] * EIGN[j] * z1);
+      d2[j] = exp(gammaRates[i] * EIGN[j] * z2);
+    }
+    for (int j = 0; j < 20; j++) {
+      left[400 * i + 20 * j] = 1.0;
+      right[400 * i + 20 * j] = 1.0;
+      for (int k = 1; k < 20; k++) {
+        left[400 * i + 20 * j + k] = d1[k] * EI[20 * j + k];
+        right[400 * i + 20 * j + k] = d2[k] * EI[20 * j + k];
+      }
+    }
+  }
+}
+
+/* calcDiagptableFlex_LG4 (evaluateGenericSpecial.c:122): takes RAW z */
+extern "C" void examl_host_calc_diag_lg4(double z, const double *gammaRates,
+                                         const double *EIGN4, double *diag) {
+  const double kZMIN = 1.0E-15; /* axml.h zmin */
+  const double lz = (z < kZMIN) ? log(kZMIN) : log(z);
+  for (int i = 0; i < 4; i++) {
+    diag[i * 20] = 1.0;
+    for (int l = 1; l < 20; l++)
+      diag[i * 20 + l] = exp(gammaRates[i] * EIGN4[i * 20 + l] * lz);
+  }
+}
+
+/* coreGTRGAMMAPROT_LG4 d-tables (makenewzGenericSpecial.c:2501-2517) */
+extern "C" void examl_host_core_dtables_prot_lg4(const double *EIGN4,
+                                                 const double *gammaRates,
+                                                 double lz, double *dtab) {
+  double *d0 = dtab, *d1 = dtab + 80, *d2 = dtab + 160;
+  for (int i = 0; i < 4; i++) {
+    const double ki = gammaRates[i], kisqr = ki * ki;
+    const double *EIGN = EIGN4 + i * 20;
+    d0[i * 20] = 1.0;
+    d1[i * 20] = 0.0;
+    d2[i * 20] = 0.0;
+    for (int l = 1; l < 20; l++) {
+      d0[i * 20 + l] = exp(EIGN[l] * ki * lz);
+      d1[i * 20 + l] = EIGN[l] * ki;
+      d2[i * 20 + l] = EIGN[l] * EIGN[l] * kisqr;
+    }
+  }
+}

@@ -367,3 +367,85 @@ class DnaCatEngine(DnaGammaEngine):
             torch.distributed.all_reduce(self.d_out2)
         v = self.d_out2.cpu()
         return float(v[0]), float(v[1])
+
+
+class Lg4Engine(DnaGammaEngine):
+    """LG4M/LG4X variant: per-gamma-category matrices (EV4/tipVector4
+    device buffers, LG4 kernels; the LG4 branches of newviewIterative /
+    evaluateIterative / makenewzIterative)."""
+
+    def __init__(self, tips, wgt, model, device="cuda", max_ops=None):
+        super().__init__(tips, wgt, model, device=device, max_ops=max_ops)
+        dev = self.device
+        # per-category model buffers replace the base d_EV/d_tipVector
+        self.d_EV4 = torch.from_numpy(model.EV4).to(dev)
+        self.d_tipVector4 = torch.from_numpy(model.tipVector4).to(dev)
+        self.d_diag = torch.empty(84, dtype=torch.float64, device=dev)
+        self.d_dtab = torch.empty(244, dtype=torch.float64, device=dev)
+
+    def upload_model(self):
+        m = self.model
+        self.d_EV4.copy_(torch.from_numpy(m.EV4))
+        self.d_tipVector4.copy_(torch.from_numpy(m.tipVector4))
+
+    def newview_traversal(self, entries):
+        if not entries:
+            return
+        assert len(entries) <= self._max_ops, "grow max_ops"
+        arr = (TravEntry * len(entries))(*entries)
+        m = self.model
+        check(lib().examl_hip_newview_traversal_prot_lg4(
+            ctypes.cast(arr, ctypes.c_void_p), len(entries),
+            _np_vp(m.EIGN4), _np_vp(m.EI4), _np_vp(m.gammaRates),
+            _vp(self.d_EV4), _vp(self.d_tipVector4), _vp(self.d_clv),
+            ctypes.c_long(self.width * 80), _vp(self.d_tips),
+            ctypes.c_long(self.width), _vp(self.d_wgt),
+            ctypes.c_long(self.width), _vp(self.d_scalers), _vp(self.d_inc),
+            _vp(self.d_pbuf), self._stream()), "newview_traversal_lg4")
+
+    def evaluate_root(self, tree, p, q, z, all_reduce=False):
+        tc, x1s, x2s, tslot, _, pn, qn = self._root_case(tree, p, q)
+        self.d_lnl.zero_()
+        m = self.model
+        check(lib().examl_hip_evaluate_root_prot_lg4(
+            tc, pn, qn, x1s, x2s, tslot, ctypes.c_double(z),
+            _np_vp(m.EIGN4), _np_vp(m.gammaRates), _np_vp(m.weights),
+            _vp(self.d_tipVector4), _vp(self.d_clv),
+            ctypes.c_long(self.width * 80), _vp(self.d_tips),
+            ctypes.c_long(self.width), _vp(self.d_wgt),
+            ctypes.c_long(self.width), _vp(self.d_scalers), _vp(self.d_diag),
+            _vp(self.d_partials), _vp(self.d_lnl), self._stream()),
+            "evaluate_root_lg4")
+        if all_reduce and torch.distributed.is_initialized():
+            torch.distributed.all_reduce(self.d_lnl)
+        return self.d_lnl
+
+    def sum_root(self, tree, p, q):
+        if self.d_sum is None:
+            self.d_sum = torch.empty(self.width * 80, dtype=torch.float64,
+                                     device=self.device)
+        p_tip, q_tip = tree.is_tip(p), tree.is_tip(q)
+        if p_tip and q_tip:
+            tc, x1s, x2s, t1, t2 = TIP_TIP, -1, -1, p, q
+        elif q_tip:
+            tc, x1s, x2s, t1, t2 = TIP_INNER, -1, tree.clv_slot(p), q, -1
+        elif p_tip:
+            tc, x1s, x2s, t1, t2 = TIP_INNER, -1, tree.clv_slot(q), p, -1
+        else:
+            tc, x1s, x2s, t1, t2 = (INNER_INNER, tree.clv_slot(p),
+                                    tree.clv_slot(q), -1, -1)
+        check(lib().examl_hip_sum_root_prot_lg4(
+            tc, x1s, x2s, t1, t2, _vp(self.d_tipVector4), _vp(self.d_clv),
+            ctypes.c_long(self.width * 80), _vp(self.d_tips),
+            ctypes.c_long(self.width), _vp(self.d_sum),
+            ctypes.c_long(self.width), self._stream()), "sum_root_lg4")
+
+    def core_derivs_async(self, lz):
+        m = self.model
+        self.d_out2.zero_()
+        check(lib().examl_hip_core_root_prot_lg4(
+            ctypes.c_long(self.width), _vp(self.d_sum), _np_vp(m.EIGN4),
+            _np_vp(m.gammaRates), _np_vp(m.weights), ctypes.c_double(lz),
+            _vp(self.d_wgt), _vp(self.d_dtab), _vp(self.d_partials),
+            _vp(self.d_out2), self._stream()), "core_root_lg4")
+        return self.d_out2

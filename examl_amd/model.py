@@ -102,3 +102,109 @@ class ProtGtrModel:
         d = np.load(os.path.join(os.path.dirname(os.path.abspath(__file__)),
                                  "data", "lg_model.npz"))
         return ProtGtrModel(d["frequencies"], d["rates190"], alpha)
+
+
+class Lg4Model:
+    """LG4M / LG4X model block (Le, Dang & Gascuel 2012): one 20-state
+    matrix per gamma category — the LG4 branch of initReversibleGTR
+    (examl/models.c:3556-3570).  Per-category arrays are stored
+    concatenated: EIGN4 stride 20, EV4/EI4 stride 400, tipVector4 stride
+    460.  EIGN4 is the SCALED eigensystem (scaleLG4X_EIGN,
+    optimizeModel.c:342: raw/sum(w_i*gamma_i)); EIGN4_raw keeps the
+    per-matrix initGeneric output."""
+
+    states = 20
+    n_codes = 23
+    lg4 = True
+
+    def __init__(self, frequencies4, rates190_4, alpha, lg4x=False):
+        self.frequencies4 = np.ascontiguousarray(frequencies4,
+                                                 dtype=np.float64)
+        self.rates190_4 = np.ascontiguousarray(rates190_4, dtype=np.float64)
+        assert self.frequencies4.shape == (4, 20)
+        assert self.rates190_4.shape == (4, 190)
+        self.lg4x = lg4x
+        self.alpha = float(alpha)
+        self.EIGN4_raw = np.zeros(80)
+        self.EIGN4 = np.zeros(80)
+        self.EV4 = np.zeros(1600)
+        self.EI4 = np.zeros(1600)
+        self.tipVector4 = np.zeros(4 * 460)
+        self.gammaRates = np.zeros(4)
+        # LG4X weight state (models.c:4229: 0.25 / 0.0)
+        self.weights = np.full(4, 0.25)
+        self.weightExponents = np.zeros(4)
+        lib().examl_host_make_gamma_cats(self.alpha, _dp(self.gammaRates), 4)
+        self.reinit()
+
+    # aliases so shared engine code (base-engine buffer setup) keeps
+    # working; the LG4 kernels read the per-category buffers instead
+    @property
+    def frequencies(self):
+        return self.frequencies4[0]
+
+    @property
+    def EV(self):
+        return self.EV4
+
+    @property
+    def tipVector(self):
+        return self.tipVector4
+
+    def reinit(self):
+        """4x initGeneric into the raw eigensystems + rescale
+        (models.c:3562-3569)."""
+        L = lib()
+        for k in range(4):
+            f = np.ascontiguousarray(self.frequencies4[k])
+            r = np.ascontiguousarray(self.rates190_4[k])
+            EIGN = np.zeros(20)
+            EV = np.zeros(400)
+            EI = np.zeros(400)
+            tv = np.zeros(460)
+            L.examl_host_init_gtr_aa(_dp(f), _dp(r), _dp(EIGN), _dp(EV),
+                                     _dp(EI), _dp(tv))
+            self.EIGN4_raw[k * 20:(k + 1) * 20] = EIGN
+            self.EV4[k * 400:(k + 1) * 400] = EV
+            self.EI4[k * 400:(k + 1) * 400] = EI
+            self.tipVector4[k * 460:(k + 1) * 460] = tv
+        self.scale_eign()
+
+    def scale_eign(self):
+        """scaleLG4X_EIGN (optimizeModel.c:342)."""
+        acc = 1.0 / float((self.weights * self.gammaRates).sum())
+        self.EIGN4[:] = self.EIGN4_raw * acc
+
+    def set_alpha(self, alpha):
+        """ALPHA_F for LG4M: makeGammaCats ONLY — the reference does NOT
+        rescale EIGN_LG4 here (changeModelParameters, optimizeModel.c:428;
+        harmless because the discrete gamma rates have mean 1)."""
+        self.alpha = float(alpha)
+        lib().examl_host_make_gamma_cats(self.alpha, _dp(self.gammaRates), 4)
+
+    def set_lg4x_rate(self, i, value):
+        """LXRATE_F (optimizeModel.c:451)."""
+        self.gammaRates[i] = value
+        self.scale_eign()
+
+    def set_weight_exponent(self, i, value):
+        """LXWEIGHT_F -> updateWeights (optimizeModel.c:370)."""
+        self.weightExponents[i] = value
+        w = np.exp(self.weightExponents)
+        self.weights[:] = w / w.sum()
+        self.scale_eign()
+
+    @staticmethod
+    def lg4m(alpha=1.0):
+        import os
+        d = np.load(os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                                 "data", "lg4_models.npz"))
+        return Lg4Model(d["lg4m_frequencies"], d["lg4m_rates190"], alpha)
+
+    @staticmethod
+    def lg4x(alpha=1.0):
+        import os
+        d = np.load(os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                                 "data", "lg4_models.npz"))
+        return Lg4Model(d["lg4x_frequencies"], d["lg4x_rates190"], alpha,
+                        lg4x=True)

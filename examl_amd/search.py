@@ -32,7 +32,8 @@ MNBRAK_TINY = 1.0e-20
 MNBRAK_GLIMIT = 100.0
 UNLIKELY = -1.0e300
 
-RATE_F, ALPHA_F, FREQ_F = 0, 1, 2
+RATE_F, ALPHA_F, FREQ_F, LXRATE_F, LXWEIGHT_F = 0, 1, 2, 3, 4
+LG4X_RATE_MIN, LG4X_RATE_MAX = 0.0000001, 1000.0  # axml.h:178
 
 
 def _sign(a, b):
@@ -441,6 +442,14 @@ class TreeSearch:
             eng.upload_model()
         elif which == ALPHA_F:
             model.set_alpha(value)
+        elif which == LXRATE_F:
+            # optimizeModel.c:451: gammaRates[rate]=value + scaleLG4X_EIGN
+            model.set_lg4x_rate(rate_number, value)
+            eng.upload_model()
+        elif which == LXWEIGHT_F:
+            # optimizeModel.c:455: updateWeights + scaleLG4X_EIGN
+            model.set_weight_exponent(rate_number, value)
+            eng.upload_model()
         elif which == FREQ_F:
             w = self.freq_exponents[m]
             w[rate_number] = value
@@ -727,6 +736,12 @@ class TreeSearch:
                     rates = (model.rates6 if model.states == 4
                              else model.rates190)
                     start_values[pos] = rates[rate_number]
+                elif which == LXRATE_F:
+                    lim_inf[pos], lim_sup[pos] = lim_inf_s, lim_sup_s
+                    start_values[pos] = model.gammaRates[rate_number]
+                elif which == LXWEIGHT_F:
+                    lim_inf[pos], lim_sup[pos] = lim_inf_s, lim_sup_s
+                    start_values[pos] = model.weightExponents[rate_number]
                 elif which == FREQ_F:
                     lim_inf[pos] = self._min_freq(m, rate_number, lim_inf_s)
                     lim_sup[pos] = self._max_freq(m, rate_number, lim_sup_s)
@@ -779,11 +794,36 @@ class TreeSearch:
         # partitions (AAisGTR, optimizeModel.c:1614) — LG et al. are fixed.
 
     def opt_alphas_generic(self, model_epsilon):
-        """optAlphasGeneric (optimizeModel.c:1136)."""
+        """optAlphasGeneric (optimizeModel.c:1136): plain-alpha partitions
+        via ALPHA_F, then LG4X partitions via optLG4X (LXRATE_F sweeps +
+        weight optimization)."""
         groups = [[m] for m in range(self.M)]
-        valid = [True] * self.M
-        self._opt_param_generic(groups, valid, -1, ALPHA_MIN, ALPHA_MAX,
-                                ALPHA_F, model_epsilon)
+        is_lg4x = [getattr(self.engines[m].model, "lg4x", False)
+                   for m in range(self.M)]
+        non_lg4x = [not b for b in is_lg4x]
+        if any(non_lg4x):
+            self._opt_param_generic(groups, non_lg4x, -1, ALPHA_MIN,
+                                    ALPHA_MAX, ALPHA_F, model_epsilon)
+        if any(is_lg4x):
+            self._opt_lg4x(model_epsilon, groups, is_lg4x)
+
+    def _opt_lg4x(self, model_epsilon, groups, valid):
+        """optLG4X (optimizeModel.c:1116)."""
+        for i in range(4):
+            self._opt_param_generic(groups, valid, i, LG4X_RATE_MIN,
+                                    LG4X_RATE_MAX, LXRATE_F, model_epsilon)
+            self._optimize_weights(model_epsilon, groups, valid)
+
+    def _optimize_weights(self, model_epsilon, groups, valid):
+        """optimizeWeights (optimizeModel.c:389)."""
+        self.evaluate_generic(full=True)
+        initial = self.likelihood
+        for i in range(4):
+            self._opt_param_generic(groups, valid, i, -1000000.0, 200.0,
+                                    LXWEIGHT_F, model_epsilon)
+        self.evaluate_generic(full=True)
+        assert self.likelihood >= initial - 1e-9
+
 
     def opt_base_freqs(self, model_epsilon):
         """optBaseFreqs + optFreqs (optimizeModel.c:1501/1594)."""

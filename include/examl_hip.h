@@ -393,6 +393,47 @@ int examl_hip_core_root_prot_gamma(long n, const double *dev_sum,
                                    double *dev_partials, double *dev_out2,
                                    void *stream);
 
+
+/* ---- LG4 (LG4M/LG4X): per-gamma-category matrices --------------------- */
+/* Host model math (model_prep.cpp): EIGN4 stride 20 (scaled), EI4 stride
+ * 400, tipVector4 stride 460, EV4 stride 400. */
+void examl_host_make_p_lg4(double z1, double z2, const double *gammaRates,
+                           const double *EI4, const double *EIGN4,
+                           double *left, double *right);
+void examl_host_calc_diag_lg4(double z, const double *gammaRates,
+                              const double *EIGN4, double *diag /*80*/);
+void examl_host_core_dtables_prot_lg4(const double *EIGN4,
+                                      const double *gammaRates, double lz,
+                                      double *dtab /*240*/);
+
+/* L1 executors (replacing the LG4 branches of newviewIterative /
+ * evaluateIterative / makenewzIterative+execCore). */
+int examl_hip_newview_traversal_prot_lg4(
+    const void *ops, int numOps, const double *EIGN4, const double *EI4,
+    const double *gammaRates, const double *dev_EV4,
+    const double *dev_tipVec4, double *dev_clv, long clvStride,
+    const unsigned char *dev_tips, long tipStride, const int *dev_wgt,
+    long n, unsigned int *dev_scalers, unsigned int *dev_inc,
+    double *dev_pbuf, void *stream);
+int examl_hip_evaluate_root_prot_lg4(
+    int rootTipCase, int pNumber, int qNumber, int x1Slot, int x2Slot,
+    int tipSlot, double z, const double *EIGN4, const double *gammaRates,
+    const double *weights, const double *dev_tipVec4, double *dev_clv,
+    long clvStride, const unsigned char *dev_tips, long tipStride,
+    const int *dev_wgt, long n, const unsigned int *dev_scalers,
+    double *dev_diag /*>=84*/, double *dev_partials, double *dev_lnl,
+    void *stream);
+int examl_hip_sum_root_prot_lg4(
+    int rootTipCase, int x1Slot, int x2Slot, int tipSlot, int tipSlot2,
+    const double *dev_tipVec4, double *dev_clv, long clvStride,
+    const unsigned char *dev_tips, long tipStride, double *dev_sum, long n,
+    void *stream);
+int examl_hip_core_root_prot_lg4(
+    long n, const double *dev_sum, const double *EIGN4,
+    const double *gammaRates, const double *weights, double lz,
+    const int *dev_wgt, double *dev_dtab /*>=244*/, double *dev_partials,
+    double *dev_out2, void *stream);
+
 #ifdef __cplusplus
 }
 #endif

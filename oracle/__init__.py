@@ -402,3 +402,79 @@ def ref_make_gamma_cats(alpha, k=4):
     _ref.makeGammaCats(ctypes.c_double(alpha), _dp(rates), ctypes.c_int(k),
                        ctypes.c_int(0))
     return rates
+
+
+# ---------------------------------------------------------------------------
+# Protein LG4 (one matrix per gamma category)
+# ---------------------------------------------------------------------------
+
+def make_p_lg4(z1, z2, rptr, EI4, EIGN4, num_cats=4):
+    left = aligned(num_cats * 400)
+    right = aligned(num_cats * 400)
+    _orc.oracle_make_p_lg4(ctypes.c_double(z1), ctypes.c_double(z2),
+                           _dp(rptr), _dp(EI4), _dp(EIGN4),
+                           ctypes.c_int(num_cats), _dp(left), _dp(right))
+    return left, right
+
+
+def calc_diagptable_lg4(z, rptr, EIGN4):
+    diag = aligned(80)
+    _orc.oracle_calc_diagptable_lg4(ctypes.c_double(z), _dp(rptr),
+                                    _dp(EIGN4), _dp(diag))
+    return diag
+
+
+def newview_prot_lg4(tip_case, x1, x2, EV4, tv4, tipX1, tipX2, n, left,
+                     right, wgt):
+    x3 = aligned(n * 80)
+    inc = ctypes.c_int(0)
+    nullp = ctypes.cast(None, _c_d)
+    nullb = ctypes.cast(None, _c_u8)
+    _orc.oracle_newview_prot_lg4(
+        ctypes.c_int(tip_case),
+        _dp(x1) if x1 is not None else nullp,
+        _dp(x2) if x2 is not None else nullp,
+        _dp(x3), _dp(EV4), _dp(tv4),
+        _u8p(tipX1) if tipX1 is not None else nullb,
+        _u8p(tipX2) if tipX2 is not None else nullb,
+        ctypes.c_int(n), _dp(left), _dp(right), _ip(wgt),
+        ctypes.byref(inc))
+    return x3, inc.value
+
+
+def evaluate_prot_lg4(wptr, x1, x2, tv4, tipX1, n, diag, weights):
+    _orc.oracle_evaluate_prot_lg4.restype = ctypes.c_double
+    nullp = ctypes.cast(None, _c_d)
+    nullb = ctypes.cast(None, _c_u8)
+    return _orc.oracle_evaluate_prot_lg4(
+        _ip(wptr),
+        _dp(x1) if x1 is not None else nullp,
+        _dp(x2) if x2 is not None else nullp,
+        _dp(tv4),
+        _u8p(tipX1) if tipX1 is not None else nullb,
+        ctypes.c_int(n), _dp(diag), _dp(weights))
+
+
+def sum_prot_lg4(tip_case, x1, x2, tv4, tipX1, tipX2, n):
+    st = aligned(n * 80)
+    nullp = ctypes.cast(None, _c_d)
+    nullb = ctypes.cast(None, _c_u8)
+    _orc.oracle_sum_prot_lg4(
+        ctypes.c_int(tip_case), _dp(st),
+        _dp(x1) if x1 is not None else nullp,
+        _dp(x2) if x2 is not None else nullp,
+        _dp(tv4),
+        _u8p(tipX1) if tipX1 is not None else nullb,
+        _u8p(tipX2) if tipX2 is not None else nullb,
+        ctypes.c_int(n))
+    return st
+
+
+def core_prot_lg4(n, sumtable, EIGN4, gammaRates, weights, lz, wgt):
+    d1 = ctypes.c_double(0.0)
+    d2 = ctypes.c_double(0.0)
+    _orc.oracle_core_prot_lg4(
+        ctypes.c_int(n), _dp(sumtable), ctypes.byref(d1), ctypes.byref(d2),
+        _dp(EIGN4), _dp(gammaRates), _dp(weights), ctypes.c_double(lz),
+        _ip(wgt))
+    return d1.value, d2.value

@@ -1066,3 +1066,234 @@ EXPORT void oracle_init_gtr(int n, const unsigned int *valueVector, int vlen,
       if (tipVector[i * n + j] > ORC_MAX_TIP_EV)
         tipVector[i * n + j] = ORC_MAX_TIP_EV;
 }
+
+/* ==========================================================================
+ * Protein LG4 (LG4M/LG4X) kernels — one matrix per gamma category
+ * (Le/Dang/Gascuel 2012).  The AVX build uses newviewGTRGAMMAPROT_AVX_LG4
+ * (avxLikelihood.c:814, 4-lane hadd3 order) for newview and the
+ * __SIM_SSE3 generic kernels for the rest: evaluateGTRGAMMAPROT_LG4
+ * (evaluateGenericSpecial.c:1164), sumGAMMAPROT_LG4
+ * (makenewzGenericSpecial.c:1999), coreGTRGAMMAPROT_LG4
+ * (makenewzGenericSpecial.c:2489), makeP_FlexLG4
+ * (newviewGenericSpecial.c:170), calcDiagptableFlex_LG4
+ * (evaluateGenericSpecial.c:122).  SSE 2-lane even/odd accumulation with
+ * a final hadd; per-category buffers passed with strides 460 (tipVector),
+ * 400 (EV/EI), 20 (EIGN).
+ * ==========================================================================*/
+
+/* 20-dot in the SSE even/odd order: acc_e/acc_o over l += 2, then sum */
+static inline double dot20_sse(const double *a, const double *b) {
+  double e = 0.0, o = 0.0;
+  int l;
+  for (l = 0; l < 20; l += 2) {
+    e += a[l] * b[l];
+    o += a[l + 1] * b[l + 1];
+  }
+  return e + o;
+}
+
+EXPORT void oracle_make_p_lg4(double z1, double z2, const double *rptr,
+                              const double *EI4, const double *EIGN4,
+                              int numCats, double *left, double *right) {
+  int i, j, k;
+  double d1[20], d2[20];
+  for (i = 0; i < numCats; i++) {
+    const double *EI = EI4 + i * 400;
+    const double *EIGN = EIGN4 + i * 20;
+    for (j = 1; j < 20; j++) {
+      d1[j] = exp(rptr[i] * EIGN[j] * z1);
+      d2[j] = exp(rptr[i] * EIGN[j] * z2);
+    }
+    for (j = 0; j < 20; j++) {
+      left[400 * i + 20 * j] = 1.0;
+      right[400 * i + 20 * j] = 1.0;
+      for (k = 1; k < 20; k++) {
+        left[400 * i + 20 * j + k] = d1[k] * EI[20 * j + k];
+        right[400 * i + 20 * j + k] = d2[k] * EI[20 * j + k];
+      }
+    }
+  }
+}
+
+EXPORT void oracle_calc_diagptable_lg4(double z, const double *rptr,
+                                       const double *EIGN4, double *diag) {
+  /* takes the RAW branch length; the zmin clamp + log happen here
+   * (evaluateGenericSpecial.c:133-136) */
+  int i, l;
+  const double lz = (z < ORC_ZMIN) ? log(ORC_ZMIN) : log(z);
+  for (i = 0; i < 4; i++) {
+    const double ki = rptr[i];
+    diag[i * 20] = 1.0;
+    for (l = 1; l < 20; l++)
+      diag[i * 20 + l] = exp(rptr[i] * EIGN4[i * 20 + l] * lz);
+  }
+}
+
+EXPORT void oracle_newview_prot_lg4(int tipCase, const double *x1,
+                                    const double *x2, double *x3,
+                                    const double *EV4, const double *tv4,
+                                    const unsigned char *tipX1,
+                                    const unsigned char *tipX2, int n,
+                                    const double *left, const double *right,
+                                    const int *wgt, int *scalerIncrement) {
+  int i, k, l, s;
+  int addScale = 0;
+  static double umpX1[23 * 80], umpX2[23 * 80];
+
+  if (tipCase != ORC_INNER_INNER) {
+    /* tip precompute: tipVector[k/20] (per-category) rows */
+    for (i = 0; i < 23; i++) {
+      for (k = 0; k < 80; k++) {
+        const double *v = &tv4[(k / 20) * 460 + 20 * i];
+        umpX1[80 * i + k] = dot20_avx(v, &left[k * 20]);
+        if (tipCase == ORC_TIP_TIP)
+          umpX2[80 * i + k] = dot20_avx(v, &right[k * 20]);
+      }
+    }
+  }
+
+  for (i = 0; i < n; i++) {
+    double xv[80];
+    for (k = 0; k < 4; k++) {
+      const double *EV = EV4 + k * 400;
+      double acc[20];
+      for (s = 0; s < 20; s++) acc[s] = 0.0;
+      for (l = 0; l < 20; l++) {
+        double u1, u2;
+        if (tipCase == ORC_TIP_TIP) {
+          u1 = umpX1[80 * tipX1[i] + k * 20 + l];
+          u2 = umpX2[80 * tipX2[i] + k * 20 + l];
+        } else if (tipCase == ORC_TIP_INNER) {
+          u1 = umpX1[80 * tipX1[i] + k * 20 + l];
+          u2 = dot20_avx(&x2[80 * i + 20 * k], &right[k * 400 + l * 20]);
+        } else {
+          u1 = dot20_avx(&x1[80 * i + 20 * k], &left[k * 400 + l * 20]);
+          u2 = dot20_avx(&x2[80 * i + 20 * k], &right[k * 400 + l * 20]);
+        }
+        const double t = u1 * u2;
+        for (s = 0; s < 20; s++) acc[s] += t * EV[20 * l + s];
+      }
+      for (s = 0; s < 20; s++) xv[k * 20 + s] = acc[s];
+    }
+    if (tipCase != ORC_TIP_TIP) {
+      int scale = 1;
+      for (l = 0; scale && l < 80; l++)
+        if (!(fabs(xv[l]) < ORC_MINLIKELIHOOD)) scale = 0;
+      if (scale) {
+        for (l = 0; l < 80; l++) xv[l] *= ORC_TWOTOTHE256;
+        addScale += wgt[i];
+      }
+    }
+    for (l = 0; l < 80; l++) x3[80 * i + l] = xv[l];
+  }
+  *scalerIncrement = addScale;
+}
+
+EXPORT double oracle_evaluate_prot_lg4(const int *wptr, const double *x1_start,
+                                       const double *x2_start,
+                                       const double *tv4,
+                                       const unsigned char *tipX1, int n,
+                                       const double *diagptable,
+                                       const double *weights) {
+  double sum = 0.0;
+  int i, j, l;
+  for (i = 0; i < n; i++) {
+    /* tv 2-lane accumulator across categories: per category a weighted
+     * even/odd partial pair, final hadd */
+    double tv_e = 0.0, tv_o = 0.0;
+    for (j = 0; j < 4; j++) {
+      const double *d = &diagptable[j * 20];
+      const double *le = tipX1 ? &tv4[j * 460 + 20 * tipX1[i]]
+                               : &x1_start[80 * i + 20 * j];
+      const double *ri = &x2_start[80 * i + 20 * j];
+      double t_e = 0.0, t_o = 0.0;
+      for (l = 0; l < 20; l += 2) {
+        t_e += le[l] * ri[l] * d[l];
+        t_o += le[l + 1] * ri[l + 1] * d[l + 1];
+      }
+      tv_e += weights[j] * t_e;
+      tv_o += weights[j] * t_o;
+    }
+    sum += wptr[i] * log(fabs(tv_e + tv_o));
+  }
+  return sum;
+}
+
+EXPORT void oracle_sum_prot_lg4(int tipCase, double *sumtable,
+                                const double *x1_start, const double *x2_start,
+                                const double *tv4, const unsigned char *tipX1,
+                                const unsigned char *tipX2, int n) {
+  int i, l, k;
+  for (i = 0; i < n; i++) {
+    for (l = 0; l < 4; l++) {
+      const double *le, *ri;
+      switch (tipCase) {
+      case ORC_TIP_TIP:
+        le = &tv4[l * 460 + 20 * tipX1[i]];
+        ri = &tv4[l * 460 + 20 * tipX2[i]];
+        break;
+      case ORC_TIP_INNER:
+        le = &tv4[l * 460 + 20 * tipX1[i]];
+        ri = &x2_start[80 * i + l * 20];
+        break;
+      default:
+        le = &x1_start[80 * i + l * 20];
+        ri = &x2_start[80 * i + l * 20];
+      }
+      for (k = 0; k < 20; k++)
+        sumtable[i * 80 + l * 20 + k] = le[k] * ri[k];
+    }
+  }
+}
+
+EXPORT void oracle_core_prot_lg4(int upper, const double *sumtable,
+                                 double *ext_dlnLdlz, double *ext_d2lnLdlz2,
+                                 const double *EIGN4,
+                                 const double *gammaRates,
+                                 const double *weights, double lz,
+                                 const int *wgt) {
+  double dlnLdlz = 0.0, d2lnLdlz2 = 0.0;
+  double d0[80], d1[80], d2[80];
+  int i, j, l;
+  for (i = 0; i < 4; i++) {
+    const double ki = gammaRates[i], kisqr = ki * ki;
+    const double *EIGN = EIGN4 + i * 20;
+    d0[i * 20] = 1.0;
+    d1[i * 20] = 0.0;
+    d2[i * 20] = 0.0;
+    for (l = 1; l < 20; l++) {
+      d0[i * 20 + l] = exp(EIGN[l] * ki * lz);
+      d1[i * 20 + l] = EIGN[l] * ki;
+      d2[i * 20 + l] = EIGN[l] * EIGN[l] * kisqr;
+    }
+  }
+  for (i = 0; i < upper; i++) {
+    const double *sum = &sumtable[i * 80];
+    /* per-category 2-lane accumulators hadd'ed per category, then
+     * weighted into the per-site terms (:2528-2563) */
+    double inv_Li = 0.0, dlnLidlz = 0.0, d2lnLidlz2 = 0.0;
+    for (j = 0; j < 4; j++) {
+      double a0e = 0, a0o = 0, a1e = 0, a1o = 0, a2e = 0, a2o = 0;
+      for (l = 0; l < 20; l += 2) {
+        const double te = d0[j * 20 + l] * sum[j * 20 + l];
+        const double to = d0[j * 20 + l + 1] * sum[j * 20 + l + 1];
+        a0e += te;
+        a0o += to;
+        a1e += te * d1[j * 20 + l];
+        a1o += to * d1[j * 20 + l + 1];
+        a2e += te * d2[j * 20 + l];
+        a2o += to * d2[j * 20 + l + 1];
+      }
+      inv_Li += weights[j] * (a0e + a0o);
+      dlnLidlz += weights[j] * (a1e + a1o);
+      d2lnLidlz2 += weights[j] * (a2e + a2o);
+    }
+    inv_Li = 1.0 / fabs(inv_Li);
+    dlnLidlz *= inv_Li;
+    d2lnLidlz2 *= inv_Li;
+    dlnLdlz += wgt[i] * dlnLidlz;
+    d2lnLdlz2 += wgt[i] * (d2lnLidlz2 - dlnLidlz * dlnLidlz);
+  }
+  *ext_dlnLdlz = dlnLdlz;
+  *ext_d2lnLdlz2 = d2lnLdlz2;
+}

@@ -420,3 +420,87 @@ def oracle_cat_full_lnl(entries, root, tree, model, tips, wgt, cptr,
     if return_state:
         return lnl, clv, scalers
     return lnl
+
+
+class OracleLg4Engine(OracleEngine):
+    """CPU LG4 (LG4M/LG4X) engine — per-category matrices through the
+    oracle LG4 kernels (the checker for the GPU LG4 path)."""
+
+    def __init__(self, tips, wgt, model):
+        super().__init__(tips, wgt, model)
+
+    def newview_traversal(self, entries):
+        m = self.model
+        for e in entries:
+            qz = math.log(e.qz) if e.qz > O.ZMIN else math.log(O.ZMIN)
+            rz = math.log(e.rz) if e.rz > O.ZMIN else math.log(O.ZMIN)
+            left, right = O.make_p_lg4(qz, rz, m.gammaRates, m.EI4, m.EIGN4)
+            if e.tipCase == TIP_TIP:
+                x3, inc = O.newview_prot_lg4(
+                    TIP_TIP, None, None, m.EV4, m.tipVector4,
+                    np.ascontiguousarray(self.tips[e.x1Slot]),
+                    np.ascontiguousarray(self.tips[e.x2Slot]), self.width,
+                    left, right, self.wgt)
+            elif e.tipCase == TIP_INNER:
+                x3, inc = O.newview_prot_lg4(
+                    TIP_INNER, None, self.clv[e.x2Slot], m.EV4, m.tipVector4,
+                    np.ascontiguousarray(self.tips[e.x1Slot]), None,
+                    self.width, left, right, self.wgt)
+            else:
+                x3, inc = O.newview_prot_lg4(
+                    INNER_INNER, self.clv[e.x1Slot], self.clv[e.x2Slot],
+                    m.EV4, m.tipVector4, None, None, self.width, left,
+                    right, self.wgt)
+            self.clv[e.x3Slot] = x3
+            self.scalers[e.pNumber] = (self.scalers[e.qNumber] +
+                                       self.scalers[e.rNumber] + inc)
+
+    def evaluate_root(self, tree, p, q, z):
+        m = self.model
+        diag = O.calc_diagptable_lg4(z, m.gammaRates, m.EIGN4)
+        p_tip, q_tip = tree.is_tip(p), tree.is_tip(q)
+        if q_tip:
+            lnl = O.evaluate_prot_lg4(
+                self.wgt, None, self.clv[tree.clv_slot(p)], m.tipVector4,
+                np.ascontiguousarray(self.tips[q]), self.width, diag,
+                m.weights)
+        elif p_tip:
+            lnl = O.evaluate_prot_lg4(
+                self.wgt, None, self.clv[tree.clv_slot(q)], m.tipVector4,
+                np.ascontiguousarray(self.tips[p]), self.width, diag,
+                m.weights)
+        else:
+            lnl = O.evaluate_prot_lg4(
+                self.wgt, self.clv[tree.clv_slot(p)],
+                self.clv[tree.clv_slot(q)], m.tipVector4, None, self.width,
+                diag, m.weights)
+        lnl += float(self.scalers[p] + self.scalers[q]) * \
+            math.log(O.MINLIKELIHOOD)
+        return lnl
+
+    def sum_root(self, tree, p, q):
+        m = self.model
+        p_tip, q_tip = tree.is_tip(p), tree.is_tip(q)
+        if p_tip and q_tip:
+            self.sumtable = O.sum_prot_lg4(
+                TIP_TIP, None, None, m.tipVector4,
+                np.ascontiguousarray(self.tips[p]),
+                np.ascontiguousarray(self.tips[q]), self.width)
+        elif q_tip:
+            self.sumtable = O.sum_prot_lg4(
+                TIP_INNER, None, self.clv[tree.clv_slot(p)], m.tipVector4,
+                np.ascontiguousarray(self.tips[q]), None, self.width)
+        elif p_tip:
+            self.sumtable = O.sum_prot_lg4(
+                TIP_INNER, None, self.clv[tree.clv_slot(q)], m.tipVector4,
+                np.ascontiguousarray(self.tips[p]), None, self.width)
+        else:
+            self.sumtable = O.sum_prot_lg4(
+                INNER_INNER, self.clv[tree.clv_slot(p)],
+                self.clv[tree.clv_slot(q)], m.tipVector4, None, None,
+                self.width)
+
+    def core_derivs(self, lz):
+        m = self.model
+        return O.core_prot_lg4(self.width, self.sumtable, m.EIGN4,
+                               m.gammaRates, m.weights, lz, self.wgt)
