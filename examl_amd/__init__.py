@@ -92,6 +92,7 @@ def _bind(lib):
     lib.examl_hip_core_root_prot_gamma.argtypes = \
         [l, p, p, p, d, p, p, p, p, p]
     # LG4 (per-category matrices)
+    lib.examl_host_make_gamma_cats_median.argtypes = [d, p, i]
     lib.examl_host_make_p_lg4.argtypes = [d, d, p, p, p, p, p]
     lib.examl_host_calc_diag_lg4.argtypes = [d, p, p, p]
     lib.examl_host_core_dtables_prot_lg4.argtypes = [p, p, d, p]

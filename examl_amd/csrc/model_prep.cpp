@@ -151,6 +151,20 @@ static double PointChi2_(double prob, double v) {
   return ch;
 }
 
+/* the useMedian=TRUE branch of makeGammaCats (models.c:3795) */
+extern "C" void examl_host_make_gamma_cats_median(double alpha,
+                                                  double *gammaRates,
+                                                  int K) {
+  const double factor = alpha / alpha * K, alfa = alpha, beta = alpha;
+  const double middle = 1.0 / (2.0 * K);
+  double t = 0.0;
+  for (int i = 0; i < K; i++)
+    gammaRates[i] =
+        PointChi2_((double)(i * 2 + 1) * middle, 2.0 * alfa) / (2.0 * beta);
+  for (int i = 0; i < K; i++) t += gammaRates[i];
+  for (int i = 0; i < K; i++) gammaRates[i] *= factor / t;
+}
+
 extern "C" void examl_host_make_gamma_cats(double alpha, double *gammaRates,
                                            int K) {
   const double factor = alpha / alpha * K, alfa = alpha, beta = alpha;

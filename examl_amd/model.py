@@ -20,10 +20,11 @@ class DnaGtrModel:
     states = 4
     n_codes = 16
 
-    def __init__(self, frequencies, rates6, alpha):
+    def __init__(self, frequencies, rates6, alpha, use_median=False):
         self.frequencies = np.ascontiguousarray(frequencies, dtype=np.float64)
         self.rates6 = np.ascontiguousarray(rates6, dtype=np.float64)
         self.alpha = float(alpha)
+        self.use_median = use_median
         assert self.frequencies.shape == (4,)
         assert self.rates6.shape == (6,)
         self.EIGN = np.zeros(4)
@@ -35,11 +36,20 @@ class DnaGtrModel:
         L.examl_host_init_gtr_dna(_dp(self.frequencies), _dp(self.rates6),
                                   _dp(self.EIGN), _dp(self.EV), _dp(self.EI),
                                   _dp(self.tipVector))
-        L.examl_host_make_gamma_cats(self.alpha, _dp(self.gammaRates), 4)
+        self._gamma_cats()
+
+    def _gamma_cats(self):
+        # makeGammaCats(...,useMedian) (models.c:3795; -a option)
+        if getattr(self, "use_median", False):
+            lib().examl_host_make_gamma_cats_median(self.alpha,
+                                                    _dp(self.gammaRates), 4)
+        else:
+            lib().examl_host_make_gamma_cats(self.alpha,
+                                             _dp(self.gammaRates), 4)
 
     def set_alpha(self, alpha):
         self.alpha = float(alpha)
-        lib().examl_host_make_gamma_cats(self.alpha, _dp(self.gammaRates), 4)
+        self._gamma_cats()
 
     def reinit(self):
         """Recompute the eigendecomposition after mutating rates6 or
@@ -62,10 +72,11 @@ class ProtGtrModel:
     states = 20
     n_codes = 23
 
-    def __init__(self, frequencies, rates190, alpha):
+    def __init__(self, frequencies, rates190, alpha, use_median=False):
         self.frequencies = np.ascontiguousarray(frequencies, dtype=np.float64)
         self.rates190 = np.ascontiguousarray(rates190, dtype=np.float64)
         self.alpha = float(alpha)
+        self.use_median = use_median
         assert self.frequencies.shape == (20,)
         assert self.rates190.shape == (190,)
         self.EIGN = np.zeros(20)
@@ -77,11 +88,13 @@ class ProtGtrModel:
         L.examl_host_init_gtr_aa(_dp(self.frequencies), _dp(self.rates190),
                                  _dp(self.EIGN), _dp(self.EV), _dp(self.EI),
                                  _dp(self.tipVector))
-        L.examl_host_make_gamma_cats(self.alpha, _dp(self.gammaRates), 4)
+        self._gamma_cats()
+
+    _gamma_cats = DnaGtrModel._gamma_cats
 
     def set_alpha(self, alpha):
         self.alpha = float(alpha)
-        lib().examl_host_make_gamma_cats(self.alpha, _dp(self.gammaRates), 4)
+        self._gamma_cats()
 
     def reinit(self):
         lib().examl_host_init_gtr_aa(_dp(self.frequencies),

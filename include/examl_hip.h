@@ -397,6 +397,8 @@ int examl_hip_core_root_prot_gamma(long n, const double *dev_sum,
 /* ---- LG4 (LG4M/LG4X): per-gamma-category matrices --------------------- */
 /* Host model math (model_prep.cpp): EIGN4 stride 20 (scaled), EI4 stride
  * 400, tipVector4 stride 460, EV4 stride 400. */
+void examl_host_make_gamma_cats_median(double alpha, double *gammaRates,
+                                       int K);
 void examl_host_make_p_lg4(double z1, double z2, const double *gammaRates,
                            const double *EI4, const double *EIGN4,
                            double *left, double *right);
