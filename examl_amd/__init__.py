@@ -104,6 +104,14 @@ def _bind(lib):
         [i, i, i, i, i, p, p, l, p, l, p, l, p]
     lib.examl_hip_core_root_prot_lg4.argtypes = \
         [l, p, p, p, p, d, p, p, p, p, p]
+    # -S (saveMemory) DNA kernels
+    lib.examl_hip_gap_and_prefix.argtypes = [p, p, p, p, i, l, p]
+    lib.examl_hip_newview_dna_save.argtypes = \
+        [i, p, p, p, p, p, p, p, p, p, l, p, p, p, p, p, p, p, p, p, p, p, p]
+    lib.examl_hip_evaluate_dna_save.argtypes = \
+        [i, p, p, p, p, p, p, l, p, p, p, p, p, p, i, i, p, p, p, p]
+    lib.examl_hip_sum_dna_save.argtypes = \
+        [i, p, p, p, p, p, p, l, p, p, p, p, p, p, p]
     lib.examl_hip_profile_enable.argtypes = [i]
     lib.examl_hip_profile_reset.argtypes = []
     lib.examl_hip_profile_get.argtypes = [p, p]
@@ -156,11 +164,13 @@ def check(rc, what):
 
 from .model import DnaGtrModel, Lg4Model, ProtGtrModel  # noqa: E402
 from .tree import PhyloTree             # noqa: E402
-from .engine import DnaCatEngine, DnaGammaEngine, Lg4Engine  # noqa: E402
+from .engine import (DnaCatEngine, DnaGammaEngine, Lg4Engine,  # noqa: E402
+                     SaveDnaEngine)
 
 __all__ = [
     "lib", "check", "TravEntry", "DnaGtrModel", "ProtGtrModel",
     "Lg4Model", "PhyloTree",
-    "DnaGammaEngine", "DnaCatEngine", "Lg4Engine", "TIP_TIP", "TIP_INNER",
+    "DnaGammaEngine", "DnaCatEngine", "Lg4Engine", "SaveDnaEngine",
+    "TIP_TIP", "TIP_INNER",
     "INNER_INNER", "ZMIN", "ZMAX",
 ]

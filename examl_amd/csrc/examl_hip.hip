@@ -2862,3 +2862,467 @@ extern "C" int examl_hip_core_root_prot_lg4(
 }
 
 #undef CHK
+
+/* ===========================================================================
+ * -S (saveMemory / SEV) DNA GTRGAMMA: gap-bit-compacted CLVs + per-node gap
+ * columns (Izquierdo-Carrasco et al.; newviewGTRGAMMA_AVX_GAPPED_SAVE,
+ * avxLikelihood.c:1806; evaluateGTRGAMMA_GAPPED_SAVE / sumGAMMA_GAPPED_SAVE).
+ *
+ * GPU layout: the reference's sequential compaction pointers become O(1)
+ * per-thread indexing through per-node NON-GAP PREFIX arrays: for site i,
+ * compact(i) = prefix[i/32] + popcount(~gap bits of word i/32 below i%32).
+ * k_gap_and_prefix ANDs the child gap vectors into x3's and rebuilds x3's
+ * prefix (single workgroup; gap vectors are a few KB per node).
+ * ==========================================================================*/
+
+__device__ __forceinline__ long save_cidx(const unsigned int *gap,
+                                          const int *prefix, long i) {
+  const unsigned int w = gap[i / 32];
+  const unsigned int below = (i % 32) ? (~w) << (32 - (i % 32)) : 0u;
+  return (long)prefix[i / 32] + __popc(below);
+}
+
+__global__ void k_gap_and_prefix(const unsigned int *__restrict__ g1,
+                                 const unsigned int *__restrict__ g2,
+                                 unsigned int *__restrict__ g3,
+                                 int *__restrict__ prefix, int gvl, long n) {
+  /* one workgroup: parallel AND, then a block-level exclusive scan of
+   * per-word non-gap counts (LDS chunks of blockDim) */
+  __shared__ int sChunk[256];
+  const int tid = threadIdx.x;
+  for (int w = tid; w < gvl; w += blockDim.x) g3[w] = g1[w] & g2[w];
+  __syncthreads();
+  int running = 0;
+  for (int base = 0; base < gvl; base += blockDim.x) {
+    const int w = base + tid;
+    int cnt = 0;
+    if (w < gvl) {
+      unsigned int bits = ~g3[w];
+      const long rem = n - (long)w * 32;
+      if (rem < 32) bits &= (rem <= 0) ? 0u : ((1u << rem) - 1u);
+      cnt = __popc(bits);
+    }
+    sChunk[tid] = cnt;
+    __syncthreads();
+    /* inclusive scan in LDS (blockDim <= 256) */
+    for (int off = 1; off < blockDim.x; off <<= 1) {
+      int v = (tid >= off) ? sChunk[tid - off] : 0;
+      __syncthreads();
+      sChunk[tid] += v;
+      __syncthreads();
+    }
+    if (w < gvl) prefix[w] = running + sChunk[tid] - cnt; /* exclusive */
+    running += sChunk[blockDim.x - 1];
+    __syncthreads();
+  }
+  if (tid == 0) prefix[gvl] = running; /* total non-gap sites */
+}
+
+/* gap-column pass: computes x3_gapColumn (span 16) from the child gap
+ * columns / undetermined tipVector row, plus the scaleGap flag (TT: no
+ * scaling, avx:1879) */
+template <int TC>
+__global__ void k_gapcol_dna_save(const double *__restrict__ P,
+                                  const double *__restrict__ EV,
+                                  const double *__restrict__ tipVec,
+                                  const double *__restrict__ x1_gapcol,
+                                  const double *__restrict__ x2_gapcol,
+                                  double *__restrict__ x3_gapcol,
+                                  int *__restrict__ scaleGap) {
+  if (threadIdx.x != 0 || blockIdx.x != 0) return;
+  const double *L = P, *R = P + 64;
+  double xv[16];
+  const double *tvU = &tipVec[15 * 4];
+  int scale = 1;
+  for (int k = 0; k < 4; k++) {
+    double acc[4] = {0, 0, 0, 0};
+    for (int l = 0; l < 4; l++) {
+      double t;
+      if (TC == EXAML_TIP_TIP) {
+        double p1 = 0, p2 = 0, q1 = 0, q2 = 0;
+        for (int s = 0; s < 2; s++) {
+          p1 += tvU[s] * L[k * 16 + l * 4 + s];
+          p2 += tvU[s + 2] * L[k * 16 + l * 4 + s + 2];
+          q1 += tvU[s] * R[k * 16 + l * 4 + s];
+          q2 += tvU[s + 2] * R[k * 16 + l * 4 + s + 2];
+        }
+        t = (p1 + p2) * (q1 + q2);
+      } else if (TC == EXAML_TIP_INNER) {
+        double p1 = 0, p2 = 0, q1 = 0, q2 = 0;
+        for (int s = 0; s < 2; s++) {
+          p1 += tvU[s] * L[k * 16 + l * 4 + s];
+          p2 += tvU[s + 2] * L[k * 16 + l * 4 + s + 2];
+          q1 += x2_gapcol[k * 4 + s] * R[k * 16 + l * 4 + s];
+          q2 += x2_gapcol[k * 4 + s + 2] * R[k * 16 + l * 4 + s + 2];
+        }
+        t = (p1 + p2) * (q1 + q2);
+      } else {
+        double p1 = 0, p2 = 0, q1 = 0, q2 = 0;
+        for (int s = 0; s < 2; s++) {
+          p1 += x1_gapcol[k * 4 + s] * L[k * 16 + l * 4 + s];
+          p2 += x1_gapcol[k * 4 + s + 2] * L[k * 16 + l * 4 + s + 2];
+          q1 += x2_gapcol[k * 4 + s] * R[k * 16 + l * 4 + s];
+          q2 += x2_gapcol[k * 4 + s + 2] * R[k * 16 + l * 4 + s + 2];
+        }
+        t = (p1 + p2) * (q1 + q2);
+      }
+      for (int s = 0; s < 4; s++) acc[s] += t * EV[l * 4 + s];
+    }
+    for (int s = 0; s < 4; s++) xv[k * 4 + s] = acc[s];
+    if (scale)
+      for (int s = 0; s < 4; s++)
+        if (!(fabs(acc[s]) < MINLIKELIHOOD)) { scale = 0; break; }
+  }
+  if (TC == EXAML_TIP_TIP) scale = 0;
+  if (scale)
+    for (int s = 0; s < 16; s++) xv[s] *= TWOTOTHE256;
+  for (int s = 0; s < 16; s++) x3_gapcol[s] = xv[s];
+  *scaleGap = scale;
+}
+
+/* thread per SITE (gap sites only count the scaler; non-gap compute) */
+template <int TC>
+__global__ __launch_bounds__(NV_BLOCK) void k_newview_dna_save(
+    const double *__restrict__ x1, const double *__restrict__ x2,
+    double *__restrict__ x3, const double *__restrict__ P,
+    const double *__restrict__ EV, const double *__restrict__ tipVec,
+    const unsigned char *__restrict__ tipX1,
+    const unsigned char *__restrict__ tipX2, const int *__restrict__ wgt,
+    long n, unsigned int *__restrict__ scalerInc,
+    const unsigned int *__restrict__ g1, const unsigned int *__restrict__ g2,
+    const unsigned int *__restrict__ g3, const int *__restrict__ pre1,
+    const int *__restrict__ pre2, const int *__restrict__ pre3,
+    const double *__restrict__ x1_gapcol, const double *__restrict__ x2_gapcol,
+    const double *__restrict__ x3_gapcol, const int *__restrict__ scaleGap) {
+  __shared__ double sL[64], sR[64], sEV[16], sTV[TC != EXAML_INNER_INNER ? 64 : 1];
+  __shared__ double sU1[TC != EXAML_INNER_INNER ? 256 : 1];
+  __shared__ double sU2[TC == EXAML_TIP_TIP ? 256 : 1];
+  const int tid = threadIdx.x;
+  if (tid < 64) {
+    sL[tid] = P[tid];
+    sR[tid] = P[64 + tid];
+  }
+  if (tid < 16) sEV[tid] = EV[tid];
+  if (TC != EXAML_INNER_INNER && tid < 64) sTV[tid] = tipVec[tid];
+  __syncthreads();
+  if (TC != EXAML_INNER_INNER) {
+    for (int j = tid; j < 256; j += NV_BLOCK) {
+      const int code = j / 16, kl = j % 16, k = kl / 4, l = kl % 4;
+      if (code == 0) {
+        sU1[j] = 0.0;
+        if (TC == EXAML_TIP_TIP) sU2[j] = 0.0;
+        continue;
+      }
+      const double *tv = &sTV[code * 4];
+      double p1 = 0, p2 = 0;
+      for (int s = 0; s < 2; s++) {
+        p1 += sL[k * 16 + l * 4 + s] * tv[s];
+        p2 += sL[k * 16 + l * 4 + s + 2] * tv[s + 2];
+      }
+      sU1[j] = p1 + p2;
+      if (TC == EXAML_TIP_TIP) {
+        p1 = 0; p2 = 0;
+        for (int s = 0; s < 2; s++) {
+          p1 += sR[k * 16 + l * 4 + s] * tv[s];
+          p2 += sR[k * 16 + l * 4 + s + 2] * tv[s + 2];
+        }
+        sU2[j] = p1 + p2;
+      }
+    }
+    __syncthreads();
+  }
+
+  for (long i = (long)blockIdx.x * NV_BLOCK + tid; i < n;
+       i += (long)gridDim.x * NV_BLOCK) {
+    const bool gap3 = (g3[i / 32] >> (i % 32)) & 1u;
+    if (gap3) {
+      if (TC != EXAML_TIP_TIP && *scaleGap)
+        atomicAdd(scalerInc, (unsigned int)wgt[i]);
+      continue;
+    }
+    const double *xl = nullptr, *xr = nullptr;
+    const double *uX1 = nullptr, *uX2 = nullptr;
+    if (TC == EXAML_TIP_TIP) {
+      uX1 = &sU1[16 * tipX1[i]];
+      uX2 = &sU2[16 * tipX2[i]];
+    } else if (TC == EXAML_TIP_INNER) {
+      uX1 = &sU1[16 * tipX1[i]];
+      xr = ((g2[i / 32] >> (i % 32)) & 1u)
+               ? x2_gapcol
+               : &x2[save_cidx(g2, pre2, i) * 16];
+    } else {
+      xl = ((g1[i / 32] >> (i % 32)) & 1u)
+               ? x1_gapcol
+               : &x1[save_cidx(g1, pre1, i) * 16];
+      xr = ((g2[i / 32] >> (i % 32)) & 1u)
+               ? x2_gapcol
+               : &x2[save_cidx(g2, pre2, i) * 16];
+    }
+    double xv[16];
+    int scale = 1;
+    for (int k = 0; k < 4; k++) {
+      double acc[4] = {0, 0, 0, 0};
+      for (int l = 0; l < 4; l++) {
+        double t;
+        if (TC == EXAML_TIP_TIP) {
+          t = uX1[k * 4 + l] * uX2[k * 4 + l];
+        } else if (TC == EXAML_TIP_INNER) {
+          double q1 = 0, q2 = 0;
+          for (int s = 0; s < 2; s++) {
+            q1 += xr[k * 4 + s] * sR[k * 16 + l * 4 + s];
+            q2 += xr[k * 4 + s + 2] * sR[k * 16 + l * 4 + s + 2];
+          }
+          t = uX1[k * 4 + l] * (q1 + q2);
+        } else {
+          double p1 = 0, p2 = 0, q1 = 0, q2 = 0;
+          for (int s = 0; s < 2; s++) {
+            p1 += xl[k * 4 + s] * sL[k * 16 + l * 4 + s];
+            p2 += xl[k * 4 + s + 2] * sL[k * 16 + l * 4 + s + 2];
+            q1 += xr[k * 4 + s] * sR[k * 16 + l * 4 + s];
+            q2 += xr[k * 4 + s + 2] * sR[k * 16 + l * 4 + s + 2];
+          }
+          t = (p1 + p2) * (q1 + q2);
+        }
+        for (int s = 0; s < 4; s++) acc[s] += t * sEV[l * 4 + s];
+      }
+      for (int s = 0; s < 4; s++) xv[k * 4 + s] = acc[s];
+      if (scale)
+        for (int s = 0; s < 4; s++)
+          if (!(fabs(acc[s]) < MINLIKELIHOOD)) { scale = 0; break; }
+    }
+    if (TC != EXAML_TIP_TIP && scale) {
+      for (int s = 0; s < 16; s++) xv[s] *= TWOTOTHE256;
+      atomicAdd(scalerInc, (unsigned int)wgt[i]);
+    }
+    double *out = &x3[save_cidx(g3, pre3, i) * 16];
+    for (int s = 0; s < 16; s++) out[s] = xv[s];
+  }
+}
+
+template <bool TIP>
+__global__ __launch_bounds__(NV_BLOCK) void k_evaluate_dna_save(
+    const double *__restrict__ x1, const double *__restrict__ x2,
+    const double *__restrict__ tipVec, const unsigned char *__restrict__ tipX1,
+    const int *__restrict__ wgt, const double *__restrict__ diag, long n,
+    double *__restrict__ partials, const unsigned int *__restrict__ g1,
+    const unsigned int *__restrict__ g2, const int *__restrict__ pre1,
+    const int *__restrict__ pre2, const double *__restrict__ x1_gapcol,
+    const double *__restrict__ x2_gapcol) {
+  __shared__ double sD[16], sTV[TIP ? 64 : 1], sRed[NV_BLOCK / 64];
+  const int tid = threadIdx.x;
+  if (tid < 16) sD[tid] = diag[tid];
+  if (TIP && tid < 64) sTV[tid] = tipVec[tid];
+  __syncthreads();
+  const int lane = tid & 63;
+  double acc = 0.0;
+  for (long i = (long)blockIdx.x * NV_BLOCK + tid; i < n;
+       i += (long)gridDim.x * NV_BLOCK) {
+    const double *le, *ri;
+    if (TIP)
+      le = &sTV[4 * tipX1[i]];
+    else
+      le = ((g1[i / 32] >> (i % 32)) & 1u)
+               ? x1_gapcol
+               : &x1[save_cidx(g1, pre1, i) * 16];
+    ri = ((g2[i / 32] >> (i % 32)) & 1u)
+             ? x2_gapcol
+             : &x2[save_cidx(g2, pre2, i) * 16];
+    double t0 = 0, t1 = 0;
+    for (int j = 0; j < 4; j++) {
+      const double *l = TIP ? le : &le[j * 4];
+      t0 += l[0] * ri[j * 4 + 0] * sD[j * 4 + 0];
+      t1 += l[1] * ri[j * 4 + 1] * sD[j * 4 + 1];
+      t0 += l[2] * ri[j * 4 + 2] * sD[j * 4 + 2];
+      t1 += l[3] * ri[j * 4 + 3] * sD[j * 4 + 3];
+    }
+    acc += (double)wgt[i] * log(0.25 * fabs(t0 + t1));
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off);
+  if (lane == 0) sRed[tid >> 6] = acc;
+  __syncthreads();
+  if (tid == 0) {
+    double s = 0;
+#pragma unroll
+    for (int w = 0; w < NV_BLOCK / 64; w++) s += sRed[w];
+    partials[blockIdx.x] = s;
+  }
+}
+
+template <int TC>
+__global__ __launch_bounds__(NV_BLOCK) void k_sum_dna_save(
+    double *__restrict__ sum, const double *__restrict__ x1,
+    const double *__restrict__ x2, const double *__restrict__ tipVec,
+    const unsigned char *__restrict__ tipX1,
+    const unsigned char *__restrict__ tipX2, long n,
+    const unsigned int *__restrict__ g1, const unsigned int *__restrict__ g2,
+    const int *__restrict__ pre1, const int *__restrict__ pre2,
+    const double *__restrict__ x1_gapcol,
+    const double *__restrict__ x2_gapcol) {
+  __shared__ double sTV[TC != EXAML_INNER_INNER ? 64 : 1];
+  const int tid = threadIdx.x;
+  if (TC != EXAML_INNER_INNER && tid < 64) sTV[tid] = tipVec[tid];
+  __syncthreads();
+  for (long i = (long)blockIdx.x * NV_BLOCK + tid; i < n;
+       i += (long)gridDim.x * NV_BLOCK) {
+    const double *a, *b;
+    if (TC == EXAML_TIP_TIP) {
+      a = &sTV[4 * tipX1[i]];
+      b = &sTV[4 * tipX2[i]];
+    } else if (TC == EXAML_TIP_INNER) {
+      a = &sTV[4 * tipX1[i]];
+      b = ((g2[i / 32] >> (i % 32)) & 1u)
+              ? x2_gapcol
+              : &x2[save_cidx(g2, pre2, i) * 16];
+    } else {
+      a = ((g1[i / 32] >> (i % 32)) & 1u)
+              ? x1_gapcol
+              : &x1[save_cidx(g1, pre1, i) * 16];
+      b = ((g2[i / 32] >> (i % 32)) & 1u)
+              ? x2_gapcol
+              : &x2[save_cidx(g2, pre2, i) * 16];
+    }
+    for (int j = 0; j < 4; j++)
+      for (int k = 0; k < 4; k++)
+        sum[i * 16 + j * 4 + k] =
+            (TC == EXAML_INNER_INNER ? a[j * 4 + k] : a[k]) *
+            (TC == EXAML_TIP_TIP ? b[k] : b[j * 4 + k]);
+  }
+}
+
+#define CHK(call)                                                              \
+  do {                                                                         \
+    hipError_t _e = (call);                                                    \
+    if (_e != hipSuccess) return set_err(_e, #call);                           \
+  } while (0)
+
+extern "C" int examl_hip_gap_and_prefix(const unsigned int *g1,
+                                        const unsigned int *g2,
+                                        unsigned int *g3, int *prefix,
+                                        int gvl, long n, void *stream) {
+  (void)hipGetLastError();
+  hipLaunchKernelGGL(k_gap_and_prefix, dim3(1), dim3(256), 0,
+                     (hipStream_t)stream, g1, g2, g3, prefix, gvl, n);
+  CHK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int examl_hip_newview_dna_save(
+    int tipCase, const double *x1, const double *x2, double *x3,
+    const double *P, const double *EV, const double *tipVec,
+    const unsigned char *tipX1, const unsigned char *tipX2, const int *wgt,
+    long n, unsigned int *scalerInc, const unsigned int *g1,
+    const unsigned int *g2, const unsigned int *g3, const int *pre1,
+    const int *pre2, const int *pre3, const double *x1_gapcol,
+    const double *x2_gapcol, double *x3_gapcol, int *scaleGap,
+    void *stream) {
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  const int grid = grid_for(n);
+  switch (tipCase) {
+  case EXAML_TIP_TIP:
+    hipLaunchKernelGGL((k_gapcol_dna_save<EXAML_TIP_TIP>), dim3(1), dim3(64),
+                       0, s, P, EV, tipVec, x1_gapcol, x2_gapcol, x3_gapcol,
+                       scaleGap);
+    CHK(hipGetLastError());
+    hipLaunchKernelGGL((k_newview_dna_save<EXAML_TIP_TIP>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, x1, x2, x3, P, EV, tipVec,
+                       tipX1, tipX2, wgt, n, scalerInc, g1, g2, g3, pre1,
+                       pre2, pre3, x1_gapcol, x2_gapcol, x3_gapcol,
+                       scaleGap);
+    break;
+  case EXAML_TIP_INNER:
+    hipLaunchKernelGGL((k_gapcol_dna_save<EXAML_TIP_INNER>), dim3(1),
+                       dim3(64), 0, s, P, EV, tipVec, x1_gapcol, x2_gapcol,
+                       x3_gapcol, scaleGap);
+    CHK(hipGetLastError());
+    hipLaunchKernelGGL((k_newview_dna_save<EXAML_TIP_INNER>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, x1, x2, x3, P, EV, tipVec,
+                       tipX1, tipX2, wgt, n, scalerInc, g1, g2, g3, pre1,
+                       pre2, pre3, x1_gapcol, x2_gapcol, x3_gapcol,
+                       scaleGap);
+    break;
+  case EXAML_INNER_INNER:
+    hipLaunchKernelGGL((k_gapcol_dna_save<EXAML_INNER_INNER>), dim3(1),
+                       dim3(64), 0, s, P, EV, tipVec, x1_gapcol, x2_gapcol,
+                       x3_gapcol, scaleGap);
+    CHK(hipGetLastError());
+    hipLaunchKernelGGL((k_newview_dna_save<EXAML_INNER_INNER>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, x1, x2, x3, P, EV, tipVec,
+                       tipX1, tipX2, wgt, n, scalerInc, g1, g2, g3, pre1,
+                       pre2, pre3, x1_gapcol, x2_gapcol, x3_gapcol,
+                       scaleGap);
+    break;
+  default:
+    snprintf(g_err, sizeof(g_err), "newview_dna_save: bad tipCase %d",
+             tipCase);
+    return -1;
+  }
+  CHK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int examl_hip_evaluate_dna_save(
+    int tipCase, const double *x1, const double *x2, const double *tipVec,
+    const unsigned char *tipX1, const int *wgt, const double *diag, long n,
+    const unsigned int *g1, const unsigned int *g2, const int *pre1,
+    const int *pre2, const double *x1_gapcol, const double *x2_gapcol,
+    int pNumber, int qNumber, const unsigned int *dev_scalers,
+    double *dev_partials, double *dev_lnl, void *stream) {
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  const int grid = grid_for(n);
+  const double log_minlik = log(MINLIKELIHOOD);
+  const unsigned int *gsP = dev_scalers ? dev_scalers + pNumber : nullptr;
+  const unsigned int *gsQ = dev_scalers ? dev_scalers + qNumber : nullptr;
+  if (tipCase == EXAML_TIP_INNER)
+    hipLaunchKernelGGL((k_evaluate_dna_save<true>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, x1, x2, tipVec, tipX1, wgt,
+                       diag, n, dev_partials, g1, g2, pre1, pre2, x1_gapcol,
+                       x2_gapcol);
+  else
+    hipLaunchKernelGGL((k_evaluate_dna_save<false>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, x1, x2, tipVec, nullptr, wgt,
+                       diag, n, dev_partials, g1, g2, pre1, pre2, x1_gapcol,
+                       x2_gapcol);
+  CHK(hipGetLastError());
+  hipLaunchKernelGGL(k_reduce_lnl, dim3(1), dim3(NV_BLOCK), 0, s,
+                     dev_partials, grid, gsP, gsQ, log_minlik, dev_lnl);
+  CHK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int examl_hip_sum_dna_save(
+    int tipCase, double *dev_sum, const double *x1, const double *x2,
+    const double *tipVec, const unsigned char *tipX1,
+    const unsigned char *tipX2, long n, const unsigned int *g1,
+    const unsigned int *g2, const int *pre1, const int *pre2,
+    const double *x1_gapcol, const double *x2_gapcol, void *stream) {
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  const int grid = grid_for(n);
+  switch (tipCase) {
+  case EXAML_TIP_TIP:
+    hipLaunchKernelGGL((k_sum_dna_save<EXAML_TIP_TIP>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, dev_sum, x1, x2, tipVec, tipX1,
+                       tipX2, n, g1, g2, pre1, pre2, x1_gapcol, x2_gapcol);
+    break;
+  case EXAML_TIP_INNER:
+    hipLaunchKernelGGL((k_sum_dna_save<EXAML_TIP_INNER>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, dev_sum, x1, x2, tipVec, tipX1,
+                       tipX2, n, g1, g2, pre1, pre2, x1_gapcol, x2_gapcol);
+    break;
+  case EXAML_INNER_INNER:
+    hipLaunchKernelGGL((k_sum_dna_save<EXAML_INNER_INNER>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, dev_sum, x1, x2, tipVec, tipX1,
+                       tipX2, n, g1, g2, pre1, pre2, x1_gapcol, x2_gapcol);
+    break;
+  default:
+    snprintf(g_err, sizeof(g_err), "sum_dna_save: bad tipCase %d", tipCase);
+    return -1;
+  }
+  CHK(hipGetLastError());
+  return 0;
+}
+
+#undef CHK

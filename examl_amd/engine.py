@@ -449,3 +449,207 @@ class Lg4Engine(DnaGammaEngine):
             _vp(self.d_wgt), _vp(self.d_dtab), _vp(self.d_partials),
             _vp(self.d_out2), self._stream()), "core_root_lg4")
         return self.d_out2
+
+
+class SaveDnaEngine(DnaGammaEngine):
+    """-S (saveMemory, SEV) DNA GTRGAMMA engine: per-node gap bit vectors,
+    gap-column CLVs and COMPACTED per-node CLV slabs (allocated at each
+    node's non-gap length, like the reference's xSpaceVector realloc,
+    newviewGenericSpecial.c:1200-1218).  The reference's sequential
+    compaction pointers become per-node non-gap prefix arrays for O(1)
+    per-thread indexing on the GPU."""
+
+    def __init__(self, tips, wgt, model, device="cuda", max_ops=None):
+        assert model.states == 4
+        super().__init__(tips, wgt, model, device=device, max_ops=max_ops)
+        dev = self.device
+        self.d_clv = None  # base dense pool unused
+        self.gvl = self.width // 32 + 1
+        nn = 2 * self.ntips
+        # gap vectors: tip rows = (code == 15) (axml.c:2168)
+        host_gap = np.zeros((nn, self.gvl), dtype=np.uint32)
+        for t in range(1, self.ntips + 1):
+            idx = np.nonzero(tips[t] == 15)[0]
+            np.bitwise_or.at(host_gap[t], idx // 32,
+                             (np.uint32(1) << (idx % 32).astype(np.uint32)))
+        self.d_gap = torch.from_numpy(host_gap.view(np.int32)).to(dev)
+        # non-gap prefixes (exclusive, per 32-site word; [gvl] = total)
+        host_pre = np.zeros((nn, self.gvl + 1), dtype=np.int32)
+        for t in range(1, self.ntips + 1):
+            cnt = np.zeros(self.gvl, dtype=np.int64)
+            for w in range(self.gvl):
+                lo, hi = w * 32, min((w + 1) * 32, self.width)
+                bits = int(host_gap[t, w])
+                cnt[w] = (hi - lo) - bin(bits & ((1 << (hi - lo)) - 1)
+                                         ).count("1")
+            host_pre[t, 1:] = np.cumsum(cnt)
+        self.d_pre = torch.from_numpy(host_pre).to(dev)
+        self.d_gapcol = torch.zeros(self.ninner * 16, dtype=torch.float64,
+                                    device=dev)
+        self.d_scalegap = torch.zeros(1, dtype=torch.int32, device=dev)
+        self.clv_slots = {}  # slot -> compacted tensor (lazy realloc)
+        self._tipvec_gapcol = None  # tipVector[15*4:] device view
+
+    def _gap_row(self, node):
+        return _vp(self.d_gap[node])
+
+    def _pre_row(self, node):
+        return _vp(self.d_pre[node])
+
+    def _gapcol_of(self, node_or_slot, is_tip):
+        if is_tip:
+            # undetermined tipVector row (gapOffset, newviewGeneric:1229)
+            return ctypes.c_void_p(self.d_tipVector.data_ptr() + 15 * 4 * 8)
+        return ctypes.c_void_p(self.d_gapcol.data_ptr() +
+                               node_or_slot * 16 * 8)
+
+    def _slot_clv(self, slot, required):
+        t = self.clv_slots.get(slot)
+        if t is None or t.numel() != required:
+            t = torch.empty(required, dtype=torch.float64,
+                            device=self.device)
+            self.clv_slots[slot] = t
+        return t
+
+    def newview_traversal(self, entries):
+        if not entries:
+            return
+        m = self.model
+        L = lib()
+        s = self._stream()
+        for e in entries:
+            qz = math.log(e.qz) if e.qz > ZMIN else math.log(ZMIN)
+            rz = math.log(e.rz) if e.rz > ZMIN else math.log(ZMIN)
+            hostP = np.empty(128)
+            L.examl_host_make_p(ctypes.c_double(qz), ctypes.c_double(rz),
+                                _np_vp(m.gammaRates), _np_vp(m.EI),
+                                _np_vp(m.EIGN), 4, _np_vp(hostP),
+                                ctypes.c_void_p(hostP.ctypes.data + 64 * 8),
+                                4)
+            d_P = torch.from_numpy(hostP).to(self.device)
+            p, q, r = e.pNumber, e.qNumber, e.rNumber
+            check(L.examl_hip_gap_and_prefix(
+                self._gap_row(q), self._gap_row(r), self._gap_row(p),
+                self._pre_row(p), self.gvl, ctypes.c_long(self.width), s),
+                "gap_and_prefix")
+            # compacted x3 slab: width - setBits sites
+            total = int(self.d_pre[p, self.gvl].item())
+            x3 = self._slot_clv(e.x3Slot, max(total, 1) * 16)
+            q_tip = e.tipCase != INNER_INNER
+            r_tip = e.tipCase == TIP_TIP
+            x1 = None if q_tip else self.clv_slots[e.x1Slot]
+            x2 = None if r_tip else self.clv_slots[e.x2Slot]
+            t1 = (ctypes.c_void_p(self.d_tips.data_ptr() +
+                                  e.x1Slot * self.width)
+                  if q_tip else None)
+            t2 = (ctypes.c_void_p(self.d_tips.data_ptr() +
+                                  e.x2Slot * self.width)
+                  if r_tip else None)
+            check(L.examl_hip_newview_dna_save(
+                e.tipCase,
+                _vp(x1) if x1 is not None else None,
+                _vp(x2) if x2 is not None else None,
+                _vp(x3), _vp(d_P), _vp(self.d_EV), _vp(self.d_tipVector),
+                t1, t2, _vp(self.d_wgt), ctypes.c_long(self.width),
+                _vp(self.d_inc), self._gap_row(q), self._gap_row(r),
+                self._gap_row(p), self._pre_row(q), self._pre_row(r),
+                self._pre_row(p),
+                self._gapcol_of(q if q_tip else self.tree_slot(e.x1Slot),
+                                q_tip),
+                self._gapcol_of(r if r_tip else self.tree_slot(e.x2Slot),
+                                r_tip),
+                self._gapcol_of(e.x3Slot, False), _vp(self.d_scalegap), s),
+                "newview_dna_save")
+            # recursive scaler accumulation on host via the shared finalize
+            # kernel is overkill here; do it with the same kernel used by
+            # the dense path (one op at a time)
+            self._finalize_scaler_one(p, q, r)
+        # d_inc consumed per op inside _finalize_scaler_one
+
+    def tree_slot(self, slot):
+        return slot
+
+    def _finalize_scaler_one(self, p, q, r):
+        # gs[p] = gs[q] + gs[r] + inc (newviewGenericSpecial.c:1503);
+        # single-op version using torch ops on the stream
+        inc = self.d_inc[0]
+        self.d_scalers[p] = self.d_scalers[q] + self.d_scalers[r] + inc
+        self.d_inc.zero_()
+
+    def evaluate_root(self, tree, p, q, z, all_reduce=False):
+        tc, x1s, x2s, tslot, _, pn, qn = self._root_case(tree, p, q)
+        self.d_lnl.zero_()
+        m = self.model
+        hostDiag = np.empty(16)
+        lib().examl_host_calc_diagptable(ctypes.c_double(z), 4, 4,
+                                         _np_vp(m.gammaRates),
+                                         _np_vp(m.EIGN), _np_vp(hostDiag))
+        d_diag = torch.from_numpy(hostDiag).to(self.device)
+        s = self._stream()
+        grid_partials = self.d_partials
+        if tc == TIP_INNER:
+            x2 = self.clv_slots[x2s]
+            inner_node = p if tree.is_tip(q) else q
+            tip_node = q if tree.is_tip(q) else p
+            check(lib().examl_hip_evaluate_dna_save(
+                TIP_INNER, None, _vp(x2), _vp(self.d_tipVector),
+                ctypes.c_void_p(self.d_tips.data_ptr() +
+                                tslot * self.width),
+                _vp(self.d_wgt), _vp(d_diag), ctypes.c_long(self.width),
+                None, self._gap_row(inner_node), None,
+                self._pre_row(inner_node),
+                None, self._gapcol_of(x2s, False), pn, qn,
+                _vp(self.d_scalers), _vp(grid_partials), _vp(self.d_lnl),
+                s), "evaluate_dna_save")
+        else:
+            x1 = self.clv_slots[x1s]
+            x2 = self.clv_slots[x2s]
+            check(lib().examl_hip_evaluate_dna_save(
+                INNER_INNER, _vp(x1), _vp(x2), _vp(self.d_tipVector), None,
+                _vp(self.d_wgt), _vp(d_diag), ctypes.c_long(self.width),
+                self._gap_row(p), self._gap_row(q), self._pre_row(p),
+                self._pre_row(q), self._gapcol_of(x1s, False),
+                self._gapcol_of(x2s, False), pn, qn, _vp(self.d_scalers),
+                _vp(grid_partials), _vp(self.d_lnl), s),
+                "evaluate_dna_save")
+        if all_reduce and torch.distributed.is_initialized():
+            torch.distributed.all_reduce(self.d_lnl)
+        return self.d_lnl
+
+    def sum_root(self, tree, p, q):
+        if self.d_sum is None:
+            self.d_sum = torch.empty(self.width * 16, dtype=torch.float64,
+                                     device=self.device)
+        p_tip, q_tip = tree.is_tip(p), tree.is_tip(q)
+        s = self._stream()
+        if p_tip and q_tip:
+            check(lib().examl_hip_sum_dna_save(
+                TIP_TIP, _vp(self.d_sum), None, None,
+                _vp(self.d_tipVector),
+                ctypes.c_void_p(self.d_tips.data_ptr() + p * self.width),
+                ctypes.c_void_p(self.d_tips.data_ptr() + q * self.width),
+                ctypes.c_long(self.width), None, None, None, None, None,
+                None, s), "sum_dna_save")
+            return
+        if q_tip or p_tip:
+            tip, inner = (q, p) if q_tip else (p, q)
+            sl = tree.clv_slot(inner)
+            check(lib().examl_hip_sum_dna_save(
+                TIP_INNER, _vp(self.d_sum), None,
+                _vp(self.clv_slots[sl]), _vp(self.d_tipVector),
+                ctypes.c_void_p(self.d_tips.data_ptr() + tip * self.width),
+                None, ctypes.c_long(self.width), None,
+                self._gap_row(inner), None, self._pre_row(inner), None,
+                self._gapcol_of(sl, False), s), "sum_dna_save")
+            return
+        s1, s2 = tree.clv_slot(p), tree.clv_slot(q)
+        check(lib().examl_hip_sum_dna_save(
+            INNER_INNER, _vp(self.d_sum), _vp(self.clv_slots[s1]),
+            _vp(self.clv_slots[s2]), _vp(self.d_tipVector), None, None,
+            ctypes.c_long(self.width), self._gap_row(p), self._gap_row(q),
+            self._pre_row(p), self._pre_row(q), self._gapcol_of(s1, False),
+            self._gapcol_of(s2, False), s), "sum_dna_save")
+
+    def clv_bytes(self):
+        """actual CLV memory footprint (the -S saving)"""
+        return sum(t.numel() * 8 for t in self.clv_slots.values())

@@ -436,6 +436,34 @@ int examl_hip_core_root_prot_lg4(
     const int *dev_wgt, double *dev_dtab /*>=244*/, double *dev_partials,
     double *dev_out2, void *stream);
 
+
+/* ---- -S (saveMemory, SEV) DNA GTRGAMMA: compacted CLVs + gap columns -- */
+int examl_hip_gap_and_prefix(const unsigned int *g1, const unsigned int *g2,
+                             unsigned int *g3, int *prefix /*gvl+1*/,
+                             int gvl, long n, void *stream);
+int examl_hip_newview_dna_save(
+    int tipCase, const double *x1, const double *x2, double *x3,
+    const double *P /*left|right, 128*/, const double *EV,
+    const double *tipVec, const unsigned char *tipX1,
+    const unsigned char *tipX2, const int *wgt, long n,
+    unsigned int *scalerInc, const unsigned int *g1, const unsigned int *g2,
+    const unsigned int *g3, const int *pre1, const int *pre2,
+    const int *pre3, const double *x1_gapcol, const double *x2_gapcol,
+    double *x3_gapcol, int *scaleGap, void *stream);
+int examl_hip_evaluate_dna_save(
+    int tipCase, const double *x1, const double *x2, const double *tipVec,
+    const unsigned char *tipX1, const int *wgt, const double *diag, long n,
+    const unsigned int *g1, const unsigned int *g2, const int *pre1,
+    const int *pre2, const double *x1_gapcol, const double *x2_gapcol,
+    int pNumber, int qNumber, const unsigned int *dev_scalers,
+    double *dev_partials, double *dev_lnl, void *stream);
+int examl_hip_sum_dna_save(
+    int tipCase, double *dev_sum, const double *x1, const double *x2,
+    const double *tipVec, const unsigned char *tipX1,
+    const unsigned char *tipX2, long n, const unsigned int *g1,
+    const unsigned int *g2, const int *pre1, const int *pre2,
+    const double *x1_gapcol, const double *x2_gapcol, void *stream);
+
 #ifdef __cplusplus
 }
 #endif

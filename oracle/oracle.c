@@ -1297,3 +1297,236 @@ EXPORT void oracle_core_prot_lg4(int upper, const double *sumtable,
   *ext_dlnLdlz = dlnLdlz;
   *ext_d2lnLdlz2 = d2lnLdlz2;
 }
+
+/* ==========================================================================
+ * -S (saveMemory / SEV) DNA GTRGAMMA kernels: gap-bit-compacted CLVs plus a
+ * per-node "gap column" holding the CLV of an all-undetermined site.
+ * Restate newviewGTRGAMMA_AVX_GAPPED_SAVE (avxLikelihood.c:1806),
+ * evaluateGTRGAMMA_GAPPED_SAVE (evaluateGenericSpecial.c:1750) and
+ * sumGAMMA_GAPPED_SAVE (makenewzGenericSpecial.c:1716).
+ * gap vectors: u32 words, bit (site%32) of word (site/32); compacted CLV
+ * arrays hold only no-gap sites in site order.
+ * ==========================================================================*/
+
+static inline int orc_is_gap(const unsigned int *g, int i) {
+  return (g[i / 32] >> (i % 32)) & 1u;
+}
+
+/* per-(site,cat) core of the AVX DNA newview, shared by the SAVE variant */
+static void orc_nv_dna_site(int tipCase, const double *uX1, const double *uX2,
+                            const double *xvl, const double *xvr,
+                            const double *left, const double *right,
+                            const double *extEV, double *xv, int *scale) {
+  int k, l, s;
+  *scale = 1;
+  for (k = 0; k < 4; k++) {
+    double acc[4] = {0, 0, 0, 0};
+    for (l = 0; l < 4; l++) {
+      double t;
+      if (tipCase == ORC_TIP_TIP) {
+        t = uX1[k * 4 + l] * uX2[k * 4 + l];
+      } else if (tipCase == ORC_TIP_INNER) {
+        double p[4];
+        for (s = 0; s < 4; s++)
+          p[s] = xvr[k * 4 + s] * right[k * 16 + l * 4 + s];
+        t = uX1[k * 4 + l] * hadd4d(p);
+      } else {
+        double pl[4], pr[4];
+        for (s = 0; s < 4; s++) {
+          pl[s] = xvl[k * 4 + s] * left[k * 16 + l * 4 + s];
+          pr[s] = xvr[k * 4 + s] * right[k * 16 + l * 4 + s];
+        }
+        t = hadd4d(pl) * hadd4d(pr);
+      }
+      for (s = 0; s < 4; s++) acc[s] += t * extEV[l * 4 + s];
+    }
+    for (s = 0; s < 4; s++) xv[k * 4 + s] = acc[s];
+    if (*scale) {
+      for (s = 0; s < 4; s++)
+        if (!(fabs(acc[s]) < ORC_MINLIKELIHOOD)) { *scale = 0; break; }
+    }
+  }
+}
+
+EXPORT void oracle_newview_dna_gamma_save(
+    int tipCase, const double *x1, const double *x2, double *x3,
+    const double *extEV, const double *tipVector,
+    const unsigned char *tipX1, const unsigned char *tipX2, int n,
+    const double *left, const double *right, const int *wgt,
+    int *scalerIncrement, const unsigned int *x1_gap,
+    const unsigned int *x2_gap, unsigned int *x3_gap,
+    const double *x1_gapColumn, const double *x2_gapColumn,
+    double *x3_gapColumn) {
+  int i, k, l, s;
+  int addScale = 0, scale, scaleGap = 0;
+  double umpX1[16 * 16], umpX2[16 * 16];
+  const double *x1_ptr = x1, *x2_ptr = x2;
+  double *x3_ptr = x3;
+
+  if (tipCase != ORC_INNER_INNER) {
+    for (i = 1; i < 16; i++) {
+      const double *tv = &tipVector[i * 4];
+      for (k = 0; k < 4; k++)
+        for (l = 0; l < 4; l++) {
+          double p[4];
+          for (s = 0; s < 4; s++) p[s] = left[k * 16 + l * 4 + s] * tv[s];
+          umpX1[i * 16 + k * 4 + l] = hadd4d(p);
+          if (tipCase == ORC_TIP_TIP) {
+            for (s = 0; s < 4; s++) p[s] = right[k * 16 + l * 4 + s] * tv[s];
+            umpX2[i * 16 + k * 4 + l] = hadd4d(p);
+          }
+        }
+    }
+  }
+
+  /* gap column first (scaleGap; TT never scales, avx:1879-1908) */
+  {
+    double xv[16];
+    if (tipCase == ORC_TIP_TIP) {
+      orc_nv_dna_site(ORC_TIP_TIP, &umpX1[240], &umpX2[240], NULL, NULL,
+                      left, right, extEV, xv, &scale);
+      scaleGap = 0;
+    } else if (tipCase == ORC_TIP_INNER) {
+      orc_nv_dna_site(ORC_TIP_INNER, &umpX1[240], NULL, NULL, x2_gapColumn,
+                      left, right, extEV, xv, &scaleGap);
+    } else {
+      orc_nv_dna_site(ORC_INNER_INNER, NULL, NULL, x1_gapColumn,
+                      x2_gapColumn, left, right, extEV, xv, &scaleGap);
+    }
+    if (scaleGap)
+      for (s = 0; s < 16; s++) xv[s] *= ORC_TWOTOTHE256;
+    for (s = 0; s < 16; s++) x3_gapColumn[s] = xv[s];
+  }
+
+  for (i = 0; i < n; i++) {
+    if (tipCase != ORC_TIP_TIP && orc_is_gap(x3_gap, i)) {
+      if (scaleGap) addScale += wgt[i];
+      continue;
+    }
+    if (tipCase == ORC_TIP_TIP && orc_is_gap(x3_gap, i))
+      continue; /* TT gap sites: nothing stored, never scaled */
+    {
+      double xv[16];
+      const double *xl = NULL, *xr = NULL;
+      const double *uX1 = NULL, *uX2 = NULL;
+      if (tipCase == ORC_TIP_TIP) {
+        uX1 = &umpX1[16 * tipX1[i]];
+        uX2 = &umpX2[16 * tipX2[i]];
+      } else if (tipCase == ORC_TIP_INNER) {
+        uX1 = &umpX1[16 * tipX1[i]];
+        if (orc_is_gap(x2_gap, i))
+          xr = x2_gapColumn;
+        else {
+          xr = x2_ptr;
+          x2_ptr += 16;
+        }
+      } else {
+        if (orc_is_gap(x1_gap, i))
+          xl = x1_gapColumn;
+        else {
+          xl = x1_ptr;
+          x1_ptr += 16;
+        }
+        if (orc_is_gap(x2_gap, i))
+          xr = x2_gapColumn;
+        else {
+          xr = x2_ptr;
+          x2_ptr += 16;
+        }
+      }
+      orc_nv_dna_site(tipCase, uX1, uX2, xl, xr, left, right, extEV, xv,
+                      &scale);
+      if (tipCase != ORC_TIP_TIP && scale) {
+        for (s = 0; s < 16; s++) xv[s] *= ORC_TWOTOTHE256;
+        addScale += wgt[i];
+      }
+      for (s = 0; s < 16; s++) x3_ptr[s] = xv[s];
+      x3_ptr += 16;
+    }
+  }
+  *scalerIncrement = addScale;
+}
+
+EXPORT double oracle_evaluate_dna_gamma_save(
+    const int *wptr, const double *x1_start, const double *x2_start,
+    const double *tipVector, const unsigned char *tipX1, int n,
+    const double *diagptable, const double *x1_gapColumn,
+    const double *x2_gapColumn, const unsigned int *x1_gap,
+    const unsigned int *x2_gap) {
+  double sum = 0.0;
+  int i, j;
+  const double *x1_ptr = x1_start, *x2_ptr = x2_start;
+  for (i = 0; i < n; i++) {
+    const double *x1, *x2;
+    if (tipX1) {
+      x1 = &tipVector[4 * tipX1[i]];
+    } else if (orc_is_gap(x1_gap, i)) {
+      x1 = x1_gapColumn;
+    } else {
+      x1 = x1_ptr;
+      x1_ptr += 16;
+    }
+    if (orc_is_gap(x2_gap, i)) {
+      x2 = x2_gapColumn;
+    } else {
+      x2 = x2_ptr;
+      x2_ptr += 16;
+    }
+    double t0 = 0.0, t1 = 0.0;
+    for (j = 0; j < 4; j++) {
+      const double *l = tipX1 ? x1 : &x1[j * 4];
+      t0 += l[0] * x2[j * 4 + 0] * diagptable[j * 4 + 0];
+      t1 += l[1] * x2[j * 4 + 1] * diagptable[j * 4 + 1];
+      t0 += l[2] * x2[j * 4 + 2] * diagptable[j * 4 + 2];
+      t1 += l[3] * x2[j * 4 + 3] * diagptable[j * 4 + 3];
+    }
+    sum += wptr[i] * log(0.25 * fabs(t0 + t1));
+  }
+  return sum;
+}
+
+EXPORT void oracle_sum_dna_gamma_save(
+    int tipCase, double *sumtable, const double *x1_start,
+    const double *x2_start, const double *tipVector,
+    const unsigned char *tipX1, const unsigned char *tipX2, int n,
+    const double *x1_gapColumn, const double *x2_gapColumn,
+    const unsigned int *x1_gap, const unsigned int *x2_gap) {
+  int i, j, k;
+  const double *x1_ptr = x1_start, *x2_ptr = x2_start;
+  for (i = 0; i < n; i++) {
+    const double *x1 = NULL, *x2 = NULL;
+    switch (tipCase) {
+    case ORC_TIP_TIP:
+      x1 = &tipVector[4 * tipX1[i]];
+      x2 = &tipVector[4 * tipX2[i]];
+      break;
+    case ORC_TIP_INNER:
+      x1 = &tipVector[4 * tipX1[i]];
+      if (orc_is_gap(x2_gap, i))
+        x2 = x2_gapColumn;
+      else {
+        x2 = x2_ptr;
+        x2_ptr += 16;
+      }
+      break;
+    default:
+      if (orc_is_gap(x1_gap, i))
+        x1 = x1_gapColumn;
+      else {
+        x1 = x1_ptr;
+        x1_ptr += 16;
+      }
+      if (orc_is_gap(x2_gap, i))
+        x2 = x2_gapColumn;
+      else {
+        x2 = x2_ptr;
+        x2_ptr += 16;
+      }
+    }
+    for (j = 0; j < 4; j++)
+      for (k = 0; k < 4; k++)
+        sumtable[i * 16 + j * 4 + k] =
+            (tipCase == ORC_INNER_INNER ? x1[j * 4 + k] : x1[k]) *
+            (tipCase == ORC_TIP_TIP ? x2[k] : x2[j * 4 + k]);
+  }
+}

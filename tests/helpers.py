@@ -504,3 +504,143 @@ class OracleLg4Engine(OracleEngine):
         m = self.model
         return O.core_prot_lg4(self.width, self.sumtable, m.EIGN4,
                                m.gammaRates, m.weights, lz, self.wgt)
+
+
+class OracleSaveEngine(OracleEngine):
+    """CPU -S (saveMemory) DNA engine over the oracle SAVE kernels: gap
+    vectors, gap columns and compacted CLV slabs (the checker for
+    SaveDnaEngine)."""
+
+    def __init__(self, tips, wgt, model):
+        super().__init__(tips, wgt, model)
+        assert model.states == 4
+        self.gvl = self.width // 32 + 1
+        nn = 2 * self.ntips
+        self.gap = np.zeros((nn, self.gvl), dtype=np.uint32)
+        for t in range(1, self.ntips + 1):
+            idx = np.nonzero(tips[t] == 15)[0]
+            np.bitwise_or.at(self.gap[t], idx // 32,
+                             (np.uint32(1) << (idx % 32).astype(np.uint32)))
+        self.gapcol = {}
+
+    def _gapcol_tip(self):
+        _, _, _, tipVector, _ = self._arrays()
+        return np.ascontiguousarray(tipVector[15 * 4:16 * 4])
+
+    def newview_traversal(self, entries):
+        EIGN, EV, EI, tipVector, g = self._arrays()
+        import ctypes as C
+        dp = lambda a: (a.ctypes.data_as(C.POINTER(C.c_double))
+                        if a is not None else C.cast(None,
+                                                     C.POINTER(C.c_double)))
+        up = lambda a: (a.ctypes.data_as(C.POINTER(C.c_uint))
+                        if a is not None else C.cast(None,
+                                                     C.POINTER(C.c_uint)))
+        u8 = lambda a: (a.ctypes.data_as(C.POINTER(C.c_ubyte))
+                        if a is not None else C.cast(None,
+                                                     C.POINTER(C.c_ubyte)))
+        for e in entries:
+            qz = math.log(e.qz) if e.qz > O.ZMIN else math.log(O.ZMIN)
+            rz = math.log(e.rz) if e.rz > O.ZMIN else math.log(O.ZMIN)
+            left, right = O.make_p(qz, rz, g, EI, EIGN, 4, 4)
+            p, q, r = e.pNumber, e.qNumber, e.rNumber
+            self.gap[p] = self.gap[q] & self.gap[r]
+            nz = self.width - int(
+                np.unpackbits(self.gap[p].view(np.uint8),
+                              bitorder="little")[:self.width].sum())
+            x3 = O.aligned(max(nz, 1) * 16)
+            gcol3 = O.aligned(16)
+            q_tip = e.tipCase != INNER_INNER
+            r_tip = e.tipCase == TIP_TIP
+            gc1 = self._gapcol_tip() if q_tip else self.gapcol[e.x1Slot]
+            gc2 = self._gapcol_tip() if r_tip else self.gapcol[e.x2Slot]
+            inc = C.c_int(0)
+            O._orc.oracle_newview_dna_gamma_save(
+                C.c_int(e.tipCase),
+                dp(None if q_tip else self.clv[e.x1Slot]),
+                dp(None if r_tip else self.clv[e.x2Slot]),
+                dp(x3), dp(EV), dp(tipVector),
+                u8(np.ascontiguousarray(self.tips[e.x1Slot])
+                   if q_tip else None),
+                u8(np.ascontiguousarray(self.tips[e.x2Slot])
+                   if r_tip else None),
+                C.c_int(self.width), dp(left), dp(right),
+                self.wgt.ctypes.data_as(C.POINTER(C.c_int)),
+                C.byref(inc), up(self.gap[q]), up(self.gap[r]),
+                up(self.gap[p]), dp(gc1), dp(gc2), dp(gcol3))
+            self.clv[e.x3Slot] = x3
+            self.gapcol[e.x3Slot] = gcol3
+            self.scalers[p] = self.scalers[q] + self.scalers[r] + inc.value
+
+    def evaluate_root(self, tree, p, q, z):
+        EIGN, EV, EI, tipVector, g = self._arrays()
+        import ctypes as C
+        diag = O.calc_diagptable(z, 4, 4, g, EIGN)
+        dp = lambda a: (a.ctypes.data_as(C.POINTER(C.c_double))
+                        if a is not None else C.cast(None,
+                                                     C.POINTER(C.c_double)))
+        up = lambda a: (a.ctypes.data_as(C.POINTER(C.c_uint))
+                        if a is not None else C.cast(None,
+                                                     C.POINTER(C.c_uint)))
+        O._orc.oracle_evaluate_dna_gamma_save.restype = C.c_double
+        p_tip, q_tip = tree.is_tip(p), tree.is_tip(q)
+        if q_tip or p_tip:
+            tip, inner = (q, p) if q_tip else (p, q)
+            sl = tree.clv_slot(inner)
+            lnl = O._orc.oracle_evaluate_dna_gamma_save(
+                self.wgt.ctypes.data_as(C.POINTER(C.c_int)), dp(None),
+                dp(self.clv[sl]), dp(tipVector),
+                np.ascontiguousarray(self.tips[tip]).ctypes.data_as(
+                    C.POINTER(C.c_ubyte)),
+                C.c_int(self.width), dp(diag), dp(None),
+                dp(self.gapcol[sl]), up(None), up(self.gap[inner]))
+        else:
+            s1, s2 = tree.clv_slot(p), tree.clv_slot(q)
+            lnl = O._orc.oracle_evaluate_dna_gamma_save(
+                self.wgt.ctypes.data_as(C.POINTER(C.c_int)),
+                dp(self.clv[s1]), dp(self.clv[s2]), dp(tipVector),
+                C.cast(None, C.POINTER(C.c_ubyte)), C.c_int(self.width),
+                dp(diag), dp(self.gapcol[s1]), dp(self.gapcol[s2]),
+                up(self.gap[p]), up(self.gap[q]))
+        lnl += float(self.scalers[p] + self.scalers[q]) * \
+            math.log(O.MINLIKELIHOOD)
+        return lnl
+
+    def sum_root(self, tree, p, q):
+        EIGN, EV, EI, tipVector, g = self._arrays()
+        import ctypes as C
+        dp = lambda a: (a.ctypes.data_as(C.POINTER(C.c_double))
+                        if a is not None else C.cast(None,
+                                                     C.POINTER(C.c_double)))
+        up = lambda a: (a.ctypes.data_as(C.POINTER(C.c_uint))
+                        if a is not None else C.cast(None,
+                                                     C.POINTER(C.c_uint)))
+        u8 = lambda a: (a.ctypes.data_as(C.POINTER(C.c_ubyte))
+                        if a is not None else C.cast(None,
+                                                     C.POINTER(C.c_ubyte)))
+        self.sumtable = O.aligned(self.width * 16)
+        p_tip, q_tip = tree.is_tip(p), tree.is_tip(q)
+        if p_tip and q_tip:
+            O._orc.oracle_sum_dna_gamma_save(
+                0, dp(self.sumtable), dp(None), dp(None), dp(tipVector),
+                u8(np.ascontiguousarray(self.tips[p])),
+                u8(np.ascontiguousarray(self.tips[q])),
+                C.c_int(self.width), dp(None), dp(None), up(None), up(None))
+        elif q_tip or p_tip:
+            tip, inner = (q, p) if q_tip else (p, q)
+            sl = tree.clv_slot(inner)
+            O._orc.oracle_sum_dna_gamma_save(
+                1, dp(self.sumtable), dp(None), dp(self.clv[sl]),
+                dp(tipVector), u8(np.ascontiguousarray(self.tips[tip])),
+                u8(None), C.c_int(self.width), dp(None),
+                dp(self.gapcol[sl]), up(None), up(self.gap[inner]))
+        else:
+            s1, s2 = tree.clv_slot(p), tree.clv_slot(q)
+            O._orc.oracle_sum_dna_gamma_save(
+                2, dp(self.sumtable), dp(self.clv[s1]), dp(self.clv[s2]),
+                dp(tipVector), u8(None), u8(None), C.c_int(self.width),
+                dp(self.gapcol[s1]), dp(self.gapcol[s2]), up(self.gap[p]),
+                up(self.gap[q]))
+
+    def clv_bytes(self):
+        return sum(v.nbytes for v in self.clv.values())
