@@ -3099,6 +3099,9 @@ __global__ __launch_bounds__(NV_BLOCK) void k_newview_dna_save(
   }
 }
 
+/* same per-site summation order as the dense k_evaluate_dna_gamma (one
+ * lane per (site,cat), pairwise cat combine) so -S stays bit-transparent
+ * against the dense GPU path */
 template <bool TIP>
 __global__ __launch_bounds__(NV_BLOCK) void k_evaluate_dna_save(
     const double *__restrict__ x1, const double *__restrict__ x2,
@@ -3114,28 +3117,30 @@ __global__ __launch_bounds__(NV_BLOCK) void k_evaluate_dna_save(
   if (TIP && tid < 64) sTV[tid] = tipVec[tid];
   __syncthreads();
   const int lane = tid & 63;
+  const long units = n * 4;
   double acc = 0.0;
-  for (long i = (long)blockIdx.x * NV_BLOCK + tid; i < n;
-       i += (long)gridDim.x * NV_BLOCK) {
+  for (long idx = (long)blockIdx.x * NV_BLOCK + tid; idx < units;
+       idx += (long)gridDim.x * NV_BLOCK) {
+    const long i = idx >> 2;
+    const int cat = (int)(idx & 3);
     const double *le, *ri;
     if (TIP)
       le = &sTV[4 * tipX1[i]];
-    else
+    else {
       le = ((g1[i / 32] >> (i % 32)) & 1u)
-               ? x1_gapcol
-               : &x1[save_cidx(g1, pre1, i) * 16];
-    ri = ((g2[i / 32] >> (i % 32)) & 1u)
-             ? x2_gapcol
-             : &x2[save_cidx(g2, pre2, i) * 16];
-    double t0 = 0, t1 = 0;
-    for (int j = 0; j < 4; j++) {
-      const double *l = TIP ? le : &le[j * 4];
-      t0 += l[0] * ri[j * 4 + 0] * sD[j * 4 + 0];
-      t1 += l[1] * ri[j * 4 + 1] * sD[j * 4 + 1];
-      t0 += l[2] * ri[j * 4 + 2] * sD[j * 4 + 2];
-      t1 += l[3] * ri[j * 4 + 3] * sD[j * 4 + 3];
+               ? &x1_gapcol[cat * 4]
+               : &x1[save_cidx(g1, pre1, i) * 16 + cat * 4];
     }
-    acc += (double)wgt[i] * log(0.25 * fabs(t0 + t1));
+    ri = ((g2[i / 32] >> (i % 32)) & 1u)
+             ? &x2_gapcol[cat * 4]
+             : &x2[save_cidx(g2, pre2, i) * 16 + cat * 4];
+    double p = ((le[0] * ri[0]) * sD[cat * 4 + 0] +
+                (le[1] * ri[1]) * sD[cat * 4 + 1]) +
+               ((le[2] * ri[2]) * sD[cat * 4 + 2] +
+                (le[3] * ri[3]) * sD[cat * 4 + 3]);
+    p += __shfl_xor(p, 1);
+    p += __shfl_xor(p, 2);
+    if ((lane & 3) == 0) acc += (double)wgt[i] * log(0.25 * fabs(p));
   }
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off);
@@ -3271,7 +3276,7 @@ extern "C" int examl_hip_evaluate_dna_save(
     double *dev_partials, double *dev_lnl, void *stream) {
   hipStream_t s = (hipStream_t)stream;
   (void)hipGetLastError();
-  const int grid = grid_for(n);
+  const int grid = grid_for(n * 4); /* one lane per (site,cat) */
   const double log_minlik = log(MINLIKELIHOOD);
   const unsigned int *gsP = dev_scalers ? dev_scalers + pNumber : nullptr;
   const unsigned int *gsQ = dev_scalers ? dev_scalers + qNumber : nullptr;
