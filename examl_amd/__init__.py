@@ -104,6 +104,19 @@ def _bind(lib):
         [i, i, i, i, i, p, p, l, p, l, p, l, p]
     lib.examl_hip_core_root_prot_lg4.argtypes = \
         [l, p, p, p, p, d, p, p, p, p, p]
+    # protein CAT (-m PSR on AA)
+    lib.examl_host_evaluate_partial_prot_cat.restype = ctypes.c_double
+    lib.examl_host_evaluate_partial_prot_cat.argtypes = \
+        lib.examl_host_evaluate_partial_dna_cat.argtypes
+    lib.examl_host_core_dtables_prot_cat.argtypes = [p, p, i, d, p]
+    lib.examl_hip_newview_traversal_prot_cat.argtypes = \
+        [p, i, p, p, p, i, p, p, p, p, l, p, l, p, l, p, p, p, p]
+    lib.examl_hip_evaluate_root_prot_cat.argtypes = \
+        [i, i, i, i, i, i, d, p, p, i, p, p, p, l, p, l, p, l, p, p, p, p, p]
+    lib.examl_hip_sum_root_prot_cat.argtypes = \
+        [i, i, i, i, i, p, p, l, p, l, p, l, p]
+    lib.examl_hip_core_root_prot_cat.argtypes = \
+        [l, p, p, p, i, d, p, p, p, p, p, p]
     # -S (saveMemory) DNA kernels
     lib.examl_hip_gap_and_prefix.argtypes = [p, p, p, p, i, l, p]
     lib.examl_hip_newview_dna_save.argtypes = \
@@ -165,12 +178,13 @@ def check(rc, what):
 from .model import DnaGtrModel, Lg4Model, ProtGtrModel  # noqa: E402
 from .tree import PhyloTree             # noqa: E402
 from .engine import (DnaCatEngine, DnaGammaEngine, Lg4Engine,  # noqa: E402
-                     SaveDnaEngine)
+                     ProtCatEngine, SaveDnaEngine)
 
 __all__ = [
     "lib", "check", "TravEntry", "DnaGtrModel", "ProtGtrModel",
     "Lg4Model", "PhyloTree",
     "DnaGammaEngine", "DnaCatEngine", "Lg4Engine", "SaveDnaEngine",
+    "ProtCatEngine",
     "TIP_TIP", "TIP_INNER",
     "INNER_INNER", "ZMIN", "ZMAX",
 ]

@@ -3331,3 +3331,487 @@ extern "C" int examl_hip_sum_dna_save(
 }
 
 #undef CHK
+
+/* ===========================================================================
+ * Protein CAT (PSR) kernels — span 20, per-site rate category.  newview in
+ * the AVX 4-lane dot order (newviewGTRCATPROT_AVX, avxLikelihood.c:487);
+ * evaluate/sum/core in the SSE generics' even/odd order
+ * (evaluateGTRCATPROT:1464, sumGTRCATPROT:2156, coreGTRCATPROT:2659).
+ * P pairs per op are numCats*400 each and stay in global memory (25 cats
+ * would need 160 KB of LDS); EV/tipVector are LDS-staged.
+ * ==========================================================================*/
+
+template <int TC>
+__global__ __launch_bounds__(NV_BLOCK) void k_newview_prot_cat(
+    const double *__restrict__ EV, const int *__restrict__ cptr,
+    const double *__restrict__ x1, const double *__restrict__ x2,
+    double *__restrict__ x3, const double *__restrict__ tipVec,
+    const unsigned char *__restrict__ tipX1,
+    const unsigned char *__restrict__ tipX2, const int *__restrict__ wgt,
+    long n, const double *__restrict__ P, int numCats,
+    unsigned int *__restrict__ scalerInc) {
+  __shared__ double sEV[400], sTV[TC != EXAML_INNER_INNER ? 460 : 1];
+  const int tid = threadIdx.x;
+  for (int j = tid; j < 400; j += NV_BLOCK) sEV[j] = EV[j];
+  if (TC != EXAML_INNER_INNER)
+    for (int j = tid; j < 460; j += NV_BLOCK) sTV[j] = tipVec[j];
+  __syncthreads();
+  const double *R = P + (long)numCats * 400;
+  for (long i = (long)blockIdx.x * NV_BLOCK + tid; i < n;
+       i += (long)gridDim.x * NV_BLOCK) {
+    const int cat = cptr[i];
+    const double *le = &P[(long)cat * 400];
+    const double *ri = &R[(long)cat * 400];
+    const double *vl, *vr;
+    if (TC == EXAML_TIP_TIP) {
+      vl = &sTV[20 * tipX1[i]];
+      vr = &sTV[20 * tipX2[i]];
+    } else if (TC == EXAML_TIP_INNER) {
+      vl = &sTV[20 * tipX1[i]];
+      vr = &x2[i * 20];
+    } else {
+      vl = &x1[i * 20];
+      vr = &x2[i * 20];
+    }
+    double xv[20];
+#pragma unroll
+    for (int s = 0; s < 20; s++) xv[s] = 0.0;
+    for (int l = 0; l < 20; l++) {
+      const double t = dot20o<false>(vl, &le[l * 20]) *
+                       dot20o<false>(vr, &ri[l * 20]);
+#pragma unroll
+      for (int s = 0; s < 20; s++) xv[s] += t * sEV[l * 20 + s];
+    }
+    if (TC != EXAML_TIP_TIP) {
+      bool small = true;
+#pragma unroll
+      for (int s = 0; s < 20; s++)
+        small &= (fabs(xv[s]) < MINLIKELIHOOD);
+      if (small) {
+#pragma unroll
+        for (int s = 0; s < 20; s++) xv[s] *= TWOTOTHE256;
+        atomicAdd(scalerInc, (unsigned int)wgt[i]);
+      }
+    }
+#pragma unroll
+    for (int s = 0; s < 20; s += 4)
+      *reinterpret_cast<double4 *>(&x3[i * 20 + s]) =
+          make_double4(xv[s], xv[s + 1], xv[s + 2], xv[s + 3]);
+  }
+}
+
+template <bool TIP>
+__global__ __launch_bounds__(NV_BLOCK) void k_evaluate_prot_cat(
+    const int *__restrict__ cptr, const int *__restrict__ wgt,
+    const double *__restrict__ x1, const double *__restrict__ x2,
+    const double *__restrict__ tipVec, const unsigned char *__restrict__ tipX1,
+    long n, const double *__restrict__ diag, int numCats,
+    double *__restrict__ partials) {
+  __shared__ double sD[MAX_CAT * 20], sTV[TIP ? 460 : 1],
+      sRed[NV_BLOCK / 64];
+  const int tid = threadIdx.x;
+  for (int j = tid; j < numCats * 20; j += NV_BLOCK) sD[j] = diag[j];
+  if (TIP)
+    for (int j = tid; j < 460; j += NV_BLOCK) sTV[j] = tipVec[j];
+  __syncthreads();
+  const int lane = tid & 63;
+  double acc = 0.0;
+  for (long i = (long)blockIdx.x * NV_BLOCK + tid; i < n;
+       i += (long)gridDim.x * NV_BLOCK) {
+    const double *le = TIP ? &sTV[20 * tipX1[i]] : &x1[i * 20];
+    const double *ri = &x2[i * 20];
+    const double *d = &sD[20 * cptr[i]];
+    double t0 = 0, t1 = 0;
+#pragma unroll
+    for (int l = 0; l < 20; l += 2) {
+      t0 += le[l] * ri[l] * d[l];
+      t1 += le[l + 1] * ri[l + 1] * d[l + 1];
+    }
+    acc += (double)wgt[i] * log(fabs(t0 + t1));
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off);
+  if (lane == 0) sRed[tid >> 6] = acc;
+  __syncthreads();
+  if (tid == 0) {
+    double s = 0;
+#pragma unroll
+    for (int w = 0; w < NV_BLOCK / 64; w++) s += sRed[w];
+    partials[blockIdx.x] = s;
+  }
+}
+
+template <int TC>
+__global__ __launch_bounds__(NV_BLOCK) void k_sum_prot_cat(
+    double *__restrict__ sum, const double *__restrict__ x1,
+    const double *__restrict__ x2, const double *__restrict__ tipVec,
+    const unsigned char *__restrict__ tipX1,
+    const unsigned char *__restrict__ tipX2, long n) {
+  __shared__ double sTV[TC != EXAML_INNER_INNER ? 460 : 1];
+  const int tid = threadIdx.x;
+  if (TC != EXAML_INNER_INNER) {
+    for (int j = tid; j < 460; j += NV_BLOCK) sTV[j] = tipVec[j];
+    __syncthreads();
+  }
+  for (long i = (long)blockIdx.x * NV_BLOCK + tid; i < n;
+       i += (long)gridDim.x * NV_BLOCK) {
+    const double *le, *ri;
+    if (TC == EXAML_TIP_TIP) {
+      le = &sTV[20 * tipX1[i]];
+      ri = &sTV[20 * tipX2[i]];
+    } else if (TC == EXAML_TIP_INNER) {
+      le = &sTV[20 * tipX1[i]];
+      ri = &x2[i * 20];
+    } else {
+      le = &x1[i * 20];
+      ri = &x2[i * 20];
+    }
+#pragma unroll
+    for (int l = 0; l < 20; l += 2)
+      *reinterpret_cast<double2 *>(&sum[i * 20 + l]) =
+          make_double2(le[l] * ri[l], le[l + 1] * ri[l + 1]);
+  }
+}
+
+/* dtab = d[numCats*20] | s[20] | e[20]; rW = perSiteRates[numCats] */
+__global__ __launch_bounds__(NV_BLOCK) void k_core_prot_cat(
+    const double *__restrict__ sum, const double *__restrict__ dtab,
+    const int *__restrict__ wgt, const int *__restrict__ cptr, int numCats,
+    long n, double *__restrict__ partials) {
+  __shared__ double sD[MAX_CAT * 20], sS[20], sE[20], sRp[MAX_CAT],
+      sRed[2][NV_BLOCK / 64];
+  const int tid = threadIdx.x;
+  for (int j = tid; j < numCats * 20; j += NV_BLOCK) sD[j] = dtab[j];
+  if (tid < 20) {
+    sS[tid] = dtab[numCats * 20 + tid];
+    sE[tid] = dtab[numCats * 20 + 20 + tid];
+  }
+  if (tid < numCats) sRp[tid] = dtab[numCats * 20 + 40 + tid];
+  __syncthreads();
+  const int lane = tid & 63;
+  double accD1 = 0.0, accD2 = 0.0;
+  for (long i = (long)blockIdx.x * NV_BLOCK + tid; i < n;
+       i += (long)gridDim.x * NV_BLOCK) {
+    const int cat = cptr[i];
+    const double r = sRp[cat];
+    const double wr1 = r * wgt[i], wr2 = r * r * wgt[i];
+    const double *d = &sD[20 * cat];
+    const double *s = &sum[i * 20];
+    double a0e = 0, a0o = 0, a1e = 0, a1o = 0, a2e = 0, a2o = 0;
+#pragma unroll
+    for (int l = 0; l < 20; l += 2) {
+      const double te = d[l] * s[l];
+      const double to = d[l + 1] * s[l + 1];
+      a0e += te;
+      a0o += to;
+      a1e += te * sS[l];
+      a1o += to * sS[l + 1];
+      a2e += te * sE[l];
+      a2o += to * sE[l + 1];
+    }
+    const double inv = 1.0 / fabs(a0e + a0o);
+    const double d1 = (a1e + a1o) * inv, d2 = (a2e + a2o) * inv;
+    accD1 += wr1 * d1;
+    accD2 += wr2 * (d2 - d1 * d1);
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    accD1 += __shfl_down(accD1, off);
+    accD2 += __shfl_down(accD2, off);
+  }
+  if (lane == 0) {
+    sRed[0][tid >> 6] = accD1;
+    sRed[1][tid >> 6] = accD2;
+  }
+  __syncthreads();
+  if (tid == 0) {
+    double s1 = 0, s2 = 0;
+#pragma unroll
+    for (int w = 0; w < NV_BLOCK / 64; w++) {
+      s1 += sRed[0][w];
+      s2 += sRed[1][w];
+    }
+    partials[blockIdx.x * 2] = s1;
+    partials[blockIdx.x * 2 + 1] = s2;
+  }
+}
+
+#define CHK(call)                                                              \
+  do {                                                                         \
+    hipError_t _e = (call);                                                    \
+    if (_e != hipSuccess) return set_err(_e, #call);                           \
+  } while (0)
+
+extern "C" void examl_host_core_dtables_prot_cat(const double *EIGN,
+                                                 const double *rptr,
+                                                 int numCats, double lz,
+                                                 double *dtab);
+
+extern "C" int examl_hip_newview_traversal_prot_cat(
+    const examl_hip_trav_entry *ops, int numOps, const double *EIGN,
+    const double *EI, const double *perSiteRates, int numCats,
+    const double *dev_EV, const double *dev_tipVec, const int *dev_cptr,
+    double *dev_clv, long clvStride, const unsigned char *dev_tips,
+    long tipStride, const int *dev_wgt, long n, unsigned int *dev_scalers,
+    unsigned int *dev_inc, double *dev_pbuf, void *stream) {
+  if (numOps <= 0 || n <= 0) return 0;
+  if (numCats > MAX_CAT) {
+    snprintf(g_err, sizeof(g_err), "traversal_prot_cat: numCats %d > %d",
+             numCats, MAX_CAT);
+    return -1;
+  }
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  const long PBLK = (long)numCats * 800; /* left|right, numCats x 400 */
+  HostPSlot *pslot = hostP_get(dev_pbuf, (size_t)numOps * PBLK);
+  double *hostP = pslot->buf;
+  for (int e = 0; e < numOps; e++) {
+    double qz = ops[e].qz, rz = ops[e].rz;
+    qz = (qz > ZMIN) ? log(qz) : log(ZMIN);
+    rz = (rz > ZMIN) ? log(rz) : log(ZMIN);
+    examl_host_make_p(qz, rz, perSiteRates, EI, EIGN, numCats,
+                      &hostP[e * PBLK], &hostP[e * PBLK + PBLK / 2], 20);
+  }
+
+  const bool want_graph =
+      g_use_graphs && !g_prof_on && s != nullptr && numOps >= 8;
+  unsigned long long key = 0;
+  bool capturing = false;
+  if (want_graph) {
+    key = trav_key(ops, numOps, n, dev_clv, dev_tips, dev_pbuf, dev_EV,
+                   dev_tipVec, dev_wgt, dev_scalers, dev_inc, (void *)s,
+                   3000 + numCats /* prot CAT tag */);
+    hipGraphExec_t exec = trav_graph_find(key);
+    if (exec) {
+      CHK(hipGraphLaunch(exec, s));
+      return 0;
+    }
+    capturing =
+        hipStreamBeginCapture(s, hipStreamCaptureModeThreadLocal) ==
+        hipSuccess;
+    (void)hipGetLastError();
+  }
+
+  int rc = 0;
+  do {
+    hipError_t err = hipMemcpyAsync(dev_pbuf, hostP,
+                                    (size_t)numOps * PBLK * sizeof(double),
+                                    hipMemcpyHostToDevice, s);
+    if (err != hipSuccess) { rc = set_err(err, "prot cat pbuf"); break; }
+    hipEventRecord(pslot->ev, s);
+    pslot->ev_valid = true;
+    err = hipMemsetAsync(dev_inc, 0, (size_t)numOps * sizeof(unsigned int),
+                         s);
+    if (err != hipSuccess) { rc = set_err(err, "prot cat inc"); break; }
+    const int grid = grid_for(n);
+    for (int e = 0; e < numOps && rc == 0; e++) {
+      const examl_hip_trav_entry *op = &ops[e];
+      hipEvent_t ev_a = nullptr, ev_b = nullptr;
+      if (g_prof_on) {
+        prof_begin(&ev_a, &ev_b);
+        hipEventRecord(ev_a, s);
+      }
+      const double *P = dev_pbuf + (long)e * PBLK;
+      double *x3 = dev_clv + (long)op->x3Slot * clvStride;
+      const double *x1 = nullptr, *x2 = nullptr;
+      const unsigned char *t1 = nullptr, *t2 = nullptr;
+      switch (op->tipCase) {
+      case EXAML_TIP_TIP:
+        t1 = dev_tips + (long)op->x1Slot * tipStride;
+        t2 = dev_tips + (long)op->x2Slot * tipStride;
+        hipLaunchKernelGGL((k_newview_prot_cat<EXAML_TIP_TIP>), dim3(grid),
+                           dim3(NV_BLOCK), 0, s, dev_EV, dev_cptr, x1, x2,
+                           x3, dev_tipVec, t1, t2, dev_wgt, n, P, numCats,
+                           dev_inc + e);
+        break;
+      case EXAML_TIP_INNER:
+        t1 = dev_tips + (long)op->x1Slot * tipStride;
+        x2 = dev_clv + (long)op->x2Slot * clvStride;
+        hipLaunchKernelGGL((k_newview_prot_cat<EXAML_TIP_INNER>), dim3(grid),
+                           dim3(NV_BLOCK), 0, s, dev_EV, dev_cptr, x1, x2,
+                           x3, dev_tipVec, t1, t2, dev_wgt, n, P, numCats,
+                           dev_inc + e);
+        break;
+      case EXAML_INNER_INNER:
+        x1 = dev_clv + (long)op->x1Slot * clvStride;
+        x2 = dev_clv + (long)op->x2Slot * clvStride;
+        hipLaunchKernelGGL((k_newview_prot_cat<EXAML_INNER_INNER>),
+                           dim3(grid), dim3(NV_BLOCK), 0, s, dev_EV,
+                           dev_cptr, x1, x2, x3, dev_tipVec, t1, t2,
+                           dev_wgt, n, P, numCats, dev_inc + e);
+        break;
+      default:
+        snprintf(g_err, sizeof(g_err), "prot cat traversal: bad tipCase %d",
+                 op->tipCase);
+        rc = -1;
+        break;
+      }
+      if (rc == 0) {
+        err = hipGetLastError();
+        if (err != hipSuccess) { rc = set_err(err, "prot cat launch"); break; }
+      }
+      if (g_prof_on) {
+        hipEventRecord(ev_b, s);
+        g_prof_pend.push_back({ev_a, ev_b, op->tipCase});
+        if (g_prof_pend.size() > 2048) prof_flush();
+      }
+    }
+    if (rc != 0) break;
+    for (int base = 0; base < numOps && rc == 0; base += FIN_CHUNK) {
+      FinMeta m;
+      m.count = (numOps - base < FIN_CHUNK) ? (numOps - base) : FIN_CHUNK;
+      m.base = base;
+      for (int e = 0; e < m.count; e++) {
+        m.p[e] = ops[base + e].pNumber;
+        m.q[e] = ops[base + e].qNumber;
+        m.r[e] = ops[base + e].rNumber;
+      }
+      hipLaunchKernelGGL(k_scaler_finalize, dim3(1), dim3(64), 0, s, m,
+                         dev_inc, dev_scalers);
+      err = hipGetLastError();
+      if (err != hipSuccess) rc = set_err(err, "prot cat finalize");
+    }
+  } while (0);
+
+  if (capturing) {
+    hipGraph_t graph = nullptr;
+    hipError_t err = hipStreamEndCapture(s, &graph);
+    if (rc != 0) {
+      if (graph) hipGraphDestroy(graph);
+      return rc;
+    }
+    hipGraphExec_t exec = nullptr;
+    if (err == hipSuccess) {
+      err = hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0);
+      hipGraphDestroy(graph);
+    }
+    if (err != hipSuccess) {
+      (void)hipGetLastError();
+      g_use_graphs = false;
+      return examl_hip_newview_traversal_prot_cat(
+          ops, numOps, EIGN, EI, perSiteRates, numCats, dev_EV, dev_tipVec,
+          dev_cptr, dev_clv, clvStride, dev_tips, tipStride, dev_wgt, n,
+          dev_scalers, dev_inc, dev_pbuf, stream);
+    }
+    trav_graph_store(key, exec);
+    CHK(hipGraphLaunch(exec, s));
+  }
+  return rc;
+}
+
+extern "C" int examl_hip_evaluate_root_prot_cat(
+    int rootTipCase, int pNumber, int qNumber, int x1Slot, int x2Slot,
+    int tipSlot, double z, const double *EIGN, const double *perSiteRates,
+    int numCats, const double *dev_tipVec, const int *dev_cptr,
+    double *dev_clv, long clvStride, const unsigned char *dev_tips,
+    long tipStride, const int *dev_wgt, long n,
+    const unsigned int *dev_scalers, double *dev_diag, double *dev_partials,
+    double *dev_lnl, void *stream) {
+  if (n <= 0) return 0;
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  double hostDiag[MAX_CAT * 20];
+  examl_host_calc_diagptable(z, 20, numCats, perSiteRates, EIGN, hostDiag);
+  CHK(hipMemcpyAsync(dev_diag, hostDiag,
+                     (size_t)numCats * 20 * sizeof(double),
+                     hipMemcpyHostToDevice, s));
+  const double log_minlik = log(MINLIKELIHOOD);
+  const unsigned int *gsP = dev_scalers ? dev_scalers + pNumber : nullptr;
+  const unsigned int *gsQ = dev_scalers ? dev_scalers + qNumber : nullptr;
+  const int grid = grid_for(n);
+  if (rootTipCase == EXAML_TIP_INNER) {
+    const unsigned char *t1 = dev_tips + (long)tipSlot * tipStride;
+    const double *x2 = dev_clv + (long)x2Slot * clvStride;
+    hipLaunchKernelGGL((k_evaluate_prot_cat<true>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, dev_cptr, dev_wgt, nullptr, x2,
+                       dev_tipVec, t1, n, dev_diag, numCats, dev_partials);
+  } else if (rootTipCase == EXAML_INNER_INNER) {
+    const double *x1 = dev_clv + (long)x1Slot * clvStride;
+    const double *x2 = dev_clv + (long)x2Slot * clvStride;
+    hipLaunchKernelGGL((k_evaluate_prot_cat<false>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, dev_cptr, dev_wgt, x1, x2,
+                       dev_tipVec, nullptr, n, dev_diag, numCats,
+                       dev_partials);
+  } else {
+    snprintf(g_err, sizeof(g_err), "evaluate_root_prot_cat: bad tipCase %d",
+             rootTipCase);
+    return -1;
+  }
+  CHK(hipGetLastError());
+  hipLaunchKernelGGL(k_reduce_lnl, dim3(1), dim3(NV_BLOCK), 0, s,
+                     dev_partials, grid, gsP, gsQ, log_minlik, dev_lnl);
+  CHK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int examl_hip_sum_root_prot_cat(
+    int rootTipCase, int x1Slot, int x2Slot, int tipSlot, int tipSlot2,
+    const double *dev_tipVec, double *dev_clv, long clvStride,
+    const unsigned char *dev_tips, long tipStride, double *dev_sum, long n,
+    void *stream) {
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  const double *x1 = nullptr, *x2 = nullptr;
+  const unsigned char *t1 = nullptr, *t2 = nullptr;
+  const int grid = grid_for(n);
+  switch (rootTipCase) {
+  case EXAML_TIP_TIP:
+    t1 = dev_tips + (long)tipSlot * tipStride;
+    t2 = dev_tips + (long)tipSlot2 * tipStride;
+    hipLaunchKernelGGL((k_sum_prot_cat<EXAML_TIP_TIP>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, dev_sum, x1, x2, dev_tipVec,
+                       t1, t2, n);
+    break;
+  case EXAML_TIP_INNER:
+    t1 = dev_tips + (long)tipSlot * tipStride;
+    x2 = dev_clv + (long)x2Slot * clvStride;
+    hipLaunchKernelGGL((k_sum_prot_cat<EXAML_TIP_INNER>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, dev_sum, x1, x2, dev_tipVec,
+                       t1, t2, n);
+    break;
+  case EXAML_INNER_INNER:
+    x1 = dev_clv + (long)x1Slot * clvStride;
+    x2 = dev_clv + (long)x2Slot * clvStride;
+    hipLaunchKernelGGL((k_sum_prot_cat<EXAML_INNER_INNER>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, dev_sum, x1, x2, dev_tipVec,
+                       t1, t2, n);
+    break;
+  default:
+    snprintf(g_err, sizeof(g_err), "sum_root_prot_cat: bad tipCase %d",
+             rootTipCase);
+    return -1;
+  }
+  CHK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int examl_hip_core_root_prot_cat(
+    long n, const double *dev_sum, const double *EIGN, const double *rptr,
+    int numCats, double lz, const int *dev_wgt, const int *dev_cptr,
+    double *dev_dtab, double *dev_partials, double *dev_out2,
+    void *stream) {
+  if (n <= 0) return 0;
+  if (numCats > MAX_CAT) {
+    snprintf(g_err, sizeof(g_err), "core_prot_cat: numCats %d > %d", numCats,
+             MAX_CAT);
+    return -1;
+  }
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  double host[MAX_CAT * 20 + 40 + MAX_CAT];
+  examl_host_core_dtables_prot_cat(EIGN, rptr, numCats, lz, host);
+  CHK(hipMemcpyAsync(dev_dtab, host,
+                     (size_t)(numCats * 20 + 40 + numCats) * sizeof(double),
+                     hipMemcpyHostToDevice, s));
+  const int grid = grid_for(n);
+  hipLaunchKernelGGL(k_core_prot_cat, dim3(grid), dim3(NV_BLOCK), 0, s,
+                     dev_sum, dev_dtab, dev_wgt, dev_cptr,
+                     numCats, n, dev_partials);
+  CHK(hipGetLastError());
+  hipLaunchKernelGGL(k_reduce_2, dim3(1), dim3(NV_BLOCK), 0, s, dev_partials,
+                     grid, dev_out2);
+  CHK(hipGetLastError());
+  return 0;
+}
+
+#undef CHK

@@ -510,6 +510,98 @@ extern "C" double examl_host_evaluate_partial_dna_cat(
   return term;
 }
 
+extern "C" double examl_host_evaluate_partial_prot_cat(
+    const void *ops_, int numOps, int rootTipNumber, int rootQNumber,
+    double root_z, long site, double ki, int w, const double *EIGN,
+    const double *EI, const double *EV, const double *tipVector,
+    const unsigned char *tips, long tipStride, int mxtips) {
+  /* computeVectorGTRCATPROT / evaluatePartialGTRCATPROT
+   * (evaluatePartialGenericSpecial.c:493/618), span 20, SSE even/odd ump
+   * sums; the reference's dead zmin clamp restated as-is */
+  struct TE {
+    int tipCase, pNumber, qNumber, rNumber;
+    int x1Slot, x2Slot, x3Slot;
+    double qz, rz;
+  };
+  const TE *ops = (const TE *)ops_;
+  const double ZMIN_ = 1.0E-15;
+  const double TWO256 =
+      115792089237316195423570985008687907853269984665640564039457584007913129639936.0;
+  const double MINLIK = 1.0 / TWO256;
+
+  double *lVector = (double *)malloc(sizeof(double) * 20 * (size_t)mxtips);
+  int scale = 0;
+
+  for (int k = 0; k < numOps; k++) {
+    const TE *t = &ops[k];
+    double qz = t->qz, rz = t->rz;
+    qz = (qz > ZMIN_) ? log(qz) : log(ZMIN_);
+    rz = (rz > ZMIN_) ? log(rz) : log(ZMIN_);
+    const double *x1, *x2;
+    double *x3 = &lVector[20 * (t->pNumber - mxtips)];
+    switch (t->tipCase) {
+    case 0:
+      x1 = &tipVector[20 * tips[(long)t->qNumber * tipStride + site]];
+      x2 = &tipVector[20 * tips[(long)t->rNumber * tipStride + site]];
+      break;
+    case 1:
+      x1 = &tipVector[20 * tips[(long)t->qNumber * tipStride + site]];
+      x2 = &lVector[20 * (t->rNumber - mxtips)];
+      break;
+    default:
+      x1 = &lVector[20 * (t->qNumber - mxtips)];
+      x2 = &lVector[20 * (t->rNumber - mxtips)];
+    }
+    const double lz1 = qz * ki, lz2 = rz * ki;
+    double e1[20], e2[20], d1[20], d2[20];
+    e1[0] = 1.0;
+    e2[0] = 1.0;
+    for (int l = 1; l < 20; l++) {
+      e1[l] = exp(EIGN[l] * lz1);
+      e2[l] = exp(EIGN[l] * lz2);
+    }
+    for (int l = 0; l < 20; l++) {
+      d1[l] = x1[l] * e1[l];
+      d2[l] = x2[l] * e2[l];
+    }
+    for (int l = 0; l < 20; l++) x3[l] = 0.0;
+    for (int l = 0; l < 20; l++) {
+      const double *ev = &EV[l * 20];
+      double u1e = 0, u1o = 0, u2e = 0, u2o = 0;
+      for (int kk = 0; kk < 20; kk += 2) {
+        u1e += d1[kk] * EI[20 * l + kk];
+        u1o += d1[kk + 1] * EI[20 * l + kk + 1];
+        u2e += d2[kk] * EI[20 * l + kk];
+        u2o += d2[kk + 1] * EI[20 * l + kk + 1];
+      }
+      const double x1px2 = (u1e + u1o) * (u2e + u2o);
+      for (int kk = 0; kk < 20; kk++) x3[kk] += x1px2 * ev[kk];
+    }
+    int sc = 1;
+    for (int l = 0; sc && l < 20; l++)
+      sc = (x3[l] < MINLIK && x3[l] > -MINLIK);
+    if (sc) {
+      for (int l = 0; l < 20; l++) x3[l] *= TWO256;
+      scale++;
+    }
+  }
+
+  const double *x1 =
+      &tipVector[20 * tips[(long)rootTipNumber * tipStride + site]];
+  const double *x2 = &lVector[20 * (rootQNumber - mxtips)];
+  double lz = log(root_z); /* the reference's zmin clamp is dead code */
+  lz *= ki;
+  double d[20];
+  d[0] = 1.0;
+  for (int l = 1; l < 20; l++) d[l] = exp(EIGN[l] * lz);
+  double term = 0.0;
+  for (int l = 0; l < 20; l++) term += x1[l] * x2[l] * d[l];
+  term = log(fabs(term)) + (scale * log(MINLIK));
+  term = term * w;
+  free(lVector);
+  return term;
+}
+
 /* =========================================================================
  * LG4 (LG4M/LG4X) host model math: per-category matrices.
  * EIGN4 stride 20 (SCALED, scaleLG4X_EIGN), EI4 stride 400.
@@ -567,5 +659,27 @@ extern "C" void examl_host_core_dtables_prot_lg4(const double *EIGN4,
       d1[i * 20 + l] = EIGN[l] * ki;
       d2[i * 20 + l] = EIGN[l] * EIGN[l] * kisqr;
     }
+  }
+}
+
+/* coreGTRCATPROT d-tables (makenewzGenericSpecial.c:2659):
+ * dtab = d[numCats*20] | s[20] | e[20] | rates[numCats] */
+extern "C" void examl_host_core_dtables_prot_cat(const double *EIGN,
+                                                 const double *rptr,
+                                                 int numCats, double lz,
+                                                 double *dtab) {
+  double *d = dtab, *s_ = dtab + numCats * 20, *e = dtab + numCats * 20 + 20;
+  double *rw = dtab + numCats * 20 + 40;
+  double dd[20];
+  e[0] = s_[0] = dd[0] = 0.0;
+  for (int l = 1; l < 20; l++) {
+    e[l] = EIGN[l] * EIGN[l];
+    s_[l] = EIGN[l];
+    dd[l] = s_[l] * lz;
+  }
+  for (int i = 0; i < numCats; i++) {
+    d[20 * i] = 1.0;
+    for (int l = 1; l < 20; l++) d[20 * i + l] = exp(dd[l] * rptr[i]);
+    rw[i] = rptr[i];
   }
 }

@@ -653,3 +653,101 @@ class SaveDnaEngine(DnaGammaEngine):
     def clv_bytes(self):
         """actual CLV memory footprint (the -S saving)"""
         return sum(t.numel() * 8 for t in self.clv_slots.values())
+
+
+class ProtCatEngine(DnaGammaEngine):
+    """Protein CAT (PSR) engine: span 20, per-site rate categories
+    (newviewGTRCATPROT_AVX family).  Mirrors DnaCatEngine with the
+    20-state kernels; scratch sized for maxCategories=25 up front."""
+
+    def __init__(self, tips, wgt, model, cptr, per_site_rates, device="cuda",
+                 max_ops=None):
+        assert model.states == 20
+        self.cptr = np.ascontiguousarray(cptr, dtype=np.int32)
+        self.per_site_rates = np.ascontiguousarray(per_site_rates,
+                                                   dtype=np.float64)
+        super().__init__(tips, wgt, model, device=device, max_ops=max_ops)
+        self.SPAN = 20
+        self.num_cats = len(self.per_site_rates)
+        dev = self.device
+        self.d_cptr = torch.from_numpy(self.cptr).to(dev)
+        MAXC = 25
+        self.d_clv = torch.empty((self.ninner, self.width * 20),
+                                 dtype=torch.float64, device=dev)
+        self.d_pbuf = torch.empty(self._max_ops * MAXC * 800,
+                                  dtype=torch.float64, device=dev)
+        self.d_diag = torch.empty(MAXC * 20, dtype=torch.float64, device=dev)
+        self.d_dtab = torch.empty(MAXC * 20 + 40 + MAXC,
+                                  dtype=torch.float64, device=dev)
+
+    def set_site_rates(self, cptr, per_site_rates):
+        assert len(per_site_rates) <= 25
+        self.cptr = np.ascontiguousarray(cptr, dtype=np.int32)
+        self.per_site_rates = np.ascontiguousarray(per_site_rates,
+                                                   dtype=np.float64)
+        self.num_cats = len(self.per_site_rates)
+        self.d_cptr.copy_(torch.from_numpy(self.cptr))
+
+    def newview_traversal(self, entries):
+        if not entries:
+            return
+        assert len(entries) <= self._max_ops, "grow max_ops"
+        arr = (TravEntry * len(entries))(*entries)
+        m = self.model
+        check(lib().examl_hip_newview_traversal_prot_cat(
+            ctypes.cast(arr, ctypes.c_void_p), len(entries),
+            _np_vp(m.EIGN), _np_vp(m.EI), _np_vp(self.per_site_rates),
+            self.num_cats, _vp(self.d_EV), _vp(self.d_tipVector),
+            _vp(self.d_cptr), _vp(self.d_clv),
+            ctypes.c_long(self.width * 20), _vp(self.d_tips),
+            ctypes.c_long(self.width), _vp(self.d_wgt),
+            ctypes.c_long(self.width), _vp(self.d_scalers), _vp(self.d_inc),
+            _vp(self.d_pbuf), self._stream()), "newview_traversal_prot_cat")
+
+    def evaluate_root(self, tree, p, q, z, all_reduce=False):
+        tc, x1s, x2s, tslot, _, pn, qn = self._root_case(tree, p, q)
+        self.d_lnl.zero_()
+        m = self.model
+        check(lib().examl_hip_evaluate_root_prot_cat(
+            tc, pn, qn, x1s, x2s, tslot, ctypes.c_double(z),
+            _np_vp(m.EIGN), _np_vp(self.per_site_rates), self.num_cats,
+            _vp(self.d_tipVector), _vp(self.d_cptr), _vp(self.d_clv),
+            ctypes.c_long(self.width * 20), _vp(self.d_tips),
+            ctypes.c_long(self.width), _vp(self.d_wgt),
+            ctypes.c_long(self.width), _vp(self.d_scalers), _vp(self.d_diag),
+            _vp(self.d_partials), _vp(self.d_lnl), self._stream()),
+            "evaluate_root_prot_cat")
+        if all_reduce and torch.distributed.is_initialized():
+            torch.distributed.all_reduce(self.d_lnl)
+        return self.d_lnl
+
+    def sum_root(self, tree, p, q):
+        if self.d_sum is None:
+            self.d_sum = torch.empty(self.width * 20, dtype=torch.float64,
+                                     device=self.device)
+        p_tip, q_tip = tree.is_tip(p), tree.is_tip(q)
+        if p_tip and q_tip:
+            tc, x1s, x2s, t1, t2 = TIP_TIP, -1, -1, p, q
+        elif q_tip:
+            tc, x1s, x2s, t1, t2 = TIP_INNER, -1, tree.clv_slot(p), q, -1
+        elif p_tip:
+            tc, x1s, x2s, t1, t2 = TIP_INNER, -1, tree.clv_slot(q), p, -1
+        else:
+            tc, x1s, x2s, t1, t2 = (INNER_INNER, tree.clv_slot(p),
+                                    tree.clv_slot(q), -1, -1)
+        check(lib().examl_hip_sum_root_prot_cat(
+            tc, x1s, x2s, t1, t2, _vp(self.d_tipVector), _vp(self.d_clv),
+            ctypes.c_long(self.width * 20), _vp(self.d_tips),
+            ctypes.c_long(self.width), _vp(self.d_sum),
+            ctypes.c_long(self.width), self._stream()), "sum_root_prot_cat")
+
+    def core_derivs_async(self, lz):
+        m = self.model
+        self.d_out2.zero_()
+        check(lib().examl_hip_core_root_prot_cat(
+            ctypes.c_long(self.width), _vp(self.d_sum), _np_vp(m.EIGN),
+            _np_vp(self.per_site_rates), self.num_cats, ctypes.c_double(lz),
+            _vp(self.d_wgt), _vp(self.d_cptr), _vp(self.d_dtab),
+            _vp(self.d_partials), _vp(self.d_out2), self._stream()),
+            "core_root_prot_cat")
+        return self.d_out2

@@ -949,7 +949,10 @@ class TreeSearch:
             arr = (TravEntry * len(src))(*src)
             self._epg_ops = (arr, key)
         arr = self._epg_ops[0]
-        return lib().examl_host_evaluate_partial_dna_cat(
+        fn = (lib().examl_host_evaluate_partial_dna_cat
+              if eng.model.states == 4
+              else lib().examl_host_evaluate_partial_prot_cat)
+        return fn(
             ctypes.cast(arr, ctypes.c_void_p), len(entries),
             ctypes.c_int(p), ctypes.c_int(q), ctypes.c_double(z),
             ctypes.c_long(site), ctypes.c_double(ki),

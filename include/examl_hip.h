@@ -464,6 +464,41 @@ int examl_hip_sum_dna_save(
     const unsigned int *g2, const int *pre1, const int *pre2,
     const double *x1_gapcol, const double *x2_gapcol, void *stream);
 
+
+/* ---- Protein CAT (-m PSR on AA partitions) ---------------------------- */
+double examl_host_evaluate_partial_prot_cat(
+    const void *ops, int numOps, int rootTipNumber, int rootQNumber,
+    double root_z, long site, double ki, int w, const double *EIGN,
+    const double *EI, const double *EV, const double *tipVector,
+    const unsigned char *tips, long tipStride, int mxtips);
+void examl_host_core_dtables_prot_cat(const double *EIGN, const double *rptr,
+                                      int numCats, double lz,
+                                      double *dtab /*cats*20+40+cats*/);
+int examl_hip_newview_traversal_prot_cat(
+    const examl_hip_trav_entry *ops, int numOps, const double *EIGN,
+    const double *EI, const double *perSiteRates, int numCats,
+    const double *dev_EV, const double *dev_tipVec, const int *dev_cptr,
+    double *dev_clv, long clvStride, const unsigned char *dev_tips,
+    long tipStride, const int *dev_wgt, long n, unsigned int *dev_scalers,
+    unsigned int *dev_inc, double *dev_pbuf, void *stream);
+int examl_hip_evaluate_root_prot_cat(
+    int rootTipCase, int pNumber, int qNumber, int x1Slot, int x2Slot,
+    int tipSlot, double z, const double *EIGN, const double *perSiteRates,
+    int numCats, const double *dev_tipVec, const int *dev_cptr,
+    double *dev_clv, long clvStride, const unsigned char *dev_tips,
+    long tipStride, const int *dev_wgt, long n,
+    const unsigned int *dev_scalers, double *dev_diag, double *dev_partials,
+    double *dev_lnl, void *stream);
+int examl_hip_sum_root_prot_cat(
+    int rootTipCase, int x1Slot, int x2Slot, int tipSlot, int tipSlot2,
+    const double *dev_tipVec, double *dev_clv, long clvStride,
+    const unsigned char *dev_tips, long tipStride, double *dev_sum, long n,
+    void *stream);
+int examl_hip_core_root_prot_cat(
+    long n, const double *dev_sum, const double *EIGN, const double *rptr,
+    int numCats, double lz, const int *dev_wgt, const int *dev_cptr,
+    double *dev_dtab, double *dev_partials, double *dev_out2, void *stream);
+
 #ifdef __cplusplus
 }
 #endif

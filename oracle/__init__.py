@@ -478,3 +478,63 @@ def core_prot_lg4(n, sumtable, EIGN4, gammaRates, weights, lz, wgt):
         _dp(EIGN4), _dp(gammaRates), _dp(weights), ctypes.c_double(lz),
         _ip(wgt))
     return d1.value, d2.value
+
+
+# ---------------------------------------------------------------------------
+# Protein CAT (PSR on AA data)
+# ---------------------------------------------------------------------------
+
+def newview_prot_cat(tip_case, EV, cptr, x1, x2, tipVector, tipX1, tipX2, n,
+                     left, right, wgt):
+    x3 = aligned(n * 20)
+    inc = ctypes.c_int(0)
+    nullp = ctypes.cast(None, _c_d)
+    nullb = ctypes.cast(None, _c_u8)
+    _orc.oracle_newview_prot_cat(
+        ctypes.c_int(tip_case), _dp(EV), _ip(cptr),
+        _dp(x1) if x1 is not None else nullp,
+        _dp(x2) if x2 is not None else nullp,
+        _dp(x3), _dp(tipVector),
+        _u8p(tipX1) if tipX1 is not None else nullb,
+        _u8p(tipX2) if tipX2 is not None else nullb,
+        ctypes.c_int(n), _dp(left), _dp(right), _ip(wgt),
+        ctypes.byref(inc))
+    return x3, inc.value
+
+
+def evaluate_prot_cat(cptr, wgt, x1, x2, tipVector, tipX1, n, diag):
+    _orc.oracle_evaluate_prot_cat.restype = ctypes.c_double
+    nullp = ctypes.cast(None, _c_d)
+    nullb = ctypes.cast(None, _c_u8)
+    return _orc.oracle_evaluate_prot_cat(
+        _ip(cptr), _ip(wgt),
+        _dp(x1) if x1 is not None else nullp,
+        _dp(x2) if x2 is not None else nullp,
+        _dp(tipVector),
+        _u8p(tipX1) if tipX1 is not None else nullb,
+        ctypes.c_int(n), _dp(diag))
+
+
+def sum_prot_cat(tip_case, x1, x2, tipVector, tipX1, tipX2, n):
+    st = aligned(n * 20)
+    nullp = ctypes.cast(None, _c_d)
+    nullb = ctypes.cast(None, _c_u8)
+    _orc.oracle_sum_prot_cat(
+        ctypes.c_int(tip_case), _dp(st),
+        _dp(x1) if x1 is not None else nullp,
+        _dp(x2) if x2 is not None else nullp,
+        _dp(tipVector),
+        _u8p(tipX1) if tipX1 is not None else nullb,
+        _u8p(tipX2) if tipX2 is not None else nullb,
+        ctypes.c_int(n))
+    return st
+
+
+def core_prot_cat(n, num_cats, sumtable, wgt, rptr, EIGN, cptr, lz):
+    d1 = ctypes.c_double()
+    d2 = ctypes.c_double()
+    _orc.oracle_core_prot_cat(
+        ctypes.c_int(n), ctypes.c_int(num_cats), _dp(sumtable), _ip(wgt),
+        _dp(rptr), _dp(EIGN), _ip(cptr), ctypes.c_double(lz),
+        ctypes.byref(d1), ctypes.byref(d2))
+    return d1.value, d2.value

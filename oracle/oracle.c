@@ -1530,3 +1530,148 @@ EXPORT void oracle_sum_dna_gamma_save(
             (tipCase == ORC_TIP_TIP ? x2[k] : x2[j * 4 + k]);
   }
 }
+
+/* ==========================================================================
+ * Protein CAT (PSR) kernels — span 20, per-site rate category.
+ * Restate newviewGTRCATPROT_AVX (avxLikelihood.c:487, 4-lane dot order),
+ * evaluateGTRCATPROT (evaluateGenericSpecial.c:1464, SSE even/odd,
+ * log|term| without 0.25), sumGTRCATPROT (makenewzGenericSpecial.c:2156)
+ * and coreGTRCATPROT (:2659, wr1/wr2-weighted derivatives).
+ * left/right: [cat*400 + row*20 + col] from the generic makeP.
+ * ==========================================================================*/
+
+EXPORT void oracle_newview_prot_cat(int tipCase, const double *extEV,
+                                    const int *cptr, const double *x1_start,
+                                    const double *x2_start, double *x3_start,
+                                    const double *tipVector,
+                                    const unsigned char *tipX1,
+                                    const unsigned char *tipX2, int n,
+                                    const double *left, const double *right,
+                                    const int *wgt, int *scalerIncrement) {
+  int i, l, s;
+  int addScale = 0;
+  for (i = 0; i < n; i++) {
+    const double *le = &left[cptr[i] * 400];
+    const double *ri = &right[cptr[i] * 400];
+    const double *vl, *vr;
+    double xv[20];
+    int scale;
+    if (tipCase == ORC_TIP_TIP) {
+      vl = &tipVector[20 * tipX1[i]];
+      vr = &tipVector[20 * tipX2[i]];
+    } else if (tipCase == ORC_TIP_INNER) {
+      vl = &tipVector[20 * tipX1[i]];
+      vr = &x2_start[20 * i];
+    } else {
+      vl = &x1_start[20 * i];
+      vr = &x2_start[20 * i];
+    }
+    for (s = 0; s < 20; s++) xv[s] = 0.0;
+    for (l = 0; l < 20; l++) {
+      const double t =
+          dot20_avx(vl, &le[l * 20]) * dot20_avx(vr, &ri[l * 20]);
+      for (s = 0; s < 20; s++) xv[s] += t * extEV[l * 20 + s];
+    }
+    if (tipCase != ORC_TIP_TIP) {
+      scale = 1;
+      for (s = 0; scale && s < 20; s++)
+        scale = (fabs(xv[s]) < ORC_MINLIKELIHOOD);
+      if (scale) {
+        for (s = 0; s < 20; s++) xv[s] *= ORC_TWOTOTHE256;
+        addScale += wgt[i];
+      }
+    }
+    for (s = 0; s < 20; s++) x3_start[20 * i + s] = xv[s];
+  }
+  *scalerIncrement = addScale;
+}
+
+EXPORT double oracle_evaluate_prot_cat(const int *cptr, const int *wptr,
+                                       const double *x1, const double *x2,
+                                       const double *tipVector,
+                                       const unsigned char *tipX1, int n,
+                                       const double *diagptable) {
+  double sum = 0.0;
+  int i, l;
+  for (i = 0; i < n; i++) {
+    const double *le = tipX1 ? &tipVector[20 * tipX1[i]] : &x1[20 * i];
+    const double *ri = &x2[20 * i];
+    const double *d = &diagptable[20 * cptr[i]];
+    double t0 = 0.0, t1 = 0.0;
+    for (l = 0; l < 20; l += 2) {
+      t0 += le[l] * ri[l] * d[l];
+      t1 += le[l + 1] * ri[l + 1] * d[l + 1];
+    }
+    sum += wptr[i] * log(fabs(t0 + t1));
+  }
+  return sum;
+}
+
+EXPORT void oracle_sum_prot_cat(int tipCase, double *sumtable,
+                                const double *x1, const double *x2,
+                                const double *tipVector,
+                                const unsigned char *tipX1,
+                                const unsigned char *tipX2, int n) {
+  int i, l;
+  for (i = 0; i < n; i++) {
+    const double *le, *ri;
+    switch (tipCase) {
+    case ORC_TIP_TIP:
+      le = &tipVector[20 * tipX1[i]];
+      ri = &tipVector[20 * tipX2[i]];
+      break;
+    case ORC_TIP_INNER:
+      le = &tipVector[20 * tipX1[i]];
+      ri = &x2[20 * i];
+      break;
+    default:
+      le = &x1[20 * i];
+      ri = &x2[20 * i];
+    }
+    for (l = 0; l < 20; l++) sumtable[20 * i + l] = le[l] * ri[l];
+  }
+}
+
+EXPORT void oracle_core_prot_cat(int upper, int numberOfCategories,
+                                 const double *sumtable, const int *wgt,
+                                 const double *rptr, const double *EIGN,
+                                 const int *cptr, double lz,
+                                 double *ext_dlnLdlz, double *ext_d2lnLdlz2) {
+  double d_start[25 * 20], e[20], s_[20], dd[20];
+  double dlnLdlz = 0.0, d2lnLdlz2 = 0.0;
+  int i, l;
+  e[0] = s_[0] = dd[0] = 0.0;
+  for (l = 1; l < 20; l++) {
+    e[l] = EIGN[l] * EIGN[l];
+    s_[l] = EIGN[l];
+    dd[l] = s_[l] * lz;
+  }
+  for (i = 0; i < numberOfCategories; i++) {
+    d_start[20 * i] = 1.0;
+    for (l = 1; l < 20; l++) d_start[20 * i + l] = exp(dd[l] * rptr[i]);
+  }
+  for (i = 0; i < upper; i++) {
+    const double r = rptr[cptr[i]];
+    const double wr1 = r * wgt[i], wr2 = r * r * wgt[i];
+    const double *d = &d_start[20 * cptr[i]];
+    const double *sum = &sumtable[20 * i];
+    double a0e = 0, a0o = 0, a1e = 0, a1o = 0, a2e = 0, a2o = 0;
+    for (l = 0; l < 20; l += 2) {
+      const double te = d[l] * sum[l];
+      const double to = d[l + 1] * sum[l + 1];
+      a0e += te;
+      a0o += to;
+      a1e += te * s_[l];
+      a1o += to * s_[l + 1];
+      a2e += te * e[l];
+      a2o += to * e[l + 1];
+    }
+    const double inv_Li = 1.0 / fabs(a0e + a0o);
+    const double dlnLidlz = (a1e + a1o) * inv_Li;
+    const double d2lnLidlz2 = (a2e + a2o) * inv_Li;
+    dlnLdlz += wr1 * dlnLidlz;
+    d2lnLdlz2 += wr2 * (d2lnLidlz2 - dlnLidlz * dlnLidlz);
+  }
+  *ext_dlnLdlz = dlnLdlz;
+  *ext_d2lnLdlz2 = d2lnLdlz2;
+}

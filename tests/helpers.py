@@ -305,6 +305,87 @@ class OracleCatEngine(OracleEngine):
                               self.wgt, self._rptr(), EIGN, self.cptr, lz)
 
 
+class OracleProtCatEngine(OracleCatEngine):
+    """CPU protein CAT (PSR) engine over the oracle prot-CAT kernels."""
+
+    def newview_traversal(self, entries):
+        EIGN, EV, EI, tipVector, _ = self._arrays()
+        rptr = self._rptr()
+        for e in entries:
+            qz = math.log(e.qz) if e.qz > O.ZMIN else math.log(O.ZMIN)
+            rz = math.log(e.rz) if e.rz > O.ZMIN else math.log(O.ZMIN)
+            left, right = O.make_p(qz, rz, rptr, EI, EIGN, self.num_cats, 20)
+            if e.tipCase == TIP_TIP:
+                x3, inc = O.newview_prot_cat(
+                    TIP_TIP, EV, self.cptr, None, None, tipVector,
+                    np.ascontiguousarray(self.tips[e.x1Slot]),
+                    np.ascontiguousarray(self.tips[e.x2Slot]), self.width,
+                    left, right, self.wgt)
+            elif e.tipCase == TIP_INNER:
+                x3, inc = O.newview_prot_cat(
+                    TIP_INNER, EV, self.cptr, None, self.clv[e.x2Slot],
+                    tipVector, np.ascontiguousarray(self.tips[e.x1Slot]),
+                    None, self.width, left, right, self.wgt)
+            else:
+                x3, inc = O.newview_prot_cat(
+                    INNER_INNER, EV, self.cptr, self.clv[e.x1Slot],
+                    self.clv[e.x2Slot], tipVector, None, None, self.width,
+                    left, right, self.wgt)
+            self.clv[e.x3Slot] = x3
+            self.scalers[e.pNumber] = (self.scalers[e.qNumber] +
+                                       self.scalers[e.rNumber] + inc)
+
+    def evaluate_root(self, tree, p, q, z):
+        EIGN, EV, EI, tipVector, _ = self._arrays()
+        diag = O.calc_diagptable(z, 20, self.num_cats, self._rptr(), EIGN)
+        p_tip, q_tip = tree.is_tip(p), tree.is_tip(q)
+        if q_tip:
+            lnl = O.evaluate_prot_cat(
+                self.cptr, self.wgt, None, self.clv[tree.clv_slot(p)],
+                tipVector, np.ascontiguousarray(self.tips[q]), self.width,
+                diag)
+        elif p_tip:
+            lnl = O.evaluate_prot_cat(
+                self.cptr, self.wgt, None, self.clv[tree.clv_slot(q)],
+                tipVector, np.ascontiguousarray(self.tips[p]), self.width,
+                diag)
+        else:
+            lnl = O.evaluate_prot_cat(
+                self.cptr, self.wgt, self.clv[tree.clv_slot(p)],
+                self.clv[tree.clv_slot(q)], tipVector, None, self.width,
+                diag)
+        lnl += float(self.scalers[p] + self.scalers[q]) * \
+            math.log(O.MINLIKELIHOOD)
+        return lnl
+
+    def sum_root(self, tree, p, q):
+        EIGN, EV, EI, tipVector, _ = self._arrays()
+        p_tip, q_tip = tree.is_tip(p), tree.is_tip(q)
+        if p_tip and q_tip:
+            self.sumtable = O.sum_prot_cat(
+                TIP_TIP, None, None, tipVector,
+                np.ascontiguousarray(self.tips[p]),
+                np.ascontiguousarray(self.tips[q]), self.width)
+        elif q_tip:
+            self.sumtable = O.sum_prot_cat(
+                TIP_INNER, None, self.clv[tree.clv_slot(p)], tipVector,
+                np.ascontiguousarray(self.tips[q]), None, self.width)
+        elif p_tip:
+            self.sumtable = O.sum_prot_cat(
+                TIP_INNER, None, self.clv[tree.clv_slot(q)], tipVector,
+                np.ascontiguousarray(self.tips[p]), None, self.width)
+        else:
+            self.sumtable = O.sum_prot_cat(
+                INNER_INNER, self.clv[tree.clv_slot(p)],
+                self.clv[tree.clv_slot(q)], tipVector, None, None,
+                self.width)
+
+    def core_derivs(self, lz):
+        EIGN, EV, EI, tipVector, _ = self._arrays()
+        return O.core_prot_cat(self.width, self.num_cats, self.sumtable,
+                               self.wgt, self._rptr(), EIGN, self.cptr, lz)
+
+
 def oracle_makenewz(entries, root, tree, model, tips, wgt, z0, maxiter=64):
     """CPU restatement of topLevelMakenewz (numBranches=1) over the oracle
     sum/core kernels — the checker for DnaGammaEngine.makenewz."""
