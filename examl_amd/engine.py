@@ -517,6 +517,7 @@ class SaveDnaEngine(DnaGammaEngine):
         m = self.model
         L = lib()
         s = self._stream()
+        self.d_inc.zero_()  # the per-op scaler accumulator starts at 0
         for e in entries:
             qz = math.log(e.qz) if e.qz > ZMIN else math.log(ZMIN)
             rz = math.log(e.rz) if e.rz > ZMIN else math.log(ZMIN)

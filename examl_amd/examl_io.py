@@ -95,13 +95,26 @@ def read_byte_file(path):
     return taxa, parts
 
 
+def read_newick_trees(path, taxa_names):
+    """All trees in the file (one per line — the -f E/-f e multi-tree
+    input, getNumberOfTrees/optimizeTrees)."""
+    with open(path) as f:
+        lines = [ln.strip() for ln in f if ln.strip()]
+    return [parse_newick_topology(ln, taxa_names) for ln in lines]
+
+
 def read_newick_topology(path, taxa_names):
+    with open(path) as f:
+        s = f.read().strip().splitlines()[0]
+    return parse_newick_topology(s, taxa_names)
+
+
+def parse_newick_topology(s, taxa_names):
     """Parse a (possibly multifurcating-root) Newick tree over the given
     taxa into a PhyloTree, branch lengths defaulting to z = 0.9 (defaultz,
     treeReadLen behavior for topology-only trees).  Ring order of inner
     nodes follows parse order (the reference's p->next chain)."""
-    with open(path) as f:
-        s = f.read().strip()
+    s = s.strip()
     if s.endswith(";"):
         s = s[:-1]
     name_to_id = {n: i + 1 for i, n in enumerate(taxa_names)}

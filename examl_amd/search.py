@@ -942,12 +942,15 @@ class TreeSearch:
         eng = self.engines[m]
         entries, p, q, zv = self._last_full
         z = float(zv[m]) if self.NB > 1 else float(zv[0])
-        key = (id(entries), m if self.NB > 1 else 0)
-        if not hasattr(self, "_epg_ops") or self._epg_ops[1] != key:
+        # cache keyed by IDENTITY of the live entries list (a strong ref is
+        # kept so the id cannot be recycled) + partition for -M
+        mkey = m if self.NB > 1 else 0
+        if (not hasattr(self, "_epg_ops") or self._epg_ops[1] is not entries
+                or self._epg_ops[2] != mkey):
             src = entries if self.NB == 1 \
                 else self._per_partition_entries(entries, m)
             arr = (TravEntry * len(src))(*src)
-            self._epg_ops = (arr, key)
+            self._epg_ops = (arr, entries, mkey)
         arr = self._epg_ops[0]
         fn = (lib().examl_host_evaluate_partial_dna_cat
               if eng.model.states == 4
@@ -1120,6 +1123,13 @@ class TreeSearch:
                 break
         return self.likelihood
 
+    def reset_branches(self):
+        """resetBranches (optimizeModel.c:2511): every branch back to
+        defaultz."""
+        from .tree import DEFAULTZ
+        for a, b in self.tree.edges():
+            self.tree.set_z(a, b, DEFAULTZ)
+
     def tree_evaluation_mode(self, log=None):
         """The -f E (slow TREE_EVALUATION) flow for one tree
         (axml.c:2316-2331): evaluate, treeEvaluate(1), modOpt(0.1)."""
@@ -1130,3 +1140,24 @@ class TreeSearch:
         if log:
             log(f"after treeEvaluate = {self.likelihood:.6f}")
         return self.mod_opt(0.1, log=log)
+
+
+def evaluate_trees(trees, engines, fast=False, log=None, **search_kwargs):
+    """optimizeTrees (axml.c:2721) for the -f E (slow) / -f e (fast)
+    multi-tree input: tree 0 gets the full treeEvaluate(1)+modOpt(0.1),
+    later trees get resetBranches and -- in fast mode -- only
+    treeEvaluate(2); model parameters persist across trees (the
+    reference keeps them in partitionData).  Returns the per-tree lnLs."""
+    out = []
+    for i, tree in enumerate(trees):
+        ts = TreeSearch(tree, engines, **search_kwargs)
+        if i > 0:
+            ts.reset_branches()
+        ts.evaluate_generic(full=True)
+        if fast and i > 0:
+            ts.tree_evaluate(2.0)
+        else:
+            ts.tree_evaluate(1.0)
+            ts.mod_opt(0.1, log=log)
+        out.append(ts.likelihood)
+    return out
