@@ -1675,3 +1675,220 @@ EXPORT void oracle_core_prot_cat(int upper, int numberOfCategories,
   *ext_dlnLdlz = dlnLdlz;
   *ext_d2lnLdlz2 = d2lnLdlz2;
 }
+
+/* ==========================================================================
+ * -S (saveMemory) protein GTRGAMMA kernels: span-80 gap-compacted CLVs +
+ * per-node gap columns; undetermined AA code 22 (gapOffset 440).
+ * Restate newviewGTRGAMMAPROT_AVX_GAPPED_SAVE (avxLikelihood.c:3125),
+ * evaluateGTRGAMMAPROT_GAPPED_SAVE (evaluateGenericSpecial.c:1291),
+ * sumGAMMAPROT_GAPPED_SAVE (makenewzGenericSpecial.c:1896).
+ * ==========================================================================*/
+
+/* per-(site,cat) body of the dense AVX prot kernel, operand-pointer form */
+static void orc_nv_prot_site(int tipCase, const double *uX1,
+                             const double *uX2, const double *xl,
+                             const double *xr, const double *left,
+                             const double *right, const double *extEV,
+                             double *xv, int *scale) {
+  int k, l, s;
+  *scale = 1;
+  for (k = 0; k < 4; k++) {
+    double acc[20];
+    for (s = 0; s < 20; s++) acc[s] = 0.0;
+    for (l = 0; l < 20; l++) {
+      double u1, u2;
+      if (tipCase == ORC_TIP_TIP) {
+        u1 = uX1[k * 20 + l];
+        u2 = uX2[k * 20 + l];
+      } else if (tipCase == ORC_TIP_INNER) {
+        u1 = uX1[k * 20 + l];
+        u2 = dot20_avx(&xr[20 * k], &right[k * 400 + l * 20]);
+      } else {
+        u1 = dot20_avx(&xl[20 * k], &left[k * 400 + l * 20]);
+        u2 = dot20_avx(&xr[20 * k], &right[k * 400 + l * 20]);
+      }
+      const double t = u1 * u2;
+      for (s = 0; s < 20; s++) acc[s] += t * extEV[20 * l + s];
+    }
+    for (s = 0; s < 20; s++) xv[k * 20 + s] = acc[s];
+  }
+  for (s = 0; s < 80 && *scale; s++)
+    if (!(fabs(xv[s]) < ORC_MINLIKELIHOOD)) *scale = 0;
+}
+
+EXPORT void oracle_newview_prot_gamma_save(
+    int tipCase, const double *x1, const double *x2, double *x3,
+    const double *extEV, const double *tipVector,
+    const unsigned char *tipX1, const unsigned char *tipX2, int n,
+    const double *left, const double *right, const int *wgt,
+    int *scalerIncrement, const unsigned int *x1_gap,
+    const unsigned int *x2_gap, unsigned int *x3_gap,
+    const double *x1_gapColumn, const double *x2_gapColumn,
+    double *x3_gapColumn) {
+  int i, k, s;
+  int addScale = 0, scale, scaleGap = 0;
+  static double umpX1[23 * 80], umpX2[23 * 80];
+  const double *x1_ptr = x1, *x2_ptr = x2;
+  double *x3_ptr = x3;
+
+  if (tipCase != ORC_INNER_INNER) {
+    for (i = 0; i < 23; i++) {
+      const double *v = &tipVector[20 * i];
+      for (k = 0; k < 80; k++) {
+        umpX1[80 * i + k] = dot20_avx(v, &left[k * 20]);
+        if (tipCase == ORC_TIP_TIP)
+          umpX2[80 * i + k] = dot20_avx(v, &right[k * 20]);
+      }
+    }
+  }
+
+  {
+    double xv[80];
+    if (tipCase == ORC_TIP_TIP) {
+      orc_nv_prot_site(ORC_TIP_TIP, &umpX1[1760], &umpX2[1760], NULL, NULL,
+                       left, right, extEV, xv, &scale);
+      scaleGap = 0;
+    } else if (tipCase == ORC_TIP_INNER) {
+      orc_nv_prot_site(ORC_TIP_INNER, &umpX1[1760], NULL, NULL,
+                       x2_gapColumn, left, right, extEV, xv, &scaleGap);
+    } else {
+      orc_nv_prot_site(ORC_INNER_INNER, NULL, NULL, x1_gapColumn,
+                       x2_gapColumn, left, right, extEV, xv, &scaleGap);
+    }
+    if (scaleGap)
+      for (s = 0; s < 80; s++) xv[s] *= ORC_TWOTOTHE256;
+    for (s = 0; s < 80; s++) x3_gapColumn[s] = xv[s];
+  }
+
+  for (i = 0; i < n; i++) {
+    if (orc_is_gap(x3_gap, i)) {
+      if (tipCase != ORC_TIP_TIP && scaleGap) addScale += wgt[i];
+      continue;
+    }
+    {
+      double xv[80];
+      const double *xl = NULL, *xr = NULL;
+      const double *uX1 = NULL, *uX2 = NULL;
+      if (tipCase == ORC_TIP_TIP) {
+        uX1 = &umpX1[80 * tipX1[i]];
+        uX2 = &umpX2[80 * tipX2[i]];
+      } else if (tipCase == ORC_TIP_INNER) {
+        uX1 = &umpX1[80 * tipX1[i]];
+        if (orc_is_gap(x2_gap, i))
+          xr = x2_gapColumn;
+        else {
+          xr = x2_ptr;
+          x2_ptr += 80;
+        }
+      } else {
+        if (orc_is_gap(x1_gap, i))
+          xl = x1_gapColumn;
+        else {
+          xl = x1_ptr;
+          x1_ptr += 80;
+        }
+        if (orc_is_gap(x2_gap, i))
+          xr = x2_gapColumn;
+        else {
+          xr = x2_ptr;
+          x2_ptr += 80;
+        }
+      }
+      orc_nv_prot_site(tipCase, uX1, uX2, xl, xr, left, right, extEV, xv,
+                       &scale);
+      if (tipCase != ORC_TIP_TIP && scale) {
+        for (s = 0; s < 80; s++) xv[s] *= ORC_TWOTOTHE256;
+        addScale += wgt[i];
+      }
+      for (s = 0; s < 80; s++) x3_ptr[s] = xv[s];
+      x3_ptr += 80;
+    }
+  }
+  *scalerIncrement = addScale;
+}
+
+EXPORT double oracle_evaluate_prot_gamma_save(
+    const int *wptr, const double *x1_start, const double *x2_start,
+    const double *tipVector, const unsigned char *tipX1, int n,
+    const double *diagptable, const double *x1_gapColumn,
+    const double *x2_gapColumn, const unsigned int *x1_gap,
+    const unsigned int *x2_gap) {
+  double sum = 0.0;
+  int i, j, l;
+  const double *x1_ptr = x1_start, *x2_ptr = x2_start;
+  for (i = 0; i < n; i++) {
+    const double *le, *ri;
+    if (tipX1) {
+      le = &tipVector[20 * tipX1[i]];
+    } else if (orc_is_gap(x1_gap, i)) {
+      le = x1_gapColumn;
+    } else {
+      le = x1_ptr;
+      x1_ptr += 80;
+    }
+    if (orc_is_gap(x2_gap, i)) {
+      ri = x2_gapColumn;
+    } else {
+      ri = x2_ptr;
+      x2_ptr += 80;
+    }
+    double t0 = 0.0, t1 = 0.0;
+    for (j = 0; j < 4; j++) {
+      const double *lrow = tipX1 ? le : &le[20 * j];
+      const double *d = &diagptable[j * 20];
+      const double *r = &ri[20 * j];
+      for (l = 0; l < 20; l += 2) {
+        t0 += lrow[l] * r[l] * d[l];
+        t1 += lrow[l + 1] * r[l + 1] * d[l + 1];
+      }
+    }
+    sum += wptr[i] * log(0.25 * fabs(t0 + t1));
+  }
+  return sum;
+}
+
+EXPORT void oracle_sum_prot_gamma_save(
+    int tipCase, double *sumtable, const double *x1_start,
+    const double *x2_start, const double *tipVector,
+    const unsigned char *tipX1, const unsigned char *tipX2, int n,
+    const double *x1_gapColumn, const double *x2_gapColumn,
+    const unsigned int *x1_gap, const unsigned int *x2_gap) {
+  int i, l, k;
+  const double *x1_ptr = x1_start, *x2_ptr = x2_start;
+  for (i = 0; i < n; i++) {
+    const double *le = NULL, *ri = NULL;
+    switch (tipCase) {
+    case ORC_TIP_TIP:
+      le = &tipVector[20 * tipX1[i]];
+      ri = &tipVector[20 * tipX2[i]];
+      break;
+    case ORC_TIP_INNER:
+      le = &tipVector[20 * tipX1[i]];
+      if (orc_is_gap(x2_gap, i))
+        ri = x2_gapColumn;
+      else {
+        ri = x2_ptr;
+        x2_ptr += 80;
+      }
+      break;
+    default:
+      if (orc_is_gap(x1_gap, i))
+        le = x1_gapColumn;
+      else {
+        le = x1_ptr;
+        x1_ptr += 80;
+      }
+      if (orc_is_gap(x2_gap, i))
+        ri = x2_gapColumn;
+      else {
+        ri = x2_ptr;
+        x2_ptr += 80;
+      }
+    }
+    for (l = 0; l < 4; l++)
+      for (k = 0; k < 20; k++)
+        sumtable[i * 80 + l * 20 + k] =
+            (tipCase == ORC_INNER_INNER ? le[l * 20 + k] : le[k]) *
+            (tipCase == ORC_TIP_TIP ? ri[k] : ri[l * 20 + k]);
+  }
+}

@@ -204,3 +204,108 @@ def test_save_engine_transparent_gpu():
     assert a1 == a2, (a1, a2)
     dense = e1.d_clv.numel() * 8
     assert e2.clv_bytes() < dense
+
+
+@pytest.mark.skipif(not O.have_ref(), reason="reference libref.so not built")
+def test_prot_save_kernels_bit_exact_vs_reference():
+    """Protein (span-80) GAPPED_SAVE kernels: newview/evaluate/sum vs the
+    reference's newviewGTRGAMMAPROT_AVX_GAPPED_SAVE family on gappy AA
+    data (undetermined code 22)."""
+    from tests.helpers import _model_arrays
+    rng = np.random.default_rng(13)
+    aa = np.load(os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "examl_amd", "data", "aa_models.npz"))
+    m = ea.ProtGtrModel(aa["frequencies"][4], aa["rates190"][4], 0.9)
+    EIGN, EV, EI, tipVector, g = _model_arrays(m)
+    n = 160
+    t1 = rng.integers(1, 23, n).astype(np.uint8)
+    t1[rng.random(n) < 0.3] = 22
+    t2 = rng.integers(1, 23, n).astype(np.uint8)
+    t2[rng.random(n) < 0.3] = 22
+    wgt = np.ones(n, dtype=np.int32)
+    gvl = n // 32 + 1
+
+    def gap_of(tips):
+        gv = np.zeros(gvl, dtype=np.uint32)
+        idx = np.nonzero(tips == 22)[0]
+        np.bitwise_or.at(gv, idx // 32,
+                         (np.uint32(1) << (idx % 32).astype(np.uint32)))
+        return gv
+
+    g1, g2 = gap_of(t1), gap_of(t2)
+    left, right = O.make_p(-0.2, -0.5, g, EI, EIGN, 4, 20)
+
+    def dp(a):
+        return a.ctypes.data_as(ctypes.POINTER(ctypes.c_double))
+
+    def u8(a):
+        return (a.ctypes.data_as(ctypes.POINTER(ctypes.c_ubyte))
+                if a is not None else
+                ctypes.cast(None, ctypes.POINTER(ctypes.c_ubyte)))
+
+    def ip(a):
+        return a.ctypes.data_as(ctypes.POINTER(ctypes.c_int))
+
+    def up(a):
+        return (a.ctypes.data_as(ctypes.POINTER(ctypes.c_uint))
+                if a is not None else
+                ctypes.cast(None, ctypes.POINTER(ctypes.c_uint)))
+
+    ndp = ctypes.cast(None, ctypes.POINTER(ctypes.c_double))
+    ref = O._ref
+    tvg = O.aligned(20)
+    tvg[:] = tipVector[22 * 20:23 * 20]
+
+    def run(tc, x1r, x1o, x2r, x2o, ga, gb, gca_r, gca_o, gcb_r, gcb_o, ta,
+            tb):
+        g3 = ga & gb
+        nz = int(n - sum(bin(int(w)).count("1") for w in g3))
+        x3r = O.aligned(nz * 80 + 80)
+        x3o = O.aligned(nz * 80 + 80)
+        gr = O.aligned(80)
+        go = O.aligned(80)
+        ir = ctypes.c_int(0)
+        io = ctypes.c_int(0)
+        ref.newviewGTRGAMMAPROT_AVX_GAPPED_SAVE(
+            tc, dp(x1r) if x1r is not None else ndp,
+            dp(x2r) if x2r is not None else ndp, dp(x3r), dp(EV),
+            dp(tipVector), None, u8(ta), u8(tb), ctypes.c_int(n), dp(left),
+            dp(right), ip(wgt), ctypes.byref(ir), ctypes.c_int(1), up(ga),
+            up(gb), up(g3), dp(gca_r), dp(gcb_r), dp(gr))
+        O._orc.oracle_newview_prot_gamma_save(
+            tc, dp(x1o) if x1o is not None else ndp,
+            dp(x2o) if x2o is not None else ndp, dp(x3o), dp(EV),
+            dp(tipVector), u8(ta), u8(tb), ctypes.c_int(n), dp(left),
+            dp(right), ip(wgt), ctypes.byref(io), up(ga), up(gb), up(g3),
+            dp(gca_o), dp(gcb_o), dp(go))
+        assert np.array_equal(x3r[:nz * 80], x3o[:nz * 80])
+        assert np.array_equal(gr, go) and ir.value == io.value
+        return g3, x3r, x3o, gr, go
+
+    g3, ar, ao, gr, go = run(0, None, None, None, None, g1, g2, tvg, tvg,
+                             tvg, tvg, t1, t2)
+    g3b, br, bo, gr2, go2 = run(1, None, None, ar, ao, g1, g3, tvg, tvg, gr,
+                                go, t1, None)
+    run(2, ar, ao, br, bo, g3, g3b, gr, go, gr2, go2, None, None)
+
+    diag = O.calc_diagptable(0.7, 20, 4, g, EIGN)
+    ref.evaluateGTRGAMMAPROT_GAPPED_SAVE.restype = ctypes.c_double
+    O._orc.oracle_evaluate_prot_gamma_save.restype = ctypes.c_double
+    lr = ref.evaluateGTRGAMMAPROT_GAPPED_SAVE(
+        ip(wgt), dp(ar), dp(br), dp(tipVector), None, ctypes.c_int(n),
+        dp(diag), dp(gr), dp(gr2), up(g3), up(g3b))
+    lo = O._orc.oracle_evaluate_prot_gamma_save(
+        ip(wgt), dp(ao), dp(bo), dp(tipVector), None, ctypes.c_int(n),
+        dp(diag), dp(go), dp(go2), up(g3), up(g3b))
+    assert lr == lo
+
+    sr = O.aligned(n * 80)
+    so = O.aligned(n * 80)
+    ref.sumGAMMAPROT_GAPPED_SAVE(2, dp(sr), dp(ar), dp(br), dp(tipVector),
+                                 None, None, ctypes.c_int(n), dp(gr),
+                                 dp(gr2), up(g3), up(g3b))
+    O._orc.oracle_sum_prot_gamma_save(2, dp(so), dp(ao), dp(bo),
+                                      dp(tipVector), u8(None), u8(None),
+                                      ctypes.c_int(n), dp(go), dp(go2),
+                                      up(g3), up(g3b))
+    assert np.array_equal(sr, so)
