@@ -177,12 +177,13 @@ def check(rc, what):
 
 from .model import DnaGtrModel, Lg4Model, ProtGtrModel  # noqa: E402
 from .tree import PhyloTree             # noqa: E402
+from .spr import SprSearch, SprTree     # noqa: E402
 from .engine import (DnaCatEngine, DnaGammaEngine, Lg4Engine,  # noqa: E402
                      ProtCatEngine, SaveDnaEngine)
 
 __all__ = [
     "lib", "check", "TravEntry", "DnaGtrModel", "ProtGtrModel",
-    "Lg4Model", "PhyloTree",
+    "Lg4Model", "PhyloTree", "SprSearch", "SprTree",
     "DnaGammaEngine", "DnaCatEngine", "Lg4Engine", "SaveDnaEngine",
     "ProtCatEngine",
     "TIP_TIP", "TIP_INNER",

@@ -89,6 +89,10 @@ class TreeSearch:
 
     def _children(self, p, parent):
         t = self.tree
+        if hasattr(t, "ring_children"):
+            # ring order from the parent-facing member (computeTraversalInfo
+            # uses p->next->back / p->next->next->back)
+            return t.ring_children(p, parent)
         nbrs = [w for w in t.adj[p] if w != parent]
         assert len(nbrs) == 2
         q, r = nbrs
@@ -96,6 +100,13 @@ class TreeSearch:
 
     def _collect(self, p, parent, partial, out):
         t = self.tree
+        if hasattr(t, "find_member"):
+            # ring tree (SPR flows): the x flag lives on a ring MEMBER —
+            # validity must follow the member, not the node number, to
+            # reproduce the reference's stale-CLV reuse after surgery
+            # (getxnode/computeTraversalInfo semantics)
+            self._collect_ring(t.find_member(p, parent), partial, out)
+            return
         if t.is_tip(p):
             return
         q, r = self._children(p, parent)
@@ -126,6 +137,54 @@ class TreeSearch:
         e.x2Slot = r if t.is_tip(r) else t.clv_slot(r)
         out.append(e)
         self.oriented[p] = parent
+
+    def _collect_gated(self, p, parent, partial, out):
+        """the x-gated top of evaluateGeneric/makenewzGeneric
+        (evaluateGenericSpecial.c:940: if(!p->x) computeTraversalInfo):
+        on ring trees a valid top member skips its whole side."""
+        t = self.tree
+        if partial and hasattr(t, "find_member") and not t.is_tip(p):
+            m = t.find_member(p, parent)
+            if self.oriented.get(p) is m:
+                return
+            self._collect_ring(m, partial, out)
+            return
+        self._collect(p, parent, partial, out)
+
+    def _collect_ring(self, m, partial, out):
+        """computeTraversalInfo (newviewGenericSpecial.c:691) on ring
+        member m: children q/r are m->next->back / m->next->next->back;
+        a child is valid iff the x flag (self.oriented) sits on exactly
+        that member."""
+        t = self.tree
+        if t.is_tip(m.number):
+            return
+        q = m.next.back
+        r = m.next.next.back
+        q_tip, r_tip = t.is_tip(q.number), t.is_tip(r.number)
+        if q_tip and r_tip:
+            tc = TIP_TIP
+        elif q_tip or r_tip:
+            if r_tip:
+                q, r = r, q
+            if self.oriented.get(r.number) is not r or not partial:
+                self._collect_ring(r, partial, out)
+            tc = TIP_INNER
+        else:
+            if self.oriented.get(q.number) is not q or not partial:
+                self._collect_ring(q, partial, out)
+            if self.oriented.get(r.number) is not r or not partial:
+                self._collect_ring(r, partial, out)
+            tc = INNER_INNER
+        e = TravEntry()
+        e.tipCase = tc
+        e.pNumber, e.qNumber, e.rNumber = m.number, q.number, r.number
+        e.qz, e.rz = q.z, r.z
+        e.x3Slot = t.clv_slot(m.number)
+        e.x1Slot = q.number if t.is_tip(q.number) else t.clv_slot(q.number)
+        e.x2Slot = r.number if t.is_tip(r.number) else t.clv_slot(r.number)
+        out.append(e)
+        self.oriented[m.number] = m
 
     def _per_partition_entries(self, entries, m):
         out = []
@@ -162,18 +221,26 @@ class TreeSearch:
         self._collect(p, parent, True, out)
         self._run(out)
 
-    def evaluate_generic(self, full=True, p=None):
+    def evaluate_generic(self, full=True, p=None, q=None, entries=None,
+                         z=None):
         """evaluateGeneric(tr, p (default tr->start), fullTraversal)
-        (evaluateGenericSpecial.c:897)."""
+        (evaluateGenericSpecial.c:897).  q selects the branch explicitly
+        (the reference's p->back); `entries`/`z` let a caller (the SPR
+        driver) supply a pre-built traversal for a branch that cannot be
+        resolved through node numbers mid-surgery."""
         t = self.tree
         if p is None:
             p = self.start
-        q = next(iter(t.adj[p]))  # p->back
-        out = []
-        self._collect(p, q, not full, out)
-        self._collect(q, p, not full, out)
+        if q is None:
+            q = next(iter(t.adj[p]))  # p->back
+        if entries is None:
+            out = []
+            self._collect_gated(p, q, not full, out)
+            self._collect_gated(q, p, not full, out)
+        else:
+            out = entries
         self._run(out)
-        zv = t.get_zv(p, q)
+        zv = np.array([z]) if z is not None else t.get_zv(p, q)
         z = float(zv[0])
         if full:
             # td[0] of the last full traversal: evaluatePartialGeneric walks
@@ -197,13 +264,16 @@ class TreeSearch:
         self.likelihood = sum(self.per_partition_lnl)
         return self.likelihood
 
-    def makenewz_generic(self, p, q, z0, maxiter):
+    def makenewz_generic(self, p, q, z0, maxiter, entries=None):
         """makenewzGeneric (makenewzGenericSpecial.c:1355, mask=FALSE) +
         topLevelMakenewz (:1133) for numBranches=1, derivatives summed over
         executing partitions (execCore, :1075)."""
-        out = []
-        self._collect(p, q, True, out)
-        self._collect(q, p, True, out)
+        if entries is None:
+            out = []
+            self._collect_gated(p, q, True, out)
+            self._collect_gated(q, p, True, out)
+        else:
+            out = entries
         self._run(out)
         for m, eng in enumerate(self.engines):
             if self.execute_model[m]:

@@ -1,0 +1,919 @@
+"""The SPR tree-search driver (-f d, BIG_RAPID_MODE): computeBIGRAPID and
+its machinery (searchAlgo.c) restated over the engine layer — the primary
+CALLER of the conditional-likelihood hot path.
+
+The reference represents each inner node as a ring of three `node`
+structs whose `next` pointers are fixed at setup and whose `back`
+pointers are rewired by SPR surgery (hookup).  SprTree reproduces that
+exactly: RingNode objects with static rings, surgery touches only backs.
+TreeSearch drives the same engines over SprTree through the PhyloTree
+interface plus ring-ordered children.
+
+Restated pieces (file:line in searchAlgo.c unless noted):
+  removeNodeBIG:440  insertBIG:482  insertRestoreBIG:574
+  restoreTopologyOnly:610  testInsertBIG:683  addTraverseBIG:784
+  rearrangeBIG:804  treeOptimizeRapid:914  testInsertRestoreBIG:1037
+  restoreTreeFast:1095  localSmooth:280  nodeRectifier (trash.c:54)
+  determineRearrangementSetting:1752  computeBIGRAPID:1914
+  infoList:318-380  bestlist (topologies.c:187-700)
+"""
+
+import math
+
+import numpy as np
+
+from . import INNER_INNER, TIP_INNER, TIP_TIP, TravEntry
+from .search import SMOOTHINGS, TreeSearch
+from .tree import DEFAULTZ
+
+UNLIKELY = -1.0e300
+ZMIN, ZMAX = 1.0e-15, 1.0 - 1.0e-6
+ITERATIONS = 10  # axml.h:90, NR iterations per insertion's makenewz
+
+
+def _cdiv(a, b):
+    """C floating division semantics for lhAVG/lhDEC with lhDEC==0
+    (the reference divides by a zero int: NaN/inf, never a trap) — a NaN
+    or +inf cutoff simply disables the cutoff comparison."""
+    if b:
+        return a / b
+    if a == 0.0:
+        return math.nan
+    return math.inf if a > 0 else -math.inf
+
+
+class RingNode:
+    __slots__ = ("number", "next", "back", "z")
+
+    def __init__(self, number):
+        self.number = number
+        self.next = self
+        self.back = None
+        self.z = DEFAULTZ
+
+    def __repr__(self):
+        return f"RN({self.number})"
+
+
+def hookup(p, q, z):
+    """hookup (axml.c:478), numBranches=1."""
+    p.back = q
+    q.back = p
+    p.z = q.z = z
+
+
+class SprTree:
+    """Ring-based unrooted tree with the PhyloTree interface (so the
+    unchanged TreeSearch drives the engines over it) plus the surgery
+    primitives of searchAlgo.c.  nodep[i] is the canonical ring member
+    (reassigned by node_rectifier, like tr->nodep)."""
+
+    def __init__(self, ntips):
+        self.ntips = ntips
+        self.nnodes = 2 * ntips - 1
+        # nodep: the tr->nodep iteration array — node_rectifier PERMUTES
+        # the inner entries (nodep[i].number need not equal i afterwards,
+        # exactly like the reference).  ring: a fixed member per node
+        # NUMBER, used for all number->ring lookups.
+        self.nodep = [None] * (2 * ntips)
+        self.ring = [None] * (2 * ntips)
+        for i in range(1, ntips + 1):
+            self.nodep[i] = self.ring[i] = RingNode(i)
+        for i in range(ntips + 1, 2 * ntips - 1):
+            # setupTree ring construction (axml.c:631-655): members built
+            # j=1..3 with p->next = previous; nodep[i] = last member
+            m1, m2, m3 = RingNode(i), RingNode(i), RingNode(i)
+            m3.next = m2
+            m2.next = m1
+            m1.next = m3
+            self.nodep[i] = self.ring[i] = m3
+        self.start = 1  # node number of tr->start
+
+    # -- construction from a PhyloTree --------------------------------
+
+    @classmethod
+    def from_phylo(cls, t):
+        """Build rings from a parsed PhyloTree: preorder from tip 1, the
+        first-reached member of each inner node faces its parent and the
+        children attach in the PhyloTree adjacency order (which mirrors
+        the reference's treeReadLen ring order)."""
+        st = cls(t.ntips)
+        used = {}  # inner number -> next free member (cycled via .next)
+
+        def member_for(num, toward_parent):
+            if num <= t.ntips:
+                return st.ring[num]
+            if num not in used:
+                used[num] = st.ring[num]  # canonical faces the parent
+                return used[num]
+            used[num] = used[num].next
+            return used[num]
+
+        seen = set()
+
+        def walk(p, parent):
+            seen.add(p)
+            for w in t.adj[p]:
+                if w == parent:
+                    continue
+                mp = member_for(p, False)
+                mw = member_for(w, True)
+                hookup(mp, mw, t.get_z(p, w))
+                if w not in seen:
+                    walk(w, p)
+
+        # root the construction at tip 1's edge
+        first = next(iter(t.adj[1]))
+        hookup(st.nodep[1], member_for(first, True), t.get_z(1, first))
+        seen.update((1,))
+        walk(first, 1)
+        return st
+
+    # -- PhyloTree interface ------------------------------------------
+
+    def is_tip(self, v):
+        return v <= self.ntips
+
+    def clv_slot(self, v):
+        assert not self.is_tip(v)
+        return v - self.ntips - 1
+
+    def members(self, p):
+        m = self.ring[p]
+        yield m
+        if not self.is_tip(p):
+            yield m.next
+            yield m.next.next
+
+    def find_member(self, p, q):
+        for m in self.members(p):
+            if m.back is not None and m.back.number == q:
+                return m
+        raise KeyError((p, q))
+
+    def get_z(self, a, b):
+        return self.find_member(a, b).z
+
+    def get_zv(self, a, b):
+        return np.array([self.get_z(a, b)])
+
+    def set_z(self, a, b, z):
+        m = self.find_member(a, b)
+        m.z = m.back.z = z
+
+    def ring_children(self, p, parent):
+        """the (q, r) of computeTraversalInfo: ring order from the
+        parent-facing member."""
+        m = self.find_member(p, parent)
+        return m.next.back.number, m.next.next.back.number
+
+    class _Nbrs:
+        def __init__(self, t, p):
+            self.t = t
+            self.p = p
+
+        def __iter__(self):
+            for m in self.t.members(self.p):
+                if m.back is not None:
+                    yield m.back.number
+
+        def __getitem__(self, q):
+            return self.t.get_z(self.p, q)
+
+    class _Adj:
+        def __init__(self, t):
+            self.t = t
+
+        def __getitem__(self, p):
+            return SprTree._Nbrs(self.t, p)
+
+    @property
+    def adj(self):
+        return SprTree._Adj(self)
+
+    def edges(self):
+        out = []
+        seen = set()
+        for i in range(1, 2 * self.ntips - 1):
+            for m in self.members(i):
+                if m.back is None:
+                    continue
+                key = (min(i, m.back.number), max(i, m.back.number))
+                if key not in seen:
+                    seen.add(key)
+                    out.append(key)
+        return out
+
+
+class Topol:
+    """topol (topologies.c): standard-order link list.  links[k] =
+    [member_p, member_q, z, descend, sibling, valptr]."""
+
+    __slots__ = ("links", "likelihood", "start")
+
+    def __init__(self):
+        self.links = []
+        self.likelihood = UNLIKELY
+        self.start = None
+
+
+def _save_subtree(p, tpl):
+    """saveSubtree (topologies.c:229): returns the link index for branch
+    p--p->back with children merged in ascending min-tip order."""
+    links = tpl.links
+    r = [p, p.back, p.z, 0, 0, None]
+    ri = len(links)
+    links.append(r)
+    q = p.back
+    if q.number <= tpl_ntips[0]:
+        r[5] = q.number
+    else:
+        s = q.next
+        while True:
+            t = _save_subtree(s, tpl)
+            t0, t1 = 0, r[3]
+            while t1 and links[t1][5] < links[t][5]:
+                t0 = t1
+                t1 = links[t1][4]
+            if t0:
+                links[t0][4] = t
+            else:
+                r[3] = t
+            links[t][4] = t1
+            s = s.next
+            if s is q:
+                break
+        r[5] = links[r[3]][5]
+    return ri
+
+
+tpl_ntips = [0]  # module-scope like the reference's numsp parameter
+
+
+def _min_subtree_tip(p0, ntips):
+    if p0.number <= ntips:
+        return p0
+    p = p0.next
+    best = _min_subtree_tip(p.back, ntips)
+    p = p.next
+    while p is not p0:
+        t = _min_subtree_tip(p.back, ntips)
+        if t.number < best.number:
+            best = t
+        p = p.next
+    return best
+
+
+def _min_tree_tip(p, ntips):
+    a = _min_subtree_tip(p, ntips)
+    b = _min_subtree_tip(p.back, ntips)
+    return a if a.number < b.number else b
+
+
+def save_tree(st, likelihood, tpl):
+    """saveTree (topologies.c:291)."""
+    tpl.links = []
+    tpl_ntips[0] = st.ntips
+    start_m = st.find_member(st.start, next(iter(st.adj[st.start])))
+    _save_subtree(_min_tree_tip(start_m, st.ntips), tpl)
+    tpl.likelihood = likelihood
+    tpl.start = st.start
+
+
+def _cmp_subtopol(l1, p1, l2, p2):
+    """cmpSubtopol (topologies.c:452)."""
+    if not p1[3] and not p2[3]:
+        return (p1[5] > p2[5]) - (p1[5] < p2[5])
+    if not p1[3]:
+        return -1
+    if not p2[3]:
+        return 1
+    d1, d2 = l1[p1[3]], l2[p2[3]]
+    while True:
+        c = _cmp_subtopol(l1, d1, l2, d2)
+        if c:
+            return c
+        if not d1[4] and not d2[4]:
+            return 0
+        if not d1[4]:
+            return -1
+        if not d2[4]:
+            return 1
+        d1, d2 = l1[d1[4]], l2[d2[4]]
+
+
+def cmp_topol(t1, t2):
+    r1, r2 = t1.links[0], t2.links[0]
+    v1, v2 = r1[0].number, r2[0].number  # root tip values
+    if v1 != v2:
+        return -1 if v1 < v2 else 1
+    return _cmp_subtopol(t1.links, r1, t2.links, r2)
+
+
+def restore_tree(tpl, st, ts):
+    """restoreTree (topologies.c:331): clear all backs, re-hookup the
+    saved links, full evaluate."""
+    for i in range(1, 2 * st.ntips - 1):
+        for m in st.members(i):
+            m.back = None
+    for r in tpl.links:
+        hookup(r[0], r[1], r[2])
+    st.start = tpl.start
+    return ts.evaluate_generic(full=True)
+
+
+class BestList:
+    """bestlist (topologies.c:370-655): trees ordered by score, duplicate
+    topologies detected via the standard-order comparison."""
+
+    def __init__(self, keep, st):
+        self.keep = keep
+        self.st = st
+        self.by_score = []  # list of Topol, best first
+        self.best = UNLIKELY
+        self.worst = UNLIKELY
+
+    def reset(self):
+        self.by_score = []
+        self.best = UNLIKELY
+        self.worst = UNLIKELY
+
+    @property
+    def nvalid(self):
+        return len(self.by_score)
+
+    def save(self, ts, keep_identical):
+        """saveBestTree semantics: returns the 1-based score rank or 0."""
+        lh = ts.likelihood
+        tpl = Topol()
+        save_tree(self.st, lh, tpl)
+        for t in self.by_score:
+            if cmp_topol(tpl, t) == 0:
+                if not keep_identical:
+                    return 0
+                # replace the stored copy (branch lengths/score updated)
+                self.by_score.remove(t)
+                break
+        else:
+            if self.nvalid >= self.keep and lh < self.worst:
+                return 0
+        # insert by score (descending)
+        pos = 0
+        while pos < len(self.by_score) and \
+                self.by_score[pos].likelihood >= lh:
+            pos += 1
+        self.by_score.insert(pos, tpl)
+        if len(self.by_score) > self.keep:
+            self.by_score.pop()
+        self.best = self.by_score[0].likelihood
+        if len(self.by_score) == self.keep:
+            self.worst = self.by_score[-1].likelihood
+        return pos + 1
+
+    def recall(self, rank, ts):
+        rank = max(1, min(rank, self.nvalid))
+        if rank > 0:
+            return restore_tree(self.by_score[rank - 1], self.st, ts)
+        return None
+
+
+class SprSearch:
+    """computeBIGRAPID over a TreeSearch on an SprTree."""
+
+    def __init__(self, ts, do_cutoff=True, big_cutoff=False, stepwidth=5,
+                 max_rearrange=21, log=None):
+        self.ts = ts
+        self.st = ts.tree
+        assert isinstance(self.st, SprTree)
+        assert ts.NB == 1, "SPR with -M not supported yet"
+        self.thorough = False
+        self.do_cutoff = do_cutoff
+        self.big_cutoff = big_cutoff
+        self.stepwidth = stepwidth
+        self.max_rearrange = max_rearrange
+        self.log = log or (lambda *_: None)
+        # tr-> search state
+        self.start_lh = 0.0
+        self.end_lh = 0.0
+        self.best_of_node = UNLIKELY
+        self.remove_node = None   # ring member
+        self.insert_node = None   # ring member
+        self.zqr = DEFAULTZ
+        self.current_zqr = DEFAULTZ
+        self.lzq = self.lzr = self.lzs = DEFAULTZ
+        self.current_lzq = self.current_lzr = self.current_lzs = DEFAULTZ
+        self.lh_cutoff = 0.0
+        self.lh_avg = 0.0
+        self.lh_dec = 0
+        self.it_count = 0
+        # infoList (n=50)
+        self.ilist_n = 50
+        self.ilist = []
+
+    # ---- member-based traversal (computeTraversalInfo on rings) ------
+    # Node numbers are ambiguous mid-surgery (makenewzGeneric is called on
+    # node pairs that are not yet adjacent, removeNodeBIG:452), so these
+    # drive the engines through member handles directly.
+
+    def _collect_m(self, m, partial, out):
+        self.ts._collect_ring(m, partial, out)
+
+    def _evaluate_branch(self, m):
+        """evaluateGeneric(tr, m, FALSE) at the branch m--m->back:
+        each side collected only if its x flag is unset (:940)."""
+        ts = self.ts
+        out = []
+        if ts.oriented.get(m.number) is not m:
+            self._collect_m(m, True, out)
+        mb = m.back
+        if ts.oriented.get(mb.number) is not mb:
+            self._collect_m(mb, True, out)
+        return ts.evaluate_generic(full=False, p=m.number, q=mb.number,
+                                   entries=out, z=m.z)
+
+    def _newview(self, m):
+        """newviewGeneric(tr, m, FALSE): CLV at m's node oriented toward
+        m->back."""
+        out = []
+        self._collect_m(m, True, out)
+        self.ts._run(out)
+
+    def _makenewz(self, m1, m2, z0):
+        """makenewzGeneric on a (possibly not-yet-hooked) member pair;
+        sides collected only when their x flag is unset (:1386)."""
+        ts = self.ts
+        out = []
+        if ts.oriented.get(m1.number) is not m1:
+            self._collect_m(m1, True, out)
+        if ts.oriented.get(m2.number) is not m2:
+            self._collect_m(m2, True, out)
+        return ts.makenewz_generic(m1.number, m2.number, z0, ITERATIONS,
+                                   entries=out)
+
+    def _update(self, m):
+        """searchAlgo update(tr, p): optimize branch m--m->back."""
+        self.ts.update(m.number, m.back.number)
+
+    def local_smooth(self, p, maxtimes):
+        """localSmooth (searchAlgo.c:280)."""
+        ts = self.ts
+        if self.st.is_tip(p.number):
+            return False
+        ts.partition_converged = False
+        while maxtimes > 0:
+            maxtimes -= 1
+            ts.partition_smoothed = True
+            q = p
+            while True:
+                self._update(q)
+                q = q.next
+                if q is p:
+                    break
+            if ts.partition_smoothed:  # allSmoothed
+                ts.partition_converged = True
+                break
+        ts.partition_smoothed = False
+        ts.partition_converged = False
+        return True
+
+    # ---- SPR surgery -------------------------------------------------
+
+    def remove_node_big(self, p):
+        """removeNodeBIG (searchAlgo.c:440)."""
+        q = p.next.back
+        r = p.next.next.back
+        z0 = q.z * r.z
+        result = self._makenewz(q, r, z0)
+        self.zqr = result
+        hookup(q, r, result)
+        p.next.back = p.next.next.back = None
+        return q
+
+    def remove_node_restore_big(self, p):
+        """removeNodeRestoreBIG (:464)."""
+        q = p.next.back
+        r = p.next.next.back
+        self._newview(q)
+        self._newview(r)
+        hookup(q, r, self.current_zqr)
+        p.next.back = p.next.next.back = None
+        return q
+
+    def insert_big(self, p, q):
+        """insertBIG (:482)."""
+        r = q.back
+        s = p.back
+        if self.thorough:
+            qz = q.z
+            zqr = self._makenewz(q, r, qz)
+            zqs = self._makenewz(q, s, DEFAULTZ)
+            zrs = self._makenewz(r, s, DEFAULTZ)
+            lzqr = math.log(zqr) if zqr > ZMIN else math.log(ZMIN)
+            lzqs = math.log(zqs) if zqs > ZMIN else math.log(ZMIN)
+            lzrs = math.log(zrs) if zrs > ZMIN else math.log(ZMIN)
+            lzsum = 0.5 * (lzqr + lzqs + lzrs)
+            lzq = lzsum - lzrs
+            lzr = lzsum - lzqs
+            lzs = lzsum - lzqr
+            lzmax = math.log(ZMAX)
+            if lzq > lzmax:
+                lzq, lzr, lzs = lzmax, lzqr, lzqs
+            elif lzr > lzmax:
+                lzr, lzq, lzs = lzmax, lzqr, lzrs
+            elif lzs > lzmax:
+                lzs, lzq, lzr = lzmax, lzqs, lzrs
+            hookup(p.next, q, math.exp(lzq))
+            hookup(p.next.next, r, math.exp(lzr))
+            hookup(p, s, math.exp(lzs))
+        else:
+            z = math.sqrt(q.z)
+            z = min(max(z, ZMIN), ZMAX)
+            hookup(p.next, q, z)
+            hookup(p.next.next, r, z)
+        self._newview(p)
+        if self.thorough:
+            self.local_smooth(p, SMOOTHINGS)
+            self.lzq = p.next.z
+            self.lzr = p.next.next.z
+            self.lzs = p.z
+        return True
+
+    def insert_restore_big(self, p, q):
+        """insertRestoreBIG (:574)."""
+        r = q.back
+        s = p.back
+        if self.thorough:
+            hookup(p.next, q, self.current_lzq)
+            hookup(p.next.next, r, self.current_lzr)
+            hookup(p, s, self.current_lzs)
+        else:
+            z = math.sqrt(q.z)
+            z = min(max(z, ZMIN), ZMAX)
+            hookup(p.next, q, z)
+            hookup(p.next.next, r, z)
+        self._newview(p)
+        return True
+
+    def test_insert_big(self, p, q):
+        """testInsertBIG (:683), no constraint tree."""
+        r = q.back
+        qz, pz = q.z, p.z
+        start_lh = self.end_lh
+        self.insert_big(p, q)
+        lnl = self._evaluate_branch(p.next.next)
+        if lnl > self.best_of_node:
+            self.best_of_node = lnl
+            self.insert_node = q
+            self.remove_node = p
+            self.current_zqr = self.zqr
+            self.current_lzr = self.lzr
+            self.current_lzq = self.lzq
+            self.current_lzs = self.lzs
+        if lnl > self.end_lh:
+            self.insert_node = q
+            self.remove_node = p
+            self.current_zqr = self.zqr
+            self.end_lh = lnl
+        hookup(q, r, qz)
+        p.next.back = p.next.next.back = None
+        if self.thorough:
+            s = p.back
+            hookup(p, s, pz)
+        if self.do_cutoff and lnl < start_lh:
+            self.lh_avg += start_lh - lnl
+            self.lh_dec += 1
+            return not (start_lh - lnl >= self.lh_cutoff)
+        return True
+
+    def add_traverse_big(self, p, q, mintrav, maxtrav):
+        """addTraverseBIG (:784)."""
+        mintrav -= 1
+        if mintrav <= 0:
+            if not self.test_insert_big(p, q):
+                return
+        maxtrav -= 1
+        if not self.st.is_tip(q.number) and maxtrav > 0:
+            self.add_traverse_big(p, q.next.back, mintrav, maxtrav)
+            self.add_traverse_big(p, q.next.next.back, mintrav, maxtrav)
+
+    def rearrange_big(self, p, mintrav, maxtrav):
+        """rearrangeBIG (:804)."""
+        st = self.st
+        if maxtrav < 1 or mintrav > maxtrav:
+            return 0
+        q = p.back
+        if not st.is_tip(p.number):
+            p1 = p.next.back
+            p2 = p.next.next.back
+            if not st.is_tip(p1.number) or not st.is_tip(p2.number):
+                p1z, p2z = p1.z, p2.z
+                self.remove_node_big(p)
+                if not st.is_tip(p1.number):
+                    self.add_traverse_big(p, p1.next.back, mintrav, maxtrav)
+                    self.add_traverse_big(p, p1.next.next.back, mintrav,
+                                          maxtrav)
+                if not st.is_tip(p2.number):
+                    self.add_traverse_big(p, p2.next.back, mintrav, maxtrav)
+                    self.add_traverse_big(p, p2.next.next.back, mintrav,
+                                          maxtrav)
+                hookup(p.next, p1, p1z)
+                hookup(p.next.next, p2, p2z)
+                self._newview(p)
+        if not st.is_tip(q.number) and maxtrav > 0:
+            q1 = q.next.back
+            q2 = q.next.next.back
+            if ((not st.is_tip(q1.number) and
+                 (not st.is_tip(q1.next.back.number) or
+                  not st.is_tip(q1.next.next.back.number))) or
+                (not st.is_tip(q2.number) and
+                 (not st.is_tip(q2.next.back.number) or
+                  not st.is_tip(q2.next.next.back.number)))):
+                q1z, q2z = q1.z, q2.z
+                self.remove_node_big(q)
+                mintrav2 = max(mintrav, 2)
+                if not st.is_tip(q1.number):
+                    self.add_traverse_big(q, q1.next.back, mintrav2, maxtrav)
+                    self.add_traverse_big(q, q1.next.next.back, mintrav2,
+                                          maxtrav)
+                if not st.is_tip(q2.number):
+                    self.add_traverse_big(q, q2.next.back, mintrav2, maxtrav)
+                    self.add_traverse_big(q, q2.next.next.back, mintrav2,
+                                          maxtrav)
+                hookup(q.next, q1, q1z)
+                hookup(q.next.next, q2, q2z)
+                self._newview(q)
+        return 1
+
+    def test_insert_restore_big(self, p, q):
+        """testInsertRestoreBIG (:1037)."""
+        if self.thorough:
+            self.insert_big(p, q)
+            self._evaluate_branch(p.next.next)
+        else:
+            self.insert_restore_big(p, q)
+            # re-establish the CLVs the reference spins on x flags for
+            x = p.next.next
+            y = p.back
+            ts, st = self.ts, self.st
+            if not st.is_tip(x.number) and ts.oriented.get(x.number) is not x:
+                self._newview(x)
+            if not st.is_tip(y.number) and ts.oriented.get(y.number) is not y:
+                self._newview(y)
+            self.ts.likelihood = self.end_lh
+        return True
+
+    def restore_tree_fast(self):
+        """restoreTreeFast (:1095)."""
+        self.remove_node_restore_big(self.remove_node)
+        self.test_insert_restore_big(self.remove_node, self.insert_node)
+
+    def restore_topology_only(self, bt):
+        """restoreTopologyOnly (:610): record the best insertion for this
+        node into bt without computing anything."""
+        p = self.remove_node
+        q = self.insert_node
+        current_lh = self.ts.likelihood
+        p1 = p.next.back
+        p2 = p.next.next.back
+        p1z, p2z = p1.z, p2.z
+        hookup(p1, p2, self.current_zqr)
+        p.next.back = p.next.next.back = None
+        qz, pz = q.z, p.z
+        r = q.back
+        s = p.back
+        if self.thorough:
+            hookup(p.next, q, self.current_lzq)
+            hookup(p.next.next, r, self.current_lzr)
+            hookup(p, s, self.current_lzs)
+        else:
+            z = math.sqrt(q.z)
+            z = min(max(z, ZMIN), ZMAX)
+            hookup(p.next, q, z)
+            hookup(p.next.next, r, z)
+        self.ts.likelihood = self.best_of_node
+        bt.save(self.ts, True)
+        self.ts.likelihood = current_lh
+        hookup(q, r, qz)
+        p.next.back = p.next.next.back = None
+        if self.thorough:
+            hookup(p, s, pz)
+        hookup(p.next, p1, p1z)
+        hookup(p.next.next, p2, p2z)
+
+    # ---- infoList ----------------------------------------------------
+
+    def reset_info_list(self):
+        self.ilist = []
+
+    def insert_info_list(self, node, likelihood):
+        """insertInfoList (:352): replace the minimum if better."""
+        if len(self.ilist) < self.ilist_n:
+            self.ilist.append([node, likelihood])
+            return
+        mn = min(range(len(self.ilist)), key=lambda i: self.ilist[i][1])
+        if likelihood > self.ilist[mn][1]:
+            self.ilist[mn] = [node, likelihood]
+
+    # ---- driver pieces -----------------------------------------------
+
+    def node_rectifier(self):
+        """nodeRectifier (trash.c:54): canonicalize nodep[] to the
+        preorder parent-facing members from nodep[1]->back."""
+        st = self.st
+        st.start = 1
+        count = [0]
+        old = {i: st.nodep[i] for i in range(st.ntips + 1, 2 * st.ntips - 1)}
+
+        def reorder(p):
+            if st.is_tip(p.number):
+                return
+            st.nodep[st.ntips + 1 + count[0]] = p
+            count[0] += 1
+            reorder(p.next.back)
+            reorder(p.next.next.back)
+
+        reorder(st.nodep[1].back)
+        assert count[0] == st.ntips - 2, (count[0], old and None)
+
+    def tree_optimize_rapid(self, mintrav, maxtrav, bt):
+        """treeOptimizeRapid (:914)."""
+        ts, st = self.ts, self.st
+        self.node_rectifier()
+        maxtrav = min(maxtrav, st.ntips - 3)
+        self.reset_info_list()
+        bt.reset()
+        self.start_lh = self.end_lh = ts.likelihood
+        if self.do_cutoff:
+            if self.big_cutoff:
+                if self.it_count == 0:
+                    self.lh_cutoff = 0.5 * (ts.likelihood / -1000.0)
+                else:
+                    self.lh_cutoff = 0.5 * _cdiv(self.lh_avg, self.lh_dec)
+            else:
+                if self.it_count == 0:
+                    self.lh_cutoff = ts.likelihood / -1000.0
+                else:
+                    self.lh_cutoff = _cdiv(self.lh_avg, self.lh_dec)
+            self.it_count += 1
+            self.lh_avg = 0.0
+            self.lh_dec = 0
+        for i in range(1, 2 * st.ntips - 1):
+            self.best_of_node = UNLIKELY
+            if self.rearrange_big(st.nodep[i], mintrav, maxtrav):
+                if self.thorough:
+                    if self.end_lh > self.start_lh:
+                        self.restore_tree_fast()
+                        self.start_lh = self.end_lh = ts.likelihood
+                        bt.save(ts, True)
+                    elif self.best_of_node != UNLIKELY:
+                        self.restore_topology_only(bt)
+                else:
+                    self.insert_info_list(st.nodep[i], self.best_of_node)
+                    if self.end_lh > self.start_lh:
+                        self.restore_tree_fast()
+                        self.start_lh = self.end_lh = ts.likelihood
+        if not self.thorough:
+            self.thorough = True
+            for node, _lh in list(self.ilist):
+                self.best_of_node = UNLIKELY
+                if self.rearrange_big(node, mintrav, maxtrav):
+                    if self.end_lh > self.start_lh:
+                        self.restore_tree_fast()
+                        self.start_lh = self.end_lh = ts.likelihood
+                        bt.save(ts, True)
+                    elif self.best_of_node != UNLIKELY:
+                        self.restore_topology_only(bt)
+            self.thorough = False
+        return self.start_lh
+
+    def determine_rearrangement_setting(self, best_t, bt):
+        """determineRearrangementSetting (:1752)."""
+        ts = self.ts
+        MAX_FAST = 26
+        maxtrav, best_trav = 5, 5
+        start_lh = ts.likelihood
+        impr = True
+        cutoff = self.do_cutoff
+        self.do_cutoff = False
+        bt.reset()
+        assert not self.thorough
+        while impr and maxtrav < MAX_FAST:
+            best_t.recall(1, ts)
+            self.node_rectifier()
+            maxtrav_eff = min(maxtrav, self.st.ntips - 3)
+            self.start_lh = self.end_lh = ts.likelihood
+            for i in range(1, 2 * self.st.ntips - 1):
+                self.best_of_node = UNLIKELY
+                if self.rearrange_big(self.st.nodep[i], 1, maxtrav_eff):
+                    if self.end_lh > self.start_lh:
+                        self.restore_tree_fast()
+                        self.start_lh = self.end_lh = ts.likelihood
+            ts.tree_evaluate(0.25)
+            bt.save(ts, True)
+            if ts.likelihood > start_lh:
+                start_lh = ts.likelihood
+                best_trav = maxtrav
+                impr = True
+            else:
+                impr = False
+            if self.do_cutoff:
+                self.lh_cutoff = _cdiv(self.lh_avg, self.lh_dec)
+                self.it_count += 1
+                self.lh_avg = 0.0
+                self.lh_dec = 0
+            maxtrav += 5
+            self.log(f"rearrangement radius {maxtrav - 5}: "
+                     f"{ts.likelihood:.6f}")
+        bt.recall(1, ts)
+        self.do_cutoff = cutoff
+        return best_trav
+
+    def compute_big_rapid(self, estimate_model=True, initial_trav=None):
+        """computeBIGRAPID (:1914), no convergence-criterion hashing."""
+        ts, st = self.ts, self.st
+        self.lh_avg = 0.0
+        self.lh_dec = 0
+        best_t = BestList(1, st)
+        bt = BestList(20, st)
+        epsilon = 0.01
+        self.thorough = False
+        # main()'s preamble before computeBIGRAPID (axml.c:2760-2764)
+        ts.evaluate_generic(full=True)
+        ts.tree_evaluate(1.0)
+        if estimate_model:
+            ts.mod_opt(10.0)
+        else:
+            ts.tree_evaluate(2.0)
+        best_t.save(ts, True)
+        if initial_trav is None:
+            best_trav = self.determine_rearrangement_setting(best_t, bt)
+            self.log(f"best rearrangement radius: {best_trav}")
+        else:
+            best_trav = initial_trav
+        if estimate_model:
+            ts.mod_opt(5.0)
+        else:
+            ts.tree_evaluate(1.0)
+        best_t.save(ts, True)
+        impr = True
+        if self.do_cutoff:
+            self.it_count = 0
+        fast_iterations = 0
+        while impr:
+            best_t.recall(1, ts)
+            fast_iterations += 1
+            ts.tree_evaluate(1.0)
+            best_t.save(ts, True)
+            lh = previous_lh = ts.likelihood
+            self.log(f"fast SPR cycle {fast_iterations}: {lh:.6f}")
+            self.tree_optimize_rapid(1, best_trav, bt)
+            impr = False
+            for i in range(1, bt.nvalid + 1):
+                bt.recall(i, ts)
+                ts.tree_evaluate(0.25)
+                difference = abs(ts.likelihood - previous_lh)
+                if ts.likelihood > lh and difference > epsilon:
+                    impr = True
+                    lh = ts.likelihood
+                    best_t.save(ts, True)
+        self.thorough = True
+        impr = True
+        best_t.recall(1, ts)
+        ts.evaluate_generic(full=True)
+        if estimate_model:
+            ts.mod_opt(1.0)
+        else:
+            ts.tree_evaluate(1.0)
+        rearrangements_min = 1
+        rearrangements_max = self.stepwidth
+        thorough_iterations = 0
+        while True:
+            best_t.recall(1, ts)
+            if impr:
+                rearrangements_min = 1
+                rearrangements_max = self.stepwidth
+                thorough_iterations += 1
+            else:
+                rearrangements_max += self.stepwidth
+                rearrangements_min += self.stepwidth
+                if rearrangements_max > self.max_rearrange:
+                    break
+            ts.tree_evaluate(1.0)
+            previous_lh = lh = ts.likelihood
+            self.log(f"thorough SPR cycle {thorough_iterations} "
+                     f"[{rearrangements_min},{rearrangements_max}]: "
+                     f"{lh:.6f}")
+            self.tree_optimize_rapid(rearrangements_min, rearrangements_max,
+                                     bt)
+            impr = False
+            for i in range(1, bt.nvalid + 1):
+                bt.recall(i, ts)
+                ts.tree_evaluate(0.25)
+                difference = abs(ts.likelihood - previous_lh)
+                if ts.likelihood > lh and difference > epsilon:
+                    impr = True
+                    lh = ts.likelihood
+                    best_t.save(ts, True)
+        ts.evaluate_generic(full=True)
+        self.log(f"likelihood of best tree: {ts.likelihood:.6f}")
+        return ts.likelihood
