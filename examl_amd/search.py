@@ -1137,15 +1137,23 @@ class TreeSearch:
                     new_cptr[i] = best
             eng.set_site_rates(new_cptr, np.array(rc_rates[:num]))
 
-        # updatePerSiteRates (:2060), numBranches=1: global mean-1 rescale
-        wsum = rsum = 0.0
-        for m, eng in enumerate(self.engines):
-            w = eng.host_wgt
-            rsum += float((w * eng.per_site_rates[eng.cptr]).sum())
-            wsum += float(w.sum())
-        scaler = 1.0 / (rsum / wsum)
-        for eng in self.engines:
-            eng.set_site_rates(eng.cptr, eng.per_site_rates * scaler)
+        # updatePerSiteRates (:2060): per-partition mean-1 rescale under
+        # -M (numBranches > 1), one global scaler otherwise
+        if self.NB > 1:
+            for eng in self.engines:
+                w = eng.host_wgt
+                scaler = float(w.sum()) / \
+                    float((w * eng.per_site_rates[eng.cptr]).sum())
+                eng.set_site_rates(eng.cptr, eng.per_site_rates * scaler)
+        else:
+            wsum = rsum = 0.0
+            for m, eng in enumerate(self.engines):
+                w = eng.host_wgt
+                rsum += float((w * eng.per_site_rates[eng.cptr]).sum())
+                wsum += float(w.sum())
+            scaler = 1.0 / (rsum / wsum)
+            for eng in self.engines:
+                eng.set_site_rates(eng.cptr, eng.per_site_rates * scaler)
 
         self.evaluate_generic(full=True)
         if self.likelihood < initial_lh:

@@ -78,3 +78,41 @@ def test_full_psr_f_E_pipeline_gpu(golden_dir):
         rate_het="CAT")
     lnl = ts.tree_evaluation_mode()
     assert abs(lnl - GOLDEN_FINAL_LNL) < TOL_ABS, lnl
+
+
+def test_full_psr_M_f_E_pipeline_cpu_oracle(golden_dir):
+    """-m PSR with -M (per-partition branch lengths): exercises the
+    vectorized NR under CAT and updatePerSiteRates' per-partition
+    rescale branch (optimizeModel.c:2072-2082).  Reference golden
+    -14532.602263; our replay lands at 1.5e-11 relative (~70 s)."""
+    from tests.helpers import OracleCatEngine
+    parts, tree = _load(golden_dir)
+    ts = TreeSearch(tree, _engines(parts, OracleCatEngine),
+                    opt_freq_flags=[bool(p.optimizeBaseFrequencies)
+                                    for p in parts],
+                    rate_het="CAT", per_gene_bl=True)
+    lnl = ts.tree_evaluation_mode()
+    golden = -14532.602263
+    assert abs(lnl - golden) < abs(golden) * 1e-6, lnl
+    # per-partition weighted mean rate == 1 under -M (checkPerSiteRates)
+    for e in ts.engines:
+        mean = float((e.host_wgt * e.per_site_rates[e.cptr]).sum()
+                     / e.host_wgt.sum())
+        assert abs(mean - 1.0) < 1e-5
+
+
+@pytest.mark.gpu
+def test_full_psr_M_f_E_pipeline_gpu(golden_dir):
+    import torch
+    assert torch.cuda.is_available()
+    parts, tree = _load(golden_dir)
+    ts = TreeSearch(
+        tree,
+        _engines(parts, lambda tips, wgt, model, cptr, rates:
+                 ea.DnaCatEngine(tips, wgt, model, cptr, rates,
+                                 device="cuda:0")),
+        opt_freq_flags=[bool(p.optimizeBaseFrequencies) for p in parts],
+        rate_het="CAT", per_gene_bl=True)
+    lnl = ts.tree_evaluation_mode()
+    golden = -14532.602263
+    assert abs(lnl - golden) < abs(golden) * 1e-6, lnl
