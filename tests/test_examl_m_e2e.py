@@ -43,6 +43,44 @@ def _run(parts, tree, engine_cls):
     return lnl
 
 
+GOLDEN_START = -17817.521734    # after treeEvaluate(1), reference -M trace
+GOLDEN_PASS_1_END = -16190.169307  # start of pass 2 in the same trace
+
+
+def test_M_f_E_first_pass_cpu_oracle(golden_dir):
+    """Bounded -M coverage (~60 s): treeEvaluate(1) + modOpt pass 1 land
+    on the reference's own -M trace; the full pipeline (2.9e-11 rel of
+    -16035.202133, ~4 min) is the opt-in test below and runs whole on
+    the GPU."""
+    from tests.helpers import OracleEngine
+    from examl_amd.search import TreeSearch
+    parts, tree = _load(golden_dir)
+    engines = [OracleEngine(p.tips, p.wgt,
+                            ea.DnaGtrModel(p.frequencies, [1.0] * 6, 1.0))
+               for p in parts]
+    ts = TreeSearch(
+        tree, engines,
+        opt_freq_flags=[bool(p.optimizeBaseFrequencies) for p in parts],
+        per_gene_bl=True)
+    ts.evaluate_generic(full=True)
+    start = ts.tree_evaluate(1.0)
+    assert abs(start - GOLDEN_START) < 5e-6, start
+    ts.opt_rates_generic(0.0001)
+    ts.evaluate_generic(full=True)
+    ts.tree_evaluate(0.0625)
+    ts.evaluate_generic(full=True)
+    ts.opt_base_freqs(0.0001)
+    ts.evaluate_generic(full=True)
+    ts.tree_evaluate(0.0625)
+    ts.opt_alphas_generic(0.0001)
+    ts.evaluate_generic(full=True)
+    ts.tree_evaluate(0.1)
+    assert abs(ts.likelihood - GOLDEN_PASS_1_END) < 5e-6, ts.likelihood
+
+
+@pytest.mark.skipif(not os.environ.get("EXAML_E2E_M"),
+                    reason="full -M -f E on CPU oracle (~4 min): set "
+                           "EXAML_E2E_M=1")
 def test_full_M_f_E_pipeline_cpu_oracle(golden_dir):
     from tests.helpers import OracleEngine
     parts, tree = _load(golden_dir)
