@@ -109,11 +109,13 @@ def read_newick_topology(path, taxa_names):
     return parse_newick_topology(s, taxa_names)
 
 
-def parse_newick_topology(s, taxa_names):
+def parse_newick_topology(s, taxa_names, read_bl=False):
     """Parse a (possibly multifurcating-root) Newick tree over the given
     taxa into a PhyloTree, branch lengths defaulting to z = 0.9 (defaultz,
-    treeReadLen behavior for topology-only trees).  Ring order of inner
-    nodes follows parse order (the reference's p->next chain)."""
+    treeReadLen behavior for topology-only trees); with read_bl the
+    branch lengths are read back as z = exp(-bl) (the inverse of
+    getBranchLength, treeIO.c:176).  Ring order of inner nodes follows
+    parse order (the reference's p->next chain)."""
     s = s.strip()
     if s.endswith(";"):
         s = s[:-1]
@@ -166,7 +168,10 @@ def parse_newick_topology(s, taxa_names):
             while k < len(s) and s[k] not in ",();":
                 k += 1
             # branch length present in file: ExaML's topology-only flow
-            # ignores it unless -t with BLs; we keep defaultz semantics
+            # ignores it (defaultz) unless read_bl is requested
+            if read_bl:
+                import math
+                z = math.exp(-float(s[j:k]))
             pos[0] = k
         return z
 
@@ -198,3 +203,38 @@ def parse_newick_topology(s, taxa_names):
         raise ValueError("unsupported root degree")
     assert next_inner[0] <= 2 * ntips - 1
     return tree
+
+
+def to_newick(tree, taxa_names, digits=20):
+    """Write the tree with branch lengths as the reference's Tree2String
+    does (treeIO.c:234, branch length = -log(z), SUMMARIZE_LH averages
+    the per-partition -log z under -M), rooted at tr->start->back with
+    the start tip as the first child."""
+    import math
+
+    def bl(a, b):
+        zv = tree.get_zv(a, b)
+        x = 0.0
+        for z in zv:
+            z = max(z, 1.0e-15)
+            x += -math.log(z)
+        return x / len(zv)
+
+    def sub(p, parent):
+        if tree.is_tip(p):
+            return f"{taxa_names[p - 1]}:{bl(p, parent):.{digits}f}"
+        if hasattr(tree, "ring_children"):
+            kids = tree.ring_children(p, parent)
+        else:
+            kids = [w for w in tree.adj[p] if w != parent]
+        inner = ",".join(sub(w, p) for w in kids)
+        return f"({inner}):{bl(p, parent):.{digits}f}"
+
+    start = 1
+    back = next(iter(tree.adj[start]))
+    if hasattr(tree, "ring_children"):
+        kids = tree.ring_children(back, start)
+    else:
+        kids = [w for w in tree.adj[back] if w != start]
+    parts = [sub(start, back)] + [sub(w, back) for w in kids]
+    return "(" + ",".join(parts) + ");"

@@ -74,3 +74,22 @@ def test_full_f_E_pipeline_gpu(golden_dir):
                      tips, wgt, model, device="cuda:0"))
     lnl = ts.tree_evaluation_mode()
     assert abs(lnl - GOLDEN_FINAL_LNL) < TOL_ABS, lnl
+
+
+def test_result_tree_matches_reference(golden_dir, optimized_49_cpu):
+    """The optimized tree (topology + branch lengths) matches the
+    reference's own ExaML_TreeFile output for the same -f E run
+    (tests/golden/49.result.tree): identical edges, branch lengths
+    within 1e-6 (to_newick writes the reference's -log(z) form)."""
+    import os
+    from examl_amd.examl_io import parse_newick_topology, to_newick
+    ts, _ = optimized_49_cpu
+    taxa, parts = read_byte_file(os.path.join(golden_dir, "49.binary"))
+    ref = parse_newick_topology(
+        open(os.path.join(golden_dir, "49.result.tree")).read(), taxa,
+        read_bl=True)
+    ours = parse_newick_topology(to_newick(ts.tree, taxa), taxa,
+                                 read_bl=True)
+    assert sorted(ref.edges()) == sorted(ours.edges())
+    for a, b in ref.edges():
+        assert abs(ref.get_z(a, b) - ts.tree.get_z(a, b)) < 1e-6, (a, b)
