@@ -561,13 +561,10 @@ class SaveDnaEngine(DnaGammaEngine):
                                 r_tip),
                 self._gapcol_of(e.x3Slot, False), _vp(self.d_scalegap), s),
                 "newview_dna_save")
-            # recursive scaler accumulation on host via the shared finalize
-            # kernel is overkill here; do it with the same kernel used by
-            # the dense path (one op at a time)
             self._finalize_scaler_one(p, q, r)
-        # d_inc consumed per op inside _finalize_scaler_one
 
     def tree_slot(self, slot):
+        # gap-column indices are CLV-slot based for inner nodes
         return slot
 
     def _finalize_scaler_one(self, p, q, r):

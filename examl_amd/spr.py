@@ -217,7 +217,7 @@ class Topol:
         self.start = None
 
 
-def _save_subtree(p, tpl):
+def _save_subtree(p, tpl, ntips):
     """saveSubtree (topologies.c:229): returns the link index for branch
     p--p->back with children merged in ascending min-tip order."""
     links = tpl.links
@@ -225,12 +225,12 @@ def _save_subtree(p, tpl):
     ri = len(links)
     links.append(r)
     q = p.back
-    if q.number <= tpl_ntips[0]:
+    if q.number <= ntips:
         r[5] = q.number
     else:
         s = q.next
         while True:
-            t = _save_subtree(s, tpl)
+            t = _save_subtree(s, tpl, ntips)
             t0, t1 = 0, r[3]
             while t1 and links[t1][5] < links[t][5]:
                 t0 = t1
@@ -245,9 +245,6 @@ def _save_subtree(p, tpl):
                 break
         r[5] = links[r[3]][5]
     return ri
-
-
-tpl_ntips = [0]  # module-scope like the reference's numsp parameter
 
 
 def _min_subtree_tip(p0, ntips):
@@ -273,9 +270,8 @@ def _min_tree_tip(p, ntips):
 def save_tree(st, likelihood, tpl):
     """saveTree (topologies.c:291)."""
     tpl.links = []
-    tpl_ntips[0] = st.ntips
     start_m = st.find_member(st.start, next(iter(st.adj[st.start])))
-    _save_subtree(_min_tree_tip(start_m, st.ntips), tpl)
+    _save_subtree(_min_tree_tip(start_m, st.ntips), tpl, st.ntips)
     tpl.likelihood = likelihood
     tpl.start = st.start
 
