@@ -1892,3 +1892,239 @@ EXPORT void oracle_sum_prot_gamma_save(
             (tipCase == ORC_TIP_TIP ? ri[k] : ri[l * 20 + k]);
   }
 }
+
+/* ==========================================================================
+ * -S CAT kernels (DNA span 4 and protein span 20): gap-compacted CLVs with
+ * per-site rate categories plus an EXTRA P-matrix pair at rate 1.0 in slot
+ * maxCats for the gap column / gap operands (makeP's saveMem branch,
+ * newviewGenericSpecial.c:140-165).  Restate
+ * newviewGTRCAT_AVX_GAPPED_SAVE (avxLikelihood.c:2306),
+ * newviewGTRCATPROT_AVX_GAPPED_SAVE (:2607), evaluateGTRCAT_SAVE
+ * (evaluateGenericSpecial.c:1537-area), evaluateGTRCATPROT_SAVE (:1537),
+ * sumCAT_SAVE / sumGTRCATPROT_SAVE (makenewzGenericSpecial.c:1648/...).
+ * ==========================================================================*/
+
+EXPORT void oracle_make_p_save(double z1, double z2, const double *rptr,
+                               const double *EI, const double *EIGN,
+                               int numCats, double *left, double *right,
+                               int maxCats, int states) {
+  int i, j, k;
+  double d1[64], d2[64], lz1[64], lz2[64];
+  for (j = 1; j < states; j++) {
+    lz1[j] = EIGN[j] * z1;
+    lz2[j] = EIGN[j] * z2;
+  }
+  const int sq = states * states;
+  for (i = 0; i < numCats; i++) {
+    for (j = 1; j < states; j++) {
+      d1[j] = exp(rptr[i] * lz1[j]);
+      d2[j] = exp(rptr[i] * lz2[j]);
+    }
+    for (j = 0; j < states; j++) {
+      left[sq * i + states * j] = 1.0;
+      right[sq * i + states * j] = 1.0;
+      for (k = 1; k < states; k++) {
+        left[sq * i + states * j + k] = d1[k] * EI[states * j + k];
+        right[sq * i + states * j + k] = d2[k] * EI[states * j + k];
+      }
+    }
+  }
+  /* saveMem extra pair at rate 1.0 (slot maxCats) */
+  i = maxCats;
+  for (j = 1; j < states; j++) {
+    d1[j] = exp(lz1[j]);
+    d2[j] = exp(lz2[j]);
+  }
+  for (j = 0; j < states; j++) {
+    left[sq * i + states * j] = 1.0;
+    right[sq * i + states * j] = 1.0;
+    for (k = 1; k < states; k++) {
+      left[sq * i + states * j + k] = d1[k] * EI[states * j + k];
+      right[sq * i + states * j + k] = d2[k] * EI[states * j + k];
+    }
+  }
+}
+
+/* per-site body shared with the dense CAT kernel: hadd4 pairwise dots */
+static void orc_nv_cat_site(const double *x1, const double *x2,
+                            const double *le, const double *ri,
+                            const double *EV, double *xv) {
+  int l, s;
+  for (s = 0; s < 4; s++) xv[s] = 0.0;
+  for (l = 0; l < 4; l++) {
+    const double a = (x1[0] * le[l * 4] + x1[1] * le[l * 4 + 1]) +
+                     (x1[2] * le[l * 4 + 2] + x1[3] * le[l * 4 + 3]);
+    const double b = (x2[0] * ri[l * 4] + x2[1] * ri[l * 4 + 1]) +
+                     (x2[2] * ri[l * 4 + 2] + x2[3] * ri[l * 4 + 3]);
+    const double t = a * b;
+    for (s = 0; s < 4; s++) xv[s] += t * EV[l * 4 + s];
+  }
+}
+
+EXPORT void oracle_newview_dna_cat_save(
+    int tipCase, const double *EV, const int *cptr, const double *x1_start,
+    const double *x2_start, double *x3_start, const double *tipVector,
+    const unsigned char *tipX1, const unsigned char *tipX2, int n,
+    const double *left, const double *right, const int *wgt,
+    int *scalerIncrement, const unsigned int *x1_gap,
+    const unsigned int *x2_gap, const unsigned int *x3_gap,
+    const double *x1_gapColumn, const double *x2_gapColumn,
+    double *x3_gapColumn, int maxCats) {
+  int i, s, scale;
+  int addScale = 0, scaleGap = 0;
+  const double *x1_ptr = x1_start, *x2_ptr = x2_start;
+  double *x3_ptr = x3_start;
+
+  /* gap column with the rate-1.0 P pair (avx:2332-2378) */
+  {
+    double xv[4];
+    orc_nv_cat_site(x1_gapColumn, x2_gapColumn, &left[maxCats * 16],
+                    &right[maxCats * 16], EV, xv);
+    if (tipCase != ORC_TIP_TIP) {
+      scale = 1;
+      for (s = 0; s < 4; s++)
+        if (!(fabs(xv[s]) < ORC_MINLIKELIHOOD)) { scale = 0; break; }
+      if (scale) {
+        for (s = 0; s < 4; s++) xv[s] *= ORC_TWOTOTHE256;
+        scaleGap = 1;
+      }
+    }
+    for (s = 0; s < 4; s++) x3_gapColumn[s] = xv[s];
+  }
+
+  for (i = 0; i < n; i++) {
+    if (orc_is_gap(x3_gap, i)) {
+      if (tipCase != ORC_TIP_TIP && scaleGap) addScale += wgt[i];
+      continue;
+    }
+    {
+      const double *x1, *x2, *le, *ri;
+      double xv[4];
+      if (tipCase == ORC_TIP_TIP) {
+        x1 = &tipVector[4 * tipX1[i]];
+        x2 = &tipVector[4 * tipX2[i]];
+        le = orc_is_gap(x1_gap, i) ? &left[maxCats * 16]
+                                   : &left[cptr[i] * 16];
+        ri = orc_is_gap(x2_gap, i) ? &right[maxCats * 16]
+                                   : &right[cptr[i] * 16];
+      } else if (tipCase == ORC_TIP_INNER) {
+        x1 = &tipVector[4 * tipX1[i]];
+        le = orc_is_gap(x1_gap, i) ? &left[maxCats * 16]
+                                   : &left[cptr[i] * 16];
+        if (orc_is_gap(x2_gap, i)) {
+          ri = &right[maxCats * 16];
+          x2 = x2_gapColumn;
+        } else {
+          ri = &right[cptr[i] * 16];
+          x2 = x2_ptr;
+          x2_ptr += 4;
+        }
+      } else {
+        if (orc_is_gap(x1_gap, i)) {
+          x1 = x1_gapColumn;
+          le = &left[maxCats * 16];
+        } else {
+          le = &left[cptr[i] * 16];
+          x1 = x1_ptr;
+          x1_ptr += 4;
+        }
+        if (orc_is_gap(x2_gap, i)) {
+          x2 = x2_gapColumn;
+          ri = &right[maxCats * 16];
+        } else {
+          ri = &right[cptr[i] * 16];
+          x2 = x2_ptr;
+          x2_ptr += 4;
+        }
+      }
+      orc_nv_cat_site(x1, x2, le, ri, EV, xv);
+      if (tipCase != ORC_TIP_TIP) {
+        scale = 1;
+        for (s = 0; s < 4; s++)
+          if (!(fabs(xv[s]) < ORC_MINLIKELIHOOD)) { scale = 0; break; }
+        if (scale) {
+          for (s = 0; s < 4; s++) xv[s] *= ORC_TWOTOTHE256;
+          addScale += wgt[i];
+        }
+      }
+      for (s = 0; s < 4; s++) x3_ptr[s] = xv[s];
+      x3_ptr += 4;
+    }
+  }
+  *scalerIncrement = addScale;
+}
+
+EXPORT double oracle_evaluate_dna_cat_save(
+    const int *cptr, const int *wptr, const double *x1_start,
+    const double *x2_start, const double *tipVector,
+    const unsigned char *tipX1, int n, const double *diagptable,
+    const double *x1_gapColumn, const double *x2_gapColumn,
+    const unsigned int *x1_gap, const unsigned int *x2_gap) {
+  double sum = 0.0;
+  int i;
+  const double *x1_ptr = x1_start, *x2_ptr = x2_start;
+  for (i = 0; i < n; i++) {
+    const double *x1, *x2;
+    if (tipX1) {
+      x1 = &tipVector[4 * tipX1[i]];
+    } else if (orc_is_gap(x1_gap, i)) {
+      x1 = x1_gapColumn;
+    } else {
+      x1 = x1_ptr;
+      x1_ptr += 4;
+    }
+    if (orc_is_gap(x2_gap, i)) {
+      x2 = x2_gapColumn;
+    } else {
+      x2 = x2_ptr;
+      x2_ptr += 4;
+    }
+    const double *d = &diagptable[4 * cptr[i]];
+    const double t0 = x1[0] * x2[0] * d[0] + x1[2] * x2[2] * d[2];
+    const double t1 = x1[1] * x2[1] * d[1] + x1[3] * x2[3] * d[3];
+    sum += wptr[i] * log(fabs(t0 + t1));
+  }
+  return sum;
+}
+
+EXPORT void oracle_sum_dna_cat_save(
+    int tipCase, double *sumtable, const double *x1_start,
+    const double *x2_start, const double *tipVector,
+    const unsigned char *tipX1, const unsigned char *tipX2, int n,
+    const double *x1_gapColumn, const double *x2_gapColumn,
+    const unsigned int *x1_gap, const unsigned int *x2_gap) {
+  int i, j;
+  const double *x1_ptr = x1_start, *x2_ptr = x2_start;
+  for (i = 0; i < n; i++) {
+    const double *x1, *x2;
+    switch (tipCase) {
+    case ORC_TIP_TIP:
+      x1 = &tipVector[4 * tipX1[i]];
+      x2 = &tipVector[4 * tipX2[i]];
+      break;
+    case ORC_TIP_INNER:
+      x1 = &tipVector[4 * tipX1[i]];
+      if (orc_is_gap(x2_gap, i))
+        x2 = x2_gapColumn;
+      else {
+        x2 = x2_ptr;
+        x2_ptr += 4;
+      }
+      break;
+    default:
+      if (orc_is_gap(x1_gap, i)) {
+        x1 = x1_gapColumn;
+      } else {
+        x1 = x1_ptr;
+        x1_ptr += 4;
+      }
+      if (orc_is_gap(x2_gap, i)) {
+        x2 = x2_gapColumn;
+      } else {
+        x2 = x2_ptr;
+        x2_ptr += 4;
+      }
+    }
+    for (j = 0; j < 4; j++) sumtable[i * 4 + j] = x1[j] * x2[j];
+  }
+}

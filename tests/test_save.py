@@ -309,3 +309,128 @@ def test_prot_save_kernels_bit_exact_vs_reference():
                                       ctypes.c_int(n), dp(go), dp(go2),
                                       up(g3), up(g3b))
     assert np.array_equal(sr, so)
+
+
+@pytest.mark.skipif(not O.have_ref(), reason="reference libref.so not built")
+def test_cat_save_kernels_bit_exact_vs_reference():
+    """DNA PSR (-S + CAT) GAPPED_SAVE kernels vs the reference's
+    newviewGTRCAT_AVX_GAPPED_SAVE (avxLikelihood.c:2306),
+    evaluateGTRCAT_SAVE (evaluateGenericSpecial.c:1537) and sumCAT_SAVE
+    (makenewzGenericSpecial.c:1648), including the saveMem extra
+    rate-1.0 P slot at maxCats (newviewGenericSpecial.c:78 makeP)."""
+    from tests.helpers import _model_arrays
+    rng = np.random.default_rng(9)
+    m = ea.DnaGtrModel([0.3, 0.2, 0.26, 0.24],
+                       [1.2, 2.4, 0.7, 0.9, 3.1, 1.0], 0.8)
+    EIGN, EV, EI, tipVector, g = _model_arrays(m)
+    n, nc, maxc = 200, 5, 7
+    rates = O.aligned(nc)
+    rates[:] = [0.2, 0.6, 1.0, 1.7, 3.0]
+    cptr = rng.integers(0, nc, n).astype(np.int32)
+    t1 = rng.integers(1, 16, n).astype(np.uint8)
+    t1[rng.random(n) < 0.3] = 15
+    t2 = rng.integers(1, 16, n).astype(np.uint8)
+    t2[rng.random(n) < 0.3] = 15
+    wgt = np.ones(n, dtype=np.int32)
+    gvl = n // 32 + 1
+
+    def gap_of(tips):
+        gv = np.zeros(gvl, dtype=np.uint32)
+        idx = np.nonzero(tips == 15)[0]
+        np.bitwise_or.at(gv, idx // 32,
+                         (np.uint32(1) << (idx % 32).astype(np.uint32)))
+        return gv
+
+    g1, g2 = gap_of(t1), gap_of(t2)
+
+    def dp(a):
+        return (a.ctypes.data_as(ctypes.POINTER(ctypes.c_double))
+                if a is not None else
+                ctypes.cast(None, ctypes.POINTER(ctypes.c_double)))
+
+    def u8(a):
+        return (a.ctypes.data_as(ctypes.POINTER(ctypes.c_ubyte))
+                if a is not None else
+                ctypes.cast(None, ctypes.POINTER(ctypes.c_ubyte)))
+
+    def ip(a):
+        return a.ctypes.data_as(ctypes.POINTER(ctypes.c_int))
+
+    def up(a):
+        return (a.ctypes.data_as(ctypes.POINTER(ctypes.c_uint))
+                if a is not None else
+                ctypes.cast(None, ctypes.POINTER(ctypes.c_uint)))
+
+    ref = O._ref
+
+    # makeP saveMem branch: extra P pair at rate 1.0 in slot maxCats
+    left_o = O.aligned((maxc + 1) * 16)
+    right_o = O.aligned((maxc + 1) * 16)
+    O._orc.oracle_make_p_save(
+        ctypes.c_double(-0.2), ctypes.c_double(-0.5), dp(rates), dp(EI),
+        dp(EIGN), ctypes.c_int(nc), dp(left_o), dp(right_o),
+        ctypes.c_int(maxc), ctypes.c_int(4))
+    left_r = O.aligned((maxc + 1) * 16)
+    right_r = O.aligned((maxc + 1) * 16)
+    ref.makeP(ctypes.c_double(-0.2), ctypes.c_double(-0.5), dp(rates),
+              dp(EI), dp(EIGN), ctypes.c_int(nc), dp(left_r), dp(right_r),
+              ctypes.c_int(1), ctypes.c_int(maxc), ctypes.c_int(4))
+    assert np.array_equal(left_o, left_r)
+    assert np.array_equal(right_o, right_r)
+
+    tv_gap = O.aligned(4)
+    tv_gap[:] = tipVector[15 * 4:16 * 4]
+
+    def run_pair(tc, x1r, x1o, x2r, x2o, ga, gb, gca_r, gca_o, gcb_r, gcb_o,
+                 ta, tb):
+        g3 = ga & gb
+        nz = int(n - sum(bin(int(w)).count("1") for w in g3))
+        x3r = O.aligned(nz * 4 + 4)
+        x3o = O.aligned(nz * 4 + 4)
+        gr = O.aligned(4)
+        go = O.aligned(4)
+        ir = ctypes.c_int(0)
+        io = ctypes.c_int(0)
+        ref.newviewGTRCAT_AVX_GAPPED_SAVE(
+            tc, dp(EV), ip(cptr), dp(x1r), dp(x2r), dp(x3r), dp(tipVector),
+            None, u8(ta), u8(tb), ctypes.c_int(n), dp(left_r), dp(right_r),
+            ip(wgt), ctypes.byref(ir), ctypes.c_int(1), up(ga), up(gb),
+            up(g3), dp(gca_r), dp(gcb_r), dp(gr), ctypes.c_int(maxc))
+        O._orc.oracle_newview_dna_cat_save(
+            tc, dp(EV), ip(cptr), dp(x1o), dp(x2o), dp(x3o), dp(tipVector),
+            u8(ta), u8(tb), ctypes.c_int(n), dp(left_o), dp(right_o),
+            ip(wgt), ctypes.byref(io), up(ga), up(gb), up(g3), dp(gca_o),
+            dp(gcb_o), dp(go), ctypes.c_int(maxc))
+        assert np.array_equal(x3r[:nz * 4], x3o[:nz * 4])
+        assert np.array_equal(gr, go)
+        assert ir.value == io.value
+        return g3, x3r, x3o, gr, go
+
+    g3, x3r, x3o, gcr, gco = run_pair(0, None, None, None, None, g1, g2,
+                                      tv_gap, tv_gap, tv_gap, tv_gap, t1, t2)
+    g3b, x3r2, x3o2, gcr2, gco2 = run_pair(1, None, None, x3r, x3o, g1, g3,
+                                           tv_gap, tv_gap, gcr, gco, t1,
+                                           None)
+    run_pair(2, x3r, x3o, x3r2, x3o2, g3, g3b, gcr, gco, gcr2, gco2, None,
+             None)
+
+    diag = O.calc_diagptable(0.7, 4, nc, rates, EIGN)
+    ref.evaluateGTRCAT_SAVE.restype = ctypes.c_double
+    O._orc.oracle_evaluate_dna_cat_save.restype = ctypes.c_double
+    lr = ref.evaluateGTRCAT_SAVE(
+        ip(cptr), ip(wgt), dp(x3r), dp(x3r2), dp(tipVector), None,
+        ctypes.c_int(n), dp(diag), dp(gcr), dp(gcr2), up(g3), up(g3b))
+    lo = O._orc.oracle_evaluate_dna_cat_save(
+        ip(cptr), ip(wgt), dp(x3o), dp(x3o2), dp(tipVector), u8(None),
+        ctypes.c_int(n), dp(diag), dp(gco), dp(gco2), up(g3), up(g3b))
+    assert lr == lo
+
+    sr = O.aligned(n * 4)
+    so = O.aligned(n * 4)
+    ref.sumCAT_SAVE(2, dp(sr), dp(x3r), dp(x3r2), dp(tipVector), None, None,
+                    ctypes.c_int(n), dp(gcr), dp(gcr2), up(g3), up(g3b))
+    O._orc.oracle_sum_dna_cat_save(2, dp(so), dp(x3o), dp(x3o2),
+                                   dp(tipVector), u8(None), u8(None),
+                                   ctypes.c_int(n), dp(gco), dp(gco2),
+                                   up(g3), up(g3b))
+    assert np.array_equal(sr, so)
