@@ -2128,3 +2128,196 @@ EXPORT void oracle_sum_dna_cat_save(
     for (j = 0; j < 4; j++) sumtable[i * 4 + j] = x1[j] * x2[j];
   }
 }
+
+/* ==========================================================================
+ * -S protein PSR (CAT) GAPPED_SAVE kernels: span-20 compacted CLVs + gap
+ * columns, per-site P by rate category with the saveMem rate-1.0 pair at
+ * slot maxCats.  Restate newviewGTRCATPROT_AVX_GAPPED_SAVE
+ * (avxLikelihood.c:2607), evaluateGTRCATPROT_SAVE
+ * (evaluateGenericSpecial.c:1537), sumGTRCATPROT_SAVE
+ * (makenewzGenericSpecial.c:2218).  The AVX per-site body is
+ * hadd4(dot20, dot20) per row l — the dot20_avx lane order.
+ * ==========================================================================*/
+
+static void orc_nv_prot_cat_site(const double *vl, const double *vr,
+                                 const double *le, const double *ri,
+                                 const double *extEV, double *xv) {
+  int l, s;
+  for (s = 0; s < 20; s++) xv[s] = 0.0;
+  for (l = 0; l < 20; l++) {
+    const double t =
+        dot20_avx(vl, &le[l * 20]) * dot20_avx(vr, &ri[l * 20]);
+    for (s = 0; s < 20; s++) xv[s] += t * extEV[l * 20 + s];
+  }
+}
+
+EXPORT void oracle_newview_prot_cat_save(
+    int tipCase, const double *extEV, const int *cptr,
+    const double *x1_start, const double *x2_start, double *x3_start,
+    const double *tipVector, const unsigned char *tipX1,
+    const unsigned char *tipX2, int n, const double *left,
+    const double *right, const int *wgt, int *scalerIncrement,
+    const unsigned int *x1_gap, const unsigned int *x2_gap,
+    const unsigned int *x3_gap, const double *x1_gapColumn,
+    const double *x2_gapColumn, double *x3_gapColumn, int maxCats) {
+  int i, s, scale;
+  int addScale = 0, scaleGap = 0;
+  const double *x1_ptr = x1_start, *x2_ptr = x2_start;
+  double *x3_ptr = x3_start;
+
+  /* gap column with the rate-1.0 P pair (avx:2636-2733) */
+  {
+    double xv[20];
+    orc_nv_prot_cat_site(x1_gapColumn, x2_gapColumn, &left[maxCats * 400],
+                         &right[maxCats * 400], extEV, xv);
+    if (tipCase != ORC_TIP_TIP) {
+      scale = 1;
+      for (s = 0; s < 20; s++)
+        if (!(fabs(xv[s]) < ORC_MINLIKELIHOOD)) { scale = 0; break; }
+      if (scale) {
+        for (s = 0; s < 20; s++) xv[s] *= ORC_TWOTOTHE256;
+        scaleGap = 1;
+      }
+    }
+    for (s = 0; s < 20; s++) x3_gapColumn[s] = xv[s];
+  }
+
+  for (i = 0; i < n; i++) {
+    if (orc_is_gap(x3_gap, i)) {
+      if (tipCase != ORC_TIP_TIP && scaleGap) addScale += wgt[i];
+      continue;
+    }
+    {
+      const double *vl, *vr, *le, *ri;
+      double xv[20];
+      if (tipCase == ORC_TIP_TIP) {
+        vl = &tipVector[20 * tipX1[i]];
+        vr = &tipVector[20 * tipX2[i]];
+        le = orc_is_gap(x1_gap, i) ? &left[maxCats * 400]
+                                   : &left[cptr[i] * 400];
+        ri = orc_is_gap(x2_gap, i) ? &right[maxCats * 400]
+                                   : &right[cptr[i] * 400];
+      } else if (tipCase == ORC_TIP_INNER) {
+        vl = &tipVector[20 * tipX1[i]];
+        le = orc_is_gap(x1_gap, i) ? &left[maxCats * 400]
+                                   : &left[cptr[i] * 400];
+        if (orc_is_gap(x2_gap, i)) {
+          ri = &right[maxCats * 400];
+          vr = x2_gapColumn;
+        } else {
+          ri = &right[cptr[i] * 400];
+          vr = x2_ptr;
+          x2_ptr += 20;
+        }
+      } else {
+        if (orc_is_gap(x1_gap, i)) {
+          vl = x1_gapColumn;
+          le = &left[maxCats * 400];
+        } else {
+          le = &left[cptr[i] * 400];
+          vl = x1_ptr;
+          x1_ptr += 20;
+        }
+        if (orc_is_gap(x2_gap, i)) {
+          vr = x2_gapColumn;
+          ri = &right[maxCats * 400];
+        } else {
+          ri = &right[cptr[i] * 400];
+          vr = x2_ptr;
+          x2_ptr += 20;
+        }
+      }
+      orc_nv_prot_cat_site(vl, vr, le, ri, extEV, xv);
+      if (tipCase != ORC_TIP_TIP) {
+        scale = 1;
+        for (s = 0; s < 20; s++)
+          if (!(fabs(xv[s]) < ORC_MINLIKELIHOOD)) { scale = 0; break; }
+        if (scale) {
+          for (s = 0; s < 20; s++) xv[s] *= ORC_TWOTOTHE256;
+          addScale += wgt[i];
+        }
+      }
+      for (s = 0; s < 20; s++) x3_ptr[s] = xv[s];
+      x3_ptr += 20;
+    }
+  }
+  *scalerIncrement = addScale;
+}
+
+EXPORT double oracle_evaluate_prot_cat_save(
+    const int *cptr, const int *wptr, const double *x1_start,
+    const double *x2_start, const double *tipVector,
+    const unsigned char *tipX1, int n, const double *diagptable,
+    const double *x1_gapColumn, const double *x2_gapColumn,
+    const unsigned int *x1_gap, const unsigned int *x2_gap) {
+  double sum = 0.0;
+  int i, l;
+  const double *x1_ptr = x1_start, *x2_ptr = x2_start;
+  for (i = 0; i < n; i++) {
+    const double *le, *ri;
+    if (tipX1) {
+      le = &tipVector[20 * tipX1[i]];
+    } else if (orc_is_gap(x1_gap, i)) {
+      le = x1_gapColumn;
+    } else {
+      le = x1_ptr;
+      x1_ptr += 20;
+    }
+    if (orc_is_gap(x2_gap, i)) {
+      ri = x2_gapColumn;
+    } else {
+      ri = x2_ptr;
+      x2_ptr += 20;
+    }
+    const double *d = &diagptable[20 * cptr[i]];
+    double t0 = 0.0, t1 = 0.0;
+    for (l = 0; l < 20; l += 2) {
+      t0 += le[l] * ri[l] * d[l];
+      t1 += le[l + 1] * ri[l + 1] * d[l + 1];
+    }
+    sum += wptr[i] * log(fabs(t0 + t1));
+  }
+  return sum;
+}
+
+EXPORT void oracle_sum_prot_cat_save(
+    int tipCase, double *sumtable, const double *x1_start,
+    const double *x2_start, const double *tipVector,
+    const unsigned char *tipX1, const unsigned char *tipX2, int n,
+    const double *x1_gapColumn, const double *x2_gapColumn,
+    const unsigned int *x1_gap, const unsigned int *x2_gap) {
+  int i, j;
+  const double *x1_ptr = x1_start, *x2_ptr = x2_start;
+  for (i = 0; i < n; i++) {
+    const double *le, *ri;
+    switch (tipCase) {
+    case ORC_TIP_TIP:
+      le = &tipVector[20 * tipX1[i]];
+      ri = &tipVector[20 * tipX2[i]];
+      break;
+    case ORC_TIP_INNER:
+      le = &tipVector[20 * tipX1[i]];
+      if (orc_is_gap(x2_gap, i))
+        ri = x2_gapColumn;
+      else {
+        ri = x2_ptr;
+        x2_ptr += 20;
+      }
+      break;
+    default:
+      if (orc_is_gap(x1_gap, i)) {
+        le = x1_gapColumn;
+      } else {
+        le = x1_ptr;
+        x1_ptr += 20;
+      }
+      if (orc_is_gap(x2_gap, i)) {
+        ri = x2_gapColumn;
+      } else {
+        ri = x2_ptr;
+        x2_ptr += 20;
+      }
+    }
+    for (j = 0; j < 20; j++) sumtable[i * 20 + j] = le[j] * ri[j];
+  }
+}

@@ -434,3 +434,142 @@ def test_cat_save_kernels_bit_exact_vs_reference():
                                    ctypes.c_int(n), dp(gco), dp(gco2),
                                    up(g3), up(g3b))
     assert np.array_equal(sr, so)
+
+
+@pytest.mark.skipif(not O.have_ref(), reason="reference libref.so not built")
+def test_prot_cat_save_kernels_bit_exact_vs_reference():
+    """Protein PSR (-S + CAT, span 20) GAPPED_SAVE kernels vs the
+    reference's newviewGTRCATPROT_AVX_GAPPED_SAVE (avxLikelihood.c:2607),
+    evaluateGTRCATPROT_SAVE (evaluateGenericSpecial.c:1537) and
+    sumGTRCATPROT_SAVE (makenewzGenericSpecial.c:2218), sharing the
+    saveMem makeP rate-1.0 slot at maxCats."""
+    from tests.helpers import _model_arrays
+    rng = np.random.default_rng(17)
+    aa = np.load(os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "examl_amd", "data", "aa_models.npz"))
+    m = ea.ProtGtrModel(aa["frequencies"][4], aa["rates190"][4], 1.0)
+    EIGN, EV, EI, tipVector, _ = _model_arrays(m)
+    n, nc, maxc = 160, 5, 7
+    rates = O.aligned(nc)
+    rates[:] = [0.2, 0.6, 1.0, 1.7, 3.0]
+    cptr = rng.integers(0, nc, n).astype(np.int32)
+    t1 = rng.integers(1, 23, n).astype(np.uint8)
+    t1[rng.random(n) < 0.3] = 22
+    t2 = rng.integers(1, 23, n).astype(np.uint8)
+    t2[rng.random(n) < 0.3] = 22
+    wgt = np.ones(n, dtype=np.int32)
+    gvl = n // 32 + 1
+
+    def gap_of(tips):
+        gv = np.zeros(gvl, dtype=np.uint32)
+        idx = np.nonzero(tips == 22)[0]
+        np.bitwise_or.at(gv, idx // 32,
+                         (np.uint32(1) << (idx % 32).astype(np.uint32)))
+        return gv
+
+    g1, g2 = gap_of(t1), gap_of(t2)
+
+    def dp(a):
+        return (a.ctypes.data_as(ctypes.POINTER(ctypes.c_double))
+                if a is not None else
+                ctypes.cast(None, ctypes.POINTER(ctypes.c_double)))
+
+    def u8(a):
+        return (a.ctypes.data_as(ctypes.POINTER(ctypes.c_ubyte))
+                if a is not None else
+                ctypes.cast(None, ctypes.POINTER(ctypes.c_ubyte)))
+
+    def ip(a):
+        return a.ctypes.data_as(ctypes.POINTER(ctypes.c_int))
+
+    def up(a):
+        return (a.ctypes.data_as(ctypes.POINTER(ctypes.c_uint))
+                if a is not None else
+                ctypes.cast(None, ctypes.POINTER(ctypes.c_uint)))
+
+    ref = O._ref
+    left_o = O.aligned((maxc + 1) * 400)
+    right_o = O.aligned((maxc + 1) * 400)
+    O._orc.oracle_make_p_save(
+        ctypes.c_double(-0.2), ctypes.c_double(-0.45), dp(rates), dp(EI),
+        dp(EIGN), ctypes.c_int(nc), dp(left_o), dp(right_o),
+        ctypes.c_int(maxc), ctypes.c_int(20))
+    left_r = O.aligned((maxc + 1) * 400)
+    right_r = O.aligned((maxc + 1) * 400)
+    ref.makeP(ctypes.c_double(-0.2), ctypes.c_double(-0.45), dp(rates),
+              dp(EI), dp(EIGN), ctypes.c_int(nc), dp(left_r), dp(right_r),
+              ctypes.c_int(1), ctypes.c_int(maxc), ctypes.c_int(20))
+    assert np.array_equal(left_o, left_r)
+    assert np.array_equal(right_o, right_r)
+
+    tvg = O.aligned(20)
+    tvg[:] = tipVector[22 * 20:23 * 20]
+
+    def run_pair(tc, x1r, x1o, x2r, x2o, ga, gb, gca_r, gca_o, gcb_r, gcb_o,
+                 ta, tb):
+        g3 = ga & gb
+        nz = int(n - sum(bin(int(w)).count("1") for w in g3))
+        x3r = O.aligned(nz * 20 + 20)
+        x3o = O.aligned(nz * 20 + 20)
+        gr = O.aligned(20)
+        go = O.aligned(20)
+        ir = ctypes.c_int(0)
+        io = ctypes.c_int(0)
+        ref.newviewGTRCATPROT_AVX_GAPPED_SAVE(
+            tc, dp(EV), ip(cptr), dp(x1r), dp(x2r), dp(x3r), dp(tipVector),
+            None, u8(ta), u8(tb), ctypes.c_int(n), dp(left_r), dp(right_r),
+            ip(wgt), ctypes.byref(ir), ctypes.c_int(1), up(ga), up(gb),
+            up(g3), dp(gca_r), dp(gcb_r), dp(gr), ctypes.c_int(maxc))
+        O._orc.oracle_newview_prot_cat_save(
+            tc, dp(EV), ip(cptr), dp(x1o), dp(x2o), dp(x3o), dp(tipVector),
+            u8(ta), u8(tb), ctypes.c_int(n), dp(left_o), dp(right_o),
+            ip(wgt), ctypes.byref(io), up(ga), up(gb), up(g3), dp(gca_o),
+            dp(gcb_o), dp(go), ctypes.c_int(maxc))
+        assert np.array_equal(x3r[:nz * 20], x3o[:nz * 20])
+        assert np.array_equal(gr, go)
+        assert ir.value == io.value
+        return g3, x3r, x3o, gr, go
+
+    g3, ar, ao, gr, go = run_pair(0, None, None, None, None, g1, g2, tvg,
+                                  tvg, tvg, tvg, t1, t2)
+    g3b, br, bo, gr2, go2 = run_pair(1, None, None, ar, ao, g1, g3, tvg,
+                                     tvg, gr, go, t1, None)
+    run_pair(2, ar, ao, br, bo, g3, g3b, gr, go, gr2, go2, None, None)
+
+    diag = O.calc_diagptable(0.6, 20, nc, rates, EIGN)
+    ref.evaluateGTRCATPROT_SAVE.restype = ctypes.c_double
+    O._orc.oracle_evaluate_prot_cat_save.restype = ctypes.c_double
+    lr = ref.evaluateGTRCATPROT_SAVE(
+        ip(cptr), ip(wgt), dp(ar), dp(br), dp(tipVector), None,
+        ctypes.c_int(n), dp(diag), dp(gr), dp(gr2), up(g3), up(g3b))
+    lo = O._orc.oracle_evaluate_prot_cat_save(
+        ip(cptr), ip(wgt), dp(ao), dp(bo), dp(tipVector), u8(None),
+        ctypes.c_int(n), dp(diag), dp(go), dp(go2), up(g3), up(g3b))
+    assert lr == lo
+    lr2 = ref.evaluateGTRCATPROT_SAVE(
+        ip(cptr), ip(wgt), dp(None), dp(ar), dp(tipVector), u8(t1),
+        ctypes.c_int(n), dp(diag), dp(None), dp(gr), up(g1), up(g3))
+    lo2 = O._orc.oracle_evaluate_prot_cat_save(
+        ip(cptr), ip(wgt), dp(None), dp(ao), dp(tipVector), u8(t1),
+        ctypes.c_int(n), dp(diag), dp(None), dp(go), up(g1), up(g3))
+    assert lr2 == lo2
+
+    sr = O.aligned(n * 20)
+    so = O.aligned(n * 20)
+    ref.sumGTRCATPROT_SAVE(2, dp(sr), dp(ar), dp(br), dp(tipVector),
+                           u8(None), u8(None), ctypes.c_int(n), dp(gr),
+                           dp(gr2), up(g3), up(g3b))
+    O._orc.oracle_sum_prot_cat_save(2, dp(so), dp(ao), dp(bo), dp(tipVector),
+                                    u8(None), u8(None), ctypes.c_int(n),
+                                    dp(go), dp(go2), up(g3), up(g3b))
+    assert np.array_equal(sr, so)
+    sr2 = O.aligned(n * 20)
+    so2 = O.aligned(n * 20)
+    ref.sumGTRCATPROT_SAVE(1, dp(sr2), dp(None), dp(ar), dp(tipVector),
+                           u8(t1), u8(None), ctypes.c_int(n), dp(None),
+                           dp(gr), up(g1), up(g3))
+    O._orc.oracle_sum_prot_cat_save(1, dp(so2), dp(None), dp(ao),
+                                    dp(tipVector), u8(t1), u8(None),
+                                    ctypes.c_int(n), dp(None), dp(go),
+                                    up(g1), up(g3))
+    assert np.array_equal(sr2, so2)
