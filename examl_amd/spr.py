@@ -373,11 +373,72 @@ class BestList:
         return None
 
 
+class RfConvergence:
+    """The -D search convergence criterion: a two-slot bipartition store
+    (tr->h) restating bipartitionList.c — insertHashRF:385,
+    cleanupHashTable:181, convergenceCriterion:541.  Keys are the tip
+    sets on the far side of each internal edge as seen from tip 1
+    (bitVectorInitravSpecial:472 roots the traversal at nodep[1]->back,
+    so an inserted vector never contains tip 1); values are the 2-bit
+    slot masks of e->treeVector[0]."""
+
+    def __init__(self, st):
+        self.st = st
+        self.table = {}  # frozenset(tip numbers) -> slot mask (bits 1|2)
+
+    def _bipartitions(self):
+        """Subtree tip sets of every inner node except the one adjacent
+        to tip 1 — the mxtips-3 internal edges (bCounter assert)."""
+        st = self.st
+        ntips = st.ntips
+        out = []
+
+        def down(m):
+            # m: the parent-facing member of the node entered
+            if m.number <= ntips:
+                return frozenset((m.number,))
+            s = down(m.next.back) | down(m.next.next.back)
+            out.append(s)  # the edge (m, m.back); m.back is never a tip
+            return s
+
+        root = st.ring[1].back  # nodep[1]->back: its edge to tip 1 is
+        down(root.next.back)    # not an internal bipartition
+        s2 = down(root.next.next.back)
+        # merge: out already holds both child subtrees' contributions
+        del s2
+        assert len(out) == ntips - 3, len(out)
+        return out
+
+    def store(self, iteration):
+        """Record the current tree in slot iteration%2, first clearing
+        that slot's stale bits (cleanupHashTable) when iteration > 1."""
+        slot = iteration % 2
+        if iteration > 1:
+            self.cleanup(slot)
+        bit = 1 << slot
+        for b in self._bipartitions():
+            self.table[b] = self.table.get(b, 0) | bit
+
+    def cleanup(self, slot):
+        keep = 2 >> slot  # slot 0 keeps bit 2, slot 1 keeps bit 1
+        self.table = {k: v & keep for k, v in self.table.items()
+                      if v & keep}
+
+    def rrf(self):
+        """convergenceCriterion: bipartitions in exactly one slot over
+        2*(mxtips-3)."""
+        rf = sum(1 for v in self.table.values() if v == 1 or v == 2)
+        return rf / (2.0 * (self.st.ntips - 3))
+
+    def clear(self):
+        self.table = {}
+
+
 class SprSearch:
     """computeBIGRAPID over a TreeSearch on an SprTree."""
 
     def __init__(self, ts, do_cutoff=True, big_cutoff=False, stepwidth=5,
-                 max_rearrange=21, log=None):
+                 max_rearrange=21, log=None, convergence_criterion=False):
         self.ts = ts
         self.st = ts.tree
         assert isinstance(self.st, SprTree)
@@ -388,6 +449,9 @@ class SprSearch:
         self.stepwidth = stepwidth
         self.max_rearrange = max_rearrange
         self.log = log or (lambda *_: None)
+        # -D: RF-distance stopping criterion (tr->searchConvergenceCriterion)
+        self.convergence_criterion = convergence_criterion
+        self.rfconv = RfConvergence(self.st)
         # tr-> search state
         self.start_lh = 0.0
         self.end_lh = 0.0
@@ -857,6 +921,20 @@ class SprSearch:
         fast_iterations = 0
         while impr:
             best_t.recall(1, ts)
+            # -D check at the top of each fast cycle (searchAlgo.c:2160):
+            # store the current best tree in slot fastIterations%2, then
+            # compare against the previous cycle's tree.
+            if self.convergence_criterion:
+                self.rfconv.store(fast_iterations)
+                if fast_iterations > 0:
+                    rrf = self.rfconv.rrf()
+                    if rrf <= 0.01:  # 1% cutoff
+                        self.log(f"converged fast cycle {fast_iterations}"
+                                 f": {rrf:.6f}")
+                        break
+                    self.log("convergence fast cycle "
+                             f"{fast_iterations - 1}->{fast_iterations}"
+                             f": {rrf:.6f}")
             fast_iterations += 1
             ts.tree_evaluate(1.0)
             best_t.save(ts, True)
@@ -872,6 +950,9 @@ class SprSearch:
                     impr = True
                     lh = ts.likelihood
                     best_t.save(ts, True)
+        if self.convergence_criterion:
+            # both exits empty the table (searchAlgo.c:2202/2303)
+            self.rfconv.clear()
         self.thorough = True
         impr = True
         best_t.recall(1, ts)
@@ -888,6 +969,18 @@ class SprSearch:
             if impr:
                 rearrangements_min = 1
                 rearrangements_max = self.stepwidth
+                # -D check (searchAlgo.c:2438): slot thoroughIterations%2
+                if self.convergence_criterion:
+                    self.rfconv.store(thorough_iterations)
+                    if thorough_iterations > 0:
+                        rrf = self.rfconv.rrf()
+                        if rrf <= 0.01:  # goto cleanup
+                            self.log("converged thorough cycle "
+                                     f"{thorough_iterations}: {rrf:.6f}")
+                            break
+                        self.log("convergence thorough cycle "
+                                 f"{thorough_iterations - 1}->"
+                                 f"{thorough_iterations}: {rrf:.6f}")
                 thorough_iterations += 1
             else:
                 rearrangements_max += self.stepwidth
@@ -896,6 +989,9 @@ class SprSearch:
                     break
             ts.tree_evaluate(1.0)
             previous_lh = lh = ts.likelihood
+            # saveBestTree refreshes the stored copy's branch lengths
+            # (searchAlgo.c:2519) — the cleanup path re-evaluates THESE
+            best_t.save(ts, True)
             self.log(f"thorough SPR cycle {thorough_iterations} "
                      f"[{rearrangements_min},{rearrangements_max}]: "
                      f"{lh:.6f}")

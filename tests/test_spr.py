@@ -101,3 +101,117 @@ def test_full_spr_search_gpu(golden_dir):
     sp = SprSearch(ts)
     lnl = sp.compute_big_rapid(estimate_model=True)
     assert abs(lnl - GOLDEN_FINAL) < TOL, lnl
+
+
+def test_rf_convergence_semantics(golden_dir):
+    """RfConvergence restates the -D bipartition bookkeeping
+    (bipartitionList.c insertHashRF/cleanupHashTable/convergenceCriterion):
+    mxtips-3 internal bipartitions per tree, two-slot masks, symmetric
+    difference over 2*(mxtips-3)."""
+    from examl_amd.examl_io import read_byte_file, read_newick_topology
+    from examl_amd.spr import RfConvergence
+    taxa, _ = read_byte_file(os.path.join(golden_dir, "49.binary"))
+    t = read_newick_topology(os.path.join(golden_dir, "49.tree"), taxa)
+    st = SprTree.from_phylo(t)
+    rc = RfConvergence(st)
+    assert len(rc._bipartitions()) == st.ntips - 3
+    rc.store(0)
+    rc.store(1)
+    assert rc.rrf() == 0.0  # identical trees
+    rc.clear()
+    rc.store(0)
+    assert len(rc.table) == st.ntips - 3
+    rc.store(1)
+    rc.cleanup(0)  # drop slot-0 bits, keep slot-1 entries
+    assert all(v == 2 for v in rc.table.values())
+    # store(2) targets slot 0 again and cleans it first
+    rc.clear()
+    rc.store(0)
+    rc.store(1)
+    rc.store(2)
+    assert rc.rrf() == 0.0
+
+
+def test_spr_search_convergence_12_cpu(golden_dir):
+    """Full -f d -D search on a 12-taxon synthetic dataset (generated
+    with the reference parser; goldens from examl-AVX -D): the RF
+    trajectory (10/18 = 0.555556 between fast cycles 0 and 1), the final
+    lnL -2741.473102 and the result topology all match the reference.
+
+    This dataset also pins the thorough-loop saveBestTree refresh
+    (searchAlgo.c:2519): without it the cleanup path re-evaluates stale
+    branch lengths and lands at -2741.515138."""
+    from examl_amd.examl_io import (parse_newick_topology, read_byte_file,
+                                    read_newick_topology)
+    from tests.helpers import OracleEngine
+    taxa, parts = read_byte_file(os.path.join(golden_dir, "12.binary"))
+    tree = read_newick_topology(os.path.join(golden_dir, "12.tree"), taxa)
+    st = SprTree.from_phylo(tree)
+    engines = [OracleEngine(p.tips, p.wgt,
+                            ea.DnaGtrModel(p.frequencies, [1.0] * 6, 1.0))
+               for p in parts]
+    ts = TreeSearch(st, engines,
+                    opt_freq_flags=[bool(p.optimizeBaseFrequencies)
+                                    for p in parts])
+    sp = SprSearch(ts, convergence_criterion=True)
+    logs = []
+    sp.log = logs.append
+    lnl = sp.compute_big_rapid(estimate_model=True)
+    assert abs(lnl - (-2741.473102)) < abs(2741.473102) * 1e-6, lnl
+    rrfs = [s for s in logs if s.startswith("convergence")]
+    assert rrfs == ["convergence fast cycle 0->1: 0.555556"], rrfs
+    # final topology == the reference's ExaML_result file
+    with open(os.path.join(golden_dir, "12.result.tree")) as f:
+        ref = parse_newick_topology(f.read(), taxa, read_bl=True)
+    from examl_amd.spr import RfConvergence
+    ours = {frozenset(b) for b in RfConvergence(st)._bipartitions()}
+    theirs = {frozenset(b)
+              for b in RfConvergence(SprTree.from_phylo(ref))
+              ._bipartitions()}
+    assert ours == theirs
+
+
+@pytest.mark.skipif(not os.environ.get("EXAML_E2E_SPR"),
+                    reason="full -D SPR search on CPU oracle (~8 min): "
+                           "set EXAML_E2E_SPR=1")
+def test_full_spr_search_convergence_cpu(golden_dir):
+    """49-taxon -f d -D: the reference's logged RF trajectory
+    (0.086957 -> 0.043478 -> 0.021739 fast, 0.021739 thorough) and the
+    final lnL, replayed exactly."""
+    from tests.helpers import OracleEngine
+    ts, st = _setup(golden_dir, OracleEngine)
+    sp = SprSearch(ts, convergence_criterion=True)
+    logs = []
+    sp.log = logs.append
+    lnl = sp.compute_big_rapid(estimate_model=True)
+    assert abs(lnl - GOLDEN_FINAL) < TOL, lnl
+    rrfs = [s for s in logs if s.startswith("convergence")]
+    assert rrfs == ["convergence fast cycle 0->1: 0.086957",
+                    "convergence fast cycle 1->2: 0.043478",
+                    "convergence fast cycle 2->3: 0.021739",
+                    "convergence thorough cycle 0->1: 0.021739"], rrfs
+
+
+@pytest.mark.gpu
+def test_spr_search_convergence_12_gpu(golden_dir):
+    """The 12-taxon -D search end-to-end on the MI355X engines."""
+    import torch
+    from examl_amd.examl_io import read_byte_file, read_newick_topology
+    assert torch.cuda.is_available()
+    taxa, parts = read_byte_file(os.path.join(golden_dir, "12.binary"))
+    tree = read_newick_topology(os.path.join(golden_dir, "12.tree"), taxa)
+    st = SprTree.from_phylo(tree)
+    engines = [ea.DnaGammaEngine(p.tips, p.wgt,
+                                 ea.DnaGtrModel(p.frequencies, [1.0] * 6,
+                                                1.0), device="cuda:0")
+               for p in parts]
+    ts = TreeSearch(st, engines,
+                    opt_freq_flags=[bool(p.optimizeBaseFrequencies)
+                                    for p in parts])
+    sp = SprSearch(ts, convergence_criterion=True)
+    logs = []
+    sp.log = logs.append
+    lnl = sp.compute_big_rapid(estimate_model=True)
+    assert abs(lnl - (-2741.473102)) < abs(2741.473102) * 1e-6, lnl
+    assert any(s == "convergence fast cycle 0->1: 0.555556"
+               for s in logs), logs
