@@ -438,7 +438,8 @@ class SprSearch:
     """computeBIGRAPID over a TreeSearch on an SprTree."""
 
     def __init__(self, ts, do_cutoff=True, big_cutoff=False, stepwidth=5,
-                 max_rearrange=21, log=None, convergence_criterion=False):
+                 max_rearrange=21, log=None, convergence_criterion=False,
+                 save_best_trees=0):
         self.ts = ts
         self.st = ts.tree
         assert isinstance(self.st, SprTree)
@@ -452,6 +453,12 @@ class SprSearch:
         # -D: RF-distance stopping criterion (tr->searchConvergenceCriterion)
         self.convergence_criterion = convergence_criterion
         self.rfconv = RfConvergence(self.st)
+        # -B: keep the N best distinct ML trees seen during the search
+        # (tr->saveBestTrees / bestML, searchAlgo.c:1944)
+        self.save_best_trees = save_best_trees
+        self.best_ml = (BestList(save_best_trees, self.st)
+                        if save_best_trees > 0 else None)
+        self.good_trees = []  # good-tree lnls after compute_big_rapid
         # tr-> search state
         self.start_lh = 0.0
         self.end_lh = 0.0
@@ -727,9 +734,10 @@ class SprSearch:
         self.remove_node_restore_big(self.remove_node)
         self.test_insert_restore_big(self.remove_node, self.insert_node)
 
-    def restore_topology_only(self, bt):
+    def restore_topology_only(self, bt, best_ml=None):
         """restoreTopologyOnly (:610): record the best insertion for this
-        node into bt without computing anything."""
+        node into bt (and, under -B, into bestML with keep_identical
+        FALSE, :664) without computing anything."""
         p = self.remove_node
         q = self.insert_node
         current_lh = self.ts.likelihood
@@ -752,6 +760,8 @@ class SprSearch:
             hookup(p.next.next, r, z)
         self.ts.likelihood = self.best_of_node
         bt.save(self.ts, True)
+        if best_ml is not None:
+            best_ml.save(self.ts, False)
         self.ts.likelihood = current_lh
         hookup(q, r, qz)
         p.next.back = p.next.next.back = None
@@ -795,8 +805,9 @@ class SprSearch:
         reorder(st.nodep[1].back)
         assert count[0] == st.ntips - 2, (count[0], old and None)
 
-    def tree_optimize_rapid(self, mintrav, maxtrav, bt):
-        """treeOptimizeRapid (:914)."""
+    def tree_optimize_rapid(self, mintrav, maxtrav, bt, best_ml=None):
+        """treeOptimizeRapid (:914); best_ml is tr->saveBestTrees' bestML
+        list (-B), fed at :979/:1015 and through restoreTopologyOnly."""
         ts, st = self.ts, self.st
         self.node_rectifier()
         maxtrav = min(maxtrav, st.ntips - 3)
@@ -825,8 +836,10 @@ class SprSearch:
                         self.restore_tree_fast()
                         self.start_lh = self.end_lh = ts.likelihood
                         bt.save(ts, True)
+                        if best_ml is not None:
+                            best_ml.save(ts, False)
                     elif self.best_of_node != UNLIKELY:
-                        self.restore_topology_only(bt)
+                        self.restore_topology_only(bt, best_ml)
                 else:
                     self.insert_info_list(st.nodep[i], self.best_of_node)
                     if self.end_lh > self.start_lh:
@@ -841,8 +854,10 @@ class SprSearch:
                         self.restore_tree_fast()
                         self.start_lh = self.end_lh = ts.likelihood
                         bt.save(ts, True)
+                        if best_ml is not None:
+                            best_ml.save(ts, False)
                     elif self.best_of_node != UNLIKELY:
-                        self.restore_topology_only(bt)
+                        self.restore_topology_only(bt, best_ml)
             self.thorough = False
         return self.start_lh
 
@@ -940,7 +955,7 @@ class SprSearch:
             best_t.save(ts, True)
             lh = previous_lh = ts.likelihood
             self.log(f"fast SPR cycle {fast_iterations}: {lh:.6f}")
-            self.tree_optimize_rapid(1, best_trav, bt)
+            self.tree_optimize_rapid(1, best_trav, bt, self.best_ml)
             impr = False
             for i in range(1, bt.nvalid + 1):
                 bt.recall(i, ts)
@@ -996,7 +1011,7 @@ class SprSearch:
                      f"[{rearrangements_min},{rearrangements_max}]: "
                      f"{lh:.6f}")
             self.tree_optimize_rapid(rearrangements_min, rearrangements_max,
-                                     bt)
+                                     bt, self.best_ml)
             impr = False
             for i in range(1, bt.nvalid + 1):
                 bt.recall(i, ts)
@@ -1008,4 +1023,13 @@ class SprSearch:
                     best_t.save(ts, True)
         ts.evaluate_generic(full=True)
         self.log(f"likelihood of best tree: {ts.likelihood:.6f}")
-        return ts.likelihood
+        final = ts.likelihood
+        # -B epilogue (searchAlgo.c:2577): re-load each good tree and
+        # report it; leaves the tree at the last entry, like the
+        # reference, after the result tree has been produced.
+        self.good_trees = []
+        if self.best_ml is not None:
+            for i in range(1, self.best_ml.nvalid + 1):
+                self.best_ml.recall(i, ts)
+                self.good_trees.append(ts.likelihood)
+        return final

@@ -215,3 +215,45 @@ def test_spr_search_convergence_12_gpu(golden_dir):
     assert abs(lnl - (-2741.473102)) < abs(2741.473102) * 1e-6, lnl
     assert any(s == "convergence fast cycle 0->1: 0.555556"
                for s in logs), logs
+
+
+def test_spr_save_best_trees_12_cpu(golden_dir):
+    """-B 5 (tr->saveBestTrees / bestML, searchAlgo.c:979/1015/664 and
+    the :2577 epilogue) on the 12-taxon golden: the five good-tree lnLs
+    and topologies match the reference's RAxML_5_goodTrees file
+    digit-for-digit."""
+    from examl_amd.examl_io import (parse_newick_topology, read_byte_file,
+                                    read_newick_topology)
+    from examl_amd.spr import RfConvergence
+    from tests.helpers import OracleEngine
+    taxa, parts = read_byte_file(os.path.join(golden_dir, "12.binary"))
+    tree = read_newick_topology(os.path.join(golden_dir, "12.tree"), taxa)
+    st = SprTree.from_phylo(tree)
+    engines = [OracleEngine(p.tips, p.wgt,
+                            ea.DnaGtrModel(p.frequencies, [1.0] * 6, 1.0))
+               for p in parts]
+    ts = TreeSearch(st, engines,
+                    opt_freq_flags=[bool(p.optimizeBaseFrequencies)
+                                    for p in parts])
+    sp = SprSearch(ts, save_best_trees=5)
+    lnl = sp.compute_big_rapid(estimate_model=True)
+    assert abs(lnl - (-2741.473102)) < abs(2741.473102) * 1e-6, lnl
+    golden = [-2741.6194320738341, -2741.6249810031436,
+              -2741.7057391519088, -2742.0444100799273,
+              -2742.1427442910435]
+    assert len(sp.good_trees) == 5
+    for g, o in zip(golden, sp.good_trees):
+        assert abs(o - g) < 1e-8, (o, g)
+    # topologies: re-load each stored tree and compare bipartitions with
+    # the corresponding line of the reference's goodTrees file
+    with open(os.path.join(golden_dir, "12.goodtrees.txt")) as f:
+        lines = [ln.strip() for ln in f if ln.strip()]
+    assert len(lines) == 5
+    for i, line in enumerate(lines):
+        sp.best_ml.recall(i + 1, ts)
+        ours = {frozenset(b) for b in RfConvergence(st)._bipartitions()}
+        ref = parse_newick_topology(line, taxa, read_bl=True)
+        theirs = {frozenset(b)
+                  for b in RfConvergence(SprTree.from_phylo(ref))
+                  ._bipartitions()}
+        assert ours == theirs, i
