@@ -453,6 +453,9 @@ class SprSearch:
         # -D: RF-distance stopping criterion (tr->searchConvergenceCriterion)
         self.convergence_criterion = convergence_criterion
         self.rfconv = RfConvergence(self.st)
+        # -g: constraintVector by node number (treeReadLenMULT labels);
+        # None when no constraint tree is in use
+        self.constraint = None
         # -B: keep the N best distinct ML trees seen during the search
         # (tr->saveBestTrees / bestML, searchAlgo.c:1944)
         self.save_best_trees = save_best_trees
@@ -621,11 +624,45 @@ class SprSearch:
         self._newview(p)
         return True
 
+    def _constraint_checker(self, p):
+        """checker (searchAlgo.c:69): first non--9 group label found in
+        the subtree behind member p."""
+        cv = self.constraint
+        group = cv[p.number]
+        if p.number <= self.st.ntips:
+            return group
+        if group != -9:
+            return group
+        group = self._constraint_checker(p.next.back)
+        if group != -9:
+            return group
+        return self._constraint_checker(p.next.next.back)
+
     def test_insert_big(self, p, q):
-        """testInsertBIG (:683), no constraint tree."""
+        """testInsertBIG (:683); with a -g constraint tree the insertion
+        is gated by the group labels of p, q and r (:697-722)."""
         r = q.back
         qz, pz = q.z, p.z
         start_lh = self.end_lh
+        if self.constraint is not None:
+            cv = self.constraint
+            do_it = False
+            r_num = cv[r.number]
+            q_num = cv[q.number]
+            p_num = cv[p.number]
+            if p_num == -9:
+                p_num = self._constraint_checker(p.back)
+            if p_num == -9:
+                do_it = True
+            else:
+                if q_num == -9:
+                    q_num = self._constraint_checker(q)
+                if r_num == -9:
+                    r_num = self._constraint_checker(r)
+                if p_num == r_num or p_num == q_num:
+                    do_it = True
+            if not do_it:
+                return True
         self.insert_big(p, q)
         lnl = self._evaluate_branch(p.next.next)
         if lnl > self.best_of_node:
@@ -1033,3 +1070,163 @@ class SprSearch:
                 self.best_ml.recall(i, ts)
                 self.good_trees.append(ts.likelihood)
         return final
+
+
+class _NewickStream:
+    """Character-level reader with the reference's treeGetCh semantics
+    (treeIO.c:60: skip whitespace, return next char)."""
+
+    def __init__(self, text):
+        self.text = text
+        self.pos = 0
+
+    def getch(self):
+        while self.pos < len(self.text):
+            c = self.text[self.pos]
+            self.pos += 1
+            if not c.isspace():
+                return c
+        return ""
+
+    def ungetc(self):
+        self.pos -= 1
+
+    def read_name(self):
+        out = []
+        while self.pos < len(self.text):
+            c = self.text[self.pos]
+            if c in "(),:;[]":
+                break
+            out.append(c)
+            self.pos += 1
+        return "".join(out).strip()
+
+    def flush_len(self):
+        """treeFlushLen: consume an optional :branchLength."""
+        c = self.getch()
+        if c == ":":
+            while self.pos < len(self.text) and \
+                    self.text[self.pos] in "0123456789.eE+-":
+                self.pos += 1
+        elif c:
+            self.ungetc()
+
+    def flush_label(self):
+        self.read_name()
+
+
+def read_constraint_tree(text, taxa, seed):
+    """treeReadLenMULT (treeIO.c:1033) + getStartingTree (:1162) for -g:
+    parse a multifurcating constraint tree over ALL taxa, resolving each
+    multifurcation randomly with the reference's srand/rand stream
+    (randomInt(10000), :916), and label every node with its constraint
+    group in constraintVector.  Returns (SprTree, constraint dict).
+
+    Only unrooted constraints (>= 3 top-level children) are supported,
+    like the reference's non-rooted path."""
+    import ctypes
+    libc = ctypes.CDLL(None)
+    libc.srand(ctypes.c_uint(seed))
+
+    def random_int(n):
+        return libc.rand() % n
+
+    ntips = len(taxa)
+    tip_no = {name: i + 1 for i, name in enumerate(taxa)}
+    st = SprTree(ntips)
+    cv = {i: -1 for i in range(2 * ntips)}
+    state = {"nextnode": ntips + 1, "ntips": 0, "partCount": 0}
+    s = _NewickStream(text)
+
+    def next_inner():
+        n = state["nextnode"]
+        state["nextnode"] += 1
+        assert n <= 2 * ntips - 2, "rooted constraint tree not supported"
+        return st.ring[n]
+
+    def resolution():
+        rn = random_int(10000)
+        return 0.0 if rn == 0 else rn / 10000.0
+
+    def add_element(p, pc):
+        # addElementLenMULT (treeIO.c:921)
+        cv[p.number] = pc
+        ch = s.getch()
+        if ch == "(":
+            state["partCount"] += 1
+            old = state["partCount"]
+            q = next_inner()
+            cv[q.number] = state["partCount"]
+            add_element(q.next, old)
+            assert s.getch() == ","
+            add_element(q.next.next, old)
+            hookup(p, q, DEFAULTZ)
+            while True:
+                ch = s.getch()
+                if ch != ",":
+                    break
+                r = next_inner()
+                cv[r.number] = state["partCount"]  # CURRENT count (:974)
+                if resolution() < 0.5:
+                    t = q.next.back
+                    r.back = q.next
+                    q.next.back = r
+                    r.next.back = t
+                    t.back = r.next
+                    add_element(r.next.next, old)
+                else:
+                    t = q.next.next.back
+                    r.back = q.next.next
+                    q.next.next.back = r
+                    r.next.back = t
+                    t.back = r.next
+                    add_element(r.next.next, old)
+            assert ch == ")", "missing ) in constraint tree"
+            s.flush_label()
+        else:
+            s.ungetc()
+            n = tip_no[s.read_name()]
+            q = st.ring[n]
+            cv[q.number] = pc
+            state["ntips"] += 1
+            hookup(p, q, DEFAULTZ)
+        s.flush_len()
+
+    p = next_inner()
+    while s.getch() != "(":
+        pass
+    add_element(p, 0)
+    assert s.getch() == ","
+    add_element(p.next, 0)
+    ch = s.getch()
+    assert ch == ",", "rooted (bifurcating-top) constraint unsupported"
+    add_element(p.next.next, 0)
+    while True:
+        ch = s.getch()
+        if ch != ",":
+            break
+        r = next_inner()
+        cv[r.number] = 0
+        if resolution() < 0.5:
+            t = p.next.next.back
+            r.back = p.next.next
+            p.next.next.back = r
+            r.next.back = t
+            t.back = r.next
+            add_element(r.next.next, 0)
+        else:
+            t = p.next.back
+            r.back = p.next
+            p.next.back = r
+            r.next.back = t
+            t.back = r.next
+            add_element(r.next.next, 0)
+    assert ch == ")"
+    s.ungetc()
+    assert s.getch() == ")"
+    s.flush_label()
+    s.flush_len()
+    assert s.getch() == ";"
+    assert state["ntips"] == ntips, "constraint must contain all taxa"
+    st.start = 1  # getStartingTree: tr->start = tr->nodep[1]
+    return st, cv

@@ -257,3 +257,41 @@ def test_spr_save_best_trees_12_cpu(golden_dir):
                   for b in RfConvergence(SprTree.from_phylo(ref))
                   ._bipartitions()}
         assert ours == theirs, i
+
+
+def test_spr_constraint_tree_12_cpu(golden_dir):
+    """-g constraint tree + -p seed: read_constraint_tree restates
+    treeReadLenMULT (treeIO.c:1033) including the reference's
+    srand/rand multifurcation resolution, and testInsertBIG's group
+    gate (searchAlgo.c:697) restricts the SPR moves.  The whole
+    constrained search on the 12-taxon golden reproduces the
+    reference's final lnL and result topology."""
+    from examl_amd.examl_io import (parse_newick_topology, read_byte_file)
+    from examl_amd.spr import RfConvergence, read_constraint_tree
+    from tests.helpers import OracleEngine
+    taxa, parts = read_byte_file(os.path.join(golden_dir, "12.binary"))
+    with open(os.path.join(golden_dir, "12.constraint.tree")) as f:
+        st, cv = read_constraint_tree(f.read(), taxa, 12345)
+    engines = [OracleEngine(p.tips, p.wgt,
+                            ea.DnaGtrModel(p.frequencies, [1.0] * 6, 1.0))
+               for p in parts]
+    ts = TreeSearch(st, engines,
+                    opt_freq_flags=[bool(p.optimizeBaseFrequencies)
+                                    for p in parts])
+    sp = SprSearch(ts)
+    sp.constraint = cv
+    lnl = sp.compute_big_rapid(estimate_model=True)
+    assert abs(lnl - (-3435.016697)) < abs(3435.016697) * 1e-6, lnl
+    # final tree == the reference's, and it honors the constraint:
+    # {T01..T04} and {T05..T07} stay monophyletic
+    ours = {frozenset(b) for b in RfConvergence(st)._bipartitions()}
+    with open(os.path.join(golden_dir,
+                           "12.constrained.result.tree")) as f:
+        ref = parse_newick_topology(f.read(), taxa, read_bl=True)
+    theirs = {frozenset(b)
+              for b in RfConvergence(SprTree.from_phylo(ref))
+              ._bipartitions()}
+    assert ours == theirs
+    assert frozenset({5, 6, 7}) in ours
+    # {T01..T04} monophyletic: the far-from-tip-1 side of that edge
+    assert frozenset(range(5, 13)) in ours
