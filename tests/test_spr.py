@@ -295,3 +295,47 @@ def test_spr_constraint_tree_12_cpu(golden_dir):
     assert frozenset({5, 6, 7}) in ours
     # {T01..T04} monophyletic: the far-from-tip-1 side of that edge
     assert frozenset(range(5, 13)) in ours
+
+
+def _setup_12(golden_dir, cat=False, **spkw):
+    import numpy as np
+    from examl_amd.examl_io import read_byte_file, read_newick_topology
+    from tests.helpers import OracleCatEngine, OracleEngine
+    taxa, parts = read_byte_file(os.path.join(golden_dir, "12.binary"))
+    tree = read_newick_topology(os.path.join(golden_dir, "12.tree"), taxa)
+    st = SprTree.from_phylo(tree)
+    engines = []
+    for p in parts:
+        m = ea.DnaGtrModel(p.frequencies, [1.0] * 6, 1.0)
+        if cat:
+            w = p.upper - p.lower
+            engines.append(OracleCatEngine(p.tips, p.wgt, m,
+                                           np.zeros(w, dtype=np.int32),
+                                           np.array([1.0])))
+        else:
+            engines.append(OracleEngine(p.tips, p.wgt, m))
+    kw = dict(opt_freq_flags=[bool(p.optimizeBaseFrequencies)
+                              for p in parts])
+    if cat:
+        kw["rate_het"] = "CAT"
+    ts = TreeSearch(st, engines, **kw)
+    return SprSearch(ts, **spkw), ts
+
+
+def test_spr_f_o_no_cutoff_12_cpu(golden_dir):
+    """-f o (BIG_RAPID without the lhCutoff heuristic, axml.c:1143):
+    reference golden -2741.473155 on the 12-taxon dataset (note: a
+    slightly different final than -f d, so this genuinely exercises the
+    doCutoff=FALSE path)."""
+    sp, ts = _setup_12(golden_dir, do_cutoff=False)
+    lnl = sp.compute_big_rapid(estimate_model=True)
+    assert abs(lnl - (-2741.473155)) < abs(2741.473155) * 1e-6, lnl
+
+
+def test_spr_psr_search_12_cpu(golden_dir):
+    """The default mode (-f d) under -m PSR: the SPR hill climber with
+    optimizeRateCategories running inside every modOpt.  Reference
+    golden -2507.657682."""
+    sp, ts = _setup_12(golden_dir, cat=True)
+    lnl = sp.compute_big_rapid(estimate_model=True)
+    assert abs(lnl - (-2507.657682)) < abs(2507.657682) * 1e-6, lnl
