@@ -79,3 +79,32 @@ def test_random_quartets_gpu(golden_dir):
         assert (a, b, c, d) == (ga, gb, gc, gd)
         assert abs(lnl - glnl) < max(abs(glnl) * 1e-6, 1e-4), \
             ((a, b, c, d), lnl, glnl)
+
+
+def test_grouped_quartets_match_reference_12(golden_dir):
+    """GROUPED_QUARTETS (-Y, quartets.c:585-600): all 81 cross-group
+    quartets x 3 topologies on the 12-taxon golden match the
+    reference's ExaML_quartets output line for line (groups file:
+    tests/golden/12.groups)."""
+    import examl_amd as ea
+    from examl_amd.examl_io import read_byte_file, read_newick_topology
+    from examl_amd.search import TreeSearch
+    from tests.helpers import OracleEngine
+    taxa, parts = read_byte_file(os.path.join(golden_dir, "12.binary"))
+    tree = read_newick_topology(os.path.join(golden_dir, "12.tree"), taxa)
+    engines = [OracleEngine(p.tips, p.wgt,
+                            ea.DnaGtrModel(p.frequencies, [1.0] * 6, 1.0))
+               for p in parts]
+    ts = TreeSearch(tree, engines,
+                    opt_freq_flags=[bool(p.optimizeBaseFrequencies)
+                                    for p in parts])
+    ts.tree_evaluation_mode()
+    out = compute_quartets(ts.engines, 12,
+                           groups=[[1, 2, 3], [4, 5, 6], [7, 8, 9],
+                                   [10, 11, 12]])
+    golden = _parse_golden(os.path.join(golden_dir,
+                                        "12.quartets.grouped.txt"))
+    assert len(golden) == 243 and len(out) == 243
+    for o, g in zip(out, golden):
+        assert (o[0], o[1], o[2], o[3]) == (g[0], g[1], g[2], g[3])
+        assert abs(o[4] - g[4]) < 1e-5, (o, g)
