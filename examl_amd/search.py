@@ -179,7 +179,12 @@ class TreeSearch:
         e = TravEntry()
         e.tipCase = tc
         e.pNumber, e.qNumber, e.rNumber = m.number, q.number, r.number
-        e.qz, e.rz = q.z, r.z
+        qz, rz = q.z, r.z
+        e.qz = float(qz[0]) if isinstance(qz, np.ndarray) else qz
+        e.rz = float(rz[0]) if isinstance(rz, np.ndarray) else rz
+        if self.NB > 1:
+            e.qzv = np.asarray(qz, dtype=float).copy()
+            e.rzv = np.asarray(rz, dtype=float).copy()
         e.x3Slot = t.clv_slot(m.number)
         e.x1Slot = q.number if t.is_tip(q.number) else t.clv_slot(q.number)
         e.x2Slot = r.number if t.is_tip(r.number) else t.clv_slot(r.number)
@@ -240,7 +245,12 @@ class TreeSearch:
         else:
             out = entries
         self._run(out)
-        zv = np.array([z]) if z is not None else t.get_zv(p, q)
+        if z is None:
+            zv = t.get_zv(p, q)
+        elif isinstance(z, np.ndarray):
+            zv = z
+        else:
+            zv = np.array([z])
         z = float(zv[0])
         if full:
             # td[0] of the last full traversal: evaluatePartialGeneric walks
@@ -331,15 +341,18 @@ class TreeSearch:
     # branch-length smoothing (searchAlgo.c:127-270,2635)
     # ------------------------------------------------------------------
 
-    def makenewz_generic_vec(self, p, q, z0, maxiter, mask):
+    def makenewz_generic_vec(self, p, q, z0, maxiter, mask, entries=None):
         """topLevelMakenewz for numBranches == NumberOfModels (-M):
         per-partition NR with curvatOK/outerConverged masks
         (makenewzGenericSpecial.c:849-1063) and the partitionConverged
         execute mask of makenewzGeneric(mask=TRUE) (:1369-1378)."""
         NB = self.NB
-        out = []
-        self._collect(p, q, True, out)
-        self._collect(q, p, True, out)
+        if entries is None:
+            out = []
+            self._collect(p, q, True, out)
+            self._collect(q, p, True, out)
+        else:
+            out = entries
         if mask:
             for i in range(NB):
                 self.execute_model[i] = not self.partition_converged[i]

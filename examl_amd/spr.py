@@ -56,10 +56,40 @@ class RingNode:
 
 
 def hookup(p, q, z):
-    """hookup (axml.c:478), numBranches=1."""
+    """hookup (axml.c:478): under -M z is a per-partition vector and the
+    reference copies it into both nodes — each edge owns its storage."""
     p.back = q
     q.back = p
+    if isinstance(z, np.ndarray):
+        z = z.copy()
     p.z = q.z = z
+
+
+def _branch_split(zqr, zqs, zrs):
+    """The three-way branch-length split of insertBIG (:512-530)."""
+    lzqr = math.log(zqr) if zqr > ZMIN else math.log(ZMIN)
+    lzqs = math.log(zqs) if zqs > ZMIN else math.log(ZMIN)
+    lzrs = math.log(zrs) if zrs > ZMIN else math.log(ZMIN)
+    lzsum = 0.5 * (lzqr + lzqs + lzrs)
+    lzq = lzsum - lzrs
+    lzr = lzsum - lzqs
+    lzs = lzsum - lzqr
+    lzmax = math.log(ZMAX)
+    if lzq > lzmax:
+        lzq, lzr, lzs = lzmax, lzqr, lzqs
+    elif lzr > lzmax:
+        lzr, lzq, lzs = lzmax, lzqr, lzrs
+    elif lzs > lzmax:
+        lzs, lzq, lzr = lzmax, lzqs, lzrs
+    return math.exp(lzq), math.exp(lzr), math.exp(lzs)
+
+
+def _sqrtz(z):
+    """sqrt + [zmin,zmax] clamp, scalar or per-partition (insertBIG:541)."""
+    if isinstance(z, np.ndarray):
+        return np.clip(np.sqrt(z), ZMIN, ZMAX)
+    z = math.sqrt(z)
+    return min(max(z, ZMIN), ZMAX)
 
 
 class SprTree:
@@ -152,14 +182,25 @@ class SprTree:
         raise KeyError((p, q))
 
     def get_z(self, a, b):
-        return self.find_member(a, b).z
+        z = self.find_member(a, b).z
+        return float(z[0]) if isinstance(z, np.ndarray) else z
 
     def get_zv(self, a, b):
-        return np.array([self.get_z(a, b)])
+        z = self.find_member(a, b).z
+        return z if isinstance(z, np.ndarray) else np.array([z])
 
     def set_z(self, a, b, z):
         m = self.find_member(a, b)
         m.z = m.back.z = z
+
+    def expand_branches(self, nb):
+        """Per-partition branch vectors (-M): every edge's two members
+        share one nb-vector, like p->z/p->back->z."""
+        for i in range(1, 2 * self.ntips - 1):
+            for m in self.members(i):
+                if m.back is not None and not isinstance(m.z, np.ndarray):
+                    v = np.full(nb, float(m.z))
+                    m.z = m.back.z = v
 
     def ring_children(self, p, parent):
         """the (q, r) of computeTraversalInfo: ring order from the
@@ -178,7 +219,8 @@ class SprTree:
                     yield m.back.number
 
         def __getitem__(self, q):
-            return self.t.get_z(self.p, q)
+            # raw member z: -M updates mutate the shared vector in place
+            return self.t.find_member(self.p, q).z
 
     class _Adj:
         def __init__(self, t):
@@ -221,7 +263,8 @@ def _save_subtree(p, tpl, ntips):
     """saveSubtree (topologies.c:229): returns the link index for branch
     p--p->back with children merged in ascending min-tip order."""
     links = tpl.links
-    r = [p, p.back, p.z, 0, 0, None]
+    z = p.z.copy() if isinstance(p.z, np.ndarray) else p.z
+    r = [p, p.back, z, 0, 0, None]
     ri = len(links)
     links.append(r)
     q = p.back
@@ -443,7 +486,6 @@ class SprSearch:
         self.ts = ts
         self.st = ts.tree
         assert isinstance(self.st, SprTree)
-        assert ts.NB == 1, "SPR with -M not supported yet"
         self.thorough = False
         self.do_cutoff = do_cutoff
         self.big_cutoff = big_cutoff
@@ -517,6 +559,10 @@ class SprSearch:
             self._collect_m(m1, True, out)
         if ts.oriented.get(m2.number) is not m2:
             self._collect_m(m2, True, out)
+        if ts.NB > 1:
+            return ts.makenewz_generic_vec(m1.number, m2.number, z0,
+                                           ITERATIONS, mask=False,
+                                           entries=out)
         return ts.makenewz_generic(m1.number, m2.number, z0, ITERATIONS,
                                    entries=out)
 
@@ -529,21 +575,37 @@ class SprSearch:
         ts = self.ts
         if self.st.is_tip(p.number):
             return False
-        ts.partition_converged = False
+        nb1 = ts.NB == 1
+        if nb1:
+            ts.partition_converged = False
+        else:
+            ts.partition_converged[:] = False
         while maxtimes > 0:
             maxtimes -= 1
-            ts.partition_smoothed = True
+            if nb1:
+                ts.partition_smoothed = True
+            else:
+                ts.partition_smoothed[:] = True
             q = p
             while True:
                 self._update(q)
                 q = q.next
                 if q is p:
                     break
-            if ts.partition_smoothed:  # allSmoothed
-                ts.partition_converged = True
+            smoothed = (ts.partition_smoothed if nb1
+                        else all(ts.partition_smoothed))
+            if smoothed:  # allSmoothed
+                if nb1:
+                    ts.partition_converged = True
+                else:
+                    ts.partition_converged[:] = True
                 break
-        ts.partition_smoothed = False
-        ts.partition_converged = False
+        if nb1:
+            ts.partition_smoothed = False
+            ts.partition_converged = False
+        else:
+            ts.partition_smoothed[:] = False
+            ts.partition_converged[:] = False
         return True
 
     # ---- SPR surgery -------------------------------------------------
@@ -574,30 +636,29 @@ class SprSearch:
         r = q.back
         s = p.back
         if self.thorough:
+            nb = self.ts.NB
             qz = q.z
+            dz = np.full(nb, DEFAULTZ) if nb > 1 else DEFAULTZ
             zqr = self._makenewz(q, r, qz)
-            zqs = self._makenewz(q, s, DEFAULTZ)
-            zrs = self._makenewz(r, s, DEFAULTZ)
-            lzqr = math.log(zqr) if zqr > ZMIN else math.log(ZMIN)
-            lzqs = math.log(zqs) if zqs > ZMIN else math.log(ZMIN)
-            lzrs = math.log(zrs) if zrs > ZMIN else math.log(ZMIN)
-            lzsum = 0.5 * (lzqr + lzqs + lzrs)
-            lzq = lzsum - lzrs
-            lzr = lzsum - lzqs
-            lzs = lzsum - lzqr
-            lzmax = math.log(ZMAX)
-            if lzq > lzmax:
-                lzq, lzr, lzs = lzmax, lzqr, lzqs
-            elif lzr > lzmax:
-                lzr, lzq, lzs = lzmax, lzqr, lzrs
-            elif lzs > lzmax:
-                lzs, lzq, lzr = lzmax, lzqs, lzrs
-            hookup(p.next, q, math.exp(lzq))
-            hookup(p.next.next, r, math.exp(lzr))
-            hookup(p, s, math.exp(lzs))
+            zqs = self._makenewz(q, s, dz)
+            zrs = self._makenewz(r, s, dz)
+            if nb > 1:
+                e1 = np.empty(nb)
+                e2 = np.empty(nb)
+                e3 = np.empty(nb)
+                for i in range(nb):
+                    e1[i], e2[i], e3[i] = _branch_split(zqr[i], zqs[i],
+                                                        zrs[i])
+                hookup(p.next, q, e1)
+                hookup(p.next.next, r, e2)
+                hookup(p, s, e3)
+            else:
+                e1, e2, e3 = _branch_split(zqr, zqs, zrs)
+                hookup(p.next, q, e1)
+                hookup(p.next.next, r, e2)
+                hookup(p, s, e3)
         else:
-            z = math.sqrt(q.z)
-            z = min(max(z, ZMIN), ZMAX)
+            z = _sqrtz(q.z)
             hookup(p.next, q, z)
             hookup(p.next.next, r, z)
         self._newview(p)
@@ -617,8 +678,7 @@ class SprSearch:
             hookup(p.next.next, r, self.current_lzr)
             hookup(p, s, self.current_lzs)
         else:
-            z = math.sqrt(q.z)
-            z = min(max(z, ZMIN), ZMAX)
+            z = _sqrtz(q.z)
             hookup(p.next, q, z)
             hookup(p.next.next, r, z)
         self._newview(p)
@@ -791,8 +851,7 @@ class SprSearch:
             hookup(p.next.next, r, self.current_lzr)
             hookup(p, s, self.current_lzs)
         else:
-            z = math.sqrt(q.z)
-            z = min(max(z, ZMIN), ZMAX)
+            z = _sqrtz(q.z)
             hookup(p.next, q, z)
             hookup(p.next.next, r, z)
         self.ts.likelihood = self.best_of_node

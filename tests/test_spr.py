@@ -339,3 +339,40 @@ def test_spr_psr_search_12_cpu(golden_dir):
     sp, ts = _setup_12(golden_dir, cat=True)
     lnl = sp.compute_big_rapid(estimate_model=True)
     assert abs(lnl - (-2507.657682)) < abs(2507.657682) * 1e-6, lnl
+
+
+def test_spr_M_search_12_cpu(golden_dir):
+    """-f d under -M (per-partition branch lengths): the SPR machinery
+    with vector z through removeNodeBIG/insertBIG (zqr/defaultz starts
+    into the vectorized NR, the per-partition three-way branch split of
+    insertBIG:512) and localSmooth's per-partition masks.  Reference
+    golden: examl-AVX -s 12m.binary -M, final -2728.477352 and the
+    result topology."""
+    from examl_amd.examl_io import (parse_newick_topology, read_byte_file,
+                                    read_newick_topology)
+    from examl_amd.spr import RfConvergence
+    from tests.helpers import OracleEngine
+    taxa, parts = read_byte_file(os.path.join(golden_dir, "12m.binary"))
+    tree = read_newick_topology(os.path.join(golden_dir, "12.tree"), taxa)
+    st = SprTree.from_phylo(tree)
+    engines = [OracleEngine(p.tips, p.wgt,
+                            ea.DnaGtrModel(p.frequencies, [1.0] * 6, 1.0))
+               for p in parts]
+    ts = TreeSearch(st, engines,
+                    opt_freq_flags=[bool(p.optimizeBaseFrequencies)
+                                    for p in parts],
+                    per_gene_bl=True)
+    sp = SprSearch(ts)
+    lnl = sp.compute_big_rapid(estimate_model=True)
+    assert abs(lnl - (-2728.477352)) < abs(2728.477352) * 1e-6, lnl
+    ours = {frozenset(b) for b in RfConvergence(st)._bipartitions()}
+    with open(os.path.join(golden_dir, "12m.result.tree")) as f:
+        ref = parse_newick_topology(f.read(), taxa, read_bl=True)
+    theirs = {frozenset(b)
+              for b in RfConvergence(SprTree.from_phylo(ref))
+              ._bipartitions()}
+    assert ours == theirs
+    # per-partition branch lengths actually differ on some edge
+    import numpy as np
+    assert any(len(set(np.round(st.get_zv(a, b), 12))) > 1
+               for a, b in st.edges())
