@@ -1,0 +1,264 @@
+"""Command-line drop-in for the reference `examl` binary.
+
+Mirrors the reference's option surface (axml.c get_args:941) over the
+MI355X engines: the same flags mean the same things, the same output
+files appear (ExaML_info/ExaML_result/ExaML_quartets/goodTrees), and a
+reference user can point their invocation here unchanged:
+
+    python -m examl_amd -s 49.binary -t 49.tree -m GAMMA -f E -n RUN
+
+Supported: -s -t -g -p -m -f(e|E|d|o|q) -n -w -a -M -S -D -B -c -e -i
+-r -Y.  This is the PRODUCT path: it requires the HIP extension and an
+AMD GPU and fails loudly without one (no CPU fallback).
+"""
+
+import os
+import sys
+
+import numpy as np
+
+
+def _usage():
+    sys.stderr.write(
+        "examl_amd (MI355X-native ExaML)\n"
+        "  -s byteFile -n runName and one of -t startingTree | "
+        "-g constraintTree -p seed\n"
+        "  [-m GAMMA|PSR] [-f e|E|d|o|q] [-w workdir] [-a] [-M] [-S]\n"
+        "  [-D] [-B numBestTrees] [-c numRateCategories] "
+        "[-e likelihoodEpsilon]\n"
+        "  [-i initialRearrangementSetting] [-r randomQuartets -p seed] "
+        "[-Y quartetGroupingFile]\n")
+
+
+def _parse_args(argv):
+    opts = {"m": "GAMMA", "f": "d", "w": os.getcwd(), "a": False,
+            "M": False, "S": False, "D": False, "B": 0, "c": 25,
+            "e": 0.1, "i": None, "r": 0, "p": None, "s": None, "t": None,
+            "g": None, "n": None, "Y": None}
+    flags = set("aMSD")
+    valued = set("stgpmfnwBceirY")
+    i = 0
+    while i < len(argv):
+        a = argv[i]
+        if not a.startswith("-") or len(a) != 2:
+            sys.exit(f"unknown argument {a!r}")
+        c = a[1]
+        if c == "h":
+            _usage()
+            sys.exit(0)
+        if c in flags:
+            opts[c] = True
+            i += 1
+        elif c in valued:
+            if i + 1 >= len(argv):
+                sys.exit(f"option -{c} needs a value")
+            v = argv[i + 1]
+            if c in "Bci":
+                v = int(v)
+            elif c == "e":
+                v = float(v)
+            elif c == "p":
+                v = int(v)
+            elif c == "r":
+                v = int(v)
+            opts[c] = v
+            i += 2
+        else:
+            sys.exit(f"unknown option -{c}")
+    return opts
+
+
+LG4M, LG4X = 20, 21
+
+
+def _build_engines(parts, opts, device):
+    """One engine per partition, by data type and rate model — the
+    initializePartitions role (axml.c:1936)."""
+    import examl_amd as ea
+    aa = np.load(os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                              "data", "aa_models.npz"))
+    psr = opts["m"] == "PSR"
+    engines = []
+    for p in parts:
+        if p.states == 4:
+            model = ea.DnaGtrModel(p.frequencies, [1.0] * 6, 1.0,
+                                   use_median=opts["a"])
+            if psr:
+                w = p.upper - p.lower
+                engines.append(ea.DnaCatEngine(
+                    p.tips, p.wgt, model, np.zeros(w, dtype=np.int32),
+                    np.array([1.0]), device=device))
+            elif opts["S"]:
+                engines.append(ea.SaveDnaEngine(p.tips, p.wgt, model,
+                                                device=device))
+            else:
+                engines.append(ea.DnaGammaEngine(p.tips, p.wgt, model,
+                                                 device=device))
+        else:
+            if p.protModels == LG4M:
+                engines.append(ea.Lg4Engine(p.tips, p.wgt,
+                                            ea.Lg4Model.lg4m(),
+                                            device=device))
+                continue
+            if p.protModels == LG4X:
+                engines.append(ea.Lg4Engine(p.tips, p.wgt,
+                                            ea.Lg4Model.lg4x(),
+                                            device=device))
+                continue
+            freqs = (aa["frequencies"][p.protModels] if p.protFreqs == 0
+                     else p.frequencies)
+            model = ea.ProtGtrModel(freqs, aa["rates190"][p.protModels],
+                                    1.0, use_median=opts["a"])
+            if psr:
+                w = p.upper - p.lower
+                engines.append(ea.ProtCatEngine(
+                    p.tips, p.wgt, model, np.zeros(w, dtype=np.int32),
+                    np.array([1.0]), device=device))
+            elif opts["S"]:
+                sys.exit("-S for protein partitions is not yet available "
+                         "on the GPU engines (GAMMA protein -S: round 2)")
+            else:
+                engines.append(ea.DnaGammaEngine(p.tips, p.wgt, model,
+                                                 device=device))
+    return engines
+
+
+def main(argv=None, device=None):
+    opts = _parse_args(sys.argv[1:] if argv is None else argv)
+    if not opts["s"] or not opts["n"]:
+        _usage()
+        sys.exit("-s and -n are required")
+    if not opts["t"] and not opts["g"]:
+        _usage()
+        sys.exit("specify a starting tree: -t treeFile or "
+                 "-g constraintTree -p seed")
+    if opts["g"] and opts["p"] is None:
+        sys.exit("you must specify a random number seed via -p when "
+                 "using a constraint tree")
+    if opts["f"] == "q" and opts["r"] == 0 and not opts["Y"]:
+        sys.exit("you must specify either -r randomQuartetNumber or "
+                 "-Y quartetGroupingFileName with -f q")
+    if device is None:
+        # PRODUCT path: MI355X only, no CPU fallback
+        import torch
+        if not torch.cuda.is_available():
+            sys.exit("examl_amd requires an AMD GPU (the HIP engines); "
+                     "torch.cuda is not available on this host")
+        device = "cuda:0"
+
+    import examl_amd as ea
+    from examl_amd.examl_io import (read_byte_file, read_newick_trees,
+                                    to_newick)
+    from examl_amd.search import TreeSearch, evaluate_trees
+    from examl_amd.spr import (SprSearch, SprTree, read_constraint_tree)
+
+    name = opts["n"]
+    wdir = opts["w"]
+    info_path = os.path.join(wdir, f"ExaML_info.{name}")
+    info_f = open(info_path, "w")
+
+    def log(msg):
+        info_f.write(msg + "\n")
+        info_f.flush()
+        print(msg)
+
+    taxa, parts = read_byte_file(opts["s"])
+    log(f"partitions: {len(parts)}, taxa: {len(taxa)}, model {opts['m']}")
+    engines = _build_engines(parts, opts, device)
+    kw = dict(opt_freq_flags=[bool(p.optimizeBaseFrequencies)
+                              for p in parts],
+              max_categories=opts["c"])
+    if opts["m"] == "PSR":
+        kw["rate_het"] = "CAT"
+    if opts["M"]:
+        kw["per_gene_bl"] = True
+
+    mode = opts["f"]
+    if mode in ("e", "E"):
+        trees = read_newick_trees(opts["t"], taxa)
+        lnls = evaluate_trees(trees, engines, fast=(mode == "e"),
+                              epsilon=opts["e"], **kw)
+        best = int(np.argmax(lnls))
+        for i, v in enumerate(lnls):
+            log(f"Likelihood tree {i}: {v:.6f}")
+        # trees are optimized in place; emit the best-scoring one
+        result = to_newick(trees[best], taxa)
+        final = lnls[best]
+    elif mode in ("d", "o"):
+        trees = read_newick_trees(opts["t"], taxa) if opts["t"] else None
+        if opts["g"]:
+            with open(opts["g"]) as f:
+                st, cv = read_constraint_tree(f.read(), taxa, opts["p"])
+        else:
+            st = SprTree.from_phylo(trees[0])
+            cv = None
+        ts = TreeSearch(st, engines, **kw)
+        sp = SprSearch(ts, do_cutoff=(mode == "d"),
+                       convergence_criterion=opts["D"],
+                       save_best_trees=opts["B"], log=log)
+        sp.constraint = cv
+        final = sp.compute_big_rapid(estimate_model=True,
+                                     initial_trav=opts["i"])
+        log(f"Likelihood of best tree: {final:.6f}")
+        # the result file holds the best tree (written BEFORE the -B
+        # good-trees replay, which leaves the tree at the last entry)
+        result = to_newick(st, taxa)
+        if opts["B"]:
+            gt_path = os.path.join(
+                wdir, f"RAxML_{len(sp.good_trees)}_goodTrees.{name}")
+            with open(gt_path, "w") as f:
+                for i, lnl in enumerate(sp.good_trees):
+                    sp.best_ml.recall(i + 1, ts)
+                    f.write(to_newick(st, taxa) + "\n")
+                    log(f"tree {i + 1} likelihood {lnl:.6f}")
+    elif mode == "q":
+        from examl_amd.quartets import compute_quartets
+        trees = read_newick_trees(opts["t"], taxa)
+        ts = TreeSearch(trees[0], engines, **kw)
+        ts.tree_evaluation_mode(epsilon=opts["e"])
+        groups = None
+        if opts["Y"]:
+            groups = _parse_groups(opts["Y"], taxa)
+        quartets = compute_quartets(
+            ts.engines, len(taxa), random_quartets=opts["r"],
+            seed=opts["p"] or 0, groups=groups,
+            per_gene_bl=opts["M"])
+        q_path = os.path.join(wdir, f"ExaML_quartets.{name}")
+        with open(q_path, "w") as f:
+            f.write("Taxon names and indices:\n\n")
+            for i, t in enumerate(taxa):
+                f.write(f"{t} {i + 1}\n")
+            f.write("\n")
+            for a, b, c, d, lnl in quartets:
+                f.write(f"{a} {b} | {c} {d}: {lnl:.6f}\n")
+        log(f"quartets written to {q_path}")
+        info_f.close()
+        return 0
+    else:
+        sys.exit(f"unknown -f mode {mode!r}")
+
+    res_path = os.path.join(wdir, f"ExaML_result.{name}")
+    with open(res_path, "w") as f:
+        f.write(result + "\n")
+    log(f"Final tree written to: {res_path}")
+    info_f.close()
+    return 0
+
+
+def _parse_groups(path, taxa):
+    """groupingParser (quartets.c:69): four parenthesized taxon lists."""
+    tip_no = {t: i + 1 for i, t in enumerate(taxa)}
+    with open(path) as f:
+        text = f.read()
+    groups = []
+    for chunk in text.split(")")[:-1]:
+        chunk = chunk.split("(")[-1]
+        names = [x.strip() for x in chunk.split(",") if x.strip()]
+        groups.append([tip_no[n] for n in names])
+    if len(groups) != 4:
+        sys.exit("quartet grouping file must define exactly 4 groups")
+    return groups
+
+
+if __name__ == "__main__":
+    sys.exit(main())

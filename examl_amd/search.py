@@ -1221,19 +1221,21 @@ class TreeSearch:
         for a, b in self.tree.edges():
             self.tree.set_z(a, b, DEFAULTZ)
 
-    def tree_evaluation_mode(self, log=None):
+    def tree_evaluation_mode(self, log=None, epsilon=0.1):
         """The -f E (slow TREE_EVALUATION) flow for one tree
-        (axml.c:2316-2331): evaluate, treeEvaluate(1), modOpt(0.1)."""
+        (axml.c:2316-2331): evaluate, treeEvaluate(1), modOpt with
+        adef->likelihoodEpsilon (-e, default 0.1)."""
         self.evaluate_generic(full=True)
         if log:
             log(f"initial lnL = {self.likelihood:.6f}")
         self.tree_evaluate(1.0)
         if log:
             log(f"after treeEvaluate = {self.likelihood:.6f}")
-        return self.mod_opt(0.1, log=log)
+        return self.mod_opt(epsilon, log=log)
 
 
-def evaluate_trees(trees, engines, fast=False, log=None, **search_kwargs):
+def evaluate_trees(trees, engines, fast=False, log=None, epsilon=0.1,
+                   **search_kwargs):
     """optimizeTrees (axml.c:2721) for the -f E (slow) / -f e (fast)
     multi-tree input: tree 0 gets the full treeEvaluate(1)+modOpt(0.1),
     later trees get resetBranches and -- in fast mode -- only
@@ -1249,6 +1251,6 @@ def evaluate_trees(trees, engines, fast=False, log=None, **search_kwargs):
             ts.tree_evaluate(2.0)
         else:
             ts.tree_evaluate(1.0)
-            ts.mod_opt(0.1, log=log)
+            ts.mod_opt(epsilon, log=log)
         out.append(ts.likelihood)
     return out
