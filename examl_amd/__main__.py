@@ -8,8 +8,11 @@ reference user can point their invocation here unchanged:
     python -m examl_amd -s 49.binary -t 49.tree -m GAMMA -f E -n RUN
 
 Supported: -s -t -g -p -m -f(e|E|d|o|q) -n -w -a -M -S -D -B -c -e -i
--r -Y.  This is the PRODUCT path: it requires the HIP extension and an
-AMD GPU and fails loudly without one (no CPU fallback).
+-r -Y -v, plus AUTO protein partitions (per-partition model selection).
+Not yet wired: -R/-I (checkpoint restart; the binary checkpoint
+read/write interchange itself lives in examl_amd.checkpoint).  This is
+the PRODUCT path: it requires the HIP extension and an AMD GPU and
+fails loudly without one (no CPU fallback).
 """
 
 import os
@@ -46,6 +49,9 @@ def _parse_args(argv):
         if c == "h":
             _usage()
             sys.exit(0)
+        if c == "v":
+            print("examl_amd: MI355X-native ExaML-compatible engine")
+            sys.exit(0)
         if c in flags:
             opts[c] = True
             i += 1
@@ -68,18 +74,24 @@ def _parse_args(argv):
     return opts
 
 
-LG4M, LG4X = 20, 21
+LG4M, LG4X, AUTO = 20, 21, 19
+WAG = 4
 
 
 def _build_engines(parts, opts, device):
     """One engine per partition, by data type and rate model — the
-    initializePartitions role (axml.c:1936)."""
+    initializePartitions role (axml.c:1936).  Returns (engines,
+    auto_flags, empirical_freqs) for the AUTO selection machinery."""
     import examl_amd as ea
     aa = np.load(os.path.join(os.path.dirname(os.path.abspath(__file__)),
                               "data", "aa_models.npz"))
     psr = opts["m"] == "PSR"
     engines = []
+    auto_flags = []
+    empirical = []
     for p in parts:
+        auto_flags.append(p.states == 20 and p.protModels == AUTO)
+        empirical.append(p.frequencies)
         if p.states == 4:
             model = ea.DnaGtrModel(p.frequencies, [1.0] * 6, 1.0,
                                    use_median=opts["a"])
@@ -95,6 +107,16 @@ def _build_engines(parts, opts, device):
                 engines.append(ea.DnaGammaEngine(p.tips, p.wgt, model,
                                                  device=device))
         else:
+            if p.protModels == AUTO:
+                # AUTO starts as WAG (models.c:4222); protFreqs==0 ->
+                # empirical frequencies initially (models.c:3528)
+                freqs = (p.frequencies if p.protFreqs == 0
+                         else aa["frequencies"][WAG])
+                model = ea.ProtGtrModel(freqs, aa["rates190"][WAG], 1.0,
+                                        use_median=opts["a"])
+                engines.append(ea.DnaGammaEngine(p.tips, p.wgt, model,
+                                                 device=device))
+                continue
             if p.protModels == LG4M:
                 engines.append(ea.Lg4Engine(p.tips, p.wgt,
                                             ea.Lg4Model.lg4m(),
@@ -120,7 +142,7 @@ def _build_engines(parts, opts, device):
             else:
                 engines.append(ea.DnaGammaEngine(p.tips, p.wgt, model,
                                                  device=device))
-    return engines
+    return engines, auto_flags, empirical
 
 
 def main(argv=None, device=None):
@@ -164,20 +186,39 @@ def main(argv=None, device=None):
 
     taxa, parts = read_byte_file(opts["s"])
     log(f"partitions: {len(parts)}, taxa: {len(taxa)}, model {opts['m']}")
-    engines = _build_engines(parts, opts, device)
+    engines, auto_flags, empirical = _build_engines(parts, opts, device)
     kw = dict(opt_freq_flags=[bool(p.optimizeBaseFrequencies)
                               for p in parts],
               max_categories=opts["c"])
+    if any(auto_flags):
+        kw["auto_flags"] = auto_flags
+        kw["empirical_freqs"] = empirical
     if opts["m"] == "PSR":
         kw["rate_het"] = "CAT"
     if opts["M"]:
         kw["per_gene_bl"] = True
 
+    prot_freqs0 = [p.protFreqs for p in parts]
+
+    def _ts(tree):
+        ts = TreeSearch(tree, engines, **kw)
+        ts.prot_freqs = list(prot_freqs0)
+        return ts
+
     mode = opts["f"]
     if mode in ("e", "E"):
         trees = read_newick_trees(opts["t"], taxa)
-        lnls = evaluate_trees(trees, engines, fast=(mode == "e"),
-                              epsilon=opts["e"], **kw)
+        if any(auto_flags):
+            # AUTO + multi-tree: evaluate per tree with fresh searches
+            lnls = []
+            for i, tree in enumerate(trees):
+                ts = _ts(tree)
+                if i > 0:
+                    ts.reset_branches()
+                lnls.append(ts.tree_evaluation_mode(epsilon=opts["e"]))
+        else:
+            lnls = evaluate_trees(trees, engines, fast=(mode == "e"),
+                                  epsilon=opts["e"], **kw)
         best = int(np.argmax(lnls))
         for i, v in enumerate(lnls):
             log(f"Likelihood tree {i}: {v:.6f}")
@@ -192,7 +233,7 @@ def main(argv=None, device=None):
         else:
             st = SprTree.from_phylo(trees[0])
             cv = None
-        ts = TreeSearch(st, engines, **kw)
+        ts = _ts(st)
         sp = SprSearch(ts, do_cutoff=(mode == "d"),
                        convergence_criterion=opts["D"],
                        save_best_trees=opts["B"], log=log)
@@ -214,7 +255,7 @@ def main(argv=None, device=None):
     elif mode == "q":
         from examl_amd.quartets import compute_quartets
         trees = read_newick_trees(opts["t"], taxa)
-        ts = TreeSearch(trees[0], engines, **kw)
+        ts = _ts(trees[0])
         ts.tree_evaluation_mode(epsilon=opts["e"])
         groups = None
         if opts["Y"]:
