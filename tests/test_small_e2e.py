@@ -91,3 +91,25 @@ def test_12_prot_wag_jtt_f_E(golden_dir):
 
 def test_12_lg4x_lg4m_f_E(golden_dir):
     _run(golden_dir, "12lg4.binary", -7387.472983, prot=True)
+
+
+def test_12_psr_c10_f_E(golden_dir):
+    """-c 10 (tr->maxCategories, axml.c:1117): the PSR pipeline under a
+    non-default category cap lands on the reference's -3268.753243
+    (vs -3233.904617 at the default 25 — the cap genuinely binds)."""
+    from tests.helpers import OracleCatEngine
+    taxa, parts = read_byte_file(os.path.join(golden_dir, "12.binary"))
+    tree = read_newick_topology(os.path.join(golden_dir, "12.tree"), taxa)
+    engines = []
+    for p in parts:
+        m = ea.DnaGtrModel(p.frequencies, [1.0] * 6, 1.0)
+        w = p.upper - p.lower
+        engines.append(OracleCatEngine(p.tips, p.wgt, m,
+                                       np.zeros(w, dtype=np.int32),
+                                       np.array([1.0])))
+    ts = TreeSearch(tree, engines,
+                    opt_freq_flags=[bool(p.optimizeBaseFrequencies)
+                                    for p in parts],
+                    rate_het="CAT", max_categories=10)
+    lnl = ts.tree_evaluation_mode()
+    assert abs(lnl - (-3268.753243)) < abs(3268.753243) * 1e-6, lnl
