@@ -77,3 +77,84 @@ def test_cli_f_E_12_gpu(golden_dir, tmp_path):
     lnl = float(line.split(":")[1])
     assert abs(lnl - (-3650.993621)) < abs(3650.993621) * 1e-6
     assert os.path.exists(os.path.join(tmp_path, "ExaML_result.CLI12"))
+
+
+def _oracle_build_engines(parts, opts, device):
+    """Test-only stand-in for the GPU engine factory: same per-partition
+    dispatch over the CPU oracle engines (the product path never does
+    this — oracle/ is test infrastructure)."""
+    import numpy as np
+
+    import examl_amd as ea
+    from tests.helpers import OracleCatEngine, OracleEngine
+    psr = opts["m"] == "PSR"
+    engines = []
+    auto_flags = []
+    empirical = []
+    for p in parts:
+        auto_flags.append(False)
+        empirical.append(p.frequencies)
+        m = ea.DnaGtrModel(p.frequencies, [1.0] * 6, 1.0,
+                           use_median=opts["a"])
+        if psr:
+            w = p.upper - p.lower
+            engines.append(OracleCatEngine(p.tips, p.wgt, m,
+                                           np.zeros(w, dtype=np.int32),
+                                           np.array([1.0])))
+        else:
+            engines.append(OracleEngine(p.tips, p.wgt, m))
+    return engines, auto_flags, empirical
+
+
+def test_cli_full_flow_cpu(golden_dir, tmp_path, monkeypatch):
+    """The whole CLI flow (parse -> engines -> mode -> output files) on
+    CPU by swapping only the engine factory for the oracle engines; the
+    numbers are the same reference goldens the engine-level tests pin."""
+    import examl_amd.__main__ as cli
+    monkeypatch.setattr(cli, "_build_engines", _oracle_build_engines)
+    # -f E
+    rc = cli.main(["-s", os.path.join(golden_dir, "12.binary"),
+                   "-t", os.path.join(golden_dir, "12.tree"),
+                   "-n", "E", "-f", "E", "-w", str(tmp_path)],
+                  device="cpu")
+    assert rc == 0
+    info = open(os.path.join(tmp_path, "ExaML_info.E")).read()
+    lnl = float([ln for ln in info.splitlines()
+                 if ln.startswith("Likelihood tree 0:")][0].split(":")[1])
+    assert abs(lnl - (-3650.993621)) < 1e-2
+    assert os.path.exists(os.path.join(tmp_path, "ExaML_result.E"))
+    # -f d with -B and -D
+    rc = cli.main(["-s", os.path.join(golden_dir, "12.binary"),
+                   "-t", os.path.join(golden_dir, "12.tree"),
+                   "-n", "D", "-f", "d", "-B", "3", "-D",
+                   "-w", str(tmp_path)], device="cpu")
+    assert rc == 0
+    info = open(os.path.join(tmp_path, "ExaML_info.D")).read()
+    lnl = float([ln for ln in info.splitlines()
+                 if ln.startswith("Likelihood of best tree:")]
+                [0].split(":")[1])
+    assert abs(lnl - (-2741.473102)) < 1e-2
+    assert os.path.exists(os.path.join(tmp_path,
+                                       "RAxML_3_goodTrees.D"))
+    # -f q with -Y
+    rc = cli.main(["-s", os.path.join(golden_dir, "12.binary"),
+                   "-t", os.path.join(golden_dir, "12.tree"),
+                   "-n", "Q", "-f", "q",
+                   "-Y", os.path.join(golden_dir, "12.groups"),
+                   "-w", str(tmp_path)], device="cpu")
+    assert rc == 0
+    qlines = [ln for ln in
+              open(os.path.join(tmp_path, "ExaML_quartets.Q"))
+              if "|" in ln]
+    assert len(qlines) == 243
+    # -g constraint path
+    rc = cli.main(["-s", os.path.join(golden_dir, "12.binary"),
+                   "-g", os.path.join(golden_dir, "12.constraint.tree"),
+                   "-p", "12345", "-n", "G", "-f", "d",
+                   "-w", str(tmp_path)], device="cpu")
+    assert rc == 0
+    info = open(os.path.join(tmp_path, "ExaML_info.G")).read()
+    lnl = float([ln for ln in info.splitlines()
+                 if ln.startswith("Likelihood of best tree:")]
+                [0].split(":")[1])
+    assert abs(lnl - (-3435.016697)) < 1e-2
