@@ -37,9 +37,9 @@ def _parse_args(argv):
     opts = {"m": "GAMMA", "f": "d", "w": os.getcwd(), "a": False,
             "M": False, "S": False, "D": False, "B": 0, "c": 25,
             "e": 0.1, "i": None, "r": 0, "p": None, "s": None, "t": None,
-            "g": None, "n": None, "Y": None}
+            "g": None, "n": None, "Y": None, "R": None}
     flags = set("aMSD")
-    valued = set("stgpmfnwBceirY")
+    valued = set("stgpmfnwBceirYR")
     i = 0
     while i < len(argv):
         a = argv[i]
@@ -150,10 +150,12 @@ def main(argv=None, device=None):
     if not opts["s"] or not opts["n"]:
         _usage()
         sys.exit("-s and -n are required")
-    if not opts["t"] and not opts["g"]:
+    if not opts["t"] and not opts["g"] and not opts["R"]:
         _usage()
-        sys.exit("specify a starting tree: -t treeFile or "
-                 "-g constraintTree -p seed")
+        sys.exit("specify a starting tree: -t treeFile, "
+                 "-g constraintTree -p seed, or -R checkpointFile")
+    if opts["R"] and (opts["M"] or opts["m"] == "PSR"):
+        sys.exit("-R restart is currently wired for GAMMA without -M")
     if opts["g"] and opts["p"] is None:
         sys.exit("you must specify a random number seed via -p when "
                  "using a constraint tree")
@@ -186,6 +188,54 @@ def main(argv=None, device=None):
 
     taxa, parts = read_byte_file(opts["s"])
     log(f"partitions: {len(parts)}, taxa: {len(taxa)}, model {opts['m']}")
+
+    if opts["R"]:
+        # restart (searchAlgo.c:1726): tree from the checkpoint's node
+        # image, engines rebuilt from the stored model arrays (our
+        # eigendecomposition reproduces the stored one bit-for-bit)
+        import examl_amd as ea
+        from examl_amd.checkpoint import (FAST_SPRS, MOD_OPT, SLOW_SPRS,
+                                          read_checkpoint, spr_tree)
+        from examl_amd.search import TreeSearch
+        from examl_amd.spr import SprSearch
+        from examl_amd.examl_io import to_newick
+        ckpt = read_checkpoint(opts["R"], len(taxa),
+                               [p.states for p in parts])
+        engines = []
+        for p, m in zip(parts, ckpt.models):
+            assert p.states == 4, "-R restart wired for DNA GAMMA"
+            model = ea.DnaGtrModel(m["frequencies"], m["substRates"],
+                                   m["alpha"], use_median=opts["a"])
+            engines.append(ea.DnaGammaEngine(p.tips, p.wgt, model,
+                                             device=device))
+        kw = dict(opt_freq_flags=[bool(p.optimizeBaseFrequencies)
+                                  for p in parts],
+                  max_categories=opts["c"])
+        res_path = os.path.join(wdir, f"ExaML_result.{name}")
+        if opts["f"] in ("d", "o"):
+            assert ckpt.state in (FAST_SPRS, SLOW_SPRS), \
+                "checkpoint state does not match -f d"
+            st = spr_tree(ckpt, len(taxa))
+            ts = TreeSearch(st, engines, **kw)
+            sp = SprSearch(ts, do_cutoff=(opts["f"] == "d"),
+                           convergence_criterion=opts["D"],
+                           save_best_trees=opts["B"], log=log)
+            final = sp.compute_big_rapid(estimate_model=True,
+                                         checkpoint=ckpt)
+            tree_out = st
+        else:
+            assert ckpt.state == MOD_OPT, "checkpoint state/mode mismatch"
+            ts = TreeSearch(ckpt.tree, engines, **kw)
+            ts.evaluate_generic(full=True)
+            final = ts.mod_opt(opts["e"])
+            tree_out = ckpt.tree
+        log(f"Likelihood of best tree: {final:.6f}")
+        with open(res_path, "w") as f:
+            f.write(to_newick(tree_out, taxa) + "\n")
+        log(f"Final tree written to: {res_path}")
+        info_f.close()
+        return 0
+
     engines, auto_flags, empirical = _build_engines(parts, opts, device)
     kw = dict(opt_freq_flags=[bool(p.optimizeBaseFrequencies)
                               for p in parts],
@@ -197,6 +247,8 @@ def main(argv=None, device=None):
         kw["rate_het"] = "CAT"
     if opts["M"]:
         kw["per_gene_bl"] = True
+
+    prot_freqs0 = [p.protFreqs for p in parts]
 
     prot_freqs0 = [p.protFreqs for p in parts]
 

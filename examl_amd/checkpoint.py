@@ -50,7 +50,10 @@ NMLNGTH = 256
 # subst)
 _PLEN = {4: (4, 16, 16, 4, 64, 6), 20: (20, 400, 400, 20, 460, 190)}
 
-MOD_OPT = 4  # ckp.state (axml.h:658)
+REARR_SETTING = 1  # ckp.state (axml.h:655-659)
+FAST_SPRS = 2
+SLOW_SPRS = 3
+MOD_OPT = 4
 
 # rateHetModel values as stored by the reference (axml.h): CAT=0, GAMMA=1
 RATE_HET_CAT = 0
@@ -81,6 +84,18 @@ def read_checkpoint(path, mxtips, states_per_model, rate_het="GAMMA",
     ck.accumulated_time = struct.unpack_from("<d", d, 56)[0]
     ck.cat_opt = struct.unpack_from("<i", d, 180)[0]
     ck.tree_iteration = struct.unpack_from("<i", d, 184)[0]
+    # search-algorithm state (checkPointState, axml.h:679-720)
+    (ck.rearrangements_max, ck.rearrangements_min, ck.thorough_iterations,
+     ck.fast_iterations, ck.tree_vector_length, ck.mintrav, ck.maxtrav,
+     ck.best_trav, ck.thorough) = struct.unpack_from("<9i", d, 12)
+    (ck.start_lh, ck.lh, ck.previous_lh, ck.difference,
+     ck.epsilon) = struct.unpack_from("<5d", d, 64)
+    ck.impr, ck.cutoff = struct.unpack_from("<2i", d, 104)
+    (ck.tr_start_lh, ck.tr_end_lh, ck.tr_likelihood, ck.tr_best_of_node,
+     ck.tr_lh_cutoff, ck.tr_lh_avg,
+     ck.tr_lh_dec) = struct.unpack_from("<7d", d, 112)
+    (ck.tr_number_of_categories, ck.tr_it_count,
+     ck.tr_do_cutoff) = struct.unpack_from("<3i", d, 168)
     constraint = struct.unpack_from("<i", d, 8)[0]
     c = 1248
     ck.likelihood_epsilon = struct.unpack_from("<d", d, c + 24)[0]
@@ -146,6 +161,7 @@ def read_checkpoint(path, mxtips, states_per_model, rate_het="GAMMA",
     ck.start_number = struct.unpack_from("<i", d, off)[0]
     off += 4
     base = struct.unpack_from("<Q", d, off)[0]
+    ck._base = base
     off += 8
     x = _node_count(mxtips)
     recs = []
@@ -181,7 +197,62 @@ def read_checkpoint(path, mxtips, states_per_model, rate_het="GAMMA",
         tree.adj[number][nbr] = zv
         tree.adj[nbr][number] = zv
     ck.tree = tree
+    ck._recs = recs
+    ck._nb = nb
     return ck
+
+
+def spr_tree(ck, mxtips):
+    """Reconstruct the EXACT ring structure (member identity, ring
+    order, back pointers, shared branch vectors) from the checkpoint's
+    node image — resumed SPR searches must traverse the very rings the
+    reference serialized (writeTree/readTree, searchAlgo.c:1311)."""
+    from .spr import RingNode, SprTree
+    recs = ck._recs
+    nb = ck._nb
+    x = len(recs)
+    st = SprTree.__new__(SprTree)
+    st.ntips = mxtips
+    st.nnodes = 2 * mxtips - 1
+    members = [RingNode(recs[k][3]) for k in range(x)]
+    # image layout (writeTree): tips 0..mxtips-1; inner node i's block
+    # b = mxtips + 3*(i - mxtips - 1) holds [b]=p->next->next, [b+1]=
+    # p->next, [b+2]=p with ring b+2 -> b+1 -> b -> b+2
+    st.nodep = [None] * (2 * mxtips)
+    st.ring = [None] * (2 * mxtips)
+    for i in range(1, mxtips + 1):
+        st.nodep[i] = st.ring[i] = members[i - 1]
+    for i in range(mxtips + 1, 2 * mxtips - 1):
+        b = mxtips + 3 * (i - mxtips - 1)
+        members[b + 2].next = members[b + 1]
+        members[b + 1].next = members[b]
+        members[b].next = members[b + 2]
+        st.nodep[i] = st.ring[i] = members[b + 2]
+    first = {}  # per-edge shared branch vector under -M
+    ck_base = ck._base
+    nsz = NODE_SIZE
+
+    def idx(ptr):
+        return int((ptr - ck_base) // nsz) if ptr else None
+
+    for k, (z, nxt, bck, number) in enumerate(recs):
+        if not (1 <= number <= 2 * mxtips - 2):
+            continue
+        b = idx(bck)
+        if b is None:
+            continue
+        m = members[k]
+        mb = members[b]
+        m.back = mb
+        key = (min(k, b), max(k, b))
+        if nb > 1:
+            if key not in first:
+                first[key] = np.array(z[:nb])
+            m.z = first[key]
+        else:
+            m.z = float(z[0])
+    st.start = ck.start_number
+    return st
 
 
 def write_checkpoint(path, tree, models, mxtips, *, state=MOD_OPT,

@@ -26,6 +26,9 @@ from . import INNER_INNER, TIP_INNER, TIP_TIP, TravEntry
 from .search import SMOOTHINGS, TreeSearch
 from .tree import DEFAULTZ
 
+# checkpoint states for -R resume (axml.h:655-657)
+CKP_REARR_SETTING, CKP_FAST_SPRS, CKP_SLOW_SPRS = 1, 2, 3
+
 UNLIKELY = -1.0e300
 ZMIN, ZMAX = 1.0e-15, 1.0 - 1.0e-6
 ITERATIONS = 10  # axml.h:90, NR iterations per insertion's makenewz
@@ -999,8 +1002,14 @@ class SprSearch:
         self.do_cutoff = cutoff
         return best_trav
 
-    def compute_big_rapid(self, estimate_model=True, initial_trav=None):
-        """computeBIGRAPID (:1914), no convergence-criterion hashing."""
+    def compute_big_rapid(self, estimate_model=True, initial_trav=None,
+                          checkpoint=None):
+        """computeBIGRAPID (:1914).  checkpoint: a parsed reference -R
+        checkpoint (state FAST_SPRS/SLOW_SPRS) whose tree and model
+        state the caller has already restored into this search — the
+        loops resume at the reference's START_FAST_SPRS /
+        START_SLOW_SPRS labels (:2074/:2346) with the stored loop
+        variables."""
         ts, st = self.ts, self.st
         self.lh_avg = 0.0
         self.lh_dec = 0
@@ -1008,30 +1017,55 @@ class SprSearch:
         bt = BestList(20, st)
         epsilon = 0.01
         self.thorough = False
-        # main()'s preamble before computeBIGRAPID (axml.c:2760-2764)
-        ts.evaluate_generic(full=True)
-        ts.tree_evaluate(1.0)
-        if estimate_model:
-            ts.mod_opt(10.0)
+        ck = checkpoint
+        resume_fast = ck is not None and ck.state == CKP_FAST_SPRS
+        resume_slow = ck is not None and ck.state == CKP_SLOW_SPRS
+        if ck is not None:
+            assert resume_fast or resume_slow, ck.state
+            # restart(): tree+models restored by the caller; the
+            # preamble, radius search and modOpt(5) are all skipped
+            ts.evaluate_generic(full=True)
+            self.log(f"restart with likelihood: {ts.likelihood:.6f}")
+            best_trav = ck.best_trav
+            best_t.save(ts, True)
         else:
-            ts.tree_evaluate(2.0)
-        best_t.save(ts, True)
-        if initial_trav is None:
-            best_trav = self.determine_rearrangement_setting(best_t, bt)
-            self.log(f"best rearrangement radius: {best_trav}")
-        else:
-            best_trav = initial_trav
-        if estimate_model:
-            ts.mod_opt(5.0)
-        else:
+            # main()'s preamble before computeBIGRAPID (axml.c:2760-2764)
+            ts.evaluate_generic(full=True)
             ts.tree_evaluate(1.0)
-        best_t.save(ts, True)
+            if estimate_model:
+                ts.mod_opt(10.0)
+            else:
+                ts.tree_evaluate(2.0)
+            best_t.save(ts, True)
+            if initial_trav is None:
+                best_trav = self.determine_rearrangement_setting(best_t, bt)
+                self.log(f"best rearrangement radius: {best_trav}")
+            else:
+                best_trav = initial_trav
+            if estimate_model:
+                ts.mod_opt(5.0)
+            else:
+                ts.tree_evaluate(1.0)
+            best_t.save(ts, True)
         impr = True
         if self.do_cutoff:
             self.it_count = 0
         fast_iterations = 0
-        while impr:
-            best_t.recall(1, ts)
+        while impr and not resume_slow:
+            if resume_fast:
+                # START_FAST_SPRS restore (:2080-2105)
+                fast_iterations = ck.fast_iterations
+                best_trav = ck.best_trav
+                epsilon = ck.epsilon
+                ts.likelihood = ck.tr_likelihood
+                self.lh_cutoff = ck.tr_lh_cutoff
+                self.lh_avg = ck.tr_lh_avg
+                self.lh_dec = ck.tr_lh_dec
+                self.it_count = ck.tr_it_count
+                impr = bool(ck.impr)
+                resume_fast = False
+            else:
+                best_t.recall(1, ts)
             # -D check at the top of each fast cycle (searchAlgo.c:2160):
             # store the current best tree in slot fastIterations%2, then
             # compare against the previous cycle's tree.
@@ -1061,22 +1095,40 @@ class SprSearch:
                     impr = True
                     lh = ts.likelihood
                     best_t.save(ts, True)
-        if self.convergence_criterion:
-            # both exits empty the table (searchAlgo.c:2202/2303)
-            self.rfconv.clear()
-        self.thorough = True
-        impr = True
-        best_t.recall(1, ts)
-        ts.evaluate_generic(full=True)
-        if estimate_model:
-            ts.mod_opt(1.0)
-        else:
-            ts.tree_evaluate(1.0)
+        if not resume_slow:
+            if self.convergence_criterion:
+                # both exits empty the table (searchAlgo.c:2202/2303)
+                self.rfconv.clear()
+            self.thorough = True
+            impr = True
+            best_t.recall(1, ts)
+            ts.evaluate_generic(full=True)
+            if estimate_model:
+                ts.mod_opt(1.0)
+            else:
+                ts.tree_evaluate(1.0)
         rearrangements_min = 1
         rearrangements_max = self.stepwidth
         thorough_iterations = 0
         while True:
-            best_t.recall(1, ts)
+            if resume_slow:
+                # START_SLOW_SPRS restore (:2346-2379)
+                impr = bool(ck.impr)
+                self.thorough = bool(ck.thorough)
+                best_trav = ck.best_trav
+                rearrangements_max = ck.rearrangements_max
+                rearrangements_min = ck.rearrangements_min
+                thorough_iterations = ck.thorough_iterations
+                fast_iterations = ck.fast_iterations
+                epsilon = ck.epsilon
+                ts.likelihood = ck.tr_likelihood
+                self.lh_cutoff = ck.tr_lh_cutoff
+                self.lh_avg = ck.tr_lh_avg
+                self.lh_dec = ck.tr_lh_dec
+                self.it_count = ck.tr_it_count
+                resume_slow = False
+            else:
+                best_t.recall(1, ts)
             if impr:
                 rearrangements_min = 1
                 rearrangements_max = self.stepwidth

@@ -109,3 +109,55 @@ def test_reference_resumes_from_our_checkpoint(golden_dir, tmp_path):
     assert abs(restored - mid) < 1e-6 * abs(mid)
     fin = float(final[0].split(":")[1])
     assert abs(fin - GOLDEN_FINAL_LNL) < TOL_ABS, fin
+
+
+RESTART_FAST_LNL = -2744.20054491362134285736829042434692382812500
+RESTART_SLOW_LNL = -2741.47310239244689000770449638366699218750000
+SPR_FINAL = -2741.473102
+
+
+def _resume_spr(golden_dir, ckpt_name, restart_lnl):
+    from examl_amd.checkpoint import spr_tree
+    from examl_amd.spr import SprSearch
+    from tests.helpers import OracleEngine
+    taxa, parts = read_byte_file(os.path.join(golden_dir, "12.binary"))
+    ck = read_checkpoint(os.path.join(golden_dir, ckpt_name), 12, [4])
+    st = spr_tree(ck, 12)
+    engines = [OracleEngine(p.tips, p.wgt,
+                            ea.DnaGtrModel(m["frequencies"],
+                                           m["substRates"], m["alpha"]))
+               for p, m in zip(parts, ck.models)]
+    ts = TreeSearch(st, engines,
+                    opt_freq_flags=[bool(p.optimizeBaseFrequencies)
+                                    for p in parts])
+    assert ts.evaluate_generic(full=True) == restart_lnl  # bit-exact
+    sp = SprSearch(ts)
+    lnl = sp.compute_big_rapid(estimate_model=True, checkpoint=ck)
+    assert abs(lnl - SPR_FINAL) < abs(SPR_FINAL) * 1e-6, lnl
+
+
+def test_resume_spr_search_from_fast_checkpoint(golden_dir):
+    """-R restart mid-FAST_SPRS: the exact ring structure and loop
+    variables come from a reference-written checkpoint
+    (ExaML_binaryCheckpoint.R1_3 of examl-AVX on the 12-taxon golden);
+    the restored lnL is bit-exact vs the reference's own "ExaML Restart
+    with likelihood" line and the resumed search lands on the same
+    final tree score the reference's -R run prints."""
+    _resume_spr(golden_dir, "12.spr_fast.ckpt.bin", RESTART_FAST_LNL)
+
+
+def test_resume_spr_search_from_slow_checkpoint(golden_dir):
+    """-R restart mid-SLOW_SPRS (thorough loop), same contract."""
+    _resume_spr(golden_dir, "12.spr_slow.ckpt.bin", RESTART_SLOW_LNL)
+
+
+def test_spr_checkpoint_fields(golden_dir):
+    """The search-state fields of checkPointState (axml.h:679-720)
+    parse at the documented offsets."""
+    ck = read_checkpoint(os.path.join(golden_dir, "12.spr_slow.ckpt.bin"),
+                         12, [4])
+    assert ck.state == 3  # SLOW_SPRS
+    assert ck.fast_iterations == 2 and ck.thorough_iterations == 1
+    assert ck.best_trav == 5 and ck.thorough == 1 and ck.impr == 0
+    assert (ck.rearrangements_min, ck.rearrangements_max) == (1, 5)
+    assert ck.tr_it_count == 3
