@@ -260,10 +260,12 @@ def write_checkpoint(path, tree, models, mxtips, *, state=MOD_OPT,
                      rate_het="GAMMA", per_gene_bl=False,
                      likelihood_epsilon=0.1, rate_category=None,
                      patrat=None, likelihoods=None, start_number=1,
-                     accumulated_time=1.0):
+                     accumulated_time=1.0, spr=None):
     """Emit a checkpoint the reference's readCheckpoint accepts.  models:
     one dict per partition with the arrays of read_checkpoint's layout
-    (build_model_entry converts our model objects)."""
+    (build_model_entry converts our model objects).  spr: the
+    search-state fields for FAST_SPRS/SLOW_SPRS checkpoints
+    (checkPointState, axml.h:679-720)."""
     tsl = tree_string_length(mxtips)
     out = bytearray(CKP_SIZE)
     struct.pack_into("<i", out, 0, state)
@@ -271,10 +273,37 @@ def write_checkpoint(path, tree, models, mxtips, *, state=MOD_OPT,
     struct.pack_into("<d", out, 56, accumulated_time)
     struct.pack_into("<i", out, 180, cat_opt)
     struct.pack_into("<i", out, 184, tree_iteration)
+    if spr is not None:
+        struct.pack_into(
+            "<9i", out, 12, spr.get("rearrangements_max", 0),
+            spr.get("rearrangements_min", 0),
+            spr.get("thorough_iterations", 0),
+            spr.get("fast_iterations", 0),
+            spr.get("tree_vector_length", 1), spr.get("mintrav", 1),
+            spr.get("maxtrav", 5), spr.get("best_trav", 5),
+            spr.get("thorough", 0))
+        struct.pack_into(
+            "<5d", out, 64, spr.get("start_lh", 0.0), spr.get("lh", 0.0),
+            spr.get("previous_lh", 0.0), spr.get("difference", 10.0),
+            spr.get("epsilon", 0.01))
+        struct.pack_into("<2i", out, 104, int(spr.get("impr", 1)),
+                         int(spr.get("cutoff", 1)))
+        struct.pack_into(
+            "<7d", out, 112, spr.get("tr_start_lh", 0.0),
+            spr.get("tr_end_lh", 0.0), spr.get("tr_likelihood", 0.0),
+            spr.get("tr_best_of_node", 0.0),
+            spr.get("tr_lh_cutoff", 0.0), spr.get("tr_lh_avg", 0.0),
+            spr.get("tr_lh_dec", 0.0))
+        struct.pack_into("<3i", out, 168, MAX_CATEGORIES,
+                         spr.get("tr_it_count", 0),
+                         int(spr.get("tr_do_cutoff", 1)))
     c = 1248
     struct.pack_into("<d", out, c + 24, likelihood_epsilon)
     struct.pack_into("<i", out, c + 32, MAX_CATEGORIES)
-    struct.pack_into("<i", out, c + 36, 0)   # mode TREE_EVALUATION
+    # adef->mode: BIG_RAPID_MODE for SPR-state checkpoints (axml.h:226)
+    struct.pack_into("<i", out, c + 36,
+                     1 if state in (REARR_SETTING, FAST_SPRS, SLOW_SPRS)
+                     else 0)
     struct.pack_into("<i", out, c + 44, 0)   # initialSet
     struct.pack_into("<i", out, c + 48, 10)  # adef->initial default
     struct.pack_into("<i", out, c + 52,
@@ -361,16 +390,11 @@ def write_checkpoint(path, tree, models, mxtips, *, state=MOD_OPT,
         used[i] += 1
         return s
 
-    seen = set()
-    for a in tree.adj:
-        for bn in tree.adj[a]:
-            if (bn, a) in seen:
-                continue
-            seen.add((a, bn))
-            ra, rb = take_slot(a), take_slot(bn)
-            rec_back[ra] = rb
-            rec_back[rb] = ra
-            rec_z[ra] = rec_z[rb] = zvec(a, bn)
+    for a, bn in tree.edges():
+        ra, rb = take_slot(a), take_slot(bn)
+        rec_back[ra] = rb
+        rec_back[rb] = ra
+        rec_z[ra] = rec_z[rb] = zvec(a, bn)
 
     buf += struct.pack("<i", start_number)
     buf += struct.pack("<Q", base)

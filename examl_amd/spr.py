@@ -506,6 +506,9 @@ class SprSearch:
         self.save_best_trees = save_best_trees
         self.best_ml = (BestList(save_best_trees, self.st)
                         if save_best_trees > 0 else None)
+        # optional writeCheckpoint hook: called as writer(state, fields)
+        # at the reference's write points (searchAlgo.c:2155/:2425)
+        self.checkpoint_writer = None
         self.good_trees = []  # good-tree lnls after compute_big_rapid
         # tr-> search state
         self.start_lh = 0.0
@@ -1051,6 +1054,7 @@ class SprSearch:
         if self.do_cutoff:
             self.it_count = 0
         fast_iterations = 0
+        lh = previous_lh = UNLIKELY
         while impr and not resume_slow:
             if resume_fast:
                 # START_FAST_SPRS restore (:2080-2105)
@@ -1066,6 +1070,15 @@ class SprSearch:
                 resume_fast = False
             else:
                 best_t.recall(1, ts)
+            if self.checkpoint_writer is not None:
+                self.checkpoint_writer(CKP_FAST_SPRS, dict(
+                    fast_iterations=fast_iterations, best_trav=best_trav,
+                    thorough=0, impr=int(impr), epsilon=epsilon,
+                    tr_likelihood=ts.likelihood,
+                    tr_lh_cutoff=self.lh_cutoff, tr_lh_avg=self.lh_avg,
+                    tr_lh_dec=float(self.lh_dec),
+                    tr_it_count=self.it_count,
+                    tr_do_cutoff=int(self.do_cutoff)))
             # -D check at the top of each fast cycle (searchAlgo.c:2160):
             # store the current best tree in slot fastIterations%2, then
             # compare against the previous cycle's tree.
@@ -1129,6 +1142,19 @@ class SprSearch:
                 resume_slow = False
             else:
                 best_t.recall(1, ts)
+            if self.checkpoint_writer is not None:
+                self.checkpoint_writer(CKP_SLOW_SPRS, dict(
+                    fast_iterations=fast_iterations,
+                    thorough_iterations=thorough_iterations,
+                    best_trav=best_trav, thorough=1, impr=int(impr),
+                    rearrangements_min=rearrangements_min,
+                    rearrangements_max=rearrangements_max,
+                    lh=lh, previous_lh=previous_lh, epsilon=epsilon,
+                    tr_likelihood=ts.likelihood,
+                    tr_lh_cutoff=self.lh_cutoff, tr_lh_avg=self.lh_avg,
+                    tr_lh_dec=float(self.lh_dec),
+                    tr_it_count=self.it_count,
+                    tr_do_cutoff=int(self.do_cutoff)))
             if impr:
                 rearrangements_min = 1
                 rearrangements_max = self.stepwidth

@@ -161,3 +161,61 @@ def test_spr_checkpoint_fields(golden_dir):
     assert ck.best_trav == 5 and ck.thorough == 1 and ck.impr == 0
     assert (ck.rearrangements_min, ck.rearrangements_max) == (1, 5)
     assert ck.tr_it_count == 3
+
+
+@pytest.mark.skipif(not os.path.exists(_REF_BIN),
+                    reason="reference examl-AVX not built")
+def test_reference_resumes_from_our_spr_checkpoint(golden_dir, tmp_path):
+    """Write interchange for SPR-state checkpoints: OUR search writes
+    FAST_SPRS/SLOW_SPRS checkpoints at the reference's write points and
+    the unmodified reference binary -R restarts from them — its printed
+    restart lnL is bit-identical to the one it prints when restarting
+    from its OWN checkpoint of the same state, and it completes to the
+    same final score."""
+    from examl_amd.checkpoint import build_model_entry, write_checkpoint
+    from examl_amd.spr import SprSearch, SprTree
+    from tests.helpers import OracleEngine
+    taxa, parts = read_byte_file(os.path.join(golden_dir, "12.binary"))
+    t = read_newick_topology(os.path.join(golden_dir, "12.tree"), taxa)
+    st = SprTree.from_phylo(t)
+    engines = [OracleEngine(p.tips, p.wgt,
+                            ea.DnaGtrModel(p.frequencies, [1.0] * 6, 1.0))
+               for p in parts]
+    ts = TreeSearch(st, engines,
+                    opt_freq_flags=[bool(p.optimizeBaseFrequencies)
+                                    for p in parts])
+    sp = SprSearch(ts)
+    written = []
+
+    def writer(state, fields):
+        path = str(tmp_path / f"our_{len(written)}")
+        write_checkpoint(path, st,
+                         [build_model_entry(e.model) for e in ts.engines],
+                         12, state=state, spr=fields,
+                         start_number=st.start)
+        written.append((path, state, fields["fast_iterations"],
+                        fields.get("thorough_iterations", 0)))
+
+    sp.checkpoint_writer = writer
+    sp.compute_big_rapid(estimate_model=True)
+    fast = next(p for p, s, fi, ti in written if s == 2 and fi == 1)
+    slow = next(p for p, s, fi, ti in written if s == 3 and ti == 1)
+    shutil.copy(os.path.join(golden_dir, "12.binary"),
+                tmp_path / "12.binary")
+    for path, expect_restart in (
+            (fast, "-2744.20054491362134285736829042434692382812500"),
+            (slow, "-2741.47310239244689000770449638366699218750000")):
+        name = "RES" + os.path.basename(path)
+        subprocess.run(
+            [_REF_BIN, "-s", "12.binary", "-R", path, "-m", "GAMMA",
+             "-n", name], cwd=tmp_path, check=True,
+            stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL,
+            timeout=300)
+        info = open(tmp_path / f"ExaML_info.{name}").read()
+        restart = [ln for ln in info.splitlines()
+                   if "Restart with likelihood" in ln][0]
+        assert expect_restart in restart, restart
+        final = float([ln for ln in info.splitlines()
+                       if ln.startswith("Likelihood of best tree")]
+                      [0].split(":")[1])
+        assert abs(final - (-2741.473102)) < abs(2741.473102) * 1e-6
