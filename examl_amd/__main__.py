@@ -290,6 +290,25 @@ def main(argv=None, device=None):
                        convergence_criterion=opts["D"],
                        save_best_trees=opts["B"], log=log)
         sp.constraint = cv
+        if (all(p.states == 4 for p in parts) and opts["m"] == "GAMMA"
+                and not opts["M"] and cv is None):
+            # resumable like the reference: one binary checkpoint per
+            # SPR cycle (writeCheckpointInner, searchAlgo.c:1153)
+            from examl_amd.checkpoint import (build_model_entry,
+                                              write_checkpoint)
+            counter = [0]
+
+            def _writer(state, fields):
+                path = os.path.join(
+                    wdir, f"ExaML_binaryCheckpoint.{name}_{counter[0]}")
+                write_checkpoint(
+                    path, st,
+                    [build_model_entry(e.model) for e in ts.engines],
+                    len(taxa), state=state, spr=fields,
+                    start_number=st.start)
+                counter[0] += 1
+
+            sp.checkpoint_writer = _writer
         final = sp.compute_big_rapid(estimate_model=True,
                                      initial_trav=opts["i"])
         log(f"Likelihood of best tree: {final:.6f}")
