@@ -963,20 +963,43 @@ class SprSearch:
             self.thorough = False
         return self.start_lh
 
-    def determine_rearrangement_setting(self, best_t, bt):
-        """determineRearrangementSetting (:1752)."""
+    def determine_rearrangement_setting(self, best_t, bt,
+                                        checkpoint=None):
+        """determineRearrangementSetting (:1752); checkpoint resumes a
+        REARR_SETTING-state -R restart (:1769-1781)."""
         ts = self.ts
         MAX_FAST = 26
         maxtrav, best_trav = 5, 5
         start_lh = ts.likelihood
         impr = True
         cutoff = self.do_cutoff
+        if checkpoint is not None:
+            maxtrav = checkpoint.maxtrav
+            best_trav = checkpoint.best_trav
+            start_lh = checkpoint.start_lh
+            impr = bool(checkpoint.impr)
+            cutoff = bool(checkpoint.cutoff)
+            # readCheckpoint's tr-field restore (:1525-1533)
+            ts.likelihood = checkpoint.tr_likelihood
+            self.lh_cutoff = checkpoint.tr_lh_cutoff
+            self.lh_avg = checkpoint.tr_lh_avg
+            self.lh_dec = checkpoint.tr_lh_dec
+            self.it_count = checkpoint.tr_it_count
         self.do_cutoff = False
         bt.reset()
         assert not self.thorough
         while impr and maxtrav < MAX_FAST:
             best_t.recall(1, ts)
             self.node_rectifier()
+            if self.checkpoint_writer is not None:
+                self.checkpoint_writer(CKP_REARR_SETTING, dict(
+                    maxtrav=maxtrav, best_trav=best_trav,
+                    start_lh=start_lh, impr=int(impr),
+                    cutoff=int(cutoff), tr_likelihood=ts.likelihood,
+                    tr_lh_cutoff=self.lh_cutoff, tr_lh_avg=self.lh_avg,
+                    tr_lh_dec=float(self.lh_dec),
+                    tr_it_count=self.it_count,
+                    tr_do_cutoff=int(self.do_cutoff)))
             maxtrav_eff = min(maxtrav, self.st.ntips - 3)
             self.start_lh = self.end_lh = ts.likelihood
             for i in range(1, 2 * self.st.ntips - 1):
@@ -987,6 +1010,8 @@ class SprSearch:
                         self.start_lh = self.end_lh = ts.likelihood
             ts.tree_evaluate(0.25)
             bt.save(ts, True)
+            if self.best_ml is not None:
+                self.best_ml.save(ts, False)
             if ts.likelihood > start_lh:
                 start_lh = ts.likelihood
                 best_trav = maxtrav
@@ -1023,14 +1048,26 @@ class SprSearch:
         ck = checkpoint
         resume_fast = ck is not None and ck.state == CKP_FAST_SPRS
         resume_slow = ck is not None and ck.state == CKP_SLOW_SPRS
+        resume_rearr = ck is not None and ck.state == CKP_REARR_SETTING
         if ck is not None:
-            assert resume_fast or resume_slow, ck.state
+            assert resume_fast or resume_slow or resume_rearr, ck.state
             # restart(): tree+models restored by the caller; the
-            # preamble, radius search and modOpt(5) are all skipped
+            # preamble is skipped (modOpt(10) ran before the earliest
+            # REARR_SETTING checkpoint was written)
             ts.evaluate_generic(full=True)
             self.log(f"restart with likelihood: {ts.likelihood:.6f}")
-            best_trav = ck.best_trav
             best_t.save(ts, True)
+            if resume_rearr:
+                best_trav = self.determine_rearrangement_setting(
+                    best_t, bt, checkpoint=ck)
+                self.log(f"best rearrangement radius: {best_trav}")
+                if estimate_model:
+                    ts.mod_opt(5.0)
+                else:
+                    ts.tree_evaluate(1.0)
+                best_t.save(ts, True)
+            else:
+                best_trav = ck.best_trav
         else:
             # main()'s preamble before computeBIGRAPID (axml.c:2760-2764)
             ts.evaluate_generic(full=True)

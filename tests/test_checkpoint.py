@@ -193,7 +193,7 @@ def test_reference_resumes_from_our_spr_checkpoint(golden_dir, tmp_path):
                          [build_model_entry(e.model) for e in ts.engines],
                          12, state=state, spr=fields,
                          start_number=st.start)
-        written.append((path, state, fields["fast_iterations"],
+        written.append((path, state, fields.get("fast_iterations", 0),
                         fields.get("thorough_iterations", 0)))
 
     sp.checkpoint_writer = writer
@@ -219,3 +219,32 @@ def test_reference_resumes_from_our_spr_checkpoint(golden_dir, tmp_path):
                        if ln.startswith("Likelihood of best tree")]
                       [0].split(":")[1])
         assert abs(final - (-2741.473102)) < abs(2741.473102) * 1e-6
+
+
+def test_resume_spr_search_from_rearr_checkpoint(golden_dir):
+    """-R restart mid-REARR_SETTING (radius search): the resumed run
+    replays the reference's own -R run from the same checkpoint — same
+    bit-exact restored lnL (-3651.20215442176...) and the same final
+    -2741.473101 the reference's restart prints (one trajectory digit
+    off the uninterrupted run's -2741.473102, matching the reference's
+    restart exactly)."""
+    from examl_amd.checkpoint import spr_tree
+    from examl_amd.spr import SprSearch
+    from tests.helpers import OracleEngine
+    taxa, parts = read_byte_file(os.path.join(golden_dir, "12.binary"))
+    ck = read_checkpoint(os.path.join(golden_dir,
+                                      "12.spr_rearr.ckpt.bin"), 12, [4])
+    assert ck.state == 1 and ck.maxtrav == 10
+    st = spr_tree(ck, 12)
+    engines = [OracleEngine(p.tips, p.wgt,
+                            ea.DnaGtrModel(m["frequencies"],
+                                           m["substRates"], m["alpha"]))
+               for p, m in zip(parts, ck.models)]
+    ts = TreeSearch(st, engines,
+                    opt_freq_flags=[bool(p.optimizeBaseFrequencies)
+                                    for p in parts])
+    assert ts.evaluate_generic(full=True) == \
+        -3651.20215442176322540035471320152282714843750
+    sp = SprSearch(ts)
+    lnl = sp.compute_big_rapid(estimate_model=True, checkpoint=ck)
+    assert abs(lnl - (-2741.473101)) < abs(2741.473101) * 1e-6, lnl
