@@ -390,11 +390,42 @@ def write_checkpoint(path, tree, models, mxtips, *, state=MOD_OPT,
         used[i] += 1
         return s
 
-    for a, bn in tree.edges():
-        ra, rb = take_slot(a), take_slot(bn)
-        rec_back[ra] = rb
-        rec_back[rb] = ra
-        rec_z[ra] = rec_z[rb] = zvec(a, bn)
+    if hasattr(tree, "ring"):
+        # SprTree: serialize the LIVE rings so a resumed reference run
+        # traverses exactly the member structure this search had —
+        # ring[i] maps to record b+2, its .next to b+1, .next.next to b
+        # (the image's fixed next-pointer pattern above)
+        memrec = {}
+        for i in range(1, mxtips + 1):
+            memrec[id(tree.ring[i])] = i - 1
+        for i in range(mxtips + 1, 2 * mxtips - 1):
+            b = mxtips + 3 * (i - mxtips - 1)
+            m3 = tree.ring[i]
+            memrec[id(m3)] = b + 2
+            memrec[id(m3.next)] = b + 1
+            memrec[id(m3.next.next)] = b
+
+        def mzvec(mz):
+            z = np.full(256, 0.9)
+            if isinstance(mz, np.ndarray):
+                z[:nb] = mz[:nb]
+            else:
+                z[0] = float(mz)
+            return z
+
+        for i in range(1, 2 * mxtips - 1):
+            for m in tree.members(i):
+                if m.back is None:
+                    continue
+                r = memrec[id(m)]
+                rec_back[r] = memrec[id(m.back)]
+                rec_z[r] = mzvec(m.z)
+    else:
+        for a, bn in tree.edges():
+            ra, rb = take_slot(a), take_slot(bn)
+            rec_back[ra] = rb
+            rec_back[rb] = ra
+            rec_z[ra] = rec_z[rb] = zvec(a, bn)
 
     buf += struct.pack("<i", start_number)
     buf += struct.pack("<Q", base)
