@@ -194,7 +194,8 @@ def main(argv=None, device=None):
         # image, engines rebuilt from the stored model arrays (our
         # eigendecomposition reproduces the stored one bit-for-bit)
         import examl_amd as ea
-        from examl_amd.checkpoint import (FAST_SPRS, MOD_OPT, SLOW_SPRS,
+        from examl_amd.checkpoint import (FAST_SPRS, MOD_OPT,
+                                          REARR_SETTING, SLOW_SPRS,
                                           read_checkpoint, spr_tree)
         from examl_amd.search import TreeSearch
         from examl_amd.spr import SprSearch
@@ -213,7 +214,8 @@ def main(argv=None, device=None):
                   max_categories=opts["c"])
         res_path = os.path.join(wdir, f"ExaML_result.{name}")
         if opts["f"] in ("d", "o"):
-            assert ckpt.state in (FAST_SPRS, SLOW_SPRS), \
+            assert ckpt.state in (REARR_SETTING, FAST_SPRS,
+                                  SLOW_SPRS), \
                 "checkpoint state does not match -f d"
             st = spr_tree(ckpt, len(taxa))
             ts = TreeSearch(st, engines, **kw)
