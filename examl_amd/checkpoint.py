@@ -54,6 +54,7 @@ REARR_SETTING = 1  # ckp.state (axml.h:655-659)
 FAST_SPRS = 2
 SLOW_SPRS = 3
 MOD_OPT = 4
+QUARTETS = 5
 
 # rateHetModel values as stored by the reference (axml.h): CAT=0, GAMMA=1
 RATE_HET_CAT = 0
@@ -96,6 +97,12 @@ def read_checkpoint(path, mxtips, states_per_model, rate_het="GAMMA",
      ck.tr_lh_dec) = struct.unpack_from("<7d", d, 112)
     (ck.tr_number_of_categories, ck.tr_it_count,
      ck.tr_do_cutoff) = struct.unpack_from("<3i", d, 168)
+    # quartet state (-f q -I): seed at 192 (8-aligned after 184+4 pad),
+    # flavor 200, quartetCounter 208, filePosition 216, fileName 224
+    ck.seed = struct.unpack_from("<q", d, 192)[0]
+    ck.flavor = struct.unpack_from("<i", d, 200)[0]
+    ck.quartet_counter = struct.unpack_from("<Q", d, 208)[0]
+    ck.file_position = struct.unpack_from("<q", d, 216)[0]
     constraint = struct.unpack_from("<i", d, 8)[0]
     c = 1248
     ck.likelihood_epsilon = struct.unpack_from("<d", d, c + 24)[0]

@@ -125,33 +125,40 @@ def _all_three(engines, mxtips, t1, t2, t3, t4, out, per_gene_bl=False):
 
 
 def compute_quartets(engines, mxtips, random_quartets=0, seed=0,
-                     groups=None, per_gene_bl=False):
-    """computeQuartets (quartets.c:349) minus the file/checkpoint plumbing:
+                     groups=None, per_gene_bl=False, start_counter=0):
+    """computeQuartets (quartets.c:349) minus the file plumbing:
     returns [(a, b, c, d, lnL)] in the reference's emission order.
 
     random_quartets == 0 -> ALL_QUARTETS; groups -> GROUPED_QUARTETS
     (four disjoint 1-based taxon lists); otherwise RANDOM_QUARTETS with
-    the reference PRNG and sub-sampling fraction."""
+    the reference PRNG and sub-sampling fraction.  start_counter: a -R
+    resume replays the deterministic enumeration (and, for RANDOM, the
+    PRNG stream from the checkpoint's stored INITIAL seed) but skips
+    the first start_counter evaluations (quartets.c:520/:560/:598)."""
     out = []
     n_quartets = mxtips * (mxtips - 1) * (mxtips - 2) * (mxtips - 3) // 24
+    count = 0
     if groups is not None:
         for i1 in groups[0]:
             for i2 in groups[1]:
                 for i3 in groups[2]:
                     for i4 in groups[3]:
-                        _all_three(engines, mxtips, i1, i2, i3, i4, out,
-                                   per_gene_bl)
+                        if count >= start_counter:
+                            _all_three(engines, mxtips, i1, i2, i3, i4,
+                                       out, per_gene_bl)
+                        count += 1
         return out
     if random_quartets == 0 or random_quartets >= n_quartets:
         for t1 in range(1, mxtips + 1):
             for t2 in range(t1 + 1, mxtips + 1):
                 for t3 in range(t2 + 1, mxtips + 1):
                     for t4 in range(t3 + 1, mxtips + 1):
-                        _all_three(engines, mxtips, t1, t2, t3, t4, out,
-                                   per_gene_bl)
+                        if count >= start_counter:
+                            _all_three(engines, mxtips, t1, t2, t3, t4,
+                                       out, per_gene_bl)
+                        count += 1
         return out
     fraction = random_quartets / n_quartets
-    count = 0
     while True:
         for t1 in range(1, mxtips + 1):
             for t2 in range(t1 + 1, mxtips + 1):
@@ -159,8 +166,9 @@ def compute_quartets(engines, mxtips, random_quartets=0, seed=0,
                     for t4 in range(t3 + 1, mxtips + 1):
                         r, seed = randum(seed)
                         if r < fraction:
-                            _all_three(engines, mxtips, t1, t2, t3, t4, out,
-                                       per_gene_bl)
+                            if count >= start_counter:
+                                _all_three(engines, mxtips, t1, t2, t3,
+                                           t4, out, per_gene_bl)
                             count += 1
                         if count == random_quartets:
                             return out

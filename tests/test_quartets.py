@@ -108,3 +108,32 @@ def test_grouped_quartets_match_reference_12(golden_dir):
     for o, g in zip(out, golden):
         assert (o[0], o[1], o[2], o[3]) == (g[0], g[1], g[2], g[3])
         assert abs(o[4] - g[4]) < 1e-5, (o, g)
+
+
+def test_quartet_resume_from_reference_checkpoint(golden_dir):
+    """-f q -I / -R: resume RANDOM_QUARTETS from a reference-written
+    QUARTETS-state checkpoint (counter 20 of 30, seed 123).  The PRNG
+    stream replays from the stored initial seed with the first
+    quartetCounter evaluations skipped (quartets.c:560), and the
+    resumed output matches the reference's own resumed file tail line
+    for line."""
+    import examl_amd as ea
+    from examl_amd.checkpoint import read_checkpoint
+    from examl_amd.examl_io import read_byte_file
+    from tests.helpers import OracleEngine
+    taxa, parts = read_byte_file(os.path.join(golden_dir, "12.binary"))
+    ck = read_checkpoint(os.path.join(golden_dir, "12.quartets.ckpt.bin"),
+                         12, [4])
+    assert ck.state == 5 and ck.quartet_counter == 20 and ck.seed == 123
+    engines = [OracleEngine(p.tips, p.wgt,
+                            ea.DnaGtrModel(m["frequencies"],
+                                           m["substRates"], m["alpha"]))
+               for p, m in zip(parts, ck.models)]
+    out = compute_quartets(engines, 12, random_quartets=30, seed=ck.seed,
+                           start_counter=ck.quartet_counter)
+    golden = _parse_golden(os.path.join(golden_dir,
+                                        "12.quartets.random.txt"))
+    assert len(golden) == 90 and len(out) == 30
+    for o, g in zip(out, golden[60:]):
+        assert (o[0], o[1], o[2], o[3]) == (g[0], g[1], g[2], g[3])
+        assert abs(o[4] - g[4]) < 1e-5, (o, g)
