@@ -74,10 +74,14 @@ class Checkpoint:
 
 
 def read_checkpoint(path, mxtips, states_per_model, rate_het="GAMMA",
-                    crunched_length=None, num_trees=1):
+                    crunched_length=None, num_trees=1, prot_models=None):
     """Parse a reference checkpoint into plain arrays.  states_per_model:
     4 or 20 per partition; crunched_length: total pattern count (CAT
-    only)."""
+    only); prot_models: the byte file's per-partition protModels ids —
+    needed to know which partitions carry the LG4 per-category array
+    block (writeCheckpointInner stores it between weightExponents and
+    alpha, searchAlgo.c:1248-1262, and only the caller knows which
+    partitions are LG4M/LG4X)."""
     d = open(path, "rb").read()
     ck = Checkpoint()
     ck.state = struct.unpack_from("<i", d, 0)[0]
@@ -126,7 +130,7 @@ def read_checkpoint(path, mxtips, states_per_model, rate_het="GAMMA",
         off += 8 * crunched_length
 
     ck.models = []
-    for states in states_per_model:
+    for mi, states in enumerate(states_per_model):
         eign, ev, ei, freq, tipvec, subst = _PLEN[states]
         m = {}
 
@@ -151,11 +155,30 @@ def read_checkpoint(path, mxtips, states_per_model, rate_het="GAMMA",
         m["substRates"] = rd(subst)
         m["weights"] = rd(4)
         m["weightExponents"] = rd(4)
+        if prot_models is not None and prot_models[mi] in (20, 21):
+            # LG4M/LG4X: four per-category eigensystems
+            # (writeCheckpointInner, searchAlgo.c:1248-1262)
+            for key, n in (("rawEIGN_LG4", eign), ("EIGN_LG4", eign),
+                           ("EV_LG4", ev), ("EI_LG4", ei),
+                           ("frequencies_LG4", freq),
+                           ("tipVector_LG4", tipvec),
+                           ("substRates_LG4", subst)):
+                m[key] = []
+            for _k in range(4):
+                m["rawEIGN_LG4"].append(rd(eign))
+                m["EIGN_LG4"].append(rd(eign))
+                m["EV_LG4"].append(rd(ev))
+                m["EI_LG4"].append(rd(ei))
+                m["frequencies_LG4"].append(rd(freq))
+                m["tipVector_LG4"].append(rd(tipvec))
+                m["substRates_LG4"].append(rd(subst))
         m["alpha"] = float(rd(1)[0])
         m["gammaRates"] = rd(4)
         m["protModels"] = rd(1, "i")
         m["autoProtModels"] = rd(1, "i")
-        assert m["protModels"] not in (20, 21), "LG4 checkpoints unsupported"
+        if m["protModels"] in (20, 21):
+            assert prot_models is not None, \
+                "pass prot_models to read LG4 checkpoints"
         ck.models.append(m)
 
     if ck.state == MOD_OPT:

@@ -257,3 +257,40 @@ def test_resume_spr_search_from_rearr_checkpoint(golden_dir):
     sp = SprSearch(ts)
     lnl = sp.compute_big_rapid(estimate_model=True, checkpoint=ck)
     assert abs(lnl - (-2741.473101)) < abs(2741.473101) * 1e-6, lnl
+
+
+def test_read_lg4_checkpoint_and_resume(golden_dir):
+    """LG4 checkpoints (per-category eigensystem block between
+    weightExponents and alpha, writeCheckpointInner
+    searchAlgo.c:1248-1262): restore an LG4X+LG4M -f E checkpoint from
+    the reference, land bit-exactly on its "ExaML Restart with
+    likelihood" value, and resume modOpt to the -f E golden."""
+    import numpy as np
+
+    from tests.helpers import OracleLg4Engine
+    taxa, parts = read_byte_file(os.path.join(golden_dir,
+                                              "12lg4.binary"))
+    pm = [p.protModels for p in parts]
+    ck = read_checkpoint(os.path.join(golden_dir, "12lg4.ckpt.bin"), 12,
+                         [20, 20], prot_models=pm)
+    engines = []
+    for p, m in zip(parts, ck.models):
+        mdl = ea.Lg4Model.lg4x() if p.protModels == 21 \
+            else ea.Lg4Model.lg4m()
+        mdl.EIGN4 = np.concatenate(m["EIGN_LG4"])
+        mdl.EIGN4_raw = np.concatenate(m["rawEIGN_LG4"])
+        mdl.EV4 = np.concatenate(m["EV_LG4"])
+        mdl.EI4 = np.concatenate(m["EI_LG4"])
+        mdl.tipVector4 = np.concatenate(m["tipVector_LG4"])
+        mdl.gammaRates = np.asarray(m["gammaRates"])
+        mdl.weights = np.asarray(m["weights"])
+        mdl.weightExponents = np.asarray(m["weightExponents"])
+        mdl.alpha = m["alpha"]
+        engines.append(OracleLg4Engine(p.tips, p.wgt, mdl))
+    ts = TreeSearch(ck.tree, engines,
+                    opt_freq_flags=[bool(p.optimizeBaseFrequencies)
+                                    for p in parts])
+    lnl = ts.evaluate_generic(full=True)
+    assert lnl == -7387.55350209685184381669387221336364746093750
+    fin = ts.mod_opt(0.1)
+    assert abs(fin - (-7387.472983)) < abs(7387.472983) * 1e-6, fin
