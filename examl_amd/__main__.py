@@ -154,8 +154,8 @@ def main(argv=None, device=None):
         _usage()
         sys.exit("specify a starting tree: -t treeFile, "
                  "-g constraintTree -p seed, or -R checkpointFile")
-    if opts["R"] and (opts["M"] or opts["m"] == "PSR"):
-        sys.exit("-R restart is currently wired for GAMMA without -M")
+    if opts["R"] and opts["m"] == "PSR":
+        sys.exit("-R restart is currently wired for GAMMA")
     if opts["g"] and opts["p"] is None:
         sys.exit("you must specify a random number seed via -p when "
                  "using a constraint tree")
@@ -212,6 +212,8 @@ def main(argv=None, device=None):
         kw = dict(opt_freq_flags=[bool(p.optimizeBaseFrequencies)
                                   for p in parts],
                   max_categories=opts["c"])
+        if opts["M"]:
+            kw["per_gene_bl"] = True
         res_path = os.path.join(wdir, f"ExaML_result.{name}")
         if opts["f"] in ("d", "o"):
             assert ckpt.state in (REARR_SETTING, FAST_SPRS,

@@ -294,3 +294,32 @@ def test_read_lg4_checkpoint_and_resume(golden_dir):
     assert lnl == -7387.55350209685184381669387221336364746093750
     fin = ts.mod_opt(0.1)
     assert abs(fin - (-7387.472983)) < abs(7387.472983) * 1e-6, fin
+
+
+def test_resume_spr_search_M_from_checkpoint(golden_dir):
+    """-R restart under -M: per-partition branch vectors come out of
+    the node image (z[0..nb-1] per edge); restored lnL is bit-exact vs
+    the reference's restart line and the resumed search lands on the
+    -M -f d golden."""
+    from examl_amd.checkpoint import spr_tree
+    from examl_amd.spr import SprSearch
+    from tests.helpers import OracleEngine
+    taxa, parts = read_byte_file(os.path.join(golden_dir, "12m.binary"))
+    ck = read_checkpoint(os.path.join(golden_dir,
+                                      "12m.spr_fast.ckpt.bin"), 12,
+                         [4, 4])
+    assert ck.per_gene_bl and ck.state == 2
+    st = spr_tree(ck, 12)
+    engines = [OracleEngine(p.tips, p.wgt,
+                            ea.DnaGtrModel(m["frequencies"],
+                                           m["substRates"], m["alpha"]))
+               for p, m in zip(parts, ck.models)]
+    ts = TreeSearch(st, engines,
+                    opt_freq_flags=[bool(p.optimizeBaseFrequencies)
+                                    for p in parts],
+                    per_gene_bl=True)
+    assert ts.evaluate_generic(full=True) == \
+        -2730.70260654699541191803291440010070800781250
+    sp = SprSearch(ts)
+    lnl = sp.compute_big_rapid(estimate_model=True, checkpoint=ck)
+    assert abs(lnl - (-2728.477352)) < abs(2728.477352) * 1e-6, lnl
