@@ -154,8 +154,7 @@ def main(argv=None, device=None):
         _usage()
         sys.exit("specify a starting tree: -t treeFile, "
                  "-g constraintTree -p seed, or -R checkpointFile")
-    if opts["R"] and opts["m"] == "PSR":
-        sys.exit("-R restart is currently wired for GAMMA")
+
     if opts["g"] and opts["p"] is None:
         sys.exit("you must specify a random number seed via -p when "
                  "using a constraint tree")
@@ -200,20 +199,35 @@ def main(argv=None, device=None):
         from examl_amd.search import TreeSearch
         from examl_amd.spr import SprSearch
         from examl_amd.examl_io import to_newick
+        psr = opts["m"] == "PSR"
+        cl = sum(p.upper - p.lower for p in parts)
         ckpt = read_checkpoint(opts["R"], len(taxa),
-                               [p.states for p in parts])
+                               [p.states for p in parts],
+                               rate_het="CAT" if psr else "GAMMA",
+                               crunched_length=cl if psr else None)
         engines = []
         for p, m in zip(parts, ckpt.models):
-            assert p.states == 4, "-R restart wired for DNA GAMMA"
+            assert p.states == 4, "-R restart wired for DNA"
             model = ea.DnaGtrModel(m["frequencies"], m["substRates"],
                                    m["alpha"], use_median=opts["a"])
-            engines.append(ea.DnaGammaEngine(p.tips, p.wgt, model,
-                                             device=device))
+            if psr:
+                cptr = np.asarray(ckpt.rate_category[p.lower:p.upper],
+                                  dtype=np.int32).copy()
+                rates = np.asarray(
+                    m["per_site_rates"][:m["num_cats"]]).copy()
+                engines.append(ea.DnaCatEngine(p.tips, p.wgt, model,
+                                               cptr, rates,
+                                               device=device))
+            else:
+                engines.append(ea.DnaGammaEngine(p.tips, p.wgt, model,
+                                                 device=device))
         kw = dict(opt_freq_flags=[bool(p.optimizeBaseFrequencies)
                                   for p in parts],
                   max_categories=opts["c"])
         if opts["M"]:
             kw["per_gene_bl"] = True
+        if psr:
+            kw["rate_het"] = "CAT"
         res_path = os.path.join(wdir, f"ExaML_result.{name}")
         if opts["f"] in ("d", "o"):
             assert ckpt.state in (REARR_SETTING, FAST_SPRS,
@@ -221,6 +235,12 @@ def main(argv=None, device=None):
                 "checkpoint state does not match -f d"
             st = spr_tree(ckpt, len(taxa))
             ts = TreeSearch(st, engines, **kw)
+            if psr:
+                ts.rate_cat_invocations = \
+                    ckpt.optimize_rate_category_invocations
+                ts.cat_patrat = [
+                    np.asarray(ckpt.patrat[p.lower:p.upper]).copy()
+                    for p in parts]
             sp = SprSearch(ts, do_cutoff=(opts["f"] == "d"),
                            convergence_criterion=opts["D"],
                            save_best_trees=opts["B"], log=log)
