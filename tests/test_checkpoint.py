@@ -362,3 +362,41 @@ def test_resume_spr_search_psr_from_checkpoint(golden_dir):
     sp = SprSearch(ts)
     lnl = sp.compute_big_rapid(estimate_model=True, checkpoint=ck)
     assert abs(lnl - (-2507.657682)) < abs(2507.657682) * 1e-6, lnl
+
+
+def test_resume_spr_search_psr_M_from_checkpoint(golden_dir):
+    """-R restart under -m PSR -M together (per-partition branch
+    vectors AND rate-category state from one checkpoint); bit-exact
+    restore, reference-golden final."""
+    import numpy as np
+
+    from examl_amd.checkpoint import spr_tree
+    from examl_amd.spr import SprSearch
+    from tests.helpers import OracleCatEngine
+    taxa, parts = read_byte_file(os.path.join(golden_dir, "12m.binary"))
+    cl = sum(p.upper - p.lower for p in parts)
+    ck = read_checkpoint(
+        os.path.join(golden_dir, "12psrm.spr_slow.ckpt.bin"), 12, [4, 4],
+        rate_het="CAT", crunched_length=cl)
+    assert ck.state == 3 and ck.per_gene_bl
+    st = spr_tree(ck, 12)
+    engines = []
+    for p, m in zip(parts, ck.models):
+        model = ea.DnaGtrModel(m["frequencies"], m["substRates"],
+                               m["alpha"])
+        cptr = np.asarray(ck.rate_category[p.lower:p.upper],
+                          dtype=np.int32).copy()
+        rates = np.asarray(m["per_site_rates"][:m["num_cats"]]).copy()
+        engines.append(OracleCatEngine(p.tips, p.wgt, model, cptr, rates))
+    ts = TreeSearch(st, engines,
+                    opt_freq_flags=[bool(p.optimizeBaseFrequencies)
+                                    for p in parts],
+                    rate_het="CAT", per_gene_bl=True)
+    ts.rate_cat_invocations = ck.optimize_rate_category_invocations
+    ts.cat_patrat = [np.asarray(ck.patrat[p.lower:p.upper]).copy()
+                     for p in parts]
+    assert ts.evaluate_generic(full=True) == \
+        -2496.48818503920119837857782840728759765625
+    sp = SprSearch(ts)
+    lnl = sp.compute_big_rapid(estimate_model=True, checkpoint=ck)
+    assert abs(lnl - (-2496.442393)) < abs(2496.442393) * 1e-6, lnl
