@@ -8,11 +8,11 @@ reference user can point their invocation here unchanged:
     python -m examl_amd -s 49.binary -t 49.tree -m GAMMA -f E -n RUN
 
 Supported: -s -t -g -p -m -f(e|E|d|o|q) -n -w -a -M -S -D -B -c -e -i
--r -Y -v, plus AUTO protein partitions (per-partition model selection).
-Not yet wired: -R/-I (checkpoint restart; the binary checkpoint
-read/write interchange itself lives in examl_amd.checkpoint).  This is
-the PRODUCT path: it requires the HIP extension and an AMD GPU and
-fails loudly without one (no CPU fallback).
+-r -Y -v -R -I, plus AUTO protein partitions (per-partition model
+selection); the binary checkpoint read/write interchange itself lives
+in examl_amd.checkpoint.  This is the PRODUCT path: it requires the
+HIP extension and an AMD GPU and fails loudly without one (no CPU
+fallback).
 """
 
 import os
@@ -244,6 +244,11 @@ def main(argv=None, device=None):
             sp = SprSearch(ts, do_cutoff=(opts["f"] == "d"),
                            convergence_criterion=opts["D"],
                            save_best_trees=opts["B"], log=log)
+            if opts["D"]:
+                from examl_amd.examl_io import to_newick_topology
+                sp.seed_rfconv_from_checkpoint(ckpt, taxa)
+                sp.topology_string_fn = (
+                    lambda: to_newick_topology(st, taxa))
             final = sp.compute_big_rapid(estimate_model=True,
                                          checkpoint=ckpt)
             tree_out = st
@@ -329,10 +334,22 @@ def main(argv=None, device=None):
                     path, st,
                     [build_model_entry(e.model) for e in ts.engines],
                     len(taxa), state=state, spr=fields,
-                    start_number=st.start)
+                    start_number=st.start,
+                    likelihood_epsilon=opts["e"],
+                    use_median=opts["a"], save_best_trees=opts["B"],
+                    save_memory=opts["S"], search_convergence=opts["D"],
+                    categories=opts["c"],
+                    initial_set=opts["i"] is not None,
+                    initial=10 if opts["i"] is None else opts["i"],
+                    tree0=sp.slot_tree_strings[0],
+                    tree1=sp.slot_tree_strings[1])
                 counter[0] += 1
 
             sp.checkpoint_writer = _writer
+            if opts["D"]:
+                from examl_amd.examl_io import to_newick_topology
+                sp.topology_string_fn = (
+                    lambda: to_newick_topology(st, taxa))
         final = sp.compute_big_rapid(estimate_model=True,
                                      initial_trav=opts["i"])
         log(f"Likelihood of best tree: {final:.6f}")

@@ -290,7 +290,11 @@ def write_checkpoint(path, tree, models, mxtips, *, state=MOD_OPT,
                      rate_het="GAMMA", per_gene_bl=False,
                      likelihood_epsilon=0.1, rate_category=None,
                      patrat=None, likelihoods=None, start_number=1,
-                     accumulated_time=1.0, spr=None):
+                     accumulated_time=1.0, spr=None, use_median=False,
+                     save_best_trees=0, save_memory=False,
+                     search_convergence=False, categories=MAX_CATEGORIES,
+                     initial_set=False, initial=10, tree0=None,
+                     tree1=None):
     """Emit a checkpoint the reference's readCheckpoint accepts.  models:
     one dict per partition with the arrays of read_checkpoint's layout
     (build_model_entry converts our model objects).  spr: the
@@ -327,22 +331,41 @@ def write_checkpoint(path, tree, models, mxtips, *, state=MOD_OPT,
         struct.pack_into("<3i", out, 168, MAX_CATEGORIES,
                          spr.get("tr_it_count", 0),
                          int(spr.get("tr_do_cutoff", 1)))
+    # commandLine block (axml.h:660-679) — the reference's
+    # checkCommandLineArguments (searchAlgo.c:1383) hard-fails on any
+    # mismatch with the restart invocation, so every option the caller
+    # actually ran with must land here.
     c = 1248
-    struct.pack_into("<d", out, c + 24, likelihood_epsilon)
-    struct.pack_into("<i", out, c + 32, MAX_CATEGORIES)
+    struct.pack_into("<i", out, c + 0, 1 if use_median else 0)        # -a
+    struct.pack_into("<i", out, c + 4, int(save_best_trees))          # -B
+    struct.pack_into("<i", out, c + 8, 1 if save_memory else 0)       # -S
+    struct.pack_into("<i", out, c + 12, 1 if search_convergence else 0)
+    struct.pack_into("<i", out, c + 16, 1 if per_gene_bl else 0)      # -M
+    struct.pack_into("<d", out, c + 24, likelihood_epsilon)           # -e
+    struct.pack_into("<i", out, c + 32, int(categories))              # -c
     # adef->mode: BIG_RAPID_MODE for SPR-state checkpoints (axml.h:226)
     struct.pack_into("<i", out, c + 36,
                      1 if state in (REARR_SETTING, FAST_SPRS, SLOW_SPRS)
                      else 0)
-    struct.pack_into("<i", out, c + 44, 0)   # initialSet
-    struct.pack_into("<i", out, c + 48, 10)  # adef->initial default
+    struct.pack_into("<i", out, c + 44, 1 if initial_set else 0)
+    struct.pack_into("<i", out, c + 48, int(initial))                 # -i
     struct.pack_into("<i", out, c + 52,
                      RATE_HET_CAT if rate_het == "CAT" else RATE_HET_GAMMA)
-    struct.pack_into("<i", out, c + 16, 1 if per_gene_bl else 0)
 
     buf = bytearray(bytes(out))
-    buf += bytes(tsl)  # tree0
-    buf += bytes(tsl)  # tree1
+
+    def tstr(t):
+        """tree0/tree1 topology string padded to treeStringLength
+        (searchAlgo.c:1207-1208); empty when the slot was never
+        stored."""
+        if t is None:
+            return bytes(tsl)
+        b = t.encode() if isinstance(t, str) else bytes(t)
+        assert len(b) < tsl, "tree string exceeds treeStringLength"
+        return b + bytes(tsl - len(b))
+
+    buf += tstr(tree0)
+    buf += tstr(tree1)
 
     if rate_het == "CAT":
         buf += np.ascontiguousarray(rate_category, np.int32).tobytes()
@@ -360,6 +383,17 @@ def write_checkpoint(path, tree, models, mxtips, *, state=MOD_OPT,
             buf += a.tobytes()
         buf += np.asarray(m.get("weights", np.full(4, 0.25))).tobytes()
         buf += np.asarray(m.get("weightExponents", np.zeros(4))).tobytes()
+        if m.get("protModels", 0) in (20, 21):
+            # LG4M/LG4X: the four per-category eigensystem blocks between
+            # weightExponents and alpha (writeCheckpointInner,
+            # searchAlgo.c:1244-1260); readCheckpoint gates this read on
+            # the byte file's protModels, so it must be present
+            for k in range(4):
+                for key in ("rawEIGN_LG4", "EIGN_LG4", "EV_LG4", "EI_LG4",
+                            "frequencies_LG4", "tipVector_LG4",
+                            "substRates_LG4"):
+                    buf += np.ascontiguousarray(m[key][k],
+                                                np.float64).tobytes()
         buf += struct.pack("<d", m["alpha"])
         buf += np.ascontiguousarray(m["gammaRates"], np.float64).tobytes()
         buf += struct.pack("<ii", m.get("protModels", 0 if states == 4

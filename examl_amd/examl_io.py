@@ -205,6 +205,30 @@ def parse_newick_topology(s, taxa_names, read_bl=False):
     return tree
 
 
+def to_newick_topology(tree, taxa_names):
+    """Topology-only Tree2String (printBranchLengths=FALSE), as used for
+    the -D convergence tree strings tr->tree0/tree1
+    (searchAlgo.c:2178)."""
+
+    def sub(p, parent):
+        if tree.is_tip(p):
+            return taxa_names[p - 1]
+        if hasattr(tree, "ring_children"):
+            kids = tree.ring_children(p, parent)
+        else:
+            kids = [w for w in tree.adj[p] if w != parent]
+        return "(" + ",".join(sub(w, p) for w in kids) + ")"
+
+    start = 1
+    back = next(iter(tree.adj[start]))
+    if hasattr(tree, "ring_children"):
+        kids = tree.ring_children(back, start)
+    else:
+        kids = [w for w in tree.adj[back] if w != start]
+    parts = [sub(start, back)] + [sub(w, back) for w in kids]
+    return "(" + ",".join(parts) + ");"
+
+
 def to_newick(tree, taxa_names, digits=20):
     """Write the tree with branch lengths as the reference's Tree2String
     does (treeIO.c:234, branch length = -log(z), SUMMARIZE_LH averages

@@ -1214,13 +1214,6 @@ class TreeSearch:
                 break
         return self.likelihood
 
-    def reset_branches(self):
-        """resetBranches (optimizeModel.c:2511): every branch back to
-        defaultz."""
-        from .tree import DEFAULTZ
-        for a, b in self.tree.edges():
-            self.tree.set_z(a, b, DEFAULTZ)
-
     def tree_evaluation_mode(self, log=None, epsilon=0.1):
         """The -f E (slow TREE_EVALUATION) flow for one tree
         (axml.c:2316-2331): evaluate, treeEvaluate(1), modOpt with

@@ -479,6 +479,37 @@ class RfConvergence:
     def clear(self):
         self.table = {}
 
+    def seed_from_newick(self, newick, slot, taxa_names):
+        """Re-populate one slot from a stored tr->tree0/tree1 topology
+        string on -R restart (readCheckpoint, searchAlgo.c:1545-1580:
+        treeReadTopologyString + bitVectorInitravSpecial into slot
+        0/1)."""
+        from .examl_io import parse_newick_topology
+        if isinstance(newick, bytes):
+            newick = newick.split(b"\0", 1)[0].decode()
+        t = parse_newick_topology(newick.strip(), taxa_names)
+        out = []
+        start = 1
+        back = next(iter(t.adj[start]))
+
+        def down(m, parent):
+            if t.is_tip(m):
+                return frozenset((m,))
+            s = frozenset()
+            for w in t.adj[m]:
+                if w != parent:
+                    s |= down(w, m)
+            out.append(s)
+            return s
+
+        for w in t.adj[back]:
+            if w != start:
+                down(w, back)
+        assert len(out) == len(taxa_names) - 3, len(out)
+        bit = 1 << slot
+        for b in out:
+            self.table[b] = self.table.get(b, 0) | bit
+
 
 class SprSearch:
     """computeBIGRAPID over a TreeSearch on an SprTree."""
@@ -509,6 +540,12 @@ class SprSearch:
         # optional writeCheckpoint hook: called as writer(state, fields)
         # at the reference's write points (searchAlgo.c:2155/:2425)
         self.checkpoint_writer = None
+        # -D bookkeeping for checkpoints: topology strings captured at
+        # the rfconv store points (tr->tree0/tree1, searchAlgo.c:2178);
+        # topology_string_fn is set by the CLI to a Tree2String-style
+        # callback
+        self.topology_string_fn = None
+        self.slot_tree_strings = [None, None]
         self.good_trees = []  # good-tree lnls after compute_big_rapid
         # tr-> search state
         self.start_lh = 0.0
@@ -1030,6 +1067,27 @@ class SprSearch:
         self.do_cutoff = cutoff
         return best_trav
 
+    def seed_rfconv_from_checkpoint(self, ck, taxa_names):
+        """-R resume with -D: re-populate the RF-convergence table from
+        the checkpoint's tree0/tree1 topology strings, with the
+        reference's gating (readCheckpoint, searchAlgo.c:1545-1580:
+        tree0 when the resumed phase's iteration count > 0, tree1 when
+        > 1)."""
+        if not self.convergence_criterion:
+            return
+        if ck.state == CKP_FAST_SPRS:
+            it = ck.fast_iterations
+        elif ck.state == CKP_SLOW_SPRS:
+            it = ck.thorough_iterations
+        else:
+            return
+        for slot, tstr in ((0, ck.tree0), (1, ck.tree1)):
+            if it > slot:
+                self.rfconv.seed_from_newick(tstr, slot, taxa_names)
+                self.slot_tree_strings[slot] = \
+                    tstr.split(b"\0", 1)[0].decode() \
+                    if isinstance(tstr, bytes) else tstr
+
     def compute_big_rapid(self, estimate_model=True, initial_trav=None,
                           checkpoint=None):
         """computeBIGRAPID (:1914).  checkpoint: a parsed reference -R
@@ -1121,6 +1179,9 @@ class SprSearch:
             # compare against the previous cycle's tree.
             if self.convergence_criterion:
                 self.rfconv.store(fast_iterations)
+                if self.topology_string_fn is not None:
+                    self.slot_tree_strings[fast_iterations % 2] = \
+                        self.topology_string_fn()
                 if fast_iterations > 0:
                     rrf = self.rfconv.rrf()
                     if rrf <= 0.01:  # 1% cutoff
@@ -1198,6 +1259,9 @@ class SprSearch:
                 # -D check (searchAlgo.c:2438): slot thoroughIterations%2
                 if self.convergence_criterion:
                     self.rfconv.store(thorough_iterations)
+                    if self.topology_string_fn is not None:
+                        self.slot_tree_strings[thorough_iterations % 2] = \
+                            self.topology_string_fn()
                     if thorough_iterations > 0:
                         rrf = self.rfconv.rrf()
                         if rrf <= 0.01:  # goto cleanup

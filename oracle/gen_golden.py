@@ -247,4 +247,9 @@ def gen_cat():
 
 
 if __name__ == "__main__":
+    # regenerate the FULL golden kernel-fixture set (model_dna,
+    # kernels_dna_gamma, kernels_prot_gamma, kernels_dna_cat); the
+    # end-to-end fixtures (12*/49/140 binaries, trees, checkpoints) come
+    # from tests/golden/gen_12.py + the reference binaries in oracle/_ref
+    main()
     gen_cat()
