@@ -230,10 +230,13 @@ def main():
     if rank != 0:
         return
 
-    # roofline of the dominant kernel (newview INNER_INNER)
+    # roofline of the dominant kernel (newview INNER_INNER).  Each launch
+    # covers ONE partition's pw = width/P sites (the profile counters sum
+    # over all P engines' launches), so bytes/launch uses pw — using the
+    # full per-GPU width here would overcount P-fold.
     ii_ms, ii_n = float(ms[2]), int(cnt[2])
     bps = BYTES_PER_SITE_II[model.states]
-    achieved = (bps * width * ii_n) / (ii_ms * 1e-3) \
+    achieved = (bps * pw * ii_n) / (ii_ms * 1e-3) \
         if ii_ms > 0 else None
     # PMC-measured HBM traffic for this exact workload (collected in a
     # separate rocprofv3 --pmc pass, corrected per MI355X_MICROARCH.md §HBM;
@@ -241,7 +244,7 @@ def main():
     traffic = None
     cal = os.path.join(os.path.dirname(os.path.abspath(__file__)),
                        "profiles", "r01_pmc_traffic.json")
-    if width == SITES_PER_GPU and model.states == 4 and os.path.exists(cal):
+    if pw == SITES_PER_GPU and model.states == 4 and os.path.exists(cal):
         with open(cal) as f:
             traffic = json.load(f)["traffic_bytes_per_launch"]
     roofline = {
