@@ -601,15 +601,23 @@ __global__ __launch_bounds__(NV_BLOCK) void k_newview_prot_gamma(
     const unsigned char *__restrict__ tipX1,
     const unsigned char *__restrict__ tipX2, const int *__restrict__ wgt,
     long n, unsigned int *__restrict__ scalerInc) {
-  __shared__ double sL[1600], sR[1600], sEV[400];
+  /* The four lanes of one site differ only in `cat`; with the natural
+   * cat stride of 400 doubles (800 LDS words, = 0 mod 32 banks) every
+   * P-row read is a 4-way bank conflict.  Padding the cat stride to 404
+   * doubles (808 words, = 8 mod 32) puts the four cats on banks
+   * 0/8/16/24 — conflict-free, which is what the round-1 "LDS-issue
+   * bound" profile was hitting. */
+  constexpr int CSTR = 404; /* padded per-cat LDS stride of sL/sR */
+  __shared__ double sL[4 * CSTR], sR[4 * CSTR], sEV[400];
   __shared__ double sTV[TC != EXAML_INNER_INNER ? 460 : 1];
   __shared__ double sU1[TC != EXAML_INNER_INNER ? 1840 : 1];
   __shared__ double sU2[TC == EXAML_TIP_TIP ? 1840 : 1];
 
   const int tid = threadIdx.x;
   for (int j = tid; j < 1600; j += NV_BLOCK) {
-    sL[j] = P[j];
-    sR[j] = P[1600 + j];
+    const int pc = j / 400, pr = j % 400;
+    sL[pc * CSTR + pr] = P[j];
+    sR[pc * CSTR + pr] = P[1600 + j];
   }
   for (int j = tid; j < 400; j += NV_BLOCK) sEV[j] = EV[j];
   if (TC != EXAML_INNER_INNER)
@@ -620,9 +628,10 @@ __global__ __launch_bounds__(NV_BLOCK) void k_newview_prot_gamma(
     /* ump tables (avxLikelihood.c:1355-1389): entry (code, cat*20+row) */
     for (int j = tid; j < 23 * 80; j += NV_BLOCK) {
       const int code = j / 80, k = j % 80;
-      sU1[j] = dot20o<FAST>(&sTV[20 * code], &sL[k * 20]);
+      const int kc = k / 20, kl = k % 20;
+      sU1[j] = dot20o<FAST>(&sTV[20 * code], &sL[kc * CSTR + kl * 20]);
       if (TC == EXAML_TIP_TIP)
-        sU2[j] = dot20o<FAST>(&sTV[20 * code], &sR[k * 20]);
+        sU2[j] = dot20o<FAST>(&sTV[20 * code], &sR[kc * CSTR + kl * 20]);
     }
     __syncthreads();
   }
@@ -659,11 +668,11 @@ __global__ __launch_bounds__(NV_BLOCK) void k_newview_prot_gamma(
     for (int l = 0; l < 20; l++) {
       double u1, u2;
       if (TC == EXAML_INNER_INNER) {
-        u1 = dot20o<FAST>(xl, &sL[cat * 400 + l * 20]);
-        u2 = dot20o<FAST>(xr, &sR[cat * 400 + l * 20]);
+        u1 = dot20o<FAST>(xl, &sL[cat * CSTR + l * 20]);
+        u2 = dot20o<FAST>(xr, &sR[cat * CSTR + l * 20]);
       } else if (TC == EXAML_TIP_INNER) {
         u1 = sU1[80 * code1 + cat * 20 + l];
-        u2 = dot20o<FAST>(xr, &sR[cat * 400 + l * 20]);
+        u2 = dot20o<FAST>(xr, &sR[cat * CSTR + l * 20]);
       } else {
         u1 = sU1[80 * code1 + cat * 20 + l];
         u2 = sU2[80 * code2 + cat * 20 + l];
