@@ -161,8 +161,7 @@ def main():
                                  wgt[i*pw:(i+1)*pw], model, device=device)
                for i in range(P)]
     eng = engines[0]
-    streams = [torch.cuda.Stream(device=device) for _ in range(P)] \
-        if P > 1 else []
+    multi = ea.MultiDnaEngine(engines) if P > 1 else None
     entries, (p, q, z) = tree.full_traversal()
     n_ops = len(entries)
     tc_counts = [sum(1 for e in entries if e.tipCase == t) for t in range(3)]
@@ -171,28 +170,18 @@ def main():
         if P == 1:
             eng.newview_traversal(entries)
             return eng.evaluate_root(tree, p, q, z, all_reduce=world > 1)
-        # partitioned: one HIP stream per partition (the per-partition loop
-        # of newviewIterative/evaluateIterative runs concurrently), then ONE
+        # partitioned: the fused mseg executors — one launch per
+        # (traversal level x tipCase) covering all P partitions, then ONE
         # all-reduce of the per-partition lnL vector (the C1 collective)
-        cur = torch.cuda.current_stream(device)
-        for e_, st in zip(engines, streams):
-            st.wait_stream(cur)
-            with torch.cuda.stream(st):
-                e_.newview_traversal(entries)
-                e_.evaluate_root(tree, p, q, z)
-        for st in streams:
-            cur.wait_stream(st)
-        lnl_vec = torch.cat([e_.d_lnl for e_ in engines])
-        if world > 1:
-            import torch.distributed as dd
-            dd.all_reduce(lnl_vec)
+        multi.newview_traversal(entries)
+        lnl_vec = multi.evaluate_root(tree, p, q, z, all_reduce=world > 1)
         return lnl_vec.sum()
 
     # warmup (also captures the traversal's hipGraph)
     for _ in range(args.warmup):
-        step()
+        lnl_t = step()
     eng.sync()
-    lnl0 = float(eng.d_lnl.cpu())
+    lnl0 = float(lnl_t.sum().cpu())
 
     # timed region: graph-replayed traversals, no profiling overhead
     if dist:
@@ -230,14 +219,24 @@ def main():
     if rank != 0:
         return
 
-    # roofline of the dominant kernel (newview INNER_INNER).  Each launch
-    # covers ONE partition's pw = width/P sites (the profile counters sum
-    # over all P engines' launches), so bytes/launch uses pw — using the
-    # full per-GPU width here would overcount P-fold.
-    ii_ms, ii_n = float(ms[2]), int(cnt[2])
+    # roofline.  P == 1: the dominant kernel (newview INNER_INNER), bytes
+    # per launch over HIP-event launch time.  P > 1 (fused mseg path): one
+    # launch covers many (op, partition) segments, so use TOTAL algorithmic
+    # newview bytes over TOTAL newview kernel time for the profiled steps —
+    # per-launch bytes would overcount.
+    BPS_TC = {4: (130.0, 260.0, 388.0), 20: (164.0, 1124.0, 1924.0)}
     bps = BYTES_PER_SITE_II[model.states]
-    achieved = (bps * pw * ii_n) / (ii_ms * 1e-3) \
-        if ii_ms > 0 else None
+    ii_ms, ii_n = float(ms[2]), int(cnt[2])
+    if P == 1:
+        achieved = (bps * pw * ii_n) / (ii_ms * 1e-3) \
+            if ii_ms > 0 else None
+    else:
+        steps_prof = max(3, args.steps // 10)
+        bytes_step = sum(tc_counts[t] * BPS_TC[model.states][t] * pw * P
+                         for t in range(3))
+        tot_ms = float(ms.sum())
+        achieved = (steps_prof * bytes_step) / (tot_ms * 1e-3) \
+            if tot_ms > 0 else None
     # PMC-measured HBM traffic for this exact workload (collected in a
     # separate rocprofv3 --pmc pass, corrected per MI355X_MICROARCH.md §HBM;
     # see profiles/r01_pmc_traffic.json)

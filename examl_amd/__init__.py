@@ -148,6 +148,16 @@ def _bind(lib):
     lib.examl_hip_use_graphs.argtypes = [i]
     lib.examl_hip_graphs_clear.argtypes = []
     lib.examl_hip_fast_math.argtypes = [i]
+    # multi-partition fused executors
+    lib.examl_hip_multi_create.argtypes = \
+        [i, i, p, p, p, p, p, p, p, p, p, i, p]
+    lib.examl_hip_multi_destroy.argtypes = [p]
+    lib.examl_hip_newview_traversal_multi.argtypes = \
+        [p, p, i, p, p, p, p, p, p, p]
+    lib.examl_hip_evaluate_root_multi.argtypes = \
+        [p, i, i, i, i, i, i, p, i, p, p, p, p, p]
+    lib.examl_hip_sum_root_multi.argtypes = [p, i, i, i, i, i, p, p]
+    lib.examl_hip_core_root_multi.argtypes = [p, p, i, p, p, p, p, p]
     return lib
 
 
@@ -179,13 +189,13 @@ from .model import DnaGtrModel, Lg4Model, ProtGtrModel  # noqa: E402
 from .tree import PhyloTree             # noqa: E402
 from .spr import SprSearch, SprTree     # noqa: E402
 from .engine import (DnaCatEngine, DnaGammaEngine, Lg4Engine,  # noqa: E402
-                     ProtCatEngine, SaveDnaEngine)
+                     MultiDnaEngine, ProtCatEngine, SaveDnaEngine)
 
 __all__ = [
     "lib", "check", "TravEntry", "DnaGtrModel", "ProtGtrModel",
     "Lg4Model", "PhyloTree", "SprSearch", "SprTree",
     "DnaGammaEngine", "DnaCatEngine", "Lg4Engine", "SaveDnaEngine",
-    "ProtCatEngine",
+    "ProtCatEngine", "MultiDnaEngine",
     "TIP_TIP", "TIP_INNER",
     "INNER_INNER", "ZMIN", "ZMAX",
 ]

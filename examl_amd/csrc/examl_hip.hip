@@ -3824,3 +3824,1053 @@ extern "C" int examl_hip_core_root_prot_cat(
 }
 
 #undef CHK
+
+/* ===========================================================================
+ * Multi-partition fused executors ("mseg"): one kernel launch per
+ * (traversal level x tipCase) covering ALL partitions, replacing the
+ * per-(partition, op) launch storm that made partitioned shapes
+ * launch-bound (config 3: 16 x 7812-site partitions; 140.model AUTO).
+ *
+ * Replaces the per-partition loops of newviewIterative
+ * (newviewGenericSpecial.c:1064), evaluateIterative
+ * (evaluateGenericSpecial.c:509), makenewzIterative
+ * (makenewzGenericSpecial.c:673) and execCore (:885) with segment-indexed
+ * fused grids: a segment is one (op, partition) work unit; blocks map to
+ * segments through a per-shape blk2seg table; masked partitions
+ * (executeModel / td[0].executeModel) exit via a per-call device flag
+ * array so CLVs and scalers stay untouched, exactly like the reference's
+ * dispatch gating.
+ *
+ * P matrices move to the DEVICE here (k_make_p_mseg) — per call only the
+ * log-branch-lengths (2 doubles per op x partition) and the per-partition
+ * model vectors (EIGN/EI/rates) are uploaded; device exp() differs from
+ * host libm in the last ulp, so the fused path is pinned to the
+ * single-partition engines at <=1e-11 relative rather than bit-exact
+ * (tests/test_multi_fused.py).
+ * ==========================================================================*/
+
+#define CHK(call)                                                              \
+  do {                                                                         \
+    hipError_t _e = (call);                                                    \
+    if (_e != hipSuccess) return set_err(_e, #call);                           \
+  } while (0)
+
+struct MSeg { /* one (op, partition) newview work unit */
+  const double *x1, *x2;
+  double *x3;
+  const unsigned char *t1, *t2;
+  const double *P;
+  const int *wgt;
+  unsigned int *inc;
+  const double *EV, *tipVec;
+  long n;
+  int blkBase, nBlocks, part, pad;
+};
+
+struct ESeg { /* one partition's root-evaluate unit */
+  const double *x1, *x2;
+  const unsigned char *t1;
+  const double *diag;
+  const int *wgt;
+  const unsigned int *gsP, *gsQ;
+  double *lnlOut;
+  long n;
+  int blkBase, nBlocks, part, pad;
+};
+
+struct SSeg { /* one partition's sumBuffer unit */
+  const double *x1, *x2;
+  const unsigned char *t1, *t2;
+  double *sum;
+  const double *tipVec;
+  long n;
+  int blkBase, nBlocks, part, pad;
+};
+
+struct CSeg { /* one partition's NR-derivative unit */
+  const double *sum;
+  const double *dtab;
+  const int *wgt;
+  double *out2;
+  long n;
+  int blkBase, nBlocks, part, pad;
+};
+
+/* device P-matrix pairs (makeP, newviewGenericSpecial.c:78):
+ * P[(e*numParts+part)*8*S^2] = [left|right], left[cat*S^2+row*S+col],
+ * col 0 = 1, else exp(rates[cat]*EIGN[col]*lz)*EI[row*S+col]. */
+template <int STATES>
+__global__ void k_make_p_mseg(const double *__restrict__ zp,
+                              const double *__restrict__ mod, int numParts,
+                              double *__restrict__ pbuf) {
+  constexpr int MSZ = STATES + STATES * STATES + 4;
+  constexpr int HALF = 4 * STATES * STATES;
+  const int gid = blockIdx.x; /* e*numParts + part */
+  const int part = gid % numParts;
+  const double *EIGN = mod + (long)part * MSZ;
+  const double *EI = EIGN + STATES;
+  const double *rates = EI + STATES * STATES;
+  const double z1 = zp[gid * 2], z2 = zp[gid * 2 + 1];
+  double *P = pbuf + (long)gid * (2 * HALF);
+  for (int j = threadIdx.x; j < 2 * HALF; j += blockDim.x) {
+    const int h = j / HALF, r = j % HALF;
+    const int cat = r / (STATES * STATES), rc = r % (STATES * STATES);
+    const int row = rc / STATES, col = rc % STATES;
+    const double z = h ? z2 : z1;
+    P[j] = (col == 0)
+               ? 1.0
+               : exp(rates[cat] * (EIGN[col] * z)) * EI[row * STATES + col];
+  }
+}
+
+/* fused newview, DNA GTRGAMMA — math identical to k_newview_dna_gamma */
+template <int TC>
+__global__ __launch_bounds__(NV_BLOCK) void k_newview_dna_mseg(
+    const MSeg *__restrict__ segs, const int *__restrict__ blk2seg,
+    const double *__restrict__ active) {
+  const int si = blk2seg[blockIdx.x];
+  const MSeg sg = segs[si];
+  if (active[sg.part] == 0.0) return;
+  __shared__ double sL[64], sR[64], sEV[16], sTV[64];
+  __shared__ double sU1[256], sU2[TC == EXAML_TIP_TIP ? 256 : 1];
+
+  const int tid = threadIdx.x;
+  if (tid < 64) {
+    sL[tid] = sg.P[tid];
+    sR[tid] = sg.P[64 + tid];
+    sTV[tid] = sg.tipVec[tid];
+  }
+  if (tid < 16) sEV[tid] = sg.EV[tid];
+  __syncthreads();
+
+  if (TC != EXAML_INNER_INNER) {
+    const int code = tid >> 4, cat = (tid >> 2) & 3, row = tid & 3;
+    const double *tv = &sTV[code * 4];
+    const double *pl = &sL[cat * 16 + row * 4];
+    sU1[tid] =
+        (pl[0] * tv[0] + pl[1] * tv[1]) + (pl[2] * tv[2] + pl[3] * tv[3]);
+    if (TC == EXAML_TIP_TIP) {
+      const double *pr = &sR[cat * 16 + row * 4];
+      sU2[tid] =
+          (pr[0] * tv[0] + pr[1] * tv[1]) + (pr[2] * tv[2] + pr[3] * tv[3]);
+    }
+    __syncthreads();
+  }
+
+  const long units = sg.n * 4;
+  const bool nt = sg.n >= 65536;
+  const int lane = tid & 63;
+  for (long idx = (long)(blockIdx.x - sg.blkBase) * NV_BLOCK + tid;
+       idx < units; idx += (long)sg.nBlocks * NV_BLOCK) {
+    const long site = idx >> 2;
+    const int cat = (int)(idx & 3);
+    double u1[4], u2[4];
+
+    if (TC == EXAML_INNER_INNER) {
+      const double4 xl = *reinterpret_cast<const double4 *>(&sg.x1[idx * 4]);
+      const double4 xr = *reinterpret_cast<const double4 *>(&sg.x2[idx * 4]);
+#pragma unroll
+      for (int l = 0; l < 4; l++) {
+        const double *pl = &sL[cat * 16 + l * 4];
+        const double *pr = &sR[cat * 16 + l * 4];
+        u1[l] = (xl.x * pl[0] + xl.y * pl[1]) + (xl.z * pl[2] + xl.w * pl[3]);
+        u2[l] = (xr.x * pr[0] + xr.y * pr[1]) + (xr.z * pr[2] + xr.w * pr[3]);
+      }
+    } else if (TC == EXAML_TIP_INNER) {
+      const int code = sg.t1[site];
+      const double4 xr = *reinterpret_cast<const double4 *>(&sg.x2[idx * 4]);
+#pragma unroll
+      for (int l = 0; l < 4; l++) {
+        const double *pr = &sR[cat * 16 + l * 4];
+        u1[l] = sU1[code * 16 + cat * 4 + l];
+        u2[l] = (xr.x * pr[0] + xr.y * pr[1]) + (xr.z * pr[2] + xr.w * pr[3]);
+      }
+    } else {
+      const int c1 = sg.t1[site], c2 = sg.t2[site];
+#pragma unroll
+      for (int l = 0; l < 4; l++) {
+        u1[l] = sU1[c1 * 16 + cat * 4 + l];
+        u2[l] = sU2[c2 * 16 + cat * 4 + l];
+      }
+    }
+
+    double a0 = 0, a1 = 0, a2 = 0, a3 = 0;
+#pragma unroll
+    for (int l = 0; l < 4; l++) {
+      const double t = u1[l] * u2[l];
+      a0 += t * sEV[l * 4 + 0];
+      a1 += t * sEV[l * 4 + 1];
+      a2 += t * sEV[l * 4 + 2];
+      a3 += t * sEV[l * 4 + 3];
+    }
+
+    if (TC != EXAML_TIP_TIP) {
+      const bool small = (fabs(a0) < MINLIKELIHOOD) &
+                         (fabs(a1) < MINLIKELIHOOD) &
+                         (fabs(a2) < MINLIKELIHOOD) &
+                         (fabs(a3) < MINLIKELIHOOD);
+      const unsigned long long m = __ballot(small);
+      if (((m >> (lane & ~3)) & 0xFULL) == 0xFULL) {
+        a0 *= TWOTOTHE256;
+        a1 *= TWOTOTHE256;
+        a2 *= TWOTOTHE256;
+        a3 *= TWOTOTHE256;
+        if ((lane & 3) == 0)
+          atomicAdd(sg.inc, (unsigned int)sg.wgt[site]);
+      }
+    }
+    if (nt)
+      __builtin_nontemporal_store((v4d){a0, a1, a2, a3},
+                                  reinterpret_cast<v4d *>(&sg.x3[idx * 4]));
+    else
+      *reinterpret_cast<double4 *>(&sg.x3[idx * 4]) =
+          make_double4(a0, a1, a2, a3);
+  }
+}
+
+/* fused evaluate, DNA — math identical to k_evaluate_dna_gamma */
+template <bool TIP>
+__global__ __launch_bounds__(NV_BLOCK) void k_evaluate_dna_mseg(
+    const ESeg *__restrict__ segs, const int *__restrict__ blk2seg,
+    const double *__restrict__ active, double *__restrict__ partials) {
+  const int si = blk2seg[blockIdx.x];
+  const ESeg sg = segs[si];
+  if (active[sg.part] == 0.0) return;
+  __shared__ double sD[16], sTV[64], sRed[NV_BLOCK / 64];
+  const int tid = threadIdx.x;
+  if (tid < 16) sD[tid] = sg.diag[tid];
+  if (TIP && tid < 64) sTV[tid] = segs[si].x1[tid]; /* x1 = tipVec for TIP */
+  __syncthreads();
+
+  const long units = sg.n * 4;
+  const int lane = tid & 63;
+  double acc = 0.0;
+  for (long idx = (long)(blockIdx.x - sg.blkBase) * NV_BLOCK + tid;
+       idx < units; idx += (long)sg.nBlocks * NV_BLOCK) {
+    const long site = idx >> 2;
+    const int cat = (int)(idx & 3);
+    const double4 b = *reinterpret_cast<const double4 *>(&sg.x2[idx * 4]);
+    double p;
+    if (TIP) {
+      const double *tv = &sTV[sg.t1[site] * 4];
+      p = ((tv[0] * b.x) * sD[cat * 4 + 0] + (tv[1] * b.y) * sD[cat * 4 + 1]) +
+          ((tv[2] * b.z) * sD[cat * 4 + 2] + (tv[3] * b.w) * sD[cat * 4 + 3]);
+    } else {
+      const double4 a = *reinterpret_cast<const double4 *>(&sg.x1[idx * 4]);
+      p = ((a.x * b.x) * sD[cat * 4 + 0] + (a.y * b.y) * sD[cat * 4 + 1]) +
+          ((a.z * b.z) * sD[cat * 4 + 2] + (a.w * b.w) * sD[cat * 4 + 3]);
+    }
+    p += __shfl_xor(p, 1);
+    p += __shfl_xor(p, 2);
+    if ((lane & 3) == 0) acc += (double)sg.wgt[site] * log(0.25 * fabs(p));
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off);
+  if (lane == 0) sRed[tid >> 6] = acc;
+  __syncthreads();
+  if (tid == 0) {
+    double s = 0;
+#pragma unroll
+    for (int w = 0; w < NV_BLOCK / 64; w++) s += sRed[w];
+    partials[blockIdx.x] = s;
+  }
+}
+
+/* deterministic per-partition second pass + scaler undo */
+__global__ __launch_bounds__(NV_BLOCK) void k_reduce_lnl_mseg(
+    const ESeg *__restrict__ segs, const double *__restrict__ partials,
+    const double *__restrict__ active, double log_minlik) {
+  const ESeg sg = segs[blockIdx.x];
+  if (active[sg.part] == 0.0) return;
+  __shared__ double sred[NV_BLOCK];
+  const int tid = threadIdx.x;
+  double v = 0;
+  for (int i = tid; i < sg.nBlocks; i += NV_BLOCK)
+    v += partials[sg.blkBase + i];
+  sred[tid] = v;
+  __syncthreads();
+  for (int off = NV_BLOCK / 2; off > 0; off >>= 1) {
+    if (tid < off) sred[tid] += sred[tid + off];
+    __syncthreads();
+  }
+  if (tid == 0) {
+    double s = sred[0];
+    s += ((double)(*sg.gsP) + (double)(*sg.gsQ)) * log_minlik;
+    *sg.lnlOut += s;
+  }
+}
+
+/* fused sumBuffer, DNA — math identical to k_sum_dna_gamma */
+template <int TC>
+__global__ __launch_bounds__(NV_BLOCK) void k_sum_dna_mseg(
+    const SSeg *__restrict__ segs, const int *__restrict__ blk2seg,
+    const double *__restrict__ active) {
+  const int si = blk2seg[blockIdx.x];
+  const SSeg sg = segs[si];
+  if (active[sg.part] == 0.0) return;
+  __shared__ double sTV[64];
+  const int tid = threadIdx.x;
+  if (TC != EXAML_INNER_INNER && tid < 64) sTV[tid] = sg.tipVec[tid];
+  if (TC != EXAML_INNER_INNER) __syncthreads();
+
+  const long units = sg.n * 4;
+  for (long idx = (long)(blockIdx.x - sg.blkBase) * NV_BLOCK + tid;
+       idx < units; idx += (long)sg.nBlocks * NV_BLOCK) {
+    const long site = idx >> 2;
+    double4 a, b;
+    if (TC == EXAML_TIP_TIP) {
+      const double *t1 = &sTV[sg.t1[site] * 4];
+      /* second tip's vector rows live in the same partition table */
+      const double *t2 = &sTV[sg.t2[site] * 4];
+      a = make_double4(t1[0], t1[1], t1[2], t1[3]);
+      b = make_double4(t2[0], t2[1], t2[2], t2[3]);
+    } else if (TC == EXAML_TIP_INNER) {
+      const double *t1 = &sTV[sg.t1[site] * 4];
+      a = make_double4(t1[0], t1[1], t1[2], t1[3]);
+      b = *reinterpret_cast<const double4 *>(&sg.x2[idx * 4]);
+    } else {
+      a = *reinterpret_cast<const double4 *>(&sg.x1[idx * 4]);
+      b = *reinterpret_cast<const double4 *>(&sg.x2[idx * 4]);
+    }
+    *reinterpret_cast<double4 *>(&sg.sum[idx * 4]) =
+        make_double4(a.x * b.x, a.y * b.y, a.z * b.z, a.w * b.w);
+  }
+}
+
+/* fused NR derivatives, DNA — math identical to k_core_dna_gamma */
+__global__ __launch_bounds__(NV_BLOCK) void k_core_dna_mseg(
+    const CSeg *__restrict__ segs, const int *__restrict__ blk2seg,
+    const double *__restrict__ active, double *__restrict__ partials) {
+  const int si = blk2seg[blockIdx.x];
+  const CSeg sg = segs[si];
+  if (active[sg.part] == 0.0) return;
+  __shared__ double sD0[16], sD1[16], sD2[16], sRed[2][NV_BLOCK / 64];
+  const int tid = threadIdx.x;
+  if (tid < 16) {
+    sD0[tid] = sg.dtab[tid];
+    sD1[tid] = sg.dtab[16 + tid];
+    sD2[tid] = sg.dtab[32 + tid];
+  }
+  __syncthreads();
+
+  const long units = sg.n * 4;
+  const int lane = tid & 63;
+  double accD1 = 0.0, accD2 = 0.0;
+  for (long idx = (long)(blockIdx.x - sg.blkBase) * NV_BLOCK + tid;
+       idx < units; idx += (long)sg.nBlocks * NV_BLOCK) {
+    const long site = idx >> 2;
+    const int cat = (int)(idx & 3);
+    const double4 s = *reinterpret_cast<const double4 *>(&sg.sum[idx * 4]);
+    const double t0 = sD0[cat * 4 + 0] * s.x, t1 = sD0[cat * 4 + 1] * s.y,
+                 t2 = sD0[cat * 4 + 2] * s.z, t3 = sD0[cat * 4 + 3] * s.w;
+    double a0 = (t0 + t1) + (t2 + t3);
+    double a1 = (t0 * sD1[cat * 4 + 0] + t1 * sD1[cat * 4 + 1]) +
+                (t2 * sD1[cat * 4 + 2] + t3 * sD1[cat * 4 + 3]);
+    double a2 = (t0 * sD2[cat * 4 + 0] + t1 * sD2[cat * 4 + 1]) +
+                (t2 * sD2[cat * 4 + 2] + t3 * sD2[cat * 4 + 3]);
+    a0 += __shfl_xor(a0, 1);
+    a0 += __shfl_xor(a0, 2);
+    a1 += __shfl_xor(a1, 1);
+    a1 += __shfl_xor(a1, 2);
+    a2 += __shfl_xor(a2, 1);
+    a2 += __shfl_xor(a2, 2);
+    if ((lane & 3) == 0) {
+      const double inv = 1.0 / fabs(a0);
+      const double d1 = a1 * inv, d2 = a2 * inv;
+      const double w = (double)sg.wgt[site];
+      accD1 += w * d1;
+      accD2 += w * (d2 - d1 * d1);
+    }
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    accD1 += __shfl_down(accD1, off);
+    accD2 += __shfl_down(accD2, off);
+  }
+  if (lane == 0) {
+    sRed[0][tid >> 6] = accD1;
+    sRed[1][tid >> 6] = accD2;
+  }
+  __syncthreads();
+  if (tid == 0) {
+    double s1 = 0, s2 = 0;
+#pragma unroll
+    for (int w = 0; w < NV_BLOCK / 64; w++) {
+      s1 += sRed[0][w];
+      s2 += sRed[1][w];
+    }
+    partials[2 * blockIdx.x] = s1;
+    partials[2 * blockIdx.x + 1] = s2;
+  }
+}
+
+__global__ __launch_bounds__(NV_BLOCK) void k_reduce_2_mseg(
+    const CSeg *__restrict__ segs, const double *__restrict__ partials,
+    const double *__restrict__ active) {
+  const CSeg sg = segs[blockIdx.x];
+  if (active[sg.part] == 0.0) return;
+  __shared__ double sred[2][NV_BLOCK];
+  const int tid = threadIdx.x;
+  double v1 = 0, v2 = 0;
+  for (int i = tid; i < sg.nBlocks; i += NV_BLOCK) {
+    v1 += partials[2 * (sg.blkBase + i)];
+    v2 += partials[2 * (sg.blkBase + i) + 1];
+  }
+  sred[0][tid] = v1;
+  sred[1][tid] = v2;
+  __syncthreads();
+  for (int off = NV_BLOCK / 2; off > 0; off >>= 1) {
+    if (tid < off) {
+      sred[0][tid] += sred[0][tid + off];
+      sred[1][tid] += sred[1][tid + off];
+    }
+    __syncthreads();
+  }
+  if (tid == 0) {
+    sg.out2[0] += sred[0][0];
+    sg.out2[1] += sred[1][0];
+  }
+}
+
+/* per-partition recursive scaler accumulation in post order
+ * (newviewGenericSpecial.c:1503); inc layout (op*numParts + part) */
+__global__ void k_scaler_finalize_mseg(FinMeta m,
+                                       const unsigned int *__restrict__ inc,
+                                       int numParts,
+                                       unsigned int *const *__restrict__ gsArr,
+                                       const double *__restrict__ active) {
+  const int b = blockIdx.x;
+  if (threadIdx.x != 0) return;
+  if (active[b] == 0.0) return;
+  unsigned int *gs = gsArr[b];
+  for (int e = 0; e < m.count; e++)
+    gs[m.p[e]] = gs[m.q[e]] + gs[m.r[e]] + inc[(size_t)(m.base + e) * numParts + b];
+}
+
+/* ---------------------------------------------------------------------------
+ * Multi-partition handle + executors (host side)
+ * ------------------------------------------------------------------------ */
+
+struct examl_hip_multi {
+  int states, numParts, maxOps;
+  std::vector<long> widths;
+  std::vector<double *> clv;
+  std::vector<long> clvStride;
+  std::vector<const unsigned char *> tips;
+  std::vector<long> tipStride;
+  std::vector<const int *> wgt;
+  std::vector<unsigned int *> scalers;
+  std::vector<const double *> EV, tipVec;
+  std::vector<int> partBlocks, partBlkBase; /* single-level geometry */
+  std::vector<int> segPart;                 /* seg idx -> partition */
+  int totalBlocks, numSegs;
+  double *d_pbuf;
+  unsigned int *d_inc;
+  double *d_stage; /* [zp | mod | active] per-call upload */
+  long zp_off, mod_off, act_off, stage_doubles;
+  double *d_partials;
+  double *d_sum_pool;
+  std::vector<double *> sum_base;
+  int *d_blk2part; /* single-level blk -> seg map */
+  unsigned int **d_gsArr;
+  char *d_callbuf; /* per-call [ESeg|SSeg|CSeg | diag | dtab | active] */
+  long callbuf_bytes;
+  struct Grp {
+    int tc, segOff, blkOff, grid;
+  };
+  struct Shape {
+    unsigned long long key;
+    MSeg *d_segs;
+    int *d_blk2seg;
+    std::vector<Grp> groups;
+    int numOps;
+  };
+  std::vector<Shape> shapes;
+};
+
+static int msz_of(int states) { return states + states * states + 4; }
+
+extern "C" int examl_hip_multi_create(
+    int states, int numParts, const long *widths, double *const *dev_clvs,
+    const long *clvStrides, const unsigned char *const *dev_tips,
+    const long *tipStrides, const int *const *dev_wgts,
+    unsigned int *const *dev_scalers, const double *const *dev_EVs,
+    const double *const *dev_tipVecs, int maxOps, void **out) {
+  if (states != 4) {
+    snprintf(g_err, sizeof(g_err), "multi_create: states %d not wired",
+             states);
+    return -1;
+  }
+  examl_hip_multi *h = new examl_hip_multi();
+  h->states = states;
+  h->numParts = numParts;
+  h->maxOps = maxOps;
+  const int span = 4 * states;
+  long sum_total = 0;
+  int blk = 0, nseg = 0;
+  for (int p = 0; p < numParts; p++) {
+    h->widths.push_back(widths[p]);
+    h->clv.push_back(dev_clvs[p]);
+    h->clvStride.push_back(clvStrides[p]);
+    h->tips.push_back(dev_tips[p]);
+    h->tipStride.push_back(tipStrides[p]);
+    h->wgt.push_back(dev_wgts[p]);
+    h->scalers.push_back(dev_scalers[p]);
+    h->EV.push_back(dev_EVs[p]);
+    h->tipVec.push_back(dev_tipVecs[p]);
+    const int nb = widths[p] > 0 ? grid_for(widths[p] * 4) : 0;
+    h->partBlocks.push_back(nb);
+    h->partBlkBase.push_back(blk);
+    if (widths[p] > 0) {
+      h->segPart.push_back(p);
+      nseg++;
+    }
+    blk += nb;
+    sum_total += widths[p] * span;
+  }
+  h->totalBlocks = blk;
+  h->numSegs = nseg;
+  const int PBLK = 8 * states * states;
+  hipError_t e = hipSuccess;
+#define MCHK(call)                                                           \
+  do {                                                                       \
+    e = (call);                                                              \
+    if (e != hipSuccess) {                                                   \
+      delete h;                                                              \
+      return set_err(e, #call);                                              \
+    }                                                                        \
+  } while (0)
+  MCHK(hipMalloc(&h->d_pbuf,
+                 (size_t)maxOps * numParts * PBLK * sizeof(double)));
+  MCHK(hipMalloc(&h->d_inc,
+                 (size_t)maxOps * numParts * sizeof(unsigned int)));
+  h->zp_off = 0;
+  h->mod_off = (long)maxOps * numParts * 2;
+  h->act_off = h->mod_off + (long)numParts * msz_of(states);
+  h->stage_doubles = h->act_off + numParts;
+  MCHK(hipMalloc(&h->d_stage, h->stage_doubles * sizeof(double)));
+  MCHK(hipMalloc(&h->d_partials,
+                 (size_t)2 * (blk > 0 ? blk : 1) * sizeof(double)));
+  MCHK(hipMalloc(&h->d_sum_pool,
+                 (size_t)(sum_total > 0 ? sum_total : 1) * sizeof(double)));
+  long off = 0;
+  for (int p = 0; p < numParts; p++) {
+    h->sum_base.push_back(h->d_sum_pool + off);
+    off += widths[p] * span;
+  }
+  /* single-level blk2seg */
+  {
+    std::vector<int> map;
+    map.reserve(blk);
+    int seg = 0;
+    for (int p = 0; p < numParts; p++) {
+      if (widths[p] == 0) continue;
+      for (int b = 0; b < h->partBlocks[p]; b++) map.push_back(seg);
+      seg++;
+    }
+    MCHK(hipMalloc(&h->d_blk2part, (size_t)(blk > 0 ? blk : 1) * sizeof(int)));
+    if (blk > 0)
+      MCHK(hipMemcpy(h->d_blk2part, map.data(), (size_t)blk * sizeof(int),
+                     hipMemcpyHostToDevice));
+  }
+  MCHK(hipMalloc(&h->d_gsArr, (size_t)numParts * sizeof(unsigned int *)));
+  MCHK(hipMemcpy(h->d_gsArr, h->scalers.data(),
+                 (size_t)numParts * sizeof(unsigned int *),
+                 hipMemcpyHostToDevice));
+  /* per-call scratch: segs + diag(4*S) + dtab(12*S) + active, in bytes */
+  const long segBytes =
+      (long)nseg *
+      (long)(sizeof(ESeg) > sizeof(SSeg)
+                 ? (sizeof(ESeg) > sizeof(CSeg) ? sizeof(ESeg) : sizeof(CSeg))
+                 : (sizeof(SSeg) > sizeof(CSeg) ? sizeof(SSeg)
+                                                : sizeof(CSeg)));
+  h->callbuf_bytes = segBytes +
+                     (long)numParts * 16 * states * sizeof(double) +
+                     (long)numParts * sizeof(double);
+  MCHK(hipMalloc(&h->d_callbuf, (size_t)h->callbuf_bytes));
+#undef MCHK
+  *out = h;
+  return 0;
+}
+
+extern "C" void examl_hip_multi_destroy(void *vh) {
+  examl_hip_multi *h = (examl_hip_multi *)vh;
+  if (!h) return;
+  examl_hip_graphs_clear(); /* cached graphs may bind these buffers */
+  hipFree(h->d_pbuf);
+  hipFree(h->d_inc);
+  hipFree(h->d_stage);
+  hipFree(h->d_partials);
+  hipFree(h->d_sum_pool);
+  hipFree(h->d_blk2part);
+  hipFree(h->d_gsArr);
+  hipFree(h->d_callbuf);
+  for (auto &sh : h->shapes) {
+    hipFree(sh.d_segs);
+    hipFree(sh.d_blk2seg);
+  }
+  delete h;
+}
+
+static unsigned long long multi_key(examl_hip_multi *h,
+                                    const examl_hip_trav_entry *ops,
+                                    int numOps, void *stream) {
+  unsigned long long x = 0x9e3779b97f4a7c15ULL ^ (unsigned long long)numOps;
+  auto mix = [&x](unsigned long long v) {
+    x ^= v + 0x9e3779b97f4a7c15ULL + (x << 6) + (x >> 2);
+  };
+  mix((unsigned long long)(uintptr_t)h);
+  mix((unsigned long long)(uintptr_t)stream);
+  mix(0x6d756c7469ULL); /* "multi" */
+  for (int e = 0; e < numOps; e++) {
+    mix(((unsigned long long)ops[e].tipCase << 48) ^
+        ((unsigned long long)(unsigned)ops[e].pNumber << 32) ^
+        ((unsigned long long)(unsigned)ops[e].x1Slot << 16) ^
+        (unsigned long long)(unsigned)ops[e].x2Slot);
+    mix(((unsigned long long)(unsigned)ops[e].qNumber << 32) ^
+        ((unsigned long long)(unsigned)ops[e].rNumber << 16) ^
+        (unsigned long long)(unsigned)ops[e].x3Slot);
+  }
+  return x;
+}
+
+/* build (or fetch) the level/tipCase-grouped segment tables for one
+ * traversal shape; segments and blk2seg live on the device, stable per
+ * shape, so cached hipGraphs can bind them. */
+static examl_hip_multi::Shape *multi_shape_get(
+    examl_hip_multi *h, const examl_hip_trav_entry *ops, int numOps,
+    unsigned long long key, int *rc) {
+  *rc = 0;
+  for (auto &sh : h->shapes)
+    if (sh.key == key) return &sh;
+
+  const int PBLK = 8 * h->states * h->states;
+  /* dependency levels: an op must run after any earlier op that wrote a
+   * CLV slot it reads; same level = independent = one fused launch */
+  std::vector<int> level(numOps, 0);
+  int maxSlot = 0;
+  for (int e = 0; e < numOps; e++)
+    if (ops[e].x3Slot > maxSlot) maxSlot = ops[e].x3Slot;
+  std::vector<int> writer((size_t)maxSlot + 1, -1);
+  for (int e = 0; e < numOps; e++) {
+    int lv = 0;
+    if (ops[e].tipCase == EXAML_INNER_INNER && ops[e].x1Slot >= 0 &&
+        ops[e].x1Slot <= maxSlot && writer[ops[e].x1Slot] >= 0)
+      lv = level[writer[ops[e].x1Slot]] + 1;
+    if (ops[e].tipCase != EXAML_TIP_TIP && ops[e].x2Slot >= 0 &&
+        ops[e].x2Slot <= maxSlot && writer[ops[e].x2Slot] >= 0 &&
+        level[writer[ops[e].x2Slot]] + 1 > lv)
+      lv = level[writer[ops[e].x2Slot]] + 1;
+    level[e] = lv;
+    writer[ops[e].x3Slot] = e;
+  }
+  int numLevels = 0;
+  for (int e = 0; e < numOps; e++)
+    if (level[e] + 1 > numLevels) numLevels = level[e] + 1;
+
+  examl_hip_multi::Shape sh;
+  sh.key = key;
+  sh.numOps = numOps;
+  std::vector<MSeg> segs;
+  std::vector<int> blk2seg;
+  for (int lv = 0; lv < numLevels; lv++) {
+    for (int tc = 0; tc < 3; tc++) {
+      examl_hip_multi::Grp g;
+      g.tc = tc;
+      g.segOff = (int)segs.size();
+      g.blkOff = (int)blk2seg.size();
+      for (int e = 0; e < numOps; e++) {
+        if (level[e] != lv || ops[e].tipCase != tc) continue;
+        for (int p = 0; p < h->numParts; p++) {
+          if (h->widths[p] == 0) continue;
+          MSeg s;
+          memset(&s, 0, sizeof(s));
+          s.x3 = h->clv[p] + (long)ops[e].x3Slot * h->clvStride[p];
+          if (tc == EXAML_TIP_TIP) {
+            s.t1 = h->tips[p] + (long)ops[e].x1Slot * h->tipStride[p];
+            s.t2 = h->tips[p] + (long)ops[e].x2Slot * h->tipStride[p];
+          } else if (tc == EXAML_TIP_INNER) {
+            s.t1 = h->tips[p] + (long)ops[e].x1Slot * h->tipStride[p];
+            s.x2 = h->clv[p] + (long)ops[e].x2Slot * h->clvStride[p];
+          } else {
+            s.x1 = h->clv[p] + (long)ops[e].x1Slot * h->clvStride[p];
+            s.x2 = h->clv[p] + (long)ops[e].x2Slot * h->clvStride[p];
+          }
+          s.P = h->d_pbuf + (long)(e * h->numParts + p) * PBLK;
+          s.wgt = h->wgt[p];
+          s.inc = h->d_inc + (size_t)e * h->numParts + p;
+          s.EV = h->EV[p];
+          s.tipVec = h->tipVec[p];
+          s.n = h->widths[p];
+          s.part = p;
+          s.blkBase = (int)blk2seg.size() - g.blkOff;
+          s.nBlocks = h->partBlocks[p];
+          const int si = (int)segs.size();
+          segs.push_back(s);
+          for (int b = 0; b < s.nBlocks; b++) blk2seg.push_back(si);
+        }
+      }
+      g.grid = (int)blk2seg.size() - g.blkOff;
+      if (g.grid > 0) sh.groups.push_back(g);
+    }
+  }
+  hipError_t e1 = hipMalloc(&sh.d_segs, segs.size() * sizeof(MSeg));
+  hipError_t e2 =
+      hipMalloc(&sh.d_blk2seg, blk2seg.size() * sizeof(int));
+  if (e1 != hipSuccess || e2 != hipSuccess) {
+    *rc = set_err(e1 != hipSuccess ? e1 : e2, "shape alloc");
+    return nullptr;
+  }
+  hipMemcpy(sh.d_segs, segs.data(), segs.size() * sizeof(MSeg),
+            hipMemcpyHostToDevice);
+  hipMemcpy(sh.d_blk2seg, blk2seg.data(), blk2seg.size() * sizeof(int),
+            hipMemcpyHostToDevice);
+  if (h->shapes.size() >= 64) {
+    for (auto &old : h->shapes) {
+      hipFree(old.d_segs);
+      hipFree(old.d_blk2seg);
+    }
+    h->shapes.clear();
+    examl_hip_graphs_clear();
+  }
+  h->shapes.push_back(sh);
+  return &h->shapes.back();
+}
+
+extern "C" int examl_hip_newview_traversal_multi(
+    void *vh, const examl_hip_trav_entry *ops, int numOps,
+    const double *const *EIGNs, const double *const *EIs,
+    const double *const *rates, const unsigned char *activeMask,
+    const double *qzOv, const double *rzOv, void *stream) {
+  examl_hip_multi *h = (examl_hip_multi *)vh;
+  if (numOps <= 0) return 0;
+  if (numOps > h->maxOps) {
+    snprintf(g_err, sizeof(g_err), "multi traversal: numOps %d > maxOps %d",
+             numOps, h->maxOps);
+    return -1;
+  }
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  const int S = h->states, NP = h->numParts, MSZ = msz_of(S);
+
+  const unsigned long long key = multi_key(h, ops, numOps, stream);
+  int src = 0;
+  examl_hip_multi::Shape *shape = multi_shape_get(h, ops, numOps, key, &src);
+  if (!shape) return src;
+
+  /* pinned stage: log-z pairs, per-partition model vectors, active flags */
+  HostPSlot *slot = hostP_get(h->d_stage, h->stage_doubles);
+  double *st = slot->buf;
+  for (int e = 0; e < numOps; e++)
+    for (int p = 0; p < NP; p++) {
+      double qz = qzOv ? qzOv[(size_t)e * NP + p] : ops[e].qz;
+      double rz = rzOv ? rzOv[(size_t)e * NP + p] : ops[e].rz;
+      st[h->zp_off + ((size_t)e * NP + p) * 2] =
+          (qz > ZMIN) ? log(qz) : log(ZMIN);
+      st[h->zp_off + ((size_t)e * NP + p) * 2 + 1] =
+          (rz > ZMIN) ? log(rz) : log(ZMIN);
+    }
+  for (int p = 0; p < NP; p++) {
+    double *m = st + h->mod_off + (size_t)p * MSZ;
+    memcpy(m, EIGNs[p], S * sizeof(double));
+    memcpy(m + S, EIs[p], S * S * sizeof(double));
+    memcpy(m + S + S * S, rates[p], 4 * sizeof(double));
+    st[h->act_off + p] =
+        (h->widths[p] > 0 && (!activeMask || activeMask[p])) ? 1.0 : 0.0;
+  }
+  const double *d_active = h->d_stage + h->act_off;
+
+  const bool want_graph = g_use_graphs && !g_prof_on && s != nullptr;
+  bool capturing = false;
+  if (want_graph) {
+    hipGraphExec_t exec = trav_graph_find(key);
+    if (exec) {
+      CHK(hipGraphLaunch(exec, s));
+      return 0;
+    }
+    capturing =
+        hipStreamBeginCapture(s, hipStreamCaptureModeThreadLocal) ==
+        hipSuccess;
+    (void)hipGetLastError();
+  }
+
+  int rc = 0;
+  do {
+    hipError_t err =
+        hipMemcpyAsync(h->d_stage, st,
+                       ((size_t)numOps * NP * 2) * sizeof(double), /* zp */
+                       hipMemcpyHostToDevice, s);
+    if (err != hipSuccess) { rc = set_err(err, "zp upload"); break; }
+    err = hipMemcpyAsync(h->d_stage + h->mod_off, st + h->mod_off,
+                         ((size_t)NP * MSZ + NP) * sizeof(double),
+                         hipMemcpyHostToDevice, s);
+    if (err != hipSuccess) { rc = set_err(err, "mod upload"); break; }
+    hipEventRecord(slot->ev, s);
+    slot->ev_valid = true;
+    err = hipMemsetAsync(h->d_inc, 0,
+                         (size_t)numOps * NP * sizeof(unsigned int), s);
+    if (err != hipSuccess) { rc = set_err(err, "inc memset"); break; }
+
+    hipLaunchKernelGGL((k_make_p_mseg<4>), dim3(numOps * NP), dim3(128), 0,
+                       s, h->d_stage + h->zp_off, h->d_stage + h->mod_off,
+                       NP, h->d_pbuf);
+    err = hipGetLastError();
+    if (err != hipSuccess) { rc = set_err(err, "make_p launch"); break; }
+
+    for (auto &g : shape->groups) {
+      hipEvent_t ev_a = nullptr, ev_b = nullptr;
+      if (g_prof_on) {
+        prof_begin(&ev_a, &ev_b);
+        hipEventRecord(ev_a, s);
+      }
+      switch (g.tc) {
+      case EXAML_TIP_TIP:
+        hipLaunchKernelGGL((k_newview_dna_mseg<EXAML_TIP_TIP>), dim3(g.grid),
+                           dim3(NV_BLOCK), 0, s, shape->d_segs,
+                           shape->d_blk2seg + g.blkOff, d_active);
+        break;
+      case EXAML_TIP_INNER:
+        hipLaunchKernelGGL((k_newview_dna_mseg<EXAML_TIP_INNER>),
+                           dim3(g.grid), dim3(NV_BLOCK), 0, s, shape->d_segs,
+                           shape->d_blk2seg + g.blkOff, d_active);
+        break;
+      default:
+        hipLaunchKernelGGL((k_newview_dna_mseg<EXAML_INNER_INNER>),
+                           dim3(g.grid), dim3(NV_BLOCK), 0, s, shape->d_segs,
+                           shape->d_blk2seg + g.blkOff, d_active);
+      }
+      err = hipGetLastError();
+      if (err != hipSuccess) { rc = set_err(err, "mseg launch"); break; }
+      if (g_prof_on) {
+        hipEventRecord(ev_b, s);
+        g_prof_pend.push_back({ev_a, ev_b, g.tc});
+        if (g_prof_pend.size() > 2048) prof_flush();
+      }
+    }
+    if (rc != 0) break;
+
+    for (int base = 0; base < numOps && rc == 0; base += FIN_CHUNK) {
+      FinMeta m;
+      m.count = (numOps - base < FIN_CHUNK) ? (numOps - base) : FIN_CHUNK;
+      m.base = base;
+      for (int e = 0; e < m.count; e++) {
+        m.p[e] = ops[base + e].pNumber;
+        m.q[e] = ops[base + e].qNumber;
+        m.r[e] = ops[base + e].rNumber;
+      }
+      hipLaunchKernelGGL(k_scaler_finalize_mseg, dim3(NP), dim3(64), 0, s, m,
+                         h->d_inc, NP, h->d_gsArr, d_active);
+      err = hipGetLastError();
+      if (err != hipSuccess) rc = set_err(err, "finalize mseg");
+    }
+  } while (0);
+
+  if (capturing) {
+    hipGraph_t graph = nullptr;
+    hipError_t err = hipStreamEndCapture(s, &graph);
+    if (rc != 0) {
+      if (graph) hipGraphDestroy(graph);
+      return rc;
+    }
+    hipGraphExec_t exec = nullptr;
+    if (err == hipSuccess) {
+      err = hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0);
+      hipGraphDestroy(graph);
+    }
+    if (err != hipSuccess) {
+      (void)hipGetLastError();
+      g_use_graphs = false;
+      return examl_hip_newview_traversal_multi(vh, ops, numOps, EIGNs, EIs,
+                                               rates, activeMask, qzOv, rzOv,
+                                               stream);
+    }
+    trav_graph_store(key, exec);
+    CHK(hipGraphLaunch(exec, s));
+  }
+  return rc;
+}
+
+extern "C" int examl_hip_evaluate_root_multi(
+    void *vh, int rootTipCase, int pNumber, int qNumber, int x1Slot,
+    int x2Slot, int tipSlot, const double *zs, int zPerPart,
+    const double *const *EIGNs, const double *const *rates,
+    const unsigned char *activeMask, double *dev_lnl, void *stream) {
+  examl_hip_multi *h = (examl_hip_multi *)vh;
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  const int S = h->states, NP = h->numParts;
+  const int dsz = 4 * S;
+
+  /* per-call device block: [ESeg x numSegs | diag x NP | active x NP] */
+  const long diagOff = (long)h->numSegs * sizeof(ESeg);
+  const long actOff = diagOff + (long)NP * dsz * sizeof(double);
+  const long total = actOff + (long)NP * sizeof(double);
+  HostPSlot *slot =
+      hostP_get(h->d_callbuf, (total + 7) / 8 + 16);
+  char *stb = (char *)slot->buf;
+  ESeg *es = (ESeg *)stb;
+  double *diag = (double *)(stb + diagOff);
+  double *act = (double *)(stb + actOff);
+  double *d_diag = (double *)(h->d_callbuf + diagOff);
+  int si = 0;
+  for (int p = 0; p < NP; p++) {
+    act[p] = (h->widths[p] > 0 && (!activeMask || activeMask[p])) ? 1.0 : 0.0;
+    if (h->widths[p] == 0) continue;
+    const double z = zPerPart ? zs[p] : zs[0];
+    examl_host_calc_diagptable(z, S, 4, rates[p], EIGNs[p], diag + p * dsz);
+    ESeg *e = &es[si];
+    memset(e, 0, sizeof(*e));
+    if (rootTipCase == EXAML_TIP_INNER) {
+      e->x1 = h->tipVec[p]; /* staged as sTV by the TIP kernel */
+      e->t1 = h->tips[p] + (long)tipSlot * h->tipStride[p];
+      e->x2 = h->clv[p] + (long)x2Slot * h->clvStride[p];
+    } else {
+      e->x1 = h->clv[p] + (long)x1Slot * h->clvStride[p];
+      e->x2 = h->clv[p] + (long)x2Slot * h->clvStride[p];
+    }
+    e->diag = d_diag + p * dsz;
+    e->wgt = h->wgt[p];
+    e->gsP = h->scalers[p] + pNumber;
+    e->gsQ = h->scalers[p] + qNumber;
+    e->lnlOut = dev_lnl + p;
+    e->n = h->widths[p];
+    e->blkBase = h->partBlkBase[p];
+    e->nBlocks = h->partBlocks[p];
+    e->part = p;
+    si++;
+  }
+  CHK(hipMemcpyAsync(h->d_callbuf, stb, (size_t)total,
+                     hipMemcpyHostToDevice, s));
+  hipEventRecord(slot->ev, s);
+  slot->ev_valid = true;
+  const double *d_active = (const double *)(h->d_callbuf + actOff);
+  const ESeg *d_es = (const ESeg *)h->d_callbuf;
+  if (rootTipCase == EXAML_TIP_INNER)
+    hipLaunchKernelGGL((k_evaluate_dna_mseg<true>), dim3(h->totalBlocks),
+                       dim3(NV_BLOCK), 0, s, d_es, h->d_blk2part, d_active,
+                       h->d_partials);
+  else
+    hipLaunchKernelGGL((k_evaluate_dna_mseg<false>), dim3(h->totalBlocks),
+                       dim3(NV_BLOCK), 0, s, d_es, h->d_blk2part, d_active,
+                       h->d_partials);
+  CHK(hipGetLastError());
+  hipLaunchKernelGGL(k_reduce_lnl_mseg, dim3(h->numSegs), dim3(NV_BLOCK), 0,
+                     s, d_es, h->d_partials, d_active, log(MINLIKELIHOOD));
+  CHK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int examl_hip_sum_root_multi(void *vh, int rootTipCase,
+                                        int x1Slot, int x2Slot, int tipSlot,
+                                        int tipSlot2,
+                                        const unsigned char *activeMask,
+                                        void *stream) {
+  examl_hip_multi *h = (examl_hip_multi *)vh;
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  const int NP = h->numParts;
+  const long actOff = (long)h->numSegs * sizeof(SSeg);
+  const long total = actOff + (long)NP * sizeof(double);
+  HostPSlot *slot = hostP_get(h->d_callbuf, (total + 7) / 8 + 16);
+  char *stb = (char *)slot->buf;
+  SSeg *ss = (SSeg *)stb;
+  double *act = (double *)(stb + actOff);
+  int si = 0;
+  for (int p = 0; p < NP; p++) {
+    act[p] = (h->widths[p] > 0 && (!activeMask || activeMask[p])) ? 1.0 : 0.0;
+    if (h->widths[p] == 0) continue;
+    SSeg *e = &ss[si];
+    memset(e, 0, sizeof(*e));
+    if (rootTipCase == EXAML_TIP_TIP) {
+      e->t1 = h->tips[p] + (long)tipSlot * h->tipStride[p];
+      e->t2 = h->tips[p] + (long)tipSlot2 * h->tipStride[p];
+    } else if (rootTipCase == EXAML_TIP_INNER) {
+      e->t1 = h->tips[p] + (long)tipSlot * h->tipStride[p];
+      e->x2 = h->clv[p] + (long)x2Slot * h->clvStride[p];
+    } else {
+      e->x1 = h->clv[p] + (long)x1Slot * h->clvStride[p];
+      e->x2 = h->clv[p] + (long)x2Slot * h->clvStride[p];
+    }
+    e->sum = h->sum_base[p];
+    e->tipVec = h->tipVec[p];
+    e->n = h->widths[p];
+    e->blkBase = h->partBlkBase[p];
+    e->nBlocks = h->partBlocks[p];
+    e->part = p;
+    si++;
+  }
+  CHK(hipMemcpyAsync(h->d_callbuf, stb, (size_t)total,
+                     hipMemcpyHostToDevice, s));
+  hipEventRecord(slot->ev, s);
+  slot->ev_valid = true;
+  const double *d_active = (const double *)(h->d_callbuf + actOff);
+  const SSeg *d_ss = (const SSeg *)h->d_callbuf;
+  switch (rootTipCase) {
+  case EXAML_TIP_TIP:
+    hipLaunchKernelGGL((k_sum_dna_mseg<EXAML_TIP_TIP>), dim3(h->totalBlocks),
+                       dim3(NV_BLOCK), 0, s, d_ss, h->d_blk2part, d_active);
+    break;
+  case EXAML_TIP_INNER:
+    hipLaunchKernelGGL((k_sum_dna_mseg<EXAML_TIP_INNER>),
+                       dim3(h->totalBlocks), dim3(NV_BLOCK), 0, s, d_ss,
+                       h->d_blk2part, d_active);
+    break;
+  default:
+    hipLaunchKernelGGL((k_sum_dna_mseg<EXAML_INNER_INNER>),
+                       dim3(h->totalBlocks), dim3(NV_BLOCK), 0, s, d_ss,
+                       h->d_blk2part, d_active);
+  }
+  CHK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int examl_hip_core_root_multi(
+    void *vh, const double *lzs, int lzPerPart, const double *const *EIGNs,
+    const double *const *rates, const unsigned char *activeMask,
+    double *dev_out2, void *stream) {
+  examl_hip_multi *h = (examl_hip_multi *)vh;
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  const int S = h->states, NP = h->numParts;
+  const int tsz = 12 * S;
+  const long dtabOff = (long)h->numSegs * sizeof(CSeg);
+  const long actOff = dtabOff + (long)NP * tsz * sizeof(double);
+  const long total = actOff + (long)NP * sizeof(double);
+  HostPSlot *slot = hostP_get(h->d_callbuf, (total + 7) / 8 + 16);
+  char *stb = (char *)slot->buf;
+  CSeg *cs = (CSeg *)stb;
+  double *dtab = (double *)(stb + dtabOff);
+  double *act = (double *)(stb + actOff);
+  double *d_dtab = (double *)(h->d_callbuf + dtabOff);
+  int si = 0;
+  for (int p = 0; p < NP; p++) {
+    act[p] = (h->widths[p] > 0 && (!activeMask || activeMask[p])) ? 1.0 : 0.0;
+    if (h->widths[p] == 0) continue;
+    const double lz = lzPerPart ? lzs[p] : lzs[0];
+    examl_host_core_dtables_dna(EIGNs[p], rates[p], lz, dtab + p * tsz);
+    CSeg *e = &cs[si];
+    memset(e, 0, sizeof(*e));
+    e->sum = h->sum_base[p];
+    e->dtab = d_dtab + p * tsz;
+    e->wgt = h->wgt[p];
+    e->out2 = dev_out2 + 2 * p;
+    e->n = h->widths[p];
+    e->blkBase = h->partBlkBase[p];
+    e->nBlocks = h->partBlocks[p];
+    e->part = p;
+    si++;
+  }
+  CHK(hipMemcpyAsync(h->d_callbuf, stb, (size_t)total,
+                     hipMemcpyHostToDevice, s));
+  hipEventRecord(slot->ev, s);
+  slot->ev_valid = true;
+  const double *d_active = (const double *)(h->d_callbuf + actOff);
+  const CSeg *d_cs = (const CSeg *)h->d_callbuf;
+  hipLaunchKernelGGL(k_core_dna_mseg, dim3(h->totalBlocks), dim3(NV_BLOCK),
+                     0, s, d_cs, h->d_blk2part, d_active, h->d_partials);
+  CHK(hipGetLastError());
+  hipLaunchKernelGGL(k_reduce_2_mseg, dim3(h->numSegs), dim3(NV_BLOCK), 0, s,
+                     d_cs, h->d_partials, d_active);
+  CHK(hipGetLastError());
+  return 0;
+}

@@ -749,3 +749,189 @@ class ProtCatEngine(DnaGammaEngine):
             _vp(self.d_partials), _vp(self.d_out2), self._stream()),
             "core_root_prot_cat")
         return self.d_out2
+
+
+class MultiDnaEngine:
+    """Fused multi-partition DNA GTRGAMMA engine: drives ALL partitions'
+    per-op work through one kernel launch per (traversal level, tipCase)
+    via the mseg executors — the GPU answer to the per-partition dispatch
+    loops of newviewIterative (newviewGenericSpecial.c:1064) and execCore
+    (makenewzGenericSpecial.c:885) that made partitioned shapes (config 3,
+    140.model) launch-bound.  Wraps per-partition DnaGammaEngine buffers;
+    P matrices are computed on device, so results agree with the
+    per-partition engines to <=1e-11 relative (device exp vs libm), not
+    bit-exactly."""
+
+    def __init__(self, engines):
+        assert engines and all(e.states == 4 for e in engines)
+        self.engines = engines
+        self.device = engines[0].device
+        NP = self.NP = len(engines)
+        self.max_ops = max(e._max_ops for e in engines)
+        self.d_lnl = torch.zeros(NP, dtype=torch.float64,
+                                 device=self.device)
+        self.d_out2 = torch.zeros(2 * NP, dtype=torch.float64,
+                                  device=self.device)
+
+        def parr(vals):
+            return (ctypes.c_void_p * NP)(*vals)
+
+        h = ctypes.c_void_p()
+        check(lib().examl_hip_multi_create(
+            4, NP, (ctypes.c_long * NP)(*[e.width for e in engines]),
+            parr([e.d_clv.data_ptr() for e in engines]),
+            (ctypes.c_long * NP)(*[e.width * 16 for e in engines]),
+            parr([e.d_tips.data_ptr() for e in engines]),
+            (ctypes.c_long * NP)(*[e.width for e in engines]),
+            parr([e.d_wgt.data_ptr() for e in engines]),
+            parr([e.d_scalers.data_ptr() for e in engines]),
+            parr([e.d_EV.data_ptr() for e in engines]),
+            parr([e.d_tipVector.data_ptr() for e in engines]),
+            self.max_ops, ctypes.byref(h)), "multi_create")
+        self.h = h
+
+    def __del__(self):
+        try:
+            if getattr(self, "h", None):
+                lib().examl_hip_multi_destroy(self.h)
+        except Exception:
+            pass
+
+    def _stream(self):
+        if self.device.type == "cuda":
+            return ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
+        return ctypes.c_void_p(0)
+
+    def sync(self):
+        if self.device.type == "cuda":
+            torch.cuda.synchronize(self.device)
+
+    def _models(self, attr):
+        return (ctypes.c_void_p * self.NP)(
+            *[getattr(e.model, attr).ctypes.data for e in self.engines])
+
+    def _active(self, active):
+        if active is None:
+            return None
+        return (ctypes.c_ubyte * self.NP)(*[1 if a else 0 for a in active])
+
+    def newview_traversal(self, entries, active=None, qz_ov=None,
+                          rz_ov=None):
+        if not entries:
+            return
+        assert len(entries) <= self.max_ops
+        arr = (TravEntry * len(entries))(*entries)
+
+        def dvec(v):
+            if v is None:
+                return None
+            a = np.ascontiguousarray(v, dtype=np.float64)
+            assert a.size == len(entries) * self.NP
+            return a
+
+        qz_a, rz_a = dvec(qz_ov), dvec(rz_ov)
+        check(lib().examl_hip_newview_traversal_multi(
+            self.h, ctypes.cast(arr, ctypes.c_void_p), len(entries),
+            self._models("EIGN"), self._models("EI"),
+            self._models("gammaRates"), self._active(active),
+            _np_vp(qz_a) if qz_a is not None else None,
+            _np_vp(rz_a) if rz_a is not None else None,
+            self._stream()), "newview_traversal_multi")
+
+    def evaluate_root(self, tree, p, q, z, active=None, all_reduce=False):
+        """z: scalar (joint) or per-partition sequence (-M).  Returns the
+        per-partition lnL device vector (sum for the total)."""
+        e0 = self.engines[0]
+        tc, x1s, x2s, tslot, _, pn, qn = e0._root_case(tree, p, q)
+        self.d_lnl.zero_()
+        zv = np.atleast_1d(np.asarray(z, dtype=np.float64))
+        per_part = 1 if zv.size > 1 else 0
+        check(lib().examl_hip_evaluate_root_multi(
+            self.h, tc, pn, qn, x1s, x2s, tslot, _np_vp(zv), per_part,
+            self._models("EIGN"), self._models("gammaRates"),
+            self._active(active), _vp(self.d_lnl), self._stream()),
+            "evaluate_root_multi")
+        if all_reduce and torch.distributed.is_initialized():
+            torch.distributed.all_reduce(self.d_lnl)
+        return self.d_lnl
+
+    def full_lnl(self, tree, root_edge=None, all_reduce=False):
+        entries, (p, q, z) = tree.full_traversal(root_edge)
+        self.newview_traversal(entries)
+        return self.evaluate_root(tree, p, q, z, all_reduce=all_reduce)
+
+    def sum_root(self, tree, p, q, active=None):
+        e0 = self.engines[0]
+        p_tip, q_tip = tree.is_tip(p), tree.is_tip(q)
+        if p_tip and q_tip:
+            tc, x1s, x2s, t1, t2 = TIP_TIP, -1, -1, p, q
+        elif q_tip:
+            tc, x1s, x2s, t1, t2 = TIP_INNER, -1, tree.clv_slot(p), q, -1
+        elif p_tip:
+            tc, x1s, x2s, t1, t2 = TIP_INNER, -1, tree.clv_slot(q), p, -1
+        else:
+            tc, x1s, x2s, t1, t2 = (INNER_INNER, tree.clv_slot(p),
+                                    tree.clv_slot(q), -1, -1)
+        del e0
+        check(lib().examl_hip_sum_root_multi(
+            self.h, tc, x1s, x2s, t1, t2, self._active(active),
+            self._stream()), "sum_root_multi")
+
+    def core_derivs_vec(self, lz, active=None):
+        """Per-partition {dlnL, d2lnL} (device vector 2*NP); lz scalar or
+        per-partition."""
+        self.d_out2.zero_()
+        lzv = np.atleast_1d(np.asarray(lz, dtype=np.float64))
+        check(lib().examl_hip_core_root_multi(
+            self.h, _np_vp(lzv), 1 if lzv.size > 1 else 0,
+            self._models("EIGN"), self._models("gammaRates"),
+            self._active(active), _vp(self.d_out2), self._stream()),
+            "core_root_multi")
+        return self.d_out2
+
+    def core_derivs(self, lz, all_reduce=False):
+        v = self.core_derivs_vec(lz)
+        if all_reduce and torch.distributed.is_initialized():
+            torch.distributed.all_reduce(v)
+        host = v.cpu().numpy().reshape(self.NP, 2)
+        return float(host[:, 0].sum()), float(host[:, 1].sum())
+
+    def makenewz(self, tree, p, q, z0, maxiter=64, all_reduce=False):
+        """Joint-branch NR (topLevelMakenewz, makenewzGenericSpecial.c:1133)
+        over all partitions with fused sum/core launches."""
+        self.sum_root(tree, p, q)
+        z = float(z0)
+        zprev = z
+        zstep = (1.0 - ZMAX) * z + ZMIN
+        curvat_ok = True
+        outer_converged = False
+        it = maxiter
+        while not outer_converged:
+            if curvat_ok:
+                curvat_ok = False
+                zprev = z
+                zstep = (1.0 - ZMAX) * z + ZMIN
+            z = min(max(z, ZMIN), ZMAX)
+            dlnL, d2lnL = self.core_derivs(math.log(z),
+                                           all_reduce=all_reduce)
+            if (d2lnL >= 0.0) and (z < ZMAX):
+                zprev = z = 0.37 * z + 0.63
+                continue
+            curvat_ok = True
+            if d2lnL < 0.0:
+                tantmp = -dlnL / d2lnL
+                if tantmp < 100:
+                    z *= math.exp(tantmp)
+                    z = max(z, ZMIN)
+                    z = min(z, 0.25 * zprev + 0.75)
+                else:
+                    z = 0.25 * zprev + 0.75
+            z = min(z, ZMAX)
+            it -= 1
+            if abs(z - zprev) > zstep:
+                if it < -20:
+                    z = float(z0)
+                    outer_converged = True
+            else:
+                outer_converged = True
+        return z

@@ -499,6 +499,46 @@ int examl_hip_core_root_prot_cat(
     int numCats, double lz, const int *dev_wgt, const int *dev_cptr,
     double *dev_dtab, double *dev_partials, double *dev_out2, void *stream);
 
+/* ---------------------------------------------------------------------------
+ * Multi-partition fused executors: one launch per (traversal level x
+ * tipCase) covering ALL partitions, replacing the per-(partition, op)
+ * launch loop for partitioned data (newviewIterative's partition loop,
+ * newviewGenericSpecial.c:1064; execCore's, makenewzGenericSpecial.c:885).
+ * P matrices are computed ON DEVICE from per-partition EIGN/EI/rates and
+ * the ops' branch lengths, so the fused path agrees with the
+ * single-partition executors to <=1e-11 relative (device exp vs libm),
+ * not bit-exact.  activeMask (host, numParts bytes, NULL = all) carries
+ * the executeModel gating; masked partitions' CLVs/scalers/outputs are
+ * untouched.  qzOv/rzOv (numOps*numParts, NULL = use ops[].qz/rz) carry
+ * per-partition branch lengths under -M.
+ * ------------------------------------------------------------------------ */
+
+int examl_hip_multi_create(
+    int states, int numParts, const long *widths, double *const *dev_clvs,
+    const long *clvStrides, const unsigned char *const *dev_tips,
+    const long *tipStrides, const int *const *dev_wgts,
+    unsigned int *const *dev_scalers, const double *const *dev_EVs,
+    const double *const *dev_tipVecs, int maxOps, void **out);
+void examl_hip_multi_destroy(void *h);
+int examl_hip_newview_traversal_multi(
+    void *h, const examl_hip_trav_entry *ops, int numOps,
+    const double *const *EIGNs, const double *const *EIs,
+    const double *const *gammaRates, const unsigned char *activeMask,
+    const double *qzOv, const double *rzOv, void *stream);
+int examl_hip_evaluate_root_multi(
+    void *h, int rootTipCase, int pNumber, int qNumber, int x1Slot,
+    int x2Slot, int tipSlot, const double *zs, int zPerPart,
+    const double *const *EIGNs, const double *const *gammaRates,
+    const unsigned char *activeMask, double *dev_lnl /* numParts, caller
+    zeroes */, void *stream);
+int examl_hip_sum_root_multi(void *h, int rootTipCase, int x1Slot,
+                             int x2Slot, int tipSlot, int tipSlot2,
+                             const unsigned char *activeMask, void *stream);
+int examl_hip_core_root_multi(
+    void *h, const double *lzs, int lzPerPart, const double *const *EIGNs,
+    const double *const *gammaRates, const unsigned char *activeMask,
+    double *dev_out2 /* 2*numParts, caller zeroes */, void *stream);
+
 #ifdef __cplusplus
 }
 #endif
