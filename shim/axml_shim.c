@@ -102,6 +102,20 @@ static ShimPart *S = NULL;
 static int S_n = 0;
 static int S_mxtips = 0;
 
+/* Fused multi-partition path (examl_hip_multi_*): one launch per
+ * (traversal level x tipCase) covering all partitions, replacing the
+ * per-partition loop below when every partition is DNA GAMMA.  This is
+ * the C-side batching of newviewIterative's partition loop
+ * (newviewGenericSpecial.c:1064) and of execCore's
+ * (makenewzGenericSpecial.c:885). */
+static void *g_multi = NULL;
+static double *g_d_lnl_vec = NULL, *g_h_lnl_vec = NULL;
+static double *g_d_out2_vec = NULL, *g_h_out2_vec = NULL;
+static examl_hip_trav_entry *g_mops = NULL;
+static double *g_qzov = NULL, *g_rzov = NULL;
+static unsigned char *g_amask = NULL;
+static const double **g_eign = NULL, **g_ei = NULL, **g_rates = NULL;
+
 static void *dmalloc(size_t bytes)
 {
   void *p = NULL;
@@ -197,6 +211,57 @@ static void shim_init(tree *tr)
   if (tr->saveMemory) {
     fprintf(stderr, "examl-HIP shim: -S not wired in the C shim\n");
     MPI_Abort(MPI_COMM_WORLD, 1);
+  }
+
+  /* all-DNA runs go through the fused multi-partition executors */
+  {
+    int allDna = 1;
+    for (m = 0; m < S_n; m++)
+      if (S[m].states != 4) allDna = 0;
+    if (allDna && S_n >= 1) {
+      long widths[S_n], clvStrides[S_n], tipStrides[S_n];
+      double *clvs[S_n];
+      const unsigned char *tips[S_n];
+      const int *wgts[S_n];
+      unsigned int *scalers[S_n];
+      const double *EVs[S_n], *tipVecs[S_n];
+      int maxOps = 0;
+      for (m = 0; m < S_n; m++) {
+        widths[m] = S[m].width;
+        clvs[m] = S[m].d_clv;
+        clvStrides[m] = S[m].width * S[m].span;
+        tips[m] = S[m].d_tips;
+        tipStrides[m] = S[m].width;
+        wgts[m] = S[m].d_wgt;
+        scalers[m] = S[m].d_scalers;
+        EVs[m] = S[m].d_EV;
+        tipVecs[m] = S[m].d_tipVector;
+        if (S[m].maxOps > maxOps) maxOps = S[m].maxOps;
+      }
+      if (examl_hip_multi_create(4, S_n, widths, clvs, clvStrides, tips,
+                                 tipStrides, wgts, scalers, EVs, tipVecs,
+                                 maxOps, &g_multi) != 0) {
+        fprintf(stderr, "examl-HIP shim: multi_create failed (%s); using "
+                        "per-partition path\n",
+                examl_hip_last_error_string());
+        g_multi = NULL;
+      } else {
+        g_d_lnl_vec = (double *)dmalloc(S_n * sizeof(double));
+        g_d_out2_vec = (double *)dmalloc(2 * S_n * sizeof(double));
+        HIP_OK(hipHostMalloc((void **)&g_h_lnl_vec, S_n * sizeof(double),
+                             0));
+        HIP_OK(hipHostMalloc((void **)&g_h_out2_vec,
+                             2 * S_n * sizeof(double), 0));
+        g_mops = (examl_hip_trav_entry *)malloc(
+            maxOps * sizeof(examl_hip_trav_entry));
+        g_qzov = (double *)malloc((size_t)maxOps * S_n * sizeof(double));
+        g_rzov = (double *)malloc((size_t)maxOps * S_n * sizeof(double));
+        g_amask = (unsigned char *)malloc(S_n);
+        g_eign = (const double **)malloc(S_n * sizeof(double *));
+        g_ei = (const double **)malloc(S_n * sizeof(double *));
+        g_rates = (const double **)malloc(S_n * sizeof(double *));
+      }
+    }
   }
 }
 
@@ -328,11 +393,69 @@ static int build_ops(tree *tr, int m, int startIndex,
  * launch per post-order entry, device-side recursive scaler accumulation.
  * ------------------------------------------------------------------------ */
 
+/* gather per-partition host model pointers + the executeModel mask */
+static void multi_prep(tree *tr, int needWidth)
+{
+  int m;
+  for (m = 0; m < tr->NumberOfModels; m++) {
+    pInfo *pd = &tr->partitionData[m];
+    g_eign[m] = pd->EIGN;
+    g_ei[m] = pd->EI;
+    g_rates[m] = pd->gammaRates;
+    g_amask[m] =
+        (tr->td[0].executeModel[m] && (!needWidth || S[m].width > 0)) ? 1
+                                                                      : 0;
+    if (S[m].width > 0) upload_model(&S[m], pd);
+  }
+}
+
 void newviewIterative(tree *tr, int startIndex)
 {
   int m;
   shim_init(tr);
   if (tr->td[0].count - startIndex <= 0) return;
+
+  if (g_multi) {
+    /* fused path: one launch per (level x tipCase) over all partitions */
+    int n = 0, i;
+    for (i = startIndex; i < tr->td[0].count; i++) {
+      traversalInfo *ti = &tr->td[0].ti[i];
+      examl_hip_trav_entry *e = &g_mops[n];
+      e->tipCase = ti->tipCase;
+      e->pNumber = ti->pNumber;
+      e->qNumber = ti->qNumber;
+      e->rNumber = ti->rNumber;
+      e->x3Slot = ti->pNumber - tr->mxtips - 1;
+      switch (ti->tipCase) {
+        case TIP_TIP:
+          e->x1Slot = ti->qNumber;
+          e->x2Slot = ti->rNumber;
+          break;
+        case TIP_INNER:
+          e->x1Slot = ti->qNumber;
+          e->x2Slot = ti->rNumber - tr->mxtips - 1;
+          break;
+        default:
+          e->x1Slot = ti->qNumber - tr->mxtips - 1;
+          e->x2Slot = ti->rNumber - tr->mxtips - 1;
+      }
+      e->qz = ti->qz[0];
+      e->rz = ti->rz[0];
+      if (tr->numBranches > 1) {
+        for (m = 0; m < S_n; m++) {
+          g_qzov[(size_t)n * S_n + m] = ti->qz[m];
+          g_rzov[(size_t)n * S_n + m] = ti->rz[m];
+        }
+      }
+      n++;
+    }
+    multi_prep(tr, 0);
+    CK(examl_hip_newview_traversal_multi(
+        g_multi, g_mops, n, g_eign, g_ei, g_rates, g_amask,
+        tr->numBranches > 1 ? g_qzov : NULL,
+        tr->numBranches > 1 ? g_rzov : NULL, 0));
+    return;
+  }
 
   for (m = 0; m < tr->NumberOfModels; m++) {
     ShimPart *p = &S[m];
@@ -403,6 +526,31 @@ void evaluateIterative(tree *tr)
 
   shim_init(tr);
   newviewIterative(tr, 1);
+
+  if (g_multi) {
+    int tc, x1s, x2s, ts, ts2;
+    double zs[NUM_BRANCHES];
+    root_case(pNumber, qNumber, tr->mxtips, &tc, &x1s, &x2s, &ts, &ts2);
+    if (tc == TIP_TIP) shim_die("evaluate at a tip-tip branch", 0);
+    for (m = 0; m < tr->numBranches; m++) zs[m] = pz[m];
+    multi_prep(tr, 0);
+    HIP_OK(hipMemsetAsync(g_d_lnl_vec, 0, S_n * sizeof(double), 0));
+    CK(examl_hip_evaluate_root_multi(
+        g_multi, tc, pNumber, qNumber, x1s, x2s, ts, zs,
+        tr->numBranches > 1 ? 1 : 0, g_eign, g_rates, g_amask, g_d_lnl_vec,
+        0));
+    HIP_OK(hipMemcpyAsync(g_h_lnl_vec, g_d_lnl_vec,
+                          S_n * sizeof(double), hipMemcpyDeviceToHost, 0));
+    HIP_OK(hipStreamSynchronize(0));
+    for (m = 0; m < tr->NumberOfModels; m++) {
+      if (tr->td[0].executeModel[m] && S[m].width > 0)
+        tr->perPartitionLH[m] = g_h_lnl_vec[m];
+      else if (S[m].width == 0)
+        tr->perPartitionLH[m] = 0.0;
+      /* masked + width > 0: keep stale (evaluateGenericSpecial.c:855) */
+    }
+    return;
+  }
 
   for (m = 0; m < tr->NumberOfModels; m++) {
     ShimPart *p = &S[m];
@@ -539,6 +687,15 @@ void makenewzIterative(tree *tr)
   shim_init(tr);
   newviewIterative(tr, 1);
 
+  if (g_multi) {
+    int tc, x1s, x2s, ts, ts2;
+    root_case(pNumber, qNumber, tr->mxtips, &tc, &x1s, &x2s, &ts, &ts2);
+    multi_prep(tr, 0);
+    CK(examl_hip_sum_root_multi(g_multi, tc, x1s, x2s, ts, ts2, g_amask,
+                                0));
+    return;
+  }
+
   for (m = 0; m < tr->NumberOfModels; m++) {
     ShimPart *p = &S[m];
     pInfo *pd = &tr->partitionData[m];
@@ -569,6 +726,32 @@ void execCore(tree *tr, volatile double *_dlnLdlz, volatile double *_d2lnLdlz2)
 {
   int m;
   shim_init(tr);
+
+  if (g_multi) {
+    double lzs[NUM_BRANCHES];
+    for (m = 0; m < tr->numBranches; m++)
+      lzs[m] = tr->td[0].parameterValues[m];
+    multi_prep(tr, 1);
+    HIP_OK(hipMemsetAsync(g_d_out2_vec, 0, 2 * S_n * sizeof(double), 0));
+    CK(examl_hip_core_root_multi(g_multi, lzs,
+                                 tr->numBranches > 1 ? 1 : 0, g_eign,
+                                 g_rates, g_amask, g_d_out2_vec, 0));
+    HIP_OK(hipMemcpyAsync(g_h_out2_vec, g_d_out2_vec,
+                          2 * S_n * sizeof(double), hipMemcpyDeviceToHost,
+                          0));
+    HIP_OK(hipStreamSynchronize(0));
+    for (m = 0; m < tr->NumberOfModels; m++) {
+      const int brIdx = (tr->numBranches > 1) ? m : 0;
+      if (brIdx == m) {
+        _dlnLdlz[brIdx] = 0.0;
+        _d2lnLdlz2[brIdx] = 0.0;
+      }
+      if (!(tr->td[0].executeModel[m] && S[m].width > 0)) continue;
+      _dlnLdlz[brIdx] += g_h_out2_vec[2 * m];
+      _d2lnLdlz2[brIdx] += g_h_out2_vec[2 * m + 1];
+    }
+    return;
+  }
 
   for (m = 0; m < tr->NumberOfModels; m++) {
     ShimPart *p = &S[m];
