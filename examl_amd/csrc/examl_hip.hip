@@ -601,17 +601,21 @@ __global__ __launch_bounds__(NV_BLOCK) void k_newview_prot_gamma(
     const unsigned char *__restrict__ tipX1,
     const unsigned char *__restrict__ tipX2, const int *__restrict__ wgt,
     long n, unsigned int *__restrict__ scalerInc) {
-  /* The four lanes of one site differ only in `cat`; with the natural
-   * cat stride of 400 doubles (800 LDS words, = 0 mod 32 banks) every
-   * P-row read is a 4-way bank conflict.  Padding the cat stride to 404
-   * doubles (808 words, = 8 mod 32) puts the four cats on banks
-   * 0/8/16/24 — conflict-free, which is what the round-1 "LDS-issue
-   * bound" profile was hitting. */
+  /* Wave <-> cat mapping: wave w of the block handles gamma category w
+   * for 64 consecutive sites (one per lane).  Every sL/sR P-row read is
+   * then WAVE-UNIFORM — a single-cycle LDS broadcast — instead of the
+   * per-lane gathers of the site*4+cat mapping, which made this kernel
+   * LDS-issue bound (round-1 profile: 141 us/200k sites, 0.34 of HBM
+   * peak).  The per-site rescale verdict ("all 80 span entries below
+   * 2^-256", avxLikelihood.c:1806) now needs the four cats of a site,
+   * which live in the four waves: each wave publishes its 64-lane ballot
+   * in sSmall and the AND across waves selects the sites to scale. */
   constexpr int CSTR = 404; /* padded per-cat LDS stride of sL/sR */
   __shared__ double sL[4 * CSTR], sR[4 * CSTR], sEV[400];
   __shared__ double sTV[TC != EXAML_INNER_INNER ? 460 : 1];
   __shared__ double sU1[TC != EXAML_INNER_INNER ? 1840 : 1];
   __shared__ double sU2[TC == EXAML_TIP_TIP ? 1840 : 1];
+  __shared__ unsigned long long sSmall[TC != EXAML_TIP_TIP ? 4 : 1];
 
   const int tid = threadIdx.x;
   for (int j = tid; j < 1600; j += NV_BLOCK) {
@@ -636,80 +640,95 @@ __global__ __launch_bounds__(NV_BLOCK) void k_newview_prot_gamma(
     __syncthreads();
   }
 
-  const long units = n * 4;
   const int lane = tid & 63;
-  for (long idx = (long)blockIdx.x * NV_BLOCK + tid; idx < units;
-       idx += (long)gridDim.x * NV_BLOCK) {
-    const long site = idx >> 2;
-    const int cat = (int)(idx & 3);
+  const int cat = tid >> 6; /* wave id = gamma category */
+  const long nChunks = (n + 63) / 64;
+  for (long chunk = blockIdx.x; chunk < nChunks; chunk += gridDim.x) {
+    const long site = chunk * 64 + lane;
+    const bool live = site < n;
+    const long idx = site * 4 + cat; /* span-80 slot (site*80 + cat*20) */
     double xl[20], xr[20], acc[20];
     int code1 = 0, code2 = 0;
-    if (TC == EXAML_INNER_INNER) {
-#pragma unroll
-      for (int s = 0; s < 20; s += 4) {
-        const double4 a = *reinterpret_cast<const double4 *>(&x1[idx * 20 + s]);
-        const double4 b = *reinterpret_cast<const double4 *>(&x2[idx * 20 + s]);
-        xl[s] = a.x; xl[s + 1] = a.y; xl[s + 2] = a.z; xl[s + 3] = a.w;
-        xr[s] = b.x; xr[s + 1] = b.y; xr[s + 2] = b.z; xr[s + 3] = b.w;
-      }
-    } else if (TC == EXAML_TIP_INNER) {
-      code1 = tipX1[site];
-#pragma unroll
-      for (int s = 0; s < 20; s += 4) {
-        const double4 b = *reinterpret_cast<const double4 *>(&x2[idx * 20 + s]);
-        xr[s] = b.x; xr[s + 1] = b.y; xr[s + 2] = b.z; xr[s + 3] = b.w;
-      }
-    } else {
-      code1 = tipX1[site];
-      code2 = tipX2[site];
-    }
-#pragma unroll
-    for (int s = 0; s < 20; s++) acc[s] = 0.0;
-    for (int l = 0; l < 20; l++) {
-      double u1, u2;
+    if (live) {
       if (TC == EXAML_INNER_INNER) {
-        u1 = dot20o<FAST>(xl, &sL[cat * CSTR + l * 20]);
-        u2 = dot20o<FAST>(xr, &sR[cat * CSTR + l * 20]);
-      } else if (TC == EXAML_TIP_INNER) {
-        u1 = sU1[80 * code1 + cat * 20 + l];
-        u2 = dot20o<FAST>(xr, &sR[cat * CSTR + l * 20]);
-      } else {
-        u1 = sU1[80 * code1 + cat * 20 + l];
-        u2 = sU2[80 * code2 + cat * 20 + l];
-      }
-      const double t = u1 * u2;
 #pragma unroll
-      for (int s = 0; s < 20; s++) {
-        if (FAST)
-          acc[s] = fma(t, sEV[l * 20 + s], acc[s]);
-        else
-          acc[s] += t * sEV[l * 20 + s];
+        for (int s = 0; s < 20; s += 4) {
+          const double4 a =
+              *reinterpret_cast<const double4 *>(&x1[idx * 20 + s]);
+          const double4 b =
+              *reinterpret_cast<const double4 *>(&x2[idx * 20 + s]);
+          xl[s] = a.x; xl[s + 1] = a.y; xl[s + 2] = a.z; xl[s + 3] = a.w;
+          xr[s] = b.x; xr[s + 1] = b.y; xr[s + 2] = b.z; xr[s + 3] = b.w;
+        }
+      } else if (TC == EXAML_TIP_INNER) {
+        code1 = tipX1[site];
+#pragma unroll
+        for (int s = 0; s < 20; s += 4) {
+          const double4 b =
+              *reinterpret_cast<const double4 *>(&x2[idx * 20 + s]);
+          xr[s] = b.x; xr[s + 1] = b.y; xr[s + 2] = b.z; xr[s + 3] = b.w;
+        }
+      } else {
+        code1 = tipX1[site];
+        code2 = tipX2[site];
+      }
+#pragma unroll
+      for (int s = 0; s < 20; s++) acc[s] = 0.0;
+      for (int l = 0; l < 20; l++) {
+        double u1, u2;
+        if (TC == EXAML_INNER_INNER) {
+          u1 = dot20o<FAST>(xl, &sL[cat * CSTR + l * 20]);
+          u2 = dot20o<FAST>(xr, &sR[cat * CSTR + l * 20]);
+        } else if (TC == EXAML_TIP_INNER) {
+          u1 = sU1[80 * code1 + cat * 20 + l];
+          u2 = dot20o<FAST>(xr, &sR[cat * CSTR + l * 20]);
+        } else {
+          u1 = sU1[80 * code1 + cat * 20 + l];
+          u2 = sU2[80 * code2 + cat * 20 + l];
+        }
+        const double t = u1 * u2;
+#pragma unroll
+        for (int s = 0; s < 20; s++) {
+          if (FAST)
+            acc[s] = fma(t, sEV[l * 20 + s], acc[s]);
+          else
+            acc[s] += t * sEV[l * 20 + s];
+        }
       }
     }
 
     if (TC != EXAML_TIP_TIP) {
+      /* cross-wave AND of the per-cat smallness over the site's 4 cats */
       bool small = true;
+      if (live) {
 #pragma unroll
-      for (int s = 0; s < 20; s++)
-        small &= (fabs(acc[s]) < MINLIKELIHOOD);
+        for (int s = 0; s < 20; s++)
+          small &= (fabs(acc[s]) < MINLIKELIHOOD);
+      }
       const unsigned long long m = __ballot(small);
-      if (((m >> (lane & ~3)) & 0xFULL) == 0xFULL) {
+      if (lane == 0) sSmall[cat] = m;
+      __syncthreads();
+      const unsigned long long all4 =
+          sSmall[0] & sSmall[1] & sSmall[2] & sSmall[3];
+      __syncthreads(); /* sSmall reused next chunk */
+      if (live && ((all4 >> lane) & 1ULL)) {
 #pragma unroll
         for (int s = 0; s < 20; s++) acc[s] *= TWOTOTHE256;
-        if ((lane & 3) == 0)
-          atomicAdd(scalerInc, (unsigned int)wgt[site]);
+        if (cat == 0) atomicAdd(scalerInc, (unsigned int)wgt[site]);
       }
     }
+    if (live) {
 #pragma unroll
-    for (int s = 0; s < 20; s += 4) {
-      const double4 v = make_double4(acc[s], acc[s + 1], acc[s + 2],
-                                     acc[s + 3]);
-      if (NT)
-        __builtin_nontemporal_store(
-            (v4d){v.x, v.y, v.z, v.w},
-            reinterpret_cast<v4d *>(&x3[idx * 20 + s]));
-      else
-        *reinterpret_cast<double4 *>(&x3[idx * 20 + s]) = v;
+      for (int s = 0; s < 20; s += 4) {
+        const double4 v = make_double4(acc[s], acc[s + 1], acc[s + 2],
+                                       acc[s + 3]);
+        if (NT)
+          __builtin_nontemporal_store(
+              (v4d){v.x, v.y, v.z, v.w},
+              reinterpret_cast<v4d *>(&x3[idx * 20 + s]));
+        else
+          *reinterpret_cast<double4 *>(&x3[idx * 20 + s]) = v;
+      }
     }
   }
 }
