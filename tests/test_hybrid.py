@@ -83,3 +83,28 @@ def test_hybrid_full_search(tmp_path):
     lnl = float(lines[0].split(":")[1])
     rel = abs(lnl - GOLDEN_F_D) / abs(GOLDEN_F_D)
     assert rel < 1e-6, f"hybrid -f d lnL {lnl} vs {GOLDEN_F_D} rel {rel}"
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(not os.path.exists(HYBRID),
+                    reason="hybrid binary not built")
+def test_hybrid_140_partitioned_protein(tmp_path):
+    """testData/140 (BASELINE.json configs[4]): WAG + two AUTO protein
+    partitions through the unmodified reference -f E — exercises the
+    protein kernels, AUTO model selection (optimizeModel.c:2669) and the
+    per-partition dispatch under the C shim.  Golden: reference examl-AVX
+    -f E final lnL -121288.814123 (858 s on the dev box CPU)."""
+    shutil.copy(os.path.join(GOLDEN, "140.binary"), str(tmp_path))
+    shutil.copy(os.path.join(GOLDEN, "140.tree"), str(tmp_path))
+    r = subprocess.run(
+        [HYBRID, "-s", "140.binary", "-t", "140.tree", "-m", "GAMMA",
+         "-f", "E", "-n", "H140"], cwd=str(tmp_path), capture_output=True,
+        text=True, timeout=1800)
+    out = r.stdout + r.stderr
+    assert r.returncode == 0, out[-3000:]
+    lines = [ln for ln in out.splitlines() if "Likelihood tree 0" in ln]
+    assert lines, out[-3000:]
+    lnl = float(lines[0].split(":")[1])
+    golden = -121288.814123
+    rel = abs(lnl - golden) / abs(golden)
+    assert rel < 1e-6, f"hybrid 140 lnL {lnl} vs {golden} rel {rel}"
