@@ -602,46 +602,44 @@ __global__ __launch_bounds__(NV_BLOCK) void k_newview_prot_gamma(
     const unsigned char *__restrict__ tipX2, const int *__restrict__ wgt,
     long n, unsigned int *__restrict__ scalerInc) {
   /* Wave <-> cat mapping: wave w of the block handles gamma category w
-   * for 64 consecutive sites (one per lane).  Every sL/sR P-row read is
-   * then WAVE-UNIFORM — a single-cycle LDS broadcast — instead of the
-   * per-lane gathers of the site*4+cat mapping, which made this kernel
-   * LDS-issue bound (round-1 profile: 141 us/200k sites, 0.34 of HBM
-   * peak).  The per-site rescale verdict ("all 80 span entries below
-   * 2^-256", avxLikelihood.c:1806) now needs the four cats of a site,
-   * which live in the four waves: each wave publishes its 64-lane ballot
-   * in sSmall and the AND across waves selects the sites to scale. */
-  constexpr int CSTR = 404; /* padded per-cat LDS stride of sL/sR */
-  __shared__ double sL[4 * CSTR], sR[4 * CSTR], sEV[400];
+   * for 64 consecutive sites (one per lane).  With `cat` wave-uniform
+   * (readfirstlane), the P-matrix and EV operands of the contraction are
+   * SCALAR loads through the constant cache — no LDS traffic and no VALU
+   * address math in the hot loop; the f64 VALU ops take (SGPR, VGPR)
+   * operand pairs directly.  This replaces the per-lane LDS gathers of
+   * the old site*4+cat mapping (round-1: 141 us/200k sites, 0.34 of HBM
+   * peak, LDS/VALU-issue bound).  The per-site rescale verdict ("all 80
+   * span entries below 2^-256", avxLikelihood.c:1806) needs the four
+   * cats of a site, which live in the four waves: each wave publishes
+   * its 64-lane ballot in sSmall and the AND across waves selects the
+   * sites to scale. */
   __shared__ double sTV[TC != EXAML_INNER_INNER ? 460 : 1];
   __shared__ double sU1[TC != EXAML_INNER_INNER ? 1840 : 1];
   __shared__ double sU2[TC == EXAML_TIP_TIP ? 1840 : 1];
   __shared__ unsigned long long sSmall[TC != EXAML_TIP_TIP ? 4 : 1];
 
   const int tid = threadIdx.x;
-  for (int j = tid; j < 1600; j += NV_BLOCK) {
-    const int pc = j / 400, pr = j % 400;
-    sL[pc * CSTR + pr] = P[j];
-    sR[pc * CSTR + pr] = P[1600 + j];
-  }
-  for (int j = tid; j < 400; j += NV_BLOCK) sEV[j] = EV[j];
-  if (TC != EXAML_INNER_INNER)
-    for (int j = tid; j < 460; j += NV_BLOCK) sTV[j] = tipVec[j];
-  __syncthreads();
-
   if (TC != EXAML_INNER_INNER) {
-    /* ump tables (avxLikelihood.c:1355-1389): entry (code, cat*20+row) */
+    for (int j = tid; j < 460; j += NV_BLOCK) sTV[j] = tipVec[j];
+    __syncthreads();
+    /* ump tables (avxLikelihood.c:1355-1389): entry (code, cat*20+row);
+     * one-time, P read straight from global (L2/constant-cached) */
     for (int j = tid; j < 23 * 80; j += NV_BLOCK) {
       const int code = j / 80, k = j % 80;
       const int kc = k / 20, kl = k % 20;
-      sU1[j] = dot20o<FAST>(&sTV[20 * code], &sL[kc * CSTR + kl * 20]);
+      sU1[j] = dot20o<FAST>(&sTV[20 * code], &P[kc * 400 + kl * 20]);
       if (TC == EXAML_TIP_TIP)
-        sU2[j] = dot20o<FAST>(&sTV[20 * code], &sR[kc * CSTR + kl * 20]);
+        sU2[j] =
+            dot20o<FAST>(&sTV[20 * code], &P[1600 + kc * 400 + kl * 20]);
     }
     __syncthreads();
   }
 
   const int lane = tid & 63;
-  const int cat = tid >> 6; /* wave id = gamma category */
+  const int cat =
+      __builtin_amdgcn_readfirstlane(tid >> 6); /* wave id = gamma cat */
+  const double *__restrict__ Pl = P + cat * 400;        /* scalar bases */
+  const double *__restrict__ Pr = P + 1600 + cat * 400;
   const long nChunks = (n + 63) / 64;
   for (long chunk = blockIdx.x; chunk < nChunks; chunk += gridDim.x) {
     const long site = chunk * 64 + lane;
@@ -677,11 +675,11 @@ __global__ __launch_bounds__(NV_BLOCK) void k_newview_prot_gamma(
       for (int l = 0; l < 20; l++) {
         double u1, u2;
         if (TC == EXAML_INNER_INNER) {
-          u1 = dot20o<FAST>(xl, &sL[cat * CSTR + l * 20]);
-          u2 = dot20o<FAST>(xr, &sR[cat * CSTR + l * 20]);
+          u1 = dot20o<FAST>(xl, &Pl[l * 20]);
+          u2 = dot20o<FAST>(xr, &Pr[l * 20]);
         } else if (TC == EXAML_TIP_INNER) {
           u1 = sU1[80 * code1 + cat * 20 + l];
-          u2 = dot20o<FAST>(xr, &sR[cat * CSTR + l * 20]);
+          u2 = dot20o<FAST>(xr, &Pr[l * 20]);
         } else {
           u1 = sU1[80 * code1 + cat * 20 + l];
           u2 = sU2[80 * code2 + cat * 20 + l];
@@ -690,9 +688,9 @@ __global__ __launch_bounds__(NV_BLOCK) void k_newview_prot_gamma(
 #pragma unroll
         for (int s = 0; s < 20; s++) {
           if (FAST)
-            acc[s] = fma(t, sEV[l * 20 + s], acc[s]);
+            acc[s] = fma(t, EV[l * 20 + s], acc[s]);
           else
-            acc[s] += t * sEV[l * 20 + s];
+            acc[s] += t * EV[l * 20 + s];
         }
       }
     }
@@ -3915,6 +3913,19 @@ struct CSeg { /* one partition's NR-derivative unit */
   int blkBase, nBlocks, part, pad;
 };
 
+/* forward declarations of the protein (20-state) mseg kernels defined
+ * after the executors */
+template <int TC, bool FAST>
+__global__ void k_newview_prot_mseg(const MSeg *, const int *,
+                                    const double *);
+template <bool TIP>
+__global__ void k_evaluate_prot_mseg(const ESeg *, const int *,
+                                     const double *, double *);
+template <int TC>
+__global__ void k_sum_prot_mseg(const SSeg *, const int *, const double *);
+__global__ void k_core_prot_mseg(const CSeg *, const int *, const double *,
+                                 double *);
+
 /* device P-matrix pairs (makeP, newviewGenericSpecial.c:78):
  * P[(e*numParts+part)*8*S^2] = [left|right], left[cat*S^2+row*S+col],
  * col 0 = 1, else exp(rates[cat]*EIGN[col]*lz)*EI[row*S+col]. */
@@ -4315,7 +4326,7 @@ extern "C" int examl_hip_multi_create(
     const long *tipStrides, const int *const *dev_wgts,
     unsigned int *const *dev_scalers, const double *const *dev_EVs,
     const double *const *dev_tipVecs, int maxOps, void **out) {
-  if (states != 4) {
+  if (states != 4 && states != 20) {
     snprintf(g_err, sizeof(g_err), "multi_create: states %d not wired",
              states);
     return -1;
@@ -4630,9 +4641,14 @@ extern "C" int examl_hip_newview_traversal_multi(
                          (size_t)numOps * NP * sizeof(unsigned int), s);
     if (err != hipSuccess) { rc = set_err(err, "inc memset"); break; }
 
-    hipLaunchKernelGGL((k_make_p_mseg<4>), dim3(numOps * NP), dim3(128), 0,
-                       s, h->d_stage + h->zp_off, h->d_stage + h->mod_off,
-                       NP, h->d_pbuf);
+    if (S == 4)
+      hipLaunchKernelGGL((k_make_p_mseg<4>), dim3(numOps * NP), dim3(128),
+                         0, s, h->d_stage + h->zp_off,
+                         h->d_stage + h->mod_off, NP, h->d_pbuf);
+    else
+      hipLaunchKernelGGL((k_make_p_mseg<20>), dim3(numOps * NP), dim3(256),
+                         0, s, h->d_stage + h->zp_off,
+                         h->d_stage + h->mod_off, NP, h->d_pbuf);
     err = hipGetLastError();
     if (err != hipSuccess) { rc = set_err(err, "make_p launch"); break; }
 
@@ -4642,22 +4658,36 @@ extern "C" int examl_hip_newview_traversal_multi(
         prof_begin(&ev_a, &ev_b);
         hipEventRecord(ev_a, s);
       }
-      switch (g.tc) {
-      case EXAML_TIP_TIP:
-        hipLaunchKernelGGL((k_newview_dna_mseg<EXAML_TIP_TIP>), dim3(g.grid),
-                           dim3(NV_BLOCK), 0, s, shape->d_segs,
-                           shape->d_blk2seg + g.blkOff, d_active);
-        break;
-      case EXAML_TIP_INNER:
-        hipLaunchKernelGGL((k_newview_dna_mseg<EXAML_TIP_INNER>),
-                           dim3(g.grid), dim3(NV_BLOCK), 0, s, shape->d_segs,
-                           shape->d_blk2seg + g.blkOff, d_active);
-        break;
-      default:
-        hipLaunchKernelGGL((k_newview_dna_mseg<EXAML_INNER_INNER>),
-                           dim3(g.grid), dim3(NV_BLOCK), 0, s, shape->d_segs,
-                           shape->d_blk2seg + g.blkOff, d_active);
+      const MSeg *dsegs = shape->d_segs;
+      const int *db2s = shape->d_blk2seg + g.blkOff;
+#define NV_MSEG(K) \
+  hipLaunchKernelGGL((K), dim3(g.grid), dim3(NV_BLOCK), 0, s, dsegs, db2s, \
+                     d_active)
+      if (S == 4) {
+        switch (g.tc) {
+        case EXAML_TIP_TIP: NV_MSEG(k_newview_dna_mseg<EXAML_TIP_TIP>); break;
+        case EXAML_TIP_INNER:
+          NV_MSEG(k_newview_dna_mseg<EXAML_TIP_INNER>); break;
+        default: NV_MSEG(k_newview_dna_mseg<EXAML_INNER_INNER>);
+        }
+      } else if (g_fast_math) {
+        switch (g.tc) {
+        case EXAML_TIP_TIP:
+          NV_MSEG((k_newview_prot_mseg<EXAML_TIP_TIP, true>)); break;
+        case EXAML_TIP_INNER:
+          NV_MSEG((k_newview_prot_mseg<EXAML_TIP_INNER, true>)); break;
+        default: NV_MSEG((k_newview_prot_mseg<EXAML_INNER_INNER, true>));
+        }
+      } else {
+        switch (g.tc) {
+        case EXAML_TIP_TIP:
+          NV_MSEG((k_newview_prot_mseg<EXAML_TIP_TIP, false>)); break;
+        case EXAML_TIP_INNER:
+          NV_MSEG((k_newview_prot_mseg<EXAML_TIP_INNER, false>)); break;
+        default: NV_MSEG((k_newview_prot_mseg<EXAML_INNER_INNER, false>));
+        }
       }
+#undef NV_MSEG
       err = hipGetLastError();
       if (err != hipSuccess) { rc = set_err(err, "mseg launch"); break; }
       if (g_prof_on) {
@@ -4764,14 +4794,25 @@ extern "C" int examl_hip_evaluate_root_multi(
   slot->ev_valid = true;
   const double *d_active = (const double *)(h->d_callbuf + actOff);
   const ESeg *d_es = (const ESeg *)h->d_callbuf;
-  if (rootTipCase == EXAML_TIP_INNER)
-    hipLaunchKernelGGL((k_evaluate_dna_mseg<true>), dim3(h->totalBlocks),
-                       dim3(NV_BLOCK), 0, s, d_es, h->d_blk2part, d_active,
-                       h->d_partials);
-  else
-    hipLaunchKernelGGL((k_evaluate_dna_mseg<false>), dim3(h->totalBlocks),
-                       dim3(NV_BLOCK), 0, s, d_es, h->d_blk2part, d_active,
-                       h->d_partials);
+  if (S == 4) {
+    if (rootTipCase == EXAML_TIP_INNER)
+      hipLaunchKernelGGL((k_evaluate_dna_mseg<true>), dim3(h->totalBlocks),
+                         dim3(NV_BLOCK), 0, s, d_es, h->d_blk2part,
+                         d_active, h->d_partials);
+    else
+      hipLaunchKernelGGL((k_evaluate_dna_mseg<false>), dim3(h->totalBlocks),
+                         dim3(NV_BLOCK), 0, s, d_es, h->d_blk2part,
+                         d_active, h->d_partials);
+  } else {
+    if (rootTipCase == EXAML_TIP_INNER)
+      hipLaunchKernelGGL((k_evaluate_prot_mseg<true>), dim3(h->totalBlocks),
+                         dim3(NV_BLOCK), 0, s, d_es, h->d_blk2part,
+                         d_active, h->d_partials);
+    else
+      hipLaunchKernelGGL((k_evaluate_prot_mseg<false>),
+                         dim3(h->totalBlocks), dim3(NV_BLOCK), 0, s, d_es,
+                         h->d_blk2part, d_active, h->d_partials);
+  }
   CHK(hipGetLastError());
   hipLaunchKernelGGL(k_reduce_lnl_mseg, dim3(h->numSegs), dim3(NV_BLOCK), 0,
                      s, d_es, h->d_partials, d_active, log(MINLIKELIHOOD));
@@ -4824,21 +4865,23 @@ extern "C" int examl_hip_sum_root_multi(void *vh, int rootTipCase,
   slot->ev_valid = true;
   const double *d_active = (const double *)(h->d_callbuf + actOff);
   const SSeg *d_ss = (const SSeg *)h->d_callbuf;
-  switch (rootTipCase) {
-  case EXAML_TIP_TIP:
-    hipLaunchKernelGGL((k_sum_dna_mseg<EXAML_TIP_TIP>), dim3(h->totalBlocks),
-                       dim3(NV_BLOCK), 0, s, d_ss, h->d_blk2part, d_active);
-    break;
-  case EXAML_TIP_INNER:
-    hipLaunchKernelGGL((k_sum_dna_mseg<EXAML_TIP_INNER>),
-                       dim3(h->totalBlocks), dim3(NV_BLOCK), 0, s, d_ss,
-                       h->d_blk2part, d_active);
-    break;
-  default:
-    hipLaunchKernelGGL((k_sum_dna_mseg<EXAML_INNER_INNER>),
-                       dim3(h->totalBlocks), dim3(NV_BLOCK), 0, s, d_ss,
-                       h->d_blk2part, d_active);
+#define SUM_MSEG(K) \
+  hipLaunchKernelGGL((K), dim3(h->totalBlocks), dim3(NV_BLOCK), 0, s, d_ss, \
+                     h->d_blk2part, d_active)
+  if (h->states == 4) {
+    switch (rootTipCase) {
+    case EXAML_TIP_TIP: SUM_MSEG(k_sum_dna_mseg<EXAML_TIP_TIP>); break;
+    case EXAML_TIP_INNER: SUM_MSEG(k_sum_dna_mseg<EXAML_TIP_INNER>); break;
+    default: SUM_MSEG(k_sum_dna_mseg<EXAML_INNER_INNER>);
+    }
+  } else {
+    switch (rootTipCase) {
+    case EXAML_TIP_TIP: SUM_MSEG(k_sum_prot_mseg<EXAML_TIP_TIP>); break;
+    case EXAML_TIP_INNER: SUM_MSEG(k_sum_prot_mseg<EXAML_TIP_INNER>); break;
+    default: SUM_MSEG(k_sum_prot_mseg<EXAML_INNER_INNER>);
+    }
   }
+#undef SUM_MSEG
   CHK(hipGetLastError());
   return 0;
 }
@@ -4885,11 +4928,315 @@ extern "C" int examl_hip_core_root_multi(
   slot->ev_valid = true;
   const double *d_active = (const double *)(h->d_callbuf + actOff);
   const CSeg *d_cs = (const CSeg *)h->d_callbuf;
-  hipLaunchKernelGGL(k_core_dna_mseg, dim3(h->totalBlocks), dim3(NV_BLOCK),
-                     0, s, d_cs, h->d_blk2part, d_active, h->d_partials);
+  if (S == 4)
+    hipLaunchKernelGGL(k_core_dna_mseg, dim3(h->totalBlocks),
+                       dim3(NV_BLOCK), 0, s, d_cs, h->d_blk2part, d_active,
+                       h->d_partials);
+  else
+    hipLaunchKernelGGL(k_core_prot_mseg, dim3(h->totalBlocks),
+                       dim3(NV_BLOCK), 0, s, d_cs, h->d_blk2part, d_active,
+                       h->d_partials);
   CHK(hipGetLastError());
   hipLaunchKernelGGL(k_reduce_2_mseg, dim3(h->numSegs), dim3(NV_BLOCK), 0, s,
                      d_cs, h->d_partials, d_active);
   CHK(hipGetLastError());
   return 0;
+}
+
+/* ---------------------------------------------------------------------------
+ * Protein (20-state) mseg kernels — the fused counterparts of the prot
+ * GAMMA family, with the same wave-per-cat mapping as
+ * k_newview_prot_gamma (P-row LDS reads are wave-uniform broadcasts).
+ * ------------------------------------------------------------------------ */
+
+template <int TC, bool FAST>
+__global__ __launch_bounds__(NV_BLOCK) void k_newview_prot_mseg(
+    const MSeg *__restrict__ segs, const int *__restrict__ blk2seg,
+    const double *__restrict__ active) {
+  const int si = blk2seg[blockIdx.x];
+  const MSeg sg = segs[si];
+  if (active[sg.part] == 0.0) return;
+  /* same scalar-operand scheme as k_newview_prot_gamma: cat is
+   * wave-uniform, so P/EV reads are s_loads through the constant cache */
+  __shared__ double sTV[TC != EXAML_INNER_INNER ? 460 : 1];
+  __shared__ double sU1[TC != EXAML_INNER_INNER ? 1840 : 1];
+  __shared__ double sU2[TC == EXAML_TIP_TIP ? 1840 : 1];
+  __shared__ unsigned long long sSmall[TC != EXAML_TIP_TIP ? 4 : 1];
+
+  const int tid = threadIdx.x;
+  if (TC != EXAML_INNER_INNER) {
+    for (int j = tid; j < 460; j += NV_BLOCK) sTV[j] = sg.tipVec[j];
+    __syncthreads();
+    for (int j = tid; j < 23 * 80; j += NV_BLOCK) {
+      const int code = j / 80, k = j % 80;
+      const int kc = k / 20, kl = k % 20;
+      sU1[j] = dot20o<FAST>(&sTV[20 * code], &sg.P[kc * 400 + kl * 20]);
+      if (TC == EXAML_TIP_TIP)
+        sU2[j] =
+            dot20o<FAST>(&sTV[20 * code], &sg.P[1600 + kc * 400 + kl * 20]);
+    }
+    __syncthreads();
+  }
+
+  const int lane = tid & 63;
+  const int cat = __builtin_amdgcn_readfirstlane(tid >> 6);
+  const double *__restrict__ Pl = sg.P + cat * 400;
+  const double *__restrict__ Pr = sg.P + 1600 + cat * 400;
+  const double *__restrict__ EV = sg.EV;
+  /* NT stores measured SLOWER for the protein span-80 strided stores
+   * (tools/kernel_ab, round 1) — keep cached stores here */
+  const bool nt = false;
+  const long nChunks = (sg.n + 63) / 64;
+  for (long chunk = (long)(blockIdx.x - sg.blkBase); chunk < nChunks;
+       chunk += sg.nBlocks) {
+    const long site = chunk * 64 + lane;
+    const bool live = site < sg.n;
+    const long idx = site * 4 + cat;
+    double xl[20], xr[20], acc[20];
+    int code1 = 0, code2 = 0;
+    if (live) {
+      if (TC == EXAML_INNER_INNER) {
+#pragma unroll
+        for (int s = 0; s < 20; s += 4) {
+          const double4 a =
+              *reinterpret_cast<const double4 *>(&sg.x1[idx * 20 + s]);
+          const double4 b =
+              *reinterpret_cast<const double4 *>(&sg.x2[idx * 20 + s]);
+          xl[s] = a.x; xl[s + 1] = a.y; xl[s + 2] = a.z; xl[s + 3] = a.w;
+          xr[s] = b.x; xr[s + 1] = b.y; xr[s + 2] = b.z; xr[s + 3] = b.w;
+        }
+      } else if (TC == EXAML_TIP_INNER) {
+        code1 = sg.t1[site];
+#pragma unroll
+        for (int s = 0; s < 20; s += 4) {
+          const double4 b =
+              *reinterpret_cast<const double4 *>(&sg.x2[idx * 20 + s]);
+          xr[s] = b.x; xr[s + 1] = b.y; xr[s + 2] = b.z; xr[s + 3] = b.w;
+        }
+      } else {
+        code1 = sg.t1[site];
+        code2 = sg.t2[site];
+      }
+#pragma unroll
+      for (int s = 0; s < 20; s++) acc[s] = 0.0;
+      for (int l = 0; l < 20; l++) {
+        double u1, u2;
+        if (TC == EXAML_INNER_INNER) {
+          u1 = dot20o<FAST>(xl, &Pl[l * 20]);
+          u2 = dot20o<FAST>(xr, &Pr[l * 20]);
+        } else if (TC == EXAML_TIP_INNER) {
+          u1 = sU1[80 * code1 + cat * 20 + l];
+          u2 = dot20o<FAST>(xr, &Pr[l * 20]);
+        } else {
+          u1 = sU1[80 * code1 + cat * 20 + l];
+          u2 = sU2[80 * code2 + cat * 20 + l];
+        }
+        const double t = u1 * u2;
+#pragma unroll
+        for (int s = 0; s < 20; s++) {
+          if (FAST)
+            acc[s] = fma(t, EV[l * 20 + s], acc[s]);
+          else
+            acc[s] += t * EV[l * 20 + s];
+        }
+      }
+    }
+
+    if (TC != EXAML_TIP_TIP) {
+      bool small = true;
+      if (live) {
+#pragma unroll
+        for (int s = 0; s < 20; s++)
+          small &= (fabs(acc[s]) < MINLIKELIHOOD);
+      }
+      const unsigned long long m = __ballot(small);
+      if (lane == 0) sSmall[cat] = m;
+      __syncthreads();
+      const unsigned long long all4 =
+          sSmall[0] & sSmall[1] & sSmall[2] & sSmall[3];
+      __syncthreads();
+      if (live && ((all4 >> lane) & 1ULL)) {
+#pragma unroll
+        for (int s = 0; s < 20; s++) acc[s] *= TWOTOTHE256;
+        if (cat == 0) atomicAdd(sg.inc, (unsigned int)sg.wgt[site]);
+      }
+    }
+    if (live) {
+#pragma unroll
+      for (int s = 0; s < 20; s += 4) {
+        const double4 v = make_double4(acc[s], acc[s + 1], acc[s + 2],
+                                       acc[s + 3]);
+        if (nt)
+          __builtin_nontemporal_store(
+              (v4d){v.x, v.y, v.z, v.w},
+              reinterpret_cast<v4d *>(&sg.x3[idx * 20 + s]));
+        else
+          *reinterpret_cast<double4 *>(&sg.x3[idx * 20 + s]) = v;
+      }
+    }
+  }
+}
+
+template <bool TIP>
+__global__ __launch_bounds__(NV_BLOCK) void k_evaluate_prot_mseg(
+    const ESeg *__restrict__ segs, const int *__restrict__ blk2seg,
+    const double *__restrict__ active, double *__restrict__ partials) {
+  const int si = blk2seg[blockIdx.x];
+  const ESeg sg = segs[si];
+  if (active[sg.part] == 0.0) return;
+  __shared__ double sD[80], sTV[TIP ? 460 : 1], sRed[NV_BLOCK / 64];
+  const int tid = threadIdx.x;
+  for (int j = tid; j < 80; j += NV_BLOCK) sD[j] = sg.diag[j];
+  if (TIP) /* x1 carries the partition's tipVector pointer */
+    for (int j = tid; j < 460; j += NV_BLOCK) sTV[j] = sg.x1[j];
+  __syncthreads();
+
+  const long units = sg.n * 4;
+  const int lane = tid & 63;
+  double acc = 0.0;
+  for (long idx = (long)(blockIdx.x - sg.blkBase) * NV_BLOCK + tid;
+       idx < units; idx += (long)sg.nBlocks * NV_BLOCK) {
+    const long site = idx >> 2;
+    const int cat = (int)(idx & 3);
+    const double *le = TIP ? &sTV[20 * sg.t1[site]] : &sg.x1[idx * 20];
+    double t0 = 0, t1 = 0;
+#pragma unroll
+    for (int l = 0; l < 20; l += 2) {
+      const double2 b =
+          *reinterpret_cast<const double2 *>(&sg.x2[idx * 20 + l]);
+      t0 += le[l] * b.x * sD[cat * 20 + l];
+      t1 += le[l + 1] * b.y * sD[cat * 20 + l + 1];
+    }
+    double p = t0 + t1;
+    p += __shfl_xor(p, 1);
+    p += __shfl_xor(p, 2);
+    if ((lane & 3) == 0) acc += (double)sg.wgt[site] * log(0.25 * fabs(p));
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off);
+  if (lane == 0) sRed[tid >> 6] = acc;
+  __syncthreads();
+  if (tid == 0) {
+    double s = 0;
+#pragma unroll
+    for (int w = 0; w < NV_BLOCK / 64; w++) s += sRed[w];
+    partials[blockIdx.x] = s;
+  }
+}
+
+template <int TC>
+__global__ __launch_bounds__(NV_BLOCK) void k_sum_prot_mseg(
+    const SSeg *__restrict__ segs, const int *__restrict__ blk2seg,
+    const double *__restrict__ active) {
+  const int si = blk2seg[blockIdx.x];
+  const SSeg sg = segs[si];
+  if (active[sg.part] == 0.0) return;
+  __shared__ double sTV[TC != EXAML_INNER_INNER ? 460 : 1];
+  const int tid = threadIdx.x;
+  if (TC != EXAML_INNER_INNER) {
+    for (int j = tid; j < 460; j += NV_BLOCK) sTV[j] = sg.tipVec[j];
+    __syncthreads();
+  }
+  const long units = sg.n * 4;
+  for (long idx = (long)(blockIdx.x - sg.blkBase) * NV_BLOCK + tid;
+       idx < units; idx += (long)sg.nBlocks * NV_BLOCK) {
+    const long site = idx >> 2;
+#pragma unroll
+    for (int l = 0; l < 20; l += 2) {
+      double a0, a1, b0, b1;
+      if (TC == EXAML_TIP_TIP) {
+        a0 = sTV[20 * sg.t1[site] + l];
+        a1 = sTV[20 * sg.t1[site] + l + 1];
+        b0 = sTV[20 * sg.t2[site] + l];
+        b1 = sTV[20 * sg.t2[site] + l + 1];
+      } else if (TC == EXAML_TIP_INNER) {
+        a0 = sTV[20 * sg.t1[site] + l];
+        a1 = sTV[20 * sg.t1[site] + l + 1];
+        const double2 b =
+            *reinterpret_cast<const double2 *>(&sg.x2[idx * 20 + l]);
+        b0 = b.x;
+        b1 = b.y;
+      } else {
+        const double2 a =
+            *reinterpret_cast<const double2 *>(&sg.x1[idx * 20 + l]);
+        const double2 b =
+            *reinterpret_cast<const double2 *>(&sg.x2[idx * 20 + l]);
+        a0 = a.x;
+        a1 = a.y;
+        b0 = b.x;
+        b1 = b.y;
+      }
+      *reinterpret_cast<double2 *>(&sg.sum[idx * 20 + l]) =
+          make_double2(a0 * b0, a1 * b1);
+    }
+  }
+}
+
+__global__ __launch_bounds__(NV_BLOCK) void k_core_prot_mseg(
+    const CSeg *__restrict__ segs, const int *__restrict__ blk2seg,
+    const double *__restrict__ active, double *__restrict__ partials) {
+  const int si = blk2seg[blockIdx.x];
+  const CSeg sg = segs[si];
+  if (active[sg.part] == 0.0) return;
+  __shared__ double sD0[80], sD1[80], sD2[80], sRed[2][NV_BLOCK / 64];
+  const int tid = threadIdx.x;
+  for (int j = tid; j < 80; j += NV_BLOCK) {
+    sD0[j] = sg.dtab[j];
+    sD1[j] = sg.dtab[80 + j];
+    sD2[j] = sg.dtab[160 + j];
+  }
+  __syncthreads();
+
+  const long units = sg.n * 4;
+  const int lane = tid & 63;
+  double accD1 = 0.0, accD2 = 0.0;
+  for (long idx = (long)(blockIdx.x - sg.blkBase) * NV_BLOCK + tid;
+       idx < units; idx += (long)sg.nBlocks * NV_BLOCK) {
+    const long site = idx >> 2;
+    const int cat = (int)(idx & 3);
+    double a0 = 0, a1 = 0, a2 = 0;
+#pragma unroll
+    for (int l = 0; l < 20; l += 2) {
+      const double2 s2 =
+          *reinterpret_cast<const double2 *>(&sg.sum[idx * 20 + l]);
+      const double te = sD0[cat * 20 + l] * s2.x;
+      const double to = sD0[cat * 20 + l + 1] * s2.y;
+      a0 += te + to;
+      a1 += te * sD1[cat * 20 + l] + to * sD1[cat * 20 + l + 1];
+      a2 += te * sD2[cat * 20 + l] + to * sD2[cat * 20 + l + 1];
+    }
+    a0 += __shfl_xor(a0, 1);
+    a0 += __shfl_xor(a0, 2);
+    a1 += __shfl_xor(a1, 1);
+    a1 += __shfl_xor(a1, 2);
+    a2 += __shfl_xor(a2, 1);
+    a2 += __shfl_xor(a2, 2);
+    if ((lane & 3) == 0) {
+      const double inv = 1.0 / fabs(a0);
+      const double d1 = a1 * inv, d2 = a2 * inv;
+      const double w = (double)sg.wgt[site];
+      accD1 += w * d1;
+      accD2 += w * (d2 - d1 * d1);
+    }
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    accD1 += __shfl_down(accD1, off);
+    accD2 += __shfl_down(accD2, off);
+  }
+  if (lane == 0) {
+    sRed[0][tid >> 6] = accD1;
+    sRed[1][tid >> 6] = accD2;
+  }
+  __syncthreads();
+  if (tid == 0) {
+    double s1 = 0, s2 = 0;
+#pragma unroll
+    for (int w = 0; w < NV_BLOCK / 64; w++) {
+      s1 += sRed[0][w];
+      s2 += sRed[1][w];
+    }
+    partials[2 * blockIdx.x] = s1;
+    partials[2 * blockIdx.x + 1] = s2;
+  }
 }
