@@ -634,132 +634,115 @@ __global__ __launch_bounds__(NV_BLOCK) void k_newview_prot_gamma(
     const unsigned char *__restrict__ tipX1,
     const unsigned char *__restrict__ tipX2, const int *__restrict__ wgt,
     long n, unsigned int *__restrict__ scalerInc) {
-  /* Wave <-> cat mapping: wave w of the block handles gamma category w
-   * for 64 consecutive sites (one per lane).  With `cat` wave-uniform
-   * (readfirstlane), the P-matrix and EV operands of the contraction are
-   * SCALAR loads through the constant cache — no LDS traffic and no VALU
-   * address math in the hot loop; the f64 VALU ops take (SGPR, VGPR)
-   * operand pairs directly.  This replaces the per-lane LDS gathers of
-   * the old site*4+cat mapping (round-1: 141 us/200k sites, 0.34 of HBM
-   * peak, LDS/VALU-issue bound).  The per-site rescale verdict ("all 80
-   * span entries below 2^-256", avxLikelihood.c:1806) needs the four
-   * cats of a site, which live in the four waves: each wave publishes
-   * its 64-lane ballot in sSmall and the AND across waves selects the
-   * sites to scale. */
+  /* The four lanes of one site differ only in `cat`; with the natural
+   * cat stride of 400 doubles (800 LDS words, = 0 mod 32 banks) every
+   * P-row read is a 4-way bank conflict.  Padding the cat stride to 404
+   * doubles (808 words, = 8 mod 32) puts the four cats on banks
+   * 0/8/16/24 — conflict-free, which is what the round-1 "LDS-issue
+   * bound" profile was hitting. */
+  constexpr int CSTR = 404; /* padded per-cat LDS stride of sL/sR */
+  __shared__ double sL[4 * CSTR], sR[4 * CSTR], sEV[400];
   __shared__ double sTV[TC != EXAML_INNER_INNER ? 460 : 1];
   __shared__ double sU1[TC != EXAML_INNER_INNER ? 1840 : 1];
   __shared__ double sU2[TC == EXAML_TIP_TIP ? 1840 : 1];
-  __shared__ unsigned long long sSmall[TC != EXAML_TIP_TIP ? 4 : 1];
 
   const int tid = threadIdx.x;
-  if (TC != EXAML_INNER_INNER) {
+  for (int j = tid; j < 1600; j += NV_BLOCK) {
+    const int pc = j / 400, pr = j % 400;
+    sL[pc * CSTR + pr] = P[j];
+    sR[pc * CSTR + pr] = P[1600 + j];
+  }
+  for (int j = tid; j < 400; j += NV_BLOCK) sEV[j] = EV[j];
+  if (TC != EXAML_INNER_INNER)
     for (int j = tid; j < 460; j += NV_BLOCK) sTV[j] = tipVec[j];
-    __syncthreads();
-    /* ump tables (avxLikelihood.c:1355-1389): entry (code, cat*20+row);
-     * one-time, P read straight from global (L2/constant-cached) */
+  __syncthreads();
+
+  if (TC != EXAML_INNER_INNER) {
+    /* ump tables (avxLikelihood.c:1355-1389): entry (code, cat*20+row) */
     for (int j = tid; j < 23 * 80; j += NV_BLOCK) {
       const int code = j / 80, k = j % 80;
       const int kc = k / 20, kl = k % 20;
-      sU1[j] = dot20o<FAST>(&sTV[20 * code], &P[kc * 400 + kl * 20]);
+      sU1[j] = dot20o<FAST>(&sTV[20 * code], &sL[kc * CSTR + kl * 20]);
       if (TC == EXAML_TIP_TIP)
-        sU2[j] =
-            dot20o<FAST>(&sTV[20 * code], &P[1600 + kc * 400 + kl * 20]);
+        sU2[j] = dot20o<FAST>(&sTV[20 * code], &sR[kc * CSTR + kl * 20]);
     }
     __syncthreads();
   }
 
+  const long units = n * 4;
   const int lane = tid & 63;
-  const int cat =
-      __builtin_amdgcn_readfirstlane(tid >> 6); /* wave id = gamma cat */
-  const double *__restrict__ Pl = P + cat * 400;        /* scalar bases */
-  const double *__restrict__ Pr = P + 1600 + cat * 400;
-  const long nChunks = (n + 63) / 64;
-  for (long chunk = blockIdx.x; chunk < nChunks; chunk += gridDim.x) {
-    const long site = chunk * 64 + lane;
-    const bool live = site < n;
-    const long idx = site * 4 + cat; /* span-80 slot (site*80 + cat*20) */
+  for (long idx = (long)blockIdx.x * NV_BLOCK + tid; idx < units;
+       idx += (long)gridDim.x * NV_BLOCK) {
+    const long site = idx >> 2;
+    const int cat = (int)(idx & 3);
     double xl[20], xr[20], acc[20];
     int code1 = 0, code2 = 0;
-    if (live) {
-      if (TC == EXAML_INNER_INNER) {
+    if (TC == EXAML_INNER_INNER) {
 #pragma unroll
-        for (int s = 0; s < 20; s += 4) {
-          const double4 a =
-              *reinterpret_cast<const double4 *>(&x1[idx * 20 + s]);
-          const double4 b =
-              *reinterpret_cast<const double4 *>(&x2[idx * 20 + s]);
-          xl[s] = a.x; xl[s + 1] = a.y; xl[s + 2] = a.z; xl[s + 3] = a.w;
-          xr[s] = b.x; xr[s + 1] = b.y; xr[s + 2] = b.z; xr[s + 3] = b.w;
-        }
-      } else if (TC == EXAML_TIP_INNER) {
-        code1 = tipX1[site];
-#pragma unroll
-        for (int s = 0; s < 20; s += 4) {
-          const double4 b =
-              *reinterpret_cast<const double4 *>(&x2[idx * 20 + s]);
-          xr[s] = b.x; xr[s + 1] = b.y; xr[s + 2] = b.z; xr[s + 3] = b.w;
-        }
-      } else {
-        code1 = tipX1[site];
-        code2 = tipX2[site];
+      for (int s = 0; s < 20; s += 4) {
+        const double4 a = *reinterpret_cast<const double4 *>(&x1[idx * 20 + s]);
+        const double4 b = *reinterpret_cast<const double4 *>(&x2[idx * 20 + s]);
+        xl[s] = a.x; xl[s + 1] = a.y; xl[s + 2] = a.z; xl[s + 3] = a.w;
+        xr[s] = b.x; xr[s + 1] = b.y; xr[s + 2] = b.z; xr[s + 3] = b.w;
       }
+    } else if (TC == EXAML_TIP_INNER) {
+      code1 = tipX1[site];
 #pragma unroll
-      for (int s = 0; s < 20; s++) acc[s] = 0.0;
-      for (int l = 0; l < 20; l++) {
-        double u1, u2;
-        if (TC == EXAML_INNER_INNER) {
-          u1 = dot20o<FAST>(xl, &Pl[l * 20]);
-          u2 = dot20o<FAST>(xr, &Pr[l * 20]);
-        } else if (TC == EXAML_TIP_INNER) {
-          u1 = sU1[80 * code1 + cat * 20 + l];
-          u2 = dot20o<FAST>(xr, &Pr[l * 20]);
-        } else {
-          u1 = sU1[80 * code1 + cat * 20 + l];
-          u2 = sU2[80 * code2 + cat * 20 + l];
-        }
-        const double t = u1 * u2;
+      for (int s = 0; s < 20; s += 4) {
+        const double4 b = *reinterpret_cast<const double4 *>(&x2[idx * 20 + s]);
+        xr[s] = b.x; xr[s + 1] = b.y; xr[s + 2] = b.z; xr[s + 3] = b.w;
+      }
+    } else {
+      code1 = tipX1[site];
+      code2 = tipX2[site];
+    }
 #pragma unroll
-        for (int s = 0; s < 20; s++) {
-          if (FAST)
-            acc[s] = fma(t, EV[l * 20 + s], acc[s]);
-          else
-            acc[s] += t * EV[l * 20 + s];
-        }
+    for (int s = 0; s < 20; s++) acc[s] = 0.0;
+    for (int l = 0; l < 20; l++) {
+      double u1, u2;
+      if (TC == EXAML_INNER_INNER) {
+        u1 = dot20o<FAST>(xl, &sL[cat * CSTR + l * 20]);
+        u2 = dot20o<FAST>(xr, &sR[cat * CSTR + l * 20]);
+      } else if (TC == EXAML_TIP_INNER) {
+        u1 = sU1[80 * code1 + cat * 20 + l];
+        u2 = dot20o<FAST>(xr, &sR[cat * CSTR + l * 20]);
+      } else {
+        u1 = sU1[80 * code1 + cat * 20 + l];
+        u2 = sU2[80 * code2 + cat * 20 + l];
+      }
+      const double t = u1 * u2;
+#pragma unroll
+      for (int s = 0; s < 20; s++) {
+        if (FAST)
+          acc[s] = fma(t, sEV[l * 20 + s], acc[s]);
+        else
+          acc[s] += t * sEV[l * 20 + s];
       }
     }
 
     if (TC != EXAML_TIP_TIP) {
-      /* cross-wave AND of the per-cat smallness over the site's 4 cats */
       bool small = true;
-      if (live) {
 #pragma unroll
-        for (int s = 0; s < 20; s++)
-          small &= (fabs(acc[s]) < MINLIKELIHOOD);
-      }
+      for (int s = 0; s < 20; s++)
+        small &= (fabs(acc[s]) < MINLIKELIHOOD);
       const unsigned long long m = __ballot(small);
-      if (lane == 0) sSmall[cat] = m;
-      __syncthreads();
-      const unsigned long long all4 =
-          sSmall[0] & sSmall[1] & sSmall[2] & sSmall[3];
-      __syncthreads(); /* sSmall reused next chunk */
-      if (live && ((all4 >> lane) & 1ULL)) {
+      if (((m >> (lane & ~3)) & 0xFULL) == 0xFULL) {
 #pragma unroll
         for (int s = 0; s < 20; s++) acc[s] *= TWOTOTHE256;
-        if (cat == 0) atomicAdd(scalerInc, (unsigned int)wgt[site]);
+        if ((lane & 3) == 0)
+          atomicAdd(scalerInc, (unsigned int)wgt[site]);
       }
     }
-    if (live) {
 #pragma unroll
-      for (int s = 0; s < 20; s += 4) {
-        const double4 v = make_double4(acc[s], acc[s + 1], acc[s + 2],
-                                       acc[s + 3]);
-        if (NT)
-          __builtin_nontemporal_store(
-              (v4d){v.x, v.y, v.z, v.w},
-              reinterpret_cast<v4d *>(&x3[idx * 20 + s]));
-        else
-          *reinterpret_cast<double4 *>(&x3[idx * 20 + s]) = v;
-      }
+    for (int s = 0; s < 20; s += 4) {
+      const double4 v = make_double4(acc[s], acc[s + 1], acc[s + 2],
+                                     acc[s + 3]);
+      if (NT)
+        __builtin_nontemporal_store(
+            (v4d){v.x, v.y, v.z, v.w},
+            reinterpret_cast<v4d *>(&x3[idx * 20 + s]));
+      else
+        *reinterpret_cast<double4 *>(&x3[idx * 20 + s]) = v;
     }
   }
 }
@@ -5014,127 +4997,111 @@ template <int TC, bool FAST>
 __global__ __launch_bounds__(NV_BLOCK) void k_newview_prot_mseg(
     const MSeg *__restrict__ segs, const int *__restrict__ blk2seg,
     const double *__restrict__ active) {
+  /* fused twin of k_newview_prot_gamma: same per-(site,cat) lane mapping
+   * and padded-LDS P staging (the best-measured of the three tried
+   * schemes — see DESIGN.md kernel notes) */
   const int si = blk2seg[blockIdx.x];
   const MSeg sg = segs[si];
   if (active[sg.part] == 0.0) return;
-  /* same scalar-operand scheme as k_newview_prot_gamma: cat is
-   * wave-uniform, so P/EV reads are s_loads through the constant cache */
+  constexpr int CSTR = 404;
+  __shared__ double sL[4 * CSTR], sR[4 * CSTR], sEV[400];
   __shared__ double sTV[TC != EXAML_INNER_INNER ? 460 : 1];
   __shared__ double sU1[TC != EXAML_INNER_INNER ? 1840 : 1];
   __shared__ double sU2[TC == EXAML_TIP_TIP ? 1840 : 1];
-  __shared__ unsigned long long sSmall[TC != EXAML_TIP_TIP ? 4 : 1];
 
   const int tid = threadIdx.x;
-  if (TC != EXAML_INNER_INNER) {
+  for (int j = tid; j < 1600; j += NV_BLOCK) {
+    const int pc = j / 400, pr = j % 400;
+    sL[pc * CSTR + pr] = sg.P[j];
+    sR[pc * CSTR + pr] = sg.P[1600 + j];
+  }
+  for (int j = tid; j < 400; j += NV_BLOCK) sEV[j] = sg.EV[j];
+  if (TC != EXAML_INNER_INNER)
     for (int j = tid; j < 460; j += NV_BLOCK) sTV[j] = sg.tipVec[j];
-    __syncthreads();
+  __syncthreads();
+
+  if (TC != EXAML_INNER_INNER) {
     for (int j = tid; j < 23 * 80; j += NV_BLOCK) {
       const int code = j / 80, k = j % 80;
       const int kc = k / 20, kl = k % 20;
-      sU1[j] = dot20o<FAST>(&sTV[20 * code], &sg.P[kc * 400 + kl * 20]);
+      sU1[j] = dot20o<FAST>(&sTV[20 * code], &sL[kc * CSTR + kl * 20]);
       if (TC == EXAML_TIP_TIP)
-        sU2[j] =
-            dot20o<FAST>(&sTV[20 * code], &sg.P[1600 + kc * 400 + kl * 20]);
+        sU2[j] = dot20o<FAST>(&sTV[20 * code], &sR[kc * CSTR + kl * 20]);
     }
     __syncthreads();
   }
 
+  const long units = sg.n * 4;
   const int lane = tid & 63;
-  const int cat = __builtin_amdgcn_readfirstlane(tid >> 6);
-  const double *__restrict__ Pl = sg.P + cat * 400;
-  const double *__restrict__ Pr = sg.P + 1600 + cat * 400;
-  const double *__restrict__ EV = sg.EV;
-  /* NT stores measured SLOWER for the protein span-80 strided stores
-   * (tools/kernel_ab, round 1) — keep cached stores here */
-  const bool nt = false;
-  const long nChunks = (sg.n + 63) / 64;
-  for (long chunk = (long)(blockIdx.x - sg.blkBase); chunk < nChunks;
-       chunk += sg.nBlocks) {
-    const long site = chunk * 64 + lane;
-    const bool live = site < sg.n;
-    const long idx = site * 4 + cat;
+  for (long idx = (long)(blockIdx.x - sg.blkBase) * NV_BLOCK + tid;
+       idx < units; idx += (long)sg.nBlocks * NV_BLOCK) {
+    const long site = idx >> 2;
+    const int cat = (int)(idx & 3);
     double xl[20], xr[20], acc[20];
     int code1 = 0, code2 = 0;
-    if (live) {
-      if (TC == EXAML_INNER_INNER) {
+    if (TC == EXAML_INNER_INNER) {
 #pragma unroll
-        for (int s = 0; s < 20; s += 4) {
-          const double4 a =
-              *reinterpret_cast<const double4 *>(&sg.x1[idx * 20 + s]);
-          const double4 b =
-              *reinterpret_cast<const double4 *>(&sg.x2[idx * 20 + s]);
-          xl[s] = a.x; xl[s + 1] = a.y; xl[s + 2] = a.z; xl[s + 3] = a.w;
-          xr[s] = b.x; xr[s + 1] = b.y; xr[s + 2] = b.z; xr[s + 3] = b.w;
-        }
-      } else if (TC == EXAML_TIP_INNER) {
-        code1 = sg.t1[site];
-#pragma unroll
-        for (int s = 0; s < 20; s += 4) {
-          const double4 b =
-              *reinterpret_cast<const double4 *>(&sg.x2[idx * 20 + s]);
-          xr[s] = b.x; xr[s + 1] = b.y; xr[s + 2] = b.z; xr[s + 3] = b.w;
-        }
-      } else {
-        code1 = sg.t1[site];
-        code2 = sg.t2[site];
+      for (int s = 0; s < 20; s += 4) {
+        const double4 a =
+            *reinterpret_cast<const double4 *>(&sg.x1[idx * 20 + s]);
+        const double4 b =
+            *reinterpret_cast<const double4 *>(&sg.x2[idx * 20 + s]);
+        xl[s] = a.x; xl[s + 1] = a.y; xl[s + 2] = a.z; xl[s + 3] = a.w;
+        xr[s] = b.x; xr[s + 1] = b.y; xr[s + 2] = b.z; xr[s + 3] = b.w;
       }
+    } else if (TC == EXAML_TIP_INNER) {
+      code1 = sg.t1[site];
 #pragma unroll
-      for (int s = 0; s < 20; s++) acc[s] = 0.0;
-      for (int l = 0; l < 20; l++) {
-        double u1, u2;
-        if (TC == EXAML_INNER_INNER) {
-          u1 = dot20o<FAST>(xl, &Pl[l * 20]);
-          u2 = dot20o<FAST>(xr, &Pr[l * 20]);
-        } else if (TC == EXAML_TIP_INNER) {
-          u1 = sU1[80 * code1 + cat * 20 + l];
-          u2 = dot20o<FAST>(xr, &Pr[l * 20]);
-        } else {
-          u1 = sU1[80 * code1 + cat * 20 + l];
-          u2 = sU2[80 * code2 + cat * 20 + l];
-        }
-        const double t = u1 * u2;
+      for (int s = 0; s < 20; s += 4) {
+        const double4 b =
+            *reinterpret_cast<const double4 *>(&sg.x2[idx * 20 + s]);
+        xr[s] = b.x; xr[s + 1] = b.y; xr[s + 2] = b.z; xr[s + 3] = b.w;
+      }
+    } else {
+      code1 = sg.t1[site];
+      code2 = sg.t2[site];
+    }
 #pragma unroll
-        for (int s = 0; s < 20; s++) {
-          if (FAST)
-            acc[s] = fma(t, EV[l * 20 + s], acc[s]);
-          else
-            acc[s] += t * EV[l * 20 + s];
-        }
+    for (int s = 0; s < 20; s++) acc[s] = 0.0;
+    for (int l = 0; l < 20; l++) {
+      double u1, u2;
+      if (TC == EXAML_INNER_INNER) {
+        u1 = dot20o<FAST>(xl, &sL[cat * CSTR + l * 20]);
+        u2 = dot20o<FAST>(xr, &sR[cat * CSTR + l * 20]);
+      } else if (TC == EXAML_TIP_INNER) {
+        u1 = sU1[80 * code1 + cat * 20 + l];
+        u2 = dot20o<FAST>(xr, &sR[cat * CSTR + l * 20]);
+      } else {
+        u1 = sU1[80 * code1 + cat * 20 + l];
+        u2 = sU2[80 * code2 + cat * 20 + l];
+      }
+      const double t = u1 * u2;
+#pragma unroll
+      for (int s = 0; s < 20; s++) {
+        if (FAST)
+          acc[s] = fma(t, sEV[l * 20 + s], acc[s]);
+        else
+          acc[s] += t * sEV[l * 20 + s];
       }
     }
 
     if (TC != EXAML_TIP_TIP) {
       bool small = true;
-      if (live) {
 #pragma unroll
-        for (int s = 0; s < 20; s++)
-          small &= (fabs(acc[s]) < MINLIKELIHOOD);
-      }
+      for (int s = 0; s < 20; s++)
+        small &= (fabs(acc[s]) < MINLIKELIHOOD);
       const unsigned long long m = __ballot(small);
-      if (lane == 0) sSmall[cat] = m;
-      __syncthreads();
-      const unsigned long long all4 =
-          sSmall[0] & sSmall[1] & sSmall[2] & sSmall[3];
-      __syncthreads();
-      if (live && ((all4 >> lane) & 1ULL)) {
+      if (((m >> (lane & ~3)) & 0xFULL) == 0xFULL) {
 #pragma unroll
         for (int s = 0; s < 20; s++) acc[s] *= TWOTOTHE256;
-        if (cat == 0) atomicAdd(sg.inc, (unsigned int)sg.wgt[site]);
+        if ((lane & 3) == 0)
+          atomicAdd(sg.inc, (unsigned int)sg.wgt[site]);
       }
     }
-    if (live) {
 #pragma unroll
-      for (int s = 0; s < 20; s += 4) {
-        const double4 v = make_double4(acc[s], acc[s + 1], acc[s + 2],
-                                       acc[s + 3]);
-        if (nt)
-          __builtin_nontemporal_store(
-              (v4d){v.x, v.y, v.z, v.w},
-              reinterpret_cast<v4d *>(&sg.x3[idx * 20 + s]));
-        else
-          *reinterpret_cast<double4 *>(&sg.x3[idx * 20 + s]) = v;
-      }
-    }
+    for (int s = 0; s < 20; s += 4)
+      *reinterpret_cast<double4 *>(&sg.x3[idx * 20 + s]) =
+          make_double4(acc[s], acc[s + 1], acc[s + 2], acc[s + 3]);
   }
 }
 
@@ -5300,4 +5267,752 @@ __global__ __launch_bounds__(NV_BLOCK) void k_core_prot_mseg(
     partials[2 * blockIdx.x] = s1;
     partials[2 * blockIdx.x + 1] = s2;
   }
+}
+
+/* ===========================================================================
+ * -S (saveMemory, SEV) GPU kernels for the remaining families:
+ *   protein GTRGAMMA — newviewGTRGAMMAPROT_AVX_GAPPED_SAVE
+ *     (avxLikelihood.c:3125), evaluateGTRGAMMAPROT_GAPPED_SAVE
+ *     (evaluateGenericSpecial.c:1291), sumGAMMAPROT_GAPPED_SAVE
+ *     (makenewzGenericSpecial.c:1896)
+ *   DNA CAT — newviewGTRCAT_AVX_GAPPED_SAVE (avxLikelihood.c:2306) family
+ *   protein CAT — newviewGTRCATPROT_AVX_GAPPED_SAVE (:2607) family
+ * Same design as the DNA GAMMA SAVE kernels above: per-node gap bit
+ * vectors, compacted CLV slabs with O(1) per-thread indexing via per-word
+ * non-gap prefixes (save_cidx), per-node gap columns, and the saveMem
+ * rate-1.0 P pair at slot maxCats for the CAT families (makeP's saveMem
+ * branch, newviewGenericSpecial.c:140-165).
+ * ==========================================================================*/
+
+/* makeP with the saveMem extra rate-1.0 slot (oracle_make_p_save twin) */
+extern "C" void examl_host_make_p_save(double z1, double z2,
+                                       const double *rptr, const double *EI,
+                                       const double *EIGN, int numCats,
+                                       double *left, double *right,
+                                       int maxCats, int states) {
+  const int sq = states * states;
+  double d1[64], d2[64];
+  for (int i = 0; i <= maxCats; i++) {
+    const int slot = i;
+    const double r = (i == maxCats) ? 1.0 : (i < numCats ? rptr[i] : 0.0);
+    if (i >= numCats && i != maxCats) continue;
+    for (int j = 1; j < states; j++) {
+      d1[j] = exp(r * (EIGN[j] * z1));
+      d2[j] = exp(r * (EIGN[j] * z2));
+    }
+    for (int j = 0; j < states; j++) {
+      left[sq * slot + states * j] = 1.0;
+      right[sq * slot + states * j] = 1.0;
+      for (int k = 1; k < states; k++) {
+        left[sq * slot + states * j + k] = d1[k] * EI[states * j + k];
+        right[sq * slot + states * j + k] = d2[k] * EI[states * j + k];
+      }
+    }
+  }
+}
+
+/* ---- protein GAMMA SAVE --------------------------------------------------*/
+
+/* gap column (span 80) from the child gap columns / the undetermined
+ * tipVector row 22; TT never scales */
+template <int TC>
+__global__ void k_gapcol_prot_save(const double *__restrict__ P,
+                                   const double *__restrict__ EV,
+                                   const double *__restrict__ tipVec,
+                                   const double *__restrict__ x1_gapcol,
+                                   const double *__restrict__ x2_gapcol,
+                                   double *__restrict__ x3_gapcol,
+                                   int *__restrict__ scaleGap) {
+  if (threadIdx.x != 0 || blockIdx.x != 0) return;
+  const double *L = P, *R = P + 1600;
+  const double *tvU = &tipVec[22 * 20];
+  double xv[80];
+  for (int k = 0; k < 4; k++) {
+    double acc[20];
+    for (int s = 0; s < 20; s++) acc[s] = 0.0;
+    for (int l = 0; l < 20; l++) {
+      double u1, u2;
+      if (TC == EXAML_TIP_TIP) {
+        u1 = dot20o<false>(tvU, &L[k * 400 + l * 20]);
+        u2 = dot20o<false>(tvU, &R[k * 400 + l * 20]);
+      } else if (TC == EXAML_TIP_INNER) {
+        u1 = dot20o<false>(tvU, &L[k * 400 + l * 20]);
+        u2 = dot20o<false>(&x2_gapcol[20 * k], &R[k * 400 + l * 20]);
+      } else {
+        u1 = dot20o<false>(&x1_gapcol[20 * k], &L[k * 400 + l * 20]);
+        u2 = dot20o<false>(&x2_gapcol[20 * k], &R[k * 400 + l * 20]);
+      }
+      const double t = u1 * u2;
+      for (int s = 0; s < 20; s++) acc[s] += t * EV[20 * l + s];
+    }
+    for (int s = 0; s < 20; s++) xv[k * 20 + s] = acc[s];
+  }
+  int scale = (TC != EXAML_TIP_TIP);
+  for (int s = 0; s < 80 && scale; s++)
+    if (!(fabs(xv[s]) < MINLIKELIHOOD)) scale = 0;
+  if (scale)
+    for (int s = 0; s < 80; s++) xv[s] *= TWOTOTHE256;
+  for (int s = 0; s < 80; s++) x3_gapcol[s] = xv[s];
+  *scaleGap = scale;
+}
+
+/* one (site,cat) per lane; the site's 4 lanes vote the rescale ballot */
+template <int TC>
+__global__ __launch_bounds__(NV_BLOCK) void k_newview_prot_save(
+    const double *__restrict__ x1, const double *__restrict__ x2,
+    double *__restrict__ x3, const double *__restrict__ P,
+    const double *__restrict__ EV, const double *__restrict__ tipVec,
+    const unsigned char *__restrict__ tipX1,
+    const unsigned char *__restrict__ tipX2, const int *__restrict__ wgt,
+    long n, unsigned int *__restrict__ scalerInc,
+    const unsigned int *__restrict__ g1, const unsigned int *__restrict__ g2,
+    const unsigned int *__restrict__ g3, const int *__restrict__ pre1,
+    const int *__restrict__ pre2, const int *__restrict__ pre3,
+    const double *__restrict__ x1_gapcol, const double *__restrict__ x2_gapcol,
+    const double *__restrict__ x3_gapcol, const int *__restrict__ scaleGap) {
+  __shared__ double sTV[TC != EXAML_INNER_INNER ? 460 : 1];
+  __shared__ double sU1[TC != EXAML_INNER_INNER ? 1840 : 1];
+  __shared__ double sU2[TC == EXAML_TIP_TIP ? 1840 : 1];
+  const int tid = threadIdx.x;
+  if (TC != EXAML_INNER_INNER) {
+    for (int j = tid; j < 460; j += NV_BLOCK) sTV[j] = tipVec[j];
+    __syncthreads();
+    for (int j = tid; j < 23 * 80; j += NV_BLOCK) {
+      const int code = j / 80, k = j % 80;
+      const int kc = k / 20, kl = k % 20;
+      sU1[j] = dot20o<false>(&sTV[20 * code], &P[kc * 400 + kl * 20]);
+      if (TC == EXAML_TIP_TIP)
+        sU2[j] =
+            dot20o<false>(&sTV[20 * code], &P[1600 + kc * 400 + kl * 20]);
+    }
+    __syncthreads();
+  }
+  const int lane = tid & 63;
+  const long units = n * 4;
+  for (long idx = (long)blockIdx.x * NV_BLOCK + tid; idx < units;
+       idx += (long)gridDim.x * NV_BLOCK) {
+    const long i = idx >> 2;
+    const int cat = (int)(idx & 3);
+    const bool gap3 = (g3[i / 32] >> (i % 32)) & 1u;
+    if (gap3) {
+      if (TC != EXAML_TIP_TIP && *scaleGap && (lane & 3) == 0)
+        atomicAdd(scalerInc, (unsigned int)wgt[i]);
+      continue;
+    }
+    const double *xl = nullptr, *xr = nullptr;
+    int code1 = 0;
+    if (TC == EXAML_TIP_TIP) {
+      code1 = tipX1[i];
+    } else if (TC == EXAML_TIP_INNER) {
+      code1 = tipX1[i];
+      xr = ((g2[i / 32] >> (i % 32)) & 1u)
+               ? &x2_gapcol[cat * 20]
+               : &x2[save_cidx(g2, pre2, i) * 80 + cat * 20];
+    } else {
+      xl = ((g1[i / 32] >> (i % 32)) & 1u)
+               ? &x1_gapcol[cat * 20]
+               : &x1[save_cidx(g1, pre1, i) * 80 + cat * 20];
+      xr = ((g2[i / 32] >> (i % 32)) & 1u)
+               ? &x2_gapcol[cat * 20]
+               : &x2[save_cidx(g2, pre2, i) * 80 + cat * 20];
+    }
+    double acc[20];
+#pragma unroll
+    for (int s = 0; s < 20; s++) acc[s] = 0.0;
+    for (int l = 0; l < 20; l++) {
+      double u1, u2;
+      if (TC == EXAML_TIP_TIP) {
+        u1 = sU1[80 * code1 + cat * 20 + l];
+        u2 = sU2[80 * tipX2[i] + cat * 20 + l];
+      } else if (TC == EXAML_TIP_INNER) {
+        u1 = sU1[80 * code1 + cat * 20 + l];
+        u2 = dot20o<false>(xr, &P[1600 + cat * 400 + l * 20]);
+      } else {
+        u1 = dot20o<false>(xl, &P[cat * 400 + l * 20]);
+        u2 = dot20o<false>(xr, &P[1600 + cat * 400 + l * 20]);
+      }
+      const double t = u1 * u2;
+#pragma unroll
+      for (int s = 0; s < 20; s++) acc[s] += t * EV[l * 20 + s];
+    }
+    if (TC != EXAML_TIP_TIP) {
+      bool small = true;
+#pragma unroll
+      for (int s = 0; s < 20; s++)
+        small &= (fabs(acc[s]) < MINLIKELIHOOD);
+      const unsigned long long m = __ballot(small);
+      if (((m >> (lane & ~3)) & 0xFULL) == 0xFULL) {
+#pragma unroll
+        for (int s = 0; s < 20; s++) acc[s] *= TWOTOTHE256;
+        if ((lane & 3) == 0)
+          atomicAdd(scalerInc, (unsigned int)wgt[i]);
+      }
+    }
+    double *out = &x3[save_cidx(g3, pre3, i) * 80 + cat * 20];
+#pragma unroll
+    for (int s = 0; s < 20; s += 4)
+      *reinterpret_cast<double4 *>(&out[s]) =
+          make_double4(acc[s], acc[s + 1], acc[s + 2], acc[s + 3]);
+  }
+}
+
+template <bool TIP>
+__global__ __launch_bounds__(NV_BLOCK) void k_evaluate_prot_save(
+    const double *__restrict__ x1, const double *__restrict__ x2,
+    const double *__restrict__ tipVec, const unsigned char *__restrict__ tipX1,
+    const int *__restrict__ wgt, const double *__restrict__ diag, long n,
+    double *__restrict__ partials, const unsigned int *__restrict__ g1,
+    const unsigned int *__restrict__ g2, const int *__restrict__ pre1,
+    const int *__restrict__ pre2, const double *__restrict__ x1_gapcol,
+    const double *__restrict__ x2_gapcol) {
+  __shared__ double sD[80], sTV[TIP ? 460 : 1], sRed[NV_BLOCK / 64];
+  const int tid = threadIdx.x;
+  for (int j = tid; j < 80; j += NV_BLOCK) sD[j] = diag[j];
+  if (TIP)
+    for (int j = tid; j < 460; j += NV_BLOCK) sTV[j] = tipVec[j];
+  __syncthreads();
+  const int lane = tid & 63;
+  const long units = n * 4;
+  double acc = 0.0;
+  for (long idx = (long)blockIdx.x * NV_BLOCK + tid; idx < units;
+       idx += (long)gridDim.x * NV_BLOCK) {
+    const long i = idx >> 2;
+    const int cat = (int)(idx & 3);
+    const double *le, *ri;
+    if (TIP)
+      le = &sTV[20 * tipX1[i]];
+    else
+      le = ((g1[i / 32] >> (i % 32)) & 1u)
+               ? &x1_gapcol[cat * 20]
+               : &x1[save_cidx(g1, pre1, i) * 80 + cat * 20];
+    ri = ((g2[i / 32] >> (i % 32)) & 1u)
+             ? &x2_gapcol[cat * 20]
+             : &x2[save_cidx(g2, pre2, i) * 80 + cat * 20];
+    double t0 = 0, t1 = 0;
+#pragma unroll
+    for (int l = 0; l < 20; l += 2) {
+      t0 += le[l] * ri[l] * sD[cat * 20 + l];
+      t1 += le[l + 1] * ri[l + 1] * sD[cat * 20 + l + 1];
+    }
+    double p = t0 + t1;
+    p += __shfl_xor(p, 1);
+    p += __shfl_xor(p, 2);
+    if ((lane & 3) == 0) acc += (double)wgt[i] * log(0.25 * fabs(p));
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off);
+  if (lane == 0) sRed[tid >> 6] = acc;
+  __syncthreads();
+  if (tid == 0) {
+    double s = 0;
+#pragma unroll
+    for (int w = 0; w < NV_BLOCK / 64; w++) s += sRed[w];
+    partials[blockIdx.x] = s;
+  }
+}
+
+template <int TC>
+__global__ __launch_bounds__(NV_BLOCK) void k_sum_prot_save(
+    double *__restrict__ sum, const double *__restrict__ x1,
+    const double *__restrict__ x2, const double *__restrict__ tipVec,
+    const unsigned char *__restrict__ tipX1,
+    const unsigned char *__restrict__ tipX2, long n,
+    const unsigned int *__restrict__ g1, const unsigned int *__restrict__ g2,
+    const int *__restrict__ pre1, const int *__restrict__ pre2,
+    const double *__restrict__ x1_gapcol,
+    const double *__restrict__ x2_gapcol) {
+  __shared__ double sTV[TC != EXAML_INNER_INNER ? 460 : 1];
+  const int tid = threadIdx.x;
+  if (TC != EXAML_INNER_INNER) {
+    for (int j = tid; j < 460; j += NV_BLOCK) sTV[j] = tipVec[j];
+    __syncthreads();
+  }
+  const long units = n * 4;
+  for (long idx = (long)blockIdx.x * NV_BLOCK + tid; idx < units;
+       idx += (long)gridDim.x * NV_BLOCK) {
+    const long i = idx >> 2;
+    const int cat = (int)(idx & 3);
+    const double *a, *b;
+    if (TC == EXAML_TIP_TIP) {
+      a = &sTV[20 * tipX1[i]];
+      b = &sTV[20 * tipX2[i]];
+    } else if (TC == EXAML_TIP_INNER) {
+      a = &sTV[20 * tipX1[i]];
+      b = ((g2[i / 32] >> (i % 32)) & 1u)
+              ? &x2_gapcol[cat * 20]
+              : &x2[save_cidx(g2, pre2, i) * 80 + cat * 20];
+    } else {
+      a = ((g1[i / 32] >> (i % 32)) & 1u)
+              ? &x1_gapcol[cat * 20]
+              : &x1[save_cidx(g1, pre1, i) * 80 + cat * 20];
+      b = ((g2[i / 32] >> (i % 32)) & 1u)
+              ? &x2_gapcol[cat * 20]
+              : &x2[save_cidx(g2, pre2, i) * 80 + cat * 20];
+    }
+#pragma unroll
+    for (int l = 0; l < 20; l++)
+      sum[idx * 20 + l] = a[l] * b[l];
+  }
+}
+
+/* ---- CAT SAVE (DNA span 4 / protein span 20), thread per site ----------- */
+
+template <int STATES, int TC>
+__global__ __launch_bounds__(NV_BLOCK) void k_newview_cat_save(
+    const double *__restrict__ EV, const int *__restrict__ cptr,
+    const double *__restrict__ x1, const double *__restrict__ x2,
+    double *__restrict__ x3, const double *__restrict__ tipVec,
+    const unsigned char *__restrict__ tipX1,
+    const unsigned char *__restrict__ tipX2, const int *__restrict__ wgt,
+    long n, const double *__restrict__ P, int maxCats,
+    unsigned int *__restrict__ scalerInc,
+    const unsigned int *__restrict__ g1, const unsigned int *__restrict__ g2,
+    const unsigned int *__restrict__ g3, const int *__restrict__ pre1,
+    const int *__restrict__ pre2, const int *__restrict__ pre3,
+    const double *__restrict__ x1_gapcol, const double *__restrict__ x2_gapcol,
+    const double *__restrict__ x3_gapcol, const int *__restrict__ scaleGap) {
+  constexpr int SQ = STATES * STATES;
+  __shared__ double sEV[SQ], sTV[STATES == 4 ? 64 : 460];
+  const int tid = threadIdx.x;
+  for (int j = tid; j < SQ; j += NV_BLOCK) sEV[j] = EV[j];
+  for (int j = tid; j < (STATES == 4 ? 64 : 460); j += NV_BLOCK)
+    sTV[j] = tipVec[j];
+  __syncthreads();
+  const double *Pr = P + (long)(maxCats + 1) * SQ;
+  for (long i = (long)blockIdx.x * NV_BLOCK + tid; i < n;
+       i += (long)gridDim.x * NV_BLOCK) {
+    const bool gap3 = (g3[i / 32] >> (i % 32)) & 1u;
+    if (gap3) {
+      if (TC != EXAML_TIP_TIP && *scaleGap)
+        atomicAdd(scalerInc, (unsigned int)wgt[i]);
+      continue;
+    }
+    const int cat = cptr[i];
+    const double *a, *b, *le, *ri;
+    if (TC == EXAML_TIP_TIP) {
+      a = &sTV[STATES * tipX1[i]];
+      b = &sTV[STATES * tipX2[i]];
+      le = ((g1[i / 32] >> (i % 32)) & 1u) ? &P[(long)maxCats * SQ]
+                                           : &P[(long)cat * SQ];
+      ri = ((g2[i / 32] >> (i % 32)) & 1u) ? &Pr[(long)maxCats * SQ]
+                                           : &Pr[(long)cat * SQ];
+    } else if (TC == EXAML_TIP_INNER) {
+      a = &sTV[STATES * tipX1[i]];
+      le = ((g1[i / 32] >> (i % 32)) & 1u) ? &P[(long)maxCats * SQ]
+                                           : &P[(long)cat * SQ];
+      if ((g2[i / 32] >> (i % 32)) & 1u) {
+        ri = &Pr[(long)maxCats * SQ];
+        b = x2_gapcol;
+      } else {
+        ri = &Pr[(long)cat * SQ];
+        b = &x2[save_cidx(g2, pre2, i) * STATES];
+      }
+    } else {
+      if ((g1[i / 32] >> (i % 32)) & 1u) {
+        a = x1_gapcol;
+        le = &P[(long)maxCats * SQ];
+      } else {
+        a = &x1[save_cidx(g1, pre1, i) * STATES];
+        le = &P[(long)cat * SQ];
+      }
+      if ((g2[i / 32] >> (i % 32)) & 1u) {
+        b = x2_gapcol;
+        ri = &Pr[(long)maxCats * SQ];
+      } else {
+        b = &x2[save_cidx(g2, pre2, i) * STATES];
+        ri = &Pr[(long)cat * SQ];
+      }
+    }
+    double xv[STATES];
+#pragma unroll
+    for (int s = 0; s < STATES; s++) xv[s] = 0.0;
+    for (int l = 0; l < STATES; l++) {
+      double u1, u2;
+      if (STATES == 4) {
+        u1 = (a[0] * le[l * 4] + a[1] * le[l * 4 + 1]) +
+             (a[2] * le[l * 4 + 2] + a[3] * le[l * 4 + 3]);
+        u2 = (b[0] * ri[l * 4] + b[1] * ri[l * 4 + 1]) +
+             (b[2] * ri[l * 4 + 2] + b[3] * ri[l * 4 + 3]);
+      } else {
+        u1 = dot20o<false>(a, &le[l * 20]);
+        u2 = dot20o<false>(b, &ri[l * 20]);
+      }
+      const double t = u1 * u2;
+#pragma unroll
+      for (int s = 0; s < STATES; s++) xv[s] += t * sEV[l * STATES + s];
+    }
+    if (TC != EXAML_TIP_TIP) {
+      bool small = true;
+#pragma unroll
+      for (int s = 0; s < STATES; s++)
+        small &= (fabs(xv[s]) < MINLIKELIHOOD);
+      if (small) {
+#pragma unroll
+        for (int s = 0; s < STATES; s++) xv[s] *= TWOTOTHE256;
+        atomicAdd(scalerInc, (unsigned int)wgt[i]);
+      }
+    }
+    double *out = &x3[save_cidx(g3, pre3, i) * STATES];
+#pragma unroll
+    for (int s = 0; s < STATES; s++) out[s] = xv[s];
+  }
+}
+
+/* gap column for CAT SAVE: the rate-1.0 P pair at slot maxCats */
+template <int STATES, int TC>
+__global__ void k_gapcol_cat_save(const double *__restrict__ P, int maxCats,
+                                  const double *__restrict__ EV,
+                                  const double *__restrict__ tipVec,
+                                  const double *__restrict__ x1_gapcol,
+                                  const double *__restrict__ x2_gapcol,
+                                  double *__restrict__ x3_gapcol,
+                                  int *__restrict__ scaleGap) {
+  if (threadIdx.x != 0 || blockIdx.x != 0) return;
+  constexpr int SQ = STATES * STATES;
+  const double *le = &P[(long)maxCats * SQ];
+  const double *ri = &P[(long)(maxCats + 1) * SQ + (long)maxCats * SQ];
+  const double *undet = &tipVec[(STATES == 4 ? 15 : 22) * STATES];
+  const double *a = (TC == EXAML_INNER_INNER) ? x1_gapcol : undet;
+  const double *b = (TC == EXAML_TIP_TIP) ? undet : x2_gapcol;
+  double xv[STATES];
+  for (int s = 0; s < STATES; s++) xv[s] = 0.0;
+  for (int l = 0; l < STATES; l++) {
+    double u1, u2;
+    if (STATES == 4) {
+      u1 = (a[0] * le[l * 4] + a[1] * le[l * 4 + 1]) +
+           (a[2] * le[l * 4 + 2] + a[3] * le[l * 4 + 3]);
+      u2 = (b[0] * ri[l * 4] + b[1] * ri[l * 4 + 1]) +
+           (b[2] * ri[l * 4 + 2] + b[3] * ri[l * 4 + 3]);
+    } else {
+      u1 = dot20o<false>(a, &le[l * 20]);
+      u2 = dot20o<false>(b, &ri[l * 20]);
+    }
+    const double t = u1 * u2;
+    for (int s = 0; s < STATES; s++) xv[s] += t * EV[l * STATES + s];
+  }
+  int scale = (TC != EXAML_TIP_TIP);
+  for (int s = 0; s < STATES && scale; s++)
+    if (!(fabs(xv[s]) < MINLIKELIHOOD)) scale = 0;
+  if (scale)
+    for (int s = 0; s < STATES; s++) xv[s] *= TWOTOTHE256;
+  for (int s = 0; s < STATES; s++) x3_gapcol[s] = xv[s];
+  *scaleGap = scale;
+}
+
+template <int STATES, bool TIP>
+__global__ __launch_bounds__(NV_BLOCK) void k_evaluate_cat_save(
+    const int *__restrict__ cptr, const int *__restrict__ wgt,
+    const double *__restrict__ x1, const double *__restrict__ x2,
+    const double *__restrict__ tipVec, const unsigned char *__restrict__ tipX1,
+    long n, const double *__restrict__ diag, double *__restrict__ partials,
+    const unsigned int *__restrict__ g1, const unsigned int *__restrict__ g2,
+    const int *__restrict__ pre1, const int *__restrict__ pre2,
+    const double *__restrict__ x1_gapcol,
+    const double *__restrict__ x2_gapcol) {
+  __shared__ double sTV[STATES == 4 ? 64 : 460], sRed[NV_BLOCK / 64];
+  const int tid = threadIdx.x;
+  if (TIP)
+    for (int j = tid; j < (STATES == 4 ? 64 : 460); j += NV_BLOCK)
+      sTV[j] = tipVec[j];
+  if (TIP) __syncthreads();
+  const int lane = tid & 63;
+  double acc = 0.0;
+  for (long i = (long)blockIdx.x * NV_BLOCK + tid; i < n;
+       i += (long)gridDim.x * NV_BLOCK) {
+    const double *a, *b;
+    if (TIP)
+      a = &sTV[STATES * tipX1[i]];
+    else
+      a = ((g1[i / 32] >> (i % 32)) & 1u)
+              ? x1_gapcol
+              : &x1[save_cidx(g1, pre1, i) * STATES];
+    b = ((g2[i / 32] >> (i % 32)) & 1u)
+            ? x2_gapcol
+            : &x2[save_cidx(g2, pre2, i) * STATES];
+    const double *d = &diag[(long)STATES * cptr[i]];
+    double p;
+    if (STATES == 4) {
+      const double t0 = a[0] * b[0] * d[0] + a[2] * b[2] * d[2];
+      const double t1 = a[1] * b[1] * d[1] + a[3] * b[3] * d[3];
+      p = t0 + t1;
+    } else {
+      double t0 = 0, t1 = 0;
+#pragma unroll
+      for (int l = 0; l < 20; l += 2) {
+        t0 += a[l] * b[l] * d[l];
+        t1 += a[l + 1] * b[l + 1] * d[l + 1];
+      }
+      p = t0 + t1;
+    }
+    acc += (double)wgt[i] * log(fabs(p));
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off);
+  if (lane == 0) sRed[tid >> 6] = acc;
+  __syncthreads();
+  if (tid == 0) {
+    double s = 0;
+#pragma unroll
+    for (int w = 0; w < NV_BLOCK / 64; w++) s += sRed[w];
+    partials[blockIdx.x] = s;
+  }
+}
+
+template <int STATES, int TC>
+__global__ __launch_bounds__(NV_BLOCK) void k_sum_cat_save(
+    double *__restrict__ sum, const double *__restrict__ x1,
+    const double *__restrict__ x2, const double *__restrict__ tipVec,
+    const unsigned char *__restrict__ tipX1,
+    const unsigned char *__restrict__ tipX2, long n,
+    const unsigned int *__restrict__ g1, const unsigned int *__restrict__ g2,
+    const int *__restrict__ pre1, const int *__restrict__ pre2,
+    const double *__restrict__ x1_gapcol,
+    const double *__restrict__ x2_gapcol) {
+  __shared__ double sTV[TC != EXAML_INNER_INNER ? (STATES == 4 ? 64 : 460)
+                                                : 1];
+  const int tid = threadIdx.x;
+  if (TC != EXAML_INNER_INNER) {
+    for (int j = tid; j < (STATES == 4 ? 64 : 460); j += NV_BLOCK)
+      sTV[j] = tipVec[j];
+    __syncthreads();
+  }
+  for (long i = (long)blockIdx.x * NV_BLOCK + tid; i < n;
+       i += (long)gridDim.x * NV_BLOCK) {
+    const double *a, *b;
+    if (TC == EXAML_TIP_TIP) {
+      a = &sTV[STATES * tipX1[i]];
+      b = &sTV[STATES * tipX2[i]];
+    } else if (TC == EXAML_TIP_INNER) {
+      a = &sTV[STATES * tipX1[i]];
+      b = ((g2[i / 32] >> (i % 32)) & 1u)
+              ? x2_gapcol
+              : &x2[save_cidx(g2, pre2, i) * STATES];
+    } else {
+      a = ((g1[i / 32] >> (i % 32)) & 1u)
+              ? x1_gapcol
+              : &x1[save_cidx(g1, pre1, i) * STATES];
+      b = ((g2[i / 32] >> (i % 32)) & 1u)
+              ? x2_gapcol
+              : &x2[save_cidx(g2, pre2, i) * STATES];
+    }
+#pragma unroll
+    for (int j = 0; j < STATES; j++) sum[i * STATES + j] = a[j] * b[j];
+  }
+}
+
+/* ---- SAVE executors: prot GAMMA + CAT families -------------------------- */
+
+extern "C" int examl_hip_newview_prot_save(
+    int tipCase, const double *x1, const double *x2, double *x3,
+    const double *P, const double *EV, const double *tipVec,
+    const unsigned char *tipX1, const unsigned char *tipX2, const int *wgt,
+    long n, unsigned int *scalerInc, const unsigned int *g1,
+    const unsigned int *g2, const unsigned int *g3, const int *pre1,
+    const int *pre2, const int *pre3, const double *x1_gapcol,
+    const double *x2_gapcol, double *x3_gapcol, int *scaleGap,
+    void *stream) {
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  const int grid = grid_for(n * 4);
+#define PSAVE(TCV)                                                           \
+  do {                                                                       \
+    hipLaunchKernelGGL((k_gapcol_prot_save<TCV>), dim3(1), dim3(64), 0, s,   \
+                       P, EV, tipVec, x1_gapcol, x2_gapcol, x3_gapcol,       \
+                       scaleGap);                                            \
+    CHK(hipGetLastError());                                                  \
+    hipLaunchKernelGGL((k_newview_prot_save<TCV>), dim3(grid),               \
+                       dim3(NV_BLOCK), 0, s, x1, x2, x3, P, EV, tipVec,      \
+                       tipX1, tipX2, wgt, n, scalerInc, g1, g2, g3, pre1,    \
+                       pre2, pre3, x1_gapcol, x2_gapcol, x3_gapcol,          \
+                       scaleGap);                                            \
+  } while (0)
+  switch (tipCase) {
+  case EXAML_TIP_TIP: PSAVE(EXAML_TIP_TIP); break;
+  case EXAML_TIP_INNER: PSAVE(EXAML_TIP_INNER); break;
+  case EXAML_INNER_INNER: PSAVE(EXAML_INNER_INNER); break;
+  default:
+    snprintf(g_err, sizeof(g_err), "newview_prot_save: bad tipCase %d",
+             tipCase);
+    return -1;
+  }
+#undef PSAVE
+  CHK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int examl_hip_evaluate_prot_save(
+    int tipCase, const double *x1, const double *x2, const double *tipVec,
+    const unsigned char *tipX1, const int *wgt, const double *diag, long n,
+    const unsigned int *g1, const unsigned int *g2, const int *pre1,
+    const int *pre2, const double *x1_gapcol, const double *x2_gapcol,
+    int pNumber, int qNumber, const unsigned int *dev_scalers,
+    double *dev_partials, double *dev_lnl, void *stream) {
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  const int grid = grid_for(n * 4);
+  const double log_minlik = log(MINLIKELIHOOD);
+  const unsigned int *gsP = dev_scalers ? dev_scalers + pNumber : nullptr;
+  const unsigned int *gsQ = dev_scalers ? dev_scalers + qNumber : nullptr;
+  if (tipCase == EXAML_TIP_INNER)
+    hipLaunchKernelGGL((k_evaluate_prot_save<true>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, x1, x2, tipVec, tipX1, wgt,
+                       diag, n, dev_partials, g1, g2, pre1, pre2, x1_gapcol,
+                       x2_gapcol);
+  else
+    hipLaunchKernelGGL((k_evaluate_prot_save<false>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, x1, x2, tipVec, nullptr, wgt,
+                       diag, n, dev_partials, g1, g2, pre1, pre2, x1_gapcol,
+                       x2_gapcol);
+  CHK(hipGetLastError());
+  hipLaunchKernelGGL(k_reduce_lnl, dim3(1), dim3(NV_BLOCK), 0, s,
+                     dev_partials, grid, gsP, gsQ, log_minlik, dev_lnl);
+  CHK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int examl_hip_sum_prot_save(
+    int tipCase, double *dev_sum, const double *x1, const double *x2,
+    const double *tipVec, const unsigned char *tipX1,
+    const unsigned char *tipX2, long n, const unsigned int *g1,
+    const unsigned int *g2, const int *pre1, const int *pre2,
+    const double *x1_gapcol, const double *x2_gapcol, void *stream) {
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  const int grid = grid_for(n * 4);
+  switch (tipCase) {
+  case EXAML_TIP_TIP:
+    hipLaunchKernelGGL((k_sum_prot_save<EXAML_TIP_TIP>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, dev_sum, x1, x2, tipVec, tipX1,
+                       tipX2, n, g1, g2, pre1, pre2, x1_gapcol, x2_gapcol);
+    break;
+  case EXAML_TIP_INNER:
+    hipLaunchKernelGGL((k_sum_prot_save<EXAML_TIP_INNER>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, dev_sum, x1, x2, tipVec, tipX1,
+                       tipX2, n, g1, g2, pre1, pre2, x1_gapcol, x2_gapcol);
+    break;
+  default:
+    hipLaunchKernelGGL((k_sum_prot_save<EXAML_INNER_INNER>), dim3(grid),
+                       dim3(NV_BLOCK), 0, s, dev_sum, x1, x2, tipVec, tipX1,
+                       tipX2, n, g1, g2, pre1, pre2, x1_gapcol, x2_gapcol);
+  }
+  CHK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int examl_hip_newview_cat_save(
+    int states, int tipCase, const double *EV, const int *cptr,
+    const double *x1, const double *x2, double *x3, const double *tipVec,
+    const unsigned char *tipX1, const unsigned char *tipX2, const int *wgt,
+    long n, const double *P, int maxCats, unsigned int *scalerInc,
+    const unsigned int *g1, const unsigned int *g2, const unsigned int *g3,
+    const int *pre1, const int *pre2, const int *pre3,
+    const double *x1_gapcol, const double *x2_gapcol, double *x3_gapcol,
+    int *scaleGap, void *stream) {
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  const int grid = grid_for(n);
+#define CSAVE(SV, TCV)                                                       \
+  do {                                                                       \
+    hipLaunchKernelGGL((k_gapcol_cat_save<SV, TCV>), dim3(1), dim3(64), 0,   \
+                       s, P, maxCats, EV, tipVec, x1_gapcol, x2_gapcol,      \
+                       x3_gapcol, scaleGap);                                 \
+    CHK(hipGetLastError());                                                  \
+    hipLaunchKernelGGL((k_newview_cat_save<SV, TCV>), dim3(grid),            \
+                       dim3(NV_BLOCK), 0, s, EV, cptr, x1, x2, x3, tipVec,   \
+                       tipX1, tipX2, wgt, n, P, maxCats, scalerInc, g1, g2,  \
+                       g3, pre1, pre2, pre3, x1_gapcol, x2_gapcol,           \
+                       x3_gapcol, scaleGap);                                 \
+  } while (0)
+  if (states == 4) {
+    switch (tipCase) {
+    case EXAML_TIP_TIP: CSAVE(4, EXAML_TIP_TIP); break;
+    case EXAML_TIP_INNER: CSAVE(4, EXAML_TIP_INNER); break;
+    default: CSAVE(4, EXAML_INNER_INNER);
+    }
+  } else {
+    switch (tipCase) {
+    case EXAML_TIP_TIP: CSAVE(20, EXAML_TIP_TIP); break;
+    case EXAML_TIP_INNER: CSAVE(20, EXAML_TIP_INNER); break;
+    default: CSAVE(20, EXAML_INNER_INNER);
+    }
+  }
+#undef CSAVE
+  CHK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int examl_hip_evaluate_cat_save(
+    int states, const int *cptr, const int *wgt, const double *x1,
+    const double *x2, const double *tipVec, const unsigned char *tipX1,
+    long n, const double *diag, const unsigned int *g1,
+    const unsigned int *g2, const int *pre1, const int *pre2,
+    const double *x1_gapcol, const double *x2_gapcol, int pNumber,
+    int qNumber, const unsigned int *dev_scalers, double *dev_partials,
+    double *dev_lnl, void *stream) {
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  const int grid = grid_for(n);
+  const double log_minlik = log(MINLIKELIHOOD);
+  const unsigned int *gsP = dev_scalers ? dev_scalers + pNumber : nullptr;
+  const unsigned int *gsQ = dev_scalers ? dev_scalers + qNumber : nullptr;
+  const bool tip = tipX1 != nullptr;
+  if (states == 4) {
+    if (tip)
+      hipLaunchKernelGGL((k_evaluate_cat_save<4, true>), dim3(grid),
+                         dim3(NV_BLOCK), 0, s, cptr, wgt, x1, x2, tipVec,
+                         tipX1, n, diag, dev_partials, g1, g2, pre1, pre2,
+                         x1_gapcol, x2_gapcol);
+    else
+      hipLaunchKernelGGL((k_evaluate_cat_save<4, false>), dim3(grid),
+                         dim3(NV_BLOCK), 0, s, cptr, wgt, x1, x2, tipVec,
+                         nullptr, n, diag, dev_partials, g1, g2, pre1, pre2,
+                         x1_gapcol, x2_gapcol);
+  } else {
+    if (tip)
+      hipLaunchKernelGGL((k_evaluate_cat_save<20, true>), dim3(grid),
+                         dim3(NV_BLOCK), 0, s, cptr, wgt, x1, x2, tipVec,
+                         tipX1, n, diag, dev_partials, g1, g2, pre1, pre2,
+                         x1_gapcol, x2_gapcol);
+    else
+      hipLaunchKernelGGL((k_evaluate_cat_save<20, false>), dim3(grid),
+                         dim3(NV_BLOCK), 0, s, cptr, wgt, x1, x2, tipVec,
+                         nullptr, n, diag, dev_partials, g1, g2, pre1,
+                         pre2, x1_gapcol, x2_gapcol);
+  }
+  CHK(hipGetLastError());
+  hipLaunchKernelGGL(k_reduce_lnl, dim3(1), dim3(NV_BLOCK), 0, s,
+                     dev_partials, grid, gsP, gsQ, log_minlik, dev_lnl);
+  CHK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int examl_hip_sum_cat_save(
+    int states, int tipCase, double *dev_sum, const double *x1,
+    const double *x2, const double *tipVec, const unsigned char *tipX1,
+    const unsigned char *tipX2, long n, const unsigned int *g1,
+    const unsigned int *g2, const int *pre1, const int *pre2,
+    const double *x1_gapcol, const double *x2_gapcol, void *stream) {
+  hipStream_t s = (hipStream_t)stream;
+  (void)hipGetLastError();
+  const int grid = grid_for(n);
+#define SSAVE(SV, TCV)                                                       \
+  hipLaunchKernelGGL((k_sum_cat_save<SV, TCV>), dim3(grid), dim3(NV_BLOCK), \
+                     0, s, dev_sum, x1, x2, tipVec, tipX1, tipX2, n, g1,     \
+                     g2, pre1, pre2, x1_gapcol, x2_gapcol)
+  if (states == 4) {
+    switch (tipCase) {
+    case EXAML_TIP_TIP: SSAVE(4, EXAML_TIP_TIP); break;
+    case EXAML_TIP_INNER: SSAVE(4, EXAML_TIP_INNER); break;
+    default: SSAVE(4, EXAML_INNER_INNER);
+    }
+  } else {
+    switch (tipCase) {
+    case EXAML_TIP_TIP: SSAVE(20, EXAML_TIP_TIP); break;
+    case EXAML_TIP_INNER: SSAVE(20, EXAML_TIP_INNER); break;
+    default: SSAVE(20, EXAML_INNER_INNER);
+    }
+  }
+#undef SSAVE
+  CHK(hipGetLastError());
+  return 0;
 }

@@ -465,6 +465,66 @@ int examl_hip_sum_dna_save(
     const double *x1_gapcol, const double *x2_gapcol, void *stream);
 
 
+/* ---- -S protein GTRGAMMA + CAT families --------------------------------
+ * Same compaction design as the DNA GAMMA SAVE entries above; each
+ * replaces the corresponding *_GAPPED_SAVE reference kernel:
+ *   newview prot  — newviewGTRGAMMAPROT_AVX_GAPPED_SAVE (avxLikelihood.c:3125)
+ *   evaluate prot — evaluateGTRGAMMAPROT_GAPPED_SAVE (evaluateGenericSpecial.c:1291)
+ *   sum prot      — sumGAMMAPROT_GAPPED_SAVE (makenewzGenericSpecial.c:1896)
+ *   CAT (states 4/20) — newviewGTRCAT_AVX_GAPPED_SAVE (avxLikelihood.c:2306),
+ *   newviewGTRCATPROT_AVX_GAPPED_SAVE (:2607) + their evaluate/sum twins.
+ * examl_host_make_p_save adds the saveMem rate-1.0 P pair at slot maxCats
+ * (makeP's saveMem branch, newviewGenericSpecial.c:140-165); CAT P blocks
+ * are [(maxCats+1)*S^2 left | (maxCats+1)*S^2 right]. */
+void examl_host_make_p_save(double z1, double z2, const double *rptr,
+                            const double *EI, const double *EIGN,
+                            int numCats, double *left, double *right,
+                            int maxCats, int states);
+int examl_hip_newview_prot_save(
+    int tipCase, const double *x1, const double *x2, double *x3,
+    const double *P /*3200*/, const double *EV, const double *tipVec,
+    const unsigned char *tipX1, const unsigned char *tipX2, const int *wgt,
+    long n, unsigned int *scalerInc, const unsigned int *g1,
+    const unsigned int *g2, const unsigned int *g3, const int *pre1,
+    const int *pre2, const int *pre3, const double *x1_gapcol,
+    const double *x2_gapcol, double *x3_gapcol, int *scaleGap, void *stream);
+int examl_hip_evaluate_prot_save(
+    int tipCase, const double *x1, const double *x2, const double *tipVec,
+    const unsigned char *tipX1, const int *wgt, const double *diag, long n,
+    const unsigned int *g1, const unsigned int *g2, const int *pre1,
+    const int *pre2, const double *x1_gapcol, const double *x2_gapcol,
+    int pNumber, int qNumber, const unsigned int *dev_scalers,
+    double *dev_partials, double *dev_lnl, void *stream);
+int examl_hip_sum_prot_save(
+    int tipCase, double *dev_sum, const double *x1, const double *x2,
+    const double *tipVec, const unsigned char *tipX1,
+    const unsigned char *tipX2, long n, const unsigned int *g1,
+    const unsigned int *g2, const int *pre1, const int *pre2,
+    const double *x1_gapcol, const double *x2_gapcol, void *stream);
+int examl_hip_newview_cat_save(
+    int states, int tipCase, const double *EV, const int *cptr,
+    const double *x1, const double *x2, double *x3, const double *tipVec,
+    const unsigned char *tipX1, const unsigned char *tipX2, const int *wgt,
+    long n, const double *P, int maxCats, unsigned int *scalerInc,
+    const unsigned int *g1, const unsigned int *g2, const unsigned int *g3,
+    const int *pre1, const int *pre2, const int *pre3,
+    const double *x1_gapcol, const double *x2_gapcol, double *x3_gapcol,
+    int *scaleGap, void *stream);
+int examl_hip_evaluate_cat_save(
+    int states, const int *cptr, const int *wgt, const double *x1,
+    const double *x2, const double *tipVec, const unsigned char *tipX1,
+    long n, const double *diag, const unsigned int *g1,
+    const unsigned int *g2, const int *pre1, const int *pre2,
+    const double *x1_gapcol, const double *x2_gapcol, int pNumber,
+    int qNumber, const unsigned int *dev_scalers, double *dev_partials,
+    double *dev_lnl, void *stream);
+int examl_hip_sum_cat_save(
+    int states, int tipCase, double *dev_sum, const double *x1,
+    const double *x2, const double *tipVec, const unsigned char *tipX1,
+    const unsigned char *tipX2, long n, const unsigned int *g1,
+    const unsigned int *g2, const int *pre1, const int *pre2,
+    const double *x1_gapcol, const double *x2_gapcol, void *stream);
+
 /* ---- Protein CAT (-m PSR on AA partitions) ---------------------------- */
 double examl_host_evaluate_partial_prot_cat(
     const void *ops, int numOps, int rootTipNumber, int rootQNumber,
