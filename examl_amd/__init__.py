@@ -148,6 +148,22 @@ def _bind(lib):
     lib.examl_hip_use_graphs.argtypes = [i]
     lib.examl_hip_graphs_clear.argtypes = []
     lib.examl_hip_fast_math.argtypes = [i]
+    # -S prot GAMMA + CAT families
+    lib.examl_host_make_p_save.argtypes = [d, d, p, p, p, i, p, p, i, i]
+    lib.examl_hip_newview_prot_save.argtypes = \
+        [i, p, p, p, p, p, p, p, p, p, l, p, p, p, p, p, p, p, p, p, p, p,
+         p]
+    lib.examl_hip_evaluate_prot_save.argtypes = \
+        [i, p, p, p, p, p, p, l, p, p, p, p, p, p, i, i, p, p, p, p]
+    lib.examl_hip_sum_prot_save.argtypes = \
+        [i, p, p, p, p, p, p, l, p, p, p, p, p, p, p]
+    lib.examl_hip_newview_cat_save.argtypes = \
+        [i, i, p, p, p, p, p, p, p, p, p, l, p, i, p, p, p, p, p, p, p, p,
+         p, p, p, p]
+    lib.examl_hip_evaluate_cat_save.argtypes = \
+        [i, p, p, p, p, p, p, l, p, p, p, p, p, p, p, i, i, p, p, p, p]
+    lib.examl_hip_sum_cat_save.argtypes = \
+        [i, i, p, p, p, p, p, p, l, p, p, p, p, p, p, p]
     # multi-partition fused executors
     lib.examl_hip_multi_create.argtypes = \
         [i, i, p, p, p, p, p, p, p, p, p, i, p]
@@ -189,13 +205,15 @@ from .model import DnaGtrModel, Lg4Model, ProtGtrModel  # noqa: E402
 from .tree import PhyloTree             # noqa: E402
 from .spr import SprSearch, SprTree     # noqa: E402
 from .engine import (DnaCatEngine, DnaGammaEngine, Lg4Engine,  # noqa: E402
-                     MultiDnaEngine, ProtCatEngine, SaveDnaEngine)
+                     MultiDnaEngine, ProtCatEngine, SaveCatEngine,
+                     SaveDnaEngine, SaveProtEngine)
 
 __all__ = [
     "lib", "check", "TravEntry", "DnaGtrModel", "ProtGtrModel",
     "Lg4Model", "PhyloTree", "SprSearch", "SprTree",
     "DnaGammaEngine", "DnaCatEngine", "Lg4Engine", "SaveDnaEngine",
-    "ProtCatEngine", "MultiDnaEngine",
+    "ProtCatEngine", "MultiDnaEngine", "SaveProtEngine",
+    "SaveCatEngine",
     "TIP_TIP", "TIP_INNER",
     "INNER_INNER", "ZMIN", "ZMAX",
 ]

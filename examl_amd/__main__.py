@@ -97,7 +97,8 @@ def _build_engines(parts, opts, device):
                                    use_median=opts["a"])
             if psr:
                 w = p.upper - p.lower
-                engines.append(ea.DnaCatEngine(
+                cat_cls = ea.SaveCatEngine if opts["S"] else ea.DnaCatEngine
+                engines.append(cat_cls(
                     p.tips, p.wgt, model, np.zeros(w, dtype=np.int32),
                     np.array([1.0]), device=device))
             elif opts["S"]:
@@ -133,12 +134,13 @@ def _build_engines(parts, opts, device):
                                     1.0, use_median=opts["a"])
             if psr:
                 w = p.upper - p.lower
-                engines.append(ea.ProtCatEngine(
+                cat_cls = ea.SaveCatEngine if opts["S"] else ea.ProtCatEngine
+                engines.append(cat_cls(
                     p.tips, p.wgt, model, np.zeros(w, dtype=np.int32),
                     np.array([1.0]), device=device))
             elif opts["S"]:
-                sys.exit("-S for protein partitions is not yet available "
-                         "on the GPU engines (GAMMA protein -S: round 2)")
+                engines.append(ea.SaveProtEngine(p.tips, p.wgt, model,
+                                                 device=device))
             else:
                 engines.append(ea.DnaGammaEngine(p.tips, p.wgt, model,
                                                  device=device))

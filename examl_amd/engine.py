@@ -935,3 +935,391 @@ class MultiDnaEngine:
             else:
                 outer_converged = True
         return z
+
+
+class SaveProtEngine(SaveDnaEngine):
+    """-S protein GTRGAMMA engine (span 80): the GPU side of the
+    *_GAPPED_SAVE protein family (newviewGTRGAMMAPROT_AVX_GAPPED_SAVE,
+    avxLikelihood.c:3125; evaluate :1291; sum makenewzGenericSpecial.c:1896)
+    with the same per-node gap-vector + prefix-compaction design as
+    SaveDnaEngine; undetermined AA code 22 (gapOffset 440)."""
+
+    def __init__(self, tips, wgt, model, device="cuda", max_ops=None):
+        assert model.states == 20
+        # SaveDnaEngine.__init__ asserts states == 4; replicate its setup
+        DnaGammaEngine.__init__(self, tips, wgt, model, device=device,
+                                max_ops=max_ops)
+        dev = self.device
+        self.d_clv = None
+        self.gvl = self.width // 32 + 1
+        nn = 2 * self.ntips
+        host_gap = np.zeros((nn, self.gvl), dtype=np.uint32)
+        for t in range(1, self.ntips + 1):
+            idx = np.nonzero(tips[t] == 22)[0]  # undetermined AA
+            np.bitwise_or.at(host_gap[t], idx // 32,
+                             (np.uint32(1) << (idx % 32).astype(np.uint32)))
+        self.d_gap = torch.from_numpy(host_gap.view(np.int32)).to(dev)
+        host_pre = np.zeros((nn, self.gvl + 1), dtype=np.int32)
+        for t in range(1, self.ntips + 1):
+            cnt = np.zeros(self.gvl, dtype=np.int64)
+            for w in range(self.gvl):
+                lo, hi = w * 32, min((w + 1) * 32, self.width)
+                bits = int(host_gap[t, w])
+                cnt[w] = (hi - lo) - bin(bits & ((1 << (hi - lo)) - 1)
+                                         ).count("1")
+            host_pre[t, 1:] = np.cumsum(cnt)
+        self.d_pre = torch.from_numpy(host_pre).to(dev)
+        self.d_gapcol = torch.zeros(self.ninner * 80, dtype=torch.float64,
+                                    device=dev)
+        self.d_scalegap = torch.zeros(1, dtype=torch.int32, device=dev)
+        self.clv_slots = {}
+
+    def _gapcol_of(self, node_or_slot, is_tip):
+        if is_tip:
+            return ctypes.c_void_p(self.d_tipVector.data_ptr() +
+                                   22 * 20 * 8)
+        return ctypes.c_void_p(self.d_gapcol.data_ptr() +
+                               node_or_slot * 80 * 8)
+
+    def newview_traversal(self, entries):
+        if not entries:
+            return
+        m = self.model
+        L = lib()
+        s = self._stream()
+        self.d_inc.zero_()
+        for e in entries:
+            qz = math.log(e.qz) if e.qz > ZMIN else math.log(ZMIN)
+            rz = math.log(e.rz) if e.rz > ZMIN else math.log(ZMIN)
+            hostP = np.empty(3200)
+            L.examl_host_make_p(ctypes.c_double(qz), ctypes.c_double(rz),
+                                _np_vp(m.gammaRates), _np_vp(m.EI),
+                                _np_vp(m.EIGN), 4, _np_vp(hostP),
+                                ctypes.c_void_p(hostP.ctypes.data + 1600*8),
+                                20)
+            d_P = torch.from_numpy(hostP).to(self.device)
+            p, q, r = e.pNumber, e.qNumber, e.rNumber
+            check(L.examl_hip_gap_and_prefix(
+                self._gap_row(q), self._gap_row(r), self._gap_row(p),
+                self._pre_row(p), self.gvl, ctypes.c_long(self.width), s),
+                "gap_and_prefix")
+            total = int(self.d_pre[p, self.gvl].item())
+            x3 = self._slot_clv(e.x3Slot, max(total, 1) * 80)
+            q_tip = e.tipCase != INNER_INNER
+            r_tip = e.tipCase == TIP_TIP
+            x1 = None if q_tip else self.clv_slots[e.x1Slot]
+            x2 = None if r_tip else self.clv_slots[e.x2Slot]
+            t1 = (ctypes.c_void_p(self.d_tips.data_ptr() +
+                                  e.x1Slot * self.width) if q_tip else None)
+            t2 = (ctypes.c_void_p(self.d_tips.data_ptr() +
+                                  e.x2Slot * self.width) if r_tip else None)
+            check(L.examl_hip_newview_prot_save(
+                e.tipCase,
+                _vp(x1) if x1 is not None else None,
+                _vp(x2) if x2 is not None else None,
+                _vp(x3), _vp(d_P), _vp(self.d_EV), _vp(self.d_tipVector),
+                t1, t2, _vp(self.d_wgt), ctypes.c_long(self.width),
+                _vp(self.d_inc), self._gap_row(q), self._gap_row(r),
+                self._gap_row(p), self._pre_row(q), self._pre_row(r),
+                self._pre_row(p),
+                self._gapcol_of(q if q_tip else self.tree_slot(e.x1Slot),
+                                q_tip),
+                self._gapcol_of(r if r_tip else self.tree_slot(e.x2Slot),
+                                r_tip),
+                self._gapcol_of(e.x3Slot, False), _vp(self.d_scalegap), s),
+                "newview_prot_save")
+            self._finalize_scaler_one(p, q, r)
+
+    def evaluate_root(self, tree, p, q, z, all_reduce=False):
+        tc, x1s, x2s, tslot, _, pn, qn = self._root_case(tree, p, q)
+        self.d_lnl.zero_()
+        m = self.model
+        hostDiag = np.empty(80)
+        lib().examl_host_calc_diagptable(ctypes.c_double(z), 20, 4,
+                                         _np_vp(m.gammaRates),
+                                         _np_vp(m.EIGN), _np_vp(hostDiag))
+        d_diag = torch.from_numpy(hostDiag).to(self.device)
+        s = self._stream()
+        if tc == TIP_INNER:
+            x2 = self.clv_slots[x2s]
+            inner_node = p if tree.is_tip(q) else q
+            check(lib().examl_hip_evaluate_prot_save(
+                TIP_INNER, None, _vp(x2), _vp(self.d_tipVector),
+                ctypes.c_void_p(self.d_tips.data_ptr() + tslot * self.width),
+                _vp(self.d_wgt), _vp(d_diag), ctypes.c_long(self.width),
+                None, self._gap_row(inner_node), None,
+                self._pre_row(inner_node), None,
+                self._gapcol_of(x2s, False), pn, qn, _vp(self.d_scalers),
+                _vp(self.d_partials), _vp(self.d_lnl), s),
+                "evaluate_prot_save")
+        else:
+            x1 = self.clv_slots[x1s]
+            x2 = self.clv_slots[x2s]
+            check(lib().examl_hip_evaluate_prot_save(
+                INNER_INNER, _vp(x1), _vp(x2), _vp(self.d_tipVector), None,
+                _vp(self.d_wgt), _vp(d_diag), ctypes.c_long(self.width),
+                self._gap_row(p), self._gap_row(q), self._pre_row(p),
+                self._pre_row(q), self._gapcol_of(x1s, False),
+                self._gapcol_of(x2s, False), pn, qn, _vp(self.d_scalers),
+                _vp(self.d_partials), _vp(self.d_lnl), s),
+                "evaluate_prot_save")
+        if all_reduce and torch.distributed.is_initialized():
+            torch.distributed.all_reduce(self.d_lnl)
+        return self.d_lnl
+
+    def sum_root(self, tree, p, q):
+        if self.d_sum is None:
+            self.d_sum = torch.empty(self.width * 80, dtype=torch.float64,
+                                     device=self.device)
+        p_tip, q_tip = tree.is_tip(p), tree.is_tip(q)
+        s = self._stream()
+        if p_tip and q_tip:
+            check(lib().examl_hip_sum_prot_save(
+                TIP_TIP, _vp(self.d_sum), None, None, _vp(self.d_tipVector),
+                ctypes.c_void_p(self.d_tips.data_ptr() + p * self.width),
+                ctypes.c_void_p(self.d_tips.data_ptr() + q * self.width),
+                ctypes.c_long(self.width), None, None, None, None, None,
+                None, s), "sum_prot_save")
+            return
+        if q_tip or p_tip:
+            tip, inner = (q, p) if q_tip else (p, q)
+            sl = tree.clv_slot(inner)
+            check(lib().examl_hip_sum_prot_save(
+                TIP_INNER, _vp(self.d_sum), None, _vp(self.clv_slots[sl]),
+                _vp(self.d_tipVector),
+                ctypes.c_void_p(self.d_tips.data_ptr() + tip * self.width),
+                None, ctypes.c_long(self.width), None, self._gap_row(inner),
+                None, self._pre_row(inner), None,
+                self._gapcol_of(sl, False), s), "sum_prot_save")
+            return
+        s1, s2 = tree.clv_slot(p), tree.clv_slot(q)
+        check(lib().examl_hip_sum_prot_save(
+            INNER_INNER, _vp(self.d_sum), _vp(self.clv_slots[s1]),
+            _vp(self.clv_slots[s2]), _vp(self.d_tipVector), None, None,
+            ctypes.c_long(self.width), self._gap_row(p), self._gap_row(q),
+            self._pre_row(p), self._pre_row(q), self._gapcol_of(s1, False),
+            self._gapcol_of(s2, False), s), "sum_prot_save")
+
+
+class SaveCatEngine(DnaCatEngine):
+    """-S CAT (PSR) engine for DNA (span 4) and protein (span 20): the GPU
+    side of newviewGTRCAT_AVX_GAPPED_SAVE (avxLikelihood.c:2306) /
+    newviewGTRCATPROT_AVX_GAPPED_SAVE (:2607) and their evaluate/sum
+    twins, with the saveMem rate-1.0 P pair at slot maxCats (makeP's
+    saveMem branch, newviewGenericSpecial.c:140-165).  Per-node gap
+    vectors + prefix compaction as in SaveDnaEngine; CAT CLVs are span
+    `states` per site."""
+
+    MAXC = 25
+
+    def __init__(self, tips, wgt, model, cptr, per_site_rates,
+                 device="cuda", max_ops=None):
+        super().__init__(tips, wgt, model, cptr, per_site_rates,
+                         device=device, max_ops=max_ops)
+        st = self.states
+        self.SPAN = st
+        dev = self.device
+        self.d_clv = None
+        if st == 20:  # DnaCatEngine sized these for DNA CAT
+            MAXC = self.MAXC
+            self.d_diag = torch.empty(MAXC * 20, dtype=torch.float64,
+                                      device=dev)
+            self.d_dtab = torch.empty(MAXC * 20 + 40 + MAXC,
+                                      dtype=torch.float64, device=dev)
+        undet = 15 if st == 4 else 22
+        self._undet = undet
+        self.gvl = self.width // 32 + 1
+        nn = 2 * self.ntips
+        host_gap = np.zeros((nn, self.gvl), dtype=np.uint32)
+        for t in range(1, self.ntips + 1):
+            idx = np.nonzero(tips[t] == undet)[0]
+            np.bitwise_or.at(host_gap[t], idx // 32,
+                             (np.uint32(1) << (idx % 32).astype(np.uint32)))
+        self.d_gap = torch.from_numpy(host_gap.view(np.int32)).to(dev)
+        host_pre = np.zeros((nn, self.gvl + 1), dtype=np.int32)
+        for t in range(1, self.ntips + 1):
+            cnt = np.zeros(self.gvl, dtype=np.int64)
+            for w in range(self.gvl):
+                lo, hi = w * 32, min((w + 1) * 32, self.width)
+                bits = int(host_gap[t, w])
+                cnt[w] = (hi - lo) - bin(bits & ((1 << (hi - lo)) - 1)
+                                         ).count("1")
+            host_pre[t, 1:] = np.cumsum(cnt)
+        self.d_pre = torch.from_numpy(host_pre).to(dev)
+        self.d_gapcol = torch.zeros(self.ninner * st, dtype=torch.float64,
+                                    device=dev)
+        self.d_scalegap = torch.zeros(1, dtype=torch.int32, device=dev)
+        self.clv_slots = {}
+
+    # gap helpers (same layout as SaveDnaEngine)
+    def _gap_row(self, node):
+        return _vp(self.d_gap[node])
+
+    def _pre_row(self, node):
+        return _vp(self.d_pre[node])
+
+    def _gapcol_of(self, node_or_slot, is_tip):
+        if is_tip:
+            return ctypes.c_void_p(self.d_tipVector.data_ptr() +
+                                   self._undet * self.states * 8)
+        return ctypes.c_void_p(self.d_gapcol.data_ptr() +
+                               node_or_slot * self.states * 8)
+
+    def _slot_clv(self, slot, required):
+        t = self.clv_slots.get(slot)
+        if t is None or t.numel() != required:
+            t = torch.empty(required, dtype=torch.float64,
+                            device=self.device)
+            self.clv_slots[slot] = t
+        return t
+
+    def tree_slot(self, slot):
+        return slot
+
+    def _finalize_scaler_one(self, p, q, r):
+        inc = self.d_inc[0]
+        self.d_scalers[p] = self.d_scalers[q] + self.d_scalers[r] + inc
+        self.d_inc.zero_()
+
+    def newview_traversal(self, entries):
+        if not entries:
+            return
+        m = self.model
+        st = self.states
+        sq = st * st
+        L = lib()
+        s = self._stream()
+        self.d_inc.zero_()
+        for e in entries:
+            qz = math.log(e.qz) if e.qz > ZMIN else math.log(ZMIN)
+            rz = math.log(e.rz) if e.rz > ZMIN else math.log(ZMIN)
+            hostP = np.zeros(2 * (self.MAXC + 1) * sq)
+            L.examl_host_make_p_save(
+                ctypes.c_double(qz), ctypes.c_double(rz),
+                _np_vp(self.per_site_rates), _np_vp(m.EI), _np_vp(m.EIGN),
+                self.num_cats, _np_vp(hostP),
+                ctypes.c_void_p(hostP.ctypes.data + (self.MAXC + 1)*sq*8),
+                self.MAXC, st)
+            d_P = torch.from_numpy(hostP).to(self.device)
+            p, q, r = e.pNumber, e.qNumber, e.rNumber
+            check(L.examl_hip_gap_and_prefix(
+                self._gap_row(q), self._gap_row(r), self._gap_row(p),
+                self._pre_row(p), self.gvl, ctypes.c_long(self.width), s),
+                "gap_and_prefix")
+            total = int(self.d_pre[p, self.gvl].item())
+            x3 = self._slot_clv(e.x3Slot, max(total, 1) * st)
+            q_tip = e.tipCase != INNER_INNER
+            r_tip = e.tipCase == TIP_TIP
+            x1 = None if q_tip else self.clv_slots[e.x1Slot]
+            x2 = None if r_tip else self.clv_slots[e.x2Slot]
+            t1 = (ctypes.c_void_p(self.d_tips.data_ptr() +
+                                  e.x1Slot * self.width) if q_tip else None)
+            t2 = (ctypes.c_void_p(self.d_tips.data_ptr() +
+                                  e.x2Slot * self.width) if r_tip else None)
+            check(L.examl_hip_newview_cat_save(
+                st, e.tipCase, _vp(self.d_EV), _vp(self.d_cptr),
+                _vp(x1) if x1 is not None else None,
+                _vp(x2) if x2 is not None else None,
+                _vp(x3), _vp(self.d_tipVector), t1, t2, _vp(self.d_wgt),
+                ctypes.c_long(self.width), _vp(d_P), self.MAXC,
+                _vp(self.d_inc), self._gap_row(q), self._gap_row(r),
+                self._gap_row(p), self._pre_row(q), self._pre_row(r),
+                self._pre_row(p),
+                self._gapcol_of(q if q_tip else self.tree_slot(e.x1Slot),
+                                q_tip),
+                self._gapcol_of(r if r_tip else self.tree_slot(e.x2Slot),
+                                r_tip),
+                self._gapcol_of(e.x3Slot, False), _vp(self.d_scalegap), s),
+                "newview_cat_save")
+            self._finalize_scaler_one(p, q, r)
+
+    def evaluate_root(self, tree, p, q, z, all_reduce=False):
+        tc, x1s, x2s, tslot, _, pn, qn = self._root_case(tree, p, q)
+        self.d_lnl.zero_()
+        m = self.model
+        st = self.states
+        hostDiag = np.empty(self.num_cats * st)
+        lib().examl_host_calc_diagptable(
+            ctypes.c_double(z), st, self.num_cats,
+            _np_vp(self.per_site_rates), _np_vp(m.EIGN), _np_vp(hostDiag))
+        d_diag = torch.from_numpy(hostDiag).to(self.device)
+        s = self._stream()
+        if tc == TIP_INNER:
+            x2 = self.clv_slots[x2s]
+            inner_node = p if tree.is_tip(q) else q
+            check(lib().examl_hip_evaluate_cat_save(
+                st, _vp(self.d_cptr), _vp(self.d_wgt), None, _vp(x2),
+                _vp(self.d_tipVector),
+                ctypes.c_void_p(self.d_tips.data_ptr() + tslot * self.width),
+                ctypes.c_long(self.width), _vp(d_diag), None,
+                self._gap_row(inner_node), None, self._pre_row(inner_node),
+                None, self._gapcol_of(x2s, False), pn, qn,
+                _vp(self.d_scalers), _vp(self.d_partials), _vp(self.d_lnl),
+                s), "evaluate_cat_save")
+        else:
+            x1 = self.clv_slots[x1s]
+            x2 = self.clv_slots[x2s]
+            check(lib().examl_hip_evaluate_cat_save(
+                st, _vp(self.d_cptr), _vp(self.d_wgt), _vp(x1), _vp(x2),
+                _vp(self.d_tipVector), None, ctypes.c_long(self.width),
+                _vp(d_diag), self._gap_row(p), self._gap_row(q),
+                self._pre_row(p), self._pre_row(q),
+                self._gapcol_of(x1s, False), self._gapcol_of(x2s, False),
+                pn, qn, _vp(self.d_scalers), _vp(self.d_partials),
+                _vp(self.d_lnl), s), "evaluate_cat_save")
+        if all_reduce and torch.distributed.is_initialized():
+            torch.distributed.all_reduce(self.d_lnl)
+        return self.d_lnl
+
+    def sum_root(self, tree, p, q):
+        st = self.states
+        if self.d_sum is None:
+            self.d_sum = torch.empty(self.width * st, dtype=torch.float64,
+                                     device=self.device)
+        p_tip, q_tip = tree.is_tip(p), tree.is_tip(q)
+        s = self._stream()
+        if p_tip and q_tip:
+            check(lib().examl_hip_sum_cat_save(
+                st, TIP_TIP, _vp(self.d_sum), None, None,
+                _vp(self.d_tipVector),
+                ctypes.c_void_p(self.d_tips.data_ptr() + p * self.width),
+                ctypes.c_void_p(self.d_tips.data_ptr() + q * self.width),
+                ctypes.c_long(self.width), None, None, None, None, None,
+                None, s), "sum_cat_save")
+            return
+        if q_tip or p_tip:
+            tip, inner = (q, p) if q_tip else (p, q)
+            sl = tree.clv_slot(inner)
+            check(lib().examl_hip_sum_cat_save(
+                st, TIP_INNER, _vp(self.d_sum), None,
+                _vp(self.clv_slots[sl]), _vp(self.d_tipVector),
+                ctypes.c_void_p(self.d_tips.data_ptr() + tip * self.width),
+                None, ctypes.c_long(self.width), None,
+                self._gap_row(inner), None, self._pre_row(inner), None,
+                self._gapcol_of(sl, False), s), "sum_cat_save")
+            return
+        s1, s2 = tree.clv_slot(p), tree.clv_slot(q)
+        check(lib().examl_hip_sum_cat_save(
+            st, INNER_INNER, _vp(self.d_sum), _vp(self.clv_slots[s1]),
+            _vp(self.clv_slots[s2]), _vp(self.d_tipVector), None, None,
+            ctypes.c_long(self.width), self._gap_row(p), self._gap_row(q),
+            self._pre_row(p), self._pre_row(q), self._gapcol_of(s1, False),
+            self._gapcol_of(s2, False), s), "sum_cat_save")
+
+    def clv_bytes(self):
+        return sum(t.numel() * 8 for t in self.clv_slots.values())
+
+    def core_derivs_async(self, lz):
+        """NR derivatives on the dense sumBuffer: the dense CAT core
+        kernels apply unchanged (sum_cat_save writes densely)."""
+        if self.states == 4:
+            return super().core_derivs_async(lz)
+        m = self.model
+        self.d_out2.zero_()
+        check(lib().examl_hip_core_root_prot_cat(
+            ctypes.c_long(self.width), _vp(self.d_sum), _np_vp(m.EIGN),
+            _np_vp(self.per_site_rates), self.num_cats, ctypes.c_double(lz),
+            _vp(self.d_wgt), _vp(self.d_cptr), _vp(self.d_dtab),
+            _vp(self.d_partials), _vp(self.d_out2), self._stream()),
+            "core_root_prot_cat")
+        return self.d_out2

@@ -92,3 +92,49 @@ def test_engine_requires_gpu_no_silent_fallback():
     m = ea.DnaGtrModel.jukes_cantor()
     with pytest.raises(RuntimeError):
         ea.DnaGammaEngine(tips, wgt, m, device="cuda")
+
+
+def test_make_p_save_matches_oracle():
+    """examl_host_make_p_save (saveMem rate-1.0 pair at slot maxCats,
+    newviewGenericSpecial.c:140-165) is bit-identical to the oracle
+    restatement for DNA and protein."""
+    import ctypes as C
+    import math
+
+    import oracle as O
+    rng = np.random.default_rng(3)
+    for states, sq in ((4, 16), (20, 400)):
+        if states == 4:
+            m = ea.DnaGtrModel([0.3, 0.2, 0.26, 0.24],
+                               [1.2, 2.4, 0.7, 0.9, 3.1, 1.0], 0.8)
+        else:
+            aa = np.load(os.path.join(os.path.dirname(os.path.dirname(
+                os.path.abspath(__file__))), "examl_amd", "data",
+                "aa_models.npz"))
+            m = ea.ProtGtrModel(aa["frequencies"][4], aa["rates190"][4],
+                                0.9)
+        from tests.helpers import _model_arrays
+        EIGN, EV, EI, tipVector, _ = _model_arrays(m)
+        MAXC, nc = 25, 6
+        rates = np.sort(rng.uniform(0.05, 4.0, nc))
+        qz, rz = math.log(0.37), math.log(0.72)
+        lr = np.zeros((MAXC + 1) * sq)
+        rr = np.zeros((MAXC + 1) * sq)
+        O._orc.oracle_make_p_save(
+            C.c_double(qz), C.c_double(rz),
+            rates.ctypes.data_as(C.POINTER(C.c_double)),
+            EI.ctypes.data_as(C.POINTER(C.c_double)),
+            EIGN.ctypes.data_as(C.POINTER(C.c_double)), C.c_int(nc),
+            lr.ctypes.data_as(C.POINTER(C.c_double)),
+            rr.ctypes.data_as(C.POINTER(C.c_double)), C.c_int(MAXC),
+            C.c_int(states))
+        lh = np.zeros((MAXC + 1) * sq)
+        rh = np.zeros((MAXC + 1) * sq)
+        ea.lib().examl_host_make_p_save(
+            C.c_double(qz), C.c_double(rz),
+            rates.ctypes.data_as(C.c_void_p),
+            EI.ctypes.data_as(C.c_void_p),
+            EIGN.ctypes.data_as(C.c_void_p), nc,
+            lh.ctypes.data_as(C.c_void_p), rh.ctypes.data_as(C.c_void_p),
+            MAXC, states)
+        assert np.array_equal(lh, lr) and np.array_equal(rh, rr), states
