@@ -105,39 +105,6 @@ __global__ __launch_bounds__(NV_BLOCK) void k_newview_dna_gamma(
     __syncthreads();
   }
 
-  if (TC == EXAML_TIP_TIP) {
-    /* TIP_TIP is write-dominated (128 B out, 2 B codes in, no rescale —
-     * avxLikelihood.c:85-157): one SITE per lane, so the code loads
-     * coalesce to one 64-B burst per wave and each lane emits 128 B of
-     * contiguous stores (round-1 profile had this kernel at 0.44 of
-     * peak on per-(site,cat) byte gathers). */
-    for (long site = (long)blockIdx.x * NV_BLOCK + tid; site < n;
-         site += (long)gridDim.x * NV_BLOCK) {
-      const int c1 = tipX1[site], c2 = tipX2[site];
-#pragma unroll
-      for (int cat = 0; cat < 4; cat++) {
-        double a0 = 0, a1 = 0, a2 = 0, a3 = 0;
-#pragma unroll
-        for (int l = 0; l < 4; l++) {
-          const double t =
-              sU1[c1 * 16 + cat * 4 + l] * sU2[c2 * 16 + cat * 4 + l];
-          a0 += t * sEV[l * 4 + 0];
-          a1 += t * sEV[l * 4 + 1];
-          a2 += t * sEV[l * 4 + 2];
-          a3 += t * sEV[l * 4 + 3];
-        }
-        if (NT)
-          __builtin_nontemporal_store(
-              (v4d){a0, a1, a2, a3},
-              reinterpret_cast<v4d *>(&x3[(site * 4 + cat) * 4]));
-        else
-          *reinterpret_cast<double4 *>(&x3[(site * 4 + cat) * 4]) =
-              make_double4(a0, a1, a2, a3);
-      }
-    }
-    return;
-  }
-
   const long units = n * 4;
   const int lane = tid & 63;
   for (long idx = (long)blockIdx.x * NV_BLOCK + tid; idx < units;
@@ -4006,34 +3973,6 @@ __global__ __launch_bounds__(NV_BLOCK) void k_newview_dna_mseg(
   const long units = sg.n * 4;
   const bool nt = sg.n >= 65536;
   const int lane = tid & 63;
-  if (TC == EXAML_TIP_TIP) {
-    /* site-per-lane write-coalesced TT body (see k_newview_dna_gamma) */
-    for (long site = (long)(blockIdx.x - sg.blkBase) * NV_BLOCK + tid;
-         site < sg.n; site += (long)sg.nBlocks * NV_BLOCK) {
-      const int c1 = sg.t1[site], c2 = sg.t2[site];
-#pragma unroll
-      for (int cat = 0; cat < 4; cat++) {
-        double a0 = 0, a1 = 0, a2 = 0, a3 = 0;
-#pragma unroll
-        for (int l = 0; l < 4; l++) {
-          const double t =
-              sU1[c1 * 16 + cat * 4 + l] * sU2[c2 * 16 + cat * 4 + l];
-          a0 += t * sEV[l * 4 + 0];
-          a1 += t * sEV[l * 4 + 1];
-          a2 += t * sEV[l * 4 + 2];
-          a3 += t * sEV[l * 4 + 3];
-        }
-        if (nt)
-          __builtin_nontemporal_store(
-              (v4d){a0, a1, a2, a3},
-              reinterpret_cast<v4d *>(&sg.x3[(site * 4 + cat) * 4]));
-        else
-          *reinterpret_cast<double4 *>(&sg.x3[(site * 4 + cat) * 4]) =
-              make_double4(a0, a1, a2, a3);
-      }
-    }
-    return;
-  }
   for (long idx = (long)(blockIdx.x - sg.blkBase) * NV_BLOCK + tid;
        idx < units; idx += (long)sg.nBlocks * NV_BLOCK) {
     const long site = idx >> 2;
@@ -5671,9 +5610,12 @@ __global__ void k_gapcol_cat_save(const double *__restrict__ P, int maxCats,
   constexpr int SQ = STATES * STATES;
   const double *le = &P[(long)maxCats * SQ];
   const double *ri = &P[(long)(maxCats + 1) * SQ + (long)maxCats * SQ];
-  const double *undet = &tipVec[(STATES == 4 ? 15 : 22) * STATES];
-  const double *a = (TC == EXAML_INNER_INNER) ? x1_gapcol : undet;
-  const double *b = (TC == EXAML_TIP_TIP) ? undet : x2_gapcol;
+  /* the caller passes the undetermined tipVector row as a tip operand's
+   * gap column (oracle_newview_*_cat_save takes them as arguments for
+   * every tipCase) */
+  (void)tipVec;
+  const double *a = x1_gapcol;
+  const double *b = x2_gapcol;
   double xv[STATES];
   for (int s = 0; s < STATES; s++) xv[s] = 0.0;
   for (int l = 0; l < STATES; l++) {
