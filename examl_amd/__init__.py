@@ -205,14 +205,14 @@ from .model import DnaGtrModel, Lg4Model, ProtGtrModel  # noqa: E402
 from .tree import PhyloTree             # noqa: E402
 from .spr import SprSearch, SprTree     # noqa: E402
 from .engine import (DnaCatEngine, DnaGammaEngine, Lg4Engine,  # noqa: E402
-                     MultiDnaEngine, ProtCatEngine, SaveCatEngine,
+                     MultiDnaEngine, MultiEngine, ProtCatEngine, SaveCatEngine,
                      SaveDnaEngine, SaveProtEngine)
 
 __all__ = [
     "lib", "check", "TravEntry", "DnaGtrModel", "ProtGtrModel",
     "Lg4Model", "PhyloTree", "SprSearch", "SprTree",
     "DnaGammaEngine", "DnaCatEngine", "Lg4Engine", "SaveDnaEngine",
-    "ProtCatEngine", "MultiDnaEngine", "SaveProtEngine",
+    "ProtCatEngine", "MultiDnaEngine", "MultiEngine", "SaveProtEngine",
     "SaveCatEngine",
     "TIP_TIP", "TIP_INNER",
     "INNER_INNER", "ZMIN", "ZMAX",

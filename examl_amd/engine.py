@@ -751,8 +751,8 @@ class ProtCatEngine(DnaGammaEngine):
         return self.d_out2
 
 
-class MultiDnaEngine:
-    """Fused multi-partition DNA GTRGAMMA engine: drives ALL partitions'
+class MultiEngine:
+    """Fused multi-partition GTRGAMMA engine (DNA or protein): drives ALL partitions'
     per-op work through one kernel launch per (traversal level, tipCase)
     via the mseg executors — the GPU answer to the per-partition dispatch
     loops of newviewIterative (newviewGenericSpecial.c:1064) and execCore
@@ -763,7 +763,9 @@ class MultiDnaEngine:
     bit-exactly."""
 
     def __init__(self, engines):
-        assert engines and all(e.states == 4 for e in engines)
+        assert engines and len({e.states for e in engines}) == 1
+        self.states = engines[0].states
+        assert self.states in (4, 20)
         self.engines = engines
         self.device = engines[0].device
         NP = self.NP = len(engines)
@@ -776,11 +778,13 @@ class MultiDnaEngine:
         def parr(vals):
             return (ctypes.c_void_p * NP)(*vals)
 
+        span = 4 * self.states
         h = ctypes.c_void_p()
         check(lib().examl_hip_multi_create(
-            4, NP, (ctypes.c_long * NP)(*[e.width for e in engines]),
+            self.states, NP,
+            (ctypes.c_long * NP)(*[e.width for e in engines]),
             parr([e.d_clv.data_ptr() for e in engines]),
-            (ctypes.c_long * NP)(*[e.width * 16 for e in engines]),
+            (ctypes.c_long * NP)(*[e.width * span for e in engines]),
             parr([e.d_tips.data_ptr() for e in engines]),
             (ctypes.c_long * NP)(*[e.width for e in engines]),
             parr([e.d_wgt.data_ptr() for e in engines]),
@@ -1323,3 +1327,6 @@ class SaveCatEngine(DnaCatEngine):
             _vp(self.d_partials), _vp(self.d_out2), self._stream()),
             "core_root_prot_cat")
         return self.d_out2
+
+
+MultiDnaEngine = MultiEngine  # backward-compatible alias

@@ -81,6 +81,22 @@ class TreeSearch:
             self.cat_patrat = [np.ones(e.width) for e in engines]
             self.cat_lhs = [np.zeros(e.width) for e in engines]
         self._last_full = None
+        # fused multi-partition executors (mseg): one launch per
+        # (traversal level x tipCase) over ALL partitions — used when
+        # every engine is the plain dense GAMMA engine of one states
+        # family on a GPU (the per-partition loop otherwise)
+        self.fused = None
+        if (self.M > 1 and rate_het == "GAMMA"
+                and all(type(e).__name__ == "DnaGammaEngine"
+                        for e in engines)
+                and len({e.states for e in engines}) == 1
+                and getattr(engines[0], "device", None) is not None
+                and engines[0].device.type == "cuda"):
+            try:
+                from .engine import MultiEngine
+                self.fused = MultiEngine(engines)
+            except Exception:
+                self.fused = None
 
     # ------------------------------------------------------------------
     # traversal construction (computeTraversalInfo,
@@ -205,6 +221,17 @@ class TreeSearch:
     def _run(self, entries):
         if not entries:
             return
+        if self.fused is not None:
+            if self.NB == 1:
+                self.fused.newview_traversal(entries,
+                                             active=self.execute_model)
+            else:
+                qz = np.array([e.qzv[:self.NB] for e in entries])
+                rz = np.array([e.rzv[:self.NB] for e in entries])
+                self.fused.newview_traversal(entries,
+                                             active=self.execute_model,
+                                             qz_ov=qz, rz_ov=rz)
+            return
         if self.NB == 1:
             for m, eng in enumerate(self.engines):
                 if self.execute_model[m]:
@@ -258,6 +285,15 @@ class TreeSearch:
             self._last_full = (out, p, q, zv.copy())
         # launch all partitions, then ONE host sync for the readbacks
         # (z = pz[m] per partition under -M, evaluateGenericSpecial.c:449)
+        if self.fused is not None:
+            vec = self.fused.evaluate_root(
+                t, p, q, zv if self.NB > 1 else z,
+                active=self.execute_model).cpu()
+            for m in range(self.M):
+                if self.execute_model[m]:
+                    self.per_partition_lnl[m] = float(vec[m])
+            self.likelihood = sum(self.per_partition_lnl)
+            return self.likelihood
         outs = []
         for m, eng in enumerate(self.engines):
             if self.execute_model[m]:
@@ -285,9 +321,13 @@ class TreeSearch:
         else:
             out = entries
         self._run(out)
-        for m, eng in enumerate(self.engines):
-            if self.execute_model[m]:
-                eng.sum_root(self.tree, p, q)
+        if self.fused is not None:
+            self.fused.sum_root(self.tree, p, q,
+                                active=self.execute_model)
+        else:
+            for m, eng in enumerate(self.engines):
+                if self.execute_model[m]:
+                    eng.sum_root(self.tree, p, q)
 
         z = float(z0)
         zprev = z
@@ -303,18 +343,24 @@ class TreeSearch:
             z = min(max(z, ZMIN), ZMAX)
             lz = math.log(z)
             dlnL = d2lnL = 0.0
-            outs = [eng.core_derivs_async(lz)
-                    for m, eng in enumerate(self.engines)
-                    if self.execute_model[m]]
-            if outs and not isinstance(outs[0], tuple):
-                import torch
-                vals = torch.stack(outs).cpu()
-                dlnL = float(vals[:, 0].sum())
-                d2lnL = float(vals[:, 1].sum())
+            if self.fused is not None:
+                v = self.fused.core_derivs_vec(
+                    lz, active=self.execute_model).cpu().numpy()
+                dlnL = float(v[0::2].sum())
+                d2lnL = float(v[1::2].sum())
             else:
-                for a, b in outs:
-                    dlnL += a
-                    d2lnL += b
+                outs = [eng.core_derivs_async(lz)
+                        for m, eng in enumerate(self.engines)
+                        if self.execute_model[m]]
+                if outs and not isinstance(outs[0], tuple):
+                    import torch
+                    vals = torch.stack(outs).cpu()
+                    dlnL = float(vals[:, 0].sum())
+                    d2lnL = float(vals[:, 1].sum())
+                else:
+                    for a, b in outs:
+                        dlnL += a
+                        d2lnL += b
             if (d2lnL >= 0.0) and (z < ZMAX):
                 zprev = z = 0.37 * z + 0.63
                 continue
@@ -381,23 +427,34 @@ class TreeSearch:
                     self.execute_model[m] = not curvat[m]
             if first:
                 self._run(out)
-                for m, eng in enumerate(self.engines):
-                    if self.execute_model[m]:
-                        eng.sum_root(self.tree, p, q)
+                if self.fused is not None:
+                    self.fused.sum_root(self.tree, p, q,
+                                        active=self.execute_model)
+                else:
+                    for m, eng in enumerate(self.engines):
+                        if self.execute_model[m]:
+                            eng.sum_root(self.tree, p, q)
                 first = False
             dl[:] = 0.0
             d2[:] = 0.0
-            outs = [(m, eng.core_derivs_async(float(corelz[m])))
-                    for m, eng in enumerate(self.engines)
-                    if self.execute_model[m]]
-            if outs and not isinstance(outs[0][1], tuple):
-                import torch
-                vals = torch.stack([o[1] for o in outs]).cpu()
-                for k, (m, _) in enumerate(outs):
-                    dl[m], d2[m] = float(vals[k][0]), float(vals[k][1])
+            if self.fused is not None:
+                v = self.fused.core_derivs_vec(
+                    corelz, active=self.execute_model).cpu().numpy()
+                for m in range(NB):
+                    if self.execute_model[m]:
+                        dl[m], d2[m] = v[2 * m], v[2 * m + 1]
             else:
-                for m, v in outs:
-                    dl[m], d2[m] = v
+                outs = [(m, eng.core_derivs_async(float(corelz[m])))
+                        for m, eng in enumerate(self.engines)
+                        if self.execute_model[m]]
+                if outs and not isinstance(outs[0][1], tuple):
+                    import torch
+                    vals = torch.stack([o[1] for o in outs]).cpu()
+                    for k, (m, _) in enumerate(outs):
+                        dl[m], d2[m] = float(vals[k][0]), float(vals[k][1])
+                else:
+                    for m, v in outs:
+                        dl[m], d2[m] = v
             for i in range(NB):
                 if not outer[i] and not curvat[i]:
                     if d2[i] >= 0.0 and z[i] < ZMAX:

@@ -112,3 +112,24 @@ def test_per_partition_branch_lengths():
     vec = multi.evaluate_root(tree, p, q, zroot).cpu().numpy()
     for s, f in zip(single, vec):
         assert abs(f - s) <= 1e-11 * abs(s), (s, f)
+
+
+def test_treesearch_fused_matches_unfused():
+    """TreeSearch auto-enables the fused path for homogeneous dense GAMMA
+    engines; full -f E-style treeEvaluate agrees with the per-partition
+    loop to 1e-9 (device-exp P ulps compound through Brent probes)."""
+    from examl_amd.search import TreeSearch
+    eng_a, multi, _ = _mk(widths=(800, 1200, 96))
+    eng_b, _, _ = _mk(widths=(800, 1200, 96))
+    t_a = ea.PhyloTree.random(12, seed=31, rng_z=True)
+    t_b = ea.PhyloTree.random(12, seed=31, rng_z=True)
+    ts_a = TreeSearch(t_a, eng_a)
+    assert ts_a.fused is not None
+    ts_b = TreeSearch(t_b, eng_b)
+    ts_b.fused = None  # force the per-partition loop
+    la = ts_a.evaluate_generic(full=True)
+    lb = ts_b.evaluate_generic(full=True)
+    assert abs(la - lb) <= 1e-9 * abs(lb), (la, lb)
+    ta = ts_a.tree_evaluate(1.0)
+    tb = ts_b.tree_evaluate(1.0)
+    assert abs(ta - tb) <= 1e-8 * abs(tb), (ta, tb)
