@@ -96,7 +96,12 @@ typedef struct {
   double *h_lnl;             /* pinned readbacks */
   double *h_out2;
   int executed; /* scratch flag per call */
+  int *d_cptr;               /* CAT: per-site rate category (refreshed per
+                                traversal from pd->rateCategory) */
+  unsigned char *h_tips;     /* host tips copy (evaluatePartialGeneric) */
 } ShimPart;
+
+#define SHIM_MAXC 25 /* maxCategories default, axml.h */
 
 static ShimPart *S = NULL;
 static int S_n = 0;
@@ -143,7 +148,7 @@ static void init_part(tree *tr, int m)
   }
 
   p->states = states;
-  p->span = 4 * states;
+  p->span = (tr->rateHetModel == CAT) ? states : 4 * states;
   p->width = w;
   p->maxOps = mxtips + 8;
   p->executed = 0;
@@ -157,10 +162,18 @@ static void init_part(tree *tr, int m)
   p->d_wgt = dmalloc((size_t)w * sizeof(int));
   p->d_scalers = dmalloc((size_t)2 * mxtips * sizeof(unsigned int));
   p->d_inc = dmalloc((size_t)p->maxOps * sizeof(unsigned int));
-  p->d_pbuf =
-      dmalloc((size_t)p->maxOps * 8 * states * states * sizeof(double));
-  p->d_diag = dmalloc((size_t)4 * states * sizeof(double));
-  p->d_dtab = dmalloc((size_t)12 * states * sizeof(double));
+  /* GAMMA: 2 P blocks of 4 cats per op; CAT: numCats<=25 pairs per op,
+   * diag/dtab sized for maxCategories up front (axml.c:1936) */
+  p->d_pbuf = dmalloc((size_t)p->maxOps *
+                      ((tr->rateHetModel == CAT) ? (size_t)2 * SHIM_MAXC
+                                                 : 8) *
+                      states * states * sizeof(double));
+  p->d_diag = dmalloc((size_t)(SHIM_MAXC + 4) * states * sizeof(double));
+  p->d_dtab = dmalloc(((size_t)SHIM_MAXC * states + 2 * states +
+                       SHIM_MAXC + 16) * sizeof(double));
+  p->d_cptr =
+      (tr->rateHetModel == CAT) ? (int *)dmalloc((size_t)w * sizeof(int))
+                                : NULL;
   p->d_partials = dmalloc((size_t)2 * 8192 * sizeof(double));
   p->d_lnl = dmalloc(sizeof(double));
   p->d_out2 = dmalloc(2 * sizeof(double));
@@ -177,7 +190,7 @@ static void init_part(tree *tr, int m)
       memcpy(h + (size_t)t * w, pd->yVector[t], (size_t)w);
     HIP_OK(hipMemcpy(p->d_tips, h, (size_t)(mxtips + 1) * w,
                      hipMemcpyHostToDevice));
-    free(h);
+    p->h_tips = h; /* kept for evaluatePartialGeneric (CAT rate search) */
   }
   HIP_OK(hipMemcpy(p->d_wgt, pd->wgt, (size_t)w * sizeof(int),
                    hipMemcpyHostToDevice));
@@ -204,8 +217,9 @@ static void shim_init(tree *tr)
   S_mxtips = tr->mxtips;
   S = (ShimPart *)calloc(S_n, sizeof(ShimPart));
   for (m = 0; m < S_n; m++) init_part(tr, m);
-  if (tr->rateHetModel != GAMMA) {
-    fprintf(stderr, "examl-HIP shim: only -m GAMMA is wired so far\n");
+  if (tr->rateHetModel != GAMMA && tr->rateHetModel != CAT) {
+    fprintf(stderr, "examl-HIP shim: unsupported rate-het model %d\n",
+            tr->rateHetModel);
     MPI_Abort(MPI_COMM_WORLD, 1);
   }
   if (tr->saveMemory) {
@@ -213,9 +227,9 @@ static void shim_init(tree *tr)
     MPI_Abort(MPI_COMM_WORLD, 1);
   }
 
-  /* all-DNA runs go through the fused multi-partition executors */
+  /* all-DNA GAMMA runs go through the fused multi-partition executors */
   {
-    int allDna = 1;
+    int allDna = (tr->rateHetModel == GAMMA);
     for (m = 0; m < S_n; m++)
       if (S[m].states != 4) allDna = 0;
     if (allDna && S_n >= 1) {
@@ -465,7 +479,25 @@ void newviewIterative(tree *tr, int startIndex)
     n = build_ops(tr, m, startIndex, p->ops);
     if (n > p->maxOps) shim_die("traversal exceeds maxOps", n);
     upload_model(p, pd);
-    if (p->states == 4)
+    if (tr->rateHetModel == CAT) {
+      /* the per-site categorization can change between calls
+       * (optimizeRateCategories writes pd->rateCategory in place) */
+      HIP_OK(hipMemcpyAsync(p->d_cptr, pd->rateCategory,
+                            (size_t)p->width * sizeof(int),
+                            hipMemcpyHostToDevice, 0));
+      if (p->states == 4)
+        CK(examl_hip_newview_traversal_dna_cat(
+            p->ops, n, pd->EIGN, pd->EI, pd->perSiteRates,
+            pd->numberOfCategories, p->d_EV, p->d_tipVector, p->d_cptr,
+            p->d_clv, p->width * p->span, p->d_tips, p->width, p->d_wgt,
+            p->width, p->d_scalers, p->d_inc, p->d_pbuf, 0));
+      else
+        CK(examl_hip_newview_traversal_prot_cat(
+            p->ops, n, pd->EIGN, pd->EI, pd->perSiteRates,
+            pd->numberOfCategories, p->d_EV, p->d_tipVector, p->d_cptr,
+            p->d_clv, p->width * p->span, p->d_tips, p->width, p->d_wgt,
+            p->width, p->d_scalers, p->d_inc, p->d_pbuf, 0));
+    } else if (p->states == 4)
       CK(examl_hip_newview_traversal_dna_gamma(
           p->ops, n, pd->EIGN, pd->EI, pd->gammaRates, p->d_EV,
           p->d_tipVector, p->d_clv, p->width * p->span, p->d_tips, p->width,
@@ -564,7 +596,22 @@ void evaluateIterative(tree *tr)
     if (tc == TIP_TIP) shim_die("evaluate at a tip-tip branch", 0);
     HIP_OK(hipMemsetAsync(p->d_lnl, 0, sizeof(double), 0));
     upload_model(p, pd);
-    if (p->states == 4)
+    if (tr->rateHetModel == CAT) {
+      if (p->states == 4)
+        CK(examl_hip_evaluate_root_dna_cat_x(
+            tc, pNumber, qNumber, x1s, x2s, ts, z, pd->EIGN,
+            pd->perSiteRates, pd->numberOfCategories, p->d_tipVector,
+            p->d_cptr, p->d_clv, p->width * p->span, p->d_tips, p->width,
+            p->d_wgt, p->width, p->d_scalers, p->d_diag, p->d_partials,
+            p->d_lnl, 0));
+      else
+        CK(examl_hip_evaluate_root_prot_cat(
+            tc, pNumber, qNumber, x1s, x2s, ts, z, pd->EIGN,
+            pd->perSiteRates, pd->numberOfCategories, p->d_tipVector,
+            p->d_cptr, p->d_clv, p->width * p->span, p->d_tips, p->width,
+            p->d_wgt, p->width, p->d_scalers, p->d_diag, p->d_partials,
+            p->d_lnl, 0));
+    } else if (p->states == 4)
       CK(examl_hip_evaluate_root_dna_gamma(
           tc, pNumber, qNumber, x1s, x2s, ts, z, pd->EIGN, pd->gammaRates,
           p->d_tipVector, p->d_clv, p->width * p->span, p->d_tips, p->width,
@@ -705,7 +752,18 @@ void makenewzIterative(tree *tr)
       p->d_sum = dmalloc((size_t)p->width * p->span * sizeof(double));
     root_case(pNumber, qNumber, tr->mxtips, &tc, &x1s, &x2s, &ts, &ts2);
     upload_model(p, pd);
-    if (p->states == 4)
+    if (tr->rateHetModel == CAT) {
+      if (p->states == 4)
+        CK(examl_hip_sum_root_dna_cat(tc, x1s, x2s, ts, ts2,
+                                      p->d_tipVector, p->d_clv,
+                                      p->width * p->span, p->d_tips,
+                                      p->width, p->d_sum, p->width, 0));
+      else
+        CK(examl_hip_sum_root_prot_cat(tc, x1s, x2s, ts, ts2,
+                                       p->d_tipVector, p->d_clv,
+                                       p->width * p->span, p->d_tips,
+                                       p->width, p->d_sum, p->width, 0));
+    } else if (p->states == 4)
       CK(examl_hip_sum_root_dna_gamma(tc, x1s, x2s, ts, ts2,
                                       p->d_tipVector, p->d_clv,
                                       p->width * p->span, p->d_tips,
@@ -765,7 +823,20 @@ void execCore(tree *tr, volatile double *_dlnLdlz, volatile double *_d2lnLdlz2)
     }
     if (!(tr->td[0].executeModel[m] && p->width > 0)) continue;
     HIP_OK(hipMemsetAsync(p->d_out2, 0, 2 * sizeof(double), 0));
-    if (p->states == 4)
+    if (tr->rateHetModel == CAT) {
+      if (p->states == 4)
+        CK(examl_hip_core_root_dna_cat(p->width, p->d_sum, pd->EIGN,
+                                       pd->perSiteRates,
+                                       pd->numberOfCategories, lz,
+                                       p->d_wgt, p->d_cptr, p->d_dtab,
+                                       p->d_partials, p->d_out2, 0));
+      else
+        CK(examl_hip_core_root_prot_cat(p->width, p->d_sum, pd->EIGN,
+                                        pd->perSiteRates,
+                                        pd->numberOfCategories, lz,
+                                        p->d_wgt, p->d_cptr, p->d_dtab,
+                                        p->d_partials, p->d_out2, 0));
+    } else if (p->states == 4)
       CK(examl_hip_core_root_dna_gamma(p->width, p->d_sum, pd->EIGN,
                                        pd->gammaRates, lz, p->d_wgt,
                                        p->d_dtab, p->d_partials, p->d_out2,
@@ -939,18 +1010,57 @@ void makenewzGeneric(tree *tr, nodeptr p, nodeptr q, double *z0, int maxiter,
   for (i = 0; i < tr->numBranches; i++) tr->executeModel[i] = TRUE;
 }
 
-/* evaluatePartialGeneric (evaluatePartialGenericSpecial.c:259) is only
- * reached through the CAT/PSR per-site rate search
- * (optimizeModel.c:1868-1892); the C shim wires -m GAMMA only, and
- * shim_init aborts on -m PSR before this can be hit. */
+/* evaluatePartialGeneric (evaluatePartialGenericSpecial.c:259): the
+ * CAT/PSR per-site rate probe (optimizeModel.c:1868-1892) — a single
+ * site's lnL at an arbitrary rate ki over the CURRENT tree.  Host math,
+ * as in the reference: a full traversal descriptor of the tree is built
+ * locally (not via tr->td, which belongs to the optimizer's own calls)
+ * and handed to the single-site recursion. */
 double evaluatePartialGeneric(tree *tr, int i, double ki, int _model)
 {
-  (void)tr;
-  (void)i;
-  (void)ki;
-  (void)_model;
-  fprintf(stderr,
-          "examl-HIP shim: evaluatePartialGeneric (-m PSR) not wired\n");
-  MPI_Abort(MPI_COMM_WORLD, 1);
-  return 0.0;
+  static traversalInfo *ti = NULL;
+  static examl_hip_trav_entry *pops = NULL;
+  ShimPart *p;
+  pInfo *pd;
+  nodeptr start, back;
+  int count = 0, n, e;
+
+  shim_init(tr);
+  p = &S[_model];
+  pd = &tr->partitionData[_model];
+  if (ti == NULL) {
+    ti = (traversalInfo *)malloc(sizeof(traversalInfo) * tr->mxtips);
+    pops = (examl_hip_trav_entry *)malloc(sizeof(examl_hip_trav_entry) *
+                                          tr->mxtips);
+  }
+  start = tr->start; /* a tip (axml.c) */
+  back = start->back;
+  computeTraversalInfo(back, ti, &count, tr->mxtips, tr->numBranches,
+                       FALSE);
+  n = count;
+  for (e = 0; e < n; e++) {
+    examl_hip_trav_entry *o = &pops[e];
+    o->tipCase = ti[e].tipCase;
+    o->pNumber = ti[e].pNumber;
+    o->qNumber = ti[e].qNumber;
+    o->rNumber = ti[e].rNumber;
+    o->x3Slot = ti[e].pNumber - tr->mxtips - 1;
+    o->x1Slot = (ti[e].tipCase != INNER_INNER)
+                    ? ti[e].qNumber
+                    : ti[e].qNumber - tr->mxtips - 1;
+    o->x2Slot = (ti[e].tipCase == TIP_TIP)
+                    ? ti[e].rNumber
+                    : ti[e].rNumber - tr->mxtips - 1;
+    o->qz = ti[e].qz[0];
+    o->rz = ti[e].rz[0];
+  }
+  if (p->states == 4)
+    return examl_host_evaluate_partial_dna_cat(
+        pops, n, start->number, back->number, start->z[0], i, ki,
+        pd->wgt[i], pd->EIGN, pd->EI, pd->EV, pd->tipVector, p->h_tips,
+        p->width, tr->mxtips);
+  return examl_host_evaluate_partial_prot_cat(
+      pops, n, start->number, back->number, start->z[0], i, ki,
+      pd->wgt[i], pd->EIGN, pd->EI, pd->EV, pd->tipVector, p->h_tips,
+      p->width, tr->mxtips);
 }

@@ -108,3 +108,24 @@ def test_hybrid_140_partitioned_protein(tmp_path):
     golden = -121288.814123
     rel = abs(lnl - golden) / abs(golden)
     assert rel < 1e-6, f"hybrid 140 lnL {lnl} vs {golden} rel {rel}"
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(not os.path.exists(HYBRID),
+                    reason="hybrid binary not built")
+def test_hybrid_psr(tmp_path):
+    """-m PSR -f E: the reference's CAT per-site-rate machinery
+    (optimizeRateCategories, optimizeModel.c:2403) driving the CAT HIP
+    kernels through the shim, incl. evaluatePartialGeneric as host math.
+    Golden: reference examl-AVX -m PSR -f E on testData/49 =
+    -14702.970620."""
+    r = _run(tmp_path, ["-s", "49.binary", "-t", "49.tree", "-m", "PSR",
+                        "-f", "E", "-n", "HYBPSR"], timeout=1200)
+    out = r.stdout + r.stderr
+    assert r.returncode == 0, out[-3000:]
+    lines = [ln for ln in out.splitlines() if "Likelihood tree 0" in ln]
+    assert lines, out[-3000:]
+    lnl = float(lines[0].split(":")[1])
+    golden = -14702.970620
+    rel = abs(lnl - golden) / abs(golden)
+    assert rel < 1e-6, f"hybrid PSR lnL {lnl} vs {golden} rel {rel}"
