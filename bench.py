@@ -279,8 +279,11 @@ def main():
         "config": {
             "workload": ("synthetic_prot_50taxa_200ksites_lg_gamma "
                          "(BASELINE.json configs[3])" if args.protein else
-                         "synthetic_dna_50taxa_1Msites_gtrgamma "
-                         "(BASELINE.json configs[1])")
+                         (f"synthetic_dna_50taxa_{P}x{pw}sites_gtrgamma "
+                          "(BASELINE.json configs[2] shard shape, fused "
+                          "mseg path)" if P > 1 else
+                          "synthetic_dna_50taxa_1Msites_gtrgamma "
+                          "(BASELINE.json configs[1])"))
                         + "; full-tree evaluateGeneric per step",
             "taxa": ntips,
             "sites_per_gpu": width,
