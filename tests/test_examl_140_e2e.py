@@ -59,7 +59,14 @@ def test_140_initial_evaluation_vs_oracle(golden_dir):
 
 @pytest.mark.gpu
 @pytest.mark.skipif(not os.environ.get("EXAML_E2E_140"),
-                    reason="long (~20-40 min): set EXAML_E2E_140=1")
+                    reason="Python-host-bound (>30 min even on the fused "
+                           "executors: ~1e5 tiny optimizer probes through "
+                           "Python/ctypes); the UNGATED hardware validation "
+                           "of this config is tests/test_hybrid.py::"
+                           "test_hybrid_140_partitioned_protein — the "
+                           "reference's own C search + AUTO selection on "
+                           "the same kernels in ~3 min. Set EXAML_E2E_140=1 "
+                           "to run this Python replay too")
 @pytest.mark.timeout(3000)
 def test_full_f_E_pipeline_140_gpu(golden_dir):
     import torch
