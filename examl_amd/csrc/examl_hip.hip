@@ -3896,6 +3896,9 @@ struct CSeg { /* one partition's NR-derivative unit */
   int blkBase, nBlocks, part, pad;
 };
 
+__global__ void k_newview_dna_mseg_rt(const MSeg *, const int *,
+                                      const double *);
+
 /* forward declarations of the protein (20-state) mseg kernels defined
  * after the executors */
 template <int TC, bool FAST>
@@ -4486,14 +4489,22 @@ static examl_hip_multi::Shape *multi_shape_get(
   sh.numOps = numOps;
   std::vector<MSeg> segs;
   std::vector<int> blk2seg;
+  /* DNA: ONE launch per level, runtime tipCase per segment (the DNA LDS
+   * footprint is tiny, so max-size allocation costs nothing and the
+   * bigger grids fill the chip on small partitioned shapes); protein:
+   * one launch per (level, tipCase) to keep the per-TC LDS sizing. */
+  const int tcGroups = (h->states == 4) ? 1 : 3;
   for (int lv = 0; lv < numLevels; lv++) {
-    for (int tc = 0; tc < 3; tc++) {
+    for (int tcg = 0; tcg < tcGroups; tcg++) {
       examl_hip_multi::Grp g;
-      g.tc = tc;
+      g.tc = (tcGroups == 1) ? -1 : tcg;
       g.segOff = (int)segs.size();
       g.blkOff = (int)blk2seg.size();
       for (int e = 0; e < numOps; e++) {
-        if (level[e] != lv || ops[e].tipCase != tc) continue;
+        if (level[e] != lv ||
+            (tcGroups != 1 && ops[e].tipCase != tcg))
+          continue;
+        const int tc = ops[e].tipCase;
         for (int p = 0; p < h->numParts; p++) {
           if (h->widths[p] == 0) continue;
           MSeg s;
@@ -4516,6 +4527,7 @@ static examl_hip_multi::Shape *multi_shape_get(
           s.tipVec = h->tipVec[p];
           s.n = h->widths[p];
           s.part = p;
+          s.pad = tc; /* tipCase for the runtime-TC kernel */
           s.blkBase = (int)blk2seg.size() - g.blkOff;
           s.nBlocks = h->partBlocks[p];
           const int si = (int)segs.size();
@@ -4647,12 +4659,7 @@ extern "C" int examl_hip_newview_traversal_multi(
   hipLaunchKernelGGL((K), dim3(g.grid), dim3(NV_BLOCK), 0, s, dsegs, db2s, \
                      d_active)
       if (S == 4) {
-        switch (g.tc) {
-        case EXAML_TIP_TIP: NV_MSEG(k_newview_dna_mseg<EXAML_TIP_TIP>); break;
-        case EXAML_TIP_INNER:
-          NV_MSEG(k_newview_dna_mseg<EXAML_TIP_INNER>); break;
-        default: NV_MSEG(k_newview_dna_mseg<EXAML_INNER_INNER>);
-        }
+        NV_MSEG(k_newview_dna_mseg_rt);
       } else if (g_fast_math) {
         switch (g.tc) {
         case EXAML_TIP_TIP:
@@ -4675,7 +4682,9 @@ extern "C" int examl_hip_newview_traversal_multi(
       if (err != hipSuccess) { rc = set_err(err, "mseg launch"); break; }
       if (g_prof_on) {
         hipEventRecord(ev_b, s);
-        g_prof_pend.push_back({ev_a, ev_b, g.tc});
+        /* level-fused DNA groups (tc == -1) mix tipCases; bucket them
+         * under II — the P>1 roofline sums all buckets anyway */
+        g_prof_pend.push_back({ev_a, ev_b, g.tc < 0 ? 2 : g.tc});
         if (g_prof_pend.size() > 2048) prof_flush();
       }
     }
@@ -5957,4 +5966,112 @@ extern "C" int examl_hip_sum_cat_save(
 #undef SSAVE
   CHK(hipGetLastError());
   return 0;
+}
+
+/* runtime-tipCase DNA mseg newview: one launch per LEVEL covering every
+ * tipCase's segments (DNA LDS footprint is ~6 KB, so the max-size
+ * allocation costs no occupancy; per-block the branch is uniform).
+ * Math identical to the templated k_newview_dna_mseg bodies. */
+__global__ __launch_bounds__(NV_BLOCK) void k_newview_dna_mseg_rt(
+    const MSeg *__restrict__ segs, const int *__restrict__ blk2seg,
+    const double *__restrict__ active) {
+  const int si = blk2seg[blockIdx.x];
+  const MSeg sg = segs[si];
+  if (active[sg.part] == 0.0) return;
+  const int TC = sg.pad; /* tipCase, stored by multi_shape_get */
+  __shared__ double sL[64], sR[64], sEV[16], sTV[64];
+  __shared__ double sU1[256], sU2[256];
+
+  const int tid = threadIdx.x;
+  if (tid < 64) {
+    sL[tid] = sg.P[tid];
+    sR[tid] = sg.P[64 + tid];
+    sTV[tid] = sg.tipVec[tid];
+  }
+  if (tid < 16) sEV[tid] = sg.EV[tid];
+  __syncthreads();
+
+  if (TC != EXAML_INNER_INNER) {
+    const int code = tid >> 4, cat = (tid >> 2) & 3, row = tid & 3;
+    const double *tv = &sTV[code * 4];
+    const double *pl = &sL[cat * 16 + row * 4];
+    sU1[tid] =
+        (pl[0] * tv[0] + pl[1] * tv[1]) + (pl[2] * tv[2] + pl[3] * tv[3]);
+    if (TC == EXAML_TIP_TIP) {
+      const double *pr = &sR[cat * 16 + row * 4];
+      sU2[tid] =
+          (pr[0] * tv[0] + pr[1] * tv[1]) + (pr[2] * tv[2] + pr[3] * tv[3]);
+    }
+    __syncthreads();
+  }
+
+  const long units = sg.n * 4;
+  const bool nt = sg.n >= 65536;
+  const int lane = tid & 63;
+  for (long idx = (long)(blockIdx.x - sg.blkBase) * NV_BLOCK + tid;
+       idx < units; idx += (long)sg.nBlocks * NV_BLOCK) {
+    const long site = idx >> 2;
+    const int cat = (int)(idx & 3);
+    double u1[4], u2[4];
+
+    if (TC == EXAML_INNER_INNER) {
+      const double4 xl = *reinterpret_cast<const double4 *>(&sg.x1[idx * 4]);
+      const double4 xr = *reinterpret_cast<const double4 *>(&sg.x2[idx * 4]);
+#pragma unroll
+      for (int l = 0; l < 4; l++) {
+        const double *pl = &sL[cat * 16 + l * 4];
+        const double *pr = &sR[cat * 16 + l * 4];
+        u1[l] = (xl.x * pl[0] + xl.y * pl[1]) + (xl.z * pl[2] + xl.w * pl[3]);
+        u2[l] = (xr.x * pr[0] + xr.y * pr[1]) + (xr.z * pr[2] + xr.w * pr[3]);
+      }
+    } else if (TC == EXAML_TIP_INNER) {
+      const int code = sg.t1[site];
+      const double4 xr = *reinterpret_cast<const double4 *>(&sg.x2[idx * 4]);
+#pragma unroll
+      for (int l = 0; l < 4; l++) {
+        const double *pr = &sR[cat * 16 + l * 4];
+        u1[l] = sU1[code * 16 + cat * 4 + l];
+        u2[l] = (xr.x * pr[0] + xr.y * pr[1]) + (xr.z * pr[2] + xr.w * pr[3]);
+      }
+    } else {
+      const int c1 = sg.t1[site], c2 = sg.t2[site];
+#pragma unroll
+      for (int l = 0; l < 4; l++) {
+        u1[l] = sU1[c1 * 16 + cat * 4 + l];
+        u2[l] = sU2[c2 * 16 + cat * 4 + l];
+      }
+    }
+
+    double a0 = 0, a1 = 0, a2 = 0, a3 = 0;
+#pragma unroll
+    for (int l = 0; l < 4; l++) {
+      const double t = u1[l] * u2[l];
+      a0 += t * sEV[l * 4 + 0];
+      a1 += t * sEV[l * 4 + 1];
+      a2 += t * sEV[l * 4 + 2];
+      a3 += t * sEV[l * 4 + 3];
+    }
+
+    if (TC != EXAML_TIP_TIP) {
+      const bool small = (fabs(a0) < MINLIKELIHOOD) &
+                         (fabs(a1) < MINLIKELIHOOD) &
+                         (fabs(a2) < MINLIKELIHOOD) &
+                         (fabs(a3) < MINLIKELIHOOD);
+      const unsigned long long m = __ballot(small);
+      if (((m >> (lane & ~3)) & 0xFULL) == 0xFULL) {
+        a0 *= TWOTOTHE256;
+        a1 *= TWOTOTHE256;
+        a2 *= TWOTOTHE256;
+        a3 *= TWOTOTHE256;
+        if ((lane & 3) == 0)
+          atomicAdd(sg.inc, (unsigned int)sg.wgt[site]);
+      }
+    }
+    if (nt)
+      __builtin_nontemporal_store((v4d){a0, a1, a2, a3},
+                                  reinterpret_cast<v4d *>(&sg.x3[idx * 4]));
+    else
+      *reinterpret_cast<double4 *>(&sg.x3[idx * 4]) =
+          make_double4(a0, a1, a2, a3);
+  }
 }
