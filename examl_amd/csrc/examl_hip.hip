@@ -3974,7 +3974,11 @@ __global__ __launch_bounds__(NV_BLOCK) void k_newview_dna_mseg(
   }
 
   const long units = sg.n * 4;
-  const bool nt = sg.n >= 65536;
+  /* NT when the LAUNCH's aggregate x3 output exceeds L2 (32 MiB): the
+   * fused grids write many partitions' slabs concurrently, so the
+   * per-segment size alone under-triggers streaming stores */
+  const bool nt = sg.n >= 65536 ||
+                  (long)gridDim.x * NV_BLOCK * 32 > (32L << 20);
   const int lane = tid & 63;
   for (long idx = (long)(blockIdx.x - sg.blkBase) * NV_BLOCK + tid;
        idx < units; idx += (long)sg.nBlocks * NV_BLOCK) {
@@ -6006,7 +6010,11 @@ __global__ __launch_bounds__(NV_BLOCK) void k_newview_dna_mseg_rt(
   }
 
   const long units = sg.n * 4;
-  const bool nt = sg.n >= 65536;
+  /* NT when the LAUNCH's aggregate x3 output exceeds L2 (32 MiB): the
+   * fused grids write many partitions' slabs concurrently, so the
+   * per-segment size alone under-triggers streaming stores */
+  const bool nt = sg.n >= 65536 ||
+                  (long)gridDim.x * NV_BLOCK * 32 > (32L << 20);
   const int lane = tid & 63;
   for (long idx = (long)(blockIdx.x - sg.blkBase) * NV_BLOCK + tid;
        idx < units; idx += (long)sg.nBlocks * NV_BLOCK) {
