@@ -47,12 +47,19 @@ def test_140_initial_evaluation_vs_oracle(golden_dir):
         models.append(m)
         engines.append(ea.DnaGammaEngine(p.tips, p.wgt, m, device="cuda:0"))
     ts = TreeSearch(tree, engines)
+    fused = ts.fused
+    ts.fused = None  # per-partition path first: 1e-11 vs the oracle
     lnl = ts.evaluate_generic(full=True)
     ref = 0.0
     entries, root = tree.full_traversal((1, next(iter(tree.adj[1]))))
     for p, m in zip(parts, models):
         ref += oracle_full_lnl(entries, root, tree, m, p.tips, p.wgt)
     assert abs(lnl - ref) / abs(ref) < 1e-11
+    if fused is not None:
+        # fused mseg path: device-exp P matrices -> <=1e-9 rel here
+        ts.fused = fused
+        lnl_f = ts.evaluate_generic(full=True)
+        assert abs(lnl_f - ref) / abs(ref) < 1e-9, (lnl_f, ref)
     after = ts.tree_evaluate(1.0)
     assert after > lnl
 
