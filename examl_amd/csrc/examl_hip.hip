@@ -3974,11 +3974,11 @@ __global__ __launch_bounds__(NV_BLOCK) void k_newview_dna_mseg(
   }
 
   const long units = sg.n * 4;
-  /* NT when the LAUNCH's aggregate x3 output exceeds L2 (32 MiB): the
-   * fused grids write many partitions' slabs concurrently, so the
-   * per-segment size alone under-triggers streaming stores */
-  const bool nt = sg.n >= 65536 ||
-                  (long)gridDim.x * NV_BLOCK * 32 > (32L << 20);
+  /* per-segment NT only: an aggregate-traffic criterion was tried and
+   * measured SLOWER on config 3 (0.513 -> 0.567 ms/step) — a level's
+   * x3 output is the next level's input, and streaming stores forfeit
+   * the cache hits those reads otherwise get */
+  const bool nt = sg.n >= 65536;
   const int lane = tid & 63;
   for (long idx = (long)(blockIdx.x - sg.blkBase) * NV_BLOCK + tid;
        idx < units; idx += (long)sg.nBlocks * NV_BLOCK) {
@@ -6010,11 +6010,11 @@ __global__ __launch_bounds__(NV_BLOCK) void k_newview_dna_mseg_rt(
   }
 
   const long units = sg.n * 4;
-  /* NT when the LAUNCH's aggregate x3 output exceeds L2 (32 MiB): the
-   * fused grids write many partitions' slabs concurrently, so the
-   * per-segment size alone under-triggers streaming stores */
-  const bool nt = sg.n >= 65536 ||
-                  (long)gridDim.x * NV_BLOCK * 32 > (32L << 20);
+  /* per-segment NT only: an aggregate-traffic criterion was tried and
+   * measured SLOWER on config 3 (0.513 -> 0.567 ms/step) — a level's
+   * x3 output is the next level's input, and streaming stores forfeit
+   * the cache hits those reads otherwise get */
+  const bool nt = sg.n >= 65536;
   const int lane = tid & 63;
   for (long idx = (long)(blockIdx.x - sg.blkBase) * NV_BLOCK + tid;
        idx < units; idx += (long)sg.nBlocks * NV_BLOCK) {
