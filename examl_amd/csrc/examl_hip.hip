@@ -4493,11 +4493,12 @@ static examl_hip_multi::Shape *multi_shape_get(
   sh.numOps = numOps;
   std::vector<MSeg> segs;
   std::vector<int> blk2seg;
-  /* DNA: ONE launch per level, runtime tipCase per segment (the DNA LDS
-   * footprint is tiny, so max-size allocation costs nothing and the
-   * bigger grids fill the chip on small partitioned shapes); protein:
-   * one launch per (level, tipCase) to keep the per-TC LDS sizing. */
-  const int tcGroups = (h->states == 4) ? 1 : 3;
+  /* one launch per (level, tipCase).  A per-level runtime-tipCase DNA
+   * variant (k_newview_dna_mseg_rt) was tried and measured SLOWER on
+   * config 3 (0.513 -> 0.565 ms/step: the fatter kernel lowers
+   * occupancy and mixes block durations); the templated per-TC groups
+   * stay. */
+  const int tcGroups = 3;
   for (int lv = 0; lv < numLevels; lv++) {
     for (int tcg = 0; tcg < tcGroups; tcg++) {
       examl_hip_multi::Grp g;
@@ -4663,7 +4664,12 @@ extern "C" int examl_hip_newview_traversal_multi(
   hipLaunchKernelGGL((K), dim3(g.grid), dim3(NV_BLOCK), 0, s, dsegs, db2s, \
                      d_active)
       if (S == 4) {
-        NV_MSEG(k_newview_dna_mseg_rt);
+        switch (g.tc) {
+        case EXAML_TIP_TIP: NV_MSEG(k_newview_dna_mseg<EXAML_TIP_TIP>); break;
+        case EXAML_TIP_INNER:
+          NV_MSEG(k_newview_dna_mseg<EXAML_TIP_INNER>); break;
+        default: NV_MSEG(k_newview_dna_mseg<EXAML_INNER_INNER>);
+        }
       } else if (g_fast_math) {
         switch (g.tc) {
         case EXAML_TIP_TIP:
