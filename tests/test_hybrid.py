@@ -129,3 +129,30 @@ def test_hybrid_psr(tmp_path):
     golden = -14702.970620
     rel = abs(lnl - golden) / abs(golden)
     assert rel < 1e-6, f"hybrid PSR lnL {lnl} vs {golden} rel {rel}"
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(not os.path.exists(HYBRID),
+                    reason="hybrid binary not built")
+def test_hybrid_two_ranks(tmp_path):
+    """2 MPI ranks (the reference's own data distribution,
+    partitionAssignment.c:398, + the C1/C2 MPI_Allreduce) with both
+    ranks' HIP contexts on one GPU: final -f E lnL must match the 1-rank
+    golden — the reference's own determinism anchor (SURVEY §8c:
+    bit-identical at 1 and 2 ranks)."""
+    shutil.copy(os.path.join(GOLDEN, "49.binary"), str(tmp_path))
+    shutil.copy(os.path.join(GOLDEN, "49.tree"), str(tmp_path))
+    mpiexec = "/opt/conda/bin/mpiexec"
+    if not os.path.exists(mpiexec):
+        pytest.skip("mpiexec not present")
+    r = subprocess.run(
+        [mpiexec, "-n", "2", HYBRID, "-s", "49.binary", "-t", "49.tree",
+         "-m", "GAMMA", "-f", "E", "-n", "HYB2"], cwd=str(tmp_path),
+        capture_output=True, text=True, timeout=900)
+    out = r.stdout + r.stderr
+    assert r.returncode == 0, out[-3000:]
+    lines = [ln for ln in out.splitlines() if "Likelihood tree 0" in ln]
+    assert lines, out[-3000:]
+    lnl = float(lines[0].split(":")[1])
+    rel = abs(lnl - GOLDEN_F_E) / abs(GOLDEN_F_E)
+    assert rel < 1e-6, f"2-rank hybrid lnL {lnl} vs {GOLDEN_F_E} rel {rel}"
