@@ -53,6 +53,7 @@ extern "C" const char *examl_hip_last_error_string(void) { return g_err; }
 /* clang ext-vector for nontemporal 32-byte stores (HIP's double4 is a
  * class type the builtin rejects) */
 typedef double v4d __attribute__((ext_vector_type(4)));
+typedef double v2d __attribute__((ext_vector_type(2)));
 #define MAX_GRID 8192
 
 static inline int grid_for(long units) {
@@ -601,13 +602,17 @@ __global__ __launch_bounds__(NV_BLOCK) void k_newview_prot_gamma(
     const unsigned char *__restrict__ tipX1,
     const unsigned char *__restrict__ tipX2, const int *__restrict__ wgt,
     long n, unsigned int *__restrict__ scalerInc) {
-  /* The four lanes of one site differ only in `cat`; with the natural
-   * cat stride of 400 doubles (800 LDS words, = 0 mod 32 banks) every
-   * P-row read is a 4-way bank conflict.  Padding the cat stride to 404
-   * doubles (808 words, = 8 mod 32) puts the four cats on banks
-   * 0/8/16/24 — conflict-free, which is what the round-1 "LDS-issue
-   * bound" profile was hitting. */
-  constexpr int CSTR = 404; /* padded per-cat LDS stride of sL/sR */
+  /* Two lanes per (site,cat), split by ACCUMULATOR PARITY: the reference
+   * dot20 is (t0+t1)+(t2+t3) with t_k accumulating c = 4j+k
+   * (avxLikelihood.c:1312 hadd structure); lane half 0 owns the t0,t1
+   * chains (c%4 in {0,1}), half 1 owns t2,t3, and the halves combine as
+   * s01 + s23 via one wave shuffle — BIT-IDENTICAL to the single-lane
+   * order.  Each lane then accumulates 10 of the 20 output states.
+   * This halves the per-lane register arrays (30 doubles vs 60), which
+   * lifts occupancy from 3 to ~5 waves/SIMD — the round-2 diagnosis of
+   * this kernel was latency-bound at 146 VGPRs.
+   * P rows stay in padded LDS (cat stride 404: conflict-free). */
+  constexpr int CSTR = 404;
   __shared__ double sL[4 * CSTR], sR[4 * CSTR], sEV[400];
   __shared__ double sTV[TC != EXAML_INNER_INNER ? 460 : 1];
   __shared__ double sU1[TC != EXAML_INNER_INNER ? 1840 : 1];
@@ -636,80 +641,171 @@ __global__ __launch_bounds__(NV_BLOCK) void k_newview_prot_gamma(
     __syncthreads();
   }
 
-  const long units = n * 4;
+  if (TC == EXAML_TIP_TIP) {
+    /* TT has no dots and no rescale — the classic per-(site,cat) lane
+     * with full-span accumulators stays the right shape (the split
+     * mapping ballooned this instantiation's registers) */
+    const long unitsTT = n * 4;
+    for (long idx = (long)blockIdx.x * NV_BLOCK + tid; idx < unitsTT;
+         idx += (long)gridDim.x * NV_BLOCK) {
+      const long site = idx >> 2;
+      const int cat = (int)(idx & 3);
+      const int c1 = tipX1[site], c2 = tipX2[site];
+      double acc[20];
+#pragma unroll
+      for (int s = 0; s < 20; s++) acc[s] = 0.0;
+      for (int l = 0; l < 20; l++) {
+        const double t =
+            sU1[80 * c1 + cat * 20 + l] * sU2[80 * c2 + cat * 20 + l];
+#pragma unroll
+        for (int s = 0; s < 20; s++) {
+          if (FAST)
+            acc[s] = fma(t, sEV[l * 20 + s], acc[s]);
+          else
+            acc[s] += t * sEV[l * 20 + s];
+        }
+      }
+#pragma unroll
+      for (int s = 0; s < 20; s += 4) {
+        const double4 v =
+            make_double4(acc[s], acc[s + 1], acc[s + 2], acc[s + 3]);
+        if (NT)
+          __builtin_nontemporal_store(
+              (v4d){v.x, v.y, v.z, v.w},
+              reinterpret_cast<v4d *>(&x3[idx * 20 + s]));
+        else
+          *reinterpret_cast<double4 *>(&x3[idx * 20 + s]) = v;
+      }
+    }
+    return;
+  }
+
+  const long units = n * 8; /* (site, cat, half) */
   const int lane = tid & 63;
   for (long idx = (long)blockIdx.x * NV_BLOCK + tid; idx < units;
        idx += (long)gridDim.x * NV_BLOCK) {
-    const long site = idx >> 2;
-    const int cat = (int)(idx & 3);
-    double xl[20], xr[20], acc[20];
+    const long site = idx >> 3;
+    const int cat = (int)((idx >> 1) & 3);
+    const int half = (int)(idx & 1);
+    const long b20 = (site * 4 + cat) * 20;
+    /* my half's operand entries: c = 4j + 2*half + {0,1} */
+    double xl[10], xr[10], acc[10];
     int code1 = 0, code2 = 0;
     if (TC == EXAML_INNER_INNER) {
 #pragma unroll
-      for (int s = 0; s < 20; s += 4) {
-        const double4 a = *reinterpret_cast<const double4 *>(&x1[idx * 20 + s]);
-        const double4 b = *reinterpret_cast<const double4 *>(&x2[idx * 20 + s]);
-        xl[s] = a.x; xl[s + 1] = a.y; xl[s + 2] = a.z; xl[s + 3] = a.w;
-        xr[s] = b.x; xr[s + 1] = b.y; xr[s + 2] = b.z; xr[s + 3] = b.w;
+      for (int j = 0; j < 5; j++) {
+        const double2 a = *reinterpret_cast<const double2 *>(
+            &x1[b20 + 4 * j + 2 * half]);
+        const double2 b = *reinterpret_cast<const double2 *>(
+            &x2[b20 + 4 * j + 2 * half]);
+        xl[2 * j] = a.x;
+        xl[2 * j + 1] = a.y;
+        xr[2 * j] = b.x;
+        xr[2 * j + 1] = b.y;
       }
     } else if (TC == EXAML_TIP_INNER) {
       code1 = tipX1[site];
 #pragma unroll
-      for (int s = 0; s < 20; s += 4) {
-        const double4 b = *reinterpret_cast<const double4 *>(&x2[idx * 20 + s]);
-        xr[s] = b.x; xr[s + 1] = b.y; xr[s + 2] = b.z; xr[s + 3] = b.w;
+      for (int j = 0; j < 5; j++) {
+        const double2 b = *reinterpret_cast<const double2 *>(
+            &x2[b20 + 4 * j + 2 * half]);
+        xr[2 * j] = b.x;
+        xr[2 * j + 1] = b.y;
       }
     } else {
       code1 = tipX1[site];
       code2 = tipX2[site];
     }
 #pragma unroll
-    for (int s = 0; s < 20; s++) acc[s] = 0.0;
+    for (int s = 0; s < 10; s++) acc[s] = 0.0;
+    const double *EVh = &sEV[0]; /* acc state s' = half*10 + s */
     for (int l = 0; l < 20; l++) {
       double u1, u2;
-      if (TC == EXAML_INNER_INNER) {
-        u1 = dot20o<FAST>(xl, &sL[cat * CSTR + l * 20]);
-        u2 = dot20o<FAST>(xr, &sR[cat * CSTR + l * 20]);
-      } else if (TC == EXAML_TIP_INNER) {
-        u1 = sU1[80 * code1 + cat * 20 + l];
-        u2 = dot20o<FAST>(xr, &sR[cat * CSTR + l * 20]);
-      } else {
+      if (TC == EXAML_TIP_TIP) {
         u1 = sU1[80 * code1 + cat * 20 + l];
         u2 = sU2[80 * code2 + cat * 20 + l];
+      } else {
+        /* split dot: my two accumulator chains, then cross-half join in
+         * the reference's (t01)+(t23) order */
+        const double *pr = &sR[cat * CSTR + l * 20 + 2 * half];
+        double te = 0, to = 0;
+#pragma unroll
+        for (int j = 0; j < 5; j++) {
+          if (FAST) {
+            te = fma(xr[2 * j], pr[4 * j], te);
+            to = fma(xr[2 * j + 1], pr[4 * j + 1], to);
+          } else {
+            te += xr[2 * j] * pr[4 * j];
+            to += xr[2 * j + 1] * pr[4 * j + 1];
+          }
+        }
+        const double mine2 = te + to;
+        const double other2 = __shfl_xor(mine2, 1);
+        u2 = half ? (other2 + mine2) : (mine2 + other2);
+        if (TC == EXAML_TIP_INNER) {
+          u1 = sU1[80 * code1 + cat * 20 + l];
+        } else {
+          const double *pl = &sL[cat * CSTR + l * 20 + 2 * half];
+          double se = 0, so = 0;
+#pragma unroll
+          for (int j = 0; j < 5; j++) {
+            if (FAST) {
+              se = fma(xl[2 * j], pl[4 * j], se);
+              so = fma(xl[2 * j + 1], pl[4 * j + 1], so);
+            } else {
+              se += xl[2 * j] * pl[4 * j];
+              so += xl[2 * j + 1] * pl[4 * j + 1];
+            }
+          }
+          const double mine1 = se + so;
+          const double other1 = __shfl_xor(mine1, 1);
+          u1 = half ? (other1 + mine1) : (mine1 + other1);
+        }
       }
       const double t = u1 * u2;
 #pragma unroll
-      for (int s = 0; s < 20; s++) {
+      for (int s = 0; s < 10; s++) {
         if (FAST)
-          acc[s] = fma(t, sEV[l * 20 + s], acc[s]);
+          acc[s] = fma(t, EVh[l * 20 + half * 10 + s], acc[s]);
         else
-          acc[s] += t * sEV[l * 20 + s];
+          acc[s] += t * EVh[l * 20 + half * 10 + s];
       }
     }
 
     if (TC != EXAML_TIP_TIP) {
+      /* all 80 span entries below threshold = all 8 lanes of the site
+       * vote small (avxLikelihood.c:1806 rule) */
       bool small = true;
 #pragma unroll
-      for (int s = 0; s < 20; s++)
+      for (int s = 0; s < 10; s++)
         small &= (fabs(acc[s]) < MINLIKELIHOOD);
       const unsigned long long m = __ballot(small);
-      if (((m >> (lane & ~3)) & 0xFULL) == 0xFULL) {
+      if (((m >> (lane & ~7)) & 0xFFULL) == 0xFFULL) {
 #pragma unroll
-        for (int s = 0; s < 20; s++) acc[s] *= TWOTOTHE256;
-        if ((lane & 3) == 0)
+        for (int s = 0; s < 10; s++) acc[s] *= TWOTOTHE256;
+        if ((lane & 7) == 0)
           atomicAdd(scalerInc, (unsigned int)wgt[site]);
       }
     }
-#pragma unroll
-    for (int s = 0; s < 20; s += 4) {
-      const double4 v = make_double4(acc[s], acc[s + 1], acc[s + 2],
-                                     acc[s + 3]);
-      if (NT)
+    {
+      double *out = &x3[b20 + half * 10];
+      if (NT) {
         __builtin_nontemporal_store(
-            (v4d){v.x, v.y, v.z, v.w},
-            reinterpret_cast<v4d *>(&x3[idx * 20 + s]));
-      else
-        *reinterpret_cast<double4 *>(&x3[idx * 20 + s]) = v;
+            (v4d){acc[0], acc[1], acc[2], acc[3]},
+            reinterpret_cast<v4d *>(&out[0]));
+        __builtin_nontemporal_store(
+            (v4d){acc[4], acc[5], acc[6], acc[7]},
+            reinterpret_cast<v4d *>(&out[4]));
+        __builtin_nontemporal_store((v2d){acc[8], acc[9]},
+                                    reinterpret_cast<v2d *>(&out[8]));
+      } else {
+        *reinterpret_cast<double4 *>(&out[0]) =
+            make_double4(acc[0], acc[1], acc[2], acc[3]);
+        *reinterpret_cast<double4 *>(&out[4]) =
+            make_double4(acc[4], acc[5], acc[6], acc[7]);
+        *reinterpret_cast<double2 *>(&out[8]) =
+            make_double2(acc[8], acc[9]);
+      }
     }
   }
 }
@@ -1422,8 +1518,9 @@ static int traversal_impl(const examl_hip_trav_entry *ops, int numOps,
                          s);
     if (err != hipSuccess) { rc = set_err(err, "inc memset"); break; }
 
-    /* 2. one newview kernel per entry, post order on one stream */
-    const int grid = grid_for(n * 4);
+    /* 2. one newview kernel per entry, post order on one stream
+     * (protein uses two lanes per (site,cat) -> 8 units per site) */
+    const int grid = grid_for(n * (STATES == 4 ? 4 : 8));
     const bool nt = (STATES == 4) && (n >= 65536);
     for (int e = 0; e < numOps && rc == 0; e++) {
       const examl_hip_trav_entry *op = &ops[e];
@@ -2074,7 +2171,7 @@ extern "C" int examl_hip_newview_prot_gamma(
   if (n <= 0) return 0;
   hipStream_t s = (hipStream_t)stream;
   (void)hipGetLastError();
-  const int grid = grid_for(n * 4);
+  const int grid = grid_for(n * 8); /* two lanes per (site,cat) */
   const bool nt = false; /* NT hurts the protein kernel: its five strided
     32-B stores per thread defeat write-combining (measured 2x slower) */
   if (right != left + 1600) {
