@@ -16,23 +16,38 @@ from tests.helpers import make_synthetic
 pytestmark = pytest.mark.gpu
 
 
-def _mk(widths=(1500, 700, 64, 2300), seed=5):
+def _mk(widths=(1500, 700, 64, 2300), seed=5, states=4):
+    import os
     rng = np.random.default_rng(seed)
     engines = []
+    if states == 20:
+        aa = np.load(os.path.join(os.path.dirname(os.path.dirname(
+            os.path.abspath(__file__))), "examl_amd", "data",
+            "aa_models.npz"))
     for i, w in enumerate(widths):
-        tips, wgt = make_synthetic(12, w, seed=seed + i)
-        freqs = rng.uniform(0.1, 0.4, 4)
-        freqs /= freqs.sum()
-        rates = list(rng.uniform(0.3, 3.0, 5)) + [1.0]
-        m = ea.DnaGtrModel(list(freqs), rates,
-                           alpha=float(rng.uniform(0.2, 1.5)))
+        if states == 4:
+            tips, wgt = make_synthetic(12, w, seed=seed + i)
+            freqs = rng.uniform(0.1, 0.4, 4)
+            freqs /= freqs.sum()
+            rates = list(rng.uniform(0.3, 3.0, 5)) + [1.0]
+            m = ea.DnaGtrModel(list(freqs), rates,
+                               alpha=float(rng.uniform(0.2, 1.5)))
+        else:
+            tips = np.zeros((13, w), dtype=np.uint8)
+            for t in range(1, 13):
+                tips[t] = rng.integers(1, 23, w).astype(np.uint8)
+            wgt = rng.integers(1, 4, w).astype(np.int32)
+            m = ea.ProtGtrModel(aa["frequencies"][4 + i % 3],
+                                aa["rates190"][4 + i % 3],
+                                float(rng.uniform(0.4, 1.2)))
         engines.append(ea.DnaGammaEngine(tips, wgt, m, device="cuda:0"))
     tree = ea.PhyloTree.random(12, seed=31, rng_z=True)
     return engines, ea.MultiDnaEngine(engines), tree
 
 
-def test_full_lnl_matches_per_partition():
-    engines, multi, tree = _mk()
+@pytest.mark.parametrize("states", [4, 20])
+def test_full_lnl_matches_per_partition(states):
+    engines, multi, tree = _mk(states=states)
     entries, (p, q, z) = tree.full_traversal()
     single = []
     for e in engines:
@@ -43,8 +58,9 @@ def test_full_lnl_matches_per_partition():
         assert abs(f - s) <= 1e-11 * abs(s), (s, f)
 
 
-def test_masked_partitions_stay_stale():
-    engines, multi, tree = _mk()
+@pytest.mark.parametrize("states", [4, 20])
+def test_masked_partitions_stay_stale(states):
+    engines, multi, tree = _mk(states=states)
     entries, (p, q, z) = tree.full_traversal()
     # first an all-partition traversal to give every CLV a value
     multi.newview_traversal(entries)
@@ -65,8 +81,9 @@ def test_masked_partitions_stay_stale():
             assert after[i] != base[i]
 
 
-def test_makenewz_matches_per_partition():
-    engines, multi, tree = _mk()
+@pytest.mark.parametrize("states", [4, 20])
+def test_makenewz_matches_per_partition(states):
+    engines, multi, tree = _mk(states=states)
     entries, (p, q, z) = tree.full_traversal()
     for e in engines:
         e.newview_traversal(entries)
