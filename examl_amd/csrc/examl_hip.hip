@@ -5008,7 +5008,11 @@ extern "C" int examl_hip_core_root_multi(
     act[p] = (h->widths[p] > 0 && (!activeMask || activeMask[p])) ? 1.0 : 0.0;
     if (h->widths[p] == 0) continue;
     const double lz = lzPerPart ? lzs[p] : lzs[0];
-    examl_host_core_dtables_dna(EIGNs[p], rates[p], lz, dtab + p * tsz);
+    if (S == 4)
+      examl_host_core_dtables_dna(EIGNs[p], rates[p], lz, dtab + p * tsz);
+    else
+      examl_host_core_dtables_prot(EIGNs[p], rates[p], lz,
+                                   dtab + p * tsz);
     CSeg *e = &cs[si];
     memset(e, 0, sizeof(*e));
     e->sum = h->sum_base[p];
