@@ -618,6 +618,13 @@ __global__ __launch_bounds__(NV_BLOCK) void k_newview_prot_gamma(
   __shared__ double sU1[TC != EXAML_INNER_INNER ? 1840 : 1];
   __shared__ double sU2[TC == EXAML_TIP_TIP ? 1840 : 1];
 
+  /* the launcher sizes the grid for the split-lane II mapping (n*8
+   * units); the classic TT/TI branches cover n*4 — their surplus blocks
+   * must exit BEFORE the LDS staging + ump build (1840 dot20s), which
+   * otherwise costs as much as the site work */
+  const long tcUnits = (TC == EXAML_INNER_INNER) ? n * 8 : n * 4;
+  if ((long)blockIdx.x * NV_BLOCK >= tcUnits) return;
+
   const int tid = threadIdx.x;
   for (int j = tid; j < 1600; j += NV_BLOCK) {
     const int pc = j / 400, pr = j % 400;
