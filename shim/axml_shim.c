@@ -99,6 +99,12 @@ typedef struct {
   int *d_cptr;               /* CAT: per-site rate category (refreshed per
                                 traversal from pd->rateCategory) */
   unsigned char *h_tips;     /* host tips copy (evaluatePartialGeneric) */
+  /* LG4M/LG4X: per-gamma-category eigensystems, packed contiguous from
+   * the reference's *_LG4[4] row pointers per call */
+  int lg4;
+  double *d_EV4, *d_tipVec4; /* 1600 / 1840 */
+  double *h_eign4, *h_ei4;   /* 80 / 1600 host staging */
+  double *h_pack;            /* 3440 host staging for EV4|tipVec4 */
 } ShimPart;
 
 #define SHIM_MAXC 25 /* maxCategories default, axml.h */
@@ -140,10 +146,6 @@ static void init_part(tree *tr, int m)
     fprintf(stderr,
             "examl-HIP shim: %d-state partitions are not supported\n",
             states);
-    MPI_Abort(MPI_COMM_WORLD, 1);
-  }
-  if (pd->protModels == LG4M || pd->protModels == LG4X) {
-    fprintf(stderr, "examl-HIP shim: LG4M/LG4X not wired in the C shim\n");
     MPI_Abort(MPI_COMM_WORLD, 1);
   }
 
@@ -195,6 +197,14 @@ static void init_part(tree *tr, int m)
   HIP_OK(hipMemcpy(p->d_wgt, pd->wgt, (size_t)w * sizeof(int),
                    hipMemcpyHostToDevice));
 
+  p->lg4 = (pd->protModels == LG4M || pd->protModels == LG4X);
+  if (p->lg4) {
+    p->d_EV4 = (double *)dmalloc(1600 * sizeof(double));
+    p->d_tipVec4 = (double *)dmalloc(1840 * sizeof(double));
+    p->h_eign4 = (double *)malloc(80 * sizeof(double));
+    p->h_ei4 = (double *)malloc(1600 * sizeof(double));
+    HIP_OK(hipHostMalloc((void **)&p->h_pack, 3440 * sizeof(double), 0));
+  }
   p->ops = (examl_hip_trav_entry *)malloc(p->maxOps *
                                           sizeof(examl_hip_trav_entry));
   HIP_OK(hipHostMalloc((void **)&p->h_lnl, sizeof(double), 0));
@@ -366,6 +376,26 @@ void computeTraversalInfo(nodeptr p, traversalInfo *ti, int *counter,
   }
 }
 
+/* LG4: pack the four per-category eigensystem rows (pInfo *_LG4[4],
+ * axml.h:566-575) into the contiguous stride layout the LG4 executors
+ * take, and refresh the device EV4/tipVector4 copies (optLG4X /
+ * initReversibleGTR rewrite them between calls). */
+static void upload_lg4(ShimPart *p, pInfo *pd)
+{
+  int k;
+  for (k = 0; k < 4; k++) {
+    memcpy(p->h_eign4 + 20 * k, pd->EIGN_LG4[k], 20 * sizeof(double));
+    memcpy(p->h_ei4 + 400 * k, pd->EI_LG4[k], 400 * sizeof(double));
+    memcpy(p->h_pack + 400 * k, pd->EV_LG4[k], 400 * sizeof(double));
+    memcpy(p->h_pack + 1600 + 460 * k, pd->tipVector_LG4[k],
+           460 * sizeof(double));
+  }
+  HIP_OK(hipMemcpyAsync(p->d_EV4, p->h_pack, 1600 * sizeof(double),
+                        hipMemcpyHostToDevice, 0));
+  HIP_OK(hipMemcpyAsync(p->d_tipVec4, p->h_pack + 1600,
+                        1840 * sizeof(double), hipMemcpyHostToDevice, 0));
+}
+
 /* ti entries -> executor ops: resolve the CLV-slot / tip-row bindings the
  * way newviewIterative does (newviewGenericSpecial.c:1221-1261): inner
  * node n -> slot n - mxtips - 1; tip operands use the tip row number. */
@@ -502,7 +532,13 @@ void newviewIterative(tree *tr, int startIndex)
           p->ops, n, pd->EIGN, pd->EI, pd->gammaRates, p->d_EV,
           p->d_tipVector, p->d_clv, p->width * p->span, p->d_tips, p->width,
           p->d_wgt, p->width, p->d_scalers, p->d_inc, p->d_pbuf, 0));
-    else
+    else if (p->lg4) {
+      upload_lg4(p, pd);
+      CK(examl_hip_newview_traversal_prot_lg4(
+          p->ops, n, p->h_eign4, p->h_ei4, pd->gammaRates, p->d_EV4,
+          p->d_tipVec4, p->d_clv, p->width * p->span, p->d_tips, p->width,
+          p->d_wgt, p->width, p->d_scalers, p->d_inc, p->d_pbuf, 0));
+    } else
       CK(examl_hip_newview_traversal_prot_gamma(
           p->ops, n, pd->EIGN, pd->EI, pd->gammaRates, p->d_EV,
           p->d_tipVector, p->d_clv, p->width * p->span, p->d_tips, p->width,
@@ -617,7 +653,14 @@ void evaluateIterative(tree *tr)
           p->d_tipVector, p->d_clv, p->width * p->span, p->d_tips, p->width,
           p->d_wgt, p->width, p->d_scalers, p->d_diag, p->d_partials,
           p->d_lnl, 0));
-    else
+    else if (p->lg4) {
+      upload_lg4(p, pd);
+      CK(examl_hip_evaluate_root_prot_lg4(
+          tc, pNumber, qNumber, x1s, x2s, ts, z, p->h_eign4,
+          pd->gammaRates, pd->weights, p->d_tipVec4, p->d_clv,
+          p->width * p->span, p->d_tips, p->width, p->d_wgt, p->width,
+          p->d_scalers, p->d_diag, p->d_partials, p->d_lnl, 0));
+    } else
       CK(examl_hip_evaluate_root_prot_gamma(
           tc, pNumber, qNumber, x1s, x2s, ts, z, pd->EIGN, pd->gammaRates,
           p->d_tipVector, p->d_clv, p->width * p->span, p->d_tips, p->width,
@@ -768,7 +811,13 @@ void makenewzIterative(tree *tr)
                                       p->d_tipVector, p->d_clv,
                                       p->width * p->span, p->d_tips,
                                       p->width, p->d_sum, p->width, 0));
-    else
+    else if (p->lg4) {
+      upload_lg4(p, pd);
+      CK(examl_hip_sum_root_prot_lg4(tc, x1s, x2s, ts, ts2, p->d_tipVec4,
+                                     p->d_clv, p->width * p->span,
+                                     p->d_tips, p->width, p->d_sum,
+                                     p->width, 0));
+    } else
       CK(examl_hip_sum_root_prot_gamma(tc, x1s, x2s, ts, ts2,
                                        p->d_tipVector, p->d_clv,
                                        p->width * p->span, p->d_tips,
@@ -841,6 +890,11 @@ void execCore(tree *tr, volatile double *_dlnLdlz, volatile double *_d2lnLdlz2)
                                        pd->gammaRates, lz, p->d_wgt,
                                        p->d_dtab, p->d_partials, p->d_out2,
                                        0));
+    else if (p->lg4)
+      CK(examl_hip_core_root_prot_lg4(p->width, p->d_sum, p->h_eign4,
+                                      pd->gammaRates, pd->weights, lz,
+                                      p->d_wgt, p->d_dtab, p->d_partials,
+                                      p->d_out2, 0));
     else
       CK(examl_hip_core_root_prot_gamma(p->width, p->d_sum, pd->EIGN,
                                         pd->gammaRates, lz, p->d_wgt,

@@ -156,3 +156,28 @@ def test_hybrid_two_ranks(tmp_path):
     lnl = float(lines[0].split(":")[1])
     rel = abs(lnl - GOLDEN_F_E) / abs(GOLDEN_F_E)
     assert rel < 1e-6, f"2-rank hybrid lnL {lnl} vs {GOLDEN_F_E} rel {rel}"
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(not os.path.exists(HYBRID),
+                    reason="hybrid binary not built")
+def test_hybrid_lg4(tmp_path):
+    """LG4M + LG4X + WAG partitions (tests/golden/140lg4.binary) through
+    the unmodified reference -f E: the per-gamma-category eigensystem
+    path (makeP_FlexLG4, newviewGenericSpecial.c:170; optLG4X,
+    optimizeModel.c:1116) on the LG4 HIP executors.  Golden: reference
+    examl-AVX -f E final lnL -120844.546570 (tests/test_lg4.py)."""
+    shutil.copy(os.path.join(GOLDEN, "140lg4.binary"), str(tmp_path))
+    shutil.copy(os.path.join(GOLDEN, "140.tree"), str(tmp_path))
+    r = subprocess.run(
+        [HYBRID, "-s", "140lg4.binary", "-t", "140.tree", "-m", "GAMMA",
+         "-f", "E", "-n", "HLG4"], cwd=str(tmp_path), capture_output=True,
+        text=True, timeout=1800)
+    out = r.stdout + r.stderr
+    assert r.returncode == 0, out[-3000:]
+    lines = [ln for ln in out.splitlines() if "Likelihood tree 0" in ln]
+    assert lines, out[-3000:]
+    lnl = float(lines[0].split(":")[1])
+    golden = -120844.546570
+    rel = abs(lnl - golden) / abs(golden)
+    assert rel < 1e-6, f"hybrid LG4 lnL {lnl} vs {golden} rel {rel}"
