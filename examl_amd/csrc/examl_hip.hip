@@ -687,6 +687,65 @@ __global__ __launch_bounds__(NV_BLOCK) void k_newview_prot_gamma(
     return;
   }
 
+  if (TC == EXAML_TIP_INNER) {
+    /* TI measured FASTER on the classic per-(site,cat) lane (130 vs 143
+     * us): the split mapping doubles the per-site ump gathers, which
+     * dominate once the left dot is a table lookup.  Full-span
+     * accumulators, wave ballot over the site's 4 lanes. */
+    const long unitsTI = n * 4;
+    const int laneTI = tid & 63;
+    for (long idx = (long)blockIdx.x * NV_BLOCK + tid; idx < unitsTI;
+         idx += (long)gridDim.x * NV_BLOCK) {
+      const long site = idx >> 2;
+      const int cat = (int)(idx & 3);
+      const int code1 = tipX1[site];
+      double xr[20], acc[20];
+#pragma unroll
+      for (int s = 0; s < 20; s += 4) {
+        const double4 b =
+            *reinterpret_cast<const double4 *>(&x2[idx * 20 + s]);
+        xr[s] = b.x; xr[s + 1] = b.y; xr[s + 2] = b.z; xr[s + 3] = b.w;
+      }
+#pragma unroll
+      for (int s = 0; s < 20; s++) acc[s] = 0.0;
+      for (int l = 0; l < 20; l++) {
+        const double u1 = sU1[80 * code1 + cat * 20 + l];
+        const double u2 = dot20o<FAST>(xr, &sR[cat * CSTR + l * 20]);
+        const double t = u1 * u2;
+#pragma unroll
+        for (int s = 0; s < 20; s++) {
+          if (FAST)
+            acc[s] = fma(t, sEV[l * 20 + s], acc[s]);
+          else
+            acc[s] += t * sEV[l * 20 + s];
+        }
+      }
+      bool small = true;
+#pragma unroll
+      for (int s = 0; s < 20; s++)
+        small &= (fabs(acc[s]) < MINLIKELIHOOD);
+      const unsigned long long m = __ballot(small);
+      if (((m >> (laneTI & ~3)) & 0xFULL) == 0xFULL) {
+#pragma unroll
+        for (int s = 0; s < 20; s++) acc[s] *= TWOTOTHE256;
+        if ((laneTI & 3) == 0)
+          atomicAdd(scalerInc, (unsigned int)wgt[site]);
+      }
+#pragma unroll
+      for (int s = 0; s < 20; s += 4) {
+        const double4 v = make_double4(acc[s], acc[s + 1], acc[s + 2],
+                                       acc[s + 3]);
+        if (NT)
+          __builtin_nontemporal_store(
+              (v4d){v.x, v.y, v.z, v.w},
+              reinterpret_cast<v4d *>(&x3[idx * 20 + s]));
+        else
+          *reinterpret_cast<double4 *>(&x3[idx * 20 + s]) = v;
+      }
+    }
+    return;
+  }
+
   const long units = n * 8; /* (site, cat, half) */
   const int lane = tid & 63;
   for (long idx = (long)blockIdx.x * NV_BLOCK + tid; idx < units;
@@ -750,11 +809,7 @@ __global__ __launch_bounds__(NV_BLOCK) void k_newview_prot_gamma(
         const double other2 = __shfl_xor(mine2, 1);
         u2 = half ? (other2 + mine2) : (mine2 + other2);
         if (TC == EXAML_TIP_INNER) {
-          /* one ump gather per lane PAIR: half 0 reads, half 1 takes it
-           * from the shuffle */
-          const double g = half ? 0.0 : sU1[80 * code1 + cat * 20 + l];
-          const double og = __shfl_xor(g, 1);
-          u1 = half ? og : g;
+          u1 = sU1[80 * code1 + cat * 20 + l];
         } else {
           const double *pl = &sL[cat * CSTR + l * 20 + 2 * half];
           double se = 0, so = 0;
