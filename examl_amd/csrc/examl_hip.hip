@@ -4059,9 +4059,6 @@ struct CSeg { /* one partition's NR-derivative unit */
   int blkBase, nBlocks, part, pad;
 };
 
-__global__ void k_newview_dna_mseg_rt(const MSeg *, const int *,
-                                      const double *);
-
 /* forward declarations of the protein (20-state) mseg kernels defined
  * after the executors */
 template <int TC, bool FAST>
@@ -6166,114 +6163,3 @@ extern "C" int examl_hip_sum_cat_save(
   return 0;
 }
 
-/* runtime-tipCase DNA mseg newview: one launch per LEVEL covering every
- * tipCase's segments (DNA LDS footprint is ~6 KB, so the max-size
- * allocation costs no occupancy; per-block the branch is uniform).
- * Math identical to the templated k_newview_dna_mseg bodies. */
-__global__ __launch_bounds__(NV_BLOCK) void k_newview_dna_mseg_rt(
-    const MSeg *__restrict__ segs, const int *__restrict__ blk2seg,
-    const double *__restrict__ active) {
-  const int si = blk2seg[blockIdx.x];
-  const MSeg sg = segs[si];
-  if (active[sg.part] == 0.0) return;
-  const int TC = sg.pad; /* tipCase, stored by multi_shape_get */
-  __shared__ double sL[64], sR[64], sEV[16], sTV[64];
-  __shared__ double sU1[256], sU2[256];
-
-  const int tid = threadIdx.x;
-  if (tid < 64) {
-    sL[tid] = sg.P[tid];
-    sR[tid] = sg.P[64 + tid];
-    sTV[tid] = sg.tipVec[tid];
-  }
-  if (tid < 16) sEV[tid] = sg.EV[tid];
-  __syncthreads();
-
-  if (TC != EXAML_INNER_INNER) {
-    const int code = tid >> 4, cat = (tid >> 2) & 3, row = tid & 3;
-    const double *tv = &sTV[code * 4];
-    const double *pl = &sL[cat * 16 + row * 4];
-    sU1[tid] =
-        (pl[0] * tv[0] + pl[1] * tv[1]) + (pl[2] * tv[2] + pl[3] * tv[3]);
-    if (TC == EXAML_TIP_TIP) {
-      const double *pr = &sR[cat * 16 + row * 4];
-      sU2[tid] =
-          (pr[0] * tv[0] + pr[1] * tv[1]) + (pr[2] * tv[2] + pr[3] * tv[3]);
-    }
-    __syncthreads();
-  }
-
-  const long units = sg.n * 4;
-  /* per-segment NT only: an aggregate-traffic criterion was tried and
-   * measured SLOWER on config 3 (0.513 -> 0.567 ms/step) — a level's
-   * x3 output is the next level's input, and streaming stores forfeit
-   * the cache hits those reads otherwise get */
-  const bool nt = sg.n >= 65536;
-  const int lane = tid & 63;
-  for (long idx = (long)(blockIdx.x - sg.blkBase) * NV_BLOCK + tid;
-       idx < units; idx += (long)sg.nBlocks * NV_BLOCK) {
-    const long site = idx >> 2;
-    const int cat = (int)(idx & 3);
-    double u1[4], u2[4];
-
-    if (TC == EXAML_INNER_INNER) {
-      const double4 xl = *reinterpret_cast<const double4 *>(&sg.x1[idx * 4]);
-      const double4 xr = *reinterpret_cast<const double4 *>(&sg.x2[idx * 4]);
-#pragma unroll
-      for (int l = 0; l < 4; l++) {
-        const double *pl = &sL[cat * 16 + l * 4];
-        const double *pr = &sR[cat * 16 + l * 4];
-        u1[l] = (xl.x * pl[0] + xl.y * pl[1]) + (xl.z * pl[2] + xl.w * pl[3]);
-        u2[l] = (xr.x * pr[0] + xr.y * pr[1]) + (xr.z * pr[2] + xr.w * pr[3]);
-      }
-    } else if (TC == EXAML_TIP_INNER) {
-      const int code = sg.t1[site];
-      const double4 xr = *reinterpret_cast<const double4 *>(&sg.x2[idx * 4]);
-#pragma unroll
-      for (int l = 0; l < 4; l++) {
-        const double *pr = &sR[cat * 16 + l * 4];
-        u1[l] = sU1[code * 16 + cat * 4 + l];
-        u2[l] = (xr.x * pr[0] + xr.y * pr[1]) + (xr.z * pr[2] + xr.w * pr[3]);
-      }
-    } else {
-      const int c1 = sg.t1[site], c2 = sg.t2[site];
-#pragma unroll
-      for (int l = 0; l < 4; l++) {
-        u1[l] = sU1[c1 * 16 + cat * 4 + l];
-        u2[l] = sU2[c2 * 16 + cat * 4 + l];
-      }
-    }
-
-    double a0 = 0, a1 = 0, a2 = 0, a3 = 0;
-#pragma unroll
-    for (int l = 0; l < 4; l++) {
-      const double t = u1[l] * u2[l];
-      a0 += t * sEV[l * 4 + 0];
-      a1 += t * sEV[l * 4 + 1];
-      a2 += t * sEV[l * 4 + 2];
-      a3 += t * sEV[l * 4 + 3];
-    }
-
-    if (TC != EXAML_TIP_TIP) {
-      const bool small = (fabs(a0) < MINLIKELIHOOD) &
-                         (fabs(a1) < MINLIKELIHOOD) &
-                         (fabs(a2) < MINLIKELIHOOD) &
-                         (fabs(a3) < MINLIKELIHOOD);
-      const unsigned long long m = __ballot(small);
-      if (((m >> (lane & ~3)) & 0xFULL) == 0xFULL) {
-        a0 *= TWOTOTHE256;
-        a1 *= TWOTOTHE256;
-        a2 *= TWOTOTHE256;
-        a3 *= TWOTOTHE256;
-        if ((lane & 3) == 0)
-          atomicAdd(sg.inc, (unsigned int)sg.wgt[site]);
-      }
-    }
-    if (nt)
-      __builtin_nontemporal_store((v4d){a0, a1, a2, a3},
-                                  reinterpret_cast<v4d *>(&sg.x3[idx * 4]));
-    else
-      *reinterpret_cast<double4 *>(&sg.x3[idx * 4]) =
-          make_double4(a0, a1, a2, a3);
-  }
-}
