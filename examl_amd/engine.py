@@ -865,7 +865,6 @@ class MultiEngine:
         return self.evaluate_root(tree, p, q, z, all_reduce=all_reduce)
 
     def sum_root(self, tree, p, q, active=None):
-        e0 = self.engines[0]
         p_tip, q_tip = tree.is_tip(p), tree.is_tip(q)
         if p_tip and q_tip:
             tc, x1s, x2s, t1, t2 = TIP_TIP, -1, -1, p, q
@@ -876,7 +875,6 @@ class MultiEngine:
         else:
             tc, x1s, x2s, t1, t2 = (INNER_INNER, tree.clv_slot(p),
                                     tree.clv_slot(q), -1, -1)
-        del e0
         check(lib().examl_hip_sum_root_multi(
             self.h, tc, x1s, x2s, t1, t2, self._active(active),
             self._stream()), "sum_root_multi")
