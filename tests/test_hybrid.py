@@ -181,3 +181,47 @@ def test_hybrid_lg4(tmp_path):
     golden = -120844.546570
     rel = abs(lnl - golden) / abs(golden)
     assert rel < 1e-6, f"hybrid LG4 lnL {lnl} vs {golden} rel {rel}"
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(not os.path.exists(HYBRID),
+                    reason="hybrid binary not built")
+def test_hybrid_quartets(tmp_path):
+    """-f q -r 30 -p 12345: the reference's quartet machinery
+    (computeQuartets, quartets.c:349) driving the HIP kernels through
+    the boundary on 4-taxon trees — the launch-latency stress shape.
+    The quartet file must match the reference's own line for line
+    (tests/golden/49.quartets.txt)."""
+    shutil.copy(os.path.join(GOLDEN, "49.binary"), str(tmp_path))
+    shutil.copy(os.path.join(GOLDEN, "49.tree"), str(tmp_path))
+    r = subprocess.run(
+        [HYBRID, "-s", "49.binary", "-t", "49.tree", "-m", "GAMMA",
+         "-f", "q", "-r", "30", "-p", "12345", "-n", "HQ"],
+        cwd=str(tmp_path), capture_output=True, text=True, timeout=900)
+    out = r.stdout + r.stderr
+    assert r.returncode == 0, out[-3000:]
+    qf = [f for f in os.listdir(tmp_path) if f.startswith("ExaML_quartets")]
+    assert qf, out[-2000:]
+    got = open(os.path.join(tmp_path, qf[0])).read().strip().splitlines()
+    ref = open(os.path.join(GOLDEN, "49.quartets.txt")
+               ).read().strip().splitlines()
+
+    def parse(lines):
+        # topology triple + lnLs: compare taxa + lnL to 1e-6 rel
+        rows = []
+        for ln in lines:
+            if not ln.strip() or ":" not in ln:
+                continue
+            rows.append(ln.split())
+        return rows
+
+    g, f = parse(got), parse(ref)
+    assert len(g) == len(f), (len(g), len(f))
+    for a, b in zip(g, f):
+        assert len(a) == len(b)
+        for xa, xb in zip(a, b):
+            try:
+                va, vb = float(xa), float(xb)
+                assert abs(va - vb) <= 1e-6 * max(1.0, abs(vb)), (a, b)
+            except ValueError:
+                assert xa == xb, (a, b)
