@@ -573,3 +573,34 @@ def test_prot_cat_save_kernels_bit_exact_vs_reference():
                                     ctypes.c_int(n), dp(None), dp(go),
                                     up(g1), up(g3))
     assert np.array_equal(sr2, so2)
+
+
+def test_save_engine_transparent_with_per_gene_bl():
+    """-S x -M interplay (uncovered combination): SaveDnaEngine under
+    per-partition branch lengths equals the dense oracle engine on the
+    same gappy 2-partition data — newview entries carry per-partition
+    qz/rz (axml.h:434 qz[i]) and evaluate/makenewz use per-partition
+    roots."""
+    from examl_amd.search import TreeSearch
+    from tests.helpers import OracleEngine, OracleSaveEngine
+    tips1, wgt1 = _gappy(10, 240, 0.4, 11)
+    tips2, wgt2 = _gappy(10, 180, 0.3, 12)
+    mk = lambda: [ea.DnaGtrModel([0.3, 0.2, 0.26, 0.24],
+                                 [1.2, 2.4, 0.7, 0.9, 3.1, 1.0], 0.8),
+                  ea.DnaGtrModel([0.22, 0.28, 0.24, 0.26],
+                                 [0.8, 1.4, 1.7, 0.9, 2.1, 1.0], 1.3)]
+    t1 = ea.PhyloTree.random(10, seed=21, rng_z=True)
+    t2 = ea.PhyloTree.random(10, seed=21, rng_z=True)
+    m1, m2 = mk(), mk()
+    ts_dense = TreeSearch(
+        t1, [OracleEngine(tips1, wgt1, m1[0]),
+             OracleEngine(tips2, wgt2, m1[1])], per_gene_bl=True)
+    ts_save = TreeSearch(
+        t2, [OracleSaveEngine(tips1, wgt1, m2[0]),
+             OracleSaveEngine(tips2, wgt2, m2[1])], per_gene_bl=True)
+    l1 = ts_dense.evaluate_generic(full=True)
+    l2 = ts_save.evaluate_generic(full=True)
+    assert l1 == l2, (l1, l2)
+    a1 = ts_dense.tree_evaluate(1.0)
+    a2 = ts_save.tree_evaluate(1.0)
+    assert a1 == a2, (a1, a2)
