@@ -321,10 +321,13 @@ def main(argv=None, device=None):
                        convergence_criterion=opts["D"],
                        save_best_trees=opts["B"], log=log)
         sp.constraint = cv
-        if (all(p.states == 4 for p in parts) and opts["m"] == "GAMMA"
+        if (all(p.states == 4 for p in parts)
+                and opts["m"] in ("GAMMA", "PSR")
                 and not opts["M"] and cv is None):
             # resumable like the reference: one binary checkpoint per
-            # SPR cycle (writeCheckpointInner, searchAlgo.c:1153)
+            # SPR cycle (writeCheckpointInner, searchAlgo.c:1153); PSR
+            # additionally carries rateCategory/patrat + the per-model
+            # category state (searchAlgo.c:1188-1201)
             from examl_amd.checkpoint import (build_model_entry,
                                               write_checkpoint)
             counter = [0]
@@ -332,9 +335,24 @@ def main(argv=None, device=None):
             def _writer(state, fields):
                 path = os.path.join(
                     wdir, f"ExaML_binaryCheckpoint.{name}_{counter[0]}")
+                kw2 = {}
+                if opts["m"] == "PSR":
+                    kw2 = dict(
+                        rate_het="CAT",
+                        invocations=ts.rate_cat_invocations,
+                        rate_category=np.concatenate(
+                            [e.cptr for e in ts.engines]),
+                        patrat=np.concatenate(ts.cat_patrat),
+                    )
+                    entries = [build_model_entry(
+                        e.model, num_cats=e.num_cats,
+                        per_site_rates=e.per_site_rates)
+                        for e in ts.engines]
+                else:
+                    entries = [build_model_entry(e.model)
+                               for e in ts.engines]
                 write_checkpoint(
-                    path, st,
-                    [build_model_entry(e.model) for e in ts.engines],
+                    path, st, entries,
                     len(taxa), state=state, spr=fields,
                     start_number=st.start,
                     likelihood_epsilon=opts["e"],
@@ -344,7 +362,7 @@ def main(argv=None, device=None):
                     initial_set=opts["i"] is not None,
                     initial=10 if opts["i"] is None else opts["i"],
                     tree0=sp.slot_tree_strings[0],
-                    tree1=sp.slot_tree_strings[1])
+                    tree1=sp.slot_tree_strings[1], **kw2)
                 counter[0] += 1
 
             sp.checkpoint_writer = _writer

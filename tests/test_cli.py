@@ -158,3 +158,53 @@ def test_cli_full_flow_cpu(golden_dir, tmp_path, monkeypatch):
                  if ln.startswith("Likelihood of best tree:")]
                 [0].split(":")[1])
     assert abs(lnl - (-3435.016697)) < 1e-2
+
+
+def test_cli_psr_search_writes_checkpoints(golden_dir, tmp_path,
+                                           monkeypatch):
+    """-m PSR -f d now writes per-cycle binary checkpoints like the
+    reference (searchAlgo.c:1188-1201: rateCategory/patrat + per-model
+    category state under CAT); the file parses back with the CAT layout
+    and, when the reference binary is present, the unmodified reference
+    restarts from it."""
+    import subprocess
+
+    import examl_amd.__main__ as cli
+    from examl_amd.checkpoint import read_checkpoint
+    monkeypatch.setattr(cli, "_build_engines", _oracle_build_engines)
+    rc = cli.main(["-s", os.path.join(golden_dir, "12.binary"),
+                   "-t", os.path.join(golden_dir, "12.tree"),
+                   "-n", "P", "-f", "d", "-m", "PSR",
+                   "-w", str(tmp_path)], device="cpu")
+    assert rc == 0
+    cks = sorted(f for f in os.listdir(tmp_path)
+                 if f.startswith("ExaML_binaryCheckpoint.P_"))
+    assert cks, os.listdir(tmp_path)
+    last = os.path.join(tmp_path, cks[-1])
+    from examl_amd.examl_io import read_byte_file
+    _, parts = read_byte_file(os.path.join(golden_dir, "12.binary"))
+    width_total = sum(p.upper - p.lower for p in parts)
+    ck = read_checkpoint(last, 12, [4] * len(parts), rate_het="CAT",
+                         crunched_length=width_total)
+    assert ck.rate_category.size == ck.patrat.size
+    assert ck.rate_category.size == width_total
+    assert all(m["num_cats"] >= 1 for m in ck.models)
+    # cmd block says CAT (rateHetModel, axml.h:672)
+    import struct as _s
+    d = open(last, "rb").read()
+    assert _s.unpack_from("<i", d, 1248 + 52)[0] == 0  # CAT
+    ref_bin = os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "oracle", "_ref", "examl-AVX")
+    if os.path.exists(ref_bin):
+        import shutil as _sh
+        _sh.copy(os.path.join(golden_dir, "12.binary"),
+                 str(tmp_path / "r.binary"))
+        _sh.copy(os.path.join(golden_dir, "12.tree"),
+                 str(tmp_path / "12.tree"))
+        r = subprocess.run(
+            [ref_bin, "-s", "r.binary", "-t", "12.tree", "-m", "PSR",
+             "-f", "d", "-R", cks[-1], "-n", "RP"], cwd=str(tmp_path),
+            capture_output=True, text=True, timeout=600)
+        out = r.stdout + r.stderr
+        assert "Restart with likelihood" in out, out[-2000:]
+        assert r.returncode == 0, out[-2000:]
