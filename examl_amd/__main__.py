@@ -323,7 +323,7 @@ def main(argv=None, device=None):
         sp.constraint = cv
         if (all(p.states == 4 for p in parts)
                 and opts["m"] in ("GAMMA", "PSR")
-                and not opts["M"] and cv is None):
+                and cv is None):
             # resumable like the reference: one binary checkpoint per
             # SPR cycle (writeCheckpointInner, searchAlgo.c:1153); PSR
             # additionally carries rateCategory/patrat + the per-model
@@ -354,7 +354,7 @@ def main(argv=None, device=None):
                 write_checkpoint(
                     path, st, entries,
                     len(taxa), state=state, spr=fields,
-                    start_number=st.start,
+                    start_number=st.start, per_gene_bl=opts["M"],
                     likelihood_epsilon=opts["e"],
                     use_median=opts["a"], save_best_trees=opts["B"],
                     save_memory=opts["S"], search_convergence=opts["D"],

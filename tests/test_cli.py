@@ -208,3 +208,41 @@ def test_cli_psr_search_writes_checkpoints(golden_dir, tmp_path,
         out = r.stdout + r.stderr
         assert "Restart with likelihood" in out, out[-2000:]
         assert r.returncode == 0, out[-2000:]
+
+
+def test_cli_m_search_writes_checkpoints(golden_dir, tmp_path,
+                                         monkeypatch):
+    """-M -f d writes checkpoints with per-partition branch vectors in
+    the node image (writeTree z[numBranches]); the reference restarts
+    from them when present."""
+    import subprocess
+
+    import examl_amd.__main__ as cli
+    monkeypatch.setattr(cli, "_build_engines", _oracle_build_engines)
+    rc = cli.main(["-s", os.path.join(golden_dir, "12m.binary"),
+                   "-t", os.path.join(golden_dir, "12.tree"),
+                   "-n", "M", "-f", "d", "-M",
+                   "-w", str(tmp_path)], device="cpu")
+    assert rc == 0
+    cks = sorted(f for f in os.listdir(tmp_path)
+                 if f.startswith("ExaML_binaryCheckpoint.M_"))
+    assert cks, os.listdir(tmp_path)
+    d = open(os.path.join(tmp_path, cks[-1]), "rb").read()
+    import struct as _s
+    assert _s.unpack_from("<i", d, 1248 + 16)[0] == 1  # perGeneBL
+    ref_bin = os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "oracle", "_ref", "examl-AVX")
+    if os.path.exists(ref_bin):
+        import shutil as _sh
+        _sh.copy(os.path.join(golden_dir, "12m.binary"),
+                 str(tmp_path / "r.binary"))
+        _sh.copy(os.path.join(golden_dir, "12.tree"),
+                 str(tmp_path / "12.tree"))
+        r = subprocess.run(
+            [ref_bin, "-s", "r.binary", "-t", "12.tree", "-M",
+             "-m", "GAMMA", "-f", "d", "-R", cks[-1], "-n", "RM"],
+            cwd=str(tmp_path),
+            capture_output=True, text=True, timeout=600)
+        out = r.stdout + r.stderr
+        assert "Restart with likelihood" in out, out[-2000:]
+        assert r.returncode == 0, out[-2000:]
