@@ -246,3 +246,48 @@ def test_cli_m_search_writes_checkpoints(golden_dir, tmp_path,
         out = r.stdout + r.stderr
         assert "Restart with likelihood" in out, out[-2000:]
         assert r.returncode == 0, out[-2000:]
+
+
+def test_cli_d_checkpoint_carries_tree_strings(golden_dir, tmp_path,
+                                               monkeypatch):
+    """-f d -D checkpoints carry the RF-convergence tree0/tree1 topology
+    strings (searchAlgo.c:2178-2185), so a reference -D restart can
+    re-populate its hash table (readCheckpoint:1545-1580)."""
+    import subprocess
+
+    import examl_amd.__main__ as cli
+    monkeypatch.setattr(cli, "_build_engines", _oracle_build_engines)
+    rc = cli.main(["-s", os.path.join(golden_dir, "12.binary"),
+                   "-t", os.path.join(golden_dir, "12.tree"),
+                   "-n", "DD", "-f", "d", "-D",
+                   "-w", str(tmp_path)], device="cpu")
+    assert rc == 0
+    cks = sorted(f for f in os.listdir(tmp_path)
+                 if f.startswith("ExaML_binaryCheckpoint.DD_"))
+    assert cks
+    # a FAST_SPRS checkpoint written after the first store has tree0
+    from examl_amd.checkpoint import read_checkpoint
+    got_string = False
+    for f in cks:
+        ck = read_checkpoint(os.path.join(tmp_path, f), 12, [4])
+        t0 = ck.tree0.split(b"\0", 1)[0]
+        if t0.startswith(b"(") and b"T" in t0:
+            got_string = True
+            break
+    assert got_string, "no checkpoint carried a tree0 topology string"
+    ref_bin = os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "oracle", "_ref", "examl-AVX")
+    if os.path.exists(ref_bin):
+        import shutil as _sh
+        _sh.copy(os.path.join(golden_dir, "12.binary"),
+                 str(tmp_path / "r.binary"))
+        _sh.copy(os.path.join(golden_dir, "12.tree"),
+                 str(tmp_path / "12.tree"))
+        r = subprocess.run(
+            [ref_bin, "-s", "r.binary", "-t", "12.tree", "-D",
+             "-m", "GAMMA", "-f", "d", "-R", cks[-1], "-n", "RD"],
+            cwd=str(tmp_path), capture_output=True, text=True,
+            timeout=600)
+        out = r.stdout + r.stderr
+        assert "Restart with likelihood" in out, out[-2000:]
+        assert r.returncode == 0, out[-2000:]
