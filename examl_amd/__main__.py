@@ -207,22 +207,22 @@ def main(argv=None, device=None):
                                [p.states for p in parts],
                                rate_het="CAT" if psr else "GAMMA",
                                crunched_length=cl if psr else None)
-        engines = []
-        for p, m in zip(parts, ckpt.models):
+        # engines via the (monkeypatchable) factory, then install the
+        # checkpoint's model/rate state in place (readCheckpoint's
+        # restore, searchAlgo.c:1502+)
+        engines, _af, _ef = _build_engines(parts, opts, device)
+        for p, m, eng in zip(parts, ckpt.models, engines):
             assert p.states == 4, "-R restart wired for DNA"
             model = ea.DnaGtrModel(m["frequencies"], m["substRates"],
                                    m["alpha"], use_median=opts["a"])
+            eng.model = model
+            eng.upload_model()
             if psr:
                 cptr = np.asarray(ckpt.rate_category[p.lower:p.upper],
                                   dtype=np.int32).copy()
                 rates = np.asarray(
                     m["per_site_rates"][:m["num_cats"]]).copy()
-                engines.append(ea.DnaCatEngine(p.tips, p.wgt, model,
-                                               cptr, rates,
-                                               device=device))
-            else:
-                engines.append(ea.DnaGammaEngine(p.tips, p.wgt, model,
-                                                 device=device))
+                eng.set_site_rates(cptr, rates)
         kw = dict(opt_freq_flags=[bool(p.optimizeBaseFrequencies)
                                   for p in parts],
                   max_categories=opts["c"])

@@ -291,3 +291,35 @@ def test_cli_d_checkpoint_carries_tree_strings(golden_dir, tmp_path,
         out = r.stdout + r.stderr
         assert "Restart with likelihood" in out, out[-2000:]
         assert r.returncode == 0, out[-2000:]
+
+
+def test_cli_psr_resume_from_own_checkpoint(golden_dir, tmp_path,
+                                            monkeypatch):
+    """Our own -R restart from a checkpoint WE wrote mid-PSR-search lands
+    on the same final lnL as the uninterrupted run (restart determinism,
+    searchAlgo.c:1726)."""
+    import examl_amd.__main__ as cli
+    monkeypatch.setattr(cli, "_build_engines", _oracle_build_engines)
+    rc = cli.main(["-s", os.path.join(golden_dir, "12.binary"),
+                   "-t", os.path.join(golden_dir, "12.tree"),
+                   "-n", "PA", "-f", "d", "-m", "PSR",
+                   "-w", str(tmp_path)], device="cpu")
+    assert rc == 0
+    info = open(os.path.join(tmp_path, "ExaML_info.PA")).read()
+    final_a = float([ln for ln in info.splitlines()
+                     if ln.startswith("Likelihood of best tree:")]
+                    [0].split(":")[1])
+    cks = sorted(f for f in os.listdir(tmp_path)
+                 if f.startswith("ExaML_binaryCheckpoint.PA_"))
+    assert len(cks) >= 2
+    mid = os.path.join(tmp_path, cks[len(cks) // 2])
+    rc = cli.main(["-s", os.path.join(golden_dir, "12.binary"),
+                   "-n", "PB", "-f", "d", "-m", "PSR", "-R", mid,
+                   "-w", str(tmp_path)], device="cpu")
+    assert rc == 0
+    info = open(os.path.join(tmp_path, "ExaML_info.PB")).read()
+    final_b = float([ln for ln in info.splitlines()
+                     if ln.startswith("Likelihood of best tree:")]
+                    [0].split(":")[1])
+    assert abs(final_a - final_b) <= 1e-6 * abs(final_a), \
+        (final_a, final_b)
