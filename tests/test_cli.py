@@ -323,3 +323,35 @@ def test_cli_psr_resume_from_own_checkpoint(golden_dir, tmp_path,
                     [0].split(":")[1])
     assert abs(final_a - final_b) <= 1e-6 * abs(final_a), \
         (final_a, final_b)
+
+
+def test_cli_m_resume_from_own_checkpoint(golden_dir, tmp_path,
+                                          monkeypatch):
+    """-M -f d: our own -R restart from our mid-search checkpoint lands
+    on the uninterrupted run's final lnL (per-partition branch vectors
+    restored from the node image)."""
+    import examl_amd.__main__ as cli
+    monkeypatch.setattr(cli, "_build_engines", _oracle_build_engines)
+    rc = cli.main(["-s", os.path.join(golden_dir, "12m.binary"),
+                   "-t", os.path.join(golden_dir, "12.tree"),
+                   "-n", "MA", "-f", "d", "-M",
+                   "-w", str(tmp_path)], device="cpu")
+    assert rc == 0
+    info = open(os.path.join(tmp_path, "ExaML_info.MA")).read()
+    final_a = float([ln for ln in info.splitlines()
+                     if ln.startswith("Likelihood of best tree:")]
+                    [0].split(":")[1])
+    cks = sorted(f for f in os.listdir(tmp_path)
+                 if f.startswith("ExaML_binaryCheckpoint.MA_"))
+    assert len(cks) >= 2
+    mid = os.path.join(tmp_path, cks[len(cks) // 2])
+    rc = cli.main(["-s", os.path.join(golden_dir, "12m.binary"),
+                   "-n", "MB", "-f", "d", "-M", "-R", mid,
+                   "-w", str(tmp_path)], device="cpu")
+    assert rc == 0
+    info = open(os.path.join(tmp_path, "ExaML_info.MB")).read()
+    final_b = float([ln for ln in info.splitlines()
+                     if ln.startswith("Likelihood of best tree:")]
+                    [0].split(":")[1])
+    assert abs(final_a - final_b) <= 1e-6 * abs(final_a), \
+        (final_a, final_b)
